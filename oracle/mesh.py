@@ -95,4 +95,5 @@ def r2c(real, geom=None):
 def c2r(cplx, geom):
     """Unnormalized inverse (so r2c . c2r == identity)."""
     shape = tuple(int(n) for n in geom.Nmesh)
-    return numpy.fft.irfftn(cplx, s=shape) * float(numpy.prod(shape))
+    return numpy.fft.irfftn(cplx, s=shape, axes=(0, 1, 2)) \
+        * float(numpy.prod(shape))

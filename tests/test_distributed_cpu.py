@@ -1,0 +1,122 @@
+"""
+Multi-process CPU coverage of the distributed layer: a real world_size-2
+gloo process group (torch.distributed over 127.0.0.1) exercising Comm
+collectives, FrontPadArray, and the MPIRandomState rank invariance the
+reference pins in nbodykit/tests/test_mpirng.py:12-89.
+"""
+import os
+import pickle
+
+import numpy
+import pytest
+import torch.multiprocessing as mp
+
+WORLD = 2
+
+
+def _worker(rank, world_size, port, fn_name, out_q):
+    import torch.distributed as dist
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    dist.init_process_group('gloo', rank=rank, world_size=world_size)
+    try:
+        from nbodykit_amd.comm import TorchComm
+        comm = TorchComm()
+        result = globals()[fn_name](comm)
+        out_q.put((rank, pickle.dumps(result)))
+    finally:
+        dist.destroy_process_group()
+
+
+def _run_world(fn_name):
+    ctx = mp.get_context('spawn')
+    out_q = ctx.Queue()
+    port = numpy.random.randint(20000, 40000)
+    procs = [ctx.Process(target=_worker, args=(r, WORLD, port, fn_name, out_q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, blob = out_q.get(timeout=180)
+        results[rank] = pickle.loads(blob)
+    for p in procs:
+        p.join(timeout=60)
+    return [results[r] for r in range(WORLD)]
+
+
+# ---- per-rank bodies (run inside workers) -------------------------------
+
+def _body_collectives(comm):
+    gathered = comm.allgather(comm.rank * 10)
+    total = comm.allreduce(comm.rank + 1)
+    arr = comm.allreduce(numpy.arange(4, dtype='f8') * (comm.rank + 1))
+    lo = comm.allreduce(float(comm.rank), op='min')
+    hi = comm.allreduce(float(comm.rank), op='max')
+    b = comm.bcast({'x': comm.rank}, root=1)
+    a2a = comm.alltoall([(comm.rank, dest) for dest in range(comm.size)])
+    return dict(gathered=gathered, total=total, arr=arr, lo=lo, hi=hi,
+                bcast=b, a2a=a2a)
+
+
+def _body_rng(comm):
+    from nbodykit_amd.mpirng import MPIRandomState
+    out = {}
+    for size, chunk in [((1, 10)), (10, 3)]:
+        sizes = [size] * comm.size
+        rng = MPIRandomState(comm, seed=1234, size=sizes[comm.rank],
+                             chunksize=chunk)
+        out[(size, chunk)] = (rng.uniform(), rng.csize)
+    # array-arg + itemshape variant (test_mpirng.py:64-76)
+    rng = MPIRandomState(comm, seed=1234, size=10, chunksize=3)
+    out['args'] = rng.uniform(low=numpy.ones(10)[:, None] * 0.5,
+                              itemshape=(3,))
+    return out
+
+
+def _body_frontpad(comm):
+    from nbodykit_amd.utils import FrontPadArray
+    arr = numpy.arange(5, dtype='f8') + 100 * comm.rank
+    padded = FrontPadArray(arr, 3 if comm.rank > 0 else 0, comm)
+    return padded
+
+
+# ---- tests --------------------------------------------------------------
+
+@pytest.mark.timeout(300)
+def test_gloo_collectives():
+    r0, r1 = _run_world('_body_collectives')
+    assert r0['gathered'] == [0, 10] and r1['gathered'] == [0, 10]
+    assert r0['total'] == 3
+    numpy.testing.assert_allclose(r0['arr'], numpy.arange(4) * 3.0)
+    assert r0['lo'] == 0.0 and r0['hi'] == 1.0
+    assert r0['bcast'] == {'x': 1}
+    assert r0['a2a'] == [(0, 0), (1, 0)]
+    assert r1['a2a'] == [(0, 1), (1, 1)]
+
+
+@pytest.mark.timeout(300)
+def test_gloo_rng_rank_invariance():
+    from nbodykit_amd.comm import SerialComm
+    from nbodykit_amd.mpirng import MPIRandomState
+    r0, r1 = _run_world('_body_rng')
+    for key in [(1, 10), (10, 3)]:
+        size, chunk = key
+        csize = r0[key][1]
+        serial = MPIRandomState(SerialComm(), seed=1234, size=csize,
+                                chunksize=chunk).uniform()
+        got = numpy.concatenate([r0[key][0], r1[key][0]])
+        numpy.testing.assert_array_equal(got, serial)
+    serial = MPIRandomState(SerialComm(), seed=1234, size=20,
+                            chunksize=3).uniform(
+        low=numpy.ones(20)[:, None] * 0.5, itemshape=(3,))
+    got = numpy.concatenate([r0['args'], r1['args']])
+    numpy.testing.assert_array_equal(got, serial)
+
+
+@pytest.mark.timeout(300)
+def test_gloo_frontpad():
+    r0, r1 = _run_world('_body_frontpad')
+    numpy.testing.assert_array_equal(r0, numpy.arange(5, dtype='f8'))
+    numpy.testing.assert_array_equal(
+        r1, numpy.concatenate([[2., 3., 4.], numpy.arange(5) + 100.]))
