@@ -1,0 +1,179 @@
+/*
+ * nbk_hip.h — C ABI of libnbk_hip.so, the hand-written HIP (gfx950)
+ * compute core of nbodykit_amd.
+ *
+ * This is the internal drop-in boundary defined in SURVEY.md §8(b): the
+ * reference (bccp/nbodykit) has no C plugin interface — its operator API
+ * is the MeshSource/CatalogSource Python duck type — so this ABI carries
+ * exactly the pmesh/pfft operations the Python layer replaces:
+ *
+ *   nbk_paint_f64        <- pmesh ParticleMesh.paint (CIC/TSC/PCS scatter;
+ *                           called at nbodykit/source/mesh/catalog.py:287,
+ *                           295-296)
+ *   nbk_fft_*            <- pfft-python distributed R2C/C2R
+ *                           (pmesh RealField.r2c/ComplexField.c2r; called
+ *                           via nbodykit/base/mesh.py:301-304), forward
+ *                           normalized by 1/Ntotal
+ *   nbk_compensate_f64   <- the Fourier-space window compensations
+ *                           (nbodykit/source/mesh/catalog.py:419-594)
+ *   nbk_interlace_combine_f64
+ *                        <- the interlaced-mesh combine
+ *                           c = c1/2 + c2/2 exp(i k.H/2)
+ *                           (nbodykit/source/mesh/catalog.py:341-347)
+ *   nbk_power3d_f64      <- p3d = c1 conj(c2) * V, zero mode cleared
+ *                           (nbodykit/algorithms/fftpower.py:114-128)
+ *   nbk_bin_power_f64    <- project_to_basis binning sums
+ *                           (nbodykit/algorithms/fftpower.py:507-701)
+ *   nbk_readout_nnb_f64 / elementwise helpers — support ops.
+ *
+ * Conventions:
+ *   - plain pointers + sizes only; all device pointers are HIP device
+ *     memory owned by the caller (torch allocations in practice, but no
+ *     torch types cross this boundary);
+ *   - `stream` is a hipStream_t passed as void* (0 = default stream);
+ *     all launches are asynchronous on that stream;
+ *   - return value: 0 on success, a negative NBK_ERR_* code otherwise;
+ *     nbk_last_error_string() describes the most recent failure;
+ *   - complex double arrays are interleaved (re,im) — C `double _Complex`
+ *     layout, i.e. numpy complex128 / torch complex128;
+ *   - the mesh is slab-partitioned along axis 0; every kernel taking
+ *     `x0/nx_local` operates on the local slab of a global
+ *     (n0, n1, n2) mesh.  Single-GPU: x0 = 0, nx_local = n0.
+ */
+#ifndef NBK_HIP_H
+#define NBK_HIP_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* error codes */
+#define NBK_OK            0
+#define NBK_ERR_HIP      -1   /* HIP runtime error, see last_error_string */
+#define NBK_ERR_ARG      -2   /* bad argument (size, window id, ...) */
+#define NBK_ERR_UNSUPPORTED -3 /* e.g. non-power-of-two FFT length */
+
+/* window ids (resampler names in the Python layer) */
+#define NBK_WINDOW_CIC 0   /* support 2 */
+#define NBK_WINDOW_TSC 1   /* support 3 */
+#define NBK_WINDOW_PCS 2   /* support 4 */
+
+/* library info ------------------------------------------------------- */
+const char* nbk_version(void);
+const char* nbk_last_error_string(void);
+/* number of HIP devices visible (also a cheap "runtime works" probe) */
+int nbk_device_count(void);
+
+/* paint --------------------------------------------------------------
+ * Scatter n particles into the local mesh slab with window `window` and
+ * half-cell interlacing shift `shift` (0.0 or 0.5, in mesh units;
+ * pm.affine.shift(0.5) at source/mesh/catalog.py:292).
+ *
+ *   pos     : device, SoA layout, 3 contiguous blocks of n doubles
+ *             (x[n], y[n], z[n]) — pos + 0, pos + n, pos + 2n
+ *   mass    : device, n doubles, or NULL for unit mass
+ *   nmesh   : global mesh size (n0, n1, n2)
+ *   box     : BoxSize per axis
+ *   x0, nx_local : local slab [x0, x0+nx_local) of axis 0; deposits
+ *             whose wrapped global x index falls outside are skipped
+ *             (ghost copies on the neighbour rank own them)
+ *   mesh    : device, local slab, nx_local*n1*n2 doubles, accumulated
+ *             into (hold=True semantics)
+ */
+int nbk_paint_f64(const double* pos, const double* mass, int64_t n,
+                  const int64_t nmesh[3], const double box[3],
+                  int window, double shift,
+                  double* mesh, int64_t x0, int64_t nx_local,
+                  void* stream);
+
+/* readout (gather dual of paint; resampler='nnb' serves the LogNormal
+ * generator, mockmaker.py:317-319; cic for FFTRecon later) */
+int nbk_readout_nnb_f64(const double* pos, int64_t n,
+                        const int64_t nmesh[3], const double box[3],
+                        const double* mesh, int64_t x0, int64_t nx_local,
+                        double* out, void* stream);
+
+/* FFT ----------------------------------------------------------------
+ * Power-of-two lengths only (8 <= N <= 4096).  The 3D transform is
+ * composed by the Python layer from these batched passes (plus the RCCL
+ * alltoall pencil transpose between ranks):
+ *
+ *   z pass  : real <-> half-complex along the contiguous last axis
+ *   strided : in-place complex pass along a strided axis (y or x)
+ *
+ * Layouts: nbk_fft_r2c_z treats `real` as nlines contiguous lines of nz
+ * doubles, writing nlines * (nz/2+1) complex outputs; `scale` multiplies
+ * every output (the Python layer passes 1/(n0*n1*n2) here so the forward
+ * 3D transform matches pmesh's normalization).  nbk_fft_c2r_z is the
+ * unnormalized inverse.  nbk_fft_c_strided transforms n_lines lines of
+ * length nfft; element j of line (o, i) lives at
+ * cplx[o*outer_stride + j*stride + i] with i < n_inner contiguous.
+ * sign: -1 forward, +1 inverse (unnormalized).
+ */
+int nbk_fft_r2c_z(const double* real, double* cplx,
+                  int64_t nlines, int64_t nz, double scale, void* stream);
+int nbk_fft_c2r_z(const double* cplx, double* real,
+                  int64_t nlines, int64_t nz, void* stream);
+int nbk_fft_c_strided(double* cplx, int64_t nfft, int64_t stride,
+                      int64_t n_outer, int64_t outer_stride,
+                      int64_t n_inner, int sign, void* stream);
+
+/* k-space elementwise ------------------------------------------------
+ * All of these address a local complex slab of logical global shape
+ * (n0, n1, n2h); `axis_map` gives, for each local axis, which global
+ * axis it carries (identity (0,1,2) untransposed; (1,0,2) for the
+ * post-alltoall transposed layout), and `off` the global start of the
+ * local block along each LOCAL axis.  dims = local shape.
+ */
+int nbk_compensate_f64(double* cplx, const int64_t nmesh[3],
+                       const int64_t dims[3], const int64_t off[3],
+                       const int axis_map[3],
+                       int window, int interlaced, void* stream);
+
+int nbk_interlace_combine_f64(double* c1, const double* c2,
+                              const int64_t nmesh[3], const double box[3],
+                              const int64_t dims[3], const int64_t off[3],
+                              const int axis_map[3], void* stream);
+
+int nbk_power3d_f64(double* out, const double* c1, const double* c2,
+                    double volume,
+                    const int64_t dims[3], const int64_t off[3],
+                    int clear_zero_mode, void* stream);
+
+/* binning ------------------------------------------------------------
+ * project_to_basis sums (fftpower.py:595-666) over the local slab:
+ *   xsum, musum            : (Nx+2)*(Nmu+2) doubles
+ *   Nsum                   : (Nx+2)*(Nmu+2) doubles (integral values)
+ *   ysum                   : nell * (Nx+2)*(Nmu+2) complex doubles
+ * kedges (device, nx_edges doubles) are digitized by binary search on
+ * k^2; muedges (device, nmu_edges doubles) likewise on mu; Hermitian
+ * double-count weights along global axis 2 (weight 2 where kz > 0,
+ * meshtools.py:188-215).  `ells` lists the multipole orders (first
+ * entry must be 0, mirroring fftpower.py:585).  los is the
+ * line-of-sight unit vector.  Outputs are accumulated (caller zeroes).
+ */
+int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
+                      const double box[3],
+                      const int64_t dims[3], const int64_t off[3],
+                      const int axis_map[3],
+                      const double* kedges, int64_t nx_edges,
+                      const double* muedges, int64_t nmu_edges,
+                      const double los[3],
+                      const int* ells, int nell,
+                      double* xsum, double* musum, double* Nsum,
+                      double* ysum, void* stream);
+
+/* small helpers ------------------------------------------------------ */
+/* out[i] += a[i]  (f64, n elements) */
+int nbk_axpy_f64(double* out, const double* a, double alpha, int64_t n,
+                 void* stream);
+/* mesh *= alpha  or mesh[:] = alpha when set_value != 0 */
+int nbk_scale_f64(double* mesh, double alpha, int set_value, int64_t n,
+                  void* stream);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* NBK_HIP_H */

@@ -1,0 +1,357 @@
+"""
+FFTPower — the headline algorithm (reference
+nbodykit/algorithms/fftpower.py): auto/cross P(k) / P(k,mu) / P_ell(k)
+of periodic-box sources.  Same construction flow as the reference
+(:194-334): cast sources to meshes with dtype='f8', compensated=True
+(:703-730), compute the compensated complex fields on the GPU, form
+p3d = c1 conj(c2) V with the zero mode cleared (:91-143, the
+``nbk_power3d_f64`` kernel), then project to (k, mu) / multipole bins
+(:507-701) with the ``nbk_bin_power_f64`` kernel + a tiny host/allreduce
+tail, and pack BinnedStatistic results.
+"""
+import logging
+
+import numpy
+
+from nbodykit_amd import CurrentMPIComm, hiplib
+from nbodykit_amd.base.catalog import CatalogSourceBase
+from nbodykit_amd.base.mesh import MeshSource
+from nbodykit_amd.binned_statistic import BinnedStatistic
+from nbodykit_amd.pm import ComplexField, RealField
+
+
+class FFTBase(object):
+    """Shared setup for periodic-box FFT algorithms (reference :12-143)."""
+
+    def __init__(self, first, second, Nmesh, BoxSize):
+        first = _cast_source(first, Nmesh=Nmesh, BoxSize=BoxSize)
+        if second is not None:
+            second = _cast_source(second, Nmesh=Nmesh, BoxSize=BoxSize)
+        else:
+            second = first
+
+        self.first = first
+        self.second = second
+        self.comm = first.comm
+        assert second.comm is first.comm, \
+            "communicator mismatch between input sources"
+
+        if not numpy.array_equal(first.attrs['BoxSize'],
+                                 second.attrs['BoxSize']):
+            raise ValueError("'BoxSize' mismatch between sources in "
+                             "FFTPower")
+
+        self.attrs = {}
+        self.attrs['Nmesh'] = first.attrs['Nmesh'].copy()
+        self.attrs['BoxSize'] = first.attrs['BoxSize'].copy()
+        self.attrs.update(zip(['Lx', 'Ly', 'Lz'], self.attrs['BoxSize']))
+        self.attrs.update({'volume': self.attrs['BoxSize'].prod()})
+
+    def save(self, output):
+        import json
+        from nbodykit_amd.utils import JSONEncoder
+        if self.comm.rank == 0:
+            self.logger.info('measurement done; saving result to %s'
+                             % output)
+            with open(output, 'w') as ff:
+                json.dump(self.__getstate__(), ff, cls=JSONEncoder)
+
+    @classmethod
+    @CurrentMPIComm.enable
+    def load(cls, output, comm=None):
+        import json
+        from nbodykit_amd.utils import JSONDecoder
+        if comm.rank == 0:
+            with open(output, 'r') as ff:
+                state = json.load(ff, cls=JSONDecoder)
+        else:
+            state = None
+        state = comm.bcast(state)
+        self = object.__new__(cls)
+        self.__setstate__(state)
+        self.comm = comm
+        return self
+
+    def _compute_3d_power(self, first, second):
+        """p3d = c1 conj(c2), zero mode cleared (but still binned), x V
+        (reference :91-143)."""
+        attrs = {}
+        attrs.update(self.attrs)
+
+        c1 = first.compute(mode='complex', Nmesh=self.attrs['Nmesh'])
+        if first is second:
+            c2 = c1
+        else:
+            c2 = second.compute(mode='complex', Nmesh=self.attrs['Nmesh'])
+
+        lib = hiplib.require()
+        p3d = ComplexField(c1.pm)
+        hiplib.check(lib.nbk_power3d_f64(
+            hiplib.dptr(p3d.value), hiplib.dptr(c1.value),
+            hiplib.dptr(c2.value), float(self.attrs['BoxSize'].prod()),
+            hiplib.i64_arr(c1.dims), hiplib.i64_arr(c1.off), 1,
+            hiplib.cur_stream()), 'nbk_power3d_f64')
+
+        N1 = c1.attrs.get('N', 0)
+        N2 = c2.attrs.get('N', 0)
+        attrs.update({'N1': N1, 'N2': N2})
+
+        Pshot = 0
+        if self.first is self.second:
+            if 'shotnoise' in c1.attrs:
+                Pshot = c1.attrs['shotnoise']
+        attrs['shotnoise'] = Pshot
+        return p3d, attrs
+
+
+class FFTPower(FFTBase):
+    """Periodic-box 1d/2d power spectrum and multipoles via FFT
+    (reference :146-359; same signature and defaults)."""
+    logger = logging.getLogger('FFTPower')
+
+    def __init__(self, first, mode, Nmesh=None, BoxSize=None, second=None,
+                 los=[0, 0, 1], Nmu=5, dk=None, kmin=0., kmax=None,
+                 poles=[]):
+        if mode not in ['1d', '2d']:
+            raise ValueError("`mode` should be either '1d' or '2d'")
+        if poles is None:
+            poles = []
+
+        if numpy.isscalar(los) or len(los) != 3:
+            raise ValueError("line-of-sight ``los`` should be vector with "
+                             "length 3")
+        if not numpy.allclose(numpy.einsum('i,i', los, los), 1.0,
+                              rtol=1e-5):
+            raise ValueError("line-of-sight ``los`` must be a unit vector")
+
+        FFTBase.__init__(self, first, second, Nmesh, BoxSize)
+
+        self.attrs['mode'] = mode
+        self.attrs['los'] = los
+        self.attrs['Nmu'] = Nmu
+        self.attrs['poles'] = poles
+
+        if dk is None:
+            dk = 2 * numpy.pi / self.attrs['BoxSize'].min()
+        self.attrs['dk'] = dk
+        self.attrs['kmin'] = kmin
+        self.attrs['kmax'] = kmax
+
+        self.power, self.poles = self.run()
+        self.attrs.update(self.power.attrs)
+
+    def run(self):
+        if self.attrs['mode'] == '1d':
+            self.attrs['Nmu'] = 1
+
+        y3d, attrs = self._compute_3d_power(self.first, self.second)
+
+        dk = self.attrs['dk']
+        kmin = self.attrs['kmin']
+        kmax = self.attrs['kmax']
+        if kmax is None:
+            kmax = numpy.pi * y3d.Nmesh.min() / y3d.BoxSize.max() + dk / 2
+
+        if dk > 0:
+            kedges = numpy.arange(kmin, kmax, dk)
+            kcoords = None
+        else:
+            kedges, kcoords = _find_unique_edges(
+                y3d.x, 2 * numpy.pi / y3d.BoxSize, kmax, y3d.pm.comm)
+
+        muedges = numpy.linspace(-1, 1, self.attrs['Nmu'] + 1,
+                                 endpoint=True)
+        edges = [kedges, muedges]
+        coords = [kcoords, None]
+        result, pole_result = project_to_basis(y3d, edges,
+                                               poles=self.attrs['poles'],
+                                               los=self.attrs['los'])
+
+        # pack the structured arrays (reference :306-334)
+        if self.attrs['mode'] == '1d':
+            cols = ['k', 'power', 'modes']
+            icols = [0, 2, 3]
+            edges = edges[0:1]
+            coords = coords[0:1]
+        else:
+            cols = ['k', 'mu', 'power', 'modes']
+            icols = [0, 1, 2, 3]
+
+        dtype = numpy.dtype([(name, result[icol].dtype.str)
+                             for icol, name in zip(icols, cols)])
+        power = numpy.squeeze(numpy.empty(result[0].shape, dtype=dtype))
+        for icol, col in zip(icols, cols):
+            power[col][:] = numpy.squeeze(result[icol])
+
+        poles = None
+        if pole_result is not None:
+            k, poles_arr, N = pole_result
+            cols = ['k'] + ['power_%d' % l for l in self.attrs['poles']] \
+                + ['modes']
+            result = [k] + [pole for pole in poles_arr] + [N]
+            dtype = numpy.dtype([(name, result[icol].dtype.str)
+                                 for icol, name in enumerate(cols)])
+            poles = numpy.empty(result[0].shape, dtype=dtype)
+            for icol, col in enumerate(cols):
+                poles[col][:] = result[icol]
+
+        return self._make_datasets(edges, poles, power, coords, attrs)
+
+    def __getstate__(self):
+        return dict(power=self.power.__getstate__(),
+                    poles=(self.poles.__getstate__()
+                           if self.poles is not None else None),
+                    attrs=self.attrs)
+
+    def __setstate__(self, state):
+        self.attrs = state['attrs']
+        self.power = BinnedStatistic.from_state(state['power'])
+        self.poles = None
+        if state['poles'] is not None:
+            self.poles = BinnedStatistic.from_state(state['poles'])
+
+    def _make_datasets(self, edges, poles, power, coords, attrs):
+        if self.attrs['mode'] == '1d':
+            power = BinnedStatistic(['k'], edges, power,
+                                    fields_to_sum=['modes'], coords=coords,
+                                    **attrs)
+        else:
+            power = BinnedStatistic(['k', 'mu'], edges, power,
+                                    fields_to_sum=['modes'], coords=coords,
+                                    **attrs)
+        if poles is not None:
+            poles = BinnedStatistic(['k'], [power.edges['k']], poles,
+                                    fields_to_sum=['modes'],
+                                    coords=[power.coords['k']], **attrs)
+        return power, poles
+
+
+def project_to_basis(y3d, edges, los=[0, 0, 1], poles=[]):
+    """
+    Project a 3D complex statistic onto (x, mu) bins and multipoles
+    (reference :507-701), with the sums done by the
+    ``nbk_bin_power_f64`` HIP kernel and the tiny fold/normalize tail on
+    host after one small allreduce (the reference's :669-679).
+    """
+    import torch
+    comm = y3d.pm.comm
+    lib = hiplib.require()
+
+    xedges, muedges = edges
+    Nx = len(xedges) - 1
+    Nmu = len(muedges) - 1
+
+    poles = list(poles)
+    do_poles = len(poles) > 0
+    _poles = [0] + sorted(poles) if 0 not in poles else sorted(poles)
+    ell_idx = [_poles.index(l) for l in poles]
+    Nell = len(_poles)
+    if any(ell < 0 for ell in _poles):
+        raise ValueError("in `project_to_basis`, multipole numbers must be "
+                         "non-negative integers")
+
+    NB = (Nx + 2) * (Nmu + 2)
+    nfields = 3 + 2 * Nell
+
+    dev = 'cuda'
+    k2edges_t = torch.as_tensor(numpy.asarray(xedges, dtype='f8') ** 2) \
+        .to(dev)
+    muedges_t = torch.as_tensor(numpy.asarray(muedges, dtype='f8')).to(dev)
+    sums = torch.zeros(nfields * NB, dtype=torch.float64, device=dev)
+
+    hiplib.check(lib.nbk_bin_power_f64(
+        hiplib.dptr(y3d.value), hiplib.i64_arr(y3d.pm.Nmesh),
+        hiplib.f64_arr(y3d.pm.BoxSize),
+        hiplib.i64_arr(y3d.dims), hiplib.i64_arr(y3d.off), None,
+        hiplib.dptr(k2edges_t), len(xedges),
+        hiplib.dptr(muedges_t), len(muedges),
+        hiplib.f64_arr(los), hiplib.int_arr(_poles), Nell,
+        hiplib.dptr(sums), hiplib.dptr(sums[NB:]),
+        hiplib.dptr(sums[2 * NB:]), hiplib.dptr(sums[3 * NB:]),
+        hiplib.cur_stream()), 'nbk_bin_power_f64')
+
+    torch.cuda.synchronize()
+    host = sums.cpu().numpy()
+    host = comm.allreduce(host)
+
+    shape = (Nx + 2, Nmu + 2)
+    xsum = host[:NB].reshape(shape)
+    musum = host[NB:2 * NB].reshape(shape)
+    Nsum = numpy.round(host[2 * NB:3 * NB]).astype('i8').reshape(shape)
+    ysum = host[3 * NB:].view('c16').reshape((Nell,) + shape)
+
+    # fold the internal mu == 1 bin into the last visible bin (:674-679)
+    ysum[..., -2] += ysum[..., -1]
+    musum[:, -2] += musum[:, -1]
+    xsum[:, -2] += xsum[:, -1]
+    Nsum[:, -2] += Nsum[:, -1]
+
+    sl = slice(1, -1)
+    with numpy.errstate(invalid='ignore', divide='ignore'):
+        y2d = (ysum[0] / Nsum)[sl, sl]
+        xmean_2d = (xsum / Nsum)[sl, sl]
+        mumean_2d = (musum / Nsum)[sl, sl]
+        N_2d = Nsum[sl, sl]
+
+        if do_poles:
+            N_1d = Nsum[sl, sl].sum(axis=-1)
+            xmean_1d = xsum[sl, sl].sum(axis=-1) / N_1d
+            pole_arr = ysum[:, sl, sl].sum(axis=-1) / N_1d
+            pole_arr = pole_arr[ell_idx, ...]
+            pole_result = (xmean_1d, pole_arr, N_1d)
+        else:
+            pole_result = None
+
+    return (xmean_2d, mumean_2d, y2d, N_2d), pole_result
+
+
+def _cast_source(source, BoxSize, Nmesh):
+    """Cast to a MeshSource; catalogs get dtype='f8', compensated=True
+    (reference :703-730)."""
+    from nbodykit_amd.source.mesh import FieldMesh
+
+    if isinstance(source, (RealField, ComplexField)):
+        source = FieldMesh(source)
+    elif isinstance(source, CatalogSourceBase):
+        if not isinstance(source, MeshSource):
+            source = source.to_mesh(BoxSize=BoxSize, Nmesh=Nmesh,
+                                    dtype='f8', compensated=True)
+
+    if not isinstance(source, MeshSource):
+        raise TypeError("Unknown type of source in FFTPower: %s"
+                        % str(type(source)))
+    if BoxSize is not None and any(source.attrs['BoxSize'] != BoxSize):
+        raise ValueError("Mismatched Boxsize between __init__ and source.attrs")
+    if Nmesh is not None and any(source.attrs['Nmesh'] != Nmesh):
+        raise ValueError("Mismatched Nmesh between __init__ and "
+                         "source.attrs; if trying to re-sample with a "
+                         "different mesh, specify `Nmesh` as keyword of "
+                         "to_mesh()")
+    return source
+
+
+def _find_unique_edges(x, x0, xmax, comm):
+    """dk=0 unique-modulus edges (reference :732-769)."""
+    fx2 = 0
+    for xi in x:
+        fx2 = fx2 + xi ** 2
+
+    def find_unique_local(fx2, binning):
+        fx2 = numpy.ravel(fx2)
+        ix2 = numpy.int64(fx2 / binning + 0.5)
+        ix2, ind = numpy.unique(ix2, return_index=True)
+        return fx2[ind]
+
+    binning = (x0.min() * 0.05) ** 2
+    fx = find_unique_local(fx2, binning) ** 0.5
+    fx = fx[fx < xmax]
+    fx = numpy.concatenate(comm.allgather(fx), axis=0)
+    minx0 = comm.allreduce(x0.min(), op='min')
+    fx = find_unique_local(fx, minx0 * 1e-5)
+
+    width = numpy.diff(fx)
+    edges = fx.copy()
+    edges[1:] -= width * 0.5
+    edges = numpy.append(edges, [fx[-1] + width[-1] * 0.5])
+    edges[0] = 0
+    return edges, fx

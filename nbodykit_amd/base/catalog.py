@@ -1,0 +1,288 @@
+"""
+CatalogSource — the particle-container side of the drop-in boundary
+(reference nbodykit/base/catalog.py).  The reference's columns are lazy
+dask arrays; dask is an implementation detail, not API (columns only
+ever reach the mesh layer as numpy via ``source.compute(columns)``,
+reference source/mesh/catalog.py:245), so here a :class:`ColumnAccessor`
+is a numpy-backed column with the same access protocol:
+
+- ``__getitem__`` dispatch order overrides > hardcolumns > defaults
+  (reference :370-390)
+- default columns ``Selection`` (True), ``Weight`` (1.0), ``Value`` (1.0)
+  (reference :1166-1216)
+- ``__setitem__`` with scalar -> ConstantArray broadcast (:955-972)
+- boolean/slice selection returning a new catalog (:275-325)
+- ``size`` / ``csize`` (:996-1011), ``compute`` (:530-560)
+- ``to_mesh(...)`` (:787-873) with the same signature and defaults
+"""
+import logging
+
+import numpy
+
+from nbodykit_amd import CurrentMPIComm
+
+
+def ConstantArray(value, size, chunks=None):
+    """Zero-stride broadcast of a scalar to ``size`` rows (reference
+    nbodykit/transform.py:89-107)."""
+    value = numpy.asarray(value)
+    return numpy.lib.stride_tricks.as_strided(
+        value, (size,) + value.shape, (0,) + value.strides)
+
+
+class ColumnAccessor(numpy.ndarray):
+    """A numpy view carrying its catalog and attrs (the reference's is a
+    dask.Array subclass, nbodykit/base/catalog.py:12-95)."""
+
+    def __new__(cls, catalog, array, is_default=False):
+        obj = numpy.asarray(array).view(cls)
+        obj.catalog = catalog
+        obj.is_default = is_default
+        obj.attrs = {}
+        return obj
+
+    def __array_finalize__(self, obj):
+        if obj is None:
+            return
+        self.catalog = getattr(obj, 'catalog', None)
+        self.is_default = getattr(obj, 'is_default', False)
+        self.attrs = dict(getattr(obj, 'attrs', {}))
+
+    def compute(self):
+        return numpy.asarray(self)
+
+    def as_numpy(self):
+        return numpy.asarray(self)
+
+
+def column(name=None, is_default=False):
+    """Decorator marking a method as a named column (reference :97-125)."""
+    def decorator(getter, name=name):
+        getter.column_name = getter.__name__ if name is None else name
+        getter.is_default = is_default
+        return getter
+    if callable(name):
+        getter = name
+        return decorator(getter, name=getter.__name__)
+    return decorator
+
+
+def find_columns(cls):
+    """Collect decorated columns from the class hierarchy (the reference
+    does this with a metaclass, ColumnFinder, :127-160)."""
+    hard, defaults = set(), set()
+    for klass in cls.__mro__:
+        for value in vars(klass).values():
+            if callable(value) and hasattr(value, 'column_name'):
+                (defaults if value.is_default else hard).add(
+                    value.column_name)
+    return hard, defaults
+
+
+class CatalogSourceBase(object):
+    logger = logging.getLogger('CatalogSourceBase')
+
+    def __new__(cls, *args, **kwargs):
+        obj = object.__new__(cls)
+        obj._overrides = {}
+        obj.base = None
+        hard, defaults = find_columns(cls)
+        obj._hardcolumns = sorted(hard)
+        obj._defaultcolumns = sorted(defaults)
+        return obj
+
+    def __init__(self, comm=None):
+        self.comm = comm if comm is not None else CurrentMPIComm.get()
+
+    @property
+    def attrs(self):
+        try:
+            return self._attrs
+        except AttributeError:
+            self._attrs = {}
+            return self._attrs
+
+    # -- column protocol --------------------------------------------------
+    @property
+    def hardcolumns(self):
+        return self._hardcolumns
+
+    @property
+    def columns(self):
+        all_cols = set(self._overrides) | set(self._hardcolumns) \
+            | set(self._defaultcolumns)
+        return sorted(all_cols)
+
+    def __contains__(self, col):
+        return col in self.columns
+
+    def __getitem__(self, sel):
+        if isinstance(sel, str):
+            # dispatch order: overrides > hardcolumns > defaults (:370-390)
+            if sel in self._overrides:
+                arr = self._overrides[sel]
+                is_default = False
+            elif sel in self._hardcolumns:
+                arr = getattr(self, sel)()
+                is_default = False
+            elif sel in self._defaultcolumns:
+                arr = getattr(self, sel)()
+                is_default = True
+            else:
+                raise KeyError("column `%s` is not defined in this source; "
+                               "try adding column via `source[column] = data`"
+                               % sel)
+            if isinstance(arr, ColumnAccessor):
+                return arr
+            return ColumnAccessor(self, arr, is_default=is_default)
+
+        # boolean mask or slice -> a new catalog view (:275-325)
+        if isinstance(sel, slice) or (
+                isinstance(sel, (numpy.ndarray, list))):
+            return self._get_slice(sel)
+        raise KeyError("unsupported selection type: %r" % (sel,))
+
+    def __setitem__(self, col, value):
+        size = getattr(self, 'size', None)
+        if numpy.isscalar(value):
+            assert size is not None, "size must be known to set a scalar"
+            value = ConstantArray(value, size)
+        value = numpy.asarray(value)
+        if size is not None and len(value) != size:
+            raise ValueError(
+                "error setting column '%s': data of shape %s does not match "
+                "catalog size %d" % (col, value.shape, size))
+        self._overrides[col] = value
+
+    def __delitem__(self, col):
+        if col not in self.columns:
+            raise ValueError("no such column, cannot delete it")
+        if col in self._overrides:
+            del self._overrides[col]
+        elif col in self.hardcolumns:
+            raise ValueError("cannot delete a hard-coded column")
+
+    def _get_slice(self, index):
+        if isinstance(index, list):
+            index = numpy.asarray(index)
+        if isinstance(index, numpy.ndarray) and index.dtype == bool:
+            if len(index) != self.size:
+                raise ValueError("boolean mask must have catalog length")
+        subset = object.__new__(CatalogSource)
+        subset._overrides = {}
+        subset.base = None
+        subset._hardcolumns = []
+        subset._defaultcolumns = list(self._defaultcolumns)
+        subset.comm = self.comm
+        subset._attrs = dict(self.attrs)
+        # materialize every non-default column, sliced
+        n = None
+        for col in self.columns:
+            if col in self._defaultcolumns and col not in self._overrides:
+                continue
+            data = numpy.asarray(self[col])[index]
+            subset._overrides[col] = data
+            n = len(data)
+        if n is None:
+            n = len(numpy.zeros(self.size)[index])
+        subset._size = n
+        subset._csize = subset.comm.allreduce(n)
+        return subset
+
+    def compute(self, *args):
+        """Materialize columns to numpy (reference :530-560; trivial for
+        numpy-backed columns but kept for API parity).  A single list
+        argument is materialized element-wise, like dask.compute."""
+        if len(args) == 1 and isinstance(args[0], (list, tuple)):
+            return [numpy.asarray(a) for a in args[0]]
+        toret = tuple(numpy.asarray(a) for a in args)
+        if len(toret) == 1:
+            return toret[0]
+        return list(toret)
+
+    def make_column(self, array):
+        return numpy.asarray(array)
+
+    # -- default columns (reference :1166-1216) ---------------------------
+    @column(is_default=True)
+    def Selection(self):
+        return ConstantArray(True, self.size)
+
+    @column(is_default=True)
+    def Weight(self):
+        return ConstantArray(1.0, self.size)
+
+    @column(is_default=True)
+    def Value(self):
+        return ConstantArray(1.0, self.size)
+
+    # -- mesh view --------------------------------------------------------
+    def to_mesh(self, Nmesh=None, BoxSize=None, dtype='f4', interlaced=False,
+                compensated=False, resampler='cic', weight='Weight',
+                value='Value', selection='Selection', position='Position',
+                window=None):
+        """Create a CatalogMesh view (reference :787-873; same defaults:
+        dtype='f4', resampler='cic', compensated=False, interlaced=False)."""
+        from nbodykit_amd.source.mesh.catalog import CatalogMesh
+
+        if window is not None:
+            import warnings
+            warnings.warn("The window argument is deprecated. Use "
+                          "`resampler=` instead", DeprecationWarning,
+                          stacklevel=2)
+            resampler = window
+
+        for col in [weight, selection]:
+            if col not in self:
+                raise ValueError("column '%s' missing; cannot create mesh"
+                                 % col)
+        if resampler not in ('cic', 'tsc', 'pcs'):
+            raise ValueError("valid resampler: ['cic', 'tsc', 'pcs']")
+
+        if BoxSize is None:
+            try:
+                BoxSize = self.attrs['BoxSize']
+            except KeyError:
+                raise ValueError(
+                    "cannot convert particle source to a mesh; 'BoxSize' "
+                    "keyword is not supplied and the CatalogSource does not "
+                    "define one in 'attrs'.")
+        if Nmesh is None:
+            try:
+                Nmesh = self.attrs['Nmesh']
+            except KeyError:
+                raise ValueError(
+                    "cannot convert particle source to a mesh; 'Nmesh' "
+                    "keyword is not supplied and the CatalogSource does not "
+                    "define one in 'attrs'.")
+
+        return CatalogMesh(self, Nmesh=Nmesh, BoxSize=BoxSize, dtype=dtype,
+                           Weight=self[weight], Selection=self[selection],
+                           Value=self[value], Position=self[position],
+                           interlaced=interlaced, compensated=compensated,
+                           resampler=resampler)
+
+
+class CatalogSource(CatalogSourceBase):
+    """A catalog with a well-defined local ``size`` (reference :875-1011)."""
+
+    def __init__(self, comm=None):
+        CatalogSourceBase.__init__(self, comm=comm)
+        if not hasattr(self, '_size'):
+            raise ValueError("CatalogSource subclasses must set _size "
+                             "before calling __init__")
+        self._csize = self.comm.allreduce(self._size)
+
+    @property
+    def size(self):
+        return self._size
+
+    @property
+    def csize(self):
+        return self._csize
+
+    def __len__(self):
+        return self.size
+
+    def __repr__(self):
+        return "%s(size=%d)" % (type(self).__name__, self.size)

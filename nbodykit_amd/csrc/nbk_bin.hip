@@ -1,0 +1,200 @@
+// project_to_basis binning sums (nbodykit/algorithms/fftpower.py:507-701
+// with the SlabIterator/Hermitian-weight semantics of meshtools.py):
+// one streaming pass over the local complex slab, digitizing |k|^2 and
+// mu against the edge arrays and accumulating
+//   xsum  += |k| * w,  musum += mu * w,  Nsum += w,
+//   ysum[ell] += (2 ell + 1) * Herm(P_ell(mu) * p3d)
+// into (Nx+2)x(Nmu+2) bin grids, where w is the Hermitian double-count
+// weight (2 where the compressed-axis frequency > 0, meshtools:188-215)
+// and Herm applies the conjugate-pair parity rule (fftpower.py:649-656).
+//
+// Two variants: LDS per-workgroup histograms flushed once (when the bin
+// grid fits in 64 KiB), else direct global f64 atomics.  The pass reads
+// 16 B/cell — far below the paint/FFT traffic — so either is cheap.
+#include "nbk_common.h"
+
+namespace {
+
+struct BinArgs {
+    int64_t n0, n1, n2;        // global REAL mesh (n2 = full length)
+    int64_t d0, d1, d2;        // local dims (d2 = n2/2+1 when untransposed)
+    int64_t o0, o1, o2;
+    int a0, a1, a2;
+    double k0x, k0y, k0z;      // 2 pi / BoxSize
+    double losx, losy, losz;
+    int nx_edges, nmu_edges;   // edge COUNTS (bins + 1)
+    int nell;
+};
+
+// numpy.digitize(x, edges) == count of edges <= x (right-open bins)
+__device__ __forceinline__ int dig(const double* __restrict__ edges,
+                                   int nedges, double x) {
+    int lo = 0, hi = nedges;
+    while (lo < hi) {
+        const int mid = (lo + hi) >> 1;
+        if (edges[mid] <= x) lo = mid + 1;
+        else hi = mid;
+    }
+    return lo;
+}
+
+// accumulate one element into hist arrays laid out as:
+//   [xsum (NB)] [musum (NB)] [Nsum (NB)] [ysum: nell * 2 * NB]
+// where NB = (Nx+2)*(Nmu+2)
+template <bool LDS>
+__device__ __forceinline__ void accum(double* __restrict__ h, int NB,
+                                      int bin, double kmag, double mu,
+                                      double w, const cdouble* yv, int nell) {
+    atomicAdd(&h[bin], kmag * w);
+    atomicAdd(&h[NB + bin], mu * w);
+    atomicAdd(&h[2 * NB + bin], w);
+    for (int e = 0; e < nell; e++) {
+        atomicAdd(&h[(3 + 2 * e) * NB + bin], yv[e].re);
+        atomicAdd(&h[(3 + 2 * e + 1) * NB + bin], yv[e].im);
+    }
+}
+
+#define NBK_MAX_ELL 8
+
+template <bool LDS>
+__global__ void kbin(const double* __restrict__ data, BinArgs A,
+                     const double* __restrict__ k2edges,
+                     const double* __restrict__ muedges,
+                     const int* __restrict__ ells,
+                     double* __restrict__ gout /* nfields * NB */)
+{
+    const int NB = (A.nx_edges + 1) * (A.nmu_edges + 1);
+    const int nfields = 3 + 2 * A.nell;
+
+    extern __shared__ double lh[];
+    double* h = gout;
+    if (LDS) {
+        for (int i = threadIdx.x; i < NB * nfields; i += blockDim.x)
+            lh[i] = 0.0;
+        __syncthreads();
+        h = lh;
+    }
+
+    const int64_t total = A.d0 * A.d1 * A.d2;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         idx < total; idx += stride) {
+        const int64_t l2 = idx % A.d2;
+        const int64_t l1 = (idx / A.d2) % A.d1;
+        const int64_t l0 = idx / (A.d2 * A.d1);
+        int64_t g[3];
+        g[A.a0] = l0 + A.o0;
+        g[A.a1] = l1 + A.o1;
+        g[A.a2] = l2 + A.o2;
+        const double fx = freq_full(g[0], A.n0);
+        const double fy = freq_full(g[1], A.n1);
+        const double fz = freq_half(g[2], A.n2);
+
+        const double kx = fx * A.k0x, ky = fy * A.k0y, kz = fz * A.k0z;
+        const double k2 = kx * kx + ky * ky + kz * kz;
+        const double kmag = sqrt(k2);
+        double mu = kx * A.losx + ky * A.losy + kz * A.losz;
+        mu = (kmag == 0.0) ? 0.0 : mu / kmag;
+
+        const bool nonsingular = fz > 0.0;      // doubled modes
+        const double w = nonsingular ? 2.0 : 1.0;
+
+        const int bx = dig(k2edges, A.nx_edges, k2);
+        const int bmu = dig(muedges, A.nmu_edges, mu);
+        const int bin = bx * (A.nmu_edges + 1) + bmu;
+
+        const cdouble v = {data[2 * idx], data[2 * idx + 1]};
+
+        // Legendre P_ell(mu) by recurrence, ells ascending with ells[0]==0
+        cdouble yv[NBK_MAX_ELL];
+        double Pm1 = 0.0, P = 1.0;   // P_{-1}, P_0
+        int e = 0;
+        for (int l = 0; e < A.nell; l++) {
+            if (l > 0) {
+                const double Pn = ((2 * l - 1) * mu * P - (l - 1) * Pm1) / l;
+                Pm1 = P;
+                P = Pn;
+            }
+            if (l == ells[e]) {
+                cdouble wy = cscale(v, P);
+                // conjugate-pair parity (fftpower.py:649-656)
+                if (nonsingular) {
+                    if (l % 2) wy = {0.0, 2.0 * wy.im};
+                    else wy = {2.0 * wy.re, 0.0};
+                }
+                yv[e] = cscale(wy, 2.0 * l + 1.0);
+                e++;
+            }
+        }
+
+        accum<LDS>(h, NB, bin, kmag, mu, w, yv, A.nell);
+    }
+
+    if (LDS) {
+        __syncthreads();
+        for (int i = threadIdx.x; i < NB * nfields; i += blockDim.x)
+            if (lh[i] != 0.0) atomicAdd(&gout[i], lh[i]);
+    }
+}
+
+}  // namespace
+
+extern "C" int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
+                                 const double box[3],
+                                 const int64_t dims[3], const int64_t off[3],
+                                 const int axis_map[3],
+                                 const double* kedges, int64_t nx_edges,
+                                 const double* muedges, int64_t nmu_edges,
+                                 const double los[3],
+                                 const int* ells, int nell,
+                                 double* xsum, double* musum, double* Nsum,
+                                 double* ysum, void* stream)
+{
+    (void)musum; (void)Nsum; (void)ysum;
+    if (nell > NBK_MAX_ELL) {
+        NBK_SET_ERR("nbk_bin_power_f64: at most %d multipoles", NBK_MAX_ELL);
+        return NBK_ERR_ARG;
+    }
+    // outputs must be one contiguous block [xsum|musum|Nsum|ysum] — the
+    // Python layer allocates them together; enforce the layout.
+    const int64_t NB = (nx_edges + 1) * (nmu_edges + 1);
+    if (musum != xsum + NB || Nsum != xsum + 2 * NB
+        || (nell > 0 && ysum != xsum + 3 * NB)) {
+        NBK_SET_ERR("nbk_bin_power_f64: outputs must be contiguous "
+                    "[xsum|musum|Nsum|ysum]");
+        return NBK_ERR_ARG;
+    }
+
+    BinArgs A;
+    A.n0 = nmesh[0]; A.n1 = nmesh[1]; A.n2 = nmesh[2];
+    A.d0 = dims[0]; A.d1 = dims[1]; A.d2 = dims[2];
+    A.o0 = off[0]; A.o1 = off[1]; A.o2 = off[2];
+    if (axis_map) { A.a0 = axis_map[0]; A.a1 = axis_map[1]; A.a2 = axis_map[2]; }
+    else { A.a0 = 0; A.a1 = 1; A.a2 = 2; }
+    A.k0x = 2.0 * M_PI / box[0];
+    A.k0y = 2.0 * M_PI / box[1];
+    A.k0z = 2.0 * M_PI / box[2];
+    A.losx = los[0]; A.losy = los[1]; A.losz = los[2];
+    A.nx_edges = (int)nx_edges;
+    A.nmu_edges = (int)nmu_edges;
+    A.nell = nell;
+
+    const int nfields = 3 + 2 * nell;
+    const size_t lds_bytes = (size_t)NB * nfields * sizeof(double);
+    const int64_t total = A.d0 * A.d1 * A.d2;
+    int64_t g = (total + 255) / 256;
+    if (g > 8192) g = 8192;    // bounded: LDS flush cost scales with grid
+    if (g < 1) g = 1;
+
+    hipStream_t s = (hipStream_t)stream;
+    if (lds_bytes <= 64 * 1024) {
+        hipLaunchKernelGGL(kbin<true>, dim3((uint32_t)g), dim3(256),
+                           lds_bytes, s, cplx, A, kedges, muedges, ells,
+                           xsum);
+    } else {
+        hipLaunchKernelGGL(kbin<false>, dim3((uint32_t)g), dim3(256), 0, s,
+                           cplx, A, kedges, muedges, ells, xsum);
+    }
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}

@@ -1,0 +1,294 @@
+// Hand-written power-of-two FFT passes for the 3D R2C/C2R transform
+// (replaces pfft-python / FFTW on the path; pmesh r2c/c2r semantics:
+// forward normalized by 1/Ntotal — nbodykit/algorithms/fftpower.py:126-128,
+// mockmaker.py:27-36 — and c2r the unnormalized inverse).
+//
+// Structure: the Python layer composes the 3D transform from
+//   nbk_fft_r2c_z     — real -> half-complex along the contiguous axis,
+//                       via the packed-real trick (N reals as N/2 complex,
+//                       one N/2-point FFT, then the untwiddle split)
+//   nbk_fft_c_strided — in-place complex pass along a strided axis
+// with the RCCL alltoall pencil transpose between ranks (Python side).
+//
+// Each line is transformed in LDS by an in-place radix-2
+// decimation-in-time network: coalesced natural-order global loads are
+// scattered bit-reversed into LDS, log2(N) butterfly stages follow, and
+// the natural-order result is stored back coalesced.  Twiddles come from
+// a cached per-length device table (W_N^j, j = 0..N/2) computed once on
+// host; stage twiddles index it as W_len^p = W_N^{p*N/len}.
+// Everything is f64 — the path is HBM-bound (~160 B/cell for the 3D
+// transform), so VALU cost of f64 butterflies is not the limit.
+#include "nbk_common.h"
+#include <map>
+#include <mutex>
+
+namespace {
+
+// ---- twiddle table cache ---------------------------------------------
+std::mutex table_mutex;
+std::map<int64_t, double*> twiddle_tables;   // N -> device W_N^[0..N/2]
+
+bool is_pow2(int64_t n) { return n > 0 && (n & (n - 1)) == 0; }
+
+int ilog2(int64_t n) {
+    int l = 0;
+    while ((int64_t(1) << l) < n) l++;
+    return l;
+}
+
+// returns device pointer to W_N table (N/2+1 cdouble), or nullptr on error
+double* get_twiddles(int64_t N) {
+    std::lock_guard<std::mutex> lock(table_mutex);
+    auto it = twiddle_tables.find(N);
+    if (it != twiddle_tables.end()) return it->second;
+
+    const int64_t m = N / 2 + 1;
+    double* host = (double*)malloc(m * 2 * sizeof(double));
+    if (!host) return nullptr;
+    for (int64_t j = 0; j < m; j++) {
+        const double ang = -2.0 * M_PI * (double)j / (double)N;
+        host[2 * j] = cos(ang);
+        host[2 * j + 1] = sin(ang);
+    }
+    double* dev = nullptr;
+    if (hipMalloc(&dev, m * 2 * sizeof(double)) != hipSuccess) {
+        free(host);
+        return nullptr;
+    }
+    if (hipMemcpy(dev, host, m * 2 * sizeof(double),
+                  hipMemcpyHostToDevice) != hipSuccess) {
+        free(host);
+        (void)hipFree(dev);
+        return nullptr;
+    }
+    free(host);
+    twiddle_tables[N] = dev;
+    return dev;
+}
+
+__device__ __forceinline__ int bitrev(int j, int bits) {
+    return (int)(__brev((unsigned)j) >> (32 - bits));
+}
+
+// in-place radix-2 DIT stages over an LDS buffer holding TI interleaved
+// columns: element (j, c) at buf[j*TI + c].  table = W_{2*m} device table
+// (so stage twiddles are W_len^p = table[2*p*(m/len)]).  INV: conjugate.
+template <bool INV>
+__device__ void lds_fft(cdouble* buf, int m, int bits, int TI,
+                        const cdouble* __restrict__ table) {
+    const int T = blockDim.x;
+    const int tid = threadIdx.x;
+    for (int len = 2; len <= m; len <<= 1) {
+        const int half = len >> 1;
+        const int tw = m / len;              // W_m exponent step
+        for (int w = tid; w < (m >> 1) * TI; w += T) {
+            const int c = w % TI;
+            const int j = w / TI;
+            const int grp = j / half;
+            const int pos = j % half;
+            const int i0 = (grp * len + pos) * TI + c;
+            const int i1 = i0 + half * TI;
+            cdouble wv = table[2 * pos * tw];   // W_m^{pos*tw} from W_{2m}
+            if (INV) wv.im = -wv.im;
+            const cdouble u = buf[i0];
+            const cdouble v = cmul(buf[i1], wv);
+            buf[i0] = cadd(u, v);
+            buf[i1] = csub(u, v);
+        }
+        __syncthreads();
+    }
+}
+
+// ---- z-axis real <-> half-complex ------------------------------------
+
+// one contiguous real line of nz doubles per block -> nz/2+1 complex
+__global__ void kfft_r2c_z(const double* __restrict__ real,
+                           double* __restrict__ cplx,
+                           int64_t nz, double scale,
+                           const cdouble* __restrict__ table /* W_nz */)
+{
+    extern __shared__ cdouble buf[];          // nz/2 entries
+    const int m = (int)(nz >> 1);
+    const int bits = 31 - __clz((unsigned)m);
+    const int64_t line = blockIdx.x;
+
+    const cdouble* g = (const cdouble*)(real + line * nz);  // packed pairs
+    for (int q = threadIdx.x; q < m; q += blockDim.x)
+        buf[bitrev(q, bits)] = g[q];
+    __syncthreads();
+
+    lds_fft<false>(buf, m, bits, 1, table);
+
+    // untwiddle split: X[k] = E[k] + W_nz^k * O[k], k = 0..m
+    cdouble* out = (cdouble*)cplx + line * (m + 1);
+    for (int k = threadIdx.x; k <= m; k += blockDim.x) {
+        const cdouble Zk = buf[k == m ? 0 : k];
+        const cdouble Zm = buf[(m - k) % m];
+        const cdouble E = cscale(cadd(Zk, cconj(Zm)), 0.5);
+        const cdouble D = csub(Zk, cconj(Zm));
+        const cdouble O = {0.5 * D.im, -0.5 * D.re};    // D * (-i/2)
+        const cdouble X = cadd(E, cmul(table[k], O));
+        out[k] = cscale(X, scale);
+    }
+}
+
+// inverse: nz/2+1 complex -> nz reals, unnormalized
+// (so that c2r(r2c(x, scale=1)) == nz * x per line)
+__global__ void kfft_c2r_z(const double* __restrict__ cplx,
+                           double* __restrict__ real,
+                           int64_t nz,
+                           const cdouble* __restrict__ table /* W_nz */)
+{
+    extern __shared__ cdouble smem[];
+    const int m = (int)(nz >> 1);
+    const int bits = 31 - __clz((unsigned)m);
+    cdouble* xin = smem;                      // m+1 entries
+    cdouble* buf = smem + (m + 1);            // m entries
+    const int64_t line = blockIdx.x;
+
+    const cdouble* g = (const cdouble*)cplx + line * (m + 1);
+    for (int k = threadIdx.x; k <= m; k += blockDim.x)
+        xin[k] = g[k];
+    __syncthreads();
+
+    // rebuild packed spectrum: Z[k] = E[k] + i * O[k],
+    // E = (X[k]+conj(X[m-k]))/2, O = conj(W^k) (X[k]-conj(X[m-k]))/2
+    for (int k = threadIdx.x; k < m; k += blockDim.x) {
+        const cdouble Xk = xin[k];
+        const cdouble Xm = cconj(xin[m - k]);
+        const cdouble E = cscale(cadd(Xk, Xm), 0.5);
+        const cdouble WO = cscale(csub(Xk, Xm), 0.5);
+        const cdouble O = cmul(cconj(table[k]), WO);
+        const cdouble Z = {E.re - O.im, E.im + O.re};    // E + i O
+        buf[bitrev(k, bits)] = Z;
+    }
+    __syncthreads();
+
+    lds_fft<true>(buf, m, bits, 1, table);
+
+    // unpack: line[2t] = 2 Re(z[t]), line[2t+1] = 2 Im(z[t])
+    cdouble* out = (cdouble*)(real + line * nz);
+    for (int t = threadIdx.x; t < m; t += blockDim.x)
+        out[t] = {2.0 * buf[t].re, 2.0 * buf[t].im};
+}
+
+// ---- strided complex pass --------------------------------------------
+
+// one (outer line, TI-column tile) per block; element j of column
+// (o, i) at cplx[o*ostride + j*stride + i]
+template <bool INV>
+__global__ void kfft_c_strided(double* __restrict__ data,
+                               int nfft, int64_t stride,
+                               int64_t ostride, int64_t n_inner,
+                               int TI, int tiles,
+                               const cdouble* __restrict__ table /* W_2nfft */)
+{
+    extern __shared__ cdouble buf[];          // nfft * TI
+    const int bits = 31 - __clz((unsigned)nfft);
+    const int64_t o = blockIdx.x / tiles;
+    const int64_t c0 = (int64_t)(blockIdx.x % tiles) * TI;
+    cdouble* g = (cdouble*)data + o * ostride + c0;
+    const int ncol = (int)min((int64_t)TI, n_inner - c0);
+
+    for (int w = threadIdx.x; w < nfft * TI; w += blockDim.x) {
+        const int c = w % TI;
+        const int j = w / TI;
+        if (c < ncol)
+            buf[bitrev(j, bits) * TI + c] = g[(int64_t)j * stride + c];
+    }
+    __syncthreads();
+
+    lds_fft<INV>(buf, nfft, bits, TI, table);
+
+    for (int w = threadIdx.x; w < nfft * TI; w += blockDim.x) {
+        const int c = w % TI;
+        const int j = w / TI;
+        if (c < ncol)
+            g[(int64_t)j * stride + c] = buf[j * TI + c];
+    }
+}
+
+int check_len(int64_t n, const char* who) {
+    if (!is_pow2(n) || n < 8 || n > 4096) {
+        NBK_SET_ERR("%s: length %lld unsupported (need power of two in "
+                    "[8, 4096])", who, (long long)n);
+        return NBK_ERR_UNSUPPORTED;
+    }
+    return NBK_OK;
+}
+
+}  // namespace
+
+extern "C" int nbk_fft_r2c_z(const double* real, double* cplx,
+                             int64_t nlines, int64_t nz, double scale,
+                             void* stream)
+{
+    int rc = check_len(nz, "nbk_fft_r2c_z");
+    if (rc) return rc;
+    if (nlines == 0) return NBK_OK;
+    double* table = get_twiddles(nz);
+    if (!table) { NBK_SET_ERR("twiddle alloc failed"); return NBK_ERR_HIP; }
+    const int block = (int)std::min<int64_t>(256, nz / 4);
+    const size_t shmem = (size_t)(nz / 2) * sizeof(cdouble);
+    hipLaunchKernelGGL(kfft_r2c_z, dim3((uint32_t)nlines), dim3(block), shmem,
+                       (hipStream_t)stream, real, cplx, nz, scale,
+                       (const cdouble*)table);
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}
+
+extern "C" int nbk_fft_c2r_z(const double* cplx, double* real,
+                             int64_t nlines, int64_t nz, void* stream)
+{
+    int rc = check_len(nz, "nbk_fft_c2r_z");
+    if (rc) return rc;
+    if (nlines == 0) return NBK_OK;
+    double* table = get_twiddles(nz);
+    if (!table) { NBK_SET_ERR("twiddle alloc failed"); return NBK_ERR_HIP; }
+    const int block = (int)std::min<int64_t>(256, nz / 4);
+    const size_t shmem = (size_t)(nz + 1) * sizeof(cdouble);
+    hipLaunchKernelGGL(kfft_c2r_z, dim3((uint32_t)nlines), dim3(block), shmem,
+                       (hipStream_t)stream, cplx, real, nz,
+                       (const cdouble*)table);
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}
+
+extern "C" int nbk_fft_c_strided(double* cplx, int64_t nfft, int64_t stride,
+                                 int64_t n_outer, int64_t outer_stride,
+                                 int64_t n_inner, int sign, void* stream)
+{
+    int rc = check_len(nfft, "nbk_fft_c_strided");
+    if (rc) return rc;
+    if (n_outer == 0 || n_inner == 0) return NBK_OK;
+    // W_{2*nfft} table so stage twiddles use even indices
+    double* table = get_twiddles(2 * nfft);
+    if (!table) { NBK_SET_ERR("twiddle alloc failed"); return NBK_ERR_HIP; }
+
+    // inner-tile width: LDS = nfft*TI*16 B, keep <= 64 KiB
+    int TI = 4;
+    while ((int64_t)nfft * TI * (int64_t)sizeof(cdouble) > 65536 && TI > 1)
+        TI >>= 1;
+    if (TI > n_inner) TI = (int)n_inner;
+    const int tiles = (int)((n_inner + TI - 1) / TI);
+    const int64_t grid = n_outer * tiles;
+    if (grid > 0x7fffffff) {
+        NBK_SET_ERR("nbk_fft_c_strided: grid too large");
+        return NBK_ERR_ARG;
+    }
+    const int block = 256;
+    const size_t shmem = (size_t)nfft * TI * sizeof(cdouble);
+
+    if (sign < 0)
+        hipLaunchKernelGGL(kfft_c_strided<false>, dim3((uint32_t)grid),
+                           dim3(block), shmem, (hipStream_t)stream,
+                           cplx, (int)nfft, stride, outer_stride, n_inner,
+                           TI, tiles, (const cdouble*)table);
+    else
+        hipLaunchKernelGGL(kfft_c_strided<true>, dim3((uint32_t)grid),
+                           dim3(block), shmem, (hipStream_t)stream,
+                           cplx, (int)nfft, stride, outer_stride, n_inner,
+                           TI, tiles, (const cdouble*)table);
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}
