@@ -1,0 +1,413 @@
+"""
+bench.py — FFTPower throughput on MI355X (the BASELINE.json metric:
+"FFTPower wall-clock (s) + particles/s painted, 1e9 pts/1024^3 mesh
+@1/2/4/8 GPU").
+
+One step = one full FFTPower through the public API (paint -> R2C FFT ->
+compensation -> |delta(k)|^2 -> k-binning) over a GPU-resident catalog
+(no PCIe inside the timed region).  Default N=1 workload = C4 on one
+GPU: LogNormal-config 1e9 particles, 1024^3 mesh, CIC, compensated,
+mode '1d'.  The catalog is generated ON-GPU with the same statistical
+config as BASELINE's (LogNormal: our whitenoise -> EH P(k) scaling ->
+Zel'dovich -> lognormal Poisson sampling, all through our own FFT
+kernels); exact-RNG parity catalogs are exercised by tests/, not here.
+
+Multi-rank: launched by torch.distributed.run with one rank per GPU
+(RCCL); total work fixed (BASELINE quotes 1e9 pts at every GPU count)
+=> scaling = "strong".
+
+Output: ONE JSON line from rank 0, with `roofline` (paint-kernel
+achieved HBM GB/s from HIP events vs the 8 TB/s peak) and
+`cpu_baseline` (the numpy oracle timed on this box's host cores on a
+bounded sample).
+"""
+import argparse
+import json
+import math
+import os
+import sys
+import time
+
+import numpy
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+HBM_PEAK = 8.0e12          # B/s, MI355X spec (MI355X_MICROARCH.md)
+BYTES_PER_PARTICLE = {'cic': 152.0, 'tsc': 456.0, 'pcs': 1048.0}
+
+WORKLOADS = {
+    # BASELINE.json configs (C1 is the CPU-oracle plumbing config)
+    'c2': dict(catalog='uniform', particles=int(1e7), nmesh=256,
+               box=1000., resampler='cic', interlaced=False, mode='1d'),
+    'c3': dict(catalog='lognormal', particles=int(1e8), nmesh=512,
+               box=2500., resampler='tsc', interlaced=True, mode='1d'),
+    'c4': dict(catalog='lognormal', particles=int(1e9), nmesh=1024,
+               box=5000., resampler='cic', interlaced=False, mode='1d'),
+    'c5': dict(catalog='lognormal', particles=int(1e9), nmesh=1024,
+               box=5000., resampler='cic', interlaced=False, mode='2d',
+               cross=True, Nmu=5),
+}
+
+
+def log(rank, *args):
+    if rank == 0:
+        print('[bench]', *args, file=sys.stderr, flush=True)
+
+
+# ---- GPU-side 3D FFT helpers for the generator -------------------------
+
+def gpu_r2c(real_t, lib, hiplib):
+    import torch
+    n0, n1, n2 = real_t.shape
+    nzh = n2 // 2 + 1
+    out = torch.empty((n0, n1, nzh), dtype=torch.complex128, device='cuda')
+    s = hiplib.cur_stream()
+    hiplib.check(lib.nbk_fft_r2c_z(hiplib.dptr(real_t), hiplib.dptr(out),
+                                   n0 * n1, n2, 1.0 / (n0 * n1 * n2), s),
+                 'r2c_z')
+    hiplib.check(lib.nbk_fft_c_strided(hiplib.dptr(out), n1, nzh, n0,
+                                       n1 * nzh, nzh, -1, s), 'y')
+    hiplib.check(lib.nbk_fft_c_strided(hiplib.dptr(out), n0, n1 * nzh, 1,
+                                       0, n1 * nzh, -1, s), 'x')
+    return out
+
+
+def gpu_c2r(cplx_t, n2, lib, hiplib):
+    import torch
+    n0, n1, nzh = cplx_t.shape
+    s = hiplib.cur_stream()
+    work = cplx_t.clone()
+    hiplib.check(lib.nbk_fft_c_strided(hiplib.dptr(work), n0, n1 * nzh, 1,
+                                       0, n1 * nzh, +1, s), 'xi')
+    hiplib.check(lib.nbk_fft_c_strided(hiplib.dptr(work), n1, nzh, n0,
+                                       n1 * nzh, nzh, +1, s), 'yi')
+    out = torch.empty((n0, n1, n2), dtype=torch.float64, device='cuda')
+    hiplib.check(lib.nbk_fft_c2r_z(hiplib.dptr(work), hiplib.dptr(out),
+                                   n0 * n1, n2, s), 'zi')
+    return out
+
+
+# ---- GPU catalog generation --------------------------------------------
+
+def gen_uniform(n_total, box, rank, ws, seed):
+    """per-rank slab-local uniform positions (n_total/ws each)"""
+    import torch
+    n_local = n_total // ws
+    g = torch.Generator(device='cuda')
+    g.manual_seed(seed * 1000 + rank)
+    pos = torch.rand((n_local, 3), generator=g, dtype=torch.float64,
+                     device='cuda')
+    pos[:, 0] = (pos[:, 0] + rank) * (box / ws)   # rank's x-slab
+    pos[:, 1] *= box
+    pos[:, 2] *= box
+    return pos
+
+
+def gen_lognormal(n_total, nmesh, box, rank, ws, seed, bias=2.0,
+                  redshift=0.55):
+    """GPU LogNormal pipeline (same statistical config as the CPU
+    generator in nbodykit_amd.mockmaker; torch RNG)."""
+    import torch
+    from nbodykit_amd import hiplib
+    from nbodykit_amd.cosmology import Planck15, LinearPower
+    lib = hiplib.require()
+
+    N = int(nmesh)
+    L = float(box)
+    V = L ** 3
+    Ntot = float(N) ** 3
+    nbar = n_total / V
+    H = L / N
+
+    # deterministic whitenoise, identical on all ranks
+    g = torch.Generator(device='cuda')
+    g.manual_seed(seed)
+    noise = torch.randn((N, N, N), generator=g, dtype=torch.float64,
+                        device='cuda')
+    delta_k = gpu_r2c(noise, lib, hiplib)
+    del noise
+    delta_k *= math.sqrt(Ntot)
+
+    # P(k) via a log-k interpolation table (EH, Planck15)
+    Plin = LinearPower(Planck15, redshift=redshift,
+                       transfer='EisensteinHu')
+    ktab = numpy.logspace(-6, numpy.log10(2 * math.pi / L * N * 2), 4096)
+    ptab = Plin(ktab)
+    lktab = torch.as_tensor(numpy.log(ktab)).to('cuda')
+    lptab = torch.as_tensor(numpy.log(ptab)).to('cuda')
+
+    def freqs(n, half=False):
+        if half:
+            f = torch.arange(n // 2 + 1, dtype=torch.float64,
+                             device='cuda')
+            f[-1] = -(n // 2)
+        else:
+            f = torch.arange(n, dtype=torch.float64, device='cuda')
+            f[f >= n // 2] -= n
+            if n % 2 == 0:
+                f[n // 2] = -(n // 2)
+        return f * (2 * math.pi / L)
+
+    kx = freqs(N).view(-1, 1, 1)
+    ky = freqs(N).view(1, -1, 1)
+    kz = freqs(N, half=True).view(1, 1, -1)
+    k2 = kx * kx + ky * ky + kz * kz
+    kmag = torch.sqrt(k2)
+    lk = torch.log(torch.clamp(kmag, min=ktab[0]))
+    # linear interpolation in log-log
+    idx = torch.clamp(torch.searchsorted(lktab, lk.reshape(-1)), 1,
+                      len(ktab) - 1)
+    lk0 = lktab[idx - 1]
+    lk1 = lktab[idx]
+    w = (lk.reshape(-1) - lk0) / (lk1 - lk0)
+    lp = lptab[idx - 1] * (1 - w) + lptab[idx] * w
+    P = torch.exp(lp).reshape(k2.shape)
+    del lk, idx, lk0, lk1, w, lp
+
+    delta_k *= torch.sqrt(P / V)
+    delta_k[0, 0, 0] = 0
+    del P
+
+    delta = gpu_c2r(delta_k, N, lib, hiplib)
+
+    # lognormal transform with lagrangian bias (mockmaker.py:284-287)
+    field = torch.exp((bias - 1.0) * delta)
+    del delta
+    field /= field.mean()
+    cellmean = field * (nbar * H ** 3)
+    del field
+
+    # this rank's x-slab
+    nx_l = N // ws
+    x0 = nx_l * rank
+    lam = cellmean[x0:x0 + nx_l].reshape(-1)
+    del cellmean
+    counts = torch.poisson(lam, generator=g).long()
+    del lam
+
+    npart = int(counts.sum().item())
+    # cell corner coordinates of each particle (repeat_interleave)
+    cell = torch.repeat_interleave(
+        torch.arange(nx_l * N * N, device='cuda'), counts)
+    del counts
+    iz = cell % N
+    iy = (cell // N) % N
+    ix = cell // (N * N) + x0
+
+    pos = torch.empty((npart, 3), dtype=torch.float64, device='cuda')
+    pos[:, 0] = ix.double() * H
+    pos[:, 1] = iy.double() * H
+    pos[:, 2] = iz.double() * H
+
+    # Zel'dovich displacement per cell, one axis at a time
+    k2c = k2.clone()
+    k2c[0, 0, 0] = 1.0
+    for i, ki in enumerate((kx, ky, kz)):
+        disp_k = delta_k * (1j * ki / k2c)
+        disp_k[0, 0, 0] = 0
+        disp = gpu_c2r(disp_k, N, lib, hiplib)
+        del disp_k
+        pos[:, i] += disp[x0:x0 + nx_l].reshape(-1)[cell - x0 * N * N]
+        del disp
+    del delta_k, k2, k2c, cell, ix, iy, iz
+
+    # in-cell uniform shift then periodic wrap
+    pos += torch.rand((npart, 3), generator=g, dtype=torch.float64,
+                      device='cuda') * H
+    pos %= L
+    return pos
+
+
+# ---- main ---------------------------------------------------------------
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument('--gpus', type=int, default=1)
+    ap.add_argument('--steps', type=int, default=3)
+    ap.add_argument('--warmup', type=int, default=1)
+    ap.add_argument('--workload', default='c4',
+                    choices=sorted(WORKLOADS))
+    ap.add_argument('--particles', type=int, default=None)
+    ap.add_argument('--nmesh', type=int, default=None)
+    ap.add_argument('--no-cpu-baseline', action='store_true')
+    ap.add_argument('--traffic-bytes', type=float, default=None,
+                    help='measured HBM bytes per paint launch from '
+                         'rocprofv3 --pmc (see profiles/)')
+    args = ap.parse_args()
+
+    import torch
+    ws = int(os.environ.get('WORLD_SIZE', '1'))
+    rank = int(os.environ.get('RANK', '0'))
+    if ws > 1:
+        import torch.distributed as dist
+        local = int(os.environ.get('LOCAL_RANK', rank))
+        torch.cuda.set_device(local)
+        dist.init_process_group('nccl')
+
+    from nbodykit_amd import profiling, set_options
+    from nbodykit_amd.lab import FFTPower
+    from nbodykit_amd.source.catalog.device import DeviceArrayCatalog
+    from nbodykit_amd.comm import default_comm
+
+    comm = default_comm()
+    assert comm.size == ws
+
+    cfg = dict(WORKLOADS[args.workload])
+    if args.particles:
+        cfg['particles'] = args.particles
+    if args.nmesh:
+        cfg['nmesh'] = args.nmesh
+
+    n_total = cfg['particles']
+    nmesh = cfg['nmesh']
+    box = cfg['box']
+    cross = cfg.get('cross', False)
+
+    log(rank, 'workload %s: %s %.0e particles, %d^3 mesh, %s%s, %s'
+        % (args.workload, cfg['catalog'], n_total, nmesh, cfg['resampler'],
+           ' interlaced' if cfg['interlaced'] else '', cfg['mode']))
+
+    t_gen = time.time()
+    if cfg['catalog'] == 'uniform':
+        pos = gen_uniform(n_total, box, rank, ws, seed=42)
+        pos2 = gen_uniform(n_total, box, rank, ws, seed=43) if cross \
+            else None
+    else:
+        half = n_total // 2 if cross else n_total
+        pos = gen_lognormal(half, nmesh, box, rank, ws, seed=42)
+        pos2 = gen_lognormal(half, nmesh, box, rank, ws, seed=43) \
+            if cross else None
+    torch.cuda.synchronize()
+    log(rank, 'generation done in %.1fs; local particles: %d'
+        % (time.time() - t_gen, len(pos)))
+
+    boxarr = numpy.array([box] * 3)
+    cat = DeviceArrayCatalog({'Position': pos}, BoxSize=boxarr)
+    cat2 = DeviceArrayCatalog({'Position': pos2}, BoxSize=boxarr) \
+        if cross else None
+
+    def make_mesh(c):
+        return c.to_mesh(Nmesh=nmesh, dtype='f8', compensated=True,
+                         resampler=cfg['resampler'],
+                         interlaced=cfg['interlaced'])
+
+    fft_kw = dict(mode=cfg['mode'])
+    if cfg['mode'] == '2d':
+        fft_kw['Nmu'] = cfg.get('Nmu', 5)
+
+    def step():
+        r = FFTPower(make_mesh(cat), second=(make_mesh(cat2) if cross
+                                             else None), **fft_kw)
+        return r
+
+    # one huge chunk: data is GPU-resident, the chunk loop is vestigial
+    with set_options(paint_chunk_size=1 << 30):
+        for _ in range(args.warmup):
+            step()
+        torch.cuda.synchronize()
+        comm.barrier()
+
+        profiling.enable()
+        t0 = time.time()
+        for _ in range(args.steps):
+            result = step()
+        torch.cuda.synchronize()
+        comm.barrier()
+        elapsed = time.time() - t0
+    elapsed = comm.allreduce(elapsed, op='max')
+
+    prof = profiling.summary()
+    profiling.disable()
+
+    n_global = comm.allreduce(len(pos)) + (comm.allreduce(len(pos2))
+                                           if cross else 0)
+    value = n_global * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+
+    # paint-kernel roofline (rank-0 local figures)
+    paint = prof.get('paint', {'ms': 0.0, 'calls': 0, 'units': 0})
+    bpp = BYTES_PER_PARTICLE[cfg['resampler']]
+    algo_bytes = paint['units'] * bpp
+    achieved = algo_bytes / (paint['ms'] * 1e-3) if paint['ms'] > 0 else 0.
+    traffic = None
+    if args.traffic_bytes is not None:
+        traffic = args.traffic_bytes
+
+    cpu_baseline = None
+    if rank == 0 and ws == 1 and not args.no_cpu_baseline:
+        cpu_baseline = run_cpu_baseline(cfg)
+
+    if rank == 0:
+        out = {
+            'metric': 'FFTPower_particles_per_s',
+            'value': value,
+            'unit': 'particles/s',
+            'n_gpus': ws,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': ms_per_step,
+            'higher_is_better': True,
+            'scaling': 'strong',
+            'vs_baseline': None,
+            'dtype': 'f64',
+            'data': 'synthetic (GPU-generated %s, BASELINE %s config, '
+                    'torch RNG)' % (cfg['catalog'], args.workload.upper()),
+            'config': {
+                'workload': args.workload.upper(),
+                'particles': n_global,
+                'nmesh': nmesh,
+                'box': box,
+                'resampler': cfg['resampler'],
+                'interlaced': cfg['interlaced'],
+                'mode': cfg['mode'],
+                'compensated': True,
+                'parallelism': 'slab-dp%d' % ws,
+            },
+            'roofline': {
+                'bound': 'hbm',
+                'achieved': achieved / 1e9,
+                'peak': HBM_PEAK / 1e9,
+                'unit': 'GB/s',
+                'frac': achieved / HBM_PEAK,
+                'traffic': traffic,
+                'kernel': 'nbk_paint_f64[%s]' % cfg['resampler'],
+                'paint_ms_per_launch': (paint['ms'] / paint['calls']
+                                        if paint['calls'] else None),
+                'algorithmic_B_per_particle': bpp,
+            },
+            'cpu_baseline': cpu_baseline,
+        }
+        print(json.dumps(out), flush=True)
+
+    if ws > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+def run_cpu_baseline(cfg):
+    """Time the CPU oracle (the restated reference algorithm) on a
+    bounded sample of the same workload shape: ~1 particle/cell like C4,
+    sized for tens of CPU-seconds."""
+    from oracle import fftpower_oracle
+    n = int(2e6)
+    nmesh = 128
+    box = cfg['box'] * nmesh / cfg['nmesh']
+    rng = numpy.random.RandomState(42)
+    pos = rng.uniform(0, box, size=(n, 3))
+    t0 = time.time()
+    fftpower_oracle(pos, Nmesh=nmesh, BoxSize=box, mode=cfg['mode'],
+                    resampler=cfg['resampler'],
+                    interlaced=cfg['interlaced'], compensated=True)
+    dt = time.time() - t0
+    return {
+        'value': n / dt,
+        'unit': 'particles/s',
+        'cores': 1,
+        'kind': 'port',
+        'sample': 'oracle FFTPower, %d uniform pts / %d^3 mesh '
+                  '(same ~1 pt/cell shape), %.1fs' % (n, nmesh, dt),
+    }
+
+
+if __name__ == '__main__':
+    main()

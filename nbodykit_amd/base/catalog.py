@@ -22,6 +22,10 @@ import numpy
 from nbodykit_amd import CurrentMPIComm
 
 
+def _is_torch(obj):
+    return type(obj).__module__.startswith('torch')
+
+
 def ConstantArray(value, size, chunks=None):
     """Zero-stride broadcast of a scalar to ``size`` rows (reference
     nbodykit/transform.py:89-107)."""
@@ -132,7 +136,8 @@ class CatalogSourceBase(object):
                 raise KeyError("column `%s` is not defined in this source; "
                                "try adding column via `source[column] = data`"
                                % sel)
-            if isinstance(arr, ColumnAccessor):
+            if isinstance(arr, ColumnAccessor) or _is_torch(arr):
+                # device-resident columns pass through untouched
                 return arr
             return ColumnAccessor(self, arr, is_default=is_default)
 
@@ -147,7 +152,8 @@ class CatalogSourceBase(object):
         if numpy.isscalar(value):
             assert size is not None, "size must be known to set a scalar"
             value = ConstantArray(value, size)
-        value = numpy.asarray(value)
+        if not _is_torch(value):
+            value = numpy.asarray(value)
         if size is not None and len(value) != size:
             raise ValueError(
                 "error setting column '%s': data of shape %s does not match "
@@ -193,9 +199,11 @@ class CatalogSourceBase(object):
         """Materialize columns to numpy (reference :530-560; trivial for
         numpy-backed columns but kept for API parity).  A single list
         argument is materialized element-wise, like dask.compute."""
+        def materialize(a):
+            return a if _is_torch(a) else numpy.asarray(a)
         if len(args) == 1 and isinstance(args[0], (list, tuple)):
-            return [numpy.asarray(a) for a in args[0]]
-        toret = tuple(numpy.asarray(a) for a in args)
+            return [materialize(a) for a in args[0]]
+        toret = tuple(materialize(a) for a in args)
         if len(toret) == 1:
             return toret[0]
         return list(toret)

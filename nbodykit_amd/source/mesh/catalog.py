@@ -18,9 +18,41 @@ import warnings
 import numpy
 
 from nbodykit_amd import _global_options
-from nbodykit_amd import hiplib
+from nbodykit_amd import hiplib, profiling
 from nbodykit_amd.base.mesh import MeshSource
 from nbodykit_amd.pm import RealField, exchange_particle_arrays
+
+
+def _to_device_f64(arr):
+    """numpy or torch -> contiguous f64 CUDA tensor (no copy when it is
+    already one: GPU-resident catalogs feed the kernel directly)."""
+    import torch
+    if isinstance(arr, torch.Tensor):
+        t = arr
+    else:
+        t = torch.as_tensor(numpy.ascontiguousarray(numpy.asarray(arr)))
+    return t.to(device='cuda', dtype=torch.float64).contiguous()
+
+
+def _to_device_mask(arr):
+    import torch
+    if isinstance(arr, torch.Tensor):
+        return arr.to(device='cuda', dtype=torch.bool)
+    return torch.as_tensor(numpy.asarray(arr, dtype=bool)).to('cuda')
+
+
+def _is_trivial_true(sel):
+    """all-True zero-stride Selection default: skip the mask entirely"""
+    a = sel if isinstance(sel, numpy.ndarray) else None
+    return a is not None and a.ndim == 1 and len(a) > 0 \
+        and a.strides[0] == 0 and bool(a[0])
+
+
+def _is_trivial_unit(col):
+    """unit-valued zero-stride Weight/Value default: no device copy"""
+    a = col if isinstance(col, numpy.ndarray) else None
+    return a is not None and a.ndim == 1 and len(a) > 0 \
+        and a.strides[0] == 0 and float(a[0]) == 1.0
 
 # union of x-cell offsets a particle can deposit to, relative to
 # floor(u0): see DESIGN.md (window support 2/3/4, interlacing shifts by
@@ -146,28 +178,51 @@ class CatalogMesh(MeshSource):
                 if not isinstance(data, list):
                     data = [data]
                 sel = Ellipsis if Selection is None else data.pop()
-                value = None if Value is None else data.pop()[sel]
-                weight = None if Weight is None else data.pop()[sel]
-                position = data.pop()[sel]
+                if sel is not Ellipsis and _is_trivial_true(sel):
+                    sel = Ellipsis
+                value = None if Value is None else data.pop()
+                weight = None if Weight is None else data.pop()
+                position = data.pop()
+                if sel is not Ellipsis:
+                    sel = _to_device_mask(sel)
+                    position = _to_device_f64(position)[sel]
+                    if weight is not None:
+                        weight = _to_device_f64(weight)[sel]
+                    if value is not None:
+                        value = _to_device_f64(value)[sel]
             else:
                 position = numpy.empty((0, 3), dtype='f8')
                 weight = None
                 value = None
 
-            if weight is None:
-                weight = numpy.ones(len(position))
-            if value is None:
-                value = numpy.ones(len(position))
+            # unit-valued broadcast columns (the Weight/Value defaults)
+            # never leave the host — the kernel takes mass = NULL
+            if weight is not None and _is_trivial_unit(weight):
+                weight = None
+            if value is not None and _is_trivial_unit(value):
+                value = None
 
-            Nlocal += len(position)
-            Wlocal += float(numpy.sum(weight))
-            W2local += float(numpy.sum(numpy.asarray(weight) ** 2))
+            pos_t = _to_device_f64(position)
+            w_t = None if weight is None else _to_device_f64(weight)
+            v_t = None if value is None else _to_device_f64(value)
 
-            pos_t = torch.as_tensor(numpy.ascontiguousarray(position),
-                                    dtype=torch.float64).to('cuda')
-            mass_np = numpy.asarray(weight, dtype='f8') \
-                * numpy.asarray(value, dtype='f8')
-            mass_t = torch.as_tensor(mass_np, dtype=torch.float64).to('cuda')
+            n_chunk = len(pos_t)
+            Nlocal += n_chunk
+            if w_t is None:
+                Wlocal += float(n_chunk)
+                W2local += float(n_chunk)
+            else:
+                Wlocal += float(w_t.sum().item())
+                W2local += float((w_t * w_t).sum().item())
+
+            if w_t is None and v_t is None:
+                mass_t = None
+            elif v_t is None:
+                mass_t = w_t
+            elif w_t is None:
+                mass_t = v_t
+            else:
+                mass_t = w_t * v_t
 
             if comm.size > 1:
                 pos_t, mass_t = self._route(pos_t, mass_t)
@@ -175,17 +230,19 @@ class CatalogMesh(MeshSource):
             n = len(pos_t)
             if n > 0:
                 pos_soa = pos_t.t().contiguous()    # (3, n): x[n] y[n] z[n]
-                hiplib.check(lib.nbk_paint_f64(
-                    hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n,
-                    nmesh, box, window_id, 0.0,
-                    hiplib.dptr((real1 if interlaced else toret).value),
-                    pm.x_start, pm.nx_local, stream), 'nbk_paint_f64')
-                if interlaced:
+                with profiling.collect('paint', n * (1 + interlaced)):
                     hiplib.check(lib.nbk_paint_f64(
                         hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n,
-                        nmesh, box, window_id, 0.5,
-                        hiplib.dptr(real2.value),
+                        nmesh, box, window_id, 0.0,
+                        hiplib.dptr((real1 if interlaced else toret).value),
                         pm.x_start, pm.nx_local, stream), 'nbk_paint_f64')
+                    if interlaced:
+                        hiplib.check(lib.nbk_paint_f64(
+                            hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n,
+                            nmesh, box, window_id, 0.5,
+                            hiplib.dptr(real2.value),
+                            pm.x_start, pm.nx_local, stream),
+                            'nbk_paint_f64')
             i = i + chunksize
 
         if interlaced:
@@ -262,6 +319,9 @@ class CatalogMesh(MeshSource):
         ranks = ranks[order]
         counts = torch.bincount(ranks, minlength=ws).cpu().tolist()
 
+        if mass_t is None:
+            recv = exchange_particle_arrays(pos_t[idxs], counts, comm)
+            return recv.contiguous(), None
         payload = torch.cat([pos_t[idxs], mass_t[idxs, None]], dim=1)
         recv = exchange_particle_arrays(payload, counts, comm)
         return recv[:, :3].contiguous(), recv[:, 3].contiguous()

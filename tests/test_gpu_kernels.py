@@ -1,0 +1,215 @@
+"""
+Kernel-level GPU parity: each HIP kernel against its oracle restatement
+on the same seeded inputs.  f64 bars: FFT/elementwise ~1e-12 relative
+(pure fp ordering differences), paint 1e-12 (atomic order only).
+"""
+import numpy
+import pytest
+from numpy.testing import assert_allclose
+
+pytestmark = pytest.mark.gpu
+
+torch = pytest.importorskip('torch')
+if not torch.cuda.is_available():
+    pytest.skip('no GPU', allow_module_level=True)
+
+from nbodykit_amd import hiplib                                 # noqa: E402
+from oracle import (MeshGeometry, r2c, c2r, paint,              # noqa: E402
+                    apply_compensation, compute_3d_power)
+from oracle.mesh import complex_circular_coords                 # noqa: E402
+
+
+def dev(arr):
+    return torch.as_tensor(numpy.ascontiguousarray(arr)).to('cuda')
+
+
+def host(t):
+    torch.cuda.synchronize()
+    return t.cpu().numpy()
+
+
+@pytest.fixture(scope='module')
+def lib():
+    return hiplib.require()
+
+
+def rand_positions(n, box, seed=1):
+    rng = numpy.random.RandomState(seed)
+    return rng.uniform(0, box, size=(n, 3))
+
+
+# ---- paint --------------------------------------------------------------
+
+@pytest.mark.parametrize('window', ['cic', 'tsc', 'pcs'])
+@pytest.mark.parametrize('shift', [0.0, 0.5])
+def test_paint_matches_oracle(lib, window, shift):
+    geom = MeshGeometry(16, 32.)
+    pos = rand_positions(5000, 32., seed=3)
+    mass = numpy.random.RandomState(4).uniform(0.5, 1.5, size=len(pos))
+
+    want = numpy.zeros((16, 16, 16))
+    paint(pos, mass, want, geom, resampler=window, shift=shift)
+
+    mesh_t = torch.zeros((16, 16, 16), dtype=torch.float64, device='cuda')
+    pos_soa = dev(pos).t().contiguous()
+    hiplib.check(lib.nbk_paint_f64(
+        hiplib.dptr(pos_soa), hiplib.dptr(dev(mass)), len(pos),
+        hiplib.i64_arr(geom.Nmesh), hiplib.f64_arr(geom.BoxSize),
+        hiplib.WINDOW_IDS[window], shift,
+        hiplib.dptr(mesh_t), 0, 16, None), 'paint')
+    got = host(mesh_t)
+    assert_allclose(got, want, rtol=1e-12, atol=1e-12)
+    # total mass deposited == sum of masses
+    assert_allclose(got.sum(), mass.sum(), rtol=1e-12)
+
+
+def test_paint_unit_mass_null_pointer(lib):
+    geom = MeshGeometry(8, 8.)
+    pos = rand_positions(100, 8., seed=5)
+    want = numpy.zeros((8, 8, 8))
+    paint(pos, 1.0, want, geom, resampler='cic')
+    mesh_t = torch.zeros((8, 8, 8), dtype=torch.float64, device='cuda')
+    pos_soa = dev(pos).t().contiguous()
+    hiplib.check(lib.nbk_paint_f64(
+        hiplib.dptr(pos_soa), None, len(pos),
+        hiplib.i64_arr(geom.Nmesh), hiplib.f64_arr(geom.BoxSize),
+        0, 0.0, hiplib.dptr(mesh_t), 0, 8, None), 'paint')
+    assert_allclose(host(mesh_t), want, rtol=1e-12, atol=1e-12)
+
+
+def test_paint_slab_bounds(lib):
+    """painting into two half-slabs == painting the full mesh"""
+    geom = MeshGeometry(16, 32.)
+    pos = rand_positions(2000, 32., seed=6)
+    full = numpy.zeros((16, 16, 16))
+    paint(pos, 1.0, full, geom, resampler='tsc')
+
+    got = numpy.zeros((16, 16, 16))
+    pos_soa = dev(pos).t().contiguous()
+    for x0 in (0, 8):
+        slab = torch.zeros((8, 16, 16), dtype=torch.float64, device='cuda')
+        hiplib.check(lib.nbk_paint_f64(
+            hiplib.dptr(pos_soa), None, len(pos),
+            hiplib.i64_arr(geom.Nmesh), hiplib.f64_arr(geom.BoxSize),
+            1, 0.0, hiplib.dptr(slab), x0, 8, None), 'paint')
+        got[x0:x0 + 8] = host(slab)
+    assert_allclose(got, full, rtol=1e-12, atol=1e-12)
+
+
+# ---- FFT ----------------------------------------------------------------
+
+@pytest.mark.parametrize('N', [8, 16, 32, 64, 128])
+def test_r2c_matches_oracle(lib, N):
+    geom = MeshGeometry(N, 100.)
+    field = numpy.random.RandomState(7).normal(size=(N, N, N))
+    want = r2c(field, geom)
+
+    real_t = dev(field)
+    nzh = N // 2 + 1
+    cplx_t = torch.empty((N, N, nzh), dtype=torch.complex128,
+                         device='cuda')
+    scale = 1.0 / N ** 3
+    hiplib.check(lib.nbk_fft_r2c_z(
+        hiplib.dptr(real_t), hiplib.dptr(cplx_t), N * N, N, scale, None),
+        'z')
+    hiplib.check(lib.nbk_fft_c_strided(
+        hiplib.dptr(cplx_t), N, nzh, N, N * nzh, nzh, -1, None), 'y')
+    hiplib.check(lib.nbk_fft_c_strided(
+        hiplib.dptr(cplx_t), N, N * nzh, 1, 0, N * nzh, -1, None), 'x')
+    got = host(cplx_t)
+    assert_allclose(got, want, rtol=1e-11, atol=1e-13)
+
+
+@pytest.mark.parametrize('N', [8, 32, 64])
+def test_c2r_roundtrip(lib, N):
+    geom = MeshGeometry(N, 50.)
+    field = numpy.random.RandomState(8).normal(size=(N, N, N))
+    nzh = N // 2 + 1
+
+    real_t = dev(field)
+    cplx_t = torch.empty((N, N, nzh), dtype=torch.complex128,
+                         device='cuda')
+    hiplib.check(lib.nbk_fft_r2c_z(
+        hiplib.dptr(real_t), hiplib.dptr(cplx_t), N * N, N,
+        1.0 / N ** 3, None), 'z')
+    hiplib.check(lib.nbk_fft_c_strided(
+        hiplib.dptr(cplx_t), N, nzh, N, N * nzh, nzh, -1, None), 'y')
+    hiplib.check(lib.nbk_fft_c_strided(
+        hiplib.dptr(cplx_t), N, N * nzh, 1, 0, N * nzh, -1, None), 'x')
+
+    # inverse
+    hiplib.check(lib.nbk_fft_c_strided(
+        hiplib.dptr(cplx_t), N, N * nzh, 1, 0, N * nzh, +1, None), 'xi')
+    hiplib.check(lib.nbk_fft_c_strided(
+        hiplib.dptr(cplx_t), N, nzh, N, N * nzh, nzh, +1, None), 'yi')
+    back_t = torch.empty((N, N, N), dtype=torch.float64, device='cuda')
+    hiplib.check(lib.nbk_fft_c2r_z(
+        hiplib.dptr(cplx_t), hiplib.dptr(back_t), N * N, N, None), 'zi')
+    assert_allclose(host(back_t), field, rtol=1e-11, atol=1e-12)
+
+
+def test_fft_rejects_bad_length(lib):
+    t = torch.zeros(24, dtype=torch.float64, device='cuda')
+    c = torch.zeros(14, dtype=torch.complex128, device='cuda')
+    rc = lib.nbk_fft_r2c_z(hiplib.dptr(t), hiplib.dptr(c), 2, 12, 1.0,
+                           None)
+    assert rc == -3     # NBK_ERR_UNSUPPORTED
+
+
+# ---- k-space elementwise ------------------------------------------------
+
+@pytest.mark.parametrize('window', ['cic', 'tsc', 'pcs'])
+@pytest.mark.parametrize('interlaced', [False, True])
+def test_compensate_matches_oracle(lib, window, interlaced):
+    geom = MeshGeometry(16, 32.)
+    rng = numpy.random.RandomState(9)
+    c = (rng.normal(size=geom.cshape)
+         + 1j * rng.normal(size=geom.cshape)).astype('c16')
+    want = c.copy()
+    apply_compensation(want, geom, window, interlaced)
+
+    c_t = dev(c)
+    dims = hiplib.i64_arr(geom.cshape)
+    off = hiplib.i64_arr((0, 0, 0))
+    hiplib.check(lib.nbk_compensate_f64(
+        hiplib.dptr(c_t), hiplib.i64_arr(geom.Nmesh), dims, off, None,
+        hiplib.WINDOW_IDS[window], int(interlaced), None), 'comp')
+    assert_allclose(host(c_t), want, rtol=1e-12, atol=1e-12)
+
+
+def test_interlace_combine_matches_oracle(lib):
+    geom = MeshGeometry(16, 32.)
+    rng = numpy.random.RandomState(10)
+    c1 = (rng.normal(size=geom.cshape)
+          + 1j * rng.normal(size=geom.cshape)).astype('c16')
+    c2 = (rng.normal(size=geom.cshape)
+          + 1j * rng.normal(size=geom.cshape)).astype('c16')
+
+    k = complex_circular_coords(geom)    # w = k*H
+    kH = k[0] + k[1] + k[2]
+    want = c1 * 0.5 + c2 * 0.5 * numpy.exp(0.5j * kH)
+
+    c1_t, c2_t = dev(c1), dev(c2)
+    hiplib.check(lib.nbk_interlace_combine_f64(
+        hiplib.dptr(c1_t), hiplib.dptr(c2_t),
+        hiplib.i64_arr(geom.Nmesh), hiplib.f64_arr(geom.BoxSize),
+        hiplib.i64_arr(geom.cshape), hiplib.i64_arr((0, 0, 0)), None,
+        None), 'interlace')
+    assert_allclose(host(c1_t), want, rtol=1e-12, atol=1e-12)
+
+
+def test_power3d_matches_oracle(lib):
+    geom = MeshGeometry(16, 32.)
+    rng = numpy.random.RandomState(11)
+    c1 = (rng.normal(size=geom.cshape)
+          + 1j * rng.normal(size=geom.cshape)).astype('c16')
+    c2 = (rng.normal(size=geom.cshape)
+          + 1j * rng.normal(size=geom.cshape)).astype('c16')
+    want = compute_3d_power(c1, c2, geom)
+
+    out_t = torch.empty(geom.cshape, dtype=torch.complex128, device='cuda')
+    hiplib.check(lib.nbk_power3d_f64(
+        hiplib.dptr(out_t), hiplib.dptr(dev(c1)), hiplib.dptr(dev(c2)),
+        float(numpy.prod(geom.BoxSize)), hiplib.i64_arr(geom.cshape),
+        hiplib.i64_arr((0, 0, 0)), 1, None), 'power3d')
+    assert_allclose(host(out_t), want, rtol=1e-12, atol=1e-12)
