@@ -15,6 +15,8 @@
 
 namespace {
 
+#define NBK_MAX_ELL 8
+
 struct BinArgs {
     int64_t n0, n1, n2;        // global REAL mesh (n2 = full length)
     int64_t d0, d1, d2;        // local dims (d2 = n2/2+1 when untransposed)
@@ -24,6 +26,8 @@ struct BinArgs {
     double losx, losy, losz;
     int nx_edges, nmu_edges;   // edge COUNTS (bins + 1)
     int nell;
+    int ells[NBK_MAX_ELL];     // multipole orders, by value (kernel args
+                               // live in SGPRs — no host pointer deref)
 };
 
 // numpy.digitize(x, edges) == count of edges <= x (right-open bins)
@@ -54,13 +58,10 @@ __device__ __forceinline__ void accum(double* __restrict__ h, int NB,
     }
 }
 
-#define NBK_MAX_ELL 8
-
 template <bool LDS>
 __global__ void kbin(const double* __restrict__ data, BinArgs A,
                      const double* __restrict__ k2edges,
                      const double* __restrict__ muedges,
-                     const int* __restrict__ ells,
                      double* __restrict__ gout /* nfields * NB */)
 {
     const int NB = (A.nx_edges + 1) * (A.nmu_edges + 1);
@@ -115,7 +116,7 @@ __global__ void kbin(const double* __restrict__ data, BinArgs A,
                 Pm1 = P;
                 P = Pn;
             }
-            if (l == ells[e]) {
+            if (l == A.ells[e]) {
                 cdouble wy = cscale(v, P);
                 // conjugate-pair parity (fftpower.py:649-656)
                 if (nonsingular) {
@@ -178,6 +179,8 @@ extern "C" int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
     A.nx_edges = (int)nx_edges;
     A.nmu_edges = (int)nmu_edges;
     A.nell = nell;
+    for (int e = 0; e < NBK_MAX_ELL; e++)
+        A.ells[e] = e < nell ? ells[e] : -1;
 
     const int nfields = 3 + 2 * nell;
     const size_t lds_bytes = (size_t)NB * nfields * sizeof(double);
@@ -189,11 +192,10 @@ extern "C" int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
     hipStream_t s = (hipStream_t)stream;
     if (lds_bytes <= 64 * 1024) {
         hipLaunchKernelGGL(kbin<true>, dim3((uint32_t)g), dim3(256),
-                           lds_bytes, s, cplx, A, kedges, muedges, ells,
-                           xsum);
+                           lds_bytes, s, cplx, A, kedges, muedges, xsum);
     } else {
         hipLaunchKernelGGL(kbin<false>, dim3((uint32_t)g), dim3(256), 0, s,
-                           cplx, A, kedges, muedges, ells, xsum);
+                           cplx, A, kedges, muedges, xsum);
     }
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;

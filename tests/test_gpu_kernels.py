@@ -52,8 +52,9 @@ def test_paint_matches_oracle(lib, window, shift):
 
     mesh_t = torch.zeros((16, 16, 16), dtype=torch.float64, device='cuda')
     pos_soa = dev(pos).t().contiguous()
+    mass_t = dev(mass)     # keep alive: the kernel holds only the pointer
     hiplib.check(lib.nbk_paint_f64(
-        hiplib.dptr(pos_soa), hiplib.dptr(dev(mass)), len(pos),
+        hiplib.dptr(pos_soa), hiplib.dptr(mass_t), len(pos),
         hiplib.i64_arr(geom.Nmesh), hiplib.f64_arr(geom.BoxSize),
         hiplib.WINDOW_IDS[window], shift,
         hiplib.dptr(mesh_t), 0, 16, None), 'paint')
@@ -208,8 +209,9 @@ def test_power3d_matches_oracle(lib):
     want = compute_3d_power(c1, c2, geom)
 
     out_t = torch.empty(geom.cshape, dtype=torch.complex128, device='cuda')
+    c1_t, c2_t = dev(c1), dev(c2)    # keep alive past the async launch
     hiplib.check(lib.nbk_power3d_f64(
-        hiplib.dptr(out_t), hiplib.dptr(dev(c1)), hiplib.dptr(dev(c2)),
+        hiplib.dptr(out_t), hiplib.dptr(c1_t), hiplib.dptr(c2_t),
         float(numpy.prod(geom.BoxSize)), hiplib.i64_arr(geom.cshape),
         hiplib.i64_arr((0, 0, 0)), 1, None), 'power3d')
     assert_allclose(host(out_t), want, rtol=1e-12, atol=1e-12)
