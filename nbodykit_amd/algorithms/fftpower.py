@@ -278,7 +278,9 @@ def project_to_basis(y3d, edges, los=[0, 0, 1], poles=[]):
     xsum = host[:NB].reshape(shape)
     musum = host[NB:2 * NB].reshape(shape)
     Nsum = numpy.round(host[2 * NB:3 * NB]).astype('i8').reshape(shape)
-    ysum = host[3 * NB:].view('c16').reshape((Nell,) + shape)
+    # kernel layout is planar: [y0.re | y0.im | y1.re | ...] (NB each)
+    ys = host[3 * NB:].reshape(Nell, 2, NB)
+    ysum = (ys[:, 0] + 1j * ys[:, 1]).reshape((Nell,) + shape)
 
     # fold the internal mu == 1 bin into the last visible bin (:674-679)
     ysum[..., -2] += ysum[..., -1]

@@ -199,6 +199,76 @@ def test_interlace_combine_matches_oracle(lib):
     assert_allclose(host(c1_t), want, rtol=1e-12, atol=1e-12)
 
 
+@pytest.mark.parametrize('poles', [[], [0, 2, 4]])
+@pytest.mark.parametrize('los', [(0, 0, 1), (1, 0, 0)])
+def test_bin_kernel_matches_oracle(lib, poles, los):
+    from oracle import project_to_basis as oracle_bin
+    geom = MeshGeometry(16, 32.)
+    rng = numpy.random.RandomState(12)
+    y3d = (rng.normal(size=geom.cshape)
+           + 1j * rng.normal(size=geom.cshape)).astype('c16')
+
+    kedges = numpy.arange(0., numpy.pi * 16 / 32. + 0.3, 0.15)
+    Nmu = 4
+    muedges = numpy.linspace(-1, 1, Nmu + 1)
+    want, want_poles = oracle_bin(y3d, geom, [kedges, muedges], los=los,
+                                  poles=poles)
+
+    _poles = sorted(set([0] + list(poles)))
+    Nell = len(_poles)
+    Nx = len(kedges) - 1
+    NB = (Nx + 2) * (Nmu + 2)
+    nfields = 3 + 2 * Nell
+    sums = torch.zeros(nfields * NB, dtype=torch.float64, device='cuda')
+    y_t = dev(y3d)
+    k2_t = dev(kedges ** 2)
+    mu_t = dev(muedges)
+    hiplib.check(lib.nbk_bin_power_f64(
+        hiplib.dptr(y_t), hiplib.i64_arr(geom.Nmesh),
+        hiplib.f64_arr(geom.BoxSize), hiplib.i64_arr(geom.cshape),
+        hiplib.i64_arr((0, 0, 0)), None,
+        hiplib.dptr(k2_t), len(kedges), hiplib.dptr(mu_t), len(muedges),
+        hiplib.f64_arr(los), hiplib.int_arr(_poles), Nell,
+        hiplib.dptr(sums), hiplib.dptr(sums[NB:]),
+        hiplib.dptr(sums[2 * NB:]), hiplib.dptr(sums[3 * NB:]), None),
+        'bin')
+    h = host(sums)
+    shape = (Nx + 2, Nmu + 2)
+    xsum = h[:NB].reshape(shape)
+    musum = h[NB:2 * NB].reshape(shape)
+    Nsum = numpy.round(h[2 * NB:3 * NB]).astype('i8').reshape(shape)
+    ys = h[3 * NB:].reshape(Nell, 2, NB)
+    ysum = (ys[:, 0] + 1j * ys[:, 1]).reshape((Nell,) + shape)
+
+    # fold + normalize like the host tail
+    ysum[..., -2] += ysum[..., -1]
+    musum[:, -2] += musum[:, -1]
+    xsum[:, -2] += xsum[:, -1]
+    Nsum[:, -2] += Nsum[:, -1]
+    sl = slice(1, -1)
+    with numpy.errstate(invalid='ignore', divide='ignore'):
+        y2d = (ysum[0] / Nsum)[sl, sl]
+        xm = (xsum / Nsum)[sl, sl]
+        mum = (musum / Nsum)[sl, sl]
+    xmean_2d, mumean_2d, want_y2d, want_N = want
+    assert_allclose(Nsum[sl, sl], want_N)
+    assert_allclose(numpy.nan_to_num(xm), numpy.nan_to_num(xmean_2d),
+                    rtol=1e-12, atol=1e-12)
+    assert_allclose(numpy.nan_to_num(mum), numpy.nan_to_num(mumean_2d),
+                    rtol=1e-12, atol=1e-12)
+    assert_allclose(numpy.nan_to_num(y2d), numpy.nan_to_num(want_y2d),
+                    rtol=1e-11, atol=1e-12)
+    if poles:
+        w_k, w_poles, w_N = want_poles
+        N_1d = Nsum[sl, sl].sum(axis=-1)
+        pole_arr = ysum[:, sl, sl].sum(axis=-1) / N_1d
+        for i, ell in enumerate(poles):
+            j = _poles.index(ell)
+            assert_allclose(numpy.nan_to_num(pole_arr[j]),
+                            numpy.nan_to_num(w_poles[i]),
+                            rtol=1e-11, atol=1e-12)
+
+
 def test_power3d_matches_oracle(lib):
     geom = MeshGeometry(16, 32.)
     rng = numpy.random.RandomState(11)
