@@ -147,8 +147,15 @@ __global__ void kfft_c2r_z(const double* __restrict__ cplx,
     const int64_t line = blockIdx.x;
 
     const cdouble* g = (const cdouble*)cplx + line * (m + 1);
-    for (int k = threadIdx.x; k <= m; k += blockDim.x)
-        xin[k] = g[k];
+    for (int k = threadIdx.x; k <= m; k += blockDim.x) {
+        cdouble v = g[k];
+        // FFTW/numpy c2r convention: the self-conjugate DC and Nyquist
+        // bins contribute only their real part (pmesh sits on FFTW; the
+        // interlaced combine feeds bins with nonzero imaginary parts
+        // there, source/mesh/catalog.py:341-351)
+        if (k == 0 || k == m) v.im = 0.0;
+        xin[k] = v;
+    }
     __syncthreads();
 
     // rebuild packed spectrum: Z[k] = E[k] + i * O[k],
