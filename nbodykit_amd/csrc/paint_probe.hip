@@ -1,0 +1,268 @@
+// Standalone paint-kernel ablation probe (not part of libnbk_hip).
+// Times variants of the CIC deposit on synthetic cell-ordered particles
+// (the C4 shape: ~1 particle/cell, 1024^3 f64 mesh) to find the
+// bottleneck: atomic serialization vs atomic throughput vs read path.
+//
+//   hipcc --offload-arch=gfx950 -O3 -munsafe-fp-atomics paint_probe.hip -o paint_probe
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cstdlib>
+#include <cmath>
+#include <vector>
+
+#define CHECK(x) do { hipError_t e=(x); if(e){printf("ERR %s %s\n",#x,hipGetErrorString(e)); exit(1);} } while(0)
+
+// mode 0: sorted uniform (1/cell); 1: sorted clumpy (20% of particles
+// in ~1000-deep single-cell clumps); 2: random shuffled
+__global__ void gen_sorted(double* px, double* py, double* pz, long n,
+                           long N, double H, unsigned seed, int mode)
+{
+    long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        // cell = i * ncells / n in C order, plus a cheap hash jitter
+        const double ratio = (double)N * N * N / (double)n;
+        long cell = (long)((double)i * ratio);
+        if (mode == 1) {
+            unsigned hc = (unsigned)(((unsigned long)(i >> 10)) * 2246822519u);
+            if ((hc % 5u) == 0u)  // whole 1024-particle block -> one cell
+                cell = (long)((double)((i >> 10) << 10) * ratio);
+        } else if (mode == 2) {
+            unsigned h1 = (unsigned)(i * 2654435761L) ^ 0x9e3779b9u;
+            long c2 = ((long)h1 << 18) ^ (i * 1103515245L);
+            cell = c2 % ((long)N * N * N);
+            if (cell < 0) cell += (long)N * N * N;
+        }
+        long iz = cell % N, iy = (cell / N) % N, ix = cell / (N * N);
+        unsigned h = (unsigned)(i * 2654435761u) ^ seed;
+        double f1 = (h & 1023) / 1024.0;
+        double f2 = ((h >> 10) & 1023) / 1024.0;
+        double f3 = ((h >> 20) & 1023) / 1024.0;
+        px[i] = (ix + f1) * H;
+        py[i] = (iy + f2) * H;
+        pz[i] = (iz + f3) * H;
+    }
+}
+
+__device__ __forceinline__ long wrapi(long i, long n) {
+    i %= n;
+    return i < 0 ? i + n : i;
+}
+
+// variant A: one thread per particle, 8 global f64 atomics
+template <int RUN, bool ATOMIC>
+__global__ void paint_cic(const double* __restrict__ px,
+                          const double* __restrict__ py,
+                          const double* __restrict__ pz, long n,
+                          long N, double invH, double* __restrict__ mesh)
+{
+    const long nruns = (n + RUN - 1) / RUN;
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long r = blockIdx.x * (long)blockDim.x + threadIdx.x;
+         r < nruns; r += stride) {
+        const long ibeg = (long)r * RUN;
+        const long iend = min(ibeg + RUN, n);
+        for (long i = ibeg; i < iend; i++) {
+            const double u0 = px[i] * invH, u1 = py[i] * invH,
+                         u2 = pz[i] * invH;
+            const double f0 = floor(u0), f1 = floor(u1), f2 = floor(u2);
+            const double a0 = u0 - f0, a1 = u1 - f1, a2 = u2 - f2;
+            const long b0 = (long)f0, b1 = (long)f1, b2 = (long)f2;
+            const double w0[2] = {1.0 - a0, a0};
+            const double w1[2] = {1.0 - a1, a1};
+            const double w2[2] = {1.0 - a2, a2};
+            for (int dx = 0; dx < 2; dx++) {
+                const long gx = wrapi(b0 + dx, N);
+                for (int dy = 0; dy < 2; dy++) {
+                    const long gy = wrapi(b1 + dy, N);
+                    const double wxy = w0[dx] * w1[dy];
+                    for (int dz = 0; dz < 2; dz++) {
+                        const long gz = wrapi(b2 + dz, N);
+                        double* p = &mesh[(gx * N + gy) * N + gz];
+                        if (ATOMIC) atomicAdd(p, wxy * w2[dz]);
+                        else *p += wxy * w2[dz];
+                    }
+                }
+            }
+        }
+    }
+}
+
+// variant: stride mapping + wave-segmented merge of equal deposit
+// addresses (sorted input puts same-cell particles in adjacent lanes)
+__global__ void paint_cic_merge(const double* __restrict__ px,
+                                const double* __restrict__ py,
+                                const double* __restrict__ pz, long n,
+                                long N, double invH,
+                                double* __restrict__ mesh)
+{
+    const long stride = (long)gridDim.x * blockDim.x;
+    const int lane = threadIdx.x & 63;
+    for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        const double u0 = px[i] * invH, u1 = py[i] * invH,
+                     u2 = pz[i] * invH;
+        const double f0 = floor(u0), f1 = floor(u1), f2 = floor(u2);
+        const double a0 = u0 - f0, a1 = u1 - f1, a2 = u2 - f2;
+        const long b0 = (long)f0, b1 = (long)f1, b2 = (long)f2;
+        const double w0[2] = {1.0 - a0, a0};
+        const double w1[2] = {1.0 - a1, a1};
+        const double w2[2] = {1.0 - a2, a2};
+        for (int dx = 0; dx < 2; dx++) {
+            const long gx = wrapi(b0 + dx, N);
+            for (int dy = 0; dy < 2; dy++) {
+                const long gy = wrapi(b1 + dy, N);
+                const double wxy = w0[dx] * w1[dy];
+                for (int dz = 0; dz < 2; dz++) {
+                    const long gz = wrapi(b2 + dz, N);
+                    long addr = (gx * N + gy) * N + gz;
+                    double val = wxy * w2[dz];
+                    // segmented inclusive prefix over equal-addr runs
+                    #pragma unroll
+                    for (int d = 1; d < 64; d <<= 1) {
+                        long a_up = __shfl_up((long long)addr, d, 64);
+                        double v_up = __shfl_up(val, d, 64);
+                        if (lane >= d && a_up == addr) val += v_up;
+                    }
+                    long a_dn = __shfl_down((long long)addr, 1, 64);
+                    if (lane == 63 || a_dn != addr)
+                        atomicAdd(&mesh[addr], val);
+                }
+            }
+        }
+    }
+}
+
+// variant: read-only (bandwidth leg)
+__global__ void read_only(const double* __restrict__ px,
+                          const double* __restrict__ py,
+                          const double* __restrict__ pz, long n,
+                          double* __restrict__ sink)
+{
+    double acc = 0;
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        acc += px[i] + py[i] + pz[i];
+    if (acc == 12345.6789) *sink = acc;   // never true; defeats DCE
+}
+
+// variant: z-column register accumulation — each thread walks RUN
+// consecutive (sorted) particles keeping a [2][2] x (z,z+1) window in
+// registers, flushing atomics only when the base cell moves.
+template <int RUN>
+__global__ void paint_cic_window(const double* __restrict__ px,
+                                 const double* __restrict__ py,
+                                 const double* __restrict__ pz, long n,
+                                 long N, double invH,
+                                 double* __restrict__ mesh)
+{
+    const long nruns = (n + RUN - 1) / RUN;
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long r = blockIdx.x * (long)blockDim.x + threadIdx.x;
+         r < nruns; r += stride) {
+        const long ibeg = (long)r * RUN;
+        const long iend = min(ibeg + RUN, n);
+        long cb0 = -1, cb1 = -1, cb2 = -1;     // current window base
+        double acc[2][2][2] = {};
+        for (long i = ibeg; i < iend; i++) {
+            const double u0 = px[i] * invH, u1 = py[i] * invH,
+                         u2 = pz[i] * invH;
+            const double f0 = floor(u0), f1 = floor(u1), f2 = floor(u2);
+            const double a0 = u0 - f0, a1 = u1 - f1, a2 = u2 - f2;
+            const long b0 = (long)f0, b1 = (long)f1, b2 = (long)f2;
+            if (b0 != cb0 || b1 != cb1 || b2 != cb2) {
+                if (cb0 >= 0) {
+                    for (int dx = 0; dx < 2; dx++)
+                        for (int dy = 0; dy < 2; dy++)
+                            for (int dz = 0; dz < 2; dz++)
+                                if (acc[dx][dy][dz] != 0.0)
+                                    atomicAdd(&mesh[(wrapi(cb0+dx,N)*N
+                                        + wrapi(cb1+dy,N))*N
+                                        + wrapi(cb2+dz,N)],
+                                        acc[dx][dy][dz]);
+                }
+                for (int dx = 0; dx < 2; dx++)
+                    for (int dy = 0; dy < 2; dy++)
+                        for (int dz = 0; dz < 2; dz++)
+                            acc[dx][dy][dz] = 0.0;
+                cb0 = b0; cb1 = b1; cb2 = b2;
+            }
+            const double w0[2] = {1.0 - a0, a0};
+            const double w1[2] = {1.0 - a1, a1};
+            const double w2[2] = {1.0 - a2, a2};
+            for (int dx = 0; dx < 2; dx++)
+                for (int dy = 0; dy < 2; dy++)
+                    for (int dz = 0; dz < 2; dz++)
+                        acc[dx][dy][dz] += w0[dx] * w1[dy] * w2[dz];
+        }
+        if (cb0 >= 0)
+            for (int dx = 0; dx < 2; dx++)
+                for (int dy = 0; dy < 2; dy++)
+                    for (int dz = 0; dz < 2; dz++)
+                        if (acc[dx][dy][dz] != 0.0)
+                            atomicAdd(&mesh[(wrapi(cb0+dx,N)*N
+                                + wrapi(cb1+dy,N))*N + wrapi(cb2+dz,N)],
+                                acc[dx][dy][dz]);
+    }
+}
+
+int grid_for(long work) {
+    long g = (work + 255) / 256;
+    if (g > 1048576) g = 1048576;
+    return (int)(g < 1 ? 1 : g);
+}
+
+template <typename F>
+double timeit(F&& launch, int iters) {
+    hipEvent_t a, b;
+    CHECK(hipEventCreate(&a));
+    CHECK(hipEventCreate(&b));
+    launch();                      // warm
+    CHECK(hipDeviceSynchronize());
+    CHECK(hipEventRecord(a));
+    for (int it = 0; it < iters; it++) launch();
+    CHECK(hipEventRecord(b));
+    CHECK(hipEventSynchronize(b));
+    float ms;
+    CHECK(hipEventElapsedTime(&ms, a, b));
+    return ms / iters;
+}
+
+int main(int argc, char** argv) {
+    const long N = argc > 1 ? atol(argv[1]) : 1024;
+    const long n = argc > 2 ? atol(argv[2]) : 1000000000L;
+    const double L = 5000.0, H = L / N, invH = N / L;
+
+    double *px, *py, *pz, *mesh, *sink;
+    CHECK(hipMalloc(&px, n * 8));
+    CHECK(hipMalloc(&py, n * 8));
+    CHECK(hipMalloc(&pz, n * 8));
+    CHECK(hipMalloc(&mesh, N * N * N * 8));
+    CHECK(hipMalloc(&sink, 8));
+
+    const double GB = n * 152.0 / 1e9;   // algorithmic CIC bytes
+    auto report = [&](const char* name, double ms) {
+        printf("%-28s %8.2f ms  %7.1f GB/s(algo)  %6.2f Gpart/s\n",
+               name, ms, GB / ms * 1000.0 / 1.0, n / ms / 1e6 / 1000.0);
+    };
+
+    const char* modenames[3] = {"sorted-uniform", "sorted-clumpy", "random"};
+    for (int mode = 0; mode < 3; mode++) {
+        hipLaunchKernelGGL(gen_sorted, dim3(grid_for(n)), dim3(256), 0, 0,
+                           px, py, pz, n, N, H, 12345u, mode);
+        CHECK(hipDeviceSynchronize());
+        char buf[128];
+        snprintf(buf, 128, "atomic RUN=1 [%s]", modenames[mode]);
+        report(buf, timeit([&] {
+            hipLaunchKernelGGL((paint_cic<1, true>), dim3(grid_for(n)),
+                               dim3(256), 0, 0, px, py, pz, n, N, invH, mesh);
+        }, 3));
+        snprintf(buf, 128, "wave-merge [%s]", modenames[mode]);
+        report(buf, timeit([&] {
+            hipLaunchKernelGGL(paint_cic_merge, dim3(grid_for(n)),
+                               dim3(256), 0, 0, px, py, pz, n, N, invH, mesh);
+        }, 3));
+    }
+    return 0;
+}
