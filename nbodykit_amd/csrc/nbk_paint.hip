@@ -30,14 +30,25 @@ namespace {
 __device__ __forceinline__ void merged_deposit(double* __restrict__ mesh,
                                                long long addr, double val,
                                                int lane) {
+    // segment = maximal CONTIGUOUS run of equal addresses.  A plain
+    // "same addr at distance d" guard would also fold non-adjacent
+    // duplicates ([A A B A] merging the stray A into the first run);
+    // the head-bounded Hillis-Steele scan below only sums within the
+    // lane's own run.
+    const long long a_up1 = __shfl_up(addr, 1, 64);
+    const bool head = (lane == 0) || (a_up1 != addr);
+    const unsigned long long heads = __ballot(head);
+    const unsigned long long below = heads
+        & (~0ULL >> (63 - lane));          // head bits at lanes <= lane
+    const int myhead = 63 - __clzll(below);
     #pragma unroll
     for (int d = 1; d < 64; d <<= 1) {
-        const long long a_up = __shfl_up(addr, d, 64);
         const double v_up = __shfl_up(val, d, 64);
-        if (lane >= d && a_up == addr) val += v_up;
+        if (lane - d >= myhead) val += v_up;
     }
-    const long long a_dn = __shfl_down(addr, 1, 64);
-    if (addr >= 0 && (lane == 63 || a_dn != addr))
+    const bool next_head = ((lane < 63)
+                            && ((heads >> (lane + 1)) & 1ULL));
+    if (addr >= 0 && (lane == 63 || next_head))
         atomicAdd(&mesh[addr], val);
 }
 
