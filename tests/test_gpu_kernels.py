@@ -285,3 +285,68 @@ def test_power3d_matches_oracle(lib):
         float(numpy.prod(geom.BoxSize)), hiplib.i64_arr(geom.cshape),
         hiplib.i64_arr((0, 0, 0)), 1, None), 'power3d')
     assert_allclose(host(out_t), want, rtol=1e-12, atol=1e-12)
+
+
+# ---- partitioned layouts (the multi-GPU slab/pencil paths) -------------
+
+def test_bin_partitioned_matches_full(lib):
+    """binning two y-chunks with off=(0,y0,0) and summing == binning the
+    full field (the N>1 layout at fftpower.py:669-672)"""
+    geom = MeshGeometry(16, 32.)
+    rng = numpy.random.RandomState(13)
+    y3d = (rng.normal(size=geom.cshape)
+           + 1j * rng.normal(size=geom.cshape)).astype('c16')
+    kedges = numpy.arange(0., numpy.pi * 16 / 32. + 0.3, 0.15)
+    muedges = numpy.linspace(-1, 1, 4)
+    Nx = len(kedges) - 1
+    NB = (Nx + 2) * (len(muedges) - 1 + 2)
+    nf = 3 + 2
+
+    def run_bin(block, off):
+        sums = torch.zeros(nf * NB, dtype=torch.float64, device='cuda')
+        b_t = dev(block)
+        k2_t = dev(kedges ** 2)
+        mu_t = dev(muedges)
+        hiplib.check(lib.nbk_bin_power_f64(
+            hiplib.dptr(b_t), hiplib.i64_arr(geom.Nmesh),
+            hiplib.f64_arr(geom.BoxSize),
+            hiplib.i64_arr(block.shape), hiplib.i64_arr(off), None,
+            hiplib.dptr(k2_t), len(kedges), hiplib.dptr(mu_t),
+            len(muedges), hiplib.f64_arr((0, 0, 1)),
+            hiplib.int_arr([0]), 1,
+            hiplib.dptr(sums), hiplib.dptr(sums[NB:]),
+            hiplib.dptr(sums[2 * NB:]), hiplib.dptr(sums[3 * NB:]),
+            None), 'bin')
+        return host(sums)
+
+    full = run_bin(y3d, (0, 0, 0))
+    # partition along x (slab) and along y (transposed pencil layout)
+    hx = run_bin(y3d[:8], (0, 0, 0)) + run_bin(y3d[8:], (8, 0, 0))
+    hy = run_bin(numpy.ascontiguousarray(y3d[:, :8]), (0, 0, 0)) \
+        + run_bin(numpy.ascontiguousarray(y3d[:, 8:]), (0, 8, 0))
+    assert_allclose(hx, full, rtol=1e-12, atol=1e-12)
+    assert_allclose(hy, full, rtol=1e-12, atol=1e-12)
+
+
+def test_compensate_partitioned_matches_full(lib):
+    geom = MeshGeometry(16, 32.)
+    rng = numpy.random.RandomState(14)
+    c = (rng.normal(size=geom.cshape)
+         + 1j * rng.normal(size=geom.cshape)).astype('c16')
+
+    full_t = dev(c)
+    hiplib.check(lib.nbk_compensate_f64(
+        hiplib.dptr(full_t), hiplib.i64_arr(geom.Nmesh),
+        hiplib.i64_arr(geom.cshape), hiplib.i64_arr((0, 0, 0)), None,
+        0, 0, None), 'c')
+    want = host(full_t)
+
+    got = numpy.empty_like(c)
+    for y0 in (0, 8):
+        blk = dev(numpy.ascontiguousarray(c[:, y0:y0 + 8]))
+        hiplib.check(lib.nbk_compensate_f64(
+            hiplib.dptr(blk), hiplib.i64_arr(geom.Nmesh),
+            hiplib.i64_arr(blk.shape), hiplib.i64_arr((0, y0, 0)), None,
+            0, 0, None), 'c')
+        got[:, y0:y0 + 8] = host(blk)
+    assert_allclose(got, want, rtol=1e-14, atol=1e-14)
