@@ -389,8 +389,8 @@ def run_cpu_baseline(cfg):
     bounded sample of the same workload shape: ~1 particle/cell like C4,
     sized for tens of CPU-seconds."""
     from oracle import fftpower_oracle
-    n = int(2e6)
-    nmesh = 128
+    n = int(2e7)
+    nmesh = 256
     box = cfg['box'] * nmesh / cfg['nmesh']
     rng = numpy.random.RandomState(42)
     pos = rng.uniform(0, box, size=(n, 3))
