@@ -88,6 +88,24 @@ int nbk_paint_f64(const double* pos, const double* mass, int64_t n,
                   double* mesh, int64_t x0, int64_t nx_local,
                   void* stream);
 
+/* paint locality sort --------------------------------------------------
+ * Two-pass counting sort of particles by coarse mesh cell
+ * (bucket = wrapped ix * n1 + iy): count, then (after the caller turns
+ * counts into exclusive offsets) scatter into SoA output.  Replaces a
+ * general radix sort in the paint driver: the deposit kernel wants
+ * bucket-local order, not a total order.  `pos_aos` is the (n,3)
+ * row-major input; `offsets` (n0*n1 int64) is consumed (atomically
+ * advanced) by the scatter.  mass may be NULL.
+ */
+int nbk_bucket_count_f64(const double* pos_aos, int64_t n,
+                         const int64_t nmesh[3], const double box[3],
+                         int* counts, void* stream);
+int nbk_bucket_scatter_f64(const double* pos_aos, const double* mass,
+                           int64_t n, const int64_t nmesh[3],
+                           const double box[3], int64_t* offsets,
+                           double* pos_soa_out, double* mass_out,
+                           void* stream);
+
 /* readout (gather dual of paint; resampler='nnb' serves the LogNormal
  * generator, mockmaker.py:317-319; cic for FFTRecon later) */
 int nbk_readout_nnb_f64(const double* pos, int64_t n,
