@@ -190,7 +190,19 @@ extern "C" int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
     if (g < 1) g = 1;
 
     hipStream_t s = (hipStream_t)stream;
-    if (lds_bytes <= 64 * 1024) {
+    // gfx950 has 160 KiB LDS per CU; dynamic allocations above the 64 KiB
+    // default need the attribute raised once.  The per-block histogram is
+    // worth ~1 occupancy: the global-atomic fallback costs seconds at
+    // 1024^3 x Nmu=5 (every element 5+ HBM atomics).
+    if (lds_bytes <= 160 * 1024) {
+        static size_t raised = 0;
+        if (lds_bytes > 64 * 1024 && lds_bytes > raised) {
+            (void)hipFuncSetAttribute(
+                reinterpret_cast<const void*>(&kbin<true>),
+                hipFuncAttributeMaxDynamicSharedMemorySize,
+                (int)lds_bytes);
+            raised = lds_bytes;
+        }
         hipLaunchKernelGGL(kbin<true>, dim3((uint32_t)g), dim3(256),
                            lds_bytes, s, cplx, A, kedges, muedges, xsum);
     } else {

@@ -1,42 +1,46 @@
-// Counting bucket-sort of particles by coarse mesh cell (ix, iy) — the
+// Counting bucket-sort of particles by mesh cell — the
 // paint locality pass.  The deposit kernel (nbk_paint.hip) is ~4x
 // faster when nearby-in-space particles are nearby-in-memory; a full
 // radix sort (rocprim via torch.argsort) costs ~70 ms at 1e9 particles
 // where this two-pass counting sort needs only bucket-local order.
 // Non-stable within a bucket (ticket = atomic fetch-add), which is fine:
-// a bucket is one (ix, iy) z-line whose deposits fit in a few KB of L1.
+// all particles of one cell still land in one contiguous output range,
+// so the deposit kernel's wave-merge sees clumps as contiguous runs.
 // Output is SoA (x[n] y[n] z[n]) — exactly the layout nbk_paint_f64
 // reads — so the driver also saves its AoS->SoA transpose pass.
 #include "nbk_common.h"
 
 namespace {
 
-__device__ __forceinline__ int64_t bucket_of(double x, double y,
+__device__ __forceinline__ int64_t bucket_of(double x, double y, double z,
                                              double invH0, double invH1,
-                                             int64_t n0, int64_t n1) {
+                                             double invH2, int64_t n0,
+                                             int64_t n1, int64_t n2) {
     const int64_t ix = wrap_idx((int64_t)floor(x * invH0), n0);
     const int64_t iy = wrap_idx((int64_t)floor(y * invH1), n1);
-    return ix * n1 + iy;
+    const int64_t iz = wrap_idx((int64_t)floor(z * invH2), n2);
+    return (ix * n1 + iy) * n2 + iz;
 }
 
 __global__ void kbucket_count(const double* __restrict__ pos, int64_t n,
-                              int64_t n0, int64_t n1,
-                              double invH0, double invH1,
+                              int64_t n0, int64_t n1, int64_t n2,
+                              double invH0, double invH1, double invH2,
                               int* __restrict__ counts)
 {
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
          i < n; i += stride) {
         const int64_t b = bucket_of(pos[3 * i], pos[3 * i + 1],
-                                    invH0, invH1, n0, n1);
+                                    pos[3 * i + 2], invH0, invH1, invH2,
+                                    n0, n1, n2);
         atomicAdd(&counts[b], 1);
     }
 }
 
 __global__ void kbucket_scatter(const double* __restrict__ pos,
                                 const double* __restrict__ mass, int64_t n,
-                                int64_t n0, int64_t n1,
-                                double invH0, double invH1,
+                                int64_t n0, int64_t n1, int64_t n2,
+                                double invH0, double invH1, double invH2,
                                 int64_t* __restrict__ offsets,
                                 double* __restrict__ ox,
                                 double* __restrict__ oy,
@@ -47,7 +51,8 @@ __global__ void kbucket_scatter(const double* __restrict__ pos,
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
          i < n; i += stride) {
         const double x = pos[3 * i], y = pos[3 * i + 1], z = pos[3 * i + 2];
-        const int64_t b = bucket_of(x, y, invH0, invH1, n0, n1);
+        const int64_t b = bucket_of(x, y, z, invH0, invH1, invH2,
+                                    n0, n1, n2);
         const int64_t t = atomicAdd((unsigned long long*)&offsets[b],
                                     (unsigned long long)1);
         ox[t] = x;
@@ -73,8 +78,10 @@ extern "C" int nbk_bucket_count_f64(const double* pos_aos, int64_t n,
 {
     if (n == 0) return NBK_OK;
     hipLaunchKernelGGL(kbucket_count, dim3(sgrid(n)), dim3(256), 0,
-                       (hipStream_t)stream, pos_aos, n, nmesh[0], nmesh[1],
-                       nmesh[0] / box[0], nmesh[1] / box[1], counts);
+                       (hipStream_t)stream, pos_aos, n,
+                       nmesh[0], nmesh[1], nmesh[2],
+                       nmesh[0] / box[0], nmesh[1] / box[1],
+                       nmesh[2] / box[2], counts);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }
@@ -90,8 +97,9 @@ extern "C" int nbk_bucket_scatter_f64(const double* pos_aos,
     if (n == 0) return NBK_OK;
     hipLaunchKernelGGL(kbucket_scatter, dim3(sgrid(n)), dim3(256), 0,
                        (hipStream_t)stream, pos_aos, mass, n,
-                       nmesh[0], nmesh[1],
-                       nmesh[0] / box[0], nmesh[1] / box[1], offsets,
+                       nmesh[0], nmesh[1], nmesh[2],
+                       nmesh[0] / box[0], nmesh[1] / box[1],
+                       nmesh[2] / box[2], offsets,
                        pos_soa_out, pos_soa_out + n, pos_soa_out + 2 * n,
                        mass_out);
     NBK_CHECK_HIP(hipGetLastError());

@@ -73,10 +73,11 @@ def _prepare_particles(pos_t, mass_t, pm):
         return pos_t.t().contiguous(), mass_t
 
     lib = hiplib.require()
-    n0, n1 = int(pm.Nmesh[0]), int(pm.Nmesh[1])
+    n0, n1, n2 = (int(x) for x in pm.Nmesh)
+    ncells = n0 * n1 * n2
     invH = torch.as_tensor((pm.Nmesh / pm.BoxSize)).to(pos_t.device)
-    u = torch.floor(pos_t[:, :2] * invH[:2]).long()
-    key = u[:, 0] * n1 + u[:, 1]
+    u = torch.floor(pos_t * invH).long()
+    key = (u[:, 0] * n1 + u[:, 1]) * n2 + u[:, 2]
     mono = (key[1:] >= key[:-1]).float().mean().item()
     if mono >= _SORT_MONO_THRESHOLD:
         return pos_t.t().contiguous(), mass_t
@@ -84,12 +85,12 @@ def _prepare_particles(pos_t, mass_t, pm):
     nmesh = hiplib.i64_arr(pm.Nmesh)
     box = hiplib.f64_arr(pm.BoxSize)
     stream = hiplib.cur_stream()
-    counts = torch.zeros(n0 * n1, dtype=torch.int32, device='cuda')
+    counts = torch.zeros(ncells, dtype=torch.int32, device='cuda')
     pos_in = pos_t.contiguous()
     hiplib.check(lib.nbk_bucket_count_f64(
         hiplib.dptr(pos_in), n, nmesh, box, hiplib.dptr(counts), stream),
         'nbk_bucket_count_f64')
-    offsets = torch.zeros(n0 * n1, dtype=torch.int64, device='cuda')
+    offsets = torch.zeros(ncells, dtype=torch.int64, device='cuda')
     offsets[1:] = torch.cumsum(counts[:-1].long(), 0)  # exclusive prefix
     out_soa = torch.empty(3 * n, dtype=torch.float64, device='cuda')
     out_mass = None
