@@ -171,6 +171,10 @@ int nbk_power3d_f64(double* out, const double* c1, const double* c2,
  * meshtools.py:188-215).  `ells` lists the multipole orders (first
  * entry must be 0, mirroring fftpower.py:585).  los is the
  * line-of-sight unit vector.  Outputs are accumulated (caller zeroes).
+ * kedges must hold SQUARED bin edges (the digitize runs on |k|^2 / r^2,
+ * fftpower.py:578,615).  real_field = 1 bins a configuration-space
+ * RealField instead (FFTCorr): relative-position coordinates, all axes
+ * full length, no Hermitian double-count (fftcorr.py:150-176).
  */
 int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
                       const double box[3],
@@ -180,6 +184,7 @@ int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
                       const double* muedges, int64_t nmu_edges,
                       const double los[3],
                       const int* ells, int nell,
+                      int real_field,
                       double* xsum, double* musum, double* Nsum,
                       double* ysum, void* stream);
 

@@ -1,1 +1,2 @@
 from .fftpower import FFTPower, FFTBase, project_to_basis
+from .fftcorr import FFTCorr

@@ -228,14 +228,16 @@ class FFTPower(FFTBase):
 
 def project_to_basis(y3d, edges, los=[0, 0, 1], poles=[]):
     """
-    Project a 3D complex statistic onto (x, mu) bins and multipoles
-    (reference :507-701), with the sums done by the
+    Project a 3D statistic (ComplexField in k-space, or RealField in
+    configuration space — the FFTCorr case) onto (x, mu) bins and
+    multipoles (reference :507-701), with the sums done by the
     ``nbk_bin_power_f64`` HIP kernel and the tiny fold/normalize tail on
     host after one small allreduce (the reference's :669-679).
     """
     import torch
     comm = y3d.pm.comm
     lib = hiplib.require()
+    real_field = isinstance(y3d, RealField)
 
     xedges, muedges = edges
     Nx = len(xedges) - 1
@@ -259,13 +261,21 @@ def project_to_basis(y3d, edges, los=[0, 0, 1], poles=[]):
     muedges_t = torch.as_tensor(numpy.asarray(muedges, dtype='f8')).to(dev)
     sums = torch.zeros(nfields * NB, dtype=torch.float64, device=dev)
 
+    if real_field:
+        dims = tuple(int(d) for d in y3d.value.shape)
+        off = (y3d.x_start, 0, 0)
+    else:
+        dims = y3d.dims
+        off = y3d.off
+
     hiplib.check(lib.nbk_bin_power_f64(
         hiplib.dptr(y3d.value), hiplib.i64_arr(y3d.pm.Nmesh),
         hiplib.f64_arr(y3d.pm.BoxSize),
-        hiplib.i64_arr(y3d.dims), hiplib.i64_arr(y3d.off), None,
+        hiplib.i64_arr(dims), hiplib.i64_arr(off), None,
         hiplib.dptr(k2edges_t), len(xedges),
         hiplib.dptr(muedges_t), len(muedges),
         hiplib.f64_arr(los), hiplib.int_arr(_poles), Nell,
+        int(real_field),
         hiplib.dptr(sums), hiplib.dptr(sums[NB:]),
         hiplib.dptr(sums[2 * NB:]), hiplib.dptr(sums[3 * NB:]),
         hiplib.cur_stream()), 'nbk_bin_power_f64')
@@ -280,7 +290,12 @@ def project_to_basis(y3d, edges, los=[0, 0, 1], poles=[]):
     Nsum = numpy.round(host[2 * NB:3 * NB]).astype('i8').reshape(shape)
     # kernel layout is planar: [y0.re | y0.im | y1.re | ...] (NB each)
     ys = host[3 * NB:].reshape(Nell, 2, NB)
-    ysum = (ys[:, 0] + 1j * ys[:, 1]).reshape((Nell,) + shape)
+    if real_field:
+        # real statistic: ysum carries no imaginary part
+        # (reference ysum dtype follows y3d.dtype, fftpower.py:597)
+        ysum = ys[:, 0].reshape((Nell,) + shape).copy()
+    else:
+        ysum = (ys[:, 0] + 1j * ys[:, 1]).reshape((Nell,) + shape)
 
     # fold the internal mu == 1 bin into the last visible bin (:674-679)
     ysum[..., -2] += ysum[..., -1]

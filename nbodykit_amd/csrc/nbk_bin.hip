@@ -26,6 +26,10 @@ struct BinArgs {
     double losx, losy, losz;
     int nx_edges, nmu_edges;   // edge COUNTS (bins + 1)
     int nell;
+    int real_field;            // 1: y3d is a REAL configuration-space
+                               // field (FFTCorr, fftcorr.py:150-176):
+                               // coords are relative positions, all axes
+                               // full-length, no Hermitian double-count
     int ells[NBK_MAX_ELL];     // multipole orders, by value (kernel args
                                // live in SGPRs — no host pointer deref)
 };
@@ -89,7 +93,8 @@ __global__ void kbin(const double* __restrict__ data, BinArgs A,
         g[A.a2] = l2 + A.o2;
         const double fx = freq_full(g[0], A.n0);
         const double fy = freq_full(g[1], A.n1);
-        const double fz = freq_half(g[2], A.n2);
+        const double fz = A.real_field ? freq_full(g[2], A.n2)
+                                       : freq_half(g[2], A.n2);
 
         const double kx = fx * A.k0x, ky = fy * A.k0y, kz = fz * A.k0z;
         const double k2 = kx * kx + ky * ky + kz * kz;
@@ -97,14 +102,16 @@ __global__ void kbin(const double* __restrict__ data, BinArgs A,
         double mu = kx * A.losx + ky * A.losy + kz * A.losz;
         mu = (kmag == 0.0) ? 0.0 : mu / kmag;
 
-        const bool nonsingular = fz > 0.0;      // doubled modes
+        const bool nonsingular = !A.real_field && fz > 0.0;  // doubled
         const double w = nonsingular ? 2.0 : 1.0;
 
         const int bx = dig(k2edges, A.nx_edges, k2);
         const int bmu = dig(muedges, A.nmu_edges, mu);
         const int bin = bx * (A.nmu_edges + 1) + bmu;
 
-        const cdouble v = {data[2 * idx], data[2 * idx + 1]};
+        const cdouble v = A.real_field
+            ? cdouble{data[idx], 0.0}
+            : cdouble{data[2 * idx], data[2 * idx + 1]};
 
         // Legendre P_ell(mu) by recurrence, ells ascending with ells[0]==0
         cdouble yv[NBK_MAX_ELL];
@@ -148,6 +155,7 @@ extern "C" int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
                                  const double* muedges, int64_t nmu_edges,
                                  const double los[3],
                                  const int* ells, int nell,
+                                 int real_field,
                                  double* xsum, double* musum, double* Nsum,
                                  double* ysum, void* stream)
 {
@@ -172,9 +180,18 @@ extern "C" int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
     A.o0 = off[0]; A.o1 = off[1]; A.o2 = off[2];
     if (axis_map) { A.a0 = axis_map[0]; A.a1 = axis_map[1]; A.a2 = axis_map[2]; }
     else { A.a0 = 0; A.a1 = 1; A.a2 = 2; }
-    A.k0x = 2.0 * M_PI / box[0];
-    A.k0y = 2.0 * M_PI / box[1];
-    A.k0z = 2.0 * M_PI / box[2];
+    // coordinate scale per axis: wavenumber 2 pi f / L for the complex
+    // field, relative position f * H for the real field
+    if (real_field) {
+        A.k0x = box[0] / nmesh[0];
+        A.k0y = box[1] / nmesh[1];
+        A.k0z = box[2] / nmesh[2];
+    } else {
+        A.k0x = 2.0 * M_PI / box[0];
+        A.k0y = 2.0 * M_PI / box[1];
+        A.k0z = 2.0 * M_PI / box[2];
+    }
+    A.real_field = real_field;
     A.losx = los[0]; A.losy = los[1]; A.losz = los[2];
     A.nx_edges = (int)nx_edges;
     A.nmu_edges = (int)nmu_edges;

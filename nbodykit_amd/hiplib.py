@@ -79,8 +79,8 @@ def _declare(lib):
     lib.nbk_bin_power_f64.argtypes = [c_void, c_i64_p, c_f64_p, c_i64_p,
                                       c_i64_p, c_int_p, c_void, c_i64,
                                       c_void, c_i64, c_f64_p, c_int_p,
-                                      ctypes.c_int, c_void, c_void, c_void,
-                                      c_void, c_void]
+                                      ctypes.c_int, ctypes.c_int, c_void,
+                                      c_void, c_void, c_void, c_void]
     lib.nbk_axpy_f64.restype = ctypes.c_int
     lib.nbk_axpy_f64.argtypes = [c_void, c_void, c_f64, c_i64, c_void]
     lib.nbk_scale_f64.restype = ctypes.c_int

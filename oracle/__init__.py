@@ -22,4 +22,5 @@ from .mesh import MeshGeometry, r2c, c2r, complex_coords, real_coords
 from .paint import paint, WINDOW_SUPPORT
 from .catalogmesh import to_real_field
 from .fftpower import (compensation_filter, apply_compensation,
-                      compute_3d_power, project_to_basis, fftpower_oracle)
+                      compute_3d_power, project_to_basis,
+                      fftpower_oracle, fftcorr_oracle)

@@ -9,7 +9,8 @@ oracle and the CPU baseline.
 import numpy
 from scipy.special import legendre
 
-from .mesh import MeshGeometry, r2c, complex_coords, complex_circular_coords
+from .mesh import (MeshGeometry, r2c, c2r, complex_coords,
+                   complex_circular_coords, real_coords)
 from .catalogmesh import to_real_field
 
 
@@ -104,15 +105,20 @@ def kedges_unique(geom, kmax=None):
     return edges, fx
 
 
-def project_to_basis(y3d, geom, edges, los=(0, 0, 1), poles=()):
+def project_to_basis(y3d, geom, edges, los=(0, 0, 1), poles=(),
+                     coords=None, hermitian=True):
     """
-    Restates fftpower.py:507-701 for a single-process compressed complex
-    field: iterate y-z slabs along axis 0, digitize k^2 and mu, apply
+    Restates fftpower.py:507-701 for a single-process field: iterate
+    y-z slabs along axis 0, digitize the coordinate norm^2 and mu, apply
     Hermitian double-count weights along the compressed (last) axis
     (Nyquist/DC excluded: meshtools.py:144-215), Legendre-weight the
     requested multipoles with the odd/even conjugate-pair identity
     (:649-656), bincount into (Nx+2, Nmu+2) arrays, fold the internal
     mu==1 bin into the last visible bin (:674-679).
+
+    Default coords are the compressed wavenumber grid (FFTPower);
+    ``coords=real_coords(geom), hermitian=False`` bins a real
+    configuration-space field (FFTCorr, fftcorr.py:150-176).
 
     Returns (result, pole_result) with the same contents as the reference.
     """
@@ -135,13 +141,19 @@ def project_to_basis(y3d, geom, edges, los=(0, 0, 1), poles=()):
     ysum = numpy.zeros((Nell, Nx + 2, Nmu + 2), dtype=y3d.dtype)
     Nsum = numpy.zeros((Nx + 2, Nmu + 2), dtype='i8')
 
-    coords = complex_coords(geom)
+    if coords is None:
+        coords = complex_coords(geom)
     cy = numpy.take(coords[1], 0, axis=0)      # (Ny,1)
     cz = numpy.take(coords[2], 0, axis=0)      # (1,Nzh)
 
     # Hermitian weights for an axis-0 slab: 2 on modes with positive
-    # compressed-axis frequency, else 1 (meshtools.py:188-215)
-    nonsingular = numpy.broadcast_to(cz > 0., (cy.shape[0], cz.shape[1]))
+    # compressed-axis frequency, else 1 (meshtools.py:188-215);
+    # everything weight-1 for a non-compressed (real) field
+    if hermitian:
+        nonsingular = numpy.broadcast_to(cz > 0.,
+                                         (cy.shape[0], cz.shape[1]))
+    else:
+        nonsingular = numpy.zeros((cy.shape[0], cz.shape[1]), dtype=bool)
     hw = numpy.ones(nonsingular.shape, dtype='f8')
     hw[nonsingular] = 2.
 
@@ -172,7 +184,9 @@ def project_to_basis(y3d, geom, edges, los=(0, 0, 1), poles=()):
         for iell, ell in enumerate(_poles):
             weighted = legpoly[iell](mu) * y3d[islab]
             # conjugate-pair identity on the doubled modes (:649-656)
-            if ell % 2:
+            if not hermitian:
+                pass
+            elif ell % 2:
                 weighted = 1j * numpy.where(nonsingular,
                                             2. * weighted.imag,
                                             weighted.imag) \
@@ -299,6 +313,81 @@ def fftpower_oracle(position, Nmesh, BoxSize, mode='1d', second_position=None,
     if pole_result is not None:
         pole_k, pole_arr, pole_modes = pole_result
         out['pole_k'] = pole_k
+        out['poles'] = {ell: pole_arr[i] for i, ell in enumerate(poles)}
+        out['pole_modes'] = pole_modes
+    else:
+        out['poles'] = None
+    return out
+
+
+def fftcorr_oracle(position, Nmesh, BoxSize, mode='1d',
+                   second_position=None, weight=None, second_weight=None,
+                   resampler='cic', compensated=True, interlaced=False,
+                   los=(0, 0, 1), Nmu=5, dr=None, rmin=0., rmax=None,
+                   poles=(), paint_chunk_size=4 * 1024 * 1024):
+    """
+    End-to-end FFTCorr oracle (reference algorithms/fftcorr.py:15-235):
+    the 3D power transformed back with c2r and divided by V (:151-158),
+    binned in configuration space with mu in [0, 1] (:175) and
+    dr defaulting to BoxSize.min()/Nmesh.max() (:89).
+    """
+    geom = MeshGeometry(Nmesh, BoxSize, dtype='f8')
+    if mode not in ('1d', '2d'):
+        raise ValueError("`mode` should be either '1d' or '2d'")
+    if mode == '1d':
+        Nmu = 1
+
+    def make_complex(pos, wgt):
+        mesh, attrs = to_real_field(pos, geom, weight=wgt,
+                                    resampler=resampler,
+                                    interlaced=interlaced,
+                                    paint_chunk_size=paint_chunk_size)
+        c = r2c(mesh, geom)
+        if compensated:
+            apply_compensation(c, geom, resampler, interlaced)
+        return c, attrs
+
+    c1, attrs1 = make_complex(position, weight)
+    auto = second_position is None
+    if auto:
+        c2f, attrs2 = c1, attrs1
+    else:
+        c2f, attrs2 = make_complex(second_position, second_weight)
+
+    p3d = compute_3d_power(c1, c2f, geom)
+    y3d = c2r(p3d, geom)
+    y3d *= 1.0 / float(numpy.prod(geom.BoxSize))   # fftcorr.py:157-158
+
+    if dr is None:
+        dr = float(geom.BoxSize.min() / geom.Nmesh.max())
+    if rmax is None:
+        rmax = 0.5 * float(geom.BoxSize.min()) + dr / 2
+    redges = numpy.arange(rmin, rmax, dr)
+    muedges = numpy.linspace(0, 1, Nmu + 1, endpoint=True)
+
+    result, pole_result = project_to_basis(
+        y3d, geom, [redges, muedges], los=los, poles=poles,
+        coords=real_coords(geom), hermitian=False)
+    xmean_2d, mumean_2d, y2d, N_2d = result
+
+    attrs = {
+        'N1': attrs1['N'], 'N2': attrs2['N'],
+        'shotnoise': attrs1['shotnoise'] if auto else 0.0,
+        'mode': mode, 'los': list(los), 'Nmu': Nmu, 'poles': list(poles),
+        'dr': dr, 'rmin': rmin, 'rmax': rmax,
+        'Nmesh': geom.Nmesh.copy(), 'BoxSize': geom.BoxSize.copy(),
+    }
+    out = {
+        'redges': redges, 'muedges': muedges,
+        'r': numpy.squeeze(xmean_2d) if mode == '1d' else xmean_2d,
+        'mu': None if mode == '1d' else mumean_2d,
+        'corr': numpy.squeeze(y2d) if mode == '1d' else y2d,
+        'modes': numpy.squeeze(N_2d) if mode == '1d' else N_2d,
+        'attrs': attrs,
+    }
+    if pole_result is not None:
+        pole_k, pole_arr, pole_modes = pole_result
+        out['pole_r'] = pole_k
         out['poles'] = {ell: pole_arr[i] for i, ell in enumerate(poles)}
         out['pole_modes'] = pole_modes
     else:

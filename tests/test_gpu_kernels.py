@@ -228,7 +228,7 @@ def test_bin_kernel_matches_oracle(lib, poles, los):
         hiplib.f64_arr(geom.BoxSize), hiplib.i64_arr(geom.cshape),
         hiplib.i64_arr((0, 0, 0)), None,
         hiplib.dptr(k2_t), len(kedges), hiplib.dptr(mu_t), len(muedges),
-        hiplib.f64_arr(los), hiplib.int_arr(_poles), Nell,
+        hiplib.f64_arr(los), hiplib.int_arr(_poles), Nell, 0,
         hiplib.dptr(sums), hiplib.dptr(sums[NB:]),
         hiplib.dptr(sums[2 * NB:]), hiplib.dptr(sums[3 * NB:]), None),
         'bin')

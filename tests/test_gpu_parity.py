@@ -263,3 +263,54 @@ def test_device_catalog_resident_path():
     assert_allclose(numpy.nan_to_num(r1.power['power']),
                     numpy.nan_to_num(r2.power['power']),
                     rtol=1e-12, atol=1e-12)
+
+
+# ---- FFTCorr (xi(r), the first SURVEY §8f widening row) ----------------
+
+def test_fftcorr_parity():
+    from nbodykit_amd.lab import FFTCorr
+    from oracle import fftcorr_oracle
+    cat = UniformCatalog(nbar=3e-4, BoxSize=256., seed=42)
+    r = FFTCorr(cat, mode='1d', Nmesh=32)
+    pos = uniform_positions(3e-4, 256., 42)
+    want = fftcorr_oracle(pos, Nmesh=32, BoxSize=256., mode='1d')
+    assert_array_equal(r.corr['modes'], want['modes'])
+    got = r.corr['corr']
+    ref = want['corr']
+    ok = numpy.isfinite(ref) & (numpy.abs(ref) > 1e-12)
+    rel = numpy.abs(got[ok] - ref[ok]) / numpy.abs(ref[ok])
+    assert rel.max() < PARITY_RTOL, 'xi parity: %g' % rel.max()
+    assert_allclose(numpy.nan_to_num(r.corr['r']),
+                    numpy.nan_to_num(want['r']), rtol=1e-10, atol=1e-12)
+
+
+def test_fftcorr_parity_2d_poles():
+    from nbodykit_amd.lab import FFTCorr
+    from oracle import fftcorr_oracle
+    cat = UniformCatalog(nbar=1e-3, BoxSize=256., seed=7)
+    r = FFTCorr(cat, mode='2d', Nmesh=32, Nmu=4, poles=[0, 2])
+    pos = numpy.asarray(cat['Position'], dtype='f8')
+    want = fftcorr_oracle(pos, Nmesh=32, BoxSize=256., mode='2d', Nmu=4,
+                          poles=[0, 2])
+    assert_array_equal(r.corr['modes'], want['modes'])
+    got = numpy.ravel(r.corr['corr'])
+    ref = numpy.ravel(want['corr'])
+    ok = numpy.isfinite(ref) & (numpy.abs(ref) > 1e-12)
+    rel = numpy.abs(got[ok] - ref[ok]) / numpy.abs(ref[ok])
+    assert rel.max() < PARITY_RTOL
+    for ell in (0, 2):
+        got = r.poles['corr_%d' % ell]
+        ref = want['poles'][ell]
+        ok = numpy.isfinite(ref) & (numpy.abs(ref) > 1e-12)
+        rel = numpy.abs(got[ok] - ref[ok]) / numpy.abs(ref[ok])
+        assert rel.max() < PARITY_RTOL, 'xi_%d parity: %g' % (ell, rel.max())
+
+
+def test_fftcorr_save_load(tmp_path):
+    from nbodykit_amd.lab import FFTCorr
+    cat = UniformCatalog(nbar=3e-4, BoxSize=256., seed=42)
+    r = FFTCorr(cat, mode='1d', Nmesh=32)
+    path = str(tmp_path / 'fftcorr.json')
+    r.save(path)
+    r2 = FFTCorr.load(path)
+    assert_array_equal(r.corr['corr'], r2.corr['corr'])

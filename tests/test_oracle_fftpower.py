@@ -209,3 +209,38 @@ def test_cross_power_matches_auto(upos):
 def test_mode_validation(upos):
     with pytest.raises(ValueError):
         fftpower_oracle(upos, Nmesh=8, BoxSize=512., mode='3d')
+
+
+# ---- FFTCorr oracle (reference algorithms/fftcorr.py) ------------------
+
+def test_fftcorr_shotnoise_spike(upos):
+    """xi of a Poisson catalog: the r=0 bin holds ~SN/Vcell (the c2r of
+    the flat shot-noise spectrum), everything else ~0"""
+    from oracle import fftcorr_oracle
+    r = fftcorr_oracle(upos, Nmesh=32, BoxSize=512., mode='1d')
+    xi = r['corr']
+    Vcell = (512. / 32) ** 3
+    expect = r['attrs']['shotnoise'] / Vcell
+    assert abs(xi[0] / expect - 1) < 0.05
+    tail = xi[3:][~numpy.isnan(xi[3:])]
+    assert numpy.sqrt((tail ** 2).mean()) < 0.05 * xi[0]
+
+
+def test_fftcorr_poles_identity(upos):
+    """mu-weighted monopole == xi_0 (the FFTPower identity in r-space)"""
+    from oracle import fftcorr_oracle
+    r = fftcorr_oracle(upos, Nmesh=32, BoxSize=512., mode='2d', Nmu=4,
+                       poles=[0, 2])
+    ximu = r['corr']
+    modes = r['modes']
+    modes_1d = modes.sum(axis=-1)
+    mono = numpy.nansum(ximu * modes, axis=-1) / modes_1d
+    assert_array_equal(modes_1d, r['pole_modes'])
+    ok = numpy.isfinite(mono)
+    assert_allclose(mono[ok], r['poles'][0][ok], rtol=1e-10, atol=1e-12)
+
+
+def test_fftcorr_mode_validation(upos):
+    from oracle import fftcorr_oracle
+    with pytest.raises(ValueError):
+        fftcorr_oracle(upos, Nmesh=8, BoxSize=512., mode='3d')
