@@ -313,7 +313,7 @@ def test_bin_partitioned_matches_full(lib):
             hiplib.i64_arr(block.shape), hiplib.i64_arr(off), None,
             hiplib.dptr(k2_t), len(kedges), hiplib.dptr(mu_t),
             len(muedges), hiplib.f64_arr((0, 0, 1)),
-            hiplib.int_arr([0]), 1,
+            hiplib.int_arr([0]), 1, 0,
             hiplib.dptr(sums), hiplib.dptr(sums[NB:]),
             hiplib.dptr(sums[2 * NB:]), hiplib.dptr(sums[3 * NB:]),
             None), 'bin')
