@@ -99,7 +99,7 @@ int nbk_paint_f64(const double* pos, const double* mass, int64_t n,
  */
 int nbk_bucket_count_f64(const double* pos_aos, int64_t n,
                          const int64_t nmesh[3], const double box[3],
-                         int* counts, void* stream);
+                         int* counts, int* scrambled_flag, void* stream);
 int nbk_bucket_scatter_f64(const double* pos_aos, const double* mass,
                            int64_t n, const int64_t nmesh[3],
                            const double box[3], int64_t* offsets,

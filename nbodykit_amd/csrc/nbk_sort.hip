@@ -22,19 +22,28 @@ __device__ __forceinline__ int64_t bucket_of(double x, double y, double z,
     return (ix * n1 + iy) * n2 + iz;
 }
 
+// also detects (heuristically, lane-adjacent pairs) whether the input
+// is already cell-ordered: scrambled -> *scrambled_flag = 1.  The caller
+// skips the scatter pass for ordered input.
 __global__ void kbucket_count(const double* __restrict__ pos, int64_t n,
                               int64_t n0, int64_t n1, int64_t n2,
                               double invH0, double invH1, double invH2,
-                              int* __restrict__ counts)
+                              int* __restrict__ counts,
+                              int* __restrict__ scrambled_flag)
 {
+    const int lane = threadIdx.x & 63;
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int out_of_order = 0;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
          i < n; i += stride) {
         const int64_t b = bucket_of(pos[3 * i], pos[3 * i + 1],
                                     pos[3 * i + 2], invH0, invH1, invH2,
                                     n0, n1, n2);
         atomicAdd(&counts[b], 1);
+        const int64_t b_up = __shfl_up((long long)b, 1, 64);
+        if (lane > 0 && b_up > b) out_of_order = 1;
     }
+    if (out_of_order) atomicOr(scrambled_flag, 1);
 }
 
 __global__ void kbucket_scatter(const double* __restrict__ pos,
@@ -74,14 +83,15 @@ int sgrid(int64_t n) {
 extern "C" int nbk_bucket_count_f64(const double* pos_aos, int64_t n,
                                     const int64_t nmesh[3],
                                     const double box[3],
-                                    int* counts, void* stream)
+                                    int* counts, int* scrambled_flag,
+                                    void* stream)
 {
     if (n == 0) return NBK_OK;
     hipLaunchKernelGGL(kbucket_count, dim3(sgrid(n)), dim3(256), 0,
                        (hipStream_t)stream, pos_aos, n,
                        nmesh[0], nmesh[1], nmesh[2],
                        nmesh[0] / box[0], nmesh[1] / box[1],
-                       nmesh[2] / box[2], counts);
+                       nmesh[2] / box[2], counts, scrambled_flag);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

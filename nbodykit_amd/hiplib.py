@@ -49,7 +49,7 @@ def _declare(lib):
                                   c_void]
     lib.nbk_bucket_count_f64.restype = ctypes.c_int
     lib.nbk_bucket_count_f64.argtypes = [c_void, c_i64, c_i64_p, c_f64_p,
-                                         c_void, c_void]
+                                         c_void, c_void, c_void]
     lib.nbk_bucket_scatter_f64.restype = ctypes.c_int
     lib.nbk_bucket_scatter_f64.argtypes = [c_void, c_void, c_i64, c_i64_p,
                                            c_f64_p, c_void, c_void, c_void,

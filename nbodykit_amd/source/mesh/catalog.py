@@ -49,7 +49,6 @@ def _is_trivial_true(sel):
 
 
 _SORT_MIN_N = 1 << 21        # below this the deposit is cheap either way
-_SORT_MONO_THRESHOLD = 0.98  # fraction of already-ordered neighbours
 
 
 def _prepare_particles(pos_t, mass_t, pm):
@@ -75,21 +74,20 @@ def _prepare_particles(pos_t, mass_t, pm):
     lib = hiplib.require()
     n0, n1, n2 = (int(x) for x in pm.Nmesh)
     ncells = n0 * n1 * n2
-    invH = torch.as_tensor((pm.Nmesh / pm.BoxSize)).to(pos_t.device)
-    u = torch.floor(pos_t * invH).long()
-    key = (u[:, 0] * n1 + u[:, 1]) * n2 + u[:, 2]
-    mono = (key[1:] >= key[:-1]).float().mean().item()
-    if mono >= _SORT_MONO_THRESHOLD:
-        return pos_t.t().contiguous(), mass_t
 
     nmesh = hiplib.i64_arr(pm.Nmesh)
     box = hiplib.f64_arr(pm.BoxSize)
     stream = hiplib.cur_stream()
     counts = torch.zeros(ncells, dtype=torch.int32, device='cuda')
+    flag = torch.zeros(1, dtype=torch.int32, device='cuda')
     pos_in = pos_t.contiguous()
     hiplib.check(lib.nbk_bucket_count_f64(
-        hiplib.dptr(pos_in), n, nmesh, box, hiplib.dptr(counts), stream),
-        'nbk_bucket_count_f64')
+        hiplib.dptr(pos_in), n, nmesh, box, hiplib.dptr(counts),
+        hiplib.dptr(flag), stream), 'nbk_bucket_count_f64')
+    if int(flag.item()) == 0:
+        # already cell-ordered (the count kernel checked lane-adjacent
+        # pairs): skip the scatter, just transpose to SoA
+        return pos_t.t().contiguous(), mass_t
     offsets = torch.zeros(ncells, dtype=torch.int64, device='cuda')
     offsets[1:] = torch.cumsum(counts[:-1].long(), 0)  # exclusive prefix
     out_soa = torch.empty(3 * n, dtype=torch.float64, device='cuda')
