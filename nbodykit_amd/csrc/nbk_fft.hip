@@ -272,13 +272,12 @@ extern "C" int nbk_fft_c_strided(double* cplx, int64_t nfft, int64_t stride,
     double* table = get_twiddles(2 * nfft);
     if (!table) { NBK_SET_ERR("twiddle alloc failed"); return NBK_ERR_HIP; }
 
-    // inner-tile width: 8 columns = 128 B contiguous per element row so
-    // strided global accesses use full cachelines (TI=4 wastes half of
-    // every 128 B line: measured 1.3 vs 2.1 TB/s).  LDS = nfft*TI*16 B,
-    // up to 128 KiB of the 160 KiB gfx950 LDS (needs the dynamic-LDS
-    // attribute raised past the 64 KiB default).
-    int TI = 8;
-    while ((int64_t)nfft * TI * (int64_t)sizeof(cdouble) > 131072 && TI > 1)
+    // inner-tile width: 4 columns x 16 B = 64 B contiguous per element
+    // row; LDS = nfft*TI*16 B <= 64 KiB keeps 2 blocks (8 waves) per CU.
+    // (TI=8 with 128 KiB LDS was tried: full 128 B lines but 1 block/CU
+    // — measured 16.2 vs 13.0 ms per 1024^3 pass; latency hiding wins.)
+    int TI = 4;
+    while ((int64_t)nfft * TI * (int64_t)sizeof(cdouble) > 65536 && TI > 1)
         TI >>= 1;
     if (TI > n_inner) TI = (int)n_inner;
     const int tiles = (int)((n_inner + TI - 1) / TI);
