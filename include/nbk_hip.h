@@ -106,12 +106,24 @@ int nbk_bucket_scatter_f64(const double* pos_aos, const double* mass,
                            double* pos_soa_out, double* mass_out,
                            void* stream);
 
-/* readout (gather dual of paint; resampler='nnb' serves the LogNormal
- * generator, mockmaker.py:317-319; cic for FFTRecon later) */
-int nbk_readout_nnb_f64(const double* pos, int64_t n,
-                        const int64_t nmesh[3], const double box[3],
-                        const double* mesh, int64_t x0, int64_t nx_local,
-                        double* out, void* stream);
+/* readout (gather dual of paint; window 0/1/2 = cic/tsc/pcs, 3 = nnb).
+ * Serves FFTRecon's displacement solve (fftrecon.py:246-249) and the
+ * LogNormal generator's nnb reads (mockmaker.py:317-319).  Cells outside
+ * the local slab contribute 0 — ghost owners add their partials via the
+ * host-side exchange. */
+int nbk_readout_f64(const double* pos, int64_t n,
+                    const int64_t nmesh[3], const double box[3],
+                    int window,
+                    const double* mesh, int64_t x0, int64_t nx_local,
+                    double* out, void* stream);
+
+/* FFTRecon displacement solve (fftrecon.py:222-238):
+ * out = i k_axis/k^2 exp(-k^2 R^2/2) / (bias (1 + (f/bias) mu^2)) in */
+int nbk_recon_displacement_f64(double* out, const double* in,
+                               const int64_t nmesh[3], const double box[3],
+                               const int64_t dims[3], const int64_t off[3],
+                               int axis, double R, double bias, double f,
+                               const double los[3], void* stream);
 
 /* FFT ----------------------------------------------------------------
  * Power-of-two lengths only (8 <= N <= 4096).  The 3D transform is

@@ -1,2 +1,3 @@
 from .fftpower import FFTPower, FFTBase, project_to_basis
 from .fftcorr import FFTCorr
+from .fftrecon import FFTRecon

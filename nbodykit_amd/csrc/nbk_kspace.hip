@@ -102,6 +102,36 @@ __global__ void kinterlace(double* __restrict__ c1,
     }
 }
 
+// FFTRecon displacement solve (fftrecon.py:222-238):
+// out = i k_d / k^2 * exp(-k^2 R^2 / 2) / (bias (1 + (f/bias) mu^2)) * v
+__global__ void krecon_disp(double* __restrict__ out,
+                            const double* __restrict__ in, Layout L,
+                            double k0x, double k0y, double k0z,
+                            int axis, double R, double bias, double f,
+                            double losx, double losy, double losz)
+{
+    const int64_t total = L.d0 * L.d1 * L.d2;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         idx < total; idx += stride) {
+        double fr[3];
+        global_freqs(L, idx, fr);
+        const double kx = fr[0] * k0x, ky = fr[1] * k0y, kz = fr[2] * k0z;
+        double k2 = kx * kx + ky * ky + kz * kz;
+        const bool zero = (k2 == 0.0);
+        if (zero) k2 = 1.0;
+        const double mu = (kx * losx + ky * losy + kz * losz) / sqrt(k2);
+        const double smooth = exp(-0.5 * k2 * R * R);
+        const double frac = bias * (1.0 + f / bias * mu * mu);
+        const double kd = (axis == 0) ? kx : (axis == 1) ? ky : kz;
+        const double fac = zero ? 0.0 : kd / k2 * smooth / frac;
+        const double re = in[2 * idx], im = in[2 * idx + 1];
+        // multiply by i * fac
+        out[2 * idx] = -im * fac;
+        out[2 * idx + 1] = re * fac;
+    }
+}
+
 __global__ void kpower3d(double* __restrict__ out,
                          const double* __restrict__ c1,
                          const double* __restrict__ c2,
@@ -187,6 +217,30 @@ extern "C" int nbk_interlace_combine_f64(double* c1, const double* c2,
     const int64_t total = L.d0 * L.d1 * L.d2;
     hipLaunchKernelGGL(kinterlace, dim3(egrid(total, 256)), dim3(256), 0,
                        (hipStream_t)stream, c1, c2, L);
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}
+
+extern "C" int nbk_recon_displacement_f64(double* out, const double* in,
+                                          const int64_t nmesh[3],
+                                          const double box[3],
+                                          const int64_t dims[3],
+                                          const int64_t off[3],
+                                          int axis, double R, double bias,
+                                          double f, const double los[3],
+                                          void* stream)
+{
+    if (axis < 0 || axis > 2) {
+        NBK_SET_ERR("nbk_recon_displacement_f64: bad axis %d", axis);
+        return NBK_ERR_ARG;
+    }
+    Layout L = make_layout(nmesh, dims, off, nullptr);
+    const int64_t total = L.d0 * L.d1 * L.d2;
+    hipLaunchKernelGGL(krecon_disp, dim3(egrid(total, 256)), dim3(256), 0,
+                       (hipStream_t)stream, out, in, L,
+                       2.0 * M_PI / box[0], 2.0 * M_PI / box[1],
+                       2.0 * M_PI / box[2], axis, R, bias, f,
+                       los[0], los[1], los[2]);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

@@ -150,23 +150,105 @@ __global__ void kpaint(const double* __restrict__ px,
     }
 }
 
-__global__ void kreadout_nnb(const double* __restrict__ px,
-                             const double* __restrict__ py,
-                             const double* __restrict__ pz, int64_t n,
-                             int64_t n0, int64_t n1, int64_t n2,
-                             double invH0, double invH1, double invH2,
-                             const double* __restrict__ mesh,
-                             int64_t x0, int64_t nx_local,
-                             double* __restrict__ out)
+// windowed gather — the dual of kpaint (pmesh readout, used by
+// FFTRecon's displacement solve, fftrecon.py:246-249, and the nnb
+// variant by the LogNormal generator, mockmaker.py:317-319).  Each lane
+// sums W * mesh over its particle's neighbourhood; cells outside the
+// local slab contribute 0 (the ghost owner adds its partial — the host
+// exchanges and sums partials across ranks).
+template <int WINDOW>
+__global__ void kreadout(const double* __restrict__ px,
+                         const double* __restrict__ py,
+                         const double* __restrict__ pz, int64_t n,
+                         int64_t n0, int64_t n1, int64_t n2,
+                         double invH0, double invH1, double invH2,
+                         const double* __restrict__ mesh,
+                         int64_t x0, int64_t nx_local,
+                         double* __restrict__ out)
 {
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
          i < n; i += stride) {
-        const int64_t gx = wrap_idx((int64_t)floor(px[i] * invH0 + 0.5), n0);
-        const int64_t gy = wrap_idx((int64_t)floor(py[i] * invH1 + 0.5), n1);
-        const int64_t gz = wrap_idx((int64_t)floor(pz[i] * invH2 + 0.5), n2);
-        if (gx < x0 || gx >= x0 + nx_local) { out[i] = 0.0; continue; }
-        out[i] = mesh[((gx - x0) * n1 + gy) * n2 + gz];
+        const double u0 = px[i] * invH0, u1 = py[i] * invH1,
+                     u2 = pz[i] * invH2;
+
+        if (WINDOW == 3) {    // nnb
+            const int64_t gx = wrap_idx((int64_t)floor(u0 + 0.5), n0);
+            const int64_t gy = wrap_idx((int64_t)floor(u1 + 0.5), n1);
+            const int64_t gz = wrap_idx((int64_t)floor(u2 + 0.5), n2);
+            out[i] = (gx >= x0 && gx < x0 + nx_local)
+                ? mesh[((gx - x0) * n1 + gy) * n2 + gz] : 0.0;
+            continue;
+        }
+
+        constexpr int SUP = (WINDOW == NBK_WINDOW_CIC) ? 2
+                          : (WINDOW == NBK_WINDOW_TSC) ? 3 : 4;
+        double w0[SUP], w1[SUP], w2[SUP];
+        int64_t b0, b1, b2;
+        if (WINDOW == NBK_WINDOW_CIC) {
+            const double f0 = floor(u0), f1 = floor(u1), f2 = floor(u2);
+            b0 = (int64_t)f0; b1 = (int64_t)f1; b2 = (int64_t)f2;
+            w0[1] = u0 - f0; w0[0] = 1.0 - w0[1];
+            w1[1] = u1 - f1; w1[0] = 1.0 - w1[1];
+            w2[1] = u2 - f2; w2[0] = 1.0 - w2[1];
+        } else if (WINDOW == NBK_WINDOW_TSC) {
+            const double f0 = floor(u0 + 0.5), f1 = floor(u1 + 0.5),
+                         f2 = floor(u2 + 0.5);
+            b0 = (int64_t)f0 - 1; b1 = (int64_t)f1 - 1;
+            b2 = (int64_t)f2 - 1;
+            #pragma unroll
+            for (int d = 0; d < 3; d++) {
+                const double s0 = u0 - (f0 + d - 1);
+                const double s1 = u1 - (f1 + d - 1);
+                const double s2 = u2 - (f2 + d - 1);
+                const double a0 = fabs(s0), a1 = fabs(s1), a2 = fabs(s2);
+                w0[d] = a0 < 0.5 ? 0.75 - s0 * s0
+                                 : 0.5 * (1.5 - a0) * (1.5 - a0);
+                w1[d] = a1 < 0.5 ? 0.75 - s1 * s1
+                                 : 0.5 * (1.5 - a1) * (1.5 - a1);
+                w2[d] = a2 < 0.5 ? 0.75 - s2 * s2
+                                 : 0.5 * (1.5 - a2) * (1.5 - a2);
+            }
+        } else {
+            const double f0 = floor(u0), f1 = floor(u1), f2 = floor(u2);
+            b0 = (int64_t)f0 - 1; b1 = (int64_t)f1 - 1;
+            b2 = (int64_t)f2 - 1;
+            #pragma unroll
+            for (int d = 0; d < 4; d++) {
+                const double s0 = fabs(u0 - (f0 + d - 1));
+                const double s1 = fabs(u1 - (f1 + d - 1));
+                const double s2 = fabs(u2 - (f2 + d - 1));
+                w0[d] = s0 < 1.0
+                    ? (4.0 - 6.0 * s0 * s0 + 3.0 * s0 * s0 * s0) / 6.0
+                    : (2.0 - s0) * (2.0 - s0) * (2.0 - s0) / 6.0;
+                w1[d] = s1 < 1.0
+                    ? (4.0 - 6.0 * s1 * s1 + 3.0 * s1 * s1 * s1) / 6.0
+                    : (2.0 - s1) * (2.0 - s1) * (2.0 - s1) / 6.0;
+                w2[d] = s2 < 1.0
+                    ? (4.0 - 6.0 * s2 * s2 + 3.0 * s2 * s2 * s2) / 6.0
+                    : (2.0 - s2) * (2.0 - s2) * (2.0 - s2) / 6.0;
+            }
+        }
+
+        double acc = 0.0;
+        #pragma unroll
+        for (int dx = 0; dx < SUP; dx++) {
+            const int64_t gx = wrap_idx(b0 + dx, n0);
+            if (gx < x0 || gx >= x0 + nx_local) continue;
+            const int64_t lx = gx - x0;
+            #pragma unroll
+            for (int dy = 0; dy < SUP; dy++) {
+                const int64_t gy = wrap_idx(b1 + dy, n1);
+                const double wxy = w0[dx] * w1[dy];
+                #pragma unroll
+                for (int dz = 0; dz < SUP; dz++) {
+                    const int64_t gz = wrap_idx(b2 + dz, n2);
+                    acc += wxy * w2[dz]
+                        * mesh[(lx * n1 + gy) * n2 + gz];
+                }
+            }
+        }
+        out[i] = acc;
     }
 }
 
@@ -223,21 +305,49 @@ extern "C" int nbk_paint_f64(const double* pos, const double* mass, int64_t n,
     return NBK_OK;
 }
 
-extern "C" int nbk_readout_nnb_f64(const double* pos, int64_t n,
-                                   const int64_t nmesh[3], const double box[3],
-                                   const double* mesh, int64_t x0,
-                                   int64_t nx_local,
-                                   double* out, void* stream)
+extern "C" int nbk_readout_f64(const double* pos, int64_t n,
+                               const int64_t nmesh[3], const double box[3],
+                               int window,
+                               const double* mesh, int64_t x0,
+                               int64_t nx_local,
+                               double* out, void* stream)
 {
     if (n == 0) return NBK_OK;
     const int block = 256;
     const int grid = grid_for(n, block);
     hipStream_t s = (hipStream_t)stream;
-    hipLaunchKernelGGL(kreadout_nnb, dim3(grid), dim3(block), 0, s,
-                       pos, pos + n, pos + 2 * n, n,
-                       nmesh[0], nmesh[1], nmesh[2],
-                       nmesh[0] / box[0], nmesh[1] / box[1], nmesh[2] / box[2],
-                       mesh, x0, nx_local, out);
+    const double iH0 = nmesh[0] / box[0];
+    const double iH1 = nmesh[1] / box[1];
+    const double iH2 = nmesh[2] / box[2];
+    switch (window) {
+    case NBK_WINDOW_CIC:
+        hipLaunchKernelGGL(kreadout<NBK_WINDOW_CIC>, dim3(grid), dim3(block),
+                           0, s, pos, pos + n, pos + 2 * n, n,
+                           nmesh[0], nmesh[1], nmesh[2], iH0, iH1, iH2,
+                           mesh, x0, nx_local, out);
+        break;
+    case NBK_WINDOW_TSC:
+        hipLaunchKernelGGL(kreadout<NBK_WINDOW_TSC>, dim3(grid), dim3(block),
+                           0, s, pos, pos + n, pos + 2 * n, n,
+                           nmesh[0], nmesh[1], nmesh[2], iH0, iH1, iH2,
+                           mesh, x0, nx_local, out);
+        break;
+    case NBK_WINDOW_PCS:
+        hipLaunchKernelGGL(kreadout<NBK_WINDOW_PCS>, dim3(grid), dim3(block),
+                           0, s, pos, pos + n, pos + 2 * n, n,
+                           nmesh[0], nmesh[1], nmesh[2], iH0, iH1, iH2,
+                           mesh, x0, nx_local, out);
+        break;
+    case 3:   /* nnb */
+        hipLaunchKernelGGL(kreadout<3>, dim3(grid), dim3(block), 0, s,
+                           pos, pos + n, pos + 2 * n, n,
+                           nmesh[0], nmesh[1], nmesh[2], iH0, iH1, iH2,
+                           mesh, x0, nx_local, out);
+        break;
+    default:
+        NBK_SET_ERR("nbk_readout_f64: unknown window id %d", window);
+        return NBK_ERR_ARG;
+    }
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

@@ -133,6 +133,77 @@ __global__ void paint_cic_merge(const double* __restrict__ px,
     }
 }
 
+// TSC with/without the wave merge (C3 shape diagnosis)
+__device__ __forceinline__ void probe_merged_deposit(double* mesh,
+                                                     long long addr,
+                                                     double val, int lane) {
+    const long long a_up1 = __shfl_up(addr, 1, 64);
+    const bool head = (lane == 0) || (a_up1 != addr);
+    const unsigned long long heads = __ballot(head);
+    const unsigned long long below = heads & (~0ULL >> (63 - lane));
+    const int myhead = 63 - __clzll(below);
+    #pragma unroll
+    for (int d = 1; d < 64; d <<= 1) {
+        const double v_up = __shfl_up(val, d, 64);
+        if (lane - d >= myhead) val += v_up;
+    }
+    const bool next_head = ((lane < 63) && ((heads >> (lane + 1)) & 1ULL));
+    if (addr >= 0 && (lane == 63 || next_head)) atomicAdd(&mesh[addr], val);
+}
+
+template <bool MERGE>
+__global__ void paint_tsc(const double* __restrict__ px,
+                          const double* __restrict__ py,
+                          const double* __restrict__ pz, long n,
+                          long N, double invH, double* __restrict__ mesh)
+{
+    const int lane = threadIdx.x & 63;
+    const long stride = (long)gridDim.x * blockDim.x;
+    const long wbase0 = blockIdx.x * (long)blockDim.x + (threadIdx.x & ~63);
+    for (long wb = wbase0; wb < n; wb += stride) {
+        const long i = wb + lane;
+        const bool valid = i < n;
+        const double u0 = valid ? px[i] * invH : 0.0;
+        const double u1 = valid ? py[i] * invH : 0.0;
+        const double u2 = valid ? pz[i] * invH : 0.0;
+        const double f0 = floor(u0 + 0.5), f1 = floor(u1 + 0.5),
+                     f2 = floor(u2 + 0.5);
+        const long b0 = (long)f0 - 1, b1 = (long)f1 - 1, b2 = (long)f2 - 1;
+        double w0[3], w1[3], w2[3];
+        #pragma unroll
+        for (int d = 0; d < 3; d++) {
+            const double s0 = u0 - (f0 + d - 1), s1 = u1 - (f1 + d - 1),
+                         s2 = u2 - (f2 + d - 1);
+            const double a0 = fabs(s0), a1 = fabs(s1), a2 = fabs(s2);
+            w0[d] = a0 < 0.5 ? 0.75 - s0*s0 : 0.5*(1.5-a0)*(1.5-a0);
+            w1[d] = a1 < 0.5 ? 0.75 - s1*s1 : 0.5*(1.5-a1)*(1.5-a1);
+            w2[d] = a2 < 0.5 ? 0.75 - s2*s2 : 0.5*(1.5-a2)*(1.5-a2);
+        }
+        #pragma unroll
+        for (int dx = 0; dx < 3; dx++) {
+            const long gx = wrapi(b0 + dx, N);
+            #pragma unroll
+            for (int dy = 0; dy < 3; dy++) {
+                const long gy = wrapi(b1 + dy, N);
+                const double wxy = w0[dx] * w1[dy];
+                #pragma unroll
+                for (int dz = 0; dz < 3; dz++) {
+                    const long gz = wrapi(b2 + dz, N);
+                    if (MERGE) {
+                        const long long addr = valid
+                            ? (long long)((gx * N + gy) * N + gz)
+                            : (long long)(-1 - lane);
+                        probe_merged_deposit(mesh, addr, wxy * w2[dz], lane);
+                    } else if (valid) {
+                        atomicAdd(&mesh[(gx * N + gy) * N + gz],
+                                  wxy * w2[dz]);
+                    }
+                }
+            }
+        }
+    }
+}
+
 // variant: read-only (bandwidth leg)
 __global__ void read_only(const double* __restrict__ px,
                           const double* __restrict__ py,
@@ -248,6 +319,23 @@ int main(int argc, char** argv) {
     };
 
     const char* modenames[3] = {"sorted-uniform", "sorted-clumpy", "random"};
+    // TSC A/B at the C3 shape (pass N=512 n=1e8 on the command line)
+    for (int mode = 0; mode < 2; mode++) {
+        hipLaunchKernelGGL(gen_sorted, dim3(grid_for(n)), dim3(256), 0, 0,
+                           px, py, pz, n, N, H, 999u, mode);
+        CHECK(hipDeviceSynchronize());
+        char buf[128];
+        snprintf(buf, 128, "tsc merged [%s]", mode ? "clumpy" : "uniform");
+        report(buf, timeit([&] {
+            hipLaunchKernelGGL((paint_tsc<true>), dim3(grid_for(n)),
+                               dim3(256), 0, 0, px, py, pz, n, N, invH, mesh);
+        }, 3));
+        snprintf(buf, 128, "tsc plain  [%s]", mode ? "clumpy" : "uniform");
+        report(buf, timeit([&] {
+            hipLaunchKernelGGL((paint_tsc<false>), dim3(grid_for(n)),
+                               dim3(256), 0, 0, px, py, pz, n, N, invH, mesh);
+        }, 3));
+    }
     for (int mode = 0; mode < 3; mode++) {
         hipLaunchKernelGGL(gen_sorted, dim3(grid_for(n)), dim3(256), 0, 0,
                            px, py, pz, n, N, H, 12345u, mode);

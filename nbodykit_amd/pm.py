@@ -283,6 +283,74 @@ class RealField(_FieldBase):
         f.attrs = dict(self.attrs)
         return f
 
+    # x-cell span touched by a readout, relative to floor(u0)
+    _READOUT_RANGE = {'cic': (0, 1), 'tsc': (-1, 1), 'pcs': (-1, 2),
+                      'nnb': (0, 1)}
+
+    def readout(self, pos_t, resampler='cic'):
+        """Windowed gather at device positions (pmesh ``readout``;
+        FFTRecon's displacement reads, fftrecon.py:246-249).  Multi-rank:
+        particles are duplicated to every rank whose slab their stencil
+        touches, each rank computes its partial sum and the partials are
+        summed back at the origin rank."""
+        import torch
+        lib = hiplib.require()
+        pm = self.pm
+        window_id = {'cic': 0, 'tsc': 1, 'pcs': 2, 'nnb': 3}[resampler]
+        nmesh = hiplib.i64_arr(pm.Nmesh)
+        box = hiplib.f64_arr(pm.BoxSize)
+        stream = hiplib.cur_stream()
+
+        def run_kernel(p):
+            soa = p.t().contiguous()
+            out = torch.empty(len(p), dtype=torch.float64, device='cuda')
+            hiplib.check(lib.nbk_readout_f64(
+                hiplib.dptr(soa), len(p), nmesh, box, window_id,
+                hiplib.dptr(self.value), pm.x_start, pm.nx_local,
+                hiplib.dptr(out), stream), 'nbk_readout_f64')
+            return out
+
+        if pm.comm.size == 1:
+            return run_kernel(pos_t)
+
+        import torch.distributed as dist
+        comm = pm.comm
+        ws = comm.size
+        dmin, dmax = self._READOUT_RANGE[resampler]
+        invH0 = float(pm.Nmesh[0]) / float(pm.BoxSize[0])
+        fu = torch.floor(pos_t[:, 0] * invH0).long()
+        n0 = int(pm.Nmesh[0])
+
+        idx_list, rank_list = [], []
+        for d in range(dmin, dmax + 1):
+            cell = torch.remainder(fu + d, n0)
+            rank_list.append(torch.div(cell, pm.nx_local,
+                                       rounding_mode='floor'))
+            idx_list.append(torch.arange(len(pos_t), device=pos_t.device))
+        ranks = torch.cat(rank_list)
+        idxs = torch.cat(idx_list)
+        keys = torch.unique(idxs * ws + ranks)
+        idxs = torch.div(keys, ws, rounding_mode='floor')
+        ranks = keys - idxs * ws
+        order = torch.argsort(ranks, stable=True)
+        idxs = idxs[order]
+        ranks = ranks[order]
+        counts_send = torch.bincount(ranks, minlength=ws).cpu().tolist()
+
+        recv_pos = exchange_particle_arrays(pos_t[idxs], counts_send, comm)
+        partial = run_kernel(recv_pos)
+
+        # return the partials to their origin ranks (reverse splits)
+        counts_recv = [row[comm.rank]
+                       for row in comm.allgather(counts_send)]
+        back = torch.empty(int(sum(counts_send)), dtype=torch.float64,
+                           device=partial.device)
+        dist.all_to_all_single(back, partial, counts_send, counts_recv)
+
+        out = torch.zeros(len(pos_t), dtype=torch.float64, device='cuda')
+        out.index_add_(0, idxs, back)
+        return out
+
     def preview(self, Nmesh=None, axes=None, root=0):
         """Gather the full field (optionally summed over the axes not in
         ``axes``) on every rank (base/mesh.py:340-365 semantics)."""

@@ -338,46 +338,67 @@ class CatalogMesh(MeshSource):
         return toret
 
     def _route(self, pos_t, mass_t):
-        """Duplicate each particle to every rank whose x-slab any of its
-        deposit cells falls in (the pmesh decompose/exchange step,
-        reference :271-284, with ghost width from the window support and
-        interlacing shift), then alltoall the payloads."""
-        import torch
-        pm = self.pm
-        comm = pm.comm
-        ws = comm.size
         dmin, dmax = _GHOST_RANGE[(self.resampler, bool(self.interlaced))]
+        return route_particles(pos_t, mass_t, self.pm, dmin, dmax)
 
-        invH0 = float(pm.Nmesh[0]) / float(pm.BoxSize[0])
-        fu = torch.floor(pos_t[:, 0] * invH0).long()
-        n0 = int(pm.Nmesh[0])
-        nx_l = pm.nx_local
 
-        idx_list = []
-        rank_list = []
-        for d in range(dmin, dmax + 1):
-            cell = torch.remainder(fu + d, n0)
-            rank_list.append(torch.div(cell, nx_l, rounding_mode='floor'))
-            idx_list.append(torch.arange(len(pos_t), device=pos_t.device))
-        ranks = torch.cat(rank_list)
-        idxs = torch.cat(idx_list)
-        # dedup (particle, rank) pairs
-        keys = idxs * ws + ranks
-        keys = torch.unique(keys)
-        idxs = torch.div(keys, ws, rounding_mode='floor')
-        ranks = keys - idxs * ws
+def route_particles(pos_t, mass_t, pm, dmin, dmax):
+    """Duplicate each particle to every rank whose x-slab any of its
+    deposit cells falls in (the pmesh decompose/exchange step, reference
+    :271-284, with ghost width from the window support and interlacing
+    shift), then alltoall the payloads."""
+    import torch
+    comm = pm.comm
+    ws = comm.size
+    invH0 = float(pm.Nmesh[0]) / float(pm.BoxSize[0])
+    fu = torch.floor(pos_t[:, 0] * invH0).long()
+    n0 = int(pm.Nmesh[0])
+    nx_l = pm.nx_local
 
-        order = torch.argsort(ranks, stable=True)
-        idxs = idxs[order]
-        ranks = ranks[order]
-        counts = torch.bincount(ranks, minlength=ws).cpu().tolist()
+    idx_list = []
+    rank_list = []
+    for d in range(dmin, dmax + 1):
+        cell = torch.remainder(fu + d, n0)
+        rank_list.append(torch.div(cell, nx_l, rounding_mode='floor'))
+        idx_list.append(torch.arange(len(pos_t), device=pos_t.device))
+    ranks = torch.cat(rank_list)
+    idxs = torch.cat(idx_list)
+    # dedup (particle, rank) pairs
+    keys = torch.unique(idxs * ws + ranks)
+    idxs = torch.div(keys, ws, rounding_mode='floor')
+    ranks = keys - idxs * ws
 
-        if mass_t is None:
-            recv = exchange_particle_arrays(pos_t[idxs], counts, comm)
-            return recv.contiguous(), None
-        payload = torch.cat([pos_t[idxs], mass_t[idxs, None]], dim=1)
-        recv = exchange_particle_arrays(payload, counts, comm)
-        return recv[:, :3].contiguous(), recv[:, 3].contiguous()
+    order = torch.argsort(ranks, stable=True)
+    idxs = idxs[order]
+    ranks = ranks[order]
+    counts = torch.bincount(ranks, minlength=ws).cpu().tolist()
+
+    if mass_t is None:
+        recv = exchange_particle_arrays(pos_t[idxs], counts, comm)
+        return recv.contiguous(), None
+    payload = torch.cat([pos_t[idxs], mass_t[idxs, None]], dim=1)
+    recv = exchange_particle_arrays(payload, counts, comm)
+    return recv[:, :3].contiguous(), recv[:, 3].contiguous()
+
+
+def paint_raw(pos_t, pm, resampler='cic'):
+    """Paint device positions with unit mass into a fresh (unnormalized)
+    RealField — the bare ``pm.paint`` the reconstruction driver uses
+    (fftrecon.py:159-161).  Handles routing and the locality sort."""
+    field = RealField(pm)
+    if len(pos_t) == 0:
+        return field
+    lib = hiplib.require()
+    if pm.comm.size > 1:
+        dmin, dmax = _GHOST_RANGE[(resampler, False)]
+        pos_t, _ = route_particles(pos_t, None, pm, dmin, dmax)
+    pos_soa, _ = _prepare_particles(pos_t, None, pm)
+    hiplib.check(lib.nbk_paint_f64(
+        hiplib.dptr(pos_soa), None, len(pos_t),
+        hiplib.i64_arr(pm.Nmesh), hiplib.f64_arr(pm.BoxSize),
+        hiplib.WINDOW_IDS[resampler], 0.0, hiplib.dptr(field.value),
+        pm.x_start, pm.nx_local, hiplib.cur_stream()), 'nbk_paint_f64')
+    return field
 
     # -- compensation actions (reference :405-451) ------------------------
     @property

@@ -19,8 +19,9 @@ reference repo contains no FFTPower golden vectors and pmesh/pfft/mpi4py
 cannot be imported or built in this environment.
 """
 from .mesh import MeshGeometry, r2c, c2r, complex_coords, real_coords
-from .paint import paint, WINDOW_SUPPORT
+from .paint import paint, readout, WINDOW_SUPPORT
 from .catalogmesh import to_real_field
 from .fftpower import (compensation_filter, apply_compensation,
                       compute_3d_power, project_to_basis,
                       fftpower_oracle, fftcorr_oracle)
+from .fftrecon import fftrecon_oracle

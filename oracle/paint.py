@@ -89,3 +89,39 @@ def paint(position, mass, mesh, geom, resampler='cic', shift=0.0):
         iz = numpy.remainder(iz.astype('i8'), N[2])
         numpy.add.at(mesh, (ix, iy, iz), w * m)
     return mesh
+
+
+def readout(position, mesh, geom, resampler='cic'):
+    """Windowed gather — the dual of :func:`paint` (pmesh ``readout``;
+    FFTRecon reads displacements with the default CIC window,
+    fftrecon.py:246-249; the LogNormal generator uses 'nnb',
+    mockmaker.py:317-319)."""
+    position = numpy.asarray(position, dtype='f8')
+    if len(position) == 0:
+        return numpy.zeros(0)
+    u = position / geom.H
+    N = geom.Nmesh
+    out = numpy.zeros(len(u))
+
+    if resampler == 'nnb':
+        ix = numpy.remainder(numpy.floor(u[:, 0] + 0.5).astype('i8'), N[0])
+        iy = numpy.remainder(numpy.floor(u[:, 1] + 0.5).astype('i8'), N[1])
+        iz = numpy.remainder(numpy.floor(u[:, 2] + 0.5).astype('i8'), N[2])
+        return mesh[ix, iy, iz].astype('f8')
+
+    if resampler == 'cic':
+        gen = _cic_offsets_weights(u)
+    elif resampler == 'tsc':
+        gen = _centered_offsets_weights(u, (-1, 0, 1), _tsc_w, nearest=True)
+    elif resampler == 'pcs':
+        gen = _centered_offsets_weights(u, (-1, 0, 1, 2), _pcs_w,
+                                        nearest=False)
+    else:
+        raise ValueError("unknown resampler '%s'" % resampler)
+
+    for ix, iy, iz, w in gen:
+        ix = numpy.remainder(ix.astype('i8'), N[0])
+        iy = numpy.remainder(iy.astype('i8'), N[1])
+        iz = numpy.remainder(iz.astype('i8'), N[2])
+        out += w * mesh[ix, iy, iz]
+    return out

@@ -244,3 +244,65 @@ def test_fftcorr_mode_validation(upos):
     from oracle import fftcorr_oracle
     with pytest.raises(ValueError):
         fftcorr_oracle(upos, Nmesh=8, BoxSize=512., mode='3d')
+
+
+# ---- readout + FFTRecon oracle (reference fftrecon.py) ------------------
+
+def test_readout_at_grid_points():
+    """CIC/nnb readout exactly at grid points returns the mesh values
+    (TSC/PCS smooth over neighbours even there); every window is a
+    partition of unity, so a constant field reads back constant
+    anywhere."""
+    from oracle import readout
+    geom = MeshGeometry(8, 16.)
+    rng = numpy.random.RandomState(5)
+    mesh = rng.normal(size=(8, 8, 8))
+    ii = numpy.stack(numpy.meshgrid(*[numpy.arange(8)] * 3,
+                                    indexing='ij'), axis=-1).reshape(-1, 3)
+    pos = ii * (16. / 8)
+    for resampler in ('cic', 'nnb'):
+        vals = readout(pos, mesh, geom, resampler=resampler)
+        assert_allclose(vals, mesh.reshape(-1), rtol=1e-12, atol=1e-12)
+    anywhere = rng.uniform(0, 16., size=(200, 3))
+    const = numpy.full((8, 8, 8), 3.25)
+    for resampler in ('cic', 'tsc', 'pcs', 'nnb'):
+        vals = readout(anywhere, const, geom, resampler=resampler)
+        assert_allclose(vals, 3.25, rtol=1e-12)
+
+
+def test_readout_paint_adjoint():
+    """<paint(pos, m), mesh> == <m, readout(pos, mesh)> — paint and
+    readout are adjoint (same window weights)"""
+    from oracle import readout
+    geom = MeshGeometry(16, 32.)
+    rng = numpy.random.RandomState(6)
+    pos = rng.uniform(0, 32., size=(500, 3))
+    m = rng.uniform(0.5, 1.5, size=500)
+    mesh = rng.normal(size=(16, 16, 16))
+    for resampler in ('cic', 'tsc', 'pcs'):
+        painted = numpy.zeros((16, 16, 16))
+        paint(pos, m, painted, geom, resampler=resampler)
+        lhs = (painted * mesh).sum()
+        rhs = (m * readout(pos, mesh, geom, resampler=resampler)).sum()
+        assert_allclose(lhs, rhs, rtol=1e-12)
+
+
+def test_fftrecon_data_equals_randoms_is_null():
+    """LGS with data == randoms: identical shifts => identically zero"""
+    from oracle import fftrecon_oracle
+    pos = uniform_positions(2e-3, 64., seed=9)
+    out = fftrecon_oracle(pos, pos, Nmesh=16, BoxSize=64., bias=1.0,
+                          f=0.0, R=10.0, scheme='LGS')
+    assert_allclose(out, 0.0, atol=1e-12)
+
+
+def test_fftrecon_schemes_run_and_mean_zero():
+    from oracle import fftrecon_oracle
+    data = uniform_positions(2e-3, 64., seed=9)
+    ran = uniform_positions(4e-3, 64., seed=10)
+    for scheme in ('LGS', 'LF2', 'LRR'):
+        out = fftrecon_oracle(data, ran, Nmesh=16, BoxSize=64., bias=1.2,
+                              f=0.3, R=10.0, scheme=scheme)
+        # delta fields: mean ~ 0 (data mean 1 minus randoms mean 1)
+        assert abs(out.mean()) < 1e-10
+        assert numpy.isfinite(out).all()
