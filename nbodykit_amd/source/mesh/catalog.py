@@ -342,6 +342,19 @@ class CatalogMesh(MeshSource):
         return route_particles(pos_t, mass_t, self.pm, dmin, dmax)
 
 
+    @property
+    def actions(self):
+        actions = MeshSource.actions.fget(self)
+        if self.compensated:
+            actions = self._get_compensation() + actions
+        return actions
+
+    def _get_compensation(self):
+        return get_compensation(self.interlaced, self.resampler)
+
+
+
+
 def route_particles(pos_t, mass_t, pm, dmin, dmax):
     """Duplicate each particle to every rank whose x-slab any of its
     deposit cells falls in (the pmesh decompose/exchange step, reference
@@ -401,16 +414,6 @@ def paint_raw(pos_t, pm, resampler='cic'):
     return field
 
     # -- compensation actions (reference :405-451) ------------------------
-    @property
-    def actions(self):
-        actions = MeshSource.actions.fget(self)
-        if self.compensated:
-            actions = self._get_compensation() + actions
-        return actions
-
-    def _get_compensation(self):
-        return get_compensation(self.interlaced, self.resampler)
-
 
 def get_compensation(interlaced, resampler):
     """(mode, filter, kind) action for the window compensation
