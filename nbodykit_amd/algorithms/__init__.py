@@ -1,3 +1,4 @@
-from .fftpower import FFTPower, FFTBase, project_to_basis
+from .fftpower import (FFTPower, FFTBase, ProjectedFFTPower,
+                       project_to_basis)
 from .fftcorr import FFTCorr
 from .fftrecon import FFTRecon

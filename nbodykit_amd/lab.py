@@ -10,7 +10,7 @@ from nbodykit_amd import CurrentMPIComm, setup_logging, set_options
 from nbodykit_amd import cosmology
 from nbodykit_amd.cosmology import Cosmology, Planck15, LinearPower
 from nbodykit_amd.algorithms import (FFTPower, FFTCorr, FFTRecon,
-                                     project_to_basis)
+                                     ProjectedFFTPower, project_to_basis)
 from nbodykit_amd.source.catalog import (UniformCatalog, RandomCatalog,
                                          LogNormalCatalog, ArrayCatalog)
 from nbodykit_amd.source.mesh import CatalogMesh, FieldMesh

@@ -159,3 +159,30 @@ def test_set_options_paint_chunk_size():
     assert _global_options['paint_chunk_size'] == old
     with pytest.raises(KeyError):
         set_options(bogus=1)
+
+
+def test_compensation_action_wiring():
+    """compensated meshes must carry the compensation as their first
+    action (source/mesh/catalog.py:405-451) — guards against the action
+    pipeline silently dropping it (GPU parity then fails wholesale)."""
+    from nbodykit_amd.source.mesh.catalog import (
+        CompensateTSC, CompensateCICShotnoise, get_compensation,
+        lookup_compensation)
+    cat = UniformCatalog(nbar=1e-4, BoxSize=64., seed=42)
+    mesh = cat.to_mesh(Nmesh=8, resampler='tsc', compensated=True,
+                       interlaced=True)
+    actions = mesh.actions
+    assert len(actions) == 1
+    assert actions[0] == ('complex', CompensateTSC, 'circular')
+
+    mesh2 = cat.to_mesh(Nmesh=8, resampler='cic', compensated=True)
+    assert mesh2.actions[0] == ('complex', CompensateCICShotnoise,
+                                'circular')
+    assert cat.to_mesh(Nmesh=8, compensated=False).actions == []
+
+    # every builtin filter dispatches to the HIP kernel
+    for interlaced in (False, True):
+        for res in ('cic', 'tsc', 'pcs'):
+            (mode, func, kind), = get_compensation(interlaced, res)
+            assert mode == 'complex' and kind == 'circular'
+            assert lookup_compensation(func) == (res, interlaced)
