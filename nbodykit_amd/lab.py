@@ -18,3 +18,4 @@ from nbodykit_amd.base.catalog import CatalogSource
 from nbodykit_amd.base.mesh import MeshSource
 from nbodykit_amd.binned_statistic import BinnedStatistic
 from nbodykit_amd.pm import ParticleMesh, RealField, ComplexField
+from nbodykit_amd import transform

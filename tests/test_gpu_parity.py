@@ -399,3 +399,27 @@ def test_projected_fftpower():
     got = r.power['power']
     ok = numpy.isfinite(want.real) & (numpy.abs(want) > 0)
     assert_allclose(got[ok].real, want[ok].real, rtol=1e-10)
+
+
+def test_lab_end_to_end(tmp_path):
+    """The reference's full-pipeline smoke (nbodykit/tests/test_lab.py:
+    11-27): EH LinearPower -> LogNormalCatalog -> RSD shift via
+    transform.VectorProjection -> FFTPower(mode='2d', poles=[0,2,4]) ->
+    JSON save/load."""
+    from nbodykit_amd.lab import cosmology, transform
+    from nbodykit_amd.lab import LogNormalCatalog as LNC
+    cosmo = cosmology.Planck15
+    Plin = cosmology.LinearPower(cosmo, redshift=0.55,
+                                 transfer='EisensteinHu')
+    source = LNC(Plin=Plin, nbar=3e-5, BoxSize=690., Nmesh=16, seed=42)
+
+    source['Position'] = numpy.asarray(source['Position']) \
+        + transform.VectorProjection(source['VelocityOffset'], [0, 0, 1])
+
+    result = FFTPower(source, mode='2d', Nmesh=32, poles=[0, 2, 4],
+                      los=[0, 0, 1])
+    out = str(tmp_path / 'test_fftpower.json')
+    result.save(out)
+    back = FFTPower.load(out)
+    assert_array_equal(result.poles['power_2'], back.poles['power_2'])
+    assert result.attrs['N1'] == source.csize
