@@ -151,9 +151,24 @@ class MeshSource(object):
                 var.apply(**kwargs)
 
         var = var.cast() if isinstance(var, ComplexField) else var
-        pm = self.pm.reshape(Nmesh=Nmesh)
-        if any(pm.Nmesh != self.pm.Nmesh):
-            raise NotImplementedError("resampling to a different Nmesh")
+
+        if Nmesh is not None and any(
+                numpy.asarray(Nmesh).ravel() != self.pm.Nmesh):
+            # resample by copying overlapping Fourier modes
+            # (reference base/mesh.py:320-330)
+            from nbodykit_amd.pm import ParticleMesh, spectral_resample
+            new_pm = ParticleMesh(BoxSize=self.pm.BoxSize, Nmesh=Nmesh,
+                                  dtype=self.dtype, comm=self.comm)
+            want_real = isinstance(var, RealField)
+            if want_real:
+                var = var.r2c(out=Ellipsis)
+            var = spectral_resample(var, new_pm)
+            if want_real:
+                var = var.c2r(out=Ellipsis)
+            if self.comm.rank == 0:
+                self.logger.info('%s resampling from %s to %s done'
+                                 % (str(self), str(self.pm.Nmesh),
+                                    str(new_pm.Nmesh)))
 
         var.attrs = attrs
         var.attrs.update(self.attrs)
