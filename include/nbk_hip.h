@@ -88,6 +88,16 @@ int nbk_paint_f64(const double* pos, const double* mass, int64_t n,
                   double* mesh, int64_t x0, int64_t nx_local,
                   void* stream);
 
+/* paint for cell-sorted input: TSC/PCS deposits accumulate in an LDS
+ * window per contiguous particle run and flush once (CIC falls through
+ * to nbk_paint_f64).  Same semantics as nbk_paint_f64; the caller must
+ * know the input is cell-ordered (the bucket sort's output is). */
+int nbk_paint_sorted_f64(const double* pos, const double* mass, int64_t n,
+                         const int64_t nmesh[3], const double box[3],
+                         int window, double shift,
+                         double* mesh, int64_t x0, int64_t nx_local,
+                         void* stream);
+
 /* paint locality sort --------------------------------------------------
  * Two-pass counting sort of particles by coarse mesh cell
  * (bucket = wrapped ix * n1 + iy): count, then (after the caller turns

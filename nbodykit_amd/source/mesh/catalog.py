@@ -69,7 +69,7 @@ def _prepare_particles(pos_t, mass_t, pm):
     import torch
     n = len(pos_t)
     if n < _SORT_MIN_N:
-        return pos_t.t().contiguous(), mass_t
+        return pos_t.t().contiguous(), mass_t, False
 
     lib = hiplib.require()
     n0, n1, n2 = (int(x) for x in pm.Nmesh)
@@ -87,7 +87,7 @@ def _prepare_particles(pos_t, mass_t, pm):
     if int(flag.item()) == 0:
         # already cell-ordered (the count kernel checked lane-adjacent
         # pairs): skip the scatter, just transpose to SoA
-        return pos_t.t().contiguous(), mass_t
+        return pos_t.t().contiguous(), mass_t, True
     offsets = torch.zeros(ncells, dtype=torch.int64, device='cuda')
     offsets[1:] = torch.cumsum(counts[:-1].long(), 0)  # exclusive prefix
     out_soa = torch.empty(3 * n, dtype=torch.float64, device='cuda')
@@ -98,7 +98,7 @@ def _prepare_particles(pos_t, mass_t, pm):
         hiplib.dptr(pos_in), hiplib.dptr(mass_t), n, nmesh, box,
         hiplib.dptr(offsets), hiplib.dptr(out_soa),
         hiplib.dptr(out_mass), stream), 'nbk_bucket_scatter_f64')
-    return out_soa, out_mass
+    return out_soa, out_mass, True
 
 
 def _is_trivial_unit(col):
@@ -282,20 +282,22 @@ class CatalogMesh(MeshSource):
 
             n = len(pos_t)
             if n > 0:
-                pos_soa, mass_t = _prepare_particles(pos_t, mass_t, pm)
+                pos_soa, mass_t, sorted_ = _prepare_particles(pos_t,
+                                                              mass_t, pm)
+                paint_fn = lib.nbk_paint_sorted_f64 if sorted_                     else lib.nbk_paint_f64
                 with profiling.collect('paint', n * (1 + interlaced)):
-                    hiplib.check(lib.nbk_paint_f64(
+                    hiplib.check(paint_fn(
                         hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n,
                         nmesh, box, window_id, 0.0,
                         hiplib.dptr((real1 if interlaced else toret).value),
-                        pm.x_start, pm.nx_local, stream), 'nbk_paint_f64')
+                        pm.x_start, pm.nx_local, stream), 'nbk_paint')
                     if interlaced:
-                        hiplib.check(lib.nbk_paint_f64(
+                        hiplib.check(paint_fn(
                             hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n,
                             nmesh, box, window_id, 0.5,
                             hiplib.dptr(real2.value),
                             pm.x_start, pm.nx_local, stream),
-                            'nbk_paint_f64')
+                            'nbk_paint')
             i = i + chunksize
 
         if interlaced:
@@ -405,12 +407,13 @@ def paint_raw(pos_t, pm, resampler='cic'):
     if pm.comm.size > 1:
         dmin, dmax = _GHOST_RANGE[(resampler, False)]
         pos_t, _ = route_particles(pos_t, None, pm, dmin, dmax)
-    pos_soa, _ = _prepare_particles(pos_t, None, pm)
-    hiplib.check(lib.nbk_paint_f64(
+    pos_soa, _, sorted_ = _prepare_particles(pos_t, None, pm)
+    paint_fn = lib.nbk_paint_sorted_f64 if sorted_ else lib.nbk_paint_f64
+    hiplib.check(paint_fn(
         hiplib.dptr(pos_soa), None, len(pos_t),
         hiplib.i64_arr(pm.Nmesh), hiplib.f64_arr(pm.BoxSize),
         hiplib.WINDOW_IDS[resampler], 0.0, hiplib.dptr(field.value),
-        pm.x_start, pm.nx_local, hiplib.cur_stream()), 'nbk_paint_f64')
+        pm.x_start, pm.nx_local, hiplib.cur_stream()), 'nbk_paint')
     return field
 
     # -- compensation actions (reference :405-451) ------------------------

@@ -350,3 +350,48 @@ def test_compensate_partitioned_matches_full(lib):
             0, 0, None), 'c')
         got[:, y0:y0 + 8] = host(blk)
     assert_allclose(got, want, rtol=1e-14, atol=1e-14)
+
+
+@pytest.mark.parametrize('window', ['tsc', 'pcs'])
+@pytest.mark.parametrize('shift', [0.0, 0.5])
+def test_paint_sorted_matches_oracle(lib, window, shift):
+    """LDS-windowed sorted-paint path == oracle, incl. the interlacing
+    shift and box-edge wraps (input sorted by cell like the bucket
+    sort's output)"""
+    geom = MeshGeometry(16, 32.)
+    pos = rand_positions(20000, 32., seed=31)
+    cell = ((pos / 2.0).astype('i8') * [16 * 16, 16, 1]).sum(axis=1)
+    pos = pos[numpy.argsort(cell)]
+    mass = numpy.random.RandomState(32).uniform(0.5, 1.5, size=len(pos))
+
+    want = numpy.zeros((16, 16, 16))
+    paint(pos, mass, want, geom, resampler=window, shift=shift)
+
+    mesh_t = torch.zeros((16, 16, 16), dtype=torch.float64, device='cuda')
+    pos_soa = dev(pos).t().contiguous()
+    mass_t = dev(mass)
+    hiplib.check(lib.nbk_paint_sorted_f64(
+        hiplib.dptr(pos_soa), hiplib.dptr(mass_t), len(pos),
+        hiplib.i64_arr(geom.Nmesh), hiplib.f64_arr(geom.BoxSize),
+        hiplib.WINDOW_IDS[window], shift,
+        hiplib.dptr(mesh_t), 0, 16, None), 'paint_sorted')
+    assert_allclose(host(mesh_t), want, rtol=1e-12, atol=1e-12)
+
+
+def test_paint_sorted_slab_bounds(lib):
+    geom = MeshGeometry(16, 32.)
+    pos = rand_positions(8000, 32., seed=33)
+    cell = ((pos / 2.0).astype('i8') * [16 * 16, 16, 1]).sum(axis=1)
+    pos = pos[numpy.argsort(cell)]
+    full = numpy.zeros((16, 16, 16))
+    paint(pos, 1.0, full, geom, resampler='tsc')
+    got = numpy.zeros((16, 16, 16))
+    pos_soa = dev(pos).t().contiguous()
+    for x0 in (0, 8):
+        slab = torch.zeros((8, 16, 16), dtype=torch.float64, device='cuda')
+        hiplib.check(lib.nbk_paint_sorted_f64(
+            hiplib.dptr(pos_soa), None, len(pos),
+            hiplib.i64_arr(geom.Nmesh), hiplib.f64_arr(geom.BoxSize),
+            1, 0.0, hiplib.dptr(slab), x0, 8, None), 'paint_sorted')
+        got[x0:x0 + 8] = host(slab)
+    assert_allclose(got, full, rtol=1e-12, atol=1e-12)
