@@ -492,9 +492,13 @@ extern "C" int nbk_paint_sorted_f64(const double* pos, const double* mass,
                                     double shift, double* mesh, int64_t x0,
                                     int64_t nx_local, void* stream)
 {
-    // the LDS-window path pays off for the wide windows; CIC keeps the
-    // wave-merged direct kernel (already at 40% of the HBM roofline)
-    if (window == NBK_WINDOW_CIC || n == 0)
+    // Measured on the C3 input (1e8 TSC, bucket-sorted): the LDS-window
+    // path is 76 vs 45 ms against the wave-merged direct kernel — PMC
+    // shows all TSC variants 71-88% instruction-ISSUE-stalled, so
+    // rerouting deposits through LDS buys nothing there.  Only PCS
+    // (64 deposits, flush ~3x cheaper than direct) takes the tiled
+    // path; CIC and TSC keep the direct kernel.
+    if (window == NBK_WINDOW_CIC || window == NBK_WINDOW_TSC || n == 0)
         return nbk_paint_f64(pos, mass, n, nmesh, box, window, shift,
                              mesh, x0, nx_local, stream);
     if (window != NBK_WINDOW_TSC && window != NBK_WINDOW_PCS) {

@@ -204,6 +204,98 @@ __global__ void paint_tsc(const double* __restrict__ px,
     }
 }
 
+// tiled TSC (LDS window per sorted block) — mirror of kpaint_tiled
+#define PPB 1024
+#define MAXC 8192
+__global__ void paint_tsc_tiled(const double* __restrict__ px,
+                                const double* __restrict__ py,
+                                const double* __restrict__ pz, long n,
+                                long N, double invH,
+                                double* __restrict__ mesh)
+{
+    __shared__ double buf[MAXC];
+    __shared__ int s_min0, s_max0, s_min1, s_max1;
+    const long ibeg = (long)blockIdx.x * PPB;
+    const long iend = min(ibeg + PPB, n);
+    if (threadIdx.x == 0) {
+        s_min0 = INT_MAX; s_max0 = INT_MIN;
+        s_min1 = INT_MAX; s_max1 = INT_MIN;
+    }
+    __syncthreads();
+    int mn0 = INT_MAX, mx0 = INT_MIN, mn1 = INT_MAX, mx1 = INT_MIN;
+    for (long i = ibeg + threadIdx.x; i < iend; i += blockDim.x) {
+        const int b0 = (int)floor(px[i] * invH + 0.5) - 1;
+        const int b1 = (int)floor(py[i] * invH + 0.5) - 1;
+        mn0 = min(mn0, b0); mx0 = max(mx0, b0);
+        mn1 = min(mn1, b1); mx1 = max(mx1, b1);
+    }
+    atomicMin(&s_min0, mn0); atomicMax(&s_max0, mx0);
+    atomicMin(&s_min1, mn1); atomicMax(&s_max1, mx1);
+    __syncthreads();
+    const int wx = s_max0 - s_min0 + 3;
+    const int wy = s_max1 - s_min1 + 3;
+    const long cells = (long)wx * wy * N;
+    if (ibeg >= iend) return;
+    if (cells <= 0 || cells > MAXC) {
+        for (long i = ibeg + threadIdx.x; i < iend; i += blockDim.x) {
+            const double u0 = px[i]*invH, u1 = py[i]*invH, u2 = pz[i]*invH;
+            const double f0 = floor(u0+0.5), f1 = floor(u1+0.5), f2 = floor(u2+0.5);
+            const long b0=(long)f0-1, b1=(long)f1-1, b2=(long)f2-1;
+            double w0[3], w1[3], w2[3];
+            for (int d=0; d<3; d++) {
+                double s0=u0-(f0+d-1), s1=u1-(f1+d-1), s2=u2-(f2+d-1);
+                double a0=fabs(s0),a1=fabs(s1),a2=fabs(s2);
+                w0[d]=a0<0.5?0.75-s0*s0:0.5*(1.5-a0)*(1.5-a0);
+                w1[d]=a1<0.5?0.75-s1*s1:0.5*(1.5-a1)*(1.5-a1);
+                w2[d]=a2<0.5?0.75-s2*s2:0.5*(1.5-a2)*(1.5-a2);
+            }
+            for (int dx=0;dx<3;dx++){long gx=wrapi(b0+dx,N);
+              for(int dy=0;dy<3;dy++){long gy=wrapi(b1+dy,N);
+                double wxy=w0[dx]*w1[dy];
+                for(int dz=0;dz<3;dz++){long gz=wrapi(b2+dz,N);
+                  atomicAdd(&mesh[(gx*N+gy)*N+gz], wxy*w2[dz]);}}}
+        }
+        return;
+    }
+    for (long w = threadIdx.x; w < cells; w += blockDim.x) buf[w] = 0.0;
+    __syncthreads();
+    for (long i = ibeg + threadIdx.x; i < iend; i += blockDim.x) {
+        const double u0 = px[i]*invH, u1 = py[i]*invH, u2 = pz[i]*invH;
+        const double f0 = floor(u0+0.5), f1 = floor(u1+0.5), f2 = floor(u2+0.5);
+        const long b0=(long)f0-1, b1=(long)f1-1, b2=(long)f2-1;
+        double w0[3], w1[3], w2[3];
+        for (int d=0; d<3; d++) {
+            double s0=u0-(f0+d-1), s1=u1-(f1+d-1), s2=u2-(f2+d-1);
+            double a0=fabs(s0),a1=fabs(s1),a2=fabs(s2);
+            w0[d]=a0<0.5?0.75-s0*s0:0.5*(1.5-a0)*(1.5-a0);
+            w1[d]=a1<0.5?0.75-s1*s1:0.5*(1.5-a1)*(1.5-a1);
+            w2[d]=a2<0.5?0.75-s2*s2:0.5*(1.5-a2)*(1.5-a2);
+        }
+        const int lx = (int)(b0 - s_min0);
+        const int ly = (int)(b1 - s_min1);
+        for (int dx=0;dx<3;dx++)
+          for (int dy=0;dy<3;dy++){
+            const double wxy=w0[dx]*w1[dy];
+            const long base=((long)(lx+dx)*wy+(ly+dy))*N;
+            for (int dz=0;dz<3;dz++){
+              const long gz=wrapi(b2+dz,N);
+              unsafeAtomicAdd(&buf[base+gz], wxy*w2[dz]);
+            }
+          }
+    }
+    __syncthreads();
+    for (long w = threadIdx.x; w < cells; w += blockDim.x) {
+        const double v = buf[w];
+        if (v == 0.0) continue;
+        const long gz = w % N;
+        const long wyi = (w / N) % wy;
+        const long wxi = w / ((long)N * wy);
+        const long gx = wrapi(s_min0 + wxi, N);
+        const long gy = wrapi(s_min1 + wyi, N);
+        atomicAdd(&mesh[(gx*N+gy)*N+gz], v);
+    }
+}
+
 // variant: read-only (bandwidth leg)
 __global__ void read_only(const double* __restrict__ px,
                           const double* __restrict__ py,
@@ -333,6 +425,12 @@ int main(int argc, char** argv) {
         snprintf(buf, 128, "tsc plain  [%s]", mode ? "clumpy" : "uniform");
         report(buf, timeit([&] {
             hipLaunchKernelGGL((paint_tsc<false>), dim3(grid_for(n)),
+                               dim3(256), 0, 0, px, py, pz, n, N, invH, mesh);
+        }, 3));
+        snprintf(buf, 128, "tsc tiled  [%s]", mode ? "clumpy" : "uniform");
+        report(buf, timeit([&] {
+            hipLaunchKernelGGL(paint_tsc_tiled,
+                               dim3((unsigned)((n + PPB - 1) / PPB)),
                                dim3(256), 0, 0, px, py, pz, n, N, invH, mesh);
         }, 3));
     }
