@@ -204,6 +204,80 @@ __global__ void paint_tsc(const double* __restrict__ px,
     }
 }
 
+// hoisted-address TSC: per-axis wraps computed once, 32-bit row bases,
+// add-only inner loop — attacks the 71-88% instruction-issue stall
+template <bool MERGE>
+__global__ void paint_tsc_hoisted(const double* __restrict__ px,
+                                  const double* __restrict__ py,
+                                  const double* __restrict__ pz, long n,
+                                  long N, double invH,
+                                  double* __restrict__ mesh)
+{
+    const int lane = threadIdx.x & 63;
+    const long stride = (long)gridDim.x * blockDim.x;
+    const long wbase0 = blockIdx.x * (long)blockDim.x + (threadIdx.x & ~63);
+    const unsigned uN = (unsigned)N;
+    for (long wb = wbase0; wb < n; wb += stride) {
+        const long i = wb + lane;
+        const bool valid = i < n;
+        const double u0 = valid ? px[i]*invH : 0.0;
+        const double u1 = valid ? py[i]*invH : 0.0;
+        const double u2 = valid ? pz[i]*invH : 0.0;
+        const double f0 = floor(u0+0.5), f1 = floor(u1+0.5), f2 = floor(u2+0.5);
+        double w0[3], w1[3], w2[3];
+        #pragma unroll
+        for (int d=0; d<3; d++) {
+            double s0=u0-(f0+d-1), s1=u1-(f1+d-1), s2=u2-(f2+d-1);
+            double a0=fabs(s0),a1=fabs(s1),a2=fabs(s2);
+            w0[d]=a0<0.5?0.75-s0*s0:0.5*(1.5-a0)*(1.5-a0);
+            w1[d]=a1<0.5?0.75-s1*s1:0.5*(1.5-a1)*(1.5-a1);
+            w2[d]=a2<0.5?0.75-s2*s2:0.5*(1.5-a2)*(1.5-a2);
+        }
+        unsigned gx[3], gy[3], gz[3];
+        #pragma unroll
+        for (int d=0; d<3; d++) {
+            int v0 = (int)f0 - 1 + d; v0 = v0 < 0 ? v0 + (int)N : (v0 >= N ? v0 - (int)N : v0);
+            int v1 = (int)f1 - 1 + d; v1 = v1 < 0 ? v1 + (int)N : (v1 >= N ? v1 - (int)N : v1);
+            int v2 = (int)f2 - 1 + d; v2 = v2 < 0 ? v2 + (int)N : (v2 >= N ? v2 - (int)N : v2);
+            gx[d] = (unsigned)v0; gy[d] = (unsigned)v1; gz[d] = (unsigned)v2;
+        }
+        #pragma unroll
+        for (int dx=0; dx<3; dx++) {
+            const unsigned xbase = gx[dx] * uN;
+            #pragma unroll
+            for (int dy=0; dy<3; dy++) {
+                const unsigned rb = (xbase + gy[dy]) * uN;
+                const double wxy = w0[dx]*w1[dy];
+                #pragma unroll
+                for (int dz=0; dz<3; dz++) {
+                    const double val = wxy*w2[dz];
+                    if (MERGE) {
+                        long long addr = valid ? (long long)(rb + gz[dz])
+                                               : (long long)(-1 - lane);
+                        // head-bounded segmented merge
+                        const long long a_up1 = __shfl_up(addr, 1, 64);
+                        const bool head = (lane == 0) || (a_up1 != addr);
+                        const unsigned long long heads = __ballot(head);
+                        const unsigned long long below = heads & (~0ULL >> (63 - lane));
+                        const int myhead = 63 - __clzll(below);
+                        double v = val;
+                        #pragma unroll
+                        for (int d = 1; d < 64; d <<= 1) {
+                            const double v_up = __shfl_up(v, d, 64);
+                            if (lane - d >= myhead) v += v_up;
+                        }
+                        const bool nh = (lane < 63) && ((heads >> (lane+1)) & 1ULL);
+                        if (addr >= 0 && (lane == 63 || nh))
+                            atomicAdd(&mesh[addr], v);
+                    } else if (valid) {
+                        atomicAdd(&mesh[rb + gz[dz]], val);
+                    }
+                }
+            }
+        }
+    }
+}
+
 // tiled TSC (LDS window per sorted block) — mirror of kpaint_tiled
 #define PPB 1024
 #define MAXC 8192
@@ -425,6 +499,16 @@ int main(int argc, char** argv) {
         snprintf(buf, 128, "tsc plain  [%s]", mode ? "clumpy" : "uniform");
         report(buf, timeit([&] {
             hipLaunchKernelGGL((paint_tsc<false>), dim3(grid_for(n)),
+                               dim3(256), 0, 0, px, py, pz, n, N, invH, mesh);
+        }, 3));
+        snprintf(buf, 128, "tsc hoist-plain [%s]", mode ? "clumpy" : "uniform");
+        report(buf, timeit([&] {
+            hipLaunchKernelGGL((paint_tsc_hoisted<false>), dim3(grid_for(n)),
+                               dim3(256), 0, 0, px, py, pz, n, N, invH, mesh);
+        }, 3));
+        snprintf(buf, 128, "tsc hoist-merge [%s]", mode ? "clumpy" : "uniform");
+        report(buf, timeit([&] {
+            hipLaunchKernelGGL((paint_tsc_hoisted<true>), dim3(grid_for(n)),
                                dim3(256), 0, 0, px, py, pz, n, N, invH, mesh);
         }, 3));
         snprintf(buf, 128, "tsc tiled  [%s]", mode ? "clumpy" : "uniform");
