@@ -30,12 +30,6 @@ std::map<int64_t, double*> twiddle_tables;   // N -> device W_N^[0..N/2]
 
 bool is_pow2(int64_t n) { return n > 0 && (n & (n - 1)) == 0; }
 
-int ilog2(int64_t n) {
-    int l = 0;
-    while ((int64_t(1) << l) < n) l++;
-    return l;
-}
-
 // returns device pointer to W_N table (N/2+1 cdouble), or nullptr on error
 double* get_twiddles(int64_t N) {
     std::lock_guard<std::mutex> lock(table_mutex);
