@@ -103,17 +103,23 @@ int nbk_paint_sorted_f64(const double* pos, const double* mass, int64_t n,
  * (bucket = wrapped ix * n1 + iy): count, then (after the caller turns
  * counts into exclusive offsets) scatter into SoA output.  Replaces a
  * general radix sort in the paint driver: the deposit kernel wants
- * bucket-local order, not a total order.  `pos_aos` is the (n,3)
- * row-major input; `offsets` (n0*n1 int64) is consumed (atomically
- * advanced) by the scatter.  mass may be NULL.
+ * bucket-local order, not a total order.  Bucket = cell >> shift
+ * (shift 0 = per-cell; a coarse first level keeps both passes'
+ * scattered atomics L2-resident at 1024^3).  `pos_aos` is the (n,3)
+ * row-major input; `offsets` (ncells >> shift, int64) is consumed
+ * (atomically advanced) by the scatter; soa_out selects x/y/z planes
+ * (paint layout) vs AoS rows (the two-level intermediate).
+ * mass may be NULL.
  */
 int nbk_bucket_count_f64(const double* pos_aos, int64_t n,
                          const int64_t nmesh[3], const double box[3],
+                         int shift,
                          int* counts, int* scrambled_flag, void* stream);
 int nbk_bucket_scatter_f64(const double* pos_aos, const double* mass,
                            int64_t n, const int64_t nmesh[3],
-                           const double box[3], int64_t* offsets,
-                           double* pos_soa_out, double* mass_out,
+                           const double box[3], int shift, int soa_out,
+                           int64_t* offsets,
+                           double* pos_out, double* mass_out,
                            void* stream);
 
 /* readout (gather dual of paint; window 0/1/2 = cic/tsc/pcs, 3 = nnb).

@@ -15,11 +15,12 @@ namespace {
 __device__ __forceinline__ int64_t bucket_of(double x, double y, double z,
                                              double invH0, double invH1,
                                              double invH2, int64_t n0,
-                                             int64_t n1, int64_t n2) {
+                                             int64_t n1, int64_t n2,
+                                             int shift) {
     const int64_t ix = wrap_idx((int64_t)floor(x * invH0), n0);
     const int64_t iy = wrap_idx((int64_t)floor(y * invH1), n1);
     const int64_t iz = wrap_idx((int64_t)floor(z * invH2), n2);
-    return (ix * n1 + iy) * n2 + iz;
+    return ((ix * n1 + iy) * n2 + iz) >> shift;
 }
 
 // also detects (heuristically, lane-adjacent pairs) whether the input
@@ -28,6 +29,7 @@ __device__ __forceinline__ int64_t bucket_of(double x, double y, double z,
 __global__ void kbucket_count(const double* __restrict__ pos, int64_t n,
                               int64_t n0, int64_t n1, int64_t n2,
                               double invH0, double invH1, double invH2,
+                              int shift,
                               int* __restrict__ counts,
                               int* __restrict__ scrambled_flag)
 {
@@ -38,7 +40,7 @@ __global__ void kbucket_count(const double* __restrict__ pos, int64_t n,
          i < n; i += stride) {
         const int64_t b = bucket_of(pos[3 * i], pos[3 * i + 1],
                                     pos[3 * i + 2], invH0, invH1, invH2,
-                                    n0, n1, n2);
+                                    n0, n1, n2, shift);
         atomicAdd(&counts[b], 1);
         const int64_t b_up = __shfl_up((long long)b, 1, 64);
         if (lane > 0 && b_up > b) out_of_order = 1;
@@ -46,10 +48,14 @@ __global__ void kbucket_count(const double* __restrict__ pos, int64_t n,
     if (out_of_order) atomicOr(scrambled_flag, 1);
 }
 
+// SOA=1: write x/y/z planes (the paint layout); SOA=0: AoS rows (the
+// intermediate layout of the two-level sort's coarse pass)
+template <int SOA>
 __global__ void kbucket_scatter(const double* __restrict__ pos,
                                 const double* __restrict__ mass, int64_t n,
                                 int64_t n0, int64_t n1, int64_t n2,
                                 double invH0, double invH1, double invH2,
+                                int shift,
                                 int64_t* __restrict__ offsets,
                                 double* __restrict__ ox,
                                 double* __restrict__ oy,
@@ -61,12 +67,18 @@ __global__ void kbucket_scatter(const double* __restrict__ pos,
          i < n; i += stride) {
         const double x = pos[3 * i], y = pos[3 * i + 1], z = pos[3 * i + 2];
         const int64_t b = bucket_of(x, y, z, invH0, invH1, invH2,
-                                    n0, n1, n2);
+                                    n0, n1, n2, shift);
         const int64_t t = atomicAdd((unsigned long long*)&offsets[b],
                                     (unsigned long long)1);
-        ox[t] = x;
-        oy[t] = y;
-        oz[t] = z;
+        if (SOA) {
+            ox[t] = x;
+            oy[t] = y;
+            oz[t] = z;
+        } else {
+            ox[3 * t] = x;
+            ox[3 * t + 1] = y;
+            ox[3 * t + 2] = z;
+        }
         if (mass) om[t] = mass[i];
     }
 }
@@ -82,7 +94,7 @@ int sgrid(int64_t n) {
 
 extern "C" int nbk_bucket_count_f64(const double* pos_aos, int64_t n,
                                     const int64_t nmesh[3],
-                                    const double box[3],
+                                    const double box[3], int shift,
                                     int* counts, int* scrambled_flag,
                                     void* stream)
 {
@@ -91,7 +103,7 @@ extern "C" int nbk_bucket_count_f64(const double* pos_aos, int64_t n,
                        (hipStream_t)stream, pos_aos, n,
                        nmesh[0], nmesh[1], nmesh[2],
                        nmesh[0] / box[0], nmesh[1] / box[1],
-                       nmesh[2] / box[2], counts, scrambled_flag);
+                       nmesh[2] / box[2], shift, counts, scrambled_flag);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }
@@ -99,19 +111,29 @@ extern "C" int nbk_bucket_count_f64(const double* pos_aos, int64_t n,
 extern "C" int nbk_bucket_scatter_f64(const double* pos_aos,
                                       const double* mass, int64_t n,
                                       const int64_t nmesh[3],
-                                      const double box[3],
+                                      const double box[3], int shift,
+                                      int soa_out,
                                       int64_t* offsets,
-                                      double* pos_soa_out,
+                                      double* pos_out,
                                       double* mass_out, void* stream)
 {
     if (n == 0) return NBK_OK;
-    hipLaunchKernelGGL(kbucket_scatter, dim3(sgrid(n)), dim3(256), 0,
-                       (hipStream_t)stream, pos_aos, mass, n,
-                       nmesh[0], nmesh[1], nmesh[2],
-                       nmesh[0] / box[0], nmesh[1] / box[1],
-                       nmesh[2] / box[2], offsets,
-                       pos_soa_out, pos_soa_out + n, pos_soa_out + 2 * n,
-                       mass_out);
+    hipStream_t s = (hipStream_t)stream;
+    if (soa_out)
+        hipLaunchKernelGGL(kbucket_scatter<1>, dim3(sgrid(n)), dim3(256),
+                           0, s, pos_aos, mass, n,
+                           nmesh[0], nmesh[1], nmesh[2],
+                           nmesh[0] / box[0], nmesh[1] / box[1],
+                           nmesh[2] / box[2], shift, offsets,
+                           pos_out, pos_out + n, pos_out + 2 * n,
+                           mass_out);
+    else
+        hipLaunchKernelGGL(kbucket_scatter<0>, dim3(sgrid(n)), dim3(256),
+                           0, s, pos_aos, mass, n,
+                           nmesh[0], nmesh[1], nmesh[2],
+                           nmesh[0] / box[0], nmesh[1] / box[1],
+                           nmesh[2] / box[2], shift, offsets,
+                           pos_out, nullptr, nullptr, mass_out);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

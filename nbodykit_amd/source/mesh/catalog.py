@@ -49,22 +49,25 @@ def _is_trivial_true(sel):
 
 
 _SORT_MIN_N = 1 << 21        # below this the deposit is cheap either way
+_SORT_L2_CELLS = 1 << 24     # per-cell counts array <= 64 MiB: one level
+_SORT_COARSE_BUCKETS = 8192  # coarse-level bucket count target
 
 
 def _prepare_particles(pos_t, mass_t, pm):
-    """Return (pos_soa, mass) for the deposit kernel, bucket-sorting the
-    chunk by coarse mesh cell (ix, iy) when it arrives scrambled.
+    """Return (pos_soa, mass, cell_sorted) for the deposit kernel,
+    bucket-sorting the chunk by mesh cell when it arrives scrambled.
 
     The deposit kernel's wave-merge and its L2 locality both depend on
     nearby-in-space particles being nearby-in-memory; a cell-ordered
     chunk paints ~4x faster than a scrambled one (C4: 206 -> 48 ms
     kernel — the Zel'dovich shift scrambles even generator-ordered
-    catalogs across x-planes).  The two-pass counting sort
-    (nbk_bucket_count/scatter) needs only bucket-local order and emits
-    SoA directly, replacing both a radix argsort and the AoS->SoA
-    transpose.  Near-sorted inputs skip the sort after a cheap
-    monotonicity check.  Part of the timed paint path — nothing is
-    cached across calls.
+    catalogs across x-planes).  Counting sort (nbk_bucket_count/scatter)
+    emits SoA directly; for big meshes it runs TWO levels — a coarse
+    pass into ~8k buckets first — so the fine pass's scattered atomics
+    and writes stay L2-resident instead of thrashing a multi-GB counts
+    array.  Cell-ordered inputs skip everything after the first count
+    (in-flight order detection).  Part of the timed paint path — nothing
+    is cached across calls.
     """
     import torch
     n = len(pos_t)
@@ -78,26 +81,58 @@ def _prepare_particles(pos_t, mass_t, pm):
     nmesh = hiplib.i64_arr(pm.Nmesh)
     box = hiplib.f64_arr(pm.BoxSize)
     stream = hiplib.cur_stream()
-    counts = torch.zeros(ncells, dtype=torch.int32, device='cuda')
-    flag = torch.zeros(1, dtype=torch.int32, device='cuda')
+
+    # coarse level when the per-cell counts array would bust the L2
+    shift = 0
+    if ncells > _SORT_L2_CELLS:
+        shift = max(0, (ncells // _SORT_COARSE_BUCKETS).bit_length())
+
+    def count(pos_in, nb, sh, detect):
+        counts = torch.zeros(nb, dtype=torch.int32, device='cuda')
+        flag = torch.zeros(1, dtype=torch.int32, device='cuda')
+        hiplib.check(lib.nbk_bucket_count_f64(
+            hiplib.dptr(pos_in), n, nmesh, box, sh, hiplib.dptr(counts),
+            hiplib.dptr(flag), stream), 'nbk_bucket_count_f64')
+        scrambled = True
+        if detect:
+            scrambled = int(flag.item()) != 0
+        return counts, scrambled
+
+    def exclusive(counts):
+        offsets = torch.zeros(len(counts), dtype=torch.int64,
+                              device='cuda')
+        offsets[1:] = torch.cumsum(counts[:-1].long(), 0)
+        return offsets
+
+    def scatter(pos_in, m_in, offsets, sh, soa):
+        out = torch.empty(3 * n, dtype=torch.float64, device='cuda')
+        out_m = None
+        if m_in is not None:
+            out_m = torch.empty(n, dtype=torch.float64, device='cuda')
+        hiplib.check(lib.nbk_bucket_scatter_f64(
+            hiplib.dptr(pos_in), hiplib.dptr(m_in), n, nmesh, box, sh,
+            int(soa), hiplib.dptr(offsets), hiplib.dptr(out),
+            hiplib.dptr(out_m), stream), 'nbk_bucket_scatter_f64')
+        return out, out_m
+
     pos_in = pos_t.contiguous()
-    hiplib.check(lib.nbk_bucket_count_f64(
-        hiplib.dptr(pos_in), n, nmesh, box, hiplib.dptr(counts),
-        hiplib.dptr(flag), stream), 'nbk_bucket_count_f64')
-    if int(flag.item()) == 0:
-        # already cell-ordered (the count kernel checked lane-adjacent
-        # pairs): skip the scatter, just transpose to SoA
+
+    counts, scrambled = count(pos_in, ncells >> shift, shift, detect=True)
+    if not scrambled:
+        # already cell-ordered: no scatter needed
         return pos_t.t().contiguous(), mass_t, True
-    offsets = torch.zeros(ncells, dtype=torch.int64, device='cuda')
-    offsets[1:] = torch.cumsum(counts[:-1].long(), 0)  # exclusive prefix
-    out_soa = torch.empty(3 * n, dtype=torch.float64, device='cuda')
-    out_mass = None
-    if mass_t is not None:
-        out_mass = torch.empty(n, dtype=torch.float64, device='cuda')
-    hiplib.check(lib.nbk_bucket_scatter_f64(
-        hiplib.dptr(pos_in), hiplib.dptr(mass_t), n, nmesh, box,
-        hiplib.dptr(offsets), hiplib.dptr(out_soa),
-        hiplib.dptr(out_mass), stream), 'nbk_bucket_scatter_f64')
+
+    if shift > 0:
+        # coarse grouping pass (AoS out), then the fine per-cell pass on
+        # coarsely-local data
+        coarse, mass_c = scatter(pos_in, mass_t, exclusive(counts),
+                                 shift, soa=False)
+        pos_in = coarse
+        mass_t = mass_c
+        counts, _ = count(pos_in, ncells, 0, detect=False)
+
+    out_soa, out_mass = scatter(pos_in, mass_t, exclusive(counts), 0,
+                                soa=True)
     return out_soa, out_mass, True
 
 
