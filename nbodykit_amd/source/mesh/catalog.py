@@ -49,8 +49,6 @@ def _is_trivial_true(sel):
 
 
 _SORT_MIN_N = 1 << 21        # below this the deposit is cheap either way
-_SORT_L2_CELLS = 1 << 24     # per-cell counts array <= 64 MiB: one level
-_SORT_COARSE_BUCKETS = 8192  # coarse-level bucket count target
 
 
 def _prepare_particles(pos_t, mass_t, pm):
@@ -82,10 +80,12 @@ def _prepare_particles(pos_t, mass_t, pm):
     box = hiplib.f64_arr(pm.BoxSize)
     stream = hiplib.cur_stream()
 
-    # coarse level when the per-cell counts array would bust the L2
+    # NOTE: a two-level variant (coarse ~4k-bucket pass first) was
+    # measured 10x SLOWER at C4 — the coarse scatter's returned
+    # fetch-adds pile onto a few thousand hot counters and serialize
+    # (~2.1 s/step vs 113 ms single-level).  Single level stands until
+    # the ticket atomics are block-aggregated.
     shift = 0
-    if ncells > _SORT_L2_CELLS:
-        shift = max(0, (ncells // _SORT_COARSE_BUCKETS).bit_length())
 
     def count(pos_in, nb, sh, detect):
         counts = torch.zeros(nb, dtype=torch.int32, device='cuda')
