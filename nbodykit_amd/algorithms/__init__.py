@@ -2,3 +2,4 @@ from .fftpower import (FFTPower, FFTBase, ProjectedFFTPower,
                        project_to_basis)
 from .fftcorr import FFTCorr
 from .fftrecon import FFTRecon
+from .convpower import ConvolvedFFTPower, FKPCatalog, FKPWeightFromNbar

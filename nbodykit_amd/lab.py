@@ -19,3 +19,6 @@ from nbodykit_amd.base.mesh import MeshSource
 from nbodykit_amd.binned_statistic import BinnedStatistic
 from nbodykit_amd.pm import ParticleMesh, RealField, ComplexField
 from nbodykit_amd import transform
+from nbodykit_amd.algorithms.convpower import (ConvolvedFFTPower,
+                                               FKPCatalog,
+                                               FKPWeightFromNbar)

@@ -25,3 +25,4 @@ from .fftpower import (compensation_filter, apply_compensation,
                       compute_3d_power, project_to_basis,
                       fftpower_oracle, fftcorr_oracle)
 from .fftrecon import fftrecon_oracle
+from .convpower import convpower_oracle, real_Ylm
