@@ -181,18 +181,42 @@ class MeshSource(object):
         return field.preview(Nmesh, axes=axes)
 
     def save(self, output, dataset='Field', mode='real'):
-        """Save the computed field; bigfile is out of scope,
-        so this writes a .npy + a JSON attrs sidecar."""
+        """Save the mesh as a bigfile readable by
+        :class:`~nbodykit_amd.source.mesh.bigfile.BigFileMesh`
+        (reference base/mesh.py:444-500): the raveled local field rows
+        ordered by rank, with ``ndarray.shape``/``BoxSize``/``Nmesh``
+        attrs and the remaining attrs JSON-encoded."""
         import json
+        import warnings
+        from nbodykit_amd.io.bigfile_format import BigFile
         from nbodykit_amd.utils import JSONEncoder
+
         field = self.compute(mode=mode)
-        full = field.preview() if isinstance(field, RealField) else None
-        if self.comm.rank == 0:
-            if full is None:
-                raise NotImplementedError("save(mode='complex')")
-            numpy.save(output, full)
-            with open(output + '.attrs.json', 'w') as ff:
-                json.dump(dict(field.attrs), ff, cls=JSONEncoder)
+        data = field.value.cpu().numpy().ravel()
+
+        with BigFile(output, create=True, comm=self.comm) as ff:
+            bb = ff.create_from_array(dataset, data)
+            cshape = [int(n) for n in self.pm.Nmesh]
+            if mode == 'complex':
+                cshape[-1] = cshape[-1] // 2 + 1
+            bb.attrs['ndarray.shape'] = numpy.asarray(cshape, dtype='i8')
+            bb.attrs['BoxSize'] = self.pm.BoxSize
+            bb.attrs['Nmesh'] = self.pm.Nmesh
+            for key in field.attrs:
+                if key in bb.attrs:
+                    continue
+                value = field.attrs[key]
+                try:
+                    bb.attrs[key] = value
+                except (ValueError, TypeError):
+                    try:
+                        bb.attrs[key] = 'json://' + json.dumps(
+                            value, cls=JSONEncoder)
+                    except Exception:
+                        warnings.warn(
+                            "attribute %s of type %s is unsupported and "
+                            "lost while saving MeshSource"
+                            % (key, type(value)))
 
 
 class MeshFilter(object):

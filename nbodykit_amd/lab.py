@@ -14,6 +14,8 @@ from nbodykit_amd.algorithms import (FFTPower, FFTCorr, FFTRecon,
 from nbodykit_amd.source.catalog import (UniformCatalog, RandomCatalog,
                                          LogNormalCatalog, ArrayCatalog)
 from nbodykit_amd.source.mesh import CatalogMesh, FieldMesh
+from nbodykit_amd.source.mesh.bigfile import BigFileMesh
+from nbodykit_amd.source.catalog.bigfile import BigFileCatalog
 from nbodykit_amd.base.catalog import CatalogSource
 from nbodykit_amd.base.mesh import MeshSource
 from nbodykit_amd.binned_statistic import BinnedStatistic

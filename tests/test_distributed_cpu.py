@@ -209,3 +209,52 @@ def _body_fft_scheme(comm):
 def test_gloo_distributed_fft_scheme():
     r0, r1 = _run_world('_body_fft_scheme')
     assert r0 and r1
+
+
+def _body_bigfile(comm):
+    """Collective catalog save + partitioned load (reference
+    base/catalog.py:562-695, source/catalog/file.py:67-90)."""
+    import shutil
+    import tempfile
+    from nbodykit_amd.lab import ArrayCatalog, BigFileCatalog
+
+    path = comm.bcast(tempfile.mkdtemp() if comm.rank == 0 else None)
+    try:
+        n = 10 + 5 * comm.rank        # unequal local sizes
+        base = 100 * comm.rank
+        cat = ArrayCatalog({
+            'Position': (numpy.arange(3 * n, dtype='f8')
+                         .reshape(n, 3) + base),
+            'Mass': numpy.arange(n, dtype='f8') + base}, comm=comm)
+        cat.attrs['BoxSize'] = numpy.array([7., 7., 7.])
+        cat.save(path + '/cat')
+        comm.barrier()
+
+        loaded = BigFileCatalog(path + '/cat', comm=comm)
+        full_pos = numpy.concatenate(comm.allgather(
+            numpy.asarray(loaded['Position'])))
+        full_mass = numpy.concatenate(comm.allgather(
+            numpy.asarray(loaded['Mass'])))
+        return dict(csize=loaded.csize, size=loaded.size,
+                    pos=full_pos, mass=full_mass,
+                    box=loaded.attrs['BoxSize'])
+    finally:
+        comm.barrier()
+        if comm.rank == 0:
+            shutil.rmtree(path, ignore_errors=True)
+
+
+@pytest.mark.timeout(300)
+def test_gloo_bigfile_catalog_roundtrip():
+    r0, r1 = _run_world('_body_bigfile')
+    # global order = rank 0 rows then rank 1 rows
+    expect_mass = numpy.concatenate([numpy.arange(10, dtype='f8'),
+                                     numpy.arange(15, dtype='f8') + 100])
+    for r, size in zip((r0, r1), (12, 13)):  # 25 rows split 12/13
+        assert r['csize'] == 25 and r['size'] == size
+        numpy.testing.assert_array_equal(r['mass'], expect_mass)
+        numpy.testing.assert_array_equal(
+            r['pos'][:, 0], numpy.concatenate([
+                numpy.arange(10) * 3.0,
+                numpy.arange(15) * 3.0 + 100]))
+        numpy.testing.assert_array_equal(r['box'], [7., 7., 7.])
