@@ -216,6 +216,28 @@ int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
                       double* xsum, double* musum, double* Nsum,
                       double* ysum, void* stream);
 
+/* fused compensate + cross-power + binning: one streaming pass
+ * computing comp1(c1) * conj(comp2(c2)) * volume per element (zero mode
+ * cleared but still binned, fftpower.py:114-128) and accumulating the
+ * project_to_basis sums — replaces the nbk_compensate_f64 (x2) +
+ * nbk_power3d_f64 + nbk_bin_power_f64 sequence without materializing
+ * p3d.  window1/window2: NBK_WINDOW_* or -1 for no compensation; c2 may
+ * equal c1 (auto power) or be NULL (alias of c1).  The compensation
+ * factors are bit-identical to nbk_compensate_f64's. */
+int nbk_power_bin_f64(const double* c1, const double* c2, double volume,
+                      int window1, int interlaced1,
+                      int window2, int interlaced2,
+                      int clear_zero_mode,
+                      const int64_t nmesh[3], const double box[3],
+                      const int64_t dims[3], const int64_t off[3],
+                      const int axis_map[3],
+                      const double* kedges, int64_t nx_edges,
+                      const double* muedges, int64_t nmu_edges,
+                      const double los[3],
+                      const int* ells, int nell,
+                      double* xsum, double* musum, double* Nsum,
+                      double* ysum, void* stream);
+
 /* small helpers ------------------------------------------------------ */
 /* out[i] += a[i]  (f64, n elements) */
 int nbk_axpy_f64(double* out, const double* a, double alpha, int64_t n,

@@ -103,6 +103,35 @@ class FFTBase(object):
         attrs['shotnoise'] = Pshot
         return p3d, attrs
 
+    def _compute_complex_pair(self):
+        """The complex fields computed WITHOUT the compensation action
+        (the fused nbk_power_bin_f64 kernel applies compensation on the
+        fly), plus the same attrs _compute_3d_power returns.  Only valid
+        when _fuse_info accepted both meshes."""
+        attrs = {}
+        attrs.update(self.attrs)
+
+        def _compute_raw(mesh):
+            saved = mesh.compensated
+            mesh.compensated = False
+            try:
+                return mesh.compute(mode='complex',
+                                    Nmesh=self.attrs['Nmesh'])
+            finally:
+                mesh.compensated = saved
+
+        c1 = _compute_raw(self.first)
+        c2 = c1 if self.second is self.first else _compute_raw(self.second)
+
+        attrs.update({'N1': c1.attrs.get('N', 0),
+                      'N2': c2.attrs.get('N', 0)})
+        Pshot = 0
+        if self.first is self.second:
+            if 'shotnoise' in c1.attrs:
+                Pshot = c1.attrs['shotnoise']
+        attrs['shotnoise'] = Pshot
+        return c1, c2, attrs
+
 
 class FFTPower(FFTBase):
     """Periodic-box 1d/2d power spectrum and multipoles via FFT
@@ -144,7 +173,18 @@ class FFTPower(FFTBase):
         if self.attrs['mode'] == '1d':
             self.attrs['Nmu'] = 1
 
-        y3d, attrs = self._compute_3d_power(self.first, self.second)
+        # the fused single-pass path: compensation + cross power + binning
+        # in one kernel, p3d never materialized (nbk_power_bin_f64); falls
+        # back to the explicit sequence for meshes with user actions
+        fuse1 = _fuse_info(self.first)
+        fuse2 = fuse1 if self.second is self.first \
+            else _fuse_info(self.second)
+        fused = fuse1 is not None and fuse2 is not None
+        if fused:
+            c1, c2, attrs = self._compute_complex_pair()
+            y3d = c1     # coordinate/metadata source only
+        else:
+            y3d, attrs = self._compute_3d_power(self.first, self.second)
 
         dk = self.attrs['dk']
         kmin = self.attrs['kmin']
@@ -163,9 +203,15 @@ class FFTPower(FFTBase):
                                  endpoint=True)
         edges = [kedges, muedges]
         coords = [kcoords, None]
-        result, pole_result = project_to_basis(y3d, edges,
-                                               poles=self.attrs['poles'],
-                                               los=self.attrs['los'])
+        if fused:
+            result, pole_result = _project_power_fused(
+                c1, c2, fuse1, fuse2,
+                volume=self.attrs['BoxSize'].prod(), edges=edges,
+                poles=self.attrs['poles'], los=self.attrs['los'])
+        else:
+            result, pole_result = project_to_basis(
+                y3d, edges, poles=self.attrs['poles'],
+                los=self.attrs['los'])
 
         # pack the structured arrays (reference :306-334)
         if self.attrs['mode'] == '1d':
@@ -377,7 +423,14 @@ def project_to_basis(y3d, edges, los=[0, 0, 1], poles=[]):
     torch.cuda.synchronize()
     host = sums.cpu().numpy()
     host = comm.allreduce(host)
+    return _fold_bin_sums(host, Nx, Nmu, Nell, ell_idx, do_poles,
+                          real_field)
 
+
+def _fold_bin_sums(host, Nx, Nmu, Nell, ell_idx, do_poles, real_field):
+    """Host tail of project_to_basis: unpack the kernel's planar sums,
+    fold the internal mu == 1 bin, normalize (fftpower.py:669-701)."""
+    NB = (Nx + 2) * (Nmu + 2)
     shape = (Nx + 2, Nmu + 2)
     xsum = host[:NB].reshape(shape)
     musum = host[NB:2 * NB].reshape(shape)
@@ -414,6 +467,87 @@ def project_to_basis(y3d, edges, los=[0, 0, 1], poles=[]):
             pole_result = None
 
     return (xmean_2d, mumean_2d, y2d, N_2d), pole_result
+
+
+def _fuse_info(mesh):
+    """(window_id, interlaced) for the fused compensate+power+bin kernel,
+    or None when the mesh cannot take the fused path.  Fusable: a plain
+    CatalogMesh with no user actions, whose only k-space work is the
+    auto-prepended compensation (source/mesh/catalog.py:405-451);
+    window -1 encodes compensated=False."""
+    from nbodykit_amd.base.mesh import MeshSource
+    from nbodykit_amd.source.mesh.catalog import (CatalogMesh,
+                                                  lookup_compensation)
+    if not isinstance(mesh, CatalogMesh):
+        return None
+    if len(MeshSource.actions.fget(mesh)) != 0:
+        return None
+    if not mesh.compensated:
+        return (-1, 0)
+    try:
+        actions = mesh._get_compensation()
+    except ValueError:
+        return None
+    comp = lookup_compensation(actions[0][1])
+    if comp is None:
+        return None
+    window, interlaced = comp
+    return (hiplib.WINDOW_IDS[window], int(interlaced))
+
+
+def _project_power_fused(c1, c2, comp1, comp2, volume, edges,
+                         los=[0, 0, 1], poles=[]):
+    """project_to_basis of comp1(c1) conj(comp2(c2)) V with the zero mode
+    cleared, in ONE streaming pass over the complex field(s)
+    (nbk_power_bin_f64) — the compensate/power3d/bin sequence without
+    materializing p3d.  Numerically identical to the unfused path: the
+    kernel shares the per-element compensation code with
+    nbk_compensate_f64."""
+    import torch
+    comm = c1.pm.comm
+    lib = hiplib.require()
+
+    xedges, muedges = edges
+    Nx = len(xedges) - 1
+    Nmu = len(muedges) - 1
+
+    poles = list(poles)
+    do_poles = len(poles) > 0
+    _poles = [0] + sorted(poles) if 0 not in poles else sorted(poles)
+    ell_idx = [_poles.index(l) for l in poles]
+    Nell = len(_poles)
+    if any(ell < 0 for ell in _poles):
+        raise ValueError("in `project_to_basis`, multipole numbers must "
+                         "be non-negative integers")
+
+    NB = (Nx + 2) * (Nmu + 2)
+    nfields = 3 + 2 * Nell
+
+    dev = 'cuda'
+    k2edges_t = torch.as_tensor(numpy.asarray(xedges, dtype='f8') ** 2) \
+        .to(dev)
+    muedges_t = torch.as_tensor(numpy.asarray(muedges, dtype='f8')).to(dev)
+    sums = torch.zeros(nfields * NB, dtype=torch.float64, device=dev)
+
+    win1, interl1 = comp1
+    win2, interl2 = comp2
+    hiplib.check(lib.nbk_power_bin_f64(
+        hiplib.dptr(c1.value), hiplib.dptr(c2.value), float(volume),
+        int(win1), int(interl1), int(win2), int(interl2), 1,
+        hiplib.i64_arr(c1.pm.Nmesh), hiplib.f64_arr(c1.pm.BoxSize),
+        hiplib.i64_arr(c1.dims), hiplib.i64_arr(c1.off), None,
+        hiplib.dptr(k2edges_t), len(xedges),
+        hiplib.dptr(muedges_t), len(muedges),
+        hiplib.f64_arr(los), hiplib.int_arr(_poles), Nell,
+        hiplib.dptr(sums), hiplib.dptr(sums[NB:]),
+        hiplib.dptr(sums[2 * NB:]), hiplib.dptr(sums[3 * NB:]),
+        hiplib.cur_stream()), 'nbk_power_bin_f64')
+
+    torch.cuda.synchronize()
+    host = sums.cpu().numpy()
+    host = comm.allreduce(host)
+    return _fold_bin_sums(host, Nx, Nmu, Nell, ell_idx, do_poles,
+                          real_field=False)
 
 
 def _cast_source(source, BoxSize, Nmesh):

@@ -32,6 +32,14 @@ struct BinArgs {
                                // full-length, no Hermitian double-count
     int ells[NBK_MAX_ELL];     // multipole orders, by value (kernel args
                                // live in SGPRs — no host pointer deref)
+    // fused compensate+power mode (nbk_power_bin_f64): the element value
+    // is comp1(c1) * conj(comp2(c2)) * volume computed on the fly instead
+    // of a pre-materialized p3d — one read pass replaces the
+    // compensate/power3d/bin trio (window -1 = no compensation)
+    int fuse;
+    int win1, interl1, win2, interl2;
+    int clear_zero;
+    double volume;
 };
 
 // numpy.digitize(x, edges) == count of edges <= x (right-open bins)
@@ -63,7 +71,8 @@ __device__ __forceinline__ void accum(double* __restrict__ h, int NB,
 }
 
 template <bool LDS>
-__global__ void kbin(const double* __restrict__ data, BinArgs A,
+__global__ void kbin(const double* __restrict__ data,
+                     const double* __restrict__ data2, BinArgs A,
                      const double* __restrict__ k2edges,
                      const double* __restrict__ muedges,
                      double* __restrict__ gout /* nfields * NB */)
@@ -109,9 +118,35 @@ __global__ void kbin(const double* __restrict__ data, BinArgs A,
         const int bmu = dig(muedges, A.nmu_edges, mu);
         const int bin = bx * (A.nmu_edges + 1) + bmu;
 
-        const cdouble v = A.real_field
-            ? cdouble{data[idx], 0.0}
-            : cdouble{data[2 * idx], data[2 * idx + 1]};
+        cdouble v;
+        if (A.fuse) {
+            // comp1(c1) * conj(comp2(c2)) * V, zero mode cleared but
+            // still binned (fftpower.py:114-128); compensation factors
+            // bit-identical to the standalone nbk_compensate_f64 pass
+            const double w[3] = {2.0 * M_PI * fx / (double)A.n0,
+                                 2.0 * M_PI * fy / (double)A.n1,
+                                 2.0 * M_PI * fz / (double)A.n2};
+            cdouble a = {data[2 * idx], data[2 * idx + 1]};
+            cdouble b = (data2 == data) ? a
+                : cdouble{data2[2 * idx], data2[2 * idx + 1]};
+            if (A.win1 >= 0)
+                a = cscale(a, nbk_comp_factor(A.win1, A.interl1, w));
+            if (A.win2 >= 0) {
+                if (data2 == data && A.win2 == A.win1
+                    && A.interl2 == A.interl1)
+                    b = a;
+                else
+                    b = cscale(b, nbk_comp_factor(A.win2, A.interl2, w));
+            }
+            cdouble p = cmul(a, cconj(b));
+            v = {p.re * A.volume, p.im * A.volume};
+            if (A.clear_zero && fx == 0.0 && fy == 0.0 && fz == 0.0)
+                v = {0.0, 0.0};
+        } else {
+            v = A.real_field
+                ? cdouble{data[idx], 0.0}
+                : cdouble{data[2 * idx], data[2 * idx + 1]};
+        }
 
         // Legendre P_ell(mu) by recurrence, ells ascending with ells[0]==0
         cdouble yv[NBK_MAX_ELL];
@@ -145,6 +180,19 @@ __global__ void kbin(const double* __restrict__ data, BinArgs A,
     }
 }
 
+// shared launcher for the plain (pre-materialized p3d) and fused
+// (compensate+power on the fly) binning passes
+static int launch_bin(const double* d1, const double* d2, BinArgs& A,
+                      const double* kedges, int64_t nx_edges,
+                      const double* muedges, int64_t nmu_edges,
+                      const int64_t nmesh[3], const double box[3],
+                      const int64_t dims[3], const int64_t off[3],
+                      const int axis_map[3],
+                      const double los[3], const int* ells, int nell,
+                      int real_field,
+                      double* xsum, double* musum, double* Nsum,
+                      double* ysum, void* stream);
+
 }  // namespace
 
 extern "C" int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
@@ -158,6 +206,62 @@ extern "C" int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
                                  int real_field,
                                  double* xsum, double* musum, double* Nsum,
                                  double* ysum, void* stream)
+{
+    BinArgs A;
+    A.fuse = 0;
+    A.win1 = A.win2 = -1;
+    A.interl1 = A.interl2 = 0;
+    A.clear_zero = 0;
+    A.volume = 1.0;
+    return launch_bin(cplx, cplx, A, kedges, nx_edges, muedges, nmu_edges,
+                      nmesh, box, dims, off, axis_map, los, ells, nell,
+                      real_field, xsum, musum, Nsum, ysum, stream);
+}
+
+extern "C" int nbk_power_bin_f64(const double* c1, const double* c2,
+                                 double volume,
+                                 int window1, int interlaced1,
+                                 int window2, int interlaced2,
+                                 int clear_zero_mode,
+                                 const int64_t nmesh[3], const double box[3],
+                                 const int64_t dims[3], const int64_t off[3],
+                                 const int axis_map[3],
+                                 const double* kedges, int64_t nx_edges,
+                                 const double* muedges, int64_t nmu_edges,
+                                 const double los[3],
+                                 const int* ells, int nell,
+                                 double* xsum, double* musum, double* Nsum,
+                                 double* ysum, void* stream)
+{
+    if (window1 > 2 || window2 > 2) {
+        NBK_SET_ERR("nbk_power_bin_f64: bad window (%d, %d)",
+                    window1, window2);
+        return NBK_ERR_ARG;
+    }
+    BinArgs A;
+    A.fuse = 1;
+    A.win1 = window1; A.interl1 = interlaced1;
+    A.win2 = window2; A.interl2 = interlaced2;
+    A.clear_zero = clear_zero_mode;
+    A.volume = volume;
+    return launch_bin(c1, c2 ? c2 : c1, A, kedges, nx_edges, muedges,
+                      nmu_edges, nmesh, box, dims, off, axis_map, los,
+                      ells, nell, /*real_field=*/0,
+                      xsum, musum, Nsum, ysum, stream);
+}
+
+namespace {
+
+static int launch_bin(const double* d1, const double* d2, BinArgs& A,
+                      const double* kedges, int64_t nx_edges,
+                      const double* muedges, int64_t nmu_edges,
+                      const int64_t nmesh[3], const double box[3],
+                      const int64_t dims[3], const int64_t off[3],
+                      const int axis_map[3],
+                      const double los[3], const int* ells, int nell,
+                      int real_field,
+                      double* xsum, double* musum, double* Nsum,
+                      double* ysum, void* stream)
 {
     (void)musum; (void)Nsum; (void)ysum;
     if (nell > NBK_MAX_ELL) {
@@ -174,7 +278,6 @@ extern "C" int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
         return NBK_ERR_ARG;
     }
 
-    BinArgs A;
     A.n0 = nmesh[0]; A.n1 = nmesh[1]; A.n2 = nmesh[2];
     A.d0 = dims[0]; A.d1 = dims[1]; A.d2 = dims[2];
     A.o0 = off[0]; A.o1 = off[1]; A.o2 = off[2];
@@ -221,11 +324,13 @@ extern "C" int nbk_bin_power_f64(const double* cplx, const int64_t nmesh[3],
             raised = lds_bytes;
         }
         hipLaunchKernelGGL(kbin<true>, dim3((uint32_t)g), dim3(256),
-                           lds_bytes, s, cplx, A, kedges, muedges, xsum);
+                           lds_bytes, s, d1, d2, A, kedges, muedges, xsum);
     } else {
         hipLaunchKernelGGL(kbin<false>, dim3((uint32_t)g), dim3(256), 0, s,
-                           cplx, A, kedges, muedges, xsum);
+                           d1, d2, A, kedges, muedges, xsum);
     }
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }
+
+}  // namespace

@@ -55,3 +55,38 @@ __device__ __forceinline__ double freq_full(int64_t g, int64_t n) {
 __device__ __forceinline__ double freq_half(int64_t g, int64_t n) {
     return (double)(g == n / 2 ? -(n / 2) : g);
 }
+
+// Inverse-window compensation factor for one element (the six filters of
+// nbodykit/source/mesh/catalog.py:453-594).  w[i] = 2 pi f_i / N_i is the
+// circular frequency in [-pi, pi); interlaced selects the plain
+// 1/sinc^p inverse (Jing 2005 eq. 18), otherwise the first-order
+// aliasing-corrected eq. 20 forms.  Shared by the standalone
+// nbk_compensate_f64 pass and the fused nbk_power_bin_f64 so both paths
+// are bit-identical.
+__device__ __forceinline__ double nbk_comp_factor(int window, int interlaced,
+                                                  const double w[3]) {
+    double corr = 1.0;
+    #pragma unroll
+    for (int i = 0; i < 3; i++) {
+        if (interlaced) {
+            const double s = 0.5 * w[i];
+            const double sc = (s == 0.0) ? 1.0 : sin(s) / s;
+            const double p = (window == NBK_WINDOW_CIC) ? sc * sc
+                           : (window == NBK_WINDOW_TSC) ? sc * sc * sc
+                                                        : sc * sc * sc * sc;
+            corr /= p;
+        } else {
+            const double s2 = sin(0.5 * w[i]) * sin(0.5 * w[i]);
+            double d;
+            if (window == NBK_WINDOW_CIC)
+                d = 1.0 - 2.0 / 3.0 * s2;
+            else if (window == NBK_WINDOW_TSC)
+                d = 1.0 - s2 + 2.0 / 15.0 * s2 * s2;
+            else
+                d = 1.0 - 4.0 / 3.0 * s2 + 2.0 / 5.0 * s2 * s2
+                    - 4.0 / 315.0 * s2 * s2 * s2;
+            corr /= sqrt(d);
+        }
+    }
+    return corr;
+}

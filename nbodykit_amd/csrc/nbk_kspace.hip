@@ -31,12 +31,6 @@ __device__ __forceinline__ void global_freqs(const Layout& L, int64_t idx,
     f[2] = freq_half(g[2], n[2]);   // compressed axis, Nyquist negative
 }
 
-// sinc(w/2) = sin(w/2)/(w/2) with the w=0 guard (catalog.py:516-520)
-__device__ __forceinline__ double half_sinc(double w) {
-    const double s = 0.5 * w;
-    return s == 0.0 ? 1.0 : sin(s) / s;
-}
-
 __global__ void kcompensate(double* __restrict__ data, Layout L,
                             int window, int interlaced)
 {
@@ -46,33 +40,12 @@ __global__ void kcompensate(double* __restrict__ data, Layout L,
          idx < total; idx += stride) {
         double f[3];
         global_freqs(L, idx, f);
-        double corr = 1.0;
         const int64_t n[3] = {L.n0, L.n1, L.n2};
-        #pragma unroll
-        for (int i = 0; i < 3; i++) {
-            // circular frequency w = 2 pi f / N in [-pi, pi)
-            const double w = 2.0 * M_PI * f[i] / (double)n[i];
-            if (interlaced) {
-                // plain inverse window, Jing 2005 eq. 18, p = 2/3/4
-                const double s = half_sinc(w);
-                const double p = (window == NBK_WINDOW_CIC) ? s * s
-                               : (window == NBK_WINDOW_TSC) ? s * s * s
-                                                            : s * s * s * s;
-                corr /= p;
-            } else {
-                // first-order aliasing-corrected forms, eq. 20
-                const double s2 = sin(0.5 * w) * sin(0.5 * w);
-                double d;
-                if (window == NBK_WINDOW_CIC)
-                    d = 1.0 - 2.0 / 3.0 * s2;
-                else if (window == NBK_WINDOW_TSC)
-                    d = 1.0 - s2 + 2.0 / 15.0 * s2 * s2;
-                else
-                    d = 1.0 - 4.0 / 3.0 * s2 + 2.0 / 5.0 * s2 * s2
-                        - 4.0 / 315.0 * s2 * s2 * s2;
-                corr /= sqrt(d);
-            }
-        }
+        // circular frequency w = 2 pi f / N in [-pi, pi)
+        const double w[3] = {2.0 * M_PI * f[0] / (double)n[0],
+                             2.0 * M_PI * f[1] / (double)n[1],
+                             2.0 * M_PI * f[2] / (double)n[2]};
+        const double corr = nbk_comp_factor(window, interlaced, w);
         data[2 * idx] *= corr;
         data[2 * idx + 1] *= corr;
     }
