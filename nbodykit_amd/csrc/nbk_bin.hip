@@ -70,40 +70,64 @@ __device__ __forceinline__ void accum(double* __restrict__ h, int NB,
     }
 }
 
+// integer frequency of global index g on global axis `axis` (the
+// compressed half-spectrum convention applies to global axis 2 of a
+// complex field only)
+__device__ __forceinline__ double freq_axis(int axis, int64_t g, int64_t n,
+                                            int real_field) {
+    return (axis == 2 && !real_field) ? freq_half(g, n) : freq_full(g, n);
+}
+
 template <bool LDS>
 __global__ void kbin(const double* __restrict__ data,
                      const double* __restrict__ data2, BinArgs A,
-                     const double* __restrict__ k2edges,
-                     const double* __restrict__ muedges,
+                     const double* __restrict__ k2edges_g,
+                     const double* __restrict__ muedges_g,
                      double* __restrict__ gout /* nfields * NB */)
 {
     const int NB = (A.nx_edges + 1) * (A.nmu_edges + 1);
     const int nfields = 3 + 2 * A.nell;
 
+    // LDS: [histograms | k2 edges | mu edges] — the digitize binary
+    // search runs against LDS instead of chasing L1/L2 lines
     extern __shared__ double lh[];
     double* h = gout;
+    const double* k2edges = k2edges_g;
+    const double* muedges = muedges_g;
     if (LDS) {
         for (int i = threadIdx.x; i < NB * nfields; i += blockDim.x)
             lh[i] = 0.0;
+        double* ke = lh + NB * nfields;
+        double* me = ke + A.nx_edges;
+        for (int i = threadIdx.x; i < A.nx_edges; i += blockDim.x)
+            ke[i] = k2edges_g[i];
+        for (int i = threadIdx.x; i < A.nmu_edges; i += blockDim.x)
+            me[i] = muedges_g[i];
         __syncthreads();
         h = lh;
+        k2edges = ke;
+        muedges = me;
     }
 
-    const int64_t total = A.d0 * A.d1 * A.d2;
-    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-         idx < total; idx += stride) {
-        const int64_t l2 = idx % A.d2;
-        const int64_t l1 = (idx / A.d2) % A.d1;
-        const int64_t l0 = idx / (A.d2 * A.d1);
-        int64_t g[3];
-        g[A.a0] = l0 + A.o0;
-        g[A.a1] = l1 + A.o1;
-        g[A.a2] = l2 + A.o2;
-        const double fx = freq_full(g[0], A.n0);
-        const double fy = freq_full(g[1], A.n1);
-        const double fz = A.real_field ? freq_full(g[2], A.n2)
-                                       : freq_half(g[2], A.n2);
+    // one block per z-line (grid-stride over lines): the expensive
+    // int64 div/mod runs once per line, and the per-thread reads stay
+    // coalesced along the contiguous last axis
+    const int64_t n3[3] = {A.n0, A.n1, A.n2};
+    const int64_t nlines = A.d0 * A.d1;
+    for (int64_t line = blockIdx.x; line < nlines; line += gridDim.x) {
+        const int64_t l0 = line / A.d1;
+        const int64_t l1 = line - l0 * A.d1;
+        double f3[3];
+        f3[A.a0] = freq_axis(A.a0, l0 + A.o0, n3[A.a0], A.real_field);
+        f3[A.a1] = freq_axis(A.a1, l1 + A.o1, n3[A.a1], A.real_field);
+        const int64_t base = line * A.d2;
+
+    for (int64_t l2 = threadIdx.x; l2 < A.d2; l2 += blockDim.x) {
+        const int64_t idx = base + l2;
+        f3[A.a2] = freq_axis(A.a2, l2 + A.o2, n3[A.a2], A.real_field);
+        const double fx = f3[0];
+        const double fy = f3[1];
+        const double fz = f3[2];
 
         const double kx = fx * A.k0x, ky = fy * A.k0y, kz = fz * A.k0z;
         const double k2 = kx * kx + ky * ky + kz * kz;
@@ -171,6 +195,7 @@ __global__ void kbin(const double* __restrict__ data,
         }
 
         accum<LDS>(h, NB, bin, kmag, mu, w, yv, A.nell);
+    }
     }
 
     if (LDS) {
@@ -303,9 +328,10 @@ static int launch_bin(const double* d1, const double* d2, BinArgs& A,
         A.ells[e] = e < nell ? ells[e] : -1;
 
     const int nfields = 3 + 2 * nell;
-    const size_t lds_bytes = (size_t)NB * nfields * sizeof(double);
-    const int64_t total = A.d0 * A.d1 * A.d2;
-    int64_t g = (total + 255) / 256;
+    const size_t lds_bytes = ((size_t)NB * nfields + nx_edges + nmu_edges)
+                             * sizeof(double);
+    const int64_t nlines = A.d0 * A.d1;
+    int64_t g = nlines;
     if (g > 8192) g = 8192;    // bounded: LDS flush cost scales with grid
     if (g < 1) g = 1;
 
