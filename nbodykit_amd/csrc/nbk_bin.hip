@@ -109,21 +109,43 @@ __global__ void kbin(const double* __restrict__ data,
         muedges = me;
     }
 
-    // one block per z-line (grid-stride over lines): the expensive
-    // int64 div/mod runs once per line, and the per-thread reads stay
-    // coalesced along the contiguous last axis
+    // NL z-lines per block, flattened so every thread stays busy across
+    // line boundaries (a single line of n2/2+1 elements would leave the
+    // tail wave idle); the expensive int64 div/mod runs NL times per
+    // GROUP, cooperatively, with the per-line frequencies staged through
+    // a small static LDS array
     const int64_t n3[3] = {A.n0, A.n1, A.n2};
     const int64_t nlines = A.d0 * A.d1;
-    for (int64_t line = blockIdx.x; line < nlines; line += gridDim.x) {
-        const int64_t l0 = line / A.d1;
-        const int64_t l1 = line - l0 * A.d1;
-        double f3[3];
-        f3[A.a0] = freq_axis(A.a0, l0 + A.o0, n3[A.a0], A.real_field);
-        f3[A.a1] = freq_axis(A.a1, l1 + A.o1, n3[A.a1], A.real_field);
-        const int64_t base = line * A.d2;
+    constexpr int NL = 8;
+    __shared__ double lf0[NL], lf1[NL];
+    const int64_t ngroups = (nlines + NL - 1) / NL;
+    for (int64_t grp = blockIdx.x; grp < ngroups; grp += gridDim.x) {
+        const int64_t line0 = grp * NL;
+        __syncthreads();
+        if (threadIdx.x < NL && line0 + threadIdx.x < nlines) {
+            const int64_t line = line0 + threadIdx.x;
+            const int64_t l0 = line / A.d1;
+            const int64_t l1 = line - l0 * A.d1;
+            lf0[threadIdx.x] = freq_axis(A.a0, l0 + A.o0, n3[A.a0],
+                                         A.real_field);
+            lf1[threadIdx.x] = freq_axis(A.a1, l1 + A.o1, n3[A.a1],
+                                         A.real_field);
+        }
+        __syncthreads();
+        const int64_t nelem =
+            (nlines - line0 < NL ? nlines - line0 : (int64_t)NL) * A.d2;
+        int sub = 0;
+        int64_t l2 = threadIdx.x;
+        while (l2 >= A.d2) { l2 -= A.d2; sub++; }
 
-    for (int64_t l2 = threadIdx.x; l2 < A.d2; l2 += blockDim.x) {
-        const int64_t idx = base + l2;
+    for (int64_t i = threadIdx.x; i < nelem;
+         i += blockDim.x,
+         l2 += blockDim.x,
+         ({ while (l2 >= A.d2) { l2 -= A.d2; sub++; } })) {
+        const int64_t idx = (line0 + sub) * A.d2 + l2;
+        double f3[3];
+        f3[A.a0] = lf0[sub];
+        f3[A.a1] = lf1[sub];
         f3[A.a2] = freq_axis(A.a2, l2 + A.o2, n3[A.a2], A.real_field);
         const double fx = f3[0];
         const double fy = f3[1];
@@ -153,14 +175,19 @@ __global__ void kbin(const double* __restrict__ data,
             cdouble a = {data[2 * idx], data[2 * idx + 1]};
             cdouble b = (data2 == data) ? a
                 : cdouble{data2[2 * idx], data2[2 * idx + 1]};
-            if (A.win1 >= 0)
-                a = cscale(a, nbk_comp_factor(A.win1, A.interl1, w));
+            const bool same_comp = (A.win2 == A.win1
+                                    && A.interl2 == A.interl1);
+            double fac1 = 1.0;
+            if (A.win1 >= 0) {
+                fac1 = nbk_comp_factor(A.win1, A.interl1, w);
+                a = cscale(a, fac1);
+            }
             if (A.win2 >= 0) {
-                if (data2 == data && A.win2 == A.win1
-                    && A.interl2 == A.interl1)
+                if (data2 == data && same_comp)
                     b = a;
                 else
-                    b = cscale(b, nbk_comp_factor(A.win2, A.interl2, w));
+                    b = cscale(b, same_comp && A.win1 >= 0 ? fac1
+                               : nbk_comp_factor(A.win2, A.interl2, w));
             }
             cdouble p = cmul(a, cconj(b));
             v = {p.re * A.volume, p.im * A.volume};
