@@ -153,6 +153,12 @@ __global__ void kbin(const double* __restrict__ data,
 
         const double kx = fx * A.k0x, ky = fy * A.k0y, kz = fz * A.k0z;
         const double k2 = kx * kx + ky * ky + kz * kz;
+        // modes beyond the last edge land in the (Nx+1) overflow row,
+        // which project_to_basis discards (fftpower.py:666-668 keeps
+        // bins 1..Nx) — skip them before touching the field data.  At
+        // default edges (kmax ~ the min-axis Nyquist) this is every
+        // corner mode outside the inscribed sphere, ~48% of the volume.
+        if (k2 >= k2edges[A.nx_edges - 1]) continue;
         const double kmag = sqrt(k2);
         double mu = kx * A.losx + ky * A.losy + kz * A.losz;
         mu = (kmag == 0.0) ? 0.0 : mu / kmag;
