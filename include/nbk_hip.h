@@ -104,12 +104,11 @@ int nbk_paint_sorted_f64(const double* pos, const double* mass, int64_t n,
  * counts into exclusive offsets) scatter into SoA output.  Replaces a
  * general radix sort in the paint driver: the deposit kernel wants
  * bucket-local order, not a total order.  Bucket = cell >> shift
- * (shift 0 = per-cell; a coarse first level keeps both passes'
- * scattered atomics L2-resident at 1024^3).  `pos_aos` is the (n,3)
- * row-major input; `offsets` (ncells >> shift, int64) is consumed
- * (atomically advanced) by the scatter; soa_out selects x/y/z planes
- * (paint layout) vs AoS rows (the two-level intermediate).
- * mass may be NULL.
+ * (shift 0 = per-cell).  `pos_aos` is the (n,3) row-major input;
+ * `offsets` (ncells >> shift, int32) holds the INCLUSIVE bucket cumsum
+ * and is consumed (tickets count DOWN to the exclusive base) by the
+ * scatter; soa_out selects x/y/z planes (paint layout) vs AoS rows.
+ * mass may be NULL.  Requires n < 2^31 (int32 tickets).
  */
 int nbk_bucket_count_f64(const double* pos_aos, int64_t n,
                          const int64_t nmesh[3], const double box[3],
@@ -118,9 +117,28 @@ int nbk_bucket_count_f64(const double* pos_aos, int64_t n,
 int nbk_bucket_scatter_f64(const double* pos_aos, const double* mass,
                            int64_t n, const int64_t nmesh[3],
                            const double box[3], int shift, int soa_out,
-                           int64_t* offsets,
+                           int* offsets,
                            double* pos_out, double* mass_out,
                            void* stream);
+
+/* deterministic chunked pre-sort by wrapped x-plane (the coarse level
+ * of the two-level paint locality sort; no global atomics).  Pass A
+ * (`count`) writes one n0-wide histogram row per chunk of `chunk`
+ * particles to `mat` (ceil(n/chunk) x n0 int32, row-major) and performs
+ * the cell-order detection into scrambled_flag (may be NULL); the host
+ * exclusive-scans `mat` in bucket-major order into `bases`; pass C
+ * (`scatter`) re-reads each chunk and places rows into `pos_out` (AoS)
+ * from LDS cursors seeded with `bases`.  Running the cell sort above on
+ * this output keeps its scattered atomics within one x-plane's counter
+ * window (n1*n2*4 B).  Requires nmesh[0] <= 8192 and n < 2^31. */
+int nbk_xsort_count_f64(const double* pos_aos, int64_t n, int chunk,
+                        const int64_t nmesh[3], const double box[3],
+                        int* mat, int* scrambled_flag, void* stream);
+int nbk_xsort_scatter_f64(const double* pos_aos, const double* mass,
+                          int64_t n, int chunk, const int64_t nmesh[3],
+                          const double box[3], const int* bases,
+                          double* pos_out, double* mass_out,
+                          void* stream);
 
 /* readout (gather dual of paint; window 0/1/2 = cic/tsc/pcs, 3 = nnb).
  * Serves FFTRecon's displacement solve (fftrecon.py:246-249) and the

@@ -56,7 +56,7 @@ __global__ void kbucket_scatter(const double* __restrict__ pos,
                                 int64_t n0, int64_t n1, int64_t n2,
                                 double invH0, double invH1, double invH2,
                                 int shift,
-                                int64_t* __restrict__ offsets,
+                                int* __restrict__ offsets,
                                 double* __restrict__ ox,
                                 double* __restrict__ oy,
                                 double* __restrict__ oz,
@@ -68,8 +68,10 @@ __global__ void kbucket_scatter(const double* __restrict__ pos,
         const double x = pos[3 * i], y = pos[3 * i + 1], z = pos[3 * i + 2];
         const int64_t b = bucket_of(x, y, z, invH0, invH1, invH2,
                                     n0, n1, n2, shift);
-        const int64_t t = atomicAdd((unsigned long long*)&offsets[b],
-                                    (unsigned long long)1);
+        // offsets holds the INCLUSIVE bucket cumsum; tickets count DOWN
+        // to the exclusive base (saves the host-side shift/int64 cast —
+        // within-bucket order is arbitrary either way)
+        const int64_t t = (int64_t)(atomicSub(&offsets[b], 1) - 1);
         if (SOA) {
             ox[t] = x;
             oy[t] = y;
@@ -79,6 +81,71 @@ __global__ void kbucket_scatter(const double* __restrict__ pos,
             ox[3 * t + 1] = y;
             ox[3 * t + 2] = z;
         }
+        if (mass) om[t] = mass[i];
+    }
+}
+
+// ---- chunked x-plane pre-sort (the coarse level) ----------------------
+// Deterministic counting sort by wrapped x-plane index, with NO global
+// atomics: pass A writes one histogram row per chunk to a [nchunks x n0]
+// matrix, the host scans it bucket-major, and pass C places each chunk's
+// particles from LDS cursors seeded by the scanned bases.  Running the
+// fine cell sort on this output makes its scattered counter/ticket
+// atomics hit 64-byte lines ~16x each (one x-plane's counters span
+// n1*n2*4 B) instead of thrashing the whole multi-GB array.
+
+__global__ void kxsort_count(const double* __restrict__ pos, int64_t n,
+                             int chunk, int64_t n0, int64_t n1, int64_t n2,
+                             double invH0, double invH1, double invH2,
+                             int* __restrict__ mat,
+                             int* __restrict__ scrambled_flag)
+{
+    extern __shared__ int hist[];   // n0 ints
+    for (int b = threadIdx.x; b < n0; b += blockDim.x) hist[b] = 0;
+    __syncthreads();
+    const int64_t beg = (int64_t)blockIdx.x * chunk;
+    const int64_t end = (beg + chunk < n) ? beg + chunk : n;
+    const int lane = threadIdx.x & 63;
+    int out_of_order = 0;
+    for (int64_t i = beg + threadIdx.x; i < end; i += blockDim.x) {
+        const double x = pos[3 * i], y = pos[3 * i + 1],
+                     z = pos[3 * i + 2];
+        const int64_t ix = wrap_idx((int64_t)floor(x * invH0), n0);
+        atomicAdd(&hist[ix], 1);
+        if (scrambled_flag) {
+            const int64_t b = bucket_of(x, y, z, invH0, invH1, invH2,
+                                        n0, n1, n2, 0);
+            const int64_t b_up = __shfl_up((long long)b, 1, 64);
+            if (lane > 0 && b_up > b) out_of_order = 1;
+        }
+    }
+    if (out_of_order) atomicOr(scrambled_flag, 1);
+    __syncthreads();
+    for (int b = threadIdx.x; b < n0; b += blockDim.x)
+        mat[(int64_t)blockIdx.x * n0 + b] = hist[b];
+}
+
+__global__ void kxsort_scatter(const double* __restrict__ pos,
+                               const double* __restrict__ mass, int64_t n,
+                               int chunk, int64_t n0, double invH0,
+                               const int* __restrict__ bases,
+                               double* __restrict__ out,
+                               double* __restrict__ om)
+{
+    extern __shared__ int cur[];    // n0 running cursors
+    for (int b = threadIdx.x; b < n0; b += blockDim.x)
+        cur[b] = bases[(int64_t)blockIdx.x * n0 + b];
+    __syncthreads();
+    const int64_t beg = (int64_t)blockIdx.x * chunk;
+    const int64_t end = (beg + chunk < n) ? beg + chunk : n;
+    for (int64_t i = beg + threadIdx.x; i < end; i += blockDim.x) {
+        const double x = pos[3 * i], y = pos[3 * i + 1],
+                     z = pos[3 * i + 2];
+        const int64_t ix = wrap_idx((int64_t)floor(x * invH0), n0);
+        const int64_t t = (int64_t)atomicAdd(&cur[ix], 1);
+        out[3 * t] = x;
+        out[3 * t + 1] = y;
+        out[3 * t + 2] = z;
         if (mass) om[t] = mass[i];
     }
 }
@@ -113,7 +180,7 @@ extern "C" int nbk_bucket_scatter_f64(const double* pos_aos,
                                       const int64_t nmesh[3],
                                       const double box[3], int shift,
                                       int soa_out,
-                                      int64_t* offsets,
+                                      int* offsets,
                                       double* pos_out,
                                       double* mass_out, void* stream)
 {
@@ -134,6 +201,51 @@ extern "C" int nbk_bucket_scatter_f64(const double* pos_aos,
                            nmesh[0] / box[0], nmesh[1] / box[1],
                            nmesh[2] / box[2], shift, offsets,
                            pos_out, nullptr, nullptr, mass_out);
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}
+
+extern "C" int nbk_xsort_count_f64(const double* pos_aos, int64_t n,
+                                   int chunk, const int64_t nmesh[3],
+                                   const double box[3], int* mat,
+                                   int* scrambled_flag, void* stream)
+{
+    if (n == 0) return NBK_OK;
+    if (nmesh[0] > 8192) {
+        NBK_SET_ERR("nbk_xsort_count_f64: n0 > 8192 exceeds the LDS "
+                    "histogram");
+        return NBK_ERR_ARG;
+    }
+    const int64_t nblocks = (n + chunk - 1) / chunk;
+    const size_t lds = (size_t)nmesh[0] * sizeof(int);
+    hipLaunchKernelGGL(kxsort_count, dim3((uint32_t)nblocks), dim3(256),
+                       lds, (hipStream_t)stream, pos_aos, n, chunk,
+                       nmesh[0], nmesh[1], nmesh[2],
+                       nmesh[0] / box[0], nmesh[1] / box[1],
+                       nmesh[2] / box[2], mat, scrambled_flag);
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}
+
+extern "C" int nbk_xsort_scatter_f64(const double* pos_aos,
+                                     const double* mass, int64_t n,
+                                     int chunk, const int64_t nmesh[3],
+                                     const double box[3], const int* bases,
+                                     double* pos_out, double* mass_out,
+                                     void* stream)
+{
+    if (n == 0) return NBK_OK;
+    if (nmesh[0] > 8192) {
+        NBK_SET_ERR("nbk_xsort_scatter_f64: n0 > 8192 exceeds the LDS "
+                    "cursors");
+        return NBK_ERR_ARG;
+    }
+    const int64_t nblocks = (n + chunk - 1) / chunk;
+    const size_t lds = (size_t)nmesh[0] * sizeof(int);
+    hipLaunchKernelGGL(kxsort_scatter, dim3((uint32_t)nblocks), dim3(256),
+                       lds, (hipStream_t)stream, pos_aos, mass, n, chunk,
+                       nmesh[0], nmesh[0] / box[0], bases,
+                       pos_out, mass_out);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

@@ -19,6 +19,7 @@ EXPORTED_SYMBOLS = [
     'nbk_version', 'nbk_last_error_string', 'nbk_device_count',
     'nbk_paint_f64', 'nbk_paint_sorted_f64', 'nbk_readout_f64', 'nbk_recon_displacement_f64',
     'nbk_bucket_count_f64', 'nbk_bucket_scatter_f64',
+    'nbk_xsort_count_f64', 'nbk_xsort_scatter_f64',
     'nbk_fft_r2c_z', 'nbk_fft_c2r_z', 'nbk_fft_c_strided',
     'nbk_compensate_f64', 'nbk_interlace_combine_f64', 'nbk_power3d_f64',
     'nbk_bin_power_f64', 'nbk_power_bin_f64',
@@ -55,6 +56,14 @@ def _declare(lib):
     lib.nbk_bucket_scatter_f64.argtypes = [c_void, c_void, c_i64, c_i64_p,
                                            c_f64_p, c_void, c_void, c_void,
                                            c_void]
+    lib.nbk_xsort_count_f64.restype = ctypes.c_int
+    lib.nbk_xsort_count_f64.argtypes = [c_void, c_i64, ctypes.c_int,
+                                        c_i64_p, c_f64_p, c_void, c_void,
+                                        c_void]
+    lib.nbk_xsort_scatter_f64.restype = ctypes.c_int
+    lib.nbk_xsort_scatter_f64.argtypes = [c_void, c_void, c_i64,
+                                          ctypes.c_int, c_i64_p, c_f64_p,
+                                          c_void, c_void, c_void, c_void]
     lib.nbk_paint_sorted_f64.restype = ctypes.c_int
     lib.nbk_paint_sorted_f64.argtypes = lib.nbk_paint_f64.argtypes
     lib.nbk_readout_f64.restype = ctypes.c_int

@@ -60,12 +60,12 @@ def _prepare_particles(pos_t, mass_t, pm):
     chunk paints ~4x faster than a scrambled one (C4: 206 -> 48 ms
     kernel — the Zel'dovich shift scrambles even generator-ordered
     catalogs across x-planes).  Counting sort (nbk_bucket_count/scatter)
-    emits SoA directly; for big meshes it runs TWO levels — a coarse
-    pass into ~8k buckets first — so the fine pass's scattered atomics
-    and writes stay L2-resident instead of thrashing a multi-GB counts
-    array.  Cell-ordered inputs skip everything after the first count
-    (in-flight order detection).  Part of the timed paint path — nothing
-    is cached across calls.
+    emits SoA directly; big meshes first run the deterministic chunked
+    x-plane pre-sort (nbk_xsort_*, count-matrix based, no global
+    atomics) so the cell pass's scattered atomics stay within one
+    x-plane's line window.  Cell-ordered inputs skip everything after
+    the first count (in-flight order detection).  Part of the timed
+    paint path — nothing is cached across calls.
     """
     import torch
     n = len(pos_t)
@@ -80,13 +80,6 @@ def _prepare_particles(pos_t, mass_t, pm):
     box = hiplib.f64_arr(pm.BoxSize)
     stream = hiplib.cur_stream()
 
-    # NOTE: a two-level variant (coarse ~4k-bucket pass first) was
-    # measured 10x SLOWER at C4 — the coarse scatter's returned
-    # fetch-adds pile onto a few thousand hot counters and serialize
-    # (~2.1 s/step vs 113 ms single-level).  Single level stands until
-    # the ticket atomics are block-aggregated.
-    shift = 0
-
     def count(pos_in, nb, sh, detect):
         counts = torch.zeros(nb, dtype=torch.int32, device='cuda')
         flag = torch.zeros(1, dtype=torch.int32, device='cuda')
@@ -98,41 +91,68 @@ def _prepare_particles(pos_t, mass_t, pm):
             scrambled = int(flag.item()) != 0
         return counts, scrambled
 
-    def exclusive(counts):
-        offsets = torch.zeros(len(counts), dtype=torch.int64,
-                              device='cuda')
-        offsets[1:] = torch.cumsum(counts[:-1].long(), 0)
-        return offsets
-
-    def scatter(pos_in, m_in, offsets, sh, soa):
+    def scatter(pos_in, m_in, incl, sh, soa):
+        # incl = INCLUSIVE int32 bucket cumsum; the kernel's tickets
+        # count down to the exclusive base (no host-side shift needed)
         out = torch.empty(3 * n, dtype=torch.float64, device='cuda')
         out_m = None
         if m_in is not None:
             out_m = torch.empty(n, dtype=torch.float64, device='cuda')
         hiplib.check(lib.nbk_bucket_scatter_f64(
             hiplib.dptr(pos_in), hiplib.dptr(m_in), n, nmesh, box, sh,
-            int(soa), hiplib.dptr(offsets), hiplib.dptr(out),
+            int(soa), hiplib.dptr(incl), hiplib.dptr(out),
             hiplib.dptr(out_m), stream), 'nbk_bucket_scatter_f64')
         return out, out_m
 
     pos_in = pos_t.contiguous()
+    # the driver chunks paints at paint_chunk_size (< 2^31), so int32
+    # tickets are safe
+    assert n < 2 ** 31
 
-    counts, scrambled = count(pos_in, ncells >> shift, shift, detect=True)
-    if not scrambled:
-        # already cell-ordered: no scatter needed
-        return pos_t.t().contiguous(), mass_t, True
-
-    if shift > 0:
-        # coarse grouping pass (AoS out), then the fine per-cell pass on
-        # coarsely-local data
-        coarse, mass_c = scatter(pos_in, mass_t, exclusive(counts),
-                                 shift, soa=False)
-        pos_in = coarse
-        mass_t = mass_c
+    # Big meshes first run the deterministic chunked x-plane pre-sort
+    # (nbk_xsort_*: per-chunk count matrix + host scan + LDS-cursor
+    # placement, NO global atomics): the fine pass's scattered
+    # counter/ticket atomics then stay within one x-plane's window
+    # (n1*n2*4 B) and reuse cache lines ~16x instead of thrashing the
+    # multi-GB counts array (C4 count: 54.6 -> ~7 ms).  An earlier
+    # two-level variant using atomic tickets on ~4k coarse counters
+    # serialized catastrophically (2.1 s/step) — the count-matrix scheme
+    # has no shared cursors.
+    n0 = int(pm.Nmesh[0])
+    use_coarse = (n >= (1 << 24) and ncells > (1 << 23) and n0 <= 8192)
+    if use_coarse:
+        CH = 16384
+        nblocks = (n + CH - 1) // CH
+        mat = torch.empty(nblocks * n0, dtype=torch.int32, device='cuda')
+        flag = torch.zeros(1, dtype=torch.int32, device='cuda')
+        hiplib.check(lib.nbk_xsort_count_f64(
+            hiplib.dptr(pos_in), n, CH, nmesh, box, hiplib.dptr(mat),
+            hiplib.dptr(flag), stream), 'nbk_xsort_count_f64')
+        if int(flag.item()) == 0:
+            # already cell-ordered: no sorting needed
+            return pos_t.t().contiguous(), mass_t, True
+        tm = mat.view(nblocks, n0).t().contiguous().view(-1)
+        incl = torch.cumsum(tm, 0, dtype=torch.int32)
+        bases = (incl - tm).view(n0, nblocks).t().contiguous()
+        coarse = torch.empty(3 * n, dtype=torch.float64, device='cuda')
+        mass_c = None
+        if mass_t is not None:
+            mass_c = torch.empty(n, dtype=torch.float64, device='cuda')
+        hiplib.check(lib.nbk_xsort_scatter_f64(
+            hiplib.dptr(pos_in), hiplib.dptr(mass_t), n, CH, nmesh, box,
+            hiplib.dptr(bases), hiplib.dptr(coarse),
+            hiplib.dptr(mass_c), stream), 'nbk_xsort_scatter_f64')
+        pos_in, mass_t = coarse, mass_c
         counts, _ = count(pos_in, ncells, 0, detect=False)
+    else:
+        counts, scrambled = count(pos_in, ncells, 0, detect=True)
+        if not scrambled:
+            # already cell-ordered: no scatter needed
+            return pos_t.t().contiguous(), mass_t, True
 
-    out_soa, out_mass = scatter(pos_in, mass_t, exclusive(counts), 0,
-                                soa=True)
+    out_soa, out_mass = scatter(
+        pos_in, mass_t, torch.cumsum(counts, 0, dtype=torch.int32), 0,
+        soa=True)
     return out_soa, out_mass, True
 
 
