@@ -40,8 +40,14 @@ __device__ __forceinline__ cdouble cscale(cdouble a, double s) {
     return {a.re * s, a.im * s};
 }
 
-// periodic wrap of a (possibly negative) cell index
+// periodic wrap of a (possibly negative) cell index.  Meshes are powers
+// of two in this build (the FFT requires it), so the wrap is almost
+// always a single AND — the general int64 modulo is a ~30-op sequence
+// and the paint/count kernels issue up to 14 wraps per particle.  The
+// n is wave-uniform, so the check is scalar and predicted.
 __device__ __forceinline__ int64_t wrap_idx(int64_t i, int64_t n) {
+    if (__builtin_expect((n & (n - 1)) == 0, 1))
+        return i & (n - 1);
     i %= n;
     return i < 0 ? i + n : i;
 }
