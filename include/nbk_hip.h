@@ -121,24 +121,44 @@ int nbk_bucket_scatter_f64(const double* pos_aos, const double* mass,
                            double* pos_out, double* mass_out,
                            void* stream);
 
-/* deterministic chunked pre-sort by wrapped x-plane (the coarse level
- * of the two-level paint locality sort; no global atomics).  Pass A
- * (`count`) writes one n0-wide histogram row per chunk of `chunk`
- * particles to `mat` (ceil(n/chunk) x n0 int32, row-major) and performs
- * the cell-order detection into scrambled_flag (may be NULL); the host
- * exclusive-scans `mat` in bucket-major order into `bases`; pass C
- * (`scatter`) re-reads each chunk and places rows into `pos_out` (AoS)
- * from LDS cursors seeded with `bases`.  Running the cell sort above on
- * this output keeps its scattered atomics within one x-plane's counter
- * window (n1*n2*4 B).  Requires nmesh[0] <= 8192 and n < 2^31. */
+/* Two-level atomic-free cell sort (the paint locality pass for big
+ * meshes).  The GPU's global-atomic pipe runs at ~25 G ops/s regardless
+ * of locality (csrc/count_probe.hip), so the single-level
+ * count/scatter above is atomic-bound at 1e9 particles; this pipeline
+ * keeps every histogram/cursor in LDS:
+ *
+ *  coarse key = ix * (n1>>ys) + (iy>>ys), ys chosen so that BOTH
+ *  nbuckets = n0*(n1>>ys) and the fine window (1<<ys)*n2 are <= 40960
+ *  ints (160 KiB LDS):
+ *   - nbk_xsort_count_f64 (pass A): one nbuckets-wide histogram row per
+ *     chunk of `chunk` particles into `mat` (ceil(n/chunk) x nbuckets,
+ *     int32 row-major), plus the cell-order detection into
+ *     scrambled_flag (may be NULL);
+ *   - host: bucket-major exclusive scan of `mat` -> `bases`, and bucket
+ *     totals scan -> `bucket_bases` (nbuckets+1);
+ *   - nbk_xsort_scatter_f64 (pass C): re-reads each chunk, places AoS
+ *     rows from LDS cursors seeded with `bases` (deterministic — no
+ *     global atomics);
+ *   - nbk_bucket_fine_f64: one block per coarse bucket; counts the
+ *     bucket's cells in an LDS window, block-scans it, and emits the
+ *     exact cell-sorted SoA output (x[n] y[n] z[n]) — no global
+ *     atomics.
+ * Requires power-friendly dims (n1 divisible by 1<<ys, window
+ * divisible by 1024) and n < 2^31. */
 int nbk_xsort_count_f64(const double* pos_aos, int64_t n, int chunk,
                         const int64_t nmesh[3], const double box[3],
-                        int* mat, int* scrambled_flag, void* stream);
+                        int ys, int* mat, int* scrambled_flag,
+                        void* stream);
 int nbk_xsort_scatter_f64(const double* pos_aos, const double* mass,
                           int64_t n, int chunk, const int64_t nmesh[3],
-                          const double box[3], const int* bases,
+                          const double box[3], int ys, const int* bases,
                           double* pos_out, double* mass_out,
                           void* stream);
+int nbk_bucket_fine_f64(const double* pos_aos, const double* mass,
+                        int64_t n, const int64_t nmesh[3],
+                        const double box[3], int ys,
+                        const int* bucket_bases,
+                        double* soa_out, double* mass_out, void* stream);
 
 /* readout (gather dual of paint; window 0/1/2 = cic/tsc/pcs, 3 = nnb).
  * Serves FFTRecon's displacement solve (fftrecon.py:246-249) and the

@@ -94,14 +94,27 @@ __global__ void kbucket_scatter(const double* __restrict__ pos,
 // atomics hit 64-byte lines ~16x each (one x-plane's counters span
 // n1*n2*4 B) instead of thrashing the whole multi-GB array.
 
+// coarse key: ix * (n1 >> ys) + (iy >> ys) — ys chosen by the host so
+// that BOTH the coarse histogram (nbuckets ints) and the fine pass's
+// per-bucket cell window ((1 << ys) * n2 ints) fit in LDS
+__device__ __forceinline__ int64_t coarse_key(double x, double y,
+                                              double invH0, double invH1,
+                                              int64_t n0, int64_t n1,
+                                              int ys) {
+    const int64_t ix = wrap_idx((int64_t)floor(x * invH0), n0);
+    const int64_t iy = wrap_idx((int64_t)floor(y * invH1), n1);
+    return ix * (n1 >> ys) + (iy >> ys);
+}
+
 __global__ void kxsort_count(const double* __restrict__ pos, int64_t n,
                              int chunk, int64_t n0, int64_t n1, int64_t n2,
                              double invH0, double invH1, double invH2,
+                             int ys, int64_t nbuck,
                              int* __restrict__ mat,
                              int* __restrict__ scrambled_flag)
 {
-    extern __shared__ int hist[];   // n0 ints
-    for (int b = threadIdx.x; b < n0; b += blockDim.x) hist[b] = 0;
+    extern __shared__ int hist[];   // nbuck ints
+    for (int64_t b = threadIdx.x; b < nbuck; b += blockDim.x) hist[b] = 0;
     __syncthreads();
     const int64_t beg = (int64_t)blockIdx.x * chunk;
     const int64_t end = (beg + chunk < n) ? beg + chunk : n;
@@ -110,8 +123,7 @@ __global__ void kxsort_count(const double* __restrict__ pos, int64_t n,
     for (int64_t i = beg + threadIdx.x; i < end; i += blockDim.x) {
         const double x = pos[3 * i], y = pos[3 * i + 1],
                      z = pos[3 * i + 2];
-        const int64_t ix = wrap_idx((int64_t)floor(x * invH0), n0);
-        atomicAdd(&hist[ix], 1);
+        atomicAdd(&hist[coarse_key(x, y, invH0, invH1, n0, n1, ys)], 1);
         if (scrambled_flag) {
             const int64_t b = bucket_of(x, y, z, invH0, invH1, invH2,
                                         n0, n1, n2, 0);
@@ -121,32 +133,120 @@ __global__ void kxsort_count(const double* __restrict__ pos, int64_t n,
     }
     if (out_of_order) atomicOr(scrambled_flag, 1);
     __syncthreads();
-    for (int b = threadIdx.x; b < n0; b += blockDim.x)
-        mat[(int64_t)blockIdx.x * n0 + b] = hist[b];
+    for (int64_t b = threadIdx.x; b < nbuck; b += blockDim.x)
+        mat[(int64_t)blockIdx.x * nbuck + b] = hist[b];
 }
 
 __global__ void kxsort_scatter(const double* __restrict__ pos,
                                const double* __restrict__ mass, int64_t n,
-                               int chunk, int64_t n0, double invH0,
+                               int chunk, int64_t n0, int64_t n1,
+                               double invH0, double invH1,
+                               int ys, int64_t nbuck,
                                const int* __restrict__ bases,
                                double* __restrict__ out,
                                double* __restrict__ om)
 {
-    extern __shared__ int cur[];    // n0 running cursors
-    for (int b = threadIdx.x; b < n0; b += blockDim.x)
-        cur[b] = bases[(int64_t)blockIdx.x * n0 + b];
+    extern __shared__ int cur[];    // nbuck running cursors
+    for (int64_t b = threadIdx.x; b < nbuck; b += blockDim.x)
+        cur[b] = bases[(int64_t)blockIdx.x * nbuck + b];
     __syncthreads();
     const int64_t beg = (int64_t)blockIdx.x * chunk;
     const int64_t end = (beg + chunk < n) ? beg + chunk : n;
     for (int64_t i = beg + threadIdx.x; i < end; i += blockDim.x) {
         const double x = pos[3 * i], y = pos[3 * i + 1],
                      z = pos[3 * i + 2];
-        const int64_t ix = wrap_idx((int64_t)floor(x * invH0), n0);
-        const int64_t t = (int64_t)atomicAdd(&cur[ix], 1);
+        const int64_t k = coarse_key(x, y, invH0, invH1, n0, n1, ys);
+        const int64_t t = (int64_t)atomicAdd(&cur[k], 1);
         out[3 * t] = x;
         out[3 * t + 1] = y;
         out[3 * t + 2] = z;
         if (mass) om[t] = mass[i];
+    }
+}
+
+// fused fine pass: one block per coarse bucket.  The bucket's particles
+// are contiguous (coarse sort) and its cells span a window of
+// (1 << ys) * n2 entries, so count + exclusive scan + placement all run
+// in LDS — ZERO global atomics.  (The count probe measured the global
+// atomic pipe at ~25 G ops/s regardless of locality, which bounded the
+// old single-level sort; LDS histograms run at the read bandwidth.)
+// Output order = exact cell order, identical to the single-level sort.
+__global__ void kbucket_fine(const double* __restrict__ pos,
+                             const double* __restrict__ mass,
+                             int64_t n1, int64_t n2,
+                             double invH1, double invH2,
+                             int ys,
+                             const int* __restrict__ bbase, /* nbuck+1 */
+                             double* __restrict__ ox,
+                             double* __restrict__ oy,
+                             double* __restrict__ oz,
+                             double* __restrict__ om)
+{
+    extern __shared__ int lds[];            // win ints (counts->cursors)
+    __shared__ int ssum[16];
+    const int64_t win = ((int64_t)1 << ys) * n2;
+    const int64_t ymask = ((int64_t)1 << ys) - 1;
+    const int T = blockDim.x;
+    const int t = threadIdx.x;
+
+    for (int64_t w = t; w < win; w += T) lds[w] = 0;
+    __syncthreads();
+
+    const int64_t beg = bbase[blockIdx.x];
+    const int64_t end = bbase[blockIdx.x + 1];
+
+    for (int64_t i = beg + t; i < end; i += T) {
+        const int64_t iy = wrap_idx((int64_t)floor(pos[3 * i + 1] * invH1),
+                                    n1);
+        const int64_t iz = wrap_idx((int64_t)floor(pos[3 * i + 2] * invH2),
+                                    n2);
+        atomicAdd(&lds[(iy & ymask) * n2 + iz], 1);
+    }
+    __syncthreads();
+
+    // block-wide exclusive scan of lds[0..win), seeded with beg:
+    // stripes per thread, wave shfl scan of the stripe sums, serial
+    // scan of the wave totals
+    const int S = (int)(win / T);           // win % T == 0 (host guard)
+    int acc = 0;
+    for (int j = 0; j < S; j++) acc += lds[t * S + j];
+    const int lane = t & 63;
+    const int wave = t >> 6;
+    int v = acc;
+    #pragma unroll
+    for (int d = 1; d < 64; d <<= 1) {
+        const int u = __shfl_up(v, d, 64);
+        if (lane >= d) v += u;
+    }
+    if (lane == 63) ssum[wave] = v;
+    __syncthreads();
+    if (t == 0) {
+        int run = 0;
+        for (int w = 0; w < (T >> 6); w++) {
+            const int x = ssum[w];
+            ssum[w] = run;
+            run += x;
+        }
+    }
+    __syncthreads();
+    int run = (int)beg + ssum[wave] + (v - acc);
+    for (int j = 0; j < S; j++) {
+        const int c = lds[t * S + j];
+        lds[t * S + j] = run;
+        run += c;
+    }
+    __syncthreads();
+
+    for (int64_t i = beg + t; i < end; i += T) {
+        const double x = pos[3 * i], y = pos[3 * i + 1],
+                     z = pos[3 * i + 2];
+        const int64_t iy = wrap_idx((int64_t)floor(y * invH1), n1);
+        const int64_t iz = wrap_idx((int64_t)floor(z * invH2), n2);
+        const int slot = atomicAdd(&lds[(iy & ymask) * n2 + iz], 1);
+        ox[slot] = x;
+        oy[slot] = y;
+        oz[slot] = z;
+        if (mass) om[slot] = mass[i];
     }
 }
 
@@ -205,24 +305,36 @@ extern "C" int nbk_bucket_scatter_f64(const double* pos_aos,
     return NBK_OK;
 }
 
+namespace {
+// LDS ceiling for the sort kernels, in int entries (160 KiB on gfx950)
+const int64_t NBK_SORT_LDS_INTS = 40960;
+
+void raise_lds(const void* fn, size_t bytes) {
+    if (bytes > 64 * 1024)
+        (void)hipFuncSetAttribute(fn,
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)bytes);
+}
+}  // namespace
+
 extern "C" int nbk_xsort_count_f64(const double* pos_aos, int64_t n,
                                    int chunk, const int64_t nmesh[3],
-                                   const double box[3], int* mat,
+                                   const double box[3], int ys, int* mat,
                                    int* scrambled_flag, void* stream)
 {
     if (n == 0) return NBK_OK;
-    if (nmesh[0] > 8192) {
-        NBK_SET_ERR("nbk_xsort_count_f64: n0 > 8192 exceeds the LDS "
-                    "histogram");
+    const int64_t nbuck = nmesh[0] * (nmesh[1] >> ys);
+    if (nbuck > NBK_SORT_LDS_INTS || (nmesh[1] % ((int64_t)1 << ys))) {
+        NBK_SET_ERR("nbk_xsort_count_f64: bad ys=%d for mesh", ys);
         return NBK_ERR_ARG;
     }
     const int64_t nblocks = (n + chunk - 1) / chunk;
-    const size_t lds = (size_t)nmesh[0] * sizeof(int);
-    hipLaunchKernelGGL(kxsort_count, dim3((uint32_t)nblocks), dim3(256),
+    const size_t lds = (size_t)nbuck * sizeof(int);
+    raise_lds(reinterpret_cast<const void*>(&kxsort_count), lds);
+    hipLaunchKernelGGL(kxsort_count, dim3((uint32_t)nblocks), dim3(1024),
                        lds, (hipStream_t)stream, pos_aos, n, chunk,
                        nmesh[0], nmesh[1], nmesh[2],
                        nmesh[0] / box[0], nmesh[1] / box[1],
-                       nmesh[2] / box[2], mat, scrambled_flag);
+                       nmesh[2] / box[2], ys, nbuck, mat, scrambled_flag);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }
@@ -230,22 +342,53 @@ extern "C" int nbk_xsort_count_f64(const double* pos_aos, int64_t n,
 extern "C" int nbk_xsort_scatter_f64(const double* pos_aos,
                                      const double* mass, int64_t n,
                                      int chunk, const int64_t nmesh[3],
-                                     const double box[3], const int* bases,
+                                     const double box[3], int ys,
+                                     const int* bases,
                                      double* pos_out, double* mass_out,
                                      void* stream)
 {
     if (n == 0) return NBK_OK;
-    if (nmesh[0] > 8192) {
-        NBK_SET_ERR("nbk_xsort_scatter_f64: n0 > 8192 exceeds the LDS "
-                    "cursors");
+    const int64_t nbuck = nmesh[0] * (nmesh[1] >> ys);
+    if (nbuck > NBK_SORT_LDS_INTS || (nmesh[1] % ((int64_t)1 << ys))) {
+        NBK_SET_ERR("nbk_xsort_scatter_f64: bad ys=%d for mesh", ys);
         return NBK_ERR_ARG;
     }
     const int64_t nblocks = (n + chunk - 1) / chunk;
-    const size_t lds = (size_t)nmesh[0] * sizeof(int);
-    hipLaunchKernelGGL(kxsort_scatter, dim3((uint32_t)nblocks), dim3(256),
-                       lds, (hipStream_t)stream, pos_aos, mass, n, chunk,
-                       nmesh[0], nmesh[0] / box[0], bases,
-                       pos_out, mass_out);
+    const size_t lds = (size_t)nbuck * sizeof(int);
+    raise_lds(reinterpret_cast<const void*>(&kxsort_scatter), lds);
+    hipLaunchKernelGGL(kxsort_scatter, dim3((uint32_t)nblocks),
+                       dim3(1024), lds, (hipStream_t)stream, pos_aos,
+                       mass, n, chunk, nmesh[0], nmesh[1],
+                       nmesh[0] / box[0], nmesh[1] / box[1], ys, nbuck,
+                       bases, pos_out, mass_out);
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}
+
+extern "C" int nbk_bucket_fine_f64(const double* pos_aos,
+                                   const double* mass, int64_t n,
+                                   const int64_t nmesh[3],
+                                   const double box[3], int ys,
+                                   const int* bucket_bases,
+                                   double* soa_out, double* mass_out,
+                                   void* stream)
+{
+    if (n == 0) return NBK_OK;
+    const int64_t nbuck = nmesh[0] * (nmesh[1] >> ys);
+    const int64_t win = ((int64_t)1 << ys) * nmesh[2];
+    if (win > NBK_SORT_LDS_INTS || (win % 1024)
+        || (nmesh[1] % ((int64_t)1 << ys))) {
+        NBK_SET_ERR("nbk_bucket_fine_f64: bad ys=%d for mesh", ys);
+        return NBK_ERR_ARG;
+    }
+    const size_t lds = (size_t)win * sizeof(int);
+    raise_lds(reinterpret_cast<const void*>(&kbucket_fine), lds);
+    hipLaunchKernelGGL(kbucket_fine, dim3((uint32_t)nbuck), dim3(1024),
+                       lds, (hipStream_t)stream, pos_aos, mass,
+                       nmesh[1], nmesh[2],
+                       nmesh[1] / box[1], nmesh[2] / box[2], ys,
+                       bucket_bases, soa_out, soa_out + n,
+                       soa_out + 2 * n, mass_out);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

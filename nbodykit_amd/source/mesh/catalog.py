@@ -109,46 +109,71 @@ def _prepare_particles(pos_t, mass_t, pm):
     # tickets are safe
     assert n < 2 ** 31
 
-    # Big meshes first run the deterministic chunked x-plane pre-sort
-    # (nbk_xsort_*: per-chunk count matrix + host scan + LDS-cursor
-    # placement, NO global atomics): the fine pass's scattered
-    # counter/ticket atomics then stay within one x-plane's window
-    # (n1*n2*4 B) and reuse cache lines ~16x instead of thrashing the
-    # multi-GB counts array (C4 count: 54.6 -> ~7 ms).  An earlier
-    # two-level variant using atomic tickets on ~4k coarse counters
-    # serialized catastrophically (2.1 s/step) — the count-matrix scheme
-    # has no shared cursors.
-    n0 = int(pm.Nmesh[0])
-    use_coarse = (n >= (1 << 24) and ncells > (1 << 23) and n0 <= 8192)
-    if use_coarse:
-        CH = 16384
+    # Big meshes use the two-level ATOMIC-FREE pipeline (the global
+    # atomic pipe measures ~25 G ops/s regardless of locality —
+    # csrc/count_probe.hip — which bounded the single-level sort):
+    # chunked coarse sort by (ix, iy-group) via a per-chunk count
+    # matrix + host scan + LDS cursors, then a fused per-bucket fine
+    # kernel (LDS count + block scan + placement) emitting the exact
+    # cell order.  ys balances the two LDS budgets: coarse histogram
+    # n0*(n1>>ys) ints vs fine window (1<<ys)*n2 ints, both <= 40960.
+    n0, n1, n2 = (int(x) for x in pm.Nmesh)
+    LDSW = 40960
+    ys_fine = None
+    for ys in range(0, max(1, n1.bit_length())):
+        if (1 << ys) > n1 or n1 % (1 << ys):
+            break
+        win = (1 << ys) * n2
+        if win > LDSW:
+            break                       # grows with ys: hopeless beyond
+        if win < 1024 or win % 1024:
+            continue
+        if n0 * (n1 >> ys) <= LDSW:
+            ys_fine = ys
+            break
+    use_two = n >= (1 << 24) and ncells > (1 << 23) and ys_fine is not None
+    if use_two:
+        ys = ys_fine
+        nbuck = n0 * (n1 >> ys)
+        CH = 262144
         nblocks = (n + CH - 1) // CH
-        mat = torch.empty(nblocks * n0, dtype=torch.int32, device='cuda')
+        mat = torch.empty(nblocks * nbuck, dtype=torch.int32,
+                          device='cuda')
         flag = torch.zeros(1, dtype=torch.int32, device='cuda')
         hiplib.check(lib.nbk_xsort_count_f64(
-            hiplib.dptr(pos_in), n, CH, nmesh, box, hiplib.dptr(mat),
+            hiplib.dptr(pos_in), n, CH, nmesh, box, ys, hiplib.dptr(mat),
             hiplib.dptr(flag), stream), 'nbk_xsort_count_f64')
         if int(flag.item()) == 0:
             # already cell-ordered: no sorting needed
             return pos_t.t().contiguous(), mass_t, True
-        tm = mat.view(nblocks, n0).t().contiguous().view(-1)
+        tm = mat.view(nblocks, nbuck).t().contiguous().view(-1)
         incl = torch.cumsum(tm, 0, dtype=torch.int32)
-        bases = (incl - tm).view(n0, nblocks).t().contiguous()
+        bases = (incl - tm).view(nbuck, nblocks).t().contiguous()
+        bucket_bases = torch.zeros(nbuck + 1, dtype=torch.int32,
+                                   device='cuda')
+        bucket_bases[1:] = incl.view(nbuck, nblocks)[:, -1]
         coarse = torch.empty(3 * n, dtype=torch.float64, device='cuda')
         mass_c = None
         if mass_t is not None:
             mass_c = torch.empty(n, dtype=torch.float64, device='cuda')
         hiplib.check(lib.nbk_xsort_scatter_f64(
             hiplib.dptr(pos_in), hiplib.dptr(mass_t), n, CH, nmesh, box,
-            hiplib.dptr(bases), hiplib.dptr(coarse),
+            ys, hiplib.dptr(bases), hiplib.dptr(coarse),
             hiplib.dptr(mass_c), stream), 'nbk_xsort_scatter_f64')
-        pos_in, mass_t = coarse, mass_c
-        counts, _ = count(pos_in, ncells, 0, detect=False)
-    else:
-        counts, scrambled = count(pos_in, ncells, 0, detect=True)
-        if not scrambled:
-            # already cell-ordered: no scatter needed
-            return pos_t.t().contiguous(), mass_t, True
+        out = torch.empty(3 * n, dtype=torch.float64, device='cuda')
+        out_m = None
+        if mass_t is not None:
+            out_m = torch.empty(n, dtype=torch.float64, device='cuda')
+        hiplib.check(lib.nbk_bucket_fine_f64(
+            hiplib.dptr(coarse), hiplib.dptr(mass_c), n, nmesh, box, ys,
+            hiplib.dptr(bucket_bases), hiplib.dptr(out),
+            hiplib.dptr(out_m), stream), 'nbk_bucket_fine_f64')
+        return out, out_m, True
+
+    counts, scrambled = count(pos_in, ncells, 0, detect=True)
+    if not scrambled:
+        # already cell-ordered: no scatter needed
+        return pos_t.t().contiguous(), mass_t, True
 
     out_soa, out_mass = scatter(
         pos_in, mass_t, torch.cumsum(counts, 0, dtype=torch.int32), 0,
