@@ -98,6 +98,23 @@ int nbk_paint_sorted_f64(const double* pos, const double* mass, int64_t n,
                          double* mesh, int64_t x0, int64_t nx_local,
                          void* stream);
 
+/* ownership-gather paint for cell-sorted input: requires the row table
+ * produced by nbk_bucket_fine_f64 (rowtab[n0*n1+1]: start of each
+ * (ix, iy) row in the sorted SoA arrays, sentinel n at the end).  Each
+ * block owns an exclusive LDS mesh tile (RG y-rows x n2 of one x-plane)
+ * and gathers from the source rows whose stencils reach it — deposits
+ * are LDS f64 adds and the flush is plain stores, NO global atomics
+ * (the global atomic pipe is the ~25 G op/s bound on the scatter
+ * kernels above).  accumulate=0 overwrites the slab (fresh mesh, saves
+ * the read), accumulate=1 adds (hold semantics for chunked paints).
+ * Returns NBK_ERR_UNSUPPORTED when no LDS tile fits (fall back to
+ * nbk_paint_f64). */
+int nbk_paint_gather_f64(const double* pos, const double* mass, int64_t n,
+                         const int64_t nmesh[3], const double box[3],
+                         int window, double shift, const int* rowtab,
+                         double* mesh, int64_t x0, int64_t nx_local,
+                         int accumulate, void* stream);
+
 /* paint locality sort --------------------------------------------------
  * Two-pass counting sort of particles by coarse mesh cell
  * (bucket = wrapped ix * n1 + iy): count, then (after the caller turns
@@ -158,7 +175,8 @@ int nbk_bucket_fine_f64(const double* pos_aos, const double* mass,
                         int64_t n, const int64_t nmesh[3],
                         const double box[3], int ys,
                         const int* bucket_bases,
-                        double* soa_out, double* mass_out, void* stream);
+                        double* soa_out, double* mass_out,
+                        int* rowtab, void* stream);
 
 /* readout (gather dual of paint; window 0/1/2 = cic/tsc/pcs, 3 = nnb).
  * Serves FFTRecon's displacement solve (fftrecon.py:246-249) and the

@@ -25,6 +25,11 @@ _global_options = {
     'paint_chunk_size': 1024 * 1024 * 4,
     'dask_chunk_size': 100000,
     'global_cache_size': 1e8,
+    # paint locality-sort thresholds (tests shrink these to exercise the
+    # two-level atomic-free sort + gather paint on small inputs)
+    'sort_min_n': 1 << 21,
+    'sort_two_level_min_n': 1 << 24,
+    'sort_two_level_min_cells': 1 << 23,
 }
 
 

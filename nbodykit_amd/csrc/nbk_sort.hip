@@ -180,7 +180,11 @@ __global__ void kbucket_fine(const double* __restrict__ pos,
                              double* __restrict__ ox,
                              double* __restrict__ oy,
                              double* __restrict__ oz,
-                             double* __restrict__ om)
+                             double* __restrict__ om,
+                             int* __restrict__ rowtab /* n0*n1, or NULL:
+                                 start index of each (ix, iy) row in the
+                                 sorted output — consumed by the gather
+                                 paint kernel */)
 {
     extern __shared__ int lds[];            // win ints (counts->cursors)
     __shared__ int ssum[16];
@@ -236,6 +240,18 @@ __global__ void kbucket_fine(const double* __restrict__ pos,
         run += c;
     }
     __syncthreads();
+
+    if (rowtab) {
+        // row r of this bucket starts where its first cell's cursor
+        // points (post-scan, pre-scatter)
+        const int64_t nyb = n1 >> ys;
+        const int64_t ix = blockIdx.x / nyb;
+        const int64_t yb = blockIdx.x % nyb;
+        for (int r = t; r < (1 << ys); r += T)
+            rowtab[ix * n1 + yb * ((int64_t)1 << ys) + r] =
+                lds[(int64_t)r * n2];
+        __syncthreads();
+    }
 
     for (int64_t i = beg + t; i < end; i += T) {
         const double x = pos[3 * i], y = pos[3 * i + 1],
@@ -371,7 +387,7 @@ extern "C" int nbk_bucket_fine_f64(const double* pos_aos,
                                    const double box[3], int ys,
                                    const int* bucket_bases,
                                    double* soa_out, double* mass_out,
-                                   void* stream)
+                                   int* rowtab, void* stream)
 {
     if (n == 0) return NBK_OK;
     const int64_t nbuck = nmesh[0] * (nmesh[1] >> ys);
@@ -388,7 +404,7 @@ extern "C" int nbk_bucket_fine_f64(const double* pos_aos,
                        nmesh[1], nmesh[2],
                        nmesh[1] / box[1], nmesh[2] / box[2], ys,
                        bucket_bases, soa_out, soa_out + n,
-                       soa_out + 2 * n, mass_out);
+                       soa_out + 2 * n, mass_out, rowtab);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

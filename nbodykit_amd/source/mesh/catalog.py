@@ -48,7 +48,7 @@ def _is_trivial_true(sel):
         and a.strides[0] == 0 and bool(a[0])
 
 
-_SORT_MIN_N = 1 << 21        # below this the deposit is cheap either way
+
 
 
 def _prepare_particles(pos_t, mass_t, pm):
@@ -69,7 +69,7 @@ def _prepare_particles(pos_t, mass_t, pm):
     """
     import torch
     n = len(pos_t)
-    if n < _SORT_MIN_N:
+    if n < _global_options['sort_min_n']:
         return pos_t.t().contiguous(), mass_t, False
 
     lib = hiplib.require()
@@ -131,7 +131,9 @@ def _prepare_particles(pos_t, mass_t, pm):
         if n0 * (n1 >> ys) <= LDSW:
             ys_fine = ys
             break
-    use_two = n >= (1 << 24) and ncells > (1 << 23) and ys_fine is not None
+    use_two = (n >= _global_options['sort_two_level_min_n']
+               and ncells > _global_options['sort_two_level_min_cells']
+               and ys_fine is not None)
     if use_two:
         ys = ys_fine
         nbuck = n0 * (n1 >> ys)
@@ -362,22 +364,42 @@ class CatalogMesh(MeshSource):
 
             n = len(pos_t)
             if n > 0:
-                pos_soa, mass_t, sorted_ = _prepare_particles(pos_t,
-                                                              mass_t, pm)
-                paint_fn = lib.nbk_paint_sorted_f64 if sorted_                     else lib.nbk_paint_f64
+                pos_soa, mass_t, sorted_, rowtab = _prepare_particles(
+                    pos_t, mass_t, pm)
+                # fresh fields take plain stores on the first chunk
+                acc = 0 if (i == 0 and out is None) else 1
+                tgt = (real1 if interlaced else toret).value
                 with profiling.collect('paint', n * (1 + interlaced)):
-                    hiplib.check(paint_fn(
-                        hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n,
-                        nmesh, box, window_id, 0.0,
-                        hiplib.dptr((real1 if interlaced else toret).value),
-                        pm.x_start, pm.nx_local, stream), 'nbk_paint')
-                    if interlaced:
+                    if rowtab is not None:
+                        hiplib.check(lib.nbk_paint_gather_f64(
+                            hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n,
+                            nmesh, box, window_id, 0.0,
+                            hiplib.dptr(rowtab), hiplib.dptr(tgt),
+                            pm.x_start, pm.nx_local, acc, stream),
+                            'nbk_paint_gather_f64')
+                        if interlaced:
+                            hiplib.check(lib.nbk_paint_gather_f64(
+                                hiplib.dptr(pos_soa), hiplib.dptr(mass_t),
+                                n, nmesh, box, window_id, 0.5,
+                                hiplib.dptr(rowtab),
+                                hiplib.dptr(real2.value),
+                                pm.x_start, pm.nx_local, acc, stream),
+                                'nbk_paint_gather_f64')
+                    else:
+                        paint_fn = (lib.nbk_paint_sorted_f64 if sorted_
+                                    else lib.nbk_paint_f64)
                         hiplib.check(paint_fn(
                             hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n,
-                            nmesh, box, window_id, 0.5,
-                            hiplib.dptr(real2.value),
-                            pm.x_start, pm.nx_local, stream),
-                            'nbk_paint')
+                            nmesh, box, window_id, 0.0,
+                            hiplib.dptr(tgt),
+                            pm.x_start, pm.nx_local, stream), 'nbk_paint')
+                        if interlaced:
+                            hiplib.check(paint_fn(
+                                hiplib.dptr(pos_soa), hiplib.dptr(mass_t),
+                                n, nmesh, box, window_id, 0.5,
+                                hiplib.dptr(real2.value),
+                                pm.x_start, pm.nx_local, stream),
+                                'nbk_paint')
             i = i + chunksize
 
         if interlaced:
@@ -487,7 +509,15 @@ def paint_raw(pos_t, pm, resampler='cic'):
     if pm.comm.size > 1:
         dmin, dmax = _GHOST_RANGE[(resampler, False)]
         pos_t, _ = route_particles(pos_t, None, pm, dmin, dmax)
-    pos_soa, _, sorted_ = _prepare_particles(pos_t, None, pm)
+    pos_soa, _, sorted_, rowtab = _prepare_particles(pos_t, None, pm)
+    if rowtab is not None:
+        hiplib.check(lib.nbk_paint_gather_f64(
+            hiplib.dptr(pos_soa), None, len(pos_t),
+            hiplib.i64_arr(pm.Nmesh), hiplib.f64_arr(pm.BoxSize),
+            hiplib.WINDOW_IDS[resampler], 0.0, hiplib.dptr(rowtab),
+            hiplib.dptr(field.value), pm.x_start, pm.nx_local, 0,
+            hiplib.cur_stream()), 'nbk_paint_gather_f64')
+        return field
     paint_fn = lib.nbk_paint_sorted_f64 if sorted_ else lib.nbk_paint_f64
     hiplib.check(paint_fn(
         hiplib.dptr(pos_soa), None, len(pos_t),

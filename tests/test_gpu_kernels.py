@@ -395,3 +395,67 @@ def test_paint_sorted_slab_bounds(lib):
             1, 0.0, hiplib.dptr(slab), x0, 8, None), 'paint_sorted')
         got[x0:x0 + 8] = host(slab)
     assert_allclose(got, full, rtol=1e-12, atol=1e-12)
+
+
+# ---------------------------------------------------------------------------
+# two-level atomic-free sort + ownership-gather paint (nbk_xsort_* /
+# nbk_bucket_fine_f64 / nbk_paint_gather_f64) — production C4 path,
+# exercised here by shrinking the size thresholds
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize('resampler,interlaced', [
+    ('cic', False), ('cic', True), ('tsc', False), ('tsc', True),
+    ('pcs', False), ('pcs', True)])
+def test_two_level_gather_paint_matches_direct(resampler, interlaced):
+    from nbodykit_amd import set_options
+    from nbodykit_amd.lab import ArrayCatalog
+    rng = numpy.random.RandomState(17)
+    n = 200000
+    # clumpy + scrambled: uniform plus a dense blob, shuffled
+    pos = numpy.concatenate([
+        rng.uniform(0, 64., size=(n, 3)),
+        rng.normal(32., 1.5, size=(n // 4, 3)) % 64.])
+    rng.shuffle(pos)
+    cat = ArrayCatalog({'Position': pos})
+    kw = dict(Nmesh=64, BoxSize=64., dtype='f8', resampler=resampler,
+              interlaced=interlaced, compensated=False)
+
+    r_direct = cat.to_mesh(**kw).compute(mode='real')
+    with set_options(sort_min_n=1024, sort_two_level_min_n=1024,
+                     sort_two_level_min_cells=1):
+        r_two = cat.to_mesh(**kw).compute(mode='real')
+    assert_allclose(r_two.value.cpu().numpy(),
+                    r_direct.value.cpu().numpy(), rtol=1e-12, atol=1e-12)
+
+
+def test_two_level_gather_paint_weighted():
+    from nbodykit_amd import set_options
+    from nbodykit_amd.lab import ArrayCatalog
+    rng = numpy.random.RandomState(18)
+    n = 150000
+    cat = ArrayCatalog({'Position': rng.uniform(0, 32., size=(n, 3)),
+                        'Weight': rng.exponential(size=n)})
+    kw = dict(Nmesh=32, BoxSize=32., dtype='f8', compensated=False)
+    r_direct = cat.to_mesh(**kw).compute(mode='real')
+    with set_options(sort_min_n=1024, sort_two_level_min_n=1024,
+                     sort_two_level_min_cells=1):
+        r_two = cat.to_mesh(**kw).compute(mode='real')
+    assert_allclose(r_two.value.cpu().numpy(),
+                    r_direct.value.cpu().numpy(), rtol=1e-12, atol=1e-12)
+
+
+def test_two_level_gather_paint_chunked():
+    # chunking accumulates across gather paints (accumulate=1 path)
+    from nbodykit_amd import set_options
+    from nbodykit_amd.lab import ArrayCatalog
+    rng = numpy.random.RandomState(19)
+    n = 120000
+    cat = ArrayCatalog({'Position': rng.uniform(0, 32., size=(n, 3))})
+    kw = dict(Nmesh=32, BoxSize=32., dtype='f8', compensated=False)
+    r_direct = cat.to_mesh(**kw).compute(mode='real')
+    with set_options(sort_min_n=1024, sort_two_level_min_n=1024,
+                     sort_two_level_min_cells=1,
+                     paint_chunk_size=33333):
+        r_two = cat.to_mesh(**kw).compute(mode='real')
+    assert_allclose(r_two.value.cpu().numpy(),
+                    r_direct.value.cpu().numpy(), rtol=1e-12, atol=1e-12)
