@@ -52,7 +52,8 @@ def _is_trivial_true(sel):
 
 
 def _prepare_particles(pos_t, mass_t, pm):
-    """Return (pos_soa, mass, cell_sorted) for the deposit kernel,
+    """Return (pos_soa, mass, cell_sorted, rowtab) for the deposit
+    kernel,
     bucket-sorting the chunk by mesh cell when it arrives scrambled.
 
     The deposit kernel's wave-merge and its L2 locality both depend on
@@ -70,7 +71,7 @@ def _prepare_particles(pos_t, mass_t, pm):
     import torch
     n = len(pos_t)
     if n < _global_options['sort_min_n']:
-        return pos_t.t().contiguous(), mass_t, False
+        return pos_t.t().contiguous(), mass_t, False, None
 
     lib = hiplib.require()
     n0, n1, n2 = (int(x) for x in pm.Nmesh)
@@ -147,7 +148,7 @@ def _prepare_particles(pos_t, mass_t, pm):
             hiplib.dptr(flag), stream), 'nbk_xsort_count_f64')
         if int(flag.item()) == 0:
             # already cell-ordered: no sorting needed
-            return pos_t.t().contiguous(), mass_t, True
+            return pos_t.t().contiguous(), mass_t, True, None
         tm = mat.view(nblocks, nbuck).t().contiguous().view(-1)
         incl = torch.cumsum(tm, 0, dtype=torch.int32)
         bases = (incl - tm).view(nbuck, nblocks).t().contiguous()
@@ -166,21 +167,29 @@ def _prepare_particles(pos_t, mass_t, pm):
         out_m = None
         if mass_t is not None:
             out_m = torch.empty(n, dtype=torch.float64, device='cuda')
+        # rowtab feeds the ownership-gather paint (tile = RG rows x n2
+        # f64 in LDS; needs n2 <= 20480)
+        rowtab = None
+        if n2 <= 20480:
+            rowtab = torch.empty(n0 * n1 + 1, dtype=torch.int32,
+                                 device='cuda')
+            rowtab[-1] = n
         hiplib.check(lib.nbk_bucket_fine_f64(
             hiplib.dptr(coarse), hiplib.dptr(mass_c), n, nmesh, box, ys,
             hiplib.dptr(bucket_bases), hiplib.dptr(out),
-            hiplib.dptr(out_m), stream), 'nbk_bucket_fine_f64')
-        return out, out_m, True
+            hiplib.dptr(out_m), hiplib.dptr(rowtab), stream),
+            'nbk_bucket_fine_f64')
+        return out, out_m, True, rowtab
 
     counts, scrambled = count(pos_in, ncells, 0, detect=True)
     if not scrambled:
         # already cell-ordered: no scatter needed
-        return pos_t.t().contiguous(), mass_t, True
+        return pos_t.t().contiguous(), mass_t, True, None
 
     out_soa, out_mass = scatter(
         pos_in, mass_t, torch.cumsum(counts, 0, dtype=torch.int32), 0,
         soa=True)
-    return out_soa, out_mass, True
+    return out_soa, out_mass, True, None
 
 
 def _is_trivial_unit(col):
