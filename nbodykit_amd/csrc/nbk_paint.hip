@@ -62,7 +62,13 @@ __device__ __forceinline__ void paint_weights(double u0, double u1,
                                               double (&w2)[SUP],
                                               int64_t& b0, int64_t& b1,
                                               int64_t& b2) {
-    if (WINDOW == NBK_WINDOW_TSC) {
+    if (WINDOW == NBK_WINDOW_CIC) {
+        const double f0 = floor(u0), f1 = floor(u1), f2 = floor(u2);
+        b0 = (int64_t)f0; b1 = (int64_t)f1; b2 = (int64_t)f2;
+        w0[SUP - 1] = u0 - f0; w0[0] = 1.0 - (u0 - f0);
+        w1[SUP - 1] = u1 - f1; w1[0] = 1.0 - (u1 - f1);
+        w2[SUP - 1] = u2 - f2; w2[0] = 1.0 - (u2 - f2);
+    } else if (WINDOW == NBK_WINDOW_TSC) {
         const double f0 = floor(u0 + 0.5), f1 = floor(u1 + 0.5),
                      f2 = floor(u2 + 0.5);
         b0 = (int64_t)f0 - 1; b1 = (int64_t)f1 - 1; b2 = (int64_t)f2 - 1;
