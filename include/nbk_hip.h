@@ -176,7 +176,7 @@ int nbk_bucket_fine_f64(const double* pos_aos, const double* mass,
                         const double box[3], int ys,
                         const int* bucket_bases,
                         double* soa_out, double* mass_out,
-                        int* rowtab, void* stream);
+                        int* rowtab, int rows_only, void* stream);
 
 /* readout (gather dual of paint; window 0/1/2 = cic/tsc/pcs, 3 = nnb).
  * Serves FFTRecon's displacement solve (fftrecon.py:246-249) and the

@@ -266,6 +266,74 @@ __global__ void kbucket_fine(const double* __restrict__ pos,
     }
 }
 
+// row-only fine pass: when the ownership-gather paint consumes the
+// output, full cell order is unnecessary — the gather reads particles
+// through ROW intervals (rowtab) and z order within a row is
+// irrelevant (unsorted z even reduces its LDS same-address conflicts).
+// The window shrinks from (1<<ys)*n2 cells to (1<<ys) rows, so this
+// kernel runs at full occupancy with a trivial serial scan.
+__global__ void kbucket_fine_rows(const double* __restrict__ pos,
+                                  const double* __restrict__ mass,
+                                  int64_t n1,
+                                  double invH1,
+                                  int ys,
+                                  const int* __restrict__ bbase,
+                                  double* __restrict__ ox,
+                                  double* __restrict__ oy,
+                                  double* __restrict__ oz,
+                                  double* __restrict__ om,
+                                  int* __restrict__ rowtab)
+{
+    extern __shared__ int cur[];            // (1 << ys) cursors
+    const int nr = 1 << ys;
+    const int64_t ymask = nr - 1;
+    const int T = blockDim.x;
+    const int t = threadIdx.x;
+
+    for (int r = t; r < nr; r += T) cur[r] = 0;
+    __syncthreads();
+
+    const int64_t beg = bbase[blockIdx.x];
+    const int64_t end = bbase[blockIdx.x + 1];
+
+    for (int64_t i = beg + t; i < end; i += T) {
+        const int64_t iy = wrap_idx((int64_t)floor(pos[3 * i + 1] * invH1),
+                                    n1);
+        atomicAdd(&cur[iy & ymask], 1);
+    }
+    __syncthreads();
+
+    if (t == 0) {
+        int run = (int)beg;
+        for (int r = 0; r < nr; r++) {
+            const int c = cur[r];
+            cur[r] = run;
+            run += c;
+        }
+    }
+    __syncthreads();
+
+    if (rowtab) {
+        const int64_t nyb = n1 >> ys;
+        const int64_t ix = blockIdx.x / nyb;
+        const int64_t yb = blockIdx.x % nyb;
+        for (int r = t; r < nr; r += T)
+            rowtab[ix * n1 + yb * (int64_t)nr + r] = cur[r];
+        __syncthreads();
+    }
+
+    for (int64_t i = beg + t; i < end; i += T) {
+        const double x = pos[3 * i], y = pos[3 * i + 1],
+                     z = pos[3 * i + 2];
+        const int64_t iy = wrap_idx((int64_t)floor(y * invH1), n1);
+        const int slot = atomicAdd(&cur[iy & ymask], 1);
+        ox[slot] = x;
+        oy[slot] = y;
+        oz[slot] = z;
+        if (mass) om[slot] = mass[i];
+    }
+}
+
 int sgrid(int64_t n) {
     int64_t g = (n + 255) / 256;
     if (g > 1048576) g = 1048576;
@@ -387,10 +455,25 @@ extern "C" int nbk_bucket_fine_f64(const double* pos_aos,
                                    const double box[3], int ys,
                                    const int* bucket_bases,
                                    double* soa_out, double* mass_out,
-                                   int* rowtab, void* stream)
+                                   int* rowtab, int rows_only,
+                                   void* stream)
 {
     if (n == 0) return NBK_OK;
     const int64_t nbuck = nmesh[0] * (nmesh[1] >> ys);
+    if (rows_only) {
+        if (nmesh[1] % ((int64_t)1 << ys)) {
+            NBK_SET_ERR("nbk_bucket_fine_f64: bad ys=%d", ys);
+            return NBK_ERR_ARG;
+        }
+        const size_t lds = ((size_t)1 << ys) * sizeof(int);
+        hipLaunchKernelGGL(kbucket_fine_rows, dim3((uint32_t)nbuck),
+                           dim3(1024), lds, (hipStream_t)stream, pos_aos,
+                           mass, nmesh[1], nmesh[1] / box[1], ys,
+                           bucket_bases, soa_out, soa_out + n,
+                           soa_out + 2 * n, mass_out, rowtab);
+        NBK_CHECK_HIP(hipGetLastError());
+        return NBK_OK;
+    }
     const int64_t win = ((int64_t)1 << ys) * nmesh[2];
     if (win > NBK_SORT_LDS_INTS || (win % 1024)
         || (nmesh[1] % ((int64_t)1 << ys))) {

@@ -69,7 +69,8 @@ def _declare(lib):
     lib.nbk_bucket_fine_f64.restype = ctypes.c_int
     lib.nbk_bucket_fine_f64.argtypes = [c_void, c_void, c_i64, c_i64_p,
                                         c_f64_p, ctypes.c_int, c_void,
-                                        c_void, c_void, c_void, c_void]
+                                        c_void, c_void, c_void,
+                                        ctypes.c_int, c_void]
     lib.nbk_paint_gather_f64.restype = ctypes.c_int
     lib.nbk_paint_gather_f64.argtypes = [c_void, c_void, c_i64, c_i64_p,
                                          c_f64_p, ctypes.c_int, c_f64,

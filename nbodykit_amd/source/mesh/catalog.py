@@ -174,10 +174,14 @@ def _prepare_particles(pos_t, mass_t, pm):
             rowtab = torch.empty(n0 * n1 + 1, dtype=torch.int32,
                                  device='cuda')
             rowtab[-1] = n
+        # with the gather paint downstream only ROW grouping is needed
+        # (z order within a row is irrelevant there); the full cell sort
+        # serves the nbk_paint_sorted fallback
+        rows_only = 1 if rowtab is not None else 0
         hiplib.check(lib.nbk_bucket_fine_f64(
             hiplib.dptr(coarse), hiplib.dptr(mass_c), n, nmesh, box, ys,
             hiplib.dptr(bucket_bases), hiplib.dptr(out),
-            hiplib.dptr(out_m), hiplib.dptr(rowtab), stream),
+            hiplib.dptr(out_m), hiplib.dptr(rowtab), rows_only, stream),
             'nbk_bucket_fine_f64')
         return out, out_m, True, rowtab
 
