@@ -280,7 +280,15 @@ extern "C" int nbk_fft_c_strided(double* cplx, int64_t nfft, int64_t stride,
         NBK_SET_ERR("nbk_fft_c_strided: grid too large");
         return NBK_ERR_ARG;
     }
-    const int block = 256;
+    // block size: more waves per CU = more outstanding strided lines
+    // (the pass is latency-bound, ~1.3 TB/s at 256 threads); overridable
+    // for experiments via NBK_FFT_BLOCK
+    static int block = 0;
+    if (!block) {
+        const char* e = getenv("NBK_FFT_BLOCK");
+        block = e ? atoi(e) : 1024;
+        if (block < 64 || block > 1024) block = 1024;
+    }
     const size_t shmem = (size_t)nfft * TI * sizeof(cdouble);
     if (shmem > 65536) {
         static size_t raised_fwd = 0, raised_inv = 0;
