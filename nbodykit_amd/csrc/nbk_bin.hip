@@ -364,7 +364,8 @@ static int launch_bin(const double* d1, const double* d2, BinArgs& A,
     const size_t lds_bytes = ((size_t)NB * nfields + nx_edges + nmu_edges)
                              * sizeof(double);
     const int64_t nlines = A.d0 * A.d1;
-    int64_t g = nlines;
+    const int64_t ngroups = (nlines + 7) / 8;     // NL = 8 in the kernel
+    int64_t g = ngroups;
     if (g > 8192) g = 8192;    // bounded: LDS flush cost scales with grid
     if (g < 1) g = 1;
 
@@ -382,10 +383,10 @@ static int launch_bin(const double* d1, const double* d2, BinArgs& A,
                 (int)lds_bytes);
             raised = lds_bytes;
         }
-        hipLaunchKernelGGL(kbin<true>, dim3((uint32_t)g), dim3(256),
+        hipLaunchKernelGGL(kbin<true>, dim3((uint32_t)g), dim3(1024),
                            lds_bytes, s, d1, d2, A, kedges, muedges, xsum);
     } else {
-        hipLaunchKernelGGL(kbin<false>, dim3((uint32_t)g), dim3(256), 0, s,
+        hipLaunchKernelGGL(kbin<false>, dim3((uint32_t)g), dim3(1024), 0, s,
                            d1, d2, A, kedges, muedges, xsum);
     }
     NBK_CHECK_HIP(hipGetLastError());

@@ -229,7 +229,7 @@ extern "C" int nbk_fft_r2c_z(const double* real, double* cplx,
     if (nlines == 0) return NBK_OK;
     double* table = get_twiddles(nz);
     if (!table) { NBK_SET_ERR("twiddle alloc failed"); return NBK_ERR_HIP; }
-    const int block = (int)std::min<int64_t>(256, nz / 4);
+    const int block = (int)std::min<int64_t>(512, nz / 4);
     const size_t shmem = (size_t)(nz / 2) * sizeof(cdouble);
     hipLaunchKernelGGL(kfft_r2c_z, dim3((uint32_t)nlines), dim3(block), shmem,
                        (hipStream_t)stream, real, cplx, nz, scale,
@@ -246,7 +246,7 @@ extern "C" int nbk_fft_c2r_z(const double* cplx, double* real,
     if (nlines == 0) return NBK_OK;
     double* table = get_twiddles(nz);
     if (!table) { NBK_SET_ERR("twiddle alloc failed"); return NBK_ERR_HIP; }
-    const int block = (int)std::min<int64_t>(256, nz / 4);
+    const int block = (int)std::min<int64_t>(512, nz / 4);
     const size_t shmem = (size_t)(nz + 1) * sizeof(cdouble);
     hipLaunchKernelGGL(kfft_c2r_z, dim3((uint32_t)nlines), dim3(block), shmem,
                        (hipStream_t)stream, cplx, real, nz,
