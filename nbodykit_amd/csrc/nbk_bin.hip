@@ -56,49 +56,17 @@ __device__ __forceinline__ int dig(const double* __restrict__ edges,
 
 // accumulate one element into hist arrays laid out as:
 //   [xsum (NB)] [musum (NB)] [Nsum (NB)] [ysum: nell * 2 * NB]
-// where NB = (Nx+2)*(Nmu+2).  Wave-segmented: contiguous z runs share a
-// bin (|k| moves by ~mu*k0 per element), so lanes in the same run
-// combine via a head-bounded shfl prefix and only the tail lane issues
-// the LDS atomic — the naive version wave-serializes on the
-// same-address conflicts.  bin < 0 marks an inactive lane (loop tail or
-// an overflow-row mode); all lanes of the wave must call collectively.
+// where NB = (Nx+2)*(Nmu+2)
+template <bool LDS>
 __device__ __forceinline__ void accum(double* __restrict__ h, int NB,
                                       int bin, double kmag, double mu,
-                                      double w, const cdouble* yv, int nell,
-                                      int lane) {
-    const int b_up = __shfl_up(bin, 1, 64);
-    const bool head = (lane == 0) || (b_up != bin);
-    const unsigned long long heads = __ballot(head);
-    const unsigned long long below = heads & (~0ULL >> (63 - lane));
-    const int myhead = 63 - __clzll(below);
-    const bool next_head = (lane < 63) && ((heads >> (lane + 1)) & 1ULL);
-    const bool tail = (lane == 63) || next_head;
-
-    auto seg_sum = [&](double v) {
-        #pragma unroll
-        for (int d = 1; d < 64; d <<= 1) {
-            const double u = __shfl_up(v, d, 64);
-            if (lane - d >= myhead) v += u;
-        }
-        return v;
-    };
-
-    const double s0 = seg_sum(kmag * w);
-    const double s1 = seg_sum(mu * w);
-    const double s2 = seg_sum(w);
-    const bool out = tail && bin >= 0;
-    if (out) {
-        atomicAdd(&h[bin], s0);
-        atomicAdd(&h[NB + bin], s1);
-        atomicAdd(&h[2 * NB + bin], s2);
-    }
+                                      double w, const cdouble* yv, int nell) {
+    atomicAdd(&h[bin], kmag * w);
+    atomicAdd(&h[NB + bin], mu * w);
+    atomicAdd(&h[2 * NB + bin], w);
     for (int e = 0; e < nell; e++) {
-        const double sr = seg_sum(yv[e].re);
-        const double si = seg_sum(yv[e].im);
-        if (out) {
-            atomicAdd(&h[(3 + 2 * e) * NB + bin], sr);
-            atomicAdd(&h[(3 + 2 * e + 1) * NB + bin], si);
-        }
+        atomicAdd(&h[(3 + 2 * e) * NB + bin], yv[e].re);
+        atomicAdd(&h[(3 + 2 * e + 1) * NB + bin], yv[e].im);
     }
 }
 
@@ -166,23 +134,18 @@ __global__ void kbin(const double* __restrict__ data,
         __syncthreads();
         const int64_t nelem =
             (nlines - line0 < NL ? nlines - line0 : (int64_t)NL) * A.d2;
-        const int lane = threadIdx.x & 63;
         int sub = 0;
         int64_t l2 = threadIdx.x;
         while (l2 >= A.d2) { l2 -= A.d2; sub++; }
 
-    // wave-collective: the loop bound is per-WAVE (i - lane is the wave
-    // base) so the segmented accumulate's shuffles stay uniform; lanes
-    // past the end carry bin = -1
-    for (int64_t i = threadIdx.x; i - lane < nelem;
+    for (int64_t i = threadIdx.x; i < nelem;
          i += blockDim.x,
          l2 += blockDim.x,
          ({ while (l2 >= A.d2) { l2 -= A.d2; sub++; } })) {
-        const bool valid = i < nelem && sub < NL;
-        const int64_t idx = valid ? (line0 + sub) * A.d2 + l2 : 0;
+        const int64_t idx = (line0 + sub) * A.d2 + l2;
         double f3[3];
-        f3[A.a0] = valid ? lf0[sub] : 0.0;
-        f3[A.a1] = valid ? lf1[sub] : 0.0;
+        f3[A.a0] = lf0[sub];
+        f3[A.a1] = lf1[sub];
         f3[A.a2] = freq_axis(A.a2, l2 + A.o2, n3[A.a2], A.real_field);
         const double fx = f3[0];
         const double fy = f3[1];
@@ -192,11 +155,10 @@ __global__ void kbin(const double* __restrict__ data,
         const double k2 = kx * kx + ky * ky + kz * kz;
         // modes beyond the last edge land in the (Nx+1) overflow row,
         // which project_to_basis discards (fftpower.py:666-668 keeps
-        // bins 1..Nx) — mark them inactive before touching the field
-        // data.  At default edges (kmax ~ the min-axis Nyquist) this is
-        // every corner mode outside the inscribed sphere, ~48% of the
-        // volume.
-        const bool active = valid && k2 < k2edges[A.nx_edges - 1];
+        // bins 1..Nx) — skip them before touching the field data.  At
+        // default edges (kmax ~ the min-axis Nyquist) this is every
+        // corner mode outside the inscribed sphere, ~48% of the volume.
+        if (k2 >= k2edges[A.nx_edges - 1]) continue;
         const double kmag = sqrt(k2);
         double mu = kx * A.losx + ky * A.losy + kz * A.losz;
         mu = (kmag == 0.0) ? 0.0 : mu / kmag;
@@ -204,14 +166,12 @@ __global__ void kbin(const double* __restrict__ data,
         const bool nonsingular = !A.real_field && fz > 0.0;  // doubled
         const double w = nonsingular ? 2.0 : 1.0;
 
-        const int bx = active ? dig(k2edges, A.nx_edges, k2) : 0;
-        const int bmu = active ? dig(muedges, A.nmu_edges, mu) : 0;
-        const int bin = active ? bx * (A.nmu_edges + 1) + bmu : -1;
+        const int bx = dig(k2edges, A.nx_edges, k2);
+        const int bmu = dig(muedges, A.nmu_edges, mu);
+        const int bin = bx * (A.nmu_edges + 1) + bmu;
 
-        cdouble v = {0.0, 0.0};
-        if (!active) {
-            // inactive lanes still join the collective accumulate
-        } else if (A.fuse) {
+        cdouble v;
+        if (A.fuse) {
             // comp1(c1) * conj(comp2(c2)) * V, zero mode cleared but
             // still binned (fftpower.py:114-128); compensation factors
             // bit-identical to the standalone nbk_compensate_f64 pass
@@ -267,8 +227,7 @@ __global__ void kbin(const double* __restrict__ data,
             }
         }
 
-        accum(h, NB, bin, active ? kmag : 0.0, active ? mu : 0.0,
-              active ? w : 0.0, yv, A.nell, lane);
+        accum<LDS>(h, NB, bin, kmag, mu, w, yv, A.nell);
     }
     }
 
