@@ -33,7 +33,34 @@ import numpy
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 HBM_PEAK = 8.0e12          # B/s, MI355X spec (MI355X_MICROARCH.md)
+# scatter-kernel algorithmic bytes/particle (pos read + support^3
+# deposit read-modify-writes) — applies when the atomic scatter path
+# paints (small chunks/meshes)
 BYTES_PER_PARTICLE = {'cic': 152.0, 'tsc': 456.0, 'pcs': 1048.0}
+
+
+def gather_bpp(resampler, nmesh, n_particles):
+    """Algorithmic bytes/particle of the ownership-gather paint
+    (nbk_paint_gather_f64, the big-mesh path): each particle's 24 B
+    position row is read once per source plane of its owner tiles (2
+    planes CIC, 4 TSC/PCS at shift 0) x a (RG+span)/RG row-halo factor,
+    plus the single plain mesh write amortized per particle."""
+    planes = {'cic': 2, 'tsc': 4, 'pcs': 4}[resampler]
+    xspan = {'cic': 1, 'tsc': 3, 'pcs': 3}[resampler]
+    n1 = n2 = nmesh
+    RG = 1
+    while RG * 2 * n2 <= 20480 and n1 % (RG * 2) == 0 and RG * 2 <= n1:
+        RG *= 2
+    reads = 24.0 * planes * (RG + xspan) / float(RG)
+    mesh_bytes = 8.0 * nmesh ** 3 / float(n_particles)
+    return reads + mesh_bytes
+
+
+def paint_is_gather(nmesh, n_local):
+    """Mirrors the driver's two-level-sort thresholds
+    (source/mesh/catalog.py _prepare_particles)."""
+    return nmesh ** 3 > (1 << 23) and n_local >= (1 << 24) \
+        and nmesh * nmesh <= 20480 * 32
 
 WORKLOADS = {
     # BASELINE.json configs (C1 is the CPU-oracle plumbing config)
@@ -326,7 +353,11 @@ def main():
 
     # paint-kernel roofline (rank-0 local figures)
     paint = prof.get('paint', {'ms': 0.0, 'calls': 0, 'units': 0})
-    bpp = BYTES_PER_PARTICLE[cfg['resampler']]
+    gather = paint_is_gather(nmesh, len(pos))
+    if gather:
+        bpp = gather_bpp(cfg['resampler'], nmesh, max(1, len(pos)))
+    else:
+        bpp = BYTES_PER_PARTICLE[cfg['resampler']]
     algo_bytes = paint['units'] * bpp
     achieved = algo_bytes / (paint['ms'] * 1e-3) if paint['ms'] > 0 else 0.
     traffic = None
@@ -370,7 +401,8 @@ def main():
                 'unit': 'GB/s',
                 'frac': achieved / HBM_PEAK,
                 'traffic': traffic,
-                'kernel': 'nbk_paint_f64[%s]' % cfg['resampler'],
+                'kernel': ('nbk_paint_gather_f64[%s]' if gather else
+                           'nbk_paint_f64[%s]') % cfg['resampler'],
                 'paint_ms_per_launch': (paint['ms'] / paint['calls']
                                         if paint['calls'] else None),
                 'algorithmic_B_per_particle': bpp,
