@@ -60,7 +60,7 @@ def paint_is_gather(nmesh, n_local):
     """Mirrors the driver's two-level-sort thresholds
     (source/mesh/catalog.py _prepare_particles)."""
     return nmesh ** 3 > (1 << 23) and n_local >= (1 << 24) \
-        and nmesh * nmesh <= 20480 * 32
+        and nmesh <= 20480
 
 WORKLOADS = {
     # BASELINE.json configs (C1 is the CPU-oracle plumbing config)
