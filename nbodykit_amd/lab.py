@@ -13,7 +13,8 @@ from nbodykit_amd.algorithms import (FFTPower, FFTCorr, FFTRecon,
                                      ProjectedFFTPower, project_to_basis)
 from nbodykit_amd.source.catalog import (UniformCatalog, RandomCatalog,
                                          LogNormalCatalog, ArrayCatalog)
-from nbodykit_amd.source.mesh import CatalogMesh, FieldMesh
+from nbodykit_amd.source.mesh import (CatalogMesh, FieldMesh,
+                                      ArrayMesh, LinearMesh)
 from nbodykit_amd.source.mesh.bigfile import BigFileMesh
 from nbodykit_amd.source.catalog.bigfile import BigFileCatalog
 from nbodykit_amd.base.catalog import CatalogSource

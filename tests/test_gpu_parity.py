@@ -650,3 +650,66 @@ def test_fused_cross_power_matches_unfused():
     scale = numpy.nanmax(numpy.abs(r_plain.power['power']))
     assert_allclose(r_fused.power['power'], r_plain.power['power'],
                     rtol=1e-10, atol=1e-12 * scale, equal_nan=True)
+
+
+# ---------------------------------------------------------------------------
+# ArrayMesh / LinearMesh (reference source/mesh/array.py, linear.py)
+# ---------------------------------------------------------------------------
+
+def test_arraymesh_fftpower_parity():
+    from nbodykit_amd.lab import ArrayMesh, FFTPower
+    from oracle.mesh import MeshGeometry, r2c
+    from oracle.fftpower import project_to_basis as oracle_project
+    rng = numpy.random.RandomState(31)
+    arr = 1.0 + 0.1 * rng.standard_normal((32, 32, 32))
+    mesh = ArrayMesh(arr, BoxSize=64.)
+    got = mesh.compute(mode='real')
+    assert_allclose(got.value.cpu().numpy(), arr, rtol=1e-15)
+
+    r = FFTPower(mesh, mode='1d')
+    geom = MeshGeometry([32] * 3, 64.)
+    c = r2c(arr, geom)
+    p3d = c * numpy.conj(c) * 64. ** 3
+    p3d.flat[0] = 0.0
+    (xm, _, y2d, N2d), _ = oracle_project(
+        p3d, geom, [r.power.edges['k'], numpy.linspace(-1, 1, 2)])
+    assert_allclose(r.power['power'], numpy.squeeze(y2d),
+                    rtol=1e-10, equal_nan=True)
+    assert_array_equal(r.power['modes'], numpy.squeeze(N2d))
+
+
+def test_linearmesh_unitary_flat_power():
+    # unitary_amplitude fixes |delta(k)|^2 = P/V exactly, so the
+    # measured P(k) is exactly P0 in every populated bin
+    from nbodykit_amd.lab import LinearMesh, FFTPower
+    P0 = 123.0
+    mesh = LinearMesh(lambda k: P0 * numpy.ones_like(k), BoxSize=128.,
+                      Nmesh=32, seed=11, unitary_amplitude=True)
+    r = FFTPower(mesh, mode='1d')
+    p = r.power['power'].real
+    assert_allclose(p[~numpy.isnan(p)], P0, rtol=1e-10)
+
+
+def test_linearmesh_inverted_phase():
+    from nbodykit_amd.lab import LinearMesh
+    kw = dict(BoxSize=128., Nmesh=16, seed=5)
+    P = lambda k: 10.0 * numpy.ones_like(k)
+    m1 = LinearMesh(P, **kw).compute(mode='real')
+    m2 = LinearMesh(P, inverted_phase=True, **kw).compute(mode='real')
+    total = m1.value + m2.value
+    assert_allclose(total.cpu().numpy(), 2.0, rtol=1e-10)
+
+
+def test_linearmesh_through_fftpower_statistics():
+    # non-unitary realization: bin means scatter around P0 with
+    # ~1/sqrt(Nmodes) errors; loose statistical bound
+    from nbodykit_amd.lab import LinearMesh, FFTPower
+    P0 = 50.0
+    mesh = LinearMesh(lambda k: P0 * numpy.ones_like(k), BoxSize=256.,
+                      Nmesh=64, seed=4)
+    r = FFTPower(mesh, mode='1d')
+    p = r.power['power'].real
+    N = r.power['modes']
+    ok = N > 50
+    chi = (p[ok] - P0) / (P0 * numpy.sqrt(2.0 / N[ok]))
+    assert numpy.abs(chi).max() < 5.0
