@@ -96,42 +96,33 @@ __device__ void lds_fft(cdouble* buf, int m, int bits, int TI,
 // ---- z-axis real <-> half-complex ------------------------------------
 
 // one contiguous real line of nz doubles per block -> nz/2+1 complex
-// LPB z-lines per block, interleaved as lds_fft columns — one line per
-// block leaves the memory pipe underfed (8 ms = 2.1 TB/s at 1024^3);
-// batching lines multiplies the in-flight loads per block
 __global__ void kfft_r2c_z(const double* __restrict__ real,
                            double* __restrict__ cplx,
-                           int64_t nz, int64_t nlines, int lpb,
-                           double scale,
+                           int64_t nz, double scale,
                            const cdouble* __restrict__ table /* W_nz */)
 {
-    extern __shared__ cdouble buf[];          // (nz/2) * lpb entries
+    extern __shared__ cdouble buf[];          // nz/2 entries
     const int m = (int)(nz >> 1);
     const int bits = 31 - __clz((unsigned)m);
-    const int64_t line0 = (int64_t)blockIdx.x * lpb;
-    const int L = (int)((nlines - line0 < lpb) ? nlines - line0 : lpb);
+    const int64_t line = blockIdx.x;
 
-    for (int w = threadIdx.x; w < m * L; w += blockDim.x) {
-        const int c = w / m;
-        const int q = w - c * m;
-        const cdouble* g = (const cdouble*)(real + (line0 + c) * nz);
-        buf[bitrev(q, bits) * lpb + c] = g[q];
-    }
+    const cdouble* g = (const cdouble*)(real + line * nz);  // packed pairs
+    for (int q = threadIdx.x; q < m; q += blockDim.x)
+        buf[bitrev(q, bits)] = g[q];
     __syncthreads();
 
-    lds_fft<false>(buf, m, bits, lpb, table);
+    lds_fft<false>(buf, m, bits, 1, table);
 
     // untwiddle split: X[k] = E[k] + W_nz^k * O[k], k = 0..m
-    for (int w = threadIdx.x; w < (m + 1) * L; w += blockDim.x) {
-        const int c = w / (m + 1);
-        const int k = w - c * (m + 1);
-        const cdouble Zk = buf[(k == m ? 0 : k) * lpb + c];
-        const cdouble Zm = buf[((m - k) % m) * lpb + c];
+    cdouble* out = (cdouble*)cplx + line * (m + 1);
+    for (int k = threadIdx.x; k <= m; k += blockDim.x) {
+        const cdouble Zk = buf[k == m ? 0 : k];
+        const cdouble Zm = buf[(m - k) % m];
         const cdouble E = cscale(cadd(Zk, cconj(Zm)), 0.5);
         const cdouble D = csub(Zk, cconj(Zm));
         const cdouble O = {0.5 * D.im, -0.5 * D.re};    // D * (-i/2)
         const cdouble X = cadd(E, cmul(table[k], O));
-        ((cdouble*)cplx + (line0 + c) * (m + 1))[k] = cscale(X, scale);
+        out[k] = cscale(X, scale);
     }
 }
 
@@ -139,54 +130,47 @@ __global__ void kfft_r2c_z(const double* __restrict__ real,
 // (so that c2r(r2c(x, scale=1)) == nz * x per line)
 __global__ void kfft_c2r_z(const double* __restrict__ cplx,
                            double* __restrict__ real,
-                           int64_t nz, int64_t nlines, int lpb,
+                           int64_t nz,
                            const cdouble* __restrict__ table /* W_nz */)
 {
     extern __shared__ cdouble smem[];
     const int m = (int)(nz >> 1);
     const int bits = 31 - __clz((unsigned)m);
-    cdouble* xin = smem;                      // (m+1) * lpb entries
-    cdouble* buf = smem + (m + 1) * lpb;      // m * lpb entries
-    const int64_t line0 = (int64_t)blockIdx.x * lpb;
-    const int L = (int)((nlines - line0 < lpb) ? nlines - line0 : lpb);
+    cdouble* xin = smem;                      // m+1 entries
+    cdouble* buf = smem + (m + 1);            // m entries
+    const int64_t line = blockIdx.x;
 
-    for (int w = threadIdx.x; w < (m + 1) * L; w += blockDim.x) {
-        const int c = w / (m + 1);
-        const int k = w - c * (m + 1);
-        cdouble v = ((const cdouble*)cplx + (line0 + c) * (m + 1))[k];
+    const cdouble* g = (const cdouble*)cplx + line * (m + 1);
+    for (int k = threadIdx.x; k <= m; k += blockDim.x) {
+        cdouble v = g[k];
         // FFTW/numpy c2r convention: the self-conjugate DC and Nyquist
         // bins contribute only their real part (pmesh sits on FFTW; the
         // interlaced combine feeds bins with nonzero imaginary parts
         // there, source/mesh/catalog.py:341-351)
         if (k == 0 || k == m) v.im = 0.0;
-        xin[k * lpb + c] = v;
+        xin[k] = v;
     }
     __syncthreads();
 
     // rebuild packed spectrum: Z[k] = E[k] + i * O[k],
     // E = (X[k]+conj(X[m-k]))/2, O = conj(W^k) (X[k]-conj(X[m-k]))/2
-    for (int w = threadIdx.x; w < m * L; w += blockDim.x) {
-        const int c = w / m;
-        const int k = w - c * m;
-        const cdouble Xk = xin[k * lpb + c];
-        const cdouble Xm = cconj(xin[(m - k) * lpb + c]);
+    for (int k = threadIdx.x; k < m; k += blockDim.x) {
+        const cdouble Xk = xin[k];
+        const cdouble Xm = cconj(xin[m - k]);
         const cdouble E = cscale(cadd(Xk, Xm), 0.5);
         const cdouble WO = cscale(csub(Xk, Xm), 0.5);
         const cdouble O = cmul(cconj(table[k]), WO);
         const cdouble Z = {E.re - O.im, E.im + O.re};    // E + i O
-        buf[bitrev(k, bits) * lpb + c] = Z;
+        buf[bitrev(k, bits)] = Z;
     }
     __syncthreads();
 
-    lds_fft<true>(buf, m, bits, lpb, table);
+    lds_fft<true>(buf, m, bits, 1, table);
 
     // unpack: line[2t] = 2 Re(z[t]), line[2t+1] = 2 Im(z[t])
-    for (int w = threadIdx.x; w < m * L; w += blockDim.x) {
-        const int c = w / m;
-        const int t = w - c * m;
-        ((cdouble*)(real + (line0 + c) * nz))[t] =
-            {2.0 * buf[t * lpb + c].re, 2.0 * buf[t * lpb + c].im};
-    }
+    cdouble* out = (cdouble*)(real + line * nz);
+    for (int t = threadIdx.x; t < m; t += blockDim.x)
+        out[t] = {2.0 * buf[t].re, 2.0 * buf[t].im};
 }
 
 // ---- strided complex pass --------------------------------------------
@@ -245,17 +229,11 @@ extern "C" int nbk_fft_r2c_z(const double* real, double* cplx,
     if (nlines == 0) return NBK_OK;
     double* table = get_twiddles(nz);
     if (!table) { NBK_SET_ERR("twiddle alloc failed"); return NBK_ERR_HIP; }
-    // batch lines per block while (nz/2)*lpb cdoubles fit in 64 KiB
-    int lpb = 1;
-    while ((size_t)(nz / 2) * lpb * 2 * sizeof(cdouble) <= 65536
-           && lpb < 8 && lpb * 2 <= nlines)
-        lpb *= 2;
-    const int block = 512;
-    const size_t shmem = (size_t)(nz / 2) * lpb * sizeof(cdouble);
-    const int64_t grid = (nlines + lpb - 1) / lpb;
-    hipLaunchKernelGGL(kfft_r2c_z, dim3((uint32_t)grid), dim3(block), shmem,
-                       (hipStream_t)stream, real, cplx, nz, nlines, lpb,
-                       scale, (const cdouble*)table);
+    const int block = (int)std::min<int64_t>(512, nz / 4);
+    const size_t shmem = (size_t)(nz / 2) * sizeof(cdouble);
+    hipLaunchKernelGGL(kfft_r2c_z, dim3((uint32_t)nlines), dim3(block), shmem,
+                       (hipStream_t)stream, real, cplx, nz, scale,
+                       (const cdouble*)table);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }
@@ -268,19 +246,10 @@ extern "C" int nbk_fft_c2r_z(const double* cplx, double* real,
     if (nlines == 0) return NBK_OK;
     double* table = get_twiddles(nz);
     if (!table) { NBK_SET_ERR("twiddle alloc failed"); return NBK_ERR_HIP; }
-    int lpb = 1;
-    while ((size_t)(nz + 2) * lpb * sizeof(cdouble) * 2 <= 131072
-           && lpb < 8 && lpb * 2 <= nlines)
-        lpb *= 2;
-    const int block = 512;
-    const size_t shmem = (size_t)(nz / 2 * 2 + 1) * lpb * sizeof(cdouble);
-    const int64_t grid = (nlines + lpb - 1) / lpb;
-    if (shmem > 65536)
-        (void)hipFuncSetAttribute(
-            reinterpret_cast<const void*>(&kfft_c2r_z),
-            hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);
-    hipLaunchKernelGGL(kfft_c2r_z, dim3((uint32_t)grid), dim3(block), shmem,
-                       (hipStream_t)stream, cplx, real, nz, nlines, lpb,
+    const int block = (int)std::min<int64_t>(512, nz / 4);
+    const size_t shmem = (size_t)(nz + 1) * sizeof(cdouble);
+    hipLaunchKernelGGL(kfft_c2r_z, dim3((uint32_t)nlines), dim3(block), shmem,
+                       (hipStream_t)stream, cplx, real, nz,
                        (const cdouble*)table);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
