@@ -98,6 +98,20 @@ class Cosmology(object):
         a = 1.0 / (1.0 + numpy.asarray(z, dtype='f8'))
         return self.growth.f1(a)
 
+    def comoving_distance(self, z):
+        """Line-of-sight comoving distance in Mpc/h:
+        Dc = (c/H0) int_0^z dz'/E(z') (the reference wraps CLASS's
+        background; flat matter+Lambda integral here — the neglected
+        radiation term shifts Dc by <0.1% at survey redshifts)."""
+        z = numpy.asarray(z, dtype='f8')
+        zmax = float(z.max()) if z.size else 0.0
+        grid = numpy.linspace(0.0, max(zmax, 1e-8), 4096)
+        integrand = 1.0 / self.efunc(grid)
+        dc = numpy.concatenate([[0.0], numpy.cumsum(
+            0.5 * (integrand[1:] + integrand[:-1]) * numpy.diff(grid))])
+        # c / H0 with H0 = 100 h km/s/Mpc -> Mpc/h units
+        return 2997.92458 * numpy.interp(z, grid, dc)
+
 
 # Planck15 parameters (astropy's FlatLambdaCDM Planck15 + the sigma8/n_s
 # values nbodykit adds in cosmology/__init__.py:16-21).
