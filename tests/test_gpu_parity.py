@@ -719,3 +719,31 @@ def test_linearmesh_through_fftpower_statistics():
     ok = N > 50
     chi = (p[ok] - P0) / (P0 * numpy.sqrt(2.0 / N[ok]))
     assert numpy.abs(chi).max() < 5.0
+
+
+def test_survey_workflow_end_to_end():
+    """SkyToCartesian -> FKPCatalog -> ConvolvedFFTPower: the full
+    survey pipeline composes (transform + cosmology + FKP)."""
+    from nbodykit_amd.lab import (transform, ArrayCatalog, FKPCatalog,
+                                  ConvolvedFFTPower)
+    from nbodykit_amd.cosmology import Planck15
+    rng = numpy.random.RandomState(77)
+
+    def make(n):
+        ra = rng.uniform(20., 40., n)
+        dec = rng.uniform(-10., 10., n)
+        z = rng.uniform(0.4, 0.6, n)
+        pos = transform.SkyToCartesian(ra, dec, z, Planck15)
+        return ArrayCatalog({'Position': pos,
+                             'NZ': numpy.full(n, 3e-4)})
+
+    data, ran = make(2000), make(20000)
+    cat = FKPCatalog(data, ran, P0=1e4)
+    with pytest.warns(UserWarning):
+        mesh = cat.to_mesh(Nmesh=32, dtype='c16', compensated=True)
+    r = ConvolvedFFTPower(mesh, poles=[0, 2], dk=0.05)
+    P0m = r.poles['power_0'].real - r.attrs['shotnoise']
+    assert numpy.isfinite(r.attrs['shotnoise'])
+    assert numpy.nanmax(numpy.abs(P0m)) < 50 * r.attrs['shotnoise']
+    # bbox derived from the randoms covers the data
+    assert (numpy.asarray(r.attrs['BoxSize']) > 0).all()
