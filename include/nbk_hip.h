@@ -115,6 +115,19 @@ int nbk_paint_gather_f64(const double* pos, const double* mass, int64_t n,
                          double* mesh, int64_t x0, int64_t nx_local,
                          int accumulate, void* stream);
 
+/* gather paint with the forward z-axis FFT fused into the tile flush:
+ * writes the z half-spectrum ((nx_local, n1, n2/2+1) interleaved c128,
+ * every element scaled by `scale`) directly — the real mesh never
+ * exists in HBM.  Same source/rowtab semantics as nbk_paint_gather_f64;
+ * overwrite-only (single paint per output).  n2 must be a power of two
+ * in [8, 4096].  The FFT math is identical to nbk_fft_r2c_z. */
+int nbk_paint_gather_fft_f64(const double* pos, const double* mass,
+                             int64_t n, const int64_t nmesh[3],
+                             const double box[3],
+                             int window, double shift, const int* rowtab,
+                             double* zspec, int64_t x0, int64_t nx_local,
+                             double scale, void* stream);
+
 /* paint locality sort --------------------------------------------------
  * Two-pass counting sort of particles by coarse mesh cell
  * (bucket = wrapped ix * n1 + iy): count, then (after the caller turns

@@ -62,6 +62,15 @@ __device__ __forceinline__ double freq_half(int64_t g, int64_t n) {
     return (double)(g == n / 2 ? -(n / 2) : g);
 }
 
+// bit-reversal for the in-tile FFT (shared with nbk_fft.hip's local copy)
+__device__ __forceinline__ int nbk_bitrev(int j, int bits) {
+    return (int)(__brev((unsigned)j) >> (32 - bits));
+}
+
+// cross-TU access to nbk_fft.hip's cached device twiddle tables
+// (W_N^j for j = 0..N/2, defined in nbk_fft.hip)
+double* nbk_internal_twiddles(int64_t N);
+
 // Inverse-window compensation factor for one element (the six filters of
 // nbodykit/source/mesh/catalog.py:453-594).  w[i] = 2 pi f_i / N_i is the
 // circular frequency in [-pi, pi); interlaced selects the plain

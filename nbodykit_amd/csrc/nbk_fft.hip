@@ -316,3 +316,6 @@ extern "C" int nbk_fft_c_strided(double* cplx, int64_t nfft, int64_t stride,
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }
+
+// cross-TU handle for the fused paint+z-FFT kernel (nbk_paint.hip)
+double* nbk_internal_twiddles(int64_t N) { return get_twiddles(N); }

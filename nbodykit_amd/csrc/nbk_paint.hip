@@ -346,7 +346,12 @@ __global__ void kpaint_tiled(const double* __restrict__ px,
 // bounds the scatter kernels above at ~8 atomics/particle; here the
 // deposits are LDS ds_add_f64 and the HBM traffic is ~2.1x the particle
 // reads plus one mesh write.
-template <int WINDOW>
+// DOFFT: after the deposits the tile's rows (full z-lines) are
+// transformed in place (packed-real radix-2, identical math to
+// kfft_r2c_z) and the z half-spectrum is written directly — the real
+// mesh never touches HBM.  The row stride is padded (+4 doubles) so the
+// butterflies of different rows land in different LDS banks.
+template <int WINDOW, bool DOFFT>
 __global__ void kpaint_gather(const double* __restrict__ px,
                               const double* __restrict__ py,
                               const double* __restrict__ pz,
@@ -355,19 +360,24 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                               double invH0, double invH1, double invH2,
                               double shift,
                               const int* __restrict__ rowtab,
-                              double* __restrict__ mesh,
+                              double* __restrict__ mesh, /* or z-spectrum
+                                  (nx_local, n1, n2/2+1) cdouble when
+                                  DOFFT */
                               int64_t x0, int64_t nx_local,
-                              int RG, int xlo, int xhi, int accumulate)
+                              int RG, int xlo, int xhi, int accumulate,
+                              const cdouble* __restrict__ table /* W_n2 */,
+                              double scale)
 {
     constexpr int SUP = (WINDOW == NBK_WINDOW_CIC) ? 2
                       : (WINDOW == NBK_WINDOW_TSC) ? 3 : 4;
-    extern __shared__ double tile[];      // RG * n2 doubles
+    extern __shared__ double tile[];      // RG * (n2 [+4]) doubles
+    const int64_t sp = DOFFT ? n2 + 4 : n2;   // padded row stride
     const int64_t tiles_per_plane = n1 / RG;
     const int64_t ix = x0 + blockIdx.x / tiles_per_plane;
     const int64_t r0 = (blockIdx.x % tiles_per_plane) * RG;
     const int T = blockDim.x;
     const int t = threadIdx.x;
-    const int64_t win = (int64_t)RG * n2;
+    const int64_t win = (int64_t)RG * sp;
 
     for (int64_t w = t; w < win; w += T) tile[w] = 0.0;
     __syncthreads();
@@ -413,7 +423,7 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                         #pragma unroll
                         for (int dz = 0; dz < SUP; dz++) {
                             const int64_t gz = wrap_idx(b2 + dz, n2);
-                            unsafeAtomicAdd(&tile[ly * n2 + gz],
+                            unsafeAtomicAdd(&tile[ly * sp + gz],
                                             wxy * w2[dz]);
                         }
                     }
@@ -423,14 +433,75 @@ __global__ void kpaint_gather(const double* __restrict__ px,
     }
     __syncthreads();
 
-    // flush the exclusively-owned tile with plain stores
-    double* dst = mesh + ((ix - x0) * n1 + r0) * n2;
-    if (accumulate) {
-        for (int64_t w = t; w < win; w += T)
-            if (tile[w] != 0.0) dst[w] += tile[w];
-    } else {
-        for (int64_t w = t; w < win; w += T)
-            dst[w] = tile[w];
+    if (!DOFFT) {
+        // flush the exclusively-owned tile with plain stores
+        double* dst = mesh + ((ix - x0) * n1 + r0) * n2;
+        if (accumulate) {
+            for (int64_t w = t; w < win; w += T) {
+                const int64_t r = w / sp, z = w - r * sp;
+                if (z < n2 && tile[w] != 0.0)
+                    dst[r * n2 + z] += tile[w];
+            }
+        } else {
+            for (int64_t w = t; w < win; w += T) {
+                const int64_t r = w / sp, z = w - r * sp;
+                if (z < n2) dst[r * n2 + z] = tile[w];
+            }
+        }
+        return;
+    }
+
+    // ---- fused forward z-FFT per tile row (math identical to
+    // kfft_r2c_z: packed-real radix-2 DIT + untwiddle split) ----
+    const int m = (int)(n2 >> 1);
+    const int bits = 31 - __clz((unsigned)m);
+
+    // in-place bit-reversal permutation of each row's packed pairs
+    for (int w = t; w < RG * m; w += T) {
+        const int r = w / m, j = w - r * m;
+        const int jr = nbk_bitrev(j, bits);
+        if (j < jr) {
+            cdouble* z = (cdouble*)&tile[(int64_t)r * sp];
+            const cdouble a = z[j];
+            z[j] = z[jr];
+            z[jr] = a;
+        }
+    }
+    __syncthreads();
+
+    for (int len = 2; len <= m; len <<= 1) {
+        const int half = len >> 1;
+        const int tw = m / len;
+        for (int w = t; w < RG * (m >> 1); w += T) {
+            const int r = w / (m >> 1);
+            const int q = w - r * (m >> 1);
+            const int grp = q / half;
+            const int pos = q - grp * half;
+            cdouble* z = (cdouble*)&tile[(int64_t)r * sp];
+            const int i0 = grp * len + pos;
+            const int i1 = i0 + half;
+            const cdouble wv = table[2 * pos * tw];
+            const cdouble u = z[i0];
+            const cdouble v = cmul(z[i1], wv);
+            z[i0] = cadd(u, v);
+            z[i1] = csub(u, v);
+        }
+        __syncthreads();
+    }
+
+    // untwiddle split: X[k] = E[k] + W_n2^k O[k], k = 0..m, written
+    // straight to the z half-spectrum
+    cdouble* out = (cdouble*)mesh + ((ix - x0) * n1 + r0) * (m + 1);
+    for (int w = t; w < RG * (m + 1); w += T) {
+        const int r = w / (m + 1), k = w - r * (m + 1);
+        const cdouble* z = (const cdouble*)&tile[(int64_t)r * sp];
+        const cdouble Zk = z[k == m ? 0 : k];
+        const cdouble Zm = z[(m - k) % m];
+        const cdouble E = cscale(cadd(Zk, cconj(Zm)), 0.5);
+        const cdouble D = csub(Zk, cconj(Zm));
+        const cdouble O = {0.5 * D.im, -0.5 * D.re};    // D * (-i/2)
+        const cdouble X = cadd(E, cmul(table[k], O));
+        out[(int64_t)r * (m + 1) + k] = cscale(X, scale);
     }
 }
 
@@ -722,23 +793,103 @@ extern "C" int nbk_paint_gather_f64(const double* pos, const double* mass,
     hipStream_t s = (hipStream_t)stream;
     if (lds > 64 * 1024) {
         const void* fns[3] = {
-            reinterpret_cast<const void*>(&kpaint_gather<NBK_WINDOW_CIC>),
-            reinterpret_cast<const void*>(&kpaint_gather<NBK_WINDOW_TSC>),
-            reinterpret_cast<const void*>(&kpaint_gather<NBK_WINDOW_PCS>)};
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_CIC, false>),
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_TSC, false>),
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_PCS, false>)};
         (void)hipFuncSetAttribute(fns[window],
             hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     }
     #define NBK_LAUNCH_GATHER(W) \
-        hipLaunchKernelGGL(kpaint_gather<W>, dim3((uint32_t)grid), \
+        hipLaunchKernelGGL((kpaint_gather<W, false>), \
+                           dim3((uint32_t)grid), \
                            dim3(1024), lds, s, pos, pos + n, pos + 2 * n, \
                            mass, n, n0, n1, n2, \
                            n0 / box[0], n1 / box[1], n2 / box[2], shift, \
                            rowtab, mesh, x0, nx_local, (int)RG, xlo, xhi, \
-                           accumulate)
+                           accumulate, (const cdouble*)nullptr, 1.0)
     if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GATHER(NBK_WINDOW_CIC);
     else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GATHER(NBK_WINDOW_TSC);
     else NBK_LAUNCH_GATHER(NBK_WINDOW_PCS);
     #undef NBK_LAUNCH_GATHER
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}
+
+extern "C" int nbk_paint_gather_fft_f64(const double* pos,
+                                        const double* mass, int64_t n,
+                                        const int64_t nmesh[3],
+                                        const double box[3],
+                                        int window, double shift,
+                                        const int* rowtab,
+                                        double* zspec, int64_t x0,
+                                        int64_t nx_local, double scale,
+                                        void* stream)
+{
+    const int64_t n0 = nmesh[0], n1 = nmesh[1], n2 = nmesh[2];
+    if (n2 < 8 || n2 > 4096 || (n2 & (n2 - 1))) {
+        NBK_SET_ERR("nbk_paint_gather_fft_f64: n2=%lld not a supported "
+                    "FFT length", (long long)n2);
+        return NBK_ERR_UNSUPPORTED;
+    }
+    // padded tile: RG * (n2 + 4) f64 within 160 KiB
+    int64_t RG = 1;
+    while (RG * 2 * (n2 + 4) <= 20480 && (n1 % (RG * 2)) == 0
+           && RG * 2 <= n1)
+        RG *= 2;
+    if (RG * (n2 + 4) > 20480 || (n1 % RG)) {
+        NBK_SET_ERR("nbk_paint_gather_fft_f64: no LDS tile for n1=%lld "
+                    "n2=%lld", (long long)n1, (long long)n2);
+        return NBK_ERR_UNSUPPORTED;
+    }
+    const double* table = nbk_internal_twiddles(n2);
+    if (!table) {
+        NBK_SET_ERR("twiddle alloc failed");
+        return NBK_ERR_HIP;
+    }
+    const bool sh = shift != 0.0;
+    int sup, dmin, dmax;
+    if (window == NBK_WINDOW_CIC) {
+        sup = 2; dmin = 0; dmax = sh ? 1 : 0;
+    } else if (window == NBK_WINDOW_TSC) {
+        sup = 3; dmin = sh ? 0 : -1; dmax = 0;
+    } else if (window == NBK_WINDOW_PCS) {
+        sup = 4; dmin = -1; dmax = sh ? 0 : -1;
+    } else {
+        NBK_SET_ERR("nbk_paint_gather_fft_f64: bad window %d", window);
+        return NBK_ERR_ARG;
+    }
+    const int xlo = -sup + 1 - dmax;
+    const int xhi = -dmin;
+
+    const int64_t grid = nx_local * (n1 / RG);
+    const size_t lds = (size_t)RG * (n2 + 4) * sizeof(double);
+    hipStream_t s = (hipStream_t)stream;
+    if (lds > 64 * 1024) {
+        const void* fns[3] = {
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_CIC, true>),
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_TSC, true>),
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_PCS, true>)};
+        (void)hipFuncSetAttribute(fns[window],
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    }
+    #define NBK_LAUNCH_GFFT(W) \
+        hipLaunchKernelGGL((kpaint_gather<W, true>), \
+                           dim3((uint32_t)grid), \
+                           dim3(1024), lds, s, pos, pos + n, pos + 2 * n, \
+                           mass, n, n0, n1, n2, \
+                           n0 / box[0], n1 / box[1], n2 / box[2], shift, \
+                           rowtab, zspec, x0, nx_local, (int)RG, xlo, \
+                           xhi, 0, (const cdouble*)table, scale)
+    if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GFFT(NBK_WINDOW_CIC);
+    else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GFFT(NBK_WINDOW_TSC);
+    else NBK_LAUNCH_GFFT(NBK_WINDOW_PCS);
+    #undef NBK_LAUNCH_GFFT
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

@@ -141,6 +141,28 @@ class ParticleMesh(object):
                 and numpy.array_equal(self.BoxSize, other.BoxSize))
 
 
+def _r2c_finish(cplx, pm, s):
+    """y and x strided passes (+ the RCCL pencil transpose when
+    distributed) over an existing z half-spectrum — the tail of
+    RealField.r2c, shared with the fused paint+z-FFT path
+    (source/mesh/catalog.py to_complex_field)."""
+    lib = hiplib.require()
+    nx_l, ny, nzh = cplx.shape
+    hiplib.check(lib.nbk_fft_c_strided(
+        hiplib.dptr(cplx), ny, nzh, nx_l, ny * nzh, nzh, -1, s),
+        'nbk_fft_c_strided(y)')
+    ws = pm.comm.size
+    if ws > 1:
+        cplx = transpose_x_to_y(cplx, ws, nx_l, pm.ny_local, nzh)
+        n_inner = pm.ny_local * nzh
+    else:
+        n_inner = ny * nzh
+    hiplib.check(lib.nbk_fft_c_strided(
+        hiplib.dptr(cplx), int(pm.Nmesh[0]), n_inner, 1, 0, n_inner,
+        -1, s), 'nbk_fft_c_strided(x)')
+    return cplx
+
+
 class _FieldBase(object):
     """attrs + shared plumbing for Real/Complex fields."""
 
@@ -239,22 +261,7 @@ class RealField(_FieldBase):
         hiplib.check(lib.nbk_fft_r2c_z(
             hiplib.dptr(self.value), hiplib.dptr(cplx),
             nx_l * ny, nz, scale, s), 'nbk_fft_r2c_z')
-        # y pass: lines along axis 1
-        hiplib.check(lib.nbk_fft_c_strided(
-            hiplib.dptr(cplx), ny, nzh, nx_l, ny * nzh, nzh, -1, s),
-            'nbk_fft_c_strided(y)')
-
-        ws = pm.comm.size
-        if ws > 1:
-            cplx = transpose_x_to_y(cplx, ws, nx_l, pm.ny_local, nzh)
-            n_inner = pm.ny_local * nzh
-        else:
-            n_inner = ny * nzh
-        # x pass: lines along axis 0 of the (possibly transposed) block
-        hiplib.check(lib.nbk_fft_c_strided(
-            hiplib.dptr(cplx), int(pm.Nmesh[0]), n_inner, 1, 0, n_inner,
-            -1, s), 'nbk_fft_c_strided(x)')
-
+        cplx = _r2c_finish(cplx, pm, s)
         f = ComplexField(pm, tensor=cplx)
         f.attrs.update(self.attrs)
         return f

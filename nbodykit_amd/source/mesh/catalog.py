@@ -20,7 +20,8 @@ import numpy
 from nbodykit_amd import _global_options
 from nbodykit_amd import hiplib, profiling
 from nbodykit_amd.base.mesh import MeshSource
-from nbodykit_amd.pm import RealField, exchange_particle_arrays
+from nbodykit_amd.pm import (RealField, ComplexField,
+                             exchange_particle_arrays)
 
 
 def _to_device_f64(arr):
@@ -453,6 +454,132 @@ class CatalogMesh(MeshSource):
                 toret.value.fill_(1.0)
 
         return toret
+
+    def to_complex_field(self, out=None):
+        """Fused paint -> forward-FFT fast path: the z pass runs inside
+        the gather paint's tile flush, so the real mesh never exists in
+        HBM; interlaced meshes additionally skip the c2r + re-r2c round
+        trip of the real path (combine directly in k).  Numerically the
+        FFT is the same radix-2 code as nbk_fft_r2c_z; the 1/N^3 and
+        1/nbar (normalize) factors fold into the kernel's output scale.
+        Falls back (NotImplemented -> to_real_field().r2c() in
+        MeshSource.to_field) for multi-rank runs, user `out`, paints
+        that need chunking, non-power-of-two z, or inputs below the
+        locality-sort thresholds."""
+        import torch
+        if out is not None:
+            return NotImplemented
+        pm = self.pm
+        comm = pm.comm
+        if comm.size != 1:
+            return NotImplemented
+        n2 = int(pm.Nmesh[2])
+        if n2 < 8 or n2 > 4096 or (n2 & (n2 - 1)):
+            return NotImplemented
+        Position = self.Position
+        if len(Position) > _global_options['paint_chunk_size']:
+            return NotImplemented
+
+        lib = hiplib.require()
+        interlaced = self.interlaced
+        window_id = hiplib.WINDOW_IDS[self.resampler]
+        nmesh = hiplib.i64_arr(pm.Nmesh)
+        box = hiplib.f64_arr(pm.BoxSize)
+        stream = hiplib.cur_stream()
+
+        # single-chunk column pull (the chunk loop of to_real_field)
+        sel = None if self.Selection is None else self.Selection
+        value = None if self.Value is None else self.Value
+        weight = None if self.Weight is None else self.Weight
+        data = self.source.compute([c for c in
+                                    [Position, weight, value, sel]
+                                    if c is not None])
+        position = data[0]
+        idx = 1
+        if weight is not None:
+            weight = data[idx]; idx += 1
+        if value is not None:
+            value = data[idx]; idx += 1
+        if sel is not None:
+            sel = data[idx]
+        if sel is not None and _is_trivial_true(sel):
+            sel = None
+        if sel is not None:
+            m = _to_device_mask(sel)
+            position = _to_device_f64(position)[m]
+            if weight is not None:
+                weight = _to_device_f64(weight)[m]
+            if value is not None:
+                value = _to_device_f64(value)[m]
+        if weight is not None and _is_trivial_unit(weight):
+            weight = None
+        if value is not None and _is_trivial_unit(value):
+            value = None
+
+        pos_t = _to_device_f64(position)
+        w_t = None if weight is None else _to_device_f64(weight)
+        v_t = None if value is None else _to_device_f64(value)
+
+        N = len(pos_t)
+        if w_t is None:
+            W = float(N)
+            W2 = float(N)
+        else:
+            W = float(w_t.sum().item())
+            W2 = float((w_t * w_t).sum().item())
+        if w_t is None and v_t is None:
+            mass_t = None
+        elif v_t is None:
+            mass_t = w_t
+        elif w_t is None:
+            mass_t = v_t
+        else:
+            mass_t = w_t * v_t
+
+        nbar = W / float(numpy.prod(pm.Nmesh))
+        if N == 0 or nbar <= 0:
+            return NotImplemented
+
+        pos_soa, mass_t, sorted_, rowtab = _prepare_particles(
+            pos_t, mass_t, pm)
+        if rowtab is None:
+            return NotImplemented
+
+        scale = 1.0 / float(numpy.prod(pm.Nmesh)) / nbar
+        nzh = n2 // 2 + 1
+        shape = (pm.nx_local, int(pm.Nmesh[1]), nzh)
+
+        from nbodykit_amd.pm import _r2c_finish
+
+        def one(shift):
+            z = torch.empty(shape, dtype=torch.complex128, device='cuda')
+            hiplib.check(lib.nbk_paint_gather_fft_f64(
+                hiplib.dptr(pos_soa), hiplib.dptr(mass_t), N, nmesh, box,
+                window_id, float(shift), hiplib.dptr(rowtab),
+                hiplib.dptr(z), pm.x_start, pm.nx_local, scale, stream),
+                'nbk_paint_gather_fft_f64')
+            return _r2c_finish(z, pm, stream)
+
+        with profiling.collect('paint', N * (1 + interlaced)):
+            cplx = one(0.0)
+            if interlaced:
+                c2t = one(0.5)
+                c1f = ComplexField(pm, tensor=cplx)
+                c2f = ComplexField(pm, tensor=c2t)
+                hiplib.check(lib.nbk_interlace_combine_f64(
+                    hiplib.dptr(c1f.value), hiplib.dptr(c2f.value),
+                    nmesh, box, hiplib.i64_arr(c1f.dims),
+                    hiplib.i64_arr(c1f.off), None, stream),
+                    'nbk_interlace_combine_f64')
+                cplx = c1f.value
+
+        with numpy.errstate(divide='ignore', invalid='ignore'):
+            shotnoise = float(numpy.prod(pm.BoxSize)) * W2 / W ** 2
+
+        f = ComplexField(pm, tensor=cplx)
+        f.attrs = {'shotnoise': shotnoise, 'N': N, 'W': W, 'W2': W2,
+                   'num_per_cell': nbar}
+        return f
 
     def _route(self, pos_t, mass_t):
         dmin, dmax = _GHOST_RANGE[(self.resampler, bool(self.interlaced))]

@@ -747,3 +747,54 @@ def test_survey_workflow_end_to_end():
     assert numpy.nanmax(numpy.abs(P0m)) < 50 * r.attrs['shotnoise']
     # bbox derived from the randoms covers the data
     assert (numpy.asarray(r.attrs['BoxSize']) > 0).all()
+
+
+def test_fused_complex_paint_matches_real_path():
+    """CatalogMesh.to_complex_field (paint with the z-FFT fused into the
+    tile flush) must equal r2c of the real path, plain and interlaced,
+    once the two-level thresholds admit the gather kernel."""
+    import torch
+    from nbodykit_amd import set_options
+    from nbodykit_amd.lab import ArrayCatalog
+    rng = numpy.random.RandomState(41)
+    n = 180000
+    cat = ArrayCatalog({'Position': rng.uniform(0, 64., size=(n, 3)),
+                        'Weight': rng.exponential(size=n)})
+    for interlaced, resampler in [(False, 'cic'), (True, 'tsc'),
+                                  (False, 'pcs')]:
+        kw = dict(Nmesh=64, BoxSize=64., dtype='f8', resampler=resampler,
+                  interlaced=interlaced, compensated=False)
+        with set_options(sort_min_n=1024, sort_two_level_min_n=1024,
+                         sort_two_level_min_cells=1):
+            mesh = cat.to_mesh(**kw)
+            c_fused = mesh.to_complex_field()
+            assert c_fused is not NotImplemented
+            c_ref = mesh.to_real_field().r2c()
+        a = c_fused.value.cpu().numpy()
+        b = c_ref.value.cpu().numpy()
+        scale = numpy.abs(b).max()
+        assert_allclose(a, b, atol=1e-12 * scale, rtol=1e-10,
+                        err_msg='interlaced=%s %s' % (interlaced,
+                                                      resampler))
+        for key in ['N', 'W', 'shotnoise']:
+            assert_allclose(c_fused.attrs[key], c_ref.attrs[key],
+                            rtol=1e-12)
+
+
+def test_fused_complex_paint_through_fftpower():
+    # the whole FFTPower pipeline rides the fused path when thresholds
+    # allow; results must agree with the unfused pipeline
+    from nbodykit_amd import set_options
+    from nbodykit_amd.lab import UniformCatalog, FFTPower
+    cat = UniformCatalog(nbar=1.2, BoxSize=48., seed=3)   # ~1.3e5 pts
+    kw = dict(mode='2d', Nmu=4, poles=[0, 2], Nmesh=48)
+    r_plain = FFTPower(cat, **kw)
+    with set_options(sort_min_n=1024, sort_two_level_min_n=1024,
+                     sort_two_level_min_cells=1):
+        r_fused = FFTPower(cat, **kw)
+    scale = numpy.nanmax(numpy.abs(r_plain.power['power']))
+    assert_allclose(r_fused.power['power'], r_plain.power['power'],
+                    rtol=1e-10, atol=1e-11 * scale, equal_nan=True)
+    assert_array_equal(r_fused.power['modes'], r_plain.power['modes'])
+    assert_allclose(r_fused.attrs['shotnoise'],
+                    r_plain.attrs['shotnoise'], rtol=1e-12)
