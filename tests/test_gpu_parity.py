@@ -786,8 +786,8 @@ def test_fused_complex_paint_through_fftpower():
     # allow; results must agree with the unfused pipeline
     from nbodykit_amd import set_options
     from nbodykit_amd.lab import UniformCatalog, FFTPower
-    cat = UniformCatalog(nbar=1.2, BoxSize=48., seed=3)   # ~1.3e5 pts
-    kw = dict(mode='2d', Nmu=4, poles=[0, 2], Nmesh=48)
+    cat = UniformCatalog(nbar=0.5, BoxSize=64., seed=3)   # ~1.3e5 pts
+    kw = dict(mode='2d', Nmu=4, poles=[0, 2], Nmesh=64)
     r_plain = FFTPower(cat, **kw)
     with set_options(sort_min_n=1024, sort_two_level_min_n=1024,
                      sort_two_level_min_cells=1):
