@@ -401,7 +401,9 @@ def main():
                 'unit': 'GB/s',
                 'frac': achieved / HBM_PEAK,
                 'traffic': traffic,
-                'kernel': ('nbk_paint_gather_f64[%s]' if gather else
+                'kernel': ('nbk_paint_gather_fft_f64[%s]'
+                           if gather and ws == 1 else
+                           'nbk_paint_gather_f64[%s]' if gather else
                            'nbk_paint_f64[%s]') % cfg['resampler'],
                 'paint_ms_per_launch': (paint['ms'] / paint['calls']
                                         if paint['calls'] else None),
