@@ -268,7 +268,10 @@ def main():
     if ws > 1:
         import torch.distributed as dist
         local = int(os.environ.get('LOCAL_RANK', rank))
-        torch.cuda.set_device(local)
+        # modulo lets an N-rank run share fewer devices (single-GPU
+        # smoke testing of the multi-rank path); production launches
+        # have one rank per GPU and the modulo is a no-op
+        torch.cuda.set_device(local % max(1, torch.cuda.device_count()))
         dist.init_process_group('nccl')
 
     from nbodykit_amd import profiling, set_options
