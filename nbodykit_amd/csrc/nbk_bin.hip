@@ -366,7 +366,16 @@ static int launch_bin(const double* d1, const double* d2, BinArgs& A,
     const int64_t nlines = A.d0 * A.d1;
     const int64_t ngroups = (nlines + 7) / 8;     // NL = 8 in the kernel
     int64_t g = ngroups;
-    if (g > 8192) g = 8192;    // bounded: LDS flush cost scales with grid
+    // bounded: each block's LDS histogram flush costs NB*nfields global
+    // atomics, and the atomic pipe runs at ~25 G op/s (count_probe) —
+    // the cap is tunable for experiments via NBK_BIN_GRID
+    static int64_t gcap = 0;
+    if (!gcap) {
+        const char* e = getenv("NBK_BIN_GRID");
+        gcap = e ? atoll(e) : 2048;
+        if (gcap < 256 || gcap > 65536) gcap = 2048;
+    }
+    if (g > gcap) g = gcap;
     if (g < 1) g = 1;
 
     hipStream_t s = (hipStream_t)stream;

@@ -258,3 +258,20 @@ def test_gloo_bigfile_catalog_roundtrip():
                 numpy.arange(10) * 3.0,
                 numpy.arange(15) * 3.0 + 100]))
         numpy.testing.assert_array_equal(r['box'], [7., 7., 7.])
+
+
+def _body_fused_complex_gate(comm):
+    """to_complex_field must decline (NotImplemented) on multi-rank
+    comms BEFORE touching the GPU extension (the real path handles
+    distribution)."""
+    from nbodykit_amd.lab import ArrayCatalog
+    cat = ArrayCatalog({'Position': numpy.random.RandomState(
+        comm.rank).uniform(0, 32., size=(100, 3))}, comm=comm)
+    mesh = cat.to_mesh(Nmesh=32, BoxSize=32., dtype='f8')
+    return mesh.to_complex_field() is NotImplemented
+
+
+@pytest.mark.timeout(300)
+def test_gloo_fused_complex_gate():
+    r0, r1 = _run_world('_body_fused_complex_gate')
+    assert r0 is True and r1 is True
