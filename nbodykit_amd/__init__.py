@@ -22,7 +22,12 @@ __version__ = "0.1.0"
 # Same three option names as the reference (nbodykit/__init__.py:22-25).
 # paint_chunk_size is the GPU paint batch size analogue.
 _global_options = {
-    'paint_chunk_size': 1024 * 1024 * 4,
+    # the reference defaults to 4M particles per paint chunk (a CPU
+    # memory-safety choice); with 288 GB of HBM a 2^28 chunk fits
+    # comfortably and keeps the two-level sort + gather paint engaged
+    # for big catalogs (set_options(paint_chunk_size=...) restores any
+    # other value)
+    'paint_chunk_size': 1 << 28,
     'dask_chunk_size': 100000,
     'global_cache_size': 1e8,
     # paint locality-sort thresholds (tests shrink these to exercise the
