@@ -451,57 +451,61 @@ __global__ void kpaint_gather(const double* __restrict__ px,
         return;
     }
 
-    // ---- fused forward z-FFT per tile row (math identical to
-    // kfft_r2c_z: packed-real radix-2 DIT + untwiddle split) ----
+    // ---- fused forward z-FFT, one WAVE per tile row (math identical
+    // to kfft_r2c_z: packed-real radix-2 DIT + untwiddle split).  A
+    // wave owns a whole row, so the stage ordering needs no block
+    // syncs — CDNA wave64 executes in lockstep and LDS ops complete in
+    // program order; the wave_barrier only pins the compiler's
+    // scheduling across the cross-lane dependences. ----
     const int m = (int)(n2 >> 1);
     const int bits = 31 - __clz((unsigned)m);
-
-    // in-place bit-reversal permutation of each row's packed pairs
-    for (int w = t; w < RG * m; w += T) {
-        const int r = w / m, j = w - r * m;
-        const int jr = nbk_bitrev(j, bits);
-        if (j < jr) {
-            cdouble* z = (cdouble*)&tile[(int64_t)r * sp];
-            const cdouble a = z[j];
-            z[j] = z[jr];
-            z[jr] = a;
-        }
-    }
-    __syncthreads();
-
-    for (int len = 2; len <= m; len <<= 1) {
-        const int half = len >> 1;
-        const int tw = m / len;
-        for (int w = t; w < RG * (m >> 1); w += T) {
-            const int r = w / (m >> 1);
-            const int q = w - r * (m >> 1);
-            const int grp = q / half;
-            const int pos = q - grp * half;
-            cdouble* z = (cdouble*)&tile[(int64_t)r * sp];
-            const int i0 = grp * len + pos;
-            const int i1 = i0 + half;
-            const cdouble wv = table[2 * pos * tw];
-            const cdouble u = z[i0];
-            const cdouble v = cmul(z[i1], wv);
-            z[i0] = cadd(u, v);
-            z[i1] = csub(u, v);
-        }
-        __syncthreads();
-    }
-
-    // untwiddle split: X[k] = E[k] + W_n2^k O[k], k = 0..m, written
-    // straight to the z half-spectrum
+    const int lane = t & 63;
+    const int wave = t >> 6;
+    const int nw = T >> 6;
     cdouble* out = (cdouble*)mesh + ((ix - x0) * n1 + r0) * (m + 1);
-    for (int w = t; w < RG * (m + 1); w += T) {
-        const int r = w / (m + 1), k = w - r * (m + 1);
-        const cdouble* z = (const cdouble*)&tile[(int64_t)r * sp];
-        const cdouble Zk = z[k == m ? 0 : k];
-        const cdouble Zm = z[(m - k) % m];
-        const cdouble E = cscale(cadd(Zk, cconj(Zm)), 0.5);
-        const cdouble D = csub(Zk, cconj(Zm));
-        const cdouble O = {0.5 * D.im, -0.5 * D.re};    // D * (-i/2)
-        const cdouble X = cadd(E, cmul(table[k], O));
-        out[(int64_t)r * (m + 1) + k] = cscale(X, scale);
+
+    for (int r = wave; r < RG; r += nw) {
+        cdouble* z = (cdouble*)&tile[(int64_t)r * sp];
+
+        for (int j = lane; j < m; j += 64) {
+            const int jr = nbk_bitrev(j, bits);
+            if (j < jr) {
+                const cdouble a = z[j];
+                z[j] = z[jr];
+                z[jr] = a;
+            }
+        }
+        __builtin_amdgcn_wave_barrier();
+
+        for (int len = 2; len <= m; len <<= 1) {
+            const int half = len >> 1;
+            const int tw = m / len;
+            for (int q = lane; q < (m >> 1); q += 64) {
+                const int grp = q / half;
+                const int pos = q - grp * half;
+                const int i0 = grp * len + pos;
+                const int i1 = i0 + half;
+                const cdouble wv = table[2 * pos * tw];
+                const cdouble u = z[i0];
+                const cdouble v = cmul(z[i1], wv);
+                z[i0] = cadd(u, v);
+                z[i1] = csub(u, v);
+            }
+            __builtin_amdgcn_wave_barrier();
+        }
+
+        // untwiddle split: X[k] = E[k] + W_n2^k O[k], k = 0..m, written
+        // straight to the z half-spectrum
+        for (int k = lane; k <= m; k += 64) {
+            const cdouble Zk = z[k == m ? 0 : k];
+            const cdouble Zm = z[(m - k) % m];
+            const cdouble E = cscale(cadd(Zk, cconj(Zm)), 0.5);
+            const cdouble D = csub(Zk, cconj(Zm));
+            const cdouble O = {0.5 * D.im, -0.5 * D.re};  // D * (-i/2)
+            const cdouble X = cadd(E, cmul(table[k], O));
+            out[(int64_t)r * (m + 1) + k] = cscale(X, scale);
+        }
+        __builtin_amdgcn_wave_barrier();
     }
 }
 
