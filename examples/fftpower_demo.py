@@ -3,6 +3,11 @@ The reference's periodic-box cookbook flow (nbodykit docs,
 cookbook/fftpower), unchanged except for the import line.
 Run on an MI355X: python examples/fftpower_demo.py
 """
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), '..'))
+
 import numpy
 
 from nbodykit_amd.lab import (LogNormalCatalog, LinearPower, FFTPower,

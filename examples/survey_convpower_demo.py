@@ -3,6 +3,11 @@ The reference's survey-data flow (nbodykit docs, cookbook/convpower):
 sky coordinates -> Cartesian positions -> FKP catalog -> multipoles.
 Run on an MI355X: python examples/survey_convpower_demo.py
 """
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), '..'))
+
 import numpy
 
 from nbodykit_amd.lab import (transform, ArrayCatalog, FKPCatalog,
