@@ -211,6 +211,53 @@ class CatalogSourceBase(object):
     def make_column(self, array):
         return numpy.asarray(array)
 
+    def copy(self):
+        """Shallow copy: every column of ``self`` referenced (no data
+        copied), attrs decoupled (reference :474-507).  The copy is a
+        plain CatalogSource holding the materialized column references
+        (the reference keeps the subclass; its hard columns are lazy
+        dask graphs — ours are arrays either way)."""
+        toret = object.__new__(CatalogSource)
+        toret._overrides = {}
+        toret.base = None
+        toret._hardcolumns = []
+        toret._defaultcolumns = list(self._defaultcolumns)
+        toret.comm = self.comm
+        toret._size = self.size
+        toret._csize = self.csize
+        for col in self.columns:
+            if col in self._defaultcolumns and col not in self._overrides:
+                continue
+            toret._overrides[col] = self[col]
+        toret._attrs = dict(self.attrs)
+        return toret
+
+    def gslice(self, start, stop, end=1, redistribute=True):
+        """Global slice across ranks (reference :1013-1077); with
+        ``redistribute`` the selected rows are re-balanced evenly."""
+        counts = self.comm.allgather(self.size)
+        offset = int(numpy.sum(counts[:self.comm.rank], dtype='i8'))
+        index = numpy.zeros(self.csize, dtype=bool)
+        index[slice(start, stop, end)] = True
+        sub = self[index[offset:offset + self.size]]
+        if redistribute and self.comm.size > 1:
+            rank, ws = self.comm.rank, self.comm.size
+            n_tot = sub.csize
+            lo = n_tot * rank // ws
+            hi = n_tot * (rank + 1) // ws
+            data = {}
+            for col in sub.columns:
+                if col in sub._defaultcolumns                         and col not in sub._overrides:
+                    continue
+                full = numpy.concatenate(
+                    self.comm.allgather(numpy.asarray(sub[col])), axis=0)
+                data[col] = full[lo:hi]
+            from nbodykit_amd.source.catalog.array import ArrayCatalog
+            out = ArrayCatalog(data, comm=self.comm) if data else sub
+            out.attrs.update(sub.attrs)
+            return out
+        return sub
+
     def save(self, output, columns=None, dataset=None, datasets=None,
              header='Header', compute=True):
         """Save the catalog to a bigfile directory (reference

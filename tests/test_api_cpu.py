@@ -186,3 +186,28 @@ def test_compensation_action_wiring():
             (mode, func, kind), = get_compensation(interlaced, res)
             assert mode == 'complex' and kind == 'circular'
             assert lookup_compensation(func) == (res, interlaced)
+
+
+def test_catalog_copy():
+    from nbodykit_amd.lab import UniformCatalog
+    import numpy
+    cat = UniformCatalog(nbar=1e-3, BoxSize=64., seed=1)
+    cp = cat.copy()
+    numpy.testing.assert_array_equal(numpy.asarray(cp['Position']),
+                                     numpy.asarray(cat['Position']))
+    assert cp.size == cat.size and cp.csize == cat.csize
+    # attrs decoupled
+    cp.attrs['extra'] = 1
+    assert 'extra' not in cat.attrs
+    # default columns still served
+    assert bool(numpy.all(numpy.asarray(cp['Selection'])))
+
+
+def test_catalog_gslice():
+    from nbodykit_amd.lab import ArrayCatalog
+    import numpy
+    cat = ArrayCatalog({'Mass': numpy.arange(20.)})
+    sl = cat.gslice(5, 15, 2)
+    numpy.testing.assert_array_equal(numpy.asarray(sl['Mass']),
+                                     numpy.arange(5., 15., 2.))
+    assert sl.csize == 5
