@@ -33,7 +33,7 @@ _global_options = {
     # paint locality-sort thresholds (tests shrink these to exercise the
     # two-level atomic-free sort + gather paint on small inputs)
     'sort_min_n': 1 << 21,
-    'sort_two_level_min_n': 1 << 24,
+    'sort_two_level_min_n': 1 << 22,
     'sort_two_level_min_cells': 1 << 23,
 }
 
