@@ -59,8 +59,10 @@ def gather_bpp(resampler, nmesh, n_particles):
 def paint_is_gather(nmesh, n_local):
     """Mirrors the driver's two-level-sort thresholds
     (source/mesh/catalog.py _prepare_particles)."""
-    return nmesh ** 3 > (1 << 23) and n_local >= (1 << 24) \
-        and nmesh <= 20480
+    from nbodykit_amd import _global_options as go
+    return (nmesh ** 3 > go['sort_two_level_min_cells']
+            and n_local >= go['sort_two_level_min_n']
+            and nmesh <= 20480)
 
 WORKLOADS = {
     # BASELINE.json configs (C1 is the CPU-oracle plumbing config)
