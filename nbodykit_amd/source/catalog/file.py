@@ -4,6 +4,7 @@ source/catalog/file.py factories): the io layer and the catalog are one
 class here since columns are numpy-backed; rows are split evenly across
 ranks like the reference's FileCatalogBase (file.py:67-90).
 """
+import glob as _glob
 import logging
 import os
 
@@ -18,6 +19,23 @@ def _partition(csize, comm):
     return start, end
 
 
+def _expand_paths(path):
+    """A path, a glob pattern, or a list of either -> sorted file list
+    (the reference's FileStack, io/stack.py: multiple physical files
+    form one logical catalog)."""
+    if isinstance(path, (list, tuple)):
+        out = []
+        for p in path:
+            out.extend(_expand_paths(p))
+        return out
+    if any(ch in path for ch in '*?['):
+        hits = sorted(_glob.glob(path))
+        if not hits:
+            raise FileNotFoundError("no files match %r" % path)
+        return hits
+    return [path]
+
+
 class BinaryCatalog(CatalogSource):
     """Catalog from a COLUMN-MAJOR binary file (reference
     io/binary.py:33-144: each column stored contiguously, optional
@@ -29,7 +47,23 @@ class BinaryCatalog(CatalogSource):
 
     def __init__(self, path, dtype, offsets=None, header_size=0,
                  size=None, comm=None, attrs=None):
-        self.path = path
+        paths = _expand_paths(path)
+        if len(paths) > 1:
+            if offsets is not None or size is not None:
+                raise ValueError("offsets/size only apply to a single "
+                                 "file")
+            parts = [BinaryCatalog(p, dtype, header_size=header_size,
+                                   comm=comm, attrs=attrs)
+                     for p in paths]
+            from nbodykit_amd.transform import ConcatenateSources
+            stacked = ConcatenateSources(*parts)
+            self.path = paths
+            self._size = stacked.size
+            CatalogSource.__init__(self, comm=stacked.comm)
+            self._overrides.update(stacked._overrides)
+            self._attrs = dict(stacked.attrs)
+            return
+        self.path = path = paths[0]
         dtype = numpy.dtype(dtype)
         if dtype.names is None:
             raise ValueError("input dtype should be structured (a list "
@@ -94,7 +128,21 @@ class CSVCatalog(CatalogSource):
                  usecols=None, delim_whitespace=True, comm=None,
                  attrs=None, **config):
         import pandas as pd
-        self.path = path
+        paths = _expand_paths(path)
+        if len(paths) > 1:
+            parts = [CSVCatalog(p, names, dtype=dtype, usecols=usecols,
+                                delim_whitespace=delim_whitespace,
+                                comm=comm, attrs=attrs, **config)
+                     for p in paths]
+            from nbodykit_amd.transform import ConcatenateSources
+            stacked = ConcatenateSources(*parts)
+            self.path = paths
+            self._size = stacked.size
+            CatalogSource.__init__(self, comm=stacked.comm)
+            self._overrides.update(stacked._overrides)
+            self._attrs = dict(stacked.attrs)
+            return
+        self.path = path = paths[0]
 
         if isinstance(dtype, numpy.dtype) or numpy.isscalar(dtype) \
                 or isinstance(dtype, (str, type)):

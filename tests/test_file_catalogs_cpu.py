@@ -69,3 +69,31 @@ def test_csv_catalog_to_mesh(tmp_path):
                                              cat['z'])
     mesh = cat.to_mesh(Nmesh=16, BoxSize=32.)
     numpy.testing.assert_array_equal(mesh.attrs['Nmesh'], 16)
+
+
+def test_csv_catalog_glob(tmp_path):
+    rng = numpy.random.RandomState(5)
+    chunks = []
+    for i in range(3):
+        d = rng.random_sample((10, 2))
+        numpy.savetxt(str(tmp_path / ('part%d.txt' % i)), d)
+        chunks.append(d)
+    cat = CSVCatalog(str(tmp_path / 'part*.txt'), ['x', 'y'])
+    assert cat.size == 30
+    nt.assert_allclose(numpy.asarray(cat['x']),
+                       numpy.concatenate(chunks)[:, 0])
+
+
+def test_binary_catalog_multifile(tmp_path):
+    fns = []
+    parts = []
+    for i in range(2):
+        fn = str(tmp_path / ('b%d.bin' % i))
+        d = numpy.arange(8, dtype='f8') + 100 * i
+        d.tofile(fn)
+        fns.append(fn)
+        parts.append(d)
+    cat = BinaryCatalog(fns, [('Mass', 'f8')])
+    assert cat.size == 16
+    nt.assert_array_equal(numpy.asarray(cat['Mass']),
+                          numpy.concatenate(parts))
