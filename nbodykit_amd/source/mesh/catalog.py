@@ -570,28 +570,27 @@ class CatalogMesh(MeshSource):
                     stream), 'nbk_paint_gather_fft_f64')
             return _r2c_finish(z, pm, stream)
 
-        if True:
-            cplx = one(0.0)
-            if interlaced:
-                c2t = one(0.5)
-                c1f = ComplexField(pm, tensor=cplx)
-                c2f = ComplexField(pm, tensor=c2t)
-                hiplib.check(lib.nbk_interlace_combine_f64(
-                    hiplib.dptr(c1f.value), hiplib.dptr(c2f.value),
-                    nmesh, box, hiplib.i64_arr(c1f.dims),
-                    hiplib.i64_arr(c1f.off), None, stream),
-                    'nbk_interlace_combine_f64')
-                cplx = c1f.value
-                # the real path (and the reference, which combines in k
-                # then goes through c2r + r2c) Hermitian-projects the
-                # self-conjugate z-planes: numpy's irfft drops the
-                # imaginary parts of the kz = 0/Nyquist bins, which in k
-                # is c <- (c + conj(c(-k)))/2 on those planes
-                for kz in (0, n2 // 2):
-                    A = cplx[:, :, kz]
-                    B = torch.conj(torch.roll(torch.flip(A, (0, 1)),
-                                              (1, 1), (0, 1)))
-                    cplx[:, :, kz] = 0.5 * (A + B)
+        cplx = one(0.0)
+        if interlaced:
+            c2t = one(0.5)
+            c1f = ComplexField(pm, tensor=cplx)
+            c2f = ComplexField(pm, tensor=c2t)
+            hiplib.check(lib.nbk_interlace_combine_f64(
+                hiplib.dptr(c1f.value), hiplib.dptr(c2f.value),
+                nmesh, box, hiplib.i64_arr(c1f.dims),
+                hiplib.i64_arr(c1f.off), None, stream),
+                'nbk_interlace_combine_f64')
+            cplx = c1f.value
+            # the real path (and the reference, which combines in k
+            # then goes through c2r + r2c) Hermitian-projects the
+            # self-conjugate z-planes: numpy's irfft drops the
+            # imaginary parts of the kz = 0/Nyquist bins, which in k
+            # is c <- (c + conj(c(-k)))/2 on those planes
+            for kz in (0, n2 // 2):
+                A = cplx[:, :, kz]
+                B = torch.conj(torch.roll(torch.flip(A, (0, 1)),
+                                          (1, 1), (0, 1)))
+                cplx[:, :, kz] = 0.5 * (A + B)
 
         with numpy.errstate(divide='ignore', invalid='ignore'):
             shotnoise = float(numpy.prod(pm.BoxSize)) * W2 / W ** 2
