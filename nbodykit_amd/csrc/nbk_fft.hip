@@ -177,65 +177,35 @@ __global__ void kfft_c2r_z(const double* __restrict__ cplx,
 
 // one (outer line, TI-column tile) per block; element j of column
 // (o, i) at cplx[o*ostride + j*stride + i]
-// software-pipelined: each block walks TPB adjacent column tiles and
-// prefetches tile t+1 into registers while tile t's butterflies run in
-// LDS — the pass is global-latency-bound (strided lines), and the
-// in-flight loads ride out the FFT stages (s_barrier does not wait on
-// vmcnt).  Per-thread register footprint: nfft*TI/blockDim <= 4
-// cdouble.
-#define NBK_FFT_TPB 4
-
 template <bool INV>
 __global__ void kfft_c_strided(double* __restrict__ data,
                                int nfft, int64_t stride,
                                int64_t ostride, int64_t n_inner,
-                               int TI, int tiles, int ntb,
+                               int TI, int tiles,
                                const cdouble* __restrict__ table /* W_2nfft */)
 {
     extern __shared__ cdouble buf[];          // nfft * TI
     const int bits = 31 - __clz((unsigned)nfft);
-    const int64_t o = blockIdx.x / ntb;
-    const int tb0 = (int)(blockIdx.x % ntb) * NBK_FFT_TPB;
-    const int my_tiles = min(NBK_FFT_TPB, tiles - tb0);
-    const int n_el = nfft * TI;
-    const int T = blockDim.x;
+    const int64_t o = blockIdx.x / tiles;
+    const int64_t c0 = (int64_t)(blockIdx.x % tiles) * TI;
+    cdouble* g = (cdouble*)data + o * ostride + c0;
+    const int ncol = (int)min((int64_t)TI, n_inner - c0);
 
-    cdouble r[4];
-    auto load_tile = [&](int tb) {
-        const int64_t c0 = (int64_t)tb * TI;
-        cdouble* g = (cdouble*)data + o * ostride + c0;
-        const int ncol = (int)min((int64_t)TI, n_inner - c0);
-        int k = 0;
-        for (int w = threadIdx.x; w < n_el; w += T, k++) {
-            const int c = w % TI;
-            r[k] = (c < ncol)
-                ? g[(int64_t)(w / TI) * stride + c]
-                : cdouble{0.0, 0.0};
-        }
-    };
+    for (int w = threadIdx.x; w < nfft * TI; w += blockDim.x) {
+        const int c = w % TI;
+        const int j = w / TI;
+        if (c < ncol)
+            buf[bitrev(j, bits) * TI + c] = g[(int64_t)j * stride + c];
+    }
+    __syncthreads();
 
-    load_tile(tb0);
-    for (int t = 0; t < my_tiles; t++) {
-        __syncthreads();   // prior store phase is done reading buf
-        {
-            int k = 0;
-            for (int w = threadIdx.x; w < n_el; w += T, k++)
-                buf[bitrev(w / TI, bits) * TI + (w % TI)] = r[k];
-        }
-        __syncthreads();
-        if (t + 1 < my_tiles)
-            load_tile(tb0 + t + 1);           // overlaps the FFT below
+    lds_fft<INV>(buf, nfft, bits, TI, table);
 
-        lds_fft<INV>(buf, nfft, bits, TI, table);
-
-        const int64_t c0 = (int64_t)(tb0 + t) * TI;
-        cdouble* g = (cdouble*)data + o * ostride + c0;
-        const int ncol = (int)min((int64_t)TI, n_inner - c0);
-        for (int w = threadIdx.x; w < n_el; w += T) {
-            const int c = w % TI;
-            if (c < ncol)
-                g[(int64_t)(w / TI) * stride + c] = buf[(w / TI) * TI + c];
-        }
+    for (int w = threadIdx.x; w < nfft * TI; w += blockDim.x) {
+        const int c = w % TI;
+        const int j = w / TI;
+        if (c < ncol)
+            g[(int64_t)j * stride + c] = buf[j * TI + c];
     }
 }
 
@@ -305,10 +275,7 @@ extern "C" int nbk_fft_c_strided(double* cplx, int64_t nfft, int64_t stride,
         TI >>= 1;
     if (TI > n_inner) TI = (int)n_inner;
     const int tiles = (int)((n_inner + TI - 1) / TI);
-    const int ntb = (tiles + NBK_FFT_TPB - 1) / NBK_FFT_TPB;
-    const int64_t grid = n_outer * ntb;
-    // the prefetch registers hold nfft*TI/block elements (max 4)
-    const int min_block = ((int)(nfft * TI + 3) / 4 + 63) & ~63;
+    const int64_t grid = n_outer * tiles;
     if (grid > 0x7fffffff) {
         NBK_SET_ERR("nbk_fft_c_strided: grid too large");
         return NBK_ERR_ARG;
@@ -322,7 +289,6 @@ extern "C" int nbk_fft_c_strided(double* cplx, int64_t nfft, int64_t stride,
         block = e ? atoi(e) : 1024;
         if (block < 64 || block > 1024) block = 1024;
     }
-    if (block < min_block) block = min_block;
     const size_t shmem = (size_t)nfft * TI * sizeof(cdouble);
     if (shmem > 65536) {
         static size_t raised_fwd = 0, raised_inv = 0;
@@ -341,12 +307,12 @@ extern "C" int nbk_fft_c_strided(double* cplx, int64_t nfft, int64_t stride,
         hipLaunchKernelGGL(kfft_c_strided<false>, dim3((uint32_t)grid),
                            dim3(block), shmem, (hipStream_t)stream,
                            cplx, (int)nfft, stride, outer_stride, n_inner,
-                           TI, tiles, ntb, (const cdouble*)table);
+                           TI, tiles, (const cdouble*)table);
     else
         hipLaunchKernelGGL(kfft_c_strided<true>, dim3((uint32_t)grid),
                            dim3(block), shmem, (hipStream_t)stream,
                            cplx, (int)nfft, stride, outer_stride, n_inner,
-                           TI, tiles, ntb, (const cdouble*)table);
+                           TI, tiles, (const cdouble*)table);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }
