@@ -3,3 +3,4 @@ from .fftpower import (FFTPower, FFTBase, ProjectedFFTPower,
 from .fftcorr import FFTCorr
 from .fftrecon import FFTRecon
 from .convpower import ConvolvedFFTPower, FKPCatalog, FKPWeightFromNbar
+from .zhist import RedshiftHistogram

@@ -26,3 +26,4 @@ from nbodykit_amd import transform
 from nbodykit_amd.algorithms.convpower import (ConvolvedFFTPower,
                                                FKPCatalog,
                                                FKPWeightFromNbar)
+from nbodykit_amd.algorithms.zhist import RedshiftHistogram
