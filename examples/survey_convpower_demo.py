@@ -11,7 +11,7 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), '..'))
 import numpy
 
 from nbodykit_amd.lab import (transform, ArrayCatalog, FKPCatalog,
-                              ConvolvedFFTPower)
+                              ConvolvedFFTPower, RedshiftHistogram)
 from nbodykit_amd.cosmology import Planck15
 
 rng = numpy.random.RandomState(42)
@@ -27,6 +27,18 @@ def make_survey(n):
 
 data = make_survey(50000)
 randoms = make_survey(500000)
+
+# measure n(z) from the randoms and refresh the NZ columns, like the
+# reference's convpower cookbook (fsky of the 30x30 deg patch)
+fsky = (numpy.deg2rad(30.) * (numpy.sin(numpy.deg2rad(25.))
+                              - numpy.sin(numpy.deg2rad(-5.)))) \
+    / (4 * numpy.pi)
+randoms['Redshift'] = rng.uniform(0.4, 0.7, randoms.size)
+data['Redshift'] = rng.uniform(0.4, 0.7, data.size)
+nz = RedshiftHistogram(randoms, fsky, Planck15, redshift='Redshift')
+alpha = 1.0 * data.csize / randoms.csize
+randoms['NZ'] = nz.interpolate(numpy.asarray(randoms['Redshift'])) * alpha
+data['NZ'] = nz.interpolate(numpy.asarray(data['Redshift'])) * alpha
 
 fkp = FKPCatalog(data, randoms, P0=1e4)
 mesh = fkp.to_mesh(Nmesh=128, dtype='f8', compensated=True)
