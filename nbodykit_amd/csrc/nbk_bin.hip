@@ -8,9 +8,14 @@
 // weight (2 where the compressed-axis frequency > 0, meshtools:188-215)
 // and Herm applies the conjugate-pair parity rule (fftpower.py:649-656).
 //
-// Two variants: LDS per-workgroup histograms flushed once (when the bin
-// grid fits in 64 KiB), else direct global f64 atomics.  The pass reads
-// 16 B/cell — far below the paint/FFT traffic — so either is cheap.
+// Fused mode (nbk_power_bin_f64) computes comp1(c1) conj(comp2(c2)) V
+// per element on the fly — compensation, cross power and binning in
+// one pass, p3d never materialized — and skips the ~48% of modes
+// beyond the last k-edge before touching field data.  Line-grouped
+// block mapping keeps the expensive index math per line, with the
+// digitize edges staged through LDS.  Histograms accumulate in LDS
+// (up to 160 KiB via hipFuncSetAttribute) and flush once; the
+// global-atomic fallback serves grids beyond that.
 #include "nbk_common.h"
 
 namespace {

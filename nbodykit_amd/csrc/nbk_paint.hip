@@ -1,22 +1,26 @@
-// Window-deposit ("paint") scatter kernels — the #1 kernel of the path
-// (replaces pmesh's Cython scatter called at
-// nbodykit/source/mesh/catalog.py:287,295-296).
+// Window-deposit ("paint") kernels — the #1 op of the path (replaces
+// pmesh's Cython scatter called at
+// nbodykit/source/mesh/catalog.py:287,295-296).  Two strategies:
 //
-// HBM-bound: algorithmic traffic 24 B position read + support^3 f64
-// read-modify-writes per particle (CIC 152 B, TSC 456 B, PCS 1048 B).
-// One thread per particle in wave-contiguous order; SoA position reads
-// are fully coalesced; deposits use hardware f64 global atomics
-// (-munsafe-fp-atomics => global_atomic_add_f64).
+// 1. kpaint_gather (+optional fused z-FFT): the BIG-MESH path, fed by
+//    the atomic-free two-level locality sort (nbk_sort.hip).  Each
+//    block owns an exclusive LDS mesh tile (RG y-rows x full z of one
+//    x-plane), gathers the source rows whose stencils can reach it via
+//    the sort's row table, deposits with LDS ds_add_f64 and flushes
+//    with plain stores — ZERO global atomics (the global atomic pipe
+//    measures ~25 G op/s regardless of locality, csrc/count_probe.hip,
+//    which bounded the scatter kernel below at 47 ms for 1e9 CIC).
+//    The DOFFT variant runs the forward z-FFT on the tile's rows
+//    before flushing, so the real mesh never exists in HBM
+//    (CatalogMesh.to_complex_field).
 //
-// Clump handling: catalogs on this path are typically cell-ordered (the
-// LogNormal generator emits particles in global cell order, mirroring
-// the reference's mpsort-by-cell-id, mockmaker.py:338-345), and
-// clustered fields put hundreds of particles in one cell.  Plain
-// per-lane atomics serialize on those shared addresses (measured 3.2x
-// slowdown on a clumpy 1e9/1024^3 probe).  Each deposit therefore runs
-// a wave-level segmented merge first: adjacent lanes holding the same
-// target address combine via a shfl prefix-sum and only the run tail
-// issues the atomic (probe: clumpy 167 -> 45 ms, uniform unchanged).
+// 2. kpaint: the scatter fallback for small chunks/meshes.  One thread
+//    per particle in wave-contiguous order; deposits use hardware f64
+//    global atomics (-munsafe-fp-atomics => global_atomic_add_f64)
+//    with a wave-level segmented merge: adjacent lanes holding the
+//    same target address combine via a shfl prefix-sum and only the
+//    run tail issues the atomic (probe: clumpy 167 -> 45 ms).
+//
 // Window shapes are the B-splines fixed in-tree by their Fourier duals
 // (source/mesh/catalog.py:453-594; Jing 2005 eq. 18, p = 2/3/4);
 // a particle exactly on a grid point deposits its full mass there.

@@ -1,13 +1,22 @@
-// Counting bucket-sort of particles by mesh cell — the
-// paint locality pass.  The deposit kernel (nbk_paint.hip) is ~4x
-// faster when nearby-in-space particles are nearby-in-memory; a full
-// radix sort (rocprim via torch.argsort) costs ~70 ms at 1e9 particles
-// where this two-pass counting sort needs only bucket-local order.
-// Non-stable within a bucket (ticket = atomic fetch-add), which is fine:
-// all particles of one cell still land in one contiguous output range,
-// so the deposit kernel's wave-merge sees clumps as contiguous runs.
-// Output is SoA (x[n] y[n] z[n]) — exactly the layout nbk_paint_f64
-// reads — so the driver also saves its AoS->SoA transpose pass.
+// Particle locality sorts for the paint kernels.
+//
+// Two pipelines:
+// - nbk_xsort_count/scatter + nbk_bucket_fine (big meshes): a
+//   two-level ATOMIC-FREE counting sort.  The global atomic pipe runs
+//   at ~25 G op/s regardless of locality (csrc/count_probe.hip), so no
+//   stage may use global atomics: the coarse sort by
+//   (ix, iy-group) goes through per-chunk count MATRICES (host-scanned,
+//   no shared ticket counters) + LDS cursors, and the fine pass runs
+//   count + block scan + placement entirely in LDS, one block per
+//   coarse bucket.  Row-only fine mode serves the ownership-gather
+//   paint (which needs row grouping, not full cell order) and emits
+//   the row table it reads.
+// - nbk_bucket_count/scatter (small inputs): the single-level counting
+//   sort by full cell id with int32 countdown tickets, paired with the
+//   wave-merged scatter paint.
+//
+// Both are non-stable within a bucket/cell (fine: deposits commute)
+// and emit SoA (x[n] y[n] z[n]) — the layout the paint kernels read.
 #include "nbk_common.h"
 
 namespace {
