@@ -17,7 +17,8 @@ from nbodykit_amd.source.mesh import (CatalogMesh, FieldMesh,
                                       ArrayMesh, LinearMesh)
 from nbodykit_amd.source.mesh.bigfile import BigFileMesh
 from nbodykit_amd.source.catalog.bigfile import BigFileCatalog
-from nbodykit_amd.source.catalog.file import BinaryCatalog, CSVCatalog
+from nbodykit_amd.source.catalog.file import (BinaryCatalog,
+    CSVCatalog, Gadget1Catalog)
 from nbodykit_amd.base.catalog import CatalogSource
 from nbodykit_amd.base.mesh import MeshSource
 from nbodykit_amd.binned_statistic import BinnedStatistic
