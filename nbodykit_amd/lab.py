@@ -28,3 +28,4 @@ from nbodykit_amd.algorithms.convpower import (ConvolvedFFTPower,
                                                FKPCatalog,
                                                FKPWeightFromNbar)
 from nbodykit_amd.algorithms.zhist import RedshiftHistogram
+from nbodykit_amd import filters

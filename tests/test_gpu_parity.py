@@ -798,3 +798,24 @@ def test_fused_complex_paint_through_fftpower():
     assert_array_equal(r_fused.power['modes'], r_plain.power['modes'])
     assert_allclose(r_fused.attrs['shotnoise'],
                     r_plain.attrs['shotnoise'], rtol=1e-12)
+
+
+def test_mesh_filters():
+    """TopHat/Gaussian MeshFilters through apply() (reference
+    filters.py + tests/test_filters.py): the Gaussian-filtered power is
+    P(k) exp(-k^2 r^2)."""
+    from nbodykit_amd.lab import LinearMesh, FFTPower, filters
+    P0 = 100.0
+    mesh = LinearMesh(lambda k: P0 * numpy.ones_like(k), BoxSize=256.,
+                      Nmesh=64, seed=9, unitary_amplitude=True)
+    sm = mesh.apply(filters.Gaussian(8.0))
+    r = FFTPower(sm, mode='1d')
+    k = r.power['k']
+    p = r.power['power'].real
+    good = numpy.isfinite(p) & (k > 0) & (r.power['modes'] > 8)
+    expect = P0 * numpy.exp(-(k[good] * 8.0) ** 2)
+    # bin-averaged exp(-k^2r^2) vs the value at the mean k: few-percent
+    assert_allclose(p[good], expect, rtol=0.1)
+    # TopHat runs and leaves the k=0 normalization intact
+    th = mesh.apply(filters.TopHat(8.0)).compute(mode='real')
+    assert abs(th.value.mean().item() - 1.0) < 1e-10
