@@ -78,3 +78,29 @@ def test_concatenate_sources():
                           numpy.concatenate([numpy.arange(5.),
                                              numpy.arange(3.) + 10]))
     nt.assert_array_equal(numpy.asarray(cat['Position'])[:5], 1.0)
+
+
+def test_cartesian_sky_roundtrip_with_observer():
+    # the reference's test_cartesian_to_sky (tests/test_transform.py:59)
+    rng = numpy.random.RandomState(42)
+    pos = rng.uniform(0, 1., size=(300, 3))
+    obs = [0.5, 0.5, 0.5]
+    ra, dec, z = transform.CartesianToSky(pos, Planck15, observer=obs)
+    pos2 = transform.SkyToCartesian(ra, dec, z, Planck15, observer=obs)
+    nt.assert_allclose(pos, pos2, rtol=1e-5, atol=1e-7)
+
+
+def test_cartesian_to_sky_zmax_too_small():
+    # out-of-range distances raise (reference :116-120)
+    pos = numpy.array([[20000., 0., 0.]])
+    with pytest.raises(ValueError):
+        transform.CartesianToSky(pos, Planck15, zmax=0.5)
+
+
+def test_cartesian_to_equatorial_bounds():
+    rng = numpy.random.RandomState(7)
+    pos = rng.uniform(0, 1., size=(500, 3))
+    ra, dec = transform.CartesianToEquatorial(pos,
+                                              observer=[0.5, 0.5, 0.5])
+    assert ((ra >= 0.) & (ra < 360.)).all()
+    assert ((dec >= -90.) & (dec <= 90.)).all()
