@@ -102,15 +102,25 @@ class Cosmology(object):
         """Line-of-sight comoving distance in Mpc/h:
         Dc = (c/H0) int_0^z dz'/E(z') (the reference wraps CLASS's
         background; flat matter+Lambda integral here — the neglected
-        radiation term shifts Dc by <0.1% at survey redshifts)."""
+        radiation term shifts Dc by <0.1% at survey redshifts).  The
+        integral is tabulated once on a fixed log grid so repeated
+        calls with different ranges agree to interpolation accuracy
+        (transform.CartesianToSky inverts one call against another)."""
         z = numpy.asarray(z, dtype='f8')
         zmax = float(z.max()) if z.size else 0.0
-        grid = numpy.linspace(0.0, max(zmax, 1e-8), 4096)
-        integrand = 1.0 / self.efunc(grid)
-        dc = numpy.concatenate([[0.0], numpy.cumsum(
-            0.5 * (integrand[1:] + integrand[:-1]) * numpy.diff(grid))])
+        cached = getattr(self, '_dc_table', None)
+        if cached is None or cached[0] < zmax:
+            ztop = max(10.0, 2.0 * zmax)
+            grid = numpy.concatenate(
+                [[0.0], numpy.logspace(-8, numpy.log10(ztop), 8192)])
+            integrand = 1.0 / self.efunc(grid)
+            dc = numpy.concatenate([[0.0], numpy.cumsum(
+                0.5 * (integrand[1:] + integrand[:-1])
+                * numpy.diff(grid))])
+            self._dc_table = (ztop, grid, dc)
+            cached = self._dc_table
         # c / H0 with H0 = 100 h km/s/Mpc -> Mpc/h units
-        return 2997.92458 * numpy.interp(z, grid, dc)
+        return 2997.92458 * numpy.interp(z, cached[1], cached[2])
 
 
 # Planck15 parameters (astropy's FlatLambdaCDM Planck15 + the sigma8/n_s
