@@ -211,3 +211,77 @@ def test_catalog_gslice():
     numpy.testing.assert_array_equal(numpy.asarray(sl['Mass']),
                                      numpy.arange(5., 15., 2.))
     assert sl.csize == 5
+
+
+# ---- reference base/tests/test_catalog.py semantics ---------------------
+
+def test_ref_slice_semantics():
+    """reference test_slice (:230-250) + test_getitem KeyErrors."""
+    import numpy
+    from numpy.testing import assert_array_equal
+    from nbodykit_amd.lab import UniformCatalog
+    source = UniformCatalog(nbar=2e-4, BoxSize=512., seed=42)
+    source['NZ'] = 1
+
+    subset = source[:10]
+    assert all(col in subset for col in source.columns)
+    assert len(subset) == 10
+    assert_array_equal(numpy.asarray(subset['Position']),
+                       numpy.asarray(source['Position'])[:10])
+
+    subset = source[[0, 1, 2]]
+    assert_array_equal(numpy.asarray(subset['Position']),
+                       numpy.asarray(source['Position'])[[0, 1, 2]])
+
+    import pytest
+    with pytest.raises(KeyError):
+        source['BAD_COLUMN']
+
+
+def test_ref_delitem_semantics():
+    """reference test_delitem (:323-343)."""
+    import numpy
+    import pytest
+    from nbodykit_amd.lab import UniformCatalog
+    source = UniformCatalog(nbar=2e-4, BoxSize=512., seed=42)
+    source['test'] = numpy.ones(source.size)
+    with pytest.raises(ValueError):
+        del source['Position']
+    with pytest.raises(ValueError):
+        del source['BAD_COLUMN']
+    assert 'test' in source
+    del source['test']
+    assert 'test' not in source
+
+
+def test_ref_transform_arithmetic():
+    """reference test_transform (:285-306): self-referential column
+    reassignment resolves eagerly to the same values the reference's
+    lazy graphs produce."""
+    import numpy
+    from numpy.testing import assert_allclose
+    from nbodykit_amd.lab import ArrayCatalog
+    data = numpy.ones(100, dtype=[('Position', ('f4', 3)),
+                                  ('Velocity', ('f4', 3))])
+    source = ArrayCatalog(data, BoxSize=100, Nmesh=32)
+    source['Velocity'] = source['Position'] + source['Velocity']
+    source['Position'] = source['Position'] + source['Velocity']
+    assert_allclose(numpy.asarray(source['Position']), 3)
+    mesh = source.to_mesh()
+    numpy.testing.assert_array_equal(mesh.attrs['Nmesh'], 32)
+
+
+def test_ref_column_masked_by_selection():
+    """reference test_dask_slice (:255-268): slicing a column with a
+    boolean column."""
+    import numpy
+    from numpy.testing import assert_array_equal
+    from nbodykit_amd.lab import UniformCatalog
+    source = UniformCatalog(nbar=2e-4, BoxSize=512., seed=42)
+    index = numpy.random.RandomState(3).choice([True, False],
+                                               size=len(source))
+    source['Selection'] = index
+    pos = numpy.asarray(source['Position'])
+    pos2 = numpy.asarray(source['Position'])[
+        numpy.asarray(source['Selection'], dtype=bool)]
+    assert_array_equal(pos[index], pos2)
