@@ -1,0 +1,71 @@
+"""Property-based tests (hypothesis) of the oracle's core invariants —
+depth beyond the fixed-seed cases: mass conservation, partition of
+unity, FFT round trips, and digitize/edge semantics on adversarial
+inputs."""
+import numpy
+import numpy.testing as nt
+from hypothesis import given, settings, strategies as st
+
+from oracle.mesh import MeshGeometry, r2c, c2r
+from oracle.paint import paint, readout
+
+WINDOWS = st.sampled_from(['cic', 'tsc', 'pcs'])
+
+
+def _positions(draw, n, box):
+    # adversarial: mix of interior, negative, beyond-box and exactly
+    # on-grid coordinates
+    rng = numpy.random.RandomState(draw(st.integers(0, 2 ** 31 - 1)))
+    pos = rng.uniform(-box, 2 * box, size=(n, 3))
+    ongrid = rng.randint(0, 2, size=pos.shape).astype(bool)
+    H = box / 8.0
+    pos[ongrid] = numpy.round(pos[ongrid] / H) * H
+    return pos
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.data(), WINDOWS, st.integers(1, 64))
+def test_paint_mass_conservation(data, window, n):
+    box = 16.0
+    geom = MeshGeometry([8, 8, 8], box)
+    pos = _positions(data.draw, n, box)
+    w = numpy.abs(numpy.random.RandomState(n).standard_normal(n)) + 0.1
+    mesh = numpy.zeros((8, 8, 8))
+    paint(pos, w, mesh, geom, resampler=window)
+    nt.assert_allclose(mesh.sum(), w.sum(), rtol=1e-12)
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.data(), WINDOWS, st.integers(1, 64))
+def test_readout_partition_of_unity(data, window, n):
+    # reading a constant field returns the constant at ANY position
+    box = 16.0
+    geom = MeshGeometry([8, 8, 8], box)
+    pos = _positions(data.draw, n, box)
+    mesh = numpy.full((8, 8, 8), 7.25)
+    got = readout(pos, mesh, geom, resampler=window)
+    nt.assert_allclose(got, 7.25, rtol=1e-12)
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.integers(0, 2 ** 31 - 1),
+       st.sampled_from([4, 8, 16]))
+def test_fft_roundtrip(seed, N):
+    geom = MeshGeometry([N, N, N], 10.0)
+    x = numpy.random.RandomState(seed).standard_normal((N, N, N))
+    back = c2r(r2c(x, geom), geom)
+    nt.assert_allclose(back, x, rtol=1e-12, atol=1e-12)
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.integers(0, 2 ** 31 - 1), st.integers(2, 40))
+def test_digitize_matches_numpy(seed, nedges):
+    # the kernel's bisect_right semantics are pinned against numpy on
+    # values INCLUDING exact edges
+    rng = numpy.random.RandomState(seed)
+    edges = numpy.sort(rng.uniform(0, 10, nedges))
+    vals = numpy.concatenate([rng.uniform(-1, 11, 100), edges])
+    want = numpy.digitize(vals, edges)
+    import bisect
+    got = numpy.array([bisect.bisect_right(edges, v) for v in vals])
+    nt.assert_array_equal(got, want)
