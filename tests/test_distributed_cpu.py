@@ -275,3 +275,26 @@ def _body_fused_complex_gate(comm):
 def test_gloo_fused_complex_gate():
     r0, r1 = _run_world('_body_fused_complex_gate')
     assert r0 is True and r1 is True
+
+
+def _body_gslice(comm):
+    from nbodykit_amd.lab import ArrayCatalog
+    n = 10 + 5 * comm.rank
+    base = 100 * comm.rank
+    cat = ArrayCatalog({'Mass': numpy.arange(n, dtype='f8') + base},
+                       comm=comm)
+    out = cat.gslice(3, 20)
+    full = numpy.concatenate(comm.allgather(numpy.asarray(out['Mass'])))
+    return dict(sizes=comm.allgather(out.size), full=full)
+
+
+@pytest.mark.timeout(300)
+def test_gloo_gslice_redistribute():
+    r0, r1 = _run_world('_body_gslice')
+    # global rows: rank0 holds 0..9, rank1 holds 100..114; gslice(3, 20)
+    # selects global indices 3..19 -> values 3..9 then 100..109
+    expect = numpy.concatenate([numpy.arange(3., 10.),
+                                numpy.arange(100., 110.)])
+    numpy.testing.assert_array_equal(r0['full'], expect)
+    # redistributed evenly: 17 rows -> 8 + 9
+    assert r0['sizes'] == [8, 9]
