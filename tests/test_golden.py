@@ -53,3 +53,63 @@ def test_oracle_matches_golden(path):
 
 def test_golden_files_exist():
     assert len(GOLDEN) >= 5
+
+
+def test_oracle_reproduces_fftcorr_golden():
+    """Drift detector: the oracle still reproduces the committed
+    FFTCorr vector (the GPU parity test compares the product against
+    the live oracle)."""
+    import json
+    import os
+    import numpy
+    from numpy.testing import assert_allclose, assert_array_equal
+    from nbodykit_amd.utils import JSONDecoder
+    from oracle import fftcorr_oracle
+    from oracle.make_golden import uniform_positions
+
+    path = os.path.join(os.path.dirname(__file__), 'golden',
+                        'oracle_fftcorr_uniform_cic_1d.json')
+    with open(path) as ff:
+        want = json.load(ff, cls=JSONDecoder)
+    c = want['input']
+    pos = uniform_positions(c['nbar'], c['BoxSize'], seed=c['seed'])
+    got = fftcorr_oracle(pos, Nmesh=c['Nmesh'], BoxSize=c['BoxSize'],
+                         mode=c['mode'], resampler=c['resampler'],
+                         compensated=True)
+    assert_array_equal(got['modes'], want['modes'])
+    assert_allclose(got['corr'], want['corr'], rtol=1e-12,
+                    equal_nan=True)
+
+
+def test_oracle_reproduces_convpower_golden():
+    import json
+    import os
+    import numpy
+    from numpy.testing import assert_allclose, assert_array_equal
+    from nbodykit_amd.utils import JSONDecoder
+    from oracle.convpower import convpower_oracle
+
+    path = os.path.join(os.path.dirname(__file__), 'golden',
+                        'oracle_convpower_poles02.json')
+    with open(path) as ff:
+        want = json.load(ff, cls=JSONDecoder)
+    c = want['input']
+    rng = numpy.random.RandomState(c['seed'])
+    lo = numpy.array([900., 900., 900.])
+    span = numpy.array([200., 200., 200.])
+    dpos = lo + rng.uniform(0., 1., size=(c['nd'], 3)) * span
+    rpos = lo + rng.uniform(0., 1., size=(c['nr'], 3)) * span
+    nbar = numpy.full(c['nd'], c['nd'] / span.prod())
+    nbarr = numpy.full(c['nr'], c['nd'] / span.prod())
+    got = convpower_oracle(dpos, rpos, c['poles'], Nmesh=c['Nmesh'],
+                           BoxSize=c['BoxSize'],
+                           BoxCenter=c['BoxCenter'],
+                           nbar_data=nbar, nbar_ran=nbarr,
+                           compensated=True, dk=c['dk'])
+    assert_array_equal(got['modes'], want['modes'])
+    assert_allclose(got['attrs']['alpha'], want['alpha'], rtol=1e-12)
+    assert_allclose(got['attrs']['shotnoise'], want['shotnoise'],
+                    rtol=1e-12)
+    for ell in c['poles']:
+        assert_allclose(got['power_%d' % ell], want['power_%d' % ell],
+                        rtol=1e-6, atol=1e-8, equal_nan=True)
