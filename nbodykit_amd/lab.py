@@ -29,3 +29,4 @@ from nbodykit_amd.algorithms.convpower import (ConvolvedFFTPower,
                                                FKPWeightFromNbar)
 from nbodykit_amd.algorithms.zhist import RedshiftHistogram
 from nbodykit_amd import filters
+from nbodykit_amd.source.catalog.species import MultipleSpeciesCatalog

@@ -285,3 +285,29 @@ def test_ref_column_masked_by_selection():
     pos2 = numpy.asarray(source['Position'])[
         numpy.asarray(source['Selection'], dtype=bool)]
     assert_array_equal(pos[index], pos2)
+
+
+def test_lab_namespace_completeness():
+    """The lab namespace carries every reference symbol this build
+    supports (the drop-in checklist; reference nbodykit/lab.py)."""
+    from nbodykit_amd import lab
+    names = [
+        # algorithms
+        'FFTPower', 'ProjectedFFTPower', 'FFTCorr', 'FFTRecon',
+        'ConvolvedFFTPower', 'RedshiftHistogram',
+        # FKP
+        'FKPCatalog', 'FKPWeightFromNbar',
+        # catalogs
+        'UniformCatalog', 'RandomCatalog', 'LogNormalCatalog',
+        'ArrayCatalog', 'BigFileCatalog', 'CSVCatalog', 'BinaryCatalog',
+        'Gadget1Catalog', 'MultipleSpeciesCatalog',
+        # meshes
+        'CatalogMesh', 'FieldMesh', 'ArrayMesh', 'LinearMesh',
+        'BigFileMesh',
+        # cosmology + helpers
+        'LinearPower', 'Planck15', 'transform', 'filters',
+        'BinnedStatistic', 'setup_logging', 'set_options',
+        'CurrentMPIComm',
+    ]
+    missing = [n for n in names if not hasattr(lab, n)]
+    assert not missing, "lab namespace missing: %s" % missing
