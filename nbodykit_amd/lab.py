@@ -30,3 +30,4 @@ from nbodykit_amd.algorithms.convpower import (ConvolvedFFTPower,
 from nbodykit_amd.algorithms.zhist import RedshiftHistogram
 from nbodykit_amd import filters
 from nbodykit_amd.source.catalog.species import MultipleSpeciesCatalog
+from nbodykit_amd import io as IO

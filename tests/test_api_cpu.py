@@ -307,7 +307,7 @@ def test_lab_namespace_completeness():
         # cosmology + helpers
         'LinearPower', 'Planck15', 'transform', 'filters',
         'BinnedStatistic', 'setup_logging', 'set_options',
-        'CurrentMPIComm',
+        'CurrentMPIComm', 'cosmology', 'IO',
     ]
     missing = [n for n in names if not hasattr(lab, n)]
     assert not missing, "lab namespace missing: %s" % missing
