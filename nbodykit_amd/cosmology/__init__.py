@@ -122,11 +122,38 @@ class Cosmology(object):
         # c / H0 with H0 = 100 h km/s/Mpc -> Mpc/h units
         return 2997.92458 * numpy.interp(z, cached[1], cached[2])
 
+    def comoving_transverse_distance(self, z):
+        """Flat universe: equals the comoving distance (Mpc/h)."""
+        return self.comoving_distance(z)
 
-# Planck15 parameters (astropy's FlatLambdaCDM Planck15 + the sigma8/n_s
-# values nbodykit adds in cosmology/__init__.py:16-21).
+    def angular_diameter_distance(self, z):
+        """D_A = D_C / (1+z) for flat LCDM (Mpc/h)."""
+        z = numpy.asarray(z, dtype='f8')
+        return self.comoving_distance(z) / (1.0 + z)
+
+    def luminosity_distance(self, z):
+        """D_L = (1+z) D_C for flat LCDM (Mpc/h)."""
+        z = numpy.asarray(z, dtype='f8')
+        return self.comoving_distance(z) * (1.0 + z)
+
+
+# Named cosmologies: astropy's FlatLambdaCDM parameter sets + the
+# sigma8/n_s values nbodykit adds (reference cosmology/__init__.py:8-50;
+# the reference builds them via Cosmology.from_astropy over CLASS).
 Planck15 = Cosmology(h=0.6774, Omega0_m=0.3089, Omega0_b=0.0486,
                      n_s=0.9667, Tcmb0=2.7255, sigma8=0.8159)
+
+Planck13 = Cosmology(h=0.6777, Omega0_m=0.30712, Omega0_b=0.048252,
+                     n_s=0.9611, Tcmb0=2.7255, sigma8=0.8288)
+
+WMAP5 = Cosmology(h=0.702, Omega0_m=0.277, Omega0_b=0.0459,
+                  n_s=0.962, Tcmb0=2.725, sigma8=0.817)
+
+WMAP7 = Cosmology(h=0.704, Omega0_m=0.272, Omega0_b=0.0455,
+                  n_s=0.967, Tcmb0=2.725, sigma8=0.810)
+
+WMAP9 = Cosmology(h=0.6932, Omega0_m=0.2865, Omega0_b=0.04628,
+                  n_s=0.9608, Tcmb0=2.725, sigma8=0.820)
 
 
 class LinearPower(object):

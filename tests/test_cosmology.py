@@ -80,3 +80,21 @@ def test_cosmology_clone_and_dict():
     assert c2.sigma8 == 0.9 and Planck15.sigma8 == 0.8159
     d = dict(c2.pars)
     assert d['h'] == 0.6774
+
+
+def test_named_cosmologies_and_distances():
+    import numpy
+    from numpy.testing import assert_allclose
+    from nbodykit_amd.cosmology import (Planck13, Planck15, WMAP5,
+                                        WMAP7, WMAP9)
+    for c in (Planck13, Planck15, WMAP5, WMAP7, WMAP9):
+        assert 0.6 < c.h < 0.75
+        # LinearPower works for every named set
+        from nbodykit_amd.cosmology import LinearPower
+        P = LinearPower(c, redshift=0.0)
+        assert P(0.1) > 0
+    z = numpy.array([0.5, 1.0])
+    dc = Planck15.comoving_distance(z)
+    assert_allclose(Planck15.angular_diameter_distance(z), dc / (1 + z))
+    assert_allclose(Planck15.luminosity_distance(z), dc * (1 + z))
+    assert_allclose(Planck15.comoving_transverse_distance(z), dc)
