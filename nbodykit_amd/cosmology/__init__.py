@@ -137,6 +137,9 @@ class Cosmology(object):
         return self.comoving_distance(z) * (1.0 + z)
 
 
+from .correlation import (CorrelationFunction, pk_to_xi,   # noqa: E402
+                          xi_to_pk)
+
 # Named cosmologies: astropy's FlatLambdaCDM parameter sets + the
 # sigma8/n_s values nbodykit adds (reference cosmology/__init__.py:8-50;
 # the reference builds them via Cosmology.from_astropy over CLASS).
