@@ -139,6 +139,7 @@ class Cosmology(object):
 
 from .correlation import (CorrelationFunction, pk_to_xi,   # noqa: E402
                           xi_to_pk)
+from .zeldovich import ZeldovichPower                       # noqa: E402
 
 # Named cosmologies: astropy's FlatLambdaCDM parameter sets + the
 # sigma8/n_s values nbodykit adds (reference cosmology/__init__.py:8-50;
@@ -219,6 +220,18 @@ class LinearPower(object):
         """P(k) in (Mpc/h)^3 at ``self.redshift``; k in h/Mpc."""
         Pk = numpy.asarray(k) ** self.cosmo.n_s * self._transfer(k) ** 2
         return self._norm * Pk
+
+    def velocity_dispersion(self, kmin=1e-5, kmax=10., **kwargs):
+        r"""sigma_v in Mpc/h: sigma_v^2 = 1/(6 pi^2) \int dk P(k)
+        (reference linear.py:158-183)."""
+        from scipy.integrate import quad
+
+        def integrand(logq):
+            q = numpy.exp(logq)
+            return q * self(q)
+        sigmasq = quad(integrand, numpy.log(kmin), numpy.log(kmax),
+                       **kwargs)[0] / (6 * numpy.pi ** 2)
+        return sigmasq ** 0.5
 
     def sigma_r(self, r, kmin=1e-5, kmax=1e1):
         r"""

@@ -64,6 +64,35 @@ def _sph_bessel_fftlog(x, a, ell, prefac_exp):
     return y, G
 
 
+def _mellin_sph_bessel(x, F, ell, q):
+    """Biased FFTLog (mcfit's core): G(y) = int_0^inf F(x) j_ell(xy)
+    dx/x on the reflected log grid, with Mellin tilt ``q`` (F is
+    decomposed as x^q times a log-periodic part).  Kernel
+    M(s) = int t^{s-1} j_ell(t) dt
+         = sqrt(pi/2) 2^{s-3/2} Gamma((ell+s)/2) / Gamma((ell+3-s)/2).
+    Validated against direct quadrature in tests/test_correlation_cpu.py.
+    (Low-ringing phase rotation is not implemented; callers trim the
+    grid edges.)"""
+    N = len(x)
+    dln = (numpy.log(x[-1]) - numpy.log(x[0])) / (N - 1)
+    b = F * x ** (-q)
+    c = numpy.fft.rfft(b)
+    m = numpy.arange(len(c))
+    eta = 2 * numpy.pi * m / (N * dln)
+
+    s = q + 1j * eta
+    lnM = 0.5 * numpy.log(numpy.pi / 2) + (s - 1.5) * numpy.log(2.0) \
+        + loggamma((ell + s) / 2) - loggamma((ell + 3 - s) / 2)
+    M = numpy.exp(lnM)
+
+    y0 = 1.0 / x[-1]
+    y = y0 * numpy.exp(numpy.arange(N) * dln)
+    phase = (x[0] * y0) ** (-1j * eta)
+    g = numpy.fft.irfft(c * M * phase, N)
+    idx = (N - numpy.arange(N)) % N
+    return y, y ** (-q) * g[idx]
+
+
 def _resample_log(k, F, N=NUM_PTS):
     """Spline onto a log-even grid spanning the input range."""
     k = numpy.asarray(k, dtype='f8')

@@ -53,3 +53,68 @@ def test_correlation_function_bao():
     assert 95. < rpk < 115., rpk
     # positive at small separations, decaying
     assert CF(10.) > CF(50.) > 0
+
+
+def test_zeldovich_power():
+    """ZeldovichPower (reference cosmology/power/zeldovich.py;
+    normalizations quadrature-pinned): exact linear limit at low k,
+    smooth handoff at the low-k switch, monotone BAO damping, nmax
+    convergence."""
+    from nbodykit_amd.cosmology import ZeldovichPower
+    Pl = LinearPower(Planck15, redshift=0.0)
+    Pz = ZeldovichPower(Planck15, 0.0, nmax=16)
+
+    # low k: P_zel -> P_lin
+    k = numpy.array([1e-4, 1e-3])
+    nt.assert_allclose(Pz(k), Pl(k), rtol=1e-3)
+
+    # continuity across the k0_low switch (5e-3)
+    a, b = Pz(4.9e-3), Pz(5.1e-3)
+    assert abs(a / b - 1) < 0.05
+
+    # damping grows with k (ratio to linear decreases)
+    ks = numpy.array([0.02, 0.1, 0.3, 0.6])
+    ratio = Pz(ks) / Pl(ks)
+    assert (numpy.diff(ratio) < 0).all()
+    assert 0.3 < ratio[-1] < 1.0
+
+    # nmax converged by 16
+    Pz8 = ZeldovichPower(Planck15, 0.0, nmax=8)
+    nt.assert_allclose(Pz8(numpy.array([0.1, 0.3])),
+                       Pz(numpy.array([0.1, 0.3])), rtol=1e-3)
+
+    # redshift scaling: low-k follows the linear growth factor
+    Pz1 = ZeldovichPower(Planck15, 1.0, nmax=8)
+    Pl1 = LinearPower(Planck15, redshift=1.0)
+    nt.assert_allclose(Pz1(1e-3) / Pz(1e-3), Pl1(1e-3) / Pl(1e-3),
+                       rtol=1e-3)
+
+
+def test_zeldovich_quadrature_anchor():
+    # direct 2D quadrature anchor at k = 0.5 (converges there):
+    # class value within ~2%
+    from scipy.special import spherical_jn
+    from nbodykit_amd.cosmology import ZeldovichPower
+    Pl = LinearPower(Planck15, redshift=0.0)
+    Pz = ZeldovichPower(Planck15, 0.0, nmax=16)
+    kk = numpy.logspace(-5, 2, 8000)
+    Pk = Pl(kk)
+    sig2 = numpy.trapezoid(Pk, kk) / (6 * numpy.pi ** 2)
+    k = 0.5
+    qs = numpy.arange(0.25, 600., 0.5)
+    t = numpy.outer(qs, kk)
+    I0 = numpy.trapezoid(Pk * spherical_jn(0, t), kk, axis=1) \
+        / (2 * numpy.pi ** 2)
+    I1 = numpy.trapezoid(Pk * spherical_jn(1, t) / t, kk, axis=1) \
+        / (2 * numpy.pi ** 2)
+    X = -2 * I1 + 2 * sig2
+    Y = -2 * I0 + 6 * I1
+    mus = numpy.linspace(-1, 1, 1200)
+    P = 0.0
+    for i, q in enumerate(qs):
+        f = numpy.cos(k * q * mus) * (
+            numpy.exp(-0.5 * k * k * (X[i] + Y[i] * mus ** 2))
+            - numpy.exp(-k * k * sig2))
+        P += q * q * numpy.trapezoid(f, mus) * 0.5
+    P *= 2 * numpy.pi
+    nt.assert_allclose(Pz(k), P, rtol=0.03)
