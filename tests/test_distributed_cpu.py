@@ -28,10 +28,20 @@ def _worker(rank, world_size, port, fn_name, out_q):
         dist.destroy_process_group()
 
 
+def _free_port():
+    """Ask the kernel for a free TCP port (gloo rendezvous): random
+    picks occasionally collide with lingering test workers and hang
+    init_process_group until its timeout."""
+    import socket
+    with socket.socket() as s:
+        s.bind(('127.0.0.1', 0))
+        return s.getsockname()[1]
+
+
 def _run_world(fn_name, world=WORLD):
     ctx = mp.get_context('spawn')
     out_q = ctx.Queue()
-    port = numpy.random.randint(20000, 40000)
+    port = _free_port()
     procs = [ctx.Process(target=_worker,
                          args=(r, world, port, fn_name, out_q))
              for r in range(world)]
