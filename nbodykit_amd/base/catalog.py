@@ -313,6 +313,14 @@ class CatalogSourceBase(object):
 
     # -- default columns (reference :1166-1216) ---------------------------
     @column(is_default=True)
+    def Index(self):
+        """The global row index (reference :1178-1195: offset of the
+        lower ranks + local arange, dtype i8)."""
+        counts = self.comm.allgather(self.size)
+        offset = int(numpy.sum(counts[:self.comm.rank], dtype='i8'))
+        return offset + numpy.arange(self.size, dtype='i8')
+
+    @column(is_default=True)
     def Selection(self):
         return ConstantArray(True, self.size)
 

@@ -311,3 +311,17 @@ def test_lab_namespace_completeness():
     ]
     missing = [n for n in names if not hasattr(lab, n)]
     assert not missing, "lab namespace missing: %s" % missing
+
+
+def test_index_column():
+    """reference test_index (:270-283): Index spans range(csize), i8,
+    regenerated after gslice."""
+    import numpy
+    from nbodykit_amd.lab import UniformCatalog
+    source = UniformCatalog(nbar=2e-4, BoxSize=512., seed=42)
+    idx = numpy.asarray(source['Index'])
+    numpy.testing.assert_array_equal(idx, numpy.arange(source.csize))
+    assert idx.dtype == numpy.dtype('i8')
+    sub = source.gslice(0, 100)
+    numpy.testing.assert_array_equal(numpy.asarray(sub['Index']),
+                                     numpy.arange(sub.csize))
