@@ -311,15 +311,16 @@ class CatalogSourceBase(object):
                                 "cannot save '%s' key in attrs "
                                 "dictionary" % key)
 
-    # -- default columns (reference :1166-1216) ---------------------------
-    @column(is_default=True)
+    @property
     def Index(self):
-        """The global row index (reference :1178-1195: offset of the
-        lower ranks + local arange, dtype i8)."""
+        """The global row index, as an ATTRIBUTE like the reference's
+        (base/catalog.py:1178-1195 exposes a property, not a column):
+        offset of the lower ranks + local arange, dtype i8."""
         counts = self.comm.allgather(self.size)
         offset = int(numpy.sum(counts[:self.comm.rank], dtype='i8'))
         return offset + numpy.arange(self.size, dtype='i8')
 
+    # -- default columns (reference :1166-1216) ---------------------------
     @column(is_default=True)
     def Selection(self):
         return ConstantArray(True, self.size)

@@ -319,9 +319,8 @@ def test_index_column():
     import numpy
     from nbodykit_amd.lab import UniformCatalog
     source = UniformCatalog(nbar=2e-4, BoxSize=512., seed=42)
-    idx = numpy.asarray(source['Index'])
+    idx = source.Index
     numpy.testing.assert_array_equal(idx, numpy.arange(source.csize))
     assert idx.dtype == numpy.dtype('i8')
     sub = source.gslice(0, 100)
-    numpy.testing.assert_array_equal(numpy.asarray(sub['Index']),
-                                     numpy.arange(sub.csize))
+    numpy.testing.assert_array_equal(sub.Index, numpy.arange(sub.csize))
