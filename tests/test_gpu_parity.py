@@ -822,3 +822,21 @@ def test_mesh_filters():
     # TopHat runs and leaves the k=0 normalization intact
     th = mesh.apply(filters.TopHat(8.0)).compute(mode='real')
     assert abs(th.value.mean().item() - 1.0) < 1e-10
+
+
+def test_linearmesh_chisq_vs_theory():
+    """reference source/mesh/tests/test_linear.py:12-35: P(k) measured
+    from a LinearMesh realization agrees with the input Plin at reduced
+    chi^2 < 1.5 (variance 2 P^2 / Nmodes per bin)."""
+    from nbodykit_amd.lab import LinearMesh, FFTPower, LinearPower
+    from nbodykit_amd.cosmology import Planck15
+    Plin = LinearPower(Planck15, redshift=0.55,
+                       transfer='EisensteinHu')
+    source = LinearMesh(Plin, Nmesh=64, BoxSize=512., seed=42)
+    r = FFTPower(source, mode='1d', Nmesh=64, dk=0.01, kmin=0.005)
+    valid = r.power['modes'] > 0
+    theory = Plin(r.power['k'][valid])
+    errs = (2 * theory ** 2 / r.power['modes'][valid]) ** 0.5
+    chisq = (((r.power['power'][valid].real - theory) / errs) ** 2)
+    red = chisq.sum() / (valid.sum() - 1)
+    assert red < 1.5, red
