@@ -199,3 +199,33 @@ def test_oracle_poisson_monopole():
     P0 = r['power_0'].real - Pshot
     # mean over all bins should be small compared to the shot noise
     assert abs(numpy.nanmean(P0)) < 0.5 * Pshot
+
+
+def test_species_to_mesh_metadata():
+    """MSC.to_mesh BoxSize/Nmesh inference + consistency errors
+    (reference source/catalog/species.py:157-252)."""
+    data, ran = _mock_catalogs()
+    data.attrs['BoxSize'] = 512.
+    ran.attrs['BoxSize'] = 512.
+    data.attrs['Nmesh'] = 32
+    ran.attrs['Nmesh'] = 32
+    cat = MultipleSpeciesCatalog(['data', 'randoms'], data, ran)
+    mesh = cat.to_mesh()
+    numpy.testing.assert_array_equal(mesh.attrs['Nmesh'], 32)
+    numpy.testing.assert_array_equal(mesh.attrs['BoxSize'], 512.)
+    assert list(mesh) == ['data', 'randoms']
+    sub = mesh['data']
+    numpy.testing.assert_array_equal(sub.attrs['Nmesh'], 32)
+    with pytest.raises(KeyError):
+        mesh['other']
+
+    # inconsistent metadata -> error
+    ran2 = _mock_catalogs()[1]
+    ran2.attrs['BoxSize'] = 256.
+    ran2.attrs['Nmesh'] = 32
+    cat2 = MultipleSpeciesCatalog(['data', 'randoms'], data, ran2)
+    with pytest.raises(ValueError):
+        cat2.to_mesh()
+    # window kwarg rejected
+    with pytest.raises(RuntimeError):
+        cat.to_mesh(window='cic')
