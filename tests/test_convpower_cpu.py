@@ -229,3 +229,21 @@ def test_species_to_mesh_metadata():
     # window kwarg rejected
     with pytest.raises(RuntimeError):
         cat.to_mesh(window='cic')
+
+
+def test_fkp_catalog_no_randoms():
+    """randoms=None -> an empty randoms species (reference
+    convpower/catalog.py:47-49); bbox falls back to the data."""
+    data, _ = _mock_catalogs()
+    cat = FKPCatalog(data, None, BoxSize=512.)
+    assert cat['randoms'].csize == 0
+    assert 'NZ' in cat['randoms']
+    with pytest.warns(UserWarning):
+        mesh = cat.to_mesh(Nmesh=16, BoxCenter=1200., dtype='c16')
+    assert mesh.weighted_total('data') == data.size
+    assert mesh.weighted_total('randoms') == 0.0
+    # bbox derived from data when randoms are empty
+    cat2 = FKPCatalog(data, None)
+    with pytest.warns(UserWarning):
+        mesh2 = cat2.to_mesh(Nmesh=16, dtype='c16')
+    assert (numpy.asarray(mesh2.attrs['BoxSize']) > 0).all()
