@@ -324,3 +324,26 @@ def test_index_column():
     assert idx.dtype == numpy.dtype('i8')
     sub = source.gslice(0, 100)
     numpy.testing.assert_array_equal(sub.Index, numpy.arange(sub.csize))
+
+
+def test_mesh_setters_and_bad_window():
+    """reference source/mesh/tests/test_catalogmesh.py:99-131."""
+    import pytest
+    from nbodykit_amd.lab import UniformCatalog
+    source = UniformCatalog(nbar=3e-4, BoxSize=512., seed=42)
+    mesh = source.to_mesh(resampler='cic', Nmesh=64, interlaced=True,
+                          compensated=True)
+    assert mesh.compensated is True
+    mesh.compensated = False
+    assert mesh.compensated is False
+    assert mesh.interlaced is True
+    mesh.interlaced = False
+    assert mesh.interlaced is False
+    assert mesh.window == 'cic'
+    mesh.window = 'tsc'
+    assert mesh.window == 'tsc'
+    with pytest.raises(Exception):
+        mesh.window = 'BAD'
+    # unknown resampler at construction
+    with pytest.raises(ValueError):
+        source.to_mesh(resampler='db6', Nmesh=64)
