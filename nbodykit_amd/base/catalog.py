@@ -58,6 +58,31 @@ class ColumnAccessor(numpy.ndarray):
     def as_numpy(self):
         return numpy.asarray(self)
 
+    # in-place operators DISOWN instead of mutating: the reference's
+    # dask-backed columns never write back to the catalog (an operation
+    # yields a new array; ``cat[col] *= 10`` works via __setitem__) —
+    # a live numpy view would silently corrupt the stored column.
+    def __iadd__(self, other):
+        return numpy.asarray(self) + other
+
+    def __isub__(self, other):
+        return numpy.asarray(self) - other
+
+    def __imul__(self, other):
+        return numpy.asarray(self) * other
+
+    def __itruediv__(self, other):
+        return numpy.asarray(self) / other
+
+    def __ifloordiv__(self, other):
+        return numpy.asarray(self) // other
+
+    def __ipow__(self, other):
+        return numpy.asarray(self) ** other
+
+    def __imod__(self, other):
+        return numpy.asarray(self) % other
+
 
 def column(name=None, is_default=False):
     """Decorator marking a method as a named column (reference :97-125)."""

@@ -347,3 +347,33 @@ def test_mesh_setters_and_bad_window():
     # unknown resampler at construction
     with pytest.raises(ValueError):
         source.to_mesh(resampler='db6', Nmesh=64)
+
+
+def test_columnaccessor_semantics():
+    """reference test_columnaccessor (:345-376): operations on a fetched
+    column never write back to the catalog; explicit __setitem__ does."""
+    import numpy
+    from numpy.testing import assert_array_equal
+    from nbodykit_amd.base.catalog import ColumnAccessor
+    from nbodykit_amd.lab import UniformCatalog
+    source = UniformCatalog(nbar=2e-4, BoxSize=512., seed=42)
+
+    c = source['Position']
+    truth = numpy.array(c[0])
+    assert isinstance(c, ColumnAccessor)
+    c *= 10.
+    # c is no longer an accessor (it has transformed)...
+    assert not isinstance(c, ColumnAccessor)
+    # ...and the original is unaffected
+    assert_array_equal(numpy.asarray(source['Position'])[0], truth)
+    assert_array_equal(c[0], truth * 10.)
+
+    # explicit in-place via __setitem__ works
+    source['Position'] *= 10
+    assert_array_equal(numpy.asarray(source['Position'])[0], truth * 10)
+
+    # accessors carry their catalog
+    new_col = source['Selection']
+    assert isinstance(new_col, ColumnAccessor)
+    source['Selection2'] = new_col
+    assert source['Selection'].catalog is source
