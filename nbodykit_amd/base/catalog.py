@@ -236,6 +236,17 @@ class CatalogSourceBase(object):
     def make_column(self, array):
         return numpy.asarray(array)
 
+    def view(self, type=None):
+        """A view sharing columns and attrs with ``self`` (write-through
+        — adding a column on the view adds it to the source; reference
+        :440-472).  ``type`` optionally rebrands the class."""
+        cls = self.__class__ if type is None else type
+        obj = CatalogSourceBase.__new__(cls)
+        obj.__dict__.update(self.__dict__)
+        obj._overrides = self._overrides      # shared, not copied
+        obj.base = self
+        return obj
+
     def copy(self):
         """Shallow copy: every column of ``self`` referenced (no data
         copied), attrs decoupled (reference :474-507).  The copy is a

@@ -377,3 +377,25 @@ def test_columnaccessor_semantics():
     assert isinstance(new_col, ColumnAccessor)
     source['Selection2'] = new_col
     assert source['Selection'].catalog is source
+
+
+def test_catalog_view():
+    """reference test_view (:416-440): shared columns + attrs,
+    write-through."""
+    from nbodykit_amd.lab import UniformCatalog
+    source = UniformCatalog(nbar=2e-4, BoxSize=512., seed=42)
+    source['TEST'] = 10.
+    source.attrs['TEST'] = 10.0
+    view = source.view()
+    assert view.base is source
+    assert isinstance(view, source.__class__)
+    assert view.size == source.size and view.csize == source.csize
+    for k in source.attrs:
+        assert k in view.attrs
+    view['TEST2'] = 5.0
+    assert 'TEST2' in source
+    source.attrs['foo'] = 123
+    assert 'foo' in view.attrs
+    import numpy
+    numpy.testing.assert_array_equal(numpy.asarray(view['Position']),
+                                     numpy.asarray(source['Position']))
