@@ -192,6 +192,12 @@ class CatalogSourceBase(object):
             del self._overrides[col]
         elif col in self.hardcolumns:
             raise ValueError("cannot delete a hard-coded column")
+        else:
+            # a default column (Selection/Weight/Value) with no override
+            # is not deletable (reference base/catalog.py:944-953 raises
+            # for non-overridable columns; a silent no-op hides typos)
+            raise ValueError(
+                "cannot delete default column '%s' (no override set)" % col)
 
     def _get_slice(self, index):
         if isinstance(index, list):

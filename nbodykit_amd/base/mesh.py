@@ -85,12 +85,12 @@ class MeshSource(object):
 
     def to_real_field(self, out=None, normalize=True):
         if isinstance(self.base, MeshSource):
-            return self.base.to_real_field()
+            return self.base.to_real_field(out=out, normalize=normalize)
         return NotImplemented
 
     def to_complex_field(self, out=None):
         if isinstance(self.base, MeshSource):
-            return self.base.to_complex_field()
+            return self.base.to_complex_field(out=out)
         return NotImplemented
 
     def to_field(self, mode='real', out=None):

@@ -387,6 +387,11 @@ class BigFile(object):
             if r == self._rank() and nlocal > 0:
                 block.write(offset, array)
             self._barrier()
+        # fill the per-file sysv sums so headers interchange with tools
+        # that diff/validate them (bigfile's C writer always records them)
+        if self._rank() == 0:
+            block.update_checksums()
+        self._barrier()
         return block
 
     def __enter__(self):

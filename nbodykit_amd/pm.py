@@ -489,14 +489,19 @@ class ComplexField(_FieldBase):
         if comp is not None and kind == 'circular':
             window, interlaced = comp
             lib = hiplib.require()
+            # out=None must not mutate self (reference copy-on-apply
+            # semantics): run the kernel on a clone and return a new field.
+            if out is not None:
+                target = self
+            else:
+                target = ComplexField(self.pm, tensor=self.value.clone())
+                target.attrs = dict(self.attrs)
             hiplib.check(lib.nbk_compensate_f64(
-                hiplib.dptr(self.value), hiplib.i64_arr(self.pm.Nmesh),
+                hiplib.dptr(target.value), hiplib.i64_arr(self.pm.Nmesh),
                 hiplib.i64_arr(self.dims), hiplib.i64_arr(self.off),
                 None, hiplib.WINDOW_IDS[window], int(interlaced),
                 self._stream()), 'nbk_compensate_f64')
-            if out is not None:
-                return self
-            raise NotImplementedError("compensation filters apply in-place")
+            return target
 
         # generic host hook
         import torch

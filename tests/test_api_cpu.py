@@ -252,6 +252,13 @@ def test_ref_delitem_semantics():
     assert 'test' in source
     del source['test']
     assert 'test' not in source
+    # a default column with no override is not deletable either
+    with pytest.raises(ValueError):
+        del source['Selection']
+    # ... but becomes deletable once overridden
+    source['Selection'] = numpy.ones(source.size, dtype=bool)
+    del source['Selection']
+    assert numpy.all(numpy.asarray(source['Selection']))
 
 
 def test_ref_transform_arithmetic():
