@@ -52,10 +52,38 @@ def _is_trivial_true(sel):
 
 
 
-def _prepare_particles(pos_t, mass_t, pm):
+def _two_level_ys(pm):
+    """The y-group shift of the two-level locality sort, or None when no
+    LDS budget fits this mesh (coarse histogram n0*(n1>>ys) ints and fine
+    window (1<<ys)*n2 ints both <= 40960).  pm-only, so every rank of a
+    communicator computes the same answer — the fused to_complex_field
+    path gates on it collectively."""
+    n0, n1, n2 = (int(x) for x in pm.Nmesh)
+    LDSW = 40960
+    for ys in range(0, max(1, n1.bit_length())):
+        if (1 << ys) > n1 or n1 % (1 << ys):
+            break
+        win = (1 << ys) * n2
+        if win > LDSW:
+            break                       # grows with ys: hopeless beyond
+        if win < 1024 or win % 1024:
+            continue
+        if n0 * (n1 >> ys) <= LDSW:
+            return ys
+    return None
+
+
+def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False):
     """Return (pos_soa, mass, cell_sorted, rowtab) for the deposit
     kernel,
     bucket-sorting the chunk by mesh cell when it arrives scrambled.
+
+    ``force_rowtab`` makes the two-level path run regardless of the size
+    thresholds and always emit the row table (the fused paint+z-FFT path
+    has committed collectively and must not fall back per-rank: ws>1
+    ranks can end up with arbitrarily few local particles after
+    routing).  The caller must have checked ``_two_level_ys(pm)`` is not
+    None and n2 <= 20480.
 
     The deposit kernel's wave-merge and its L2 locality both depend on
     nearby-in-space particles being nearby-in-memory; a cell-ordered
@@ -71,11 +99,17 @@ def _prepare_particles(pos_t, mass_t, pm):
     """
     import torch
     n = len(pos_t)
-    if n < _global_options['sort_min_n']:
+    n0, n1, n2 = (int(x) for x in pm.Nmesh)
+    if force_rowtab and n == 0:
+        # empty rank (post-routing): an all-zero row table means every
+        # row range is empty and the gather kernel paints zeros
+        rowtab = torch.zeros(n0 * n1 + 1, dtype=torch.int32,
+                             device='cuda')
+        return pos_t.t().contiguous(), mass_t, True, rowtab
+    if not force_rowtab and n < _global_options['sort_min_n']:
         return pos_t.t().contiguous(), mass_t, False, None
 
     lib = hiplib.require()
-    n0, n1, n2 = (int(x) for x in pm.Nmesh)
     ncells = n0 * n1 * n2
 
     nmesh = hiplib.i64_arr(pm.Nmesh)
@@ -119,23 +153,11 @@ def _prepare_particles(pos_t, mass_t, pm):
     # kernel (LDS count + block scan + placement) emitting the exact
     # cell order.  ys balances the two LDS budgets: coarse histogram
     # n0*(n1>>ys) ints vs fine window (1<<ys)*n2 ints, both <= 40960.
-    n0, n1, n2 = (int(x) for x in pm.Nmesh)
-    LDSW = 40960
-    ys_fine = None
-    for ys in range(0, max(1, n1.bit_length())):
-        if (1 << ys) > n1 or n1 % (1 << ys):
-            break
-        win = (1 << ys) * n2
-        if win > LDSW:
-            break                       # grows with ys: hopeless beyond
-        if win < 1024 or win % 1024:
-            continue
-        if n0 * (n1 >> ys) <= LDSW:
-            ys_fine = ys
-            break
-    use_two = (n >= _global_options['sort_two_level_min_n']
-               and ncells > _global_options['sort_two_level_min_cells']
-               and ys_fine is not None)
+    ys_fine = _two_level_ys(pm)
+    use_two = ys_fine is not None and (
+        force_rowtab
+        or (n >= _global_options['sort_two_level_min_n']
+            and ncells > _global_options['sort_two_level_min_cells']))
     if use_two:
         ys = ys_fine
         nbuck = n0 * (n1 >> ys)
@@ -147,8 +169,10 @@ def _prepare_particles(pos_t, mass_t, pm):
         hiplib.check(lib.nbk_xsort_count_f64(
             hiplib.dptr(pos_in), n, CH, nmesh, box, ys, hiplib.dptr(mat),
             hiplib.dptr(flag), stream), 'nbk_xsort_count_f64')
-        if int(flag.item()) == 0:
-            # already cell-ordered: no sorting needed
+        if not force_rowtab and int(flag.item()) == 0:
+            # already cell-ordered: no sorting needed.  (With
+            # force_rowtab the sort still runs — it is the cheapest
+            # correct way to get the row table the fused path needs.)
             return pos_t.t().contiguous(), mass_t, True, None
         tm = mat.view(nblocks, nbuck).t().contiguous().view(-1)
         incl = torch.cumsum(tm, 0, dtype=torch.int32)
@@ -462,28 +486,42 @@ class CatalogMesh(MeshSource):
         trip of the real path (combine directly in k).  Numerically the
         FFT is the same radix-2 code as nbk_fft_r2c_z; the 1/N^3 and
         1/nbar (normalize) factors fold into the kernel's output scale.
+
+        Multi-rank: every gate below the collective allgather is
+        rank-invariant (pm geometry or globally-reduced sizes), so all
+        ranks commit to the fused path together — it ends in the RCCL
+        pencil transpose (_r2c_finish) and a per-rank fallback would
+        deadlock.  Particles are routed to their slab owners first
+        (the same ghost exchange as to_real_field) and the two-level
+        sort is forced so every rank has a row table even when routing
+        leaves it few (or zero) particles.
+
         Falls back (NotImplemented -> to_real_field().r2c() in
-        MeshSource.to_field) for multi-rank runs, user `out`, paints
-        that need chunking, non-power-of-two z, or inputs below the
-        locality-sort thresholds."""
+        MeshSource.to_field) for user `out`, paints that need chunking,
+        non-power-of-two z, or inputs below the locality-sort
+        thresholds."""
         import torch
         if out is not None:
             return NotImplemented
         pm = self.pm
         comm = pm.comm
-        if comm.size != 1:
-            return NotImplemented
+        ws = comm.size
         n2 = int(pm.Nmesh[2])
         if n2 < 8 or n2 > 4096 or (n2 & (n2 - 1)):
             return NotImplemented
+        if _two_level_ys(pm) is None:
+            return NotImplemented       # no rowtab possible (pm-based)
+        # collective gates: identical decision on every rank.  (The
+        # sum()-of-sizes test reduces to the old per-rank test at ws=1.)
         Position = self.Position
-        if len(Position) > _global_options['paint_chunk_size']:
+        sizes = comm.allgather(len(Position))
+        if max(sizes) > _global_options['paint_chunk_size']:
             return NotImplemented
         # cheap gate BEFORE pulling columns: below the two-level sort
         # thresholds the gather/rowtab path cannot engage and the real
         # path would redo all the preparation work (C2-size inputs)
         ncells = int(numpy.prod(pm.Nmesh))
-        if not (len(Position) >= _global_options['sort_two_level_min_n']
+        if not (sum(sizes) >= _global_options['sort_two_level_min_n']
                 and ncells > _global_options['sort_two_level_min_cells']):
             return NotImplemented
 
@@ -527,13 +565,13 @@ class CatalogMesh(MeshSource):
         w_t = None if weight is None else _to_device_f64(weight)
         v_t = None if value is None else _to_device_f64(value)
 
-        N = len(pos_t)
+        Nlocal = len(pos_t)
         if w_t is None:
-            W = float(N)
-            W2 = float(N)
+            Wlocal = float(Nlocal)
+            W2local = float(Nlocal)
         else:
-            W = float(w_t.sum().item())
-            W2 = float((w_t * w_t).sum().item())
+            Wlocal = float(w_t.sum().item())
+            W2local = float((w_t * w_t).sum().item())
         if w_t is None and v_t is None:
             mass_t = None
         elif v_t is None:
@@ -543,14 +581,22 @@ class CatalogMesh(MeshSource):
         else:
             mass_t = w_t * v_t
 
+        # global counters (pre-routing, like to_real_field's chunk loop)
+        N = comm.allreduce(Nlocal)
+        W = comm.allreduce(Wlocal)
+        W2 = comm.allreduce(W2local)
+
         nbar = W / float(numpy.prod(pm.Nmesh))
         if N == 0 or nbar <= 0:
-            return NotImplemented
+            return NotImplemented       # globally empty: rank-invariant
+
+        if ws > 1:
+            pos_t, mass_t = self._route(pos_t, mass_t)
 
         pos_soa, mass_t, sorted_, rowtab = _prepare_particles(
-            pos_t, mass_t, pm)
-        if rowtab is None:
-            return NotImplemented
+            pos_t, mass_t, pm, force_rowtab=True)
+        assert rowtab is not None
+        n_routed = len(pos_t)
 
         scale = 1.0 / float(numpy.prod(pm.Nmesh)) / nbar
         nzh = n2 // 2 + 1
@@ -562,12 +608,13 @@ class CatalogMesh(MeshSource):
             z = torch.empty(shape, dtype=torch.complex128, device='cuda')
             # time only the fused kernel as 'paint' (the y/x passes are
             # FFT work, not paint work — the bench roofline reads this)
-            with profiling.collect('paint', N):
+            with profiling.collect('paint', n_routed):
                 hiplib.check(lib.nbk_paint_gather_fft_f64(
-                    hiplib.dptr(pos_soa), hiplib.dptr(mass_t), N, nmesh,
-                    box, window_id, float(shift), hiplib.dptr(rowtab),
-                    hiplib.dptr(z), pm.x_start, pm.nx_local, scale,
-                    stream), 'nbk_paint_gather_fft_f64')
+                    hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n_routed,
+                    nmesh, box, window_id, float(shift),
+                    hiplib.dptr(rowtab), hiplib.dptr(z), pm.x_start,
+                    pm.nx_local, scale, stream),
+                    'nbk_paint_gather_fft_f64')
             return _r2c_finish(z, pm, stream)
 
         cplx = one(0.0)
@@ -585,12 +632,28 @@ class CatalogMesh(MeshSource):
             # then goes through c2r + r2c) Hermitian-projects the
             # self-conjugate z-planes: numpy's irfft drops the
             # imaginary parts of the kz = 0/Nyquist bins, which in k
-            # is c <- (c + conj(c(-k)))/2 on those planes
+            # is c <- (c + conj(c(-k)))/2 on those planes.  Multi-rank
+            # the field is y-partitioned, and c(-k) needs rows owned by
+            # other ranks: allgather the two tiny planes (16 B * Nx * Ny
+            # each), project, keep the local y-block.
             for kz in (0, n2 // 2):
-                A = cplx[:, :, kz]
-                B = torch.conj(torch.roll(torch.flip(A, (0, 1)),
-                                          (1, 1), (0, 1)))
-                cplx[:, :, kz] = 0.5 * (A + B)
+                if ws > 1:
+                    import torch.distributed as dist
+                    A = cplx[:, :, kz].contiguous()
+                    parts = [torch.empty_like(A) for _ in range(ws)]
+                    dist.all_gather([torch.view_as_real(p) for p in parts],
+                                    torch.view_as_real(A))
+                    Af = torch.cat(parts, dim=1)
+                    B = torch.conj(torch.roll(torch.flip(Af, (0, 1)),
+                                              (1, 1), (0, 1)))
+                    proj = 0.5 * (Af + B)
+                    cplx[:, :, kz] = proj[:, pm.y_start:
+                                          pm.y_start + pm.ny_local]
+                else:
+                    A = cplx[:, :, kz]
+                    B = torch.conj(torch.roll(torch.flip(A, (0, 1)),
+                                              (1, 1), (0, 1)))
+                    cplx[:, :, kz] = 0.5 * (A + B)
 
         with numpy.errstate(divide='ignore', invalid='ignore'):
             shotnoise = float(numpy.prod(pm.BoxSize)) * W2 / W ** 2

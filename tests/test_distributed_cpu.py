@@ -272,14 +272,30 @@ def test_gloo_bigfile_catalog_roundtrip():
 
 
 def _body_fused_complex_gate(comm):
-    """to_complex_field must decline (NotImplemented) on multi-rank
-    comms BEFORE touching the GPU extension (the real path handles
-    distribution)."""
+    """The fused paint+z-FFT path is multi-rank: with the size gates
+    satisfied every rank must COMMIT to it together (reaching
+    hiplib.require(), which raises on this GPU-less box) rather than
+    declining per-rank — a split decision would deadlock in the pencil
+    transpose.  Below the thresholds it must still decline collectively
+    before touching the extension."""
+    from nbodykit_amd import set_options
     from nbodykit_amd.lab import ArrayCatalog
     cat = ArrayCatalog({'Position': numpy.random.RandomState(
-        comm.rank).uniform(0, 32., size=(100, 3))}, comm=comm)
+        comm.rank).uniform(0, 32., size=(100 + 50 * comm.rank, 3))},
+        comm=comm)
     mesh = cat.to_mesh(Nmesh=32, BoxSize=32., dtype='f8')
-    return mesh.to_complex_field() is NotImplemented
+    declined = mesh.to_complex_field() is NotImplemented
+    # lower the gates so the (global) sizes qualify: the path must now
+    # be taken on every rank (ws=2 included), stopping only at the
+    # GPU-extension gate
+    committed = False
+    with set_options(sort_two_level_min_n=64,
+                     sort_two_level_min_cells=1024, sort_min_n=64):
+        try:
+            mesh.to_complex_field()
+        except RuntimeError as e:
+            committed = 'requires an AMD GPU' in str(e)
+    return declined and committed
 
 
 @pytest.mark.timeout(300)

@@ -872,3 +872,57 @@ def test_species_mesh_summed_paint():
     # and it flows through FFTPower
     r = FFTPower(mesh, mode='1d')
     assert numpy.isfinite(r.power['power'].real[1:]).any()
+
+
+# ---- at-scale parity: production-size inputs
+# through the production thresholds, against the oracle at 1e-10 --------
+
+@pytest.mark.timeout(900)
+def test_at_scale_c2_parity():
+    """C2 shape (1e7 uniform pts / 256^3, CIC compensated) at DEFAULT
+    gates: the two-level sort + ownership-gather + fused z-FFT path as
+    the bench runs it.  The small-mesh tests can't see scale-dependent
+    index/digitize/sort-threshold bugs (oracle ~11 s CPU)."""
+    n = int(1e7)
+    pos = numpy.random.RandomState(42).uniform(0, 1000., size=(n, 3))
+    cat = ArrayCatalog({'Position': pos})
+    mesh = cat.to_mesh(Nmesh=256, BoxSize=1000., dtype='f8',
+                       compensated=True, resampler='cic')
+    # must take the fused paint+z-FFT path at production thresholds
+    assert mesh.to_complex_field() is not NotImplemented
+    r = FFTPower(mesh, mode='1d')
+    want = fftpower_oracle(pos, Nmesh=256, BoxSize=1000., mode='1d',
+                           resampler='cic', compensated=True)
+    assert_array_equal(r.power['modes'], want['modes'])
+    got = r.power['power'].real
+    ref = want['power'].real
+    ok = numpy.isfinite(ref) & (numpy.abs(ref) > 0)
+    rel = numpy.abs(got[ok] - ref[ok]) / numpy.abs(ref[ok])
+    assert rel.max() < 1e-10, 'at-scale P parity: %g' % rel.max()
+
+
+@pytest.mark.timeout(900)
+def test_at_scale_512_tsc_interlaced_spot():
+    """512^3 TSC+interlaced spot check (C3's mesh size): big-mesh index
+    math, interlaced ghost range and the k-space combine at a
+    production mesh (thresholds lowered so 1e6 pts engage the fused
+    path; the kernels see the same 512^3 geometry the bench does)."""
+    n = int(1e6)
+    pos = numpy.random.RandomState(7).uniform(0, 2500., size=(n, 3))
+    cat = ArrayCatalog({'Position': pos})
+    with set_options(sort_min_n=100000, sort_two_level_min_n=100000,
+                     sort_two_level_min_cells=1 << 23):
+        mesh = cat.to_mesh(Nmesh=512, BoxSize=2500., dtype='f8',
+                           compensated=True, resampler='tsc',
+                           interlaced=True)
+        assert mesh.to_complex_field() is not NotImplemented
+        r = FFTPower(mesh, mode='1d')
+    want = fftpower_oracle(pos, Nmesh=512, BoxSize=2500., mode='1d',
+                           resampler='tsc', compensated=True,
+                           interlaced=True)
+    assert_array_equal(r.power['modes'], want['modes'])
+    got = r.power['power'].real
+    ref = want['power'].real
+    ok = numpy.isfinite(ref) & (numpy.abs(ref) > 0)
+    rel = numpy.abs(got[ok] - ref[ok]) / numpy.abs(ref[ok])
+    assert rel.max() < 1e-10, '512^3 interlaced parity: %g' % rel.max()
