@@ -272,9 +272,14 @@ def main():
         local = int(os.environ.get('LOCAL_RANK', rank))
         # modulo lets an N-rank run share fewer devices (single-GPU
         # smoke testing of the multi-rank path); production launches
-        # have one rank per GPU and the modulo is a no-op
-        torch.cuda.set_device(local % max(1, torch.cuda.device_count()))
-        dist.init_process_group('nccl')
+        # have one rank per GPU and the modulo is a no-op.  RCCL refuses
+        # duplicate devices, so shared-device smoke runs fall back to
+        # gloo collectives with host staging (pm.all_to_all_tensor) —
+        # compute stays on the GPU; such runs check correctness, not
+        # interconnect throughput.
+        ndev = max(1, torch.cuda.device_count())
+        torch.cuda.set_device(local % ndev)
+        dist.init_process_group('nccl' if ndev >= ws else 'gloo')
 
     from nbodykit_amd import profiling, set_options
     from nbodykit_amd.lab import FFTPower

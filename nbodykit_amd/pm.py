@@ -34,26 +34,59 @@ def _int_freqs(N):
 # ---- pencil-transpose helpers (pure tensor ops + one alltoall; factored
 # out so the reshape logic is CPU-testable under gloo) -------------------
 
+def all_to_all_tensor(out, inp, out_splits=None, in_splits=None):
+    """dist.all_to_all_single that also runs when the process group has
+    no transport for the tensors' device (gloo + CUDA tensors: the
+    single-GPU multi-rank harness — RCCL refuses two ranks on one
+    device, "Duplicate GPU detected") by staging through host memory.
+    nccl/RCCL process groups take the direct path (alltoall over
+    xGMI)."""
+    import torch.distributed as dist
+    if inp.is_cuda and dist.get_backend() != 'nccl':
+        import torch
+        inp_c = inp.cpu()
+        out_c = torch.empty(out.shape, dtype=out.dtype, device='cpu')
+        dist.all_to_all_single(out_c, inp_c, out_splits, in_splits)
+        out.copy_(out_c)
+    else:
+        dist.all_to_all_single(out, inp, out_splits, in_splits)
+
+
+def all_gather_tensor(parts, inp):
+    """dist.all_gather with the same gloo+CUDA staging as
+    :func:`all_to_all_tensor` (complex tensors go through
+    view_as_real — gloo has no complex type)."""
+    import torch
+    import torch.distributed as dist
+    if inp.is_cuda and dist.get_backend() != 'nccl':
+        inp_c = torch.view_as_real(inp.cpu())
+        parts_c = [torch.empty_like(inp_c) for _ in parts]
+        dist.all_gather(parts_c, inp_c)
+        for p, pc in zip(parts, parts_c):
+            p.copy_(torch.view_as_complex(pc))
+    else:
+        dist.all_gather([torch.view_as_real(p) for p in parts],
+                        torch.view_as_real(inp))
+
+
 def transpose_x_to_y(cplx, ws, nx_l, ny_l, nzh):
     """(nx_l, ny, nzh) x-slab -> (nx, ny_l, nzh) y-slab (the pfft pencil
     transpose, RCCL alltoall over xGMI in GPU runs)."""
     import torch
-    import torch.distributed as dist
     send = cplx.view(nx_l, ws, ny_l, nzh).permute(1, 0, 2, 3).contiguous()
     recv = torch.empty_like(send)
-    dist.all_to_all_single(torch.view_as_real(recv).view(-1),
-                           torch.view_as_real(send).view(-1))
+    all_to_all_tensor(torch.view_as_real(recv).view(-1),
+                      torch.view_as_real(send).view(-1))
     return recv.view(ws * nx_l, ny_l, nzh)
 
 
 def transpose_y_to_x(cplx, ws, nx_l, ny_l, nzh):
     """inverse of :func:`transpose_x_to_y`"""
     import torch
-    import torch.distributed as dist
     send = cplx.view(ws, nx_l, ny_l, nzh).contiguous()
     recv = torch.empty_like(send)
-    dist.all_to_all_single(torch.view_as_real(recv).view(-1),
-                           torch.view_as_real(send).view(-1))
+    all_to_all_tensor(torch.view_as_real(recv).view(-1),
+                      torch.view_as_real(send).view(-1))
     return recv.permute(1, 0, 2, 3).contiguous().view(nx_l, ws * ny_l, nzh)
 
 
@@ -65,13 +98,12 @@ def exchange_particle_arrays(send_flat, counts_send, comm):
     (RCCL on GPU, gloo on CPU); mirrors pmesh's layout.exchange
     (called at nbodykit/source/mesh/catalog.py:282-284)."""
     import torch
-    import torch.distributed as dist
     counts_recv = [row[comm.rank] for row in comm.allgather(counts_send)]
     width = send_flat.shape[1] if send_flat.dim() > 1 else 1
     flat = send_flat.reshape(len(send_flat), -1)
     out = torch.empty((sum(counts_recv), flat.shape[1]),
                       dtype=flat.dtype, device=flat.device)
-    dist.all_to_all_single(
+    all_to_all_tensor(
         out.view(-1), flat.contiguous().view(-1),
         [c * flat.shape[1] for c in counts_recv],
         [c * flat.shape[1] for c in counts_send])
@@ -351,7 +383,7 @@ class RealField(_FieldBase):
                        for row in comm.allgather(counts_send)]
         back = torch.empty(int(sum(counts_send)), dtype=torch.float64,
                            device=partial.device)
-        dist.all_to_all_single(back, partial, counts_send, counts_recv)
+        all_to_all_tensor(back, partial, counts_send, counts_recv)
 
         out = torch.zeros(len(pos_t), dtype=torch.float64, device='cuda')
         out.index_add_(0, idxs, back)

@@ -638,11 +638,10 @@ class CatalogMesh(MeshSource):
             # each), project, keep the local y-block.
             for kz in (0, n2 // 2):
                 if ws > 1:
-                    import torch.distributed as dist
+                    from nbodykit_amd.pm import all_gather_tensor
                     A = cplx[:, :, kz].contiguous()
                     parts = [torch.empty_like(A) for _ in range(ws)]
-                    dist.all_gather([torch.view_as_real(p) for p in parts],
-                                    torch.view_as_real(A))
+                    all_gather_tensor(parts, A)
                     Af = torch.cat(parts, dim=1)
                     B = torch.conj(torch.roll(torch.flip(Af, (0, 1)),
                                               (1, 1), (0, 1)))

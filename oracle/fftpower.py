@@ -160,7 +160,13 @@ def project_to_basis(y3d, geom, edges, los=(0, 0, 1), poles=(),
     for islab in range(y3d.shape[0]):
         cx = float(coords[0][islab, 0, 0])
 
-        xslab = cx ** 2 + cy ** 2 + cz ** 2           # norm2 (Ny,Nzh)
+        # cx*cx, not cx**2: python's scalar pow rounds differently from
+        # the product in the last ulp for a handful of values (observed
+        # at Nmesh=512, Box=2500), while the reference's norm2 is a
+        # numpy ARRAY power (meshtools.py:117), which is elementwise
+        # x*x — modes sitting exactly on a k-bin edge digitize by this
+        # ulp (the GPU kernel also computes kx*kx)
+        xslab = cx * cx + cy * cy + cz * cz           # norm2 (Ny,Nzh)
         if xslab.size == 0:
             continue
 

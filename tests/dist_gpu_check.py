@@ -53,8 +53,15 @@ def main():
     local = int(os.environ.get('LOCAL_RANK', rank))
     ndev = max(1, torch.cuda.device_count())
     torch.cuda.set_device(local % ndev)
-    dist.init_process_group('nccl')
-    log('rank %d/%d on device %d (of %d)' % (rank, ws, local % ndev, ndev))
+    # RCCL refuses two ranks on one physical device ("Duplicate GPU
+    # detected", RCCL 2.26) — on a box with fewer GPUs than ranks the
+    # collectives run over gloo with host staging (pm.all_to_all_tensor)
+    # while ALL compute (route/sort/paint/FFT/bin kernels) stays on the
+    # GPU; with one rank per GPU the backend is nccl (= RCCL over xGMI).
+    backend = 'nccl' if ndev >= ws else 'gloo'
+    dist.init_process_group(backend)
+    log('rank %d/%d on device %d (of %d), backend %s'
+        % (rank, ws, local % ndev, ndev, backend))
 
     from nbodykit_amd import set_options
     from nbodykit_amd.lab import ArrayCatalog, FFTPower
