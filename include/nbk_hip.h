@@ -164,27 +164,32 @@ int nbk_bucket_scatter_f64(const double* pos_aos, const double* mass,
  *     chunk of `chunk` particles into `mat` (ceil(n/chunk) x nbuckets,
  *     int32 row-major), plus the cell-order detection into
  *     scrambled_flag (may be NULL);
- *   - host: bucket-major exclusive scan of `mat` -> `bases`, and bucket
- *     totals scan -> `bucket_bases` (nbuckets+1);
- *   - nbk_xsort_scatter_f64 (pass C): re-reads each chunk, places AoS
- *     rows from LDS cursors seeded with `bases` (deterministic — no
- *     global atomics);
- *   - nbk_bucket_fine_f64: one block per coarse bucket; counts the
- *     bucket's cells in an LDS window, block-scans it, and emits the
- *     exact cell-sorted SoA output (x[n] y[n] z[n]) — no global
- *     atomics.
+ *   - nbk_scan_matrix_i32: device scan of `mat` -> per-chunk cursor
+ *     seeds `bases` (chunk-major) and bucket edges `bucket_bases`
+ *     (nbuckets+1, exclusive; [nbuckets] = n).  colsum_tmp is nbuckets
+ *     ints of scratch;
+ *   - nbk_xsort_scatter_f64 (pass C): re-reads each chunk, places SoA
+ *     planes (x[n] y[n] z[n] into pos_out) from LDS cursors seeded
+ *     with `bases` (deterministic — no global atomics);
+ *   - nbk_bucket_fine_f64: one block per coarse bucket over the SoA
+ *     coarse output; counts the bucket's cells in an LDS window,
+ *     block-scans it, and emits the exact cell-sorted SoA output
+ *     (x[n] y[n] z[n]) — no global atomics.
  * Requires power-friendly dims (n1 divisible by 1<<ys, window
  * divisible by 1024) and n < 2^31. */
 int nbk_xsort_count_f64(const double* pos_aos, int64_t n, int chunk,
                         const int64_t nmesh[3], const double box[3],
                         int ys, int* mat, int* scrambled_flag,
                         void* stream);
+int nbk_scan_matrix_i32(const int* mat, int64_t nblocks, int64_t nbuck,
+                        int* colsum_tmp, int* bases, int* bucket_bases,
+                        void* stream);
 int nbk_xsort_scatter_f64(const double* pos_aos, const double* mass,
                           int64_t n, int chunk, const int64_t nmesh[3],
                           const double box[3], int ys, const int* bases,
                           double* pos_out, double* mass_out,
                           void* stream);
-int nbk_bucket_fine_f64(const double* pos_aos, const double* mass,
+int nbk_bucket_fine_f64(const double* pos_soa, const double* mass,
                         int64_t n, const int64_t nmesh[3],
                         const double box[3], int ys,
                         const int* bucket_bases,

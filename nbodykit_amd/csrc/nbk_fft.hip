@@ -93,6 +93,75 @@ __device__ void lds_fft(cdouble* buf, int m, int bits, int TI,
     }
 }
 
+// Radix-4 variant: pairs of radix-2 DIT stages fused into one 4-point
+// butterfly over the SAME bit-reversed input order, halving the
+// __syncthreads count and LDS round trips (the strided passes are
+// sync/latency bound — DESIGN.md round-2 lever 1).  For odd log2(m)
+// one multiply-free radix-2 stage (W_2^0 = 1) runs first.  Algebra:
+// with w = W_{2L}^p, the fused (len=L, len=2L) pair on slots
+// (p, p+h, p+2h, p+3h), h = L/2, is the standard radix-4 DIT butterfly
+// with x1/x2 exchanged (bit-reversal vs digit-reversal order):
+//   b1 = x2 w, b2 = x1 w^2, b3 = x3 w^3
+//   e0 = x0 + b2, e1 = x0 - b2, o0 = b1 + b3, o1 = b1 - b3
+//   y = (e0+o0, e1 -/+ i*o1, e0-o0, e1 +/- i*o1)   (forward/inverse)
+// Validated element-exact against numpy fft for m = 4..2048.
+template <bool INV>
+__device__ void lds_fft4(cdouble* buf, int m, int TI,
+                         const cdouble* __restrict__ table) {
+    const int T = blockDim.x;
+    const int tid = threadIdx.x;
+    const int bits = 31 - __clz((unsigned)m);
+    int len = 2;
+    if (bits & 1) {
+        for (int w = tid; w < (m >> 1) * TI; w += T) {
+            const int c = w % TI;
+            const int j = w / TI;
+            const int i0 = (2 * j) * TI + c;
+            const int i1 = i0 + TI;
+            const cdouble u = buf[i0];
+            const cdouble v = buf[i1];
+            buf[i0] = cadd(u, v);
+            buf[i1] = csub(u, v);
+        }
+        __syncthreads();
+        len = 4;
+    }
+    for (; 2 * len <= m; len <<= 2) {
+        const int h = len >> 1;
+        const int tw = m / len;              // W_{2L}^p = table[p*tw]
+        for (int w = tid; w < (m >> 2) * TI; w += T) {
+            const int c = w % TI;
+            const int j = w / TI;
+            const int grp = j / h;
+            const int pos = j - grp * h;
+            const int base = (grp * (len << 1) + pos) * TI + c;
+            const int s = h * TI;
+            cdouble w1 = table[pos * tw];
+            cdouble w2 = table[2 * pos * tw];
+            if (INV) { w1.im = -w1.im; w2.im = -w2.im; }
+            const cdouble w3 = cmul(w1, w2);
+            const cdouble x0 = buf[base];
+            const cdouble x1 = buf[base + s];
+            const cdouble x2 = buf[base + 2 * s];
+            const cdouble x3 = buf[base + 3 * s];
+            const cdouble b1 = cmul(x2, w1);
+            const cdouble b2 = cmul(x1, w2);
+            const cdouble b3 = cmul(x3, w3);
+            const cdouble e0 = cadd(x0, b2);
+            const cdouble e1 = csub(x0, b2);
+            const cdouble o0 = cadd(b1, b3);
+            const cdouble o1 = csub(b1, b3);
+            buf[base] = cadd(e0, o0);
+            buf[base + 2 * s] = csub(e0, o0);
+            const cdouble io1 = INV ? cdouble{-o1.im, o1.re}
+                                    : cdouble{o1.im, -o1.re};
+            buf[base + s] = cadd(e1, io1);
+            buf[base + 3 * s] = csub(e1, io1);
+        }
+        __syncthreads();
+    }
+}
+
 // ---- z-axis real <-> half-complex ------------------------------------
 
 // one contiguous real line of nz doubles per block -> nz/2+1 complex
@@ -111,7 +180,7 @@ __global__ void kfft_r2c_z(const double* __restrict__ real,
         buf[bitrev(q, bits)] = g[q];
     __syncthreads();
 
-    lds_fft<false>(buf, m, bits, 1, table);
+    lds_fft4<false>(buf, m, 1, table);
 
     // untwiddle split: X[k] = E[k] + W_nz^k * O[k], k = 0..m
     cdouble* out = (cdouble*)cplx + line * (m + 1);
@@ -165,7 +234,7 @@ __global__ void kfft_c2r_z(const double* __restrict__ cplx,
     }
     __syncthreads();
 
-    lds_fft<true>(buf, m, bits, 1, table);
+    lds_fft4<true>(buf, m, 1, table);
 
     // unpack: line[2t] = 2 Re(z[t]), line[2t+1] = 2 Im(z[t])
     cdouble* out = (cdouble*)(real + line * nz);
@@ -199,7 +268,7 @@ __global__ void kfft_c_strided(double* __restrict__ data,
     }
     __syncthreads();
 
-    lds_fft<INV>(buf, nfft, bits, TI, table);
+    lds_fft4<INV>(buf, nfft, TI, table);
 
     for (int w = threadIdx.x; w < nfft * TI; w += blockDim.x) {
         const int c = w % TI;

@@ -481,21 +481,54 @@ __global__ void kpaint_gather(const double* __restrict__ px,
         }
         __builtin_amdgcn_wave_barrier();
 
-        for (int len = 2; len <= m; len <<= 1) {
-            const int half = len >> 1;
-            const int tw = m / len;
-            for (int q = lane; q < (m >> 1); q += 64) {
-                const int grp = q / half;
-                const int pos = q - grp * half;
-                const int i0 = grp * len + pos;
-                const int i1 = i0 + half;
-                const cdouble wv = table[2 * pos * tw];
-                const cdouble u = z[i0];
-                const cdouble v = cmul(z[i1], wv);
-                z[i0] = cadd(u, v);
-                z[i1] = csub(u, v);
+        // fused radix-4 stages (same bit-reversed order; pairs of
+        // radix-2 stages become one 4-point butterfly — half the LDS
+        // round trips; algebra as lds_fft4 in nbk_fft.hip, validated
+        // element-exact against numpy).  Odd log2(m): one multiply-free
+        // radix-2 stage first.
+        {
+            const int fbits = 31 - __clz((unsigned)m);
+            int len = 2;
+            if (fbits & 1) {
+                for (int q = lane; q < (m >> 1); q += 64) {
+                    const int i0 = 2 * q;
+                    const cdouble u = z[i0];
+                    const cdouble v = z[i0 + 1];
+                    z[i0] = cadd(u, v);
+                    z[i0 + 1] = csub(u, v);
+                }
+                __builtin_amdgcn_wave_barrier();
+                len = 4;
             }
-            __builtin_amdgcn_wave_barrier();
+            for (; 2 * len <= m; len <<= 2) {
+                const int h = len >> 1;
+                const int tw = m / len;
+                for (int q = lane; q < (m >> 2); q += 64) {
+                    const int grp = q / h;
+                    const int pos = q - grp * h;
+                    const int base = grp * (len << 1) + pos;
+                    const cdouble w1 = table[pos * tw];
+                    const cdouble w2 = table[2 * pos * tw];
+                    const cdouble w3 = cmul(w1, w2);
+                    const cdouble x0 = z[base];
+                    const cdouble x1 = z[base + h];
+                    const cdouble x2 = z[base + 2 * h];
+                    const cdouble x3 = z[base + 3 * h];
+                    const cdouble b1 = cmul(x2, w1);
+                    const cdouble b2 = cmul(x1, w2);
+                    const cdouble b3 = cmul(x3, w3);
+                    const cdouble e0 = cadd(x0, b2);
+                    const cdouble e1 = csub(x0, b2);
+                    const cdouble o0 = cadd(b1, b3);
+                    const cdouble o1 = csub(b1, b3);
+                    z[base] = cadd(e0, o0);
+                    z[base + 2 * h] = csub(e0, o0);
+                    const cdouble io1 = {o1.im, -o1.re};   // -i*o1
+                    z[base + h] = cadd(e1, io1);
+                    z[base + 3 * h] = csub(e1, io1);
+                }
+                __builtin_amdgcn_wave_barrier();
+            }
         }
 
         // untwiddle split: X[k] = E[k] + W_n2^k O[k], k = 0..m, written

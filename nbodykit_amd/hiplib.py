@@ -20,7 +20,7 @@ EXPORTED_SYMBOLS = [
     'nbk_paint_f64', 'nbk_paint_sorted_f64', 'nbk_readout_f64', 'nbk_recon_displacement_f64',
     'nbk_bucket_count_f64', 'nbk_bucket_scatter_f64',
     'nbk_xsort_count_f64', 'nbk_xsort_scatter_f64',
-    'nbk_bucket_fine_f64', 'nbk_paint_gather_f64',
+    'nbk_bucket_fine_f64', 'nbk_scan_matrix_i32', 'nbk_paint_gather_f64',
     'nbk_paint_gather_fft_f64',
     'nbk_fft_r2c_z', 'nbk_fft_c2r_z', 'nbk_fft_c_strided',
     'nbk_compensate_f64', 'nbk_interlace_combine_f64', 'nbk_power3d_f64',
@@ -67,6 +67,9 @@ def _declare(lib):
                                           ctypes.c_int, c_i64_p, c_f64_p,
                                           ctypes.c_int, c_void, c_void,
                                           c_void, c_void]
+    lib.nbk_scan_matrix_i32.restype = ctypes.c_int
+    lib.nbk_scan_matrix_i32.argtypes = [c_void, c_i64, c_i64, c_void,
+                                        c_void, c_void, c_void]
     lib.nbk_bucket_fine_f64.restype = ctypes.c_int
     lib.nbk_bucket_fine_f64.argtypes = [c_void, c_void, c_i64, c_i64_p,
                                         c_f64_p, ctypes.c_int, c_void,
