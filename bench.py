@@ -428,28 +428,169 @@ def main():
         dist.destroy_process_group()
 
 
-def run_cpu_baseline(cfg):
-    """Time the CPU oracle (the restated reference algorithm) on a
-    bounded sample of the same workload shape: ~1 particle/cell like C4,
-    sized for tens of CPU-seconds."""
-    from oracle import fftpower_oracle
-    n = int(2e7)
-    nmesh = 256
-    box = cfg['box'] * nmesh / cfg['nmesh']
+# ---- 8-core CPU baseline at the REAL mesh config -----------------------
+# The oracle (the restated reference algorithm) slab-decomposed across
+# fork workers, mirroring the reference's MPI design: paint into a
+# shared-memory mesh with disjoint x-slab ownership, one full-size
+# rfftn (scipy, workers=CORES), compensation + |c|^2 V, then
+# project_to_basis partial sums per x-range.  Paint is timed on a
+# bounded particle sample and scaled linearly in N (paint is
+# embarrassingly linear per particle; FFT/compensation/power/binning
+# are measured AT THE FULL mesh, never extrapolated).
+
+_BASE = {}          # fork-inherited (copy-on-write) worker inputs
+
+
+def _baseline_paint_worker(r):
+    import numpy as np
+    from multiprocessing import shared_memory
+    from oracle.paint import (_cic_offsets_weights,
+                              _centered_offsets_weights, _tsc_w, _pcs_w)
+    nmesh = _BASE['nmesh']
+    cores = _BASE['cores']
+    nx_l = nmesh // cores
+    lo, hi = r * nx_l, (r + 1) * nx_l
+    shm = shared_memory.SharedMemory(name=_BASE['shm'])
+    mesh = numpy.ndarray((nmesh, nmesh, nmesh), dtype='f8',
+                         buffer=shm.buf)
+    slab = mesh[lo:hi]
+    H = _BASE['box'] / nmesh
+    pos = _BASE['pos'][_BASE['parts'][r]]
+    resampler = _BASE['resampler']
+    CH = 1 << 22
+    for s in range(0, len(pos), CH):
+        u = pos[s:s + CH] / H
+        if resampler == 'cic':
+            gen = _cic_offsets_weights(u)
+        elif resampler == 'tsc':
+            gen = _centered_offsets_weights(u, (-1, 0, 1), _tsc_w, True)
+        else:
+            gen = _centered_offsets_weights(u, (-1, 0, 1, 2), _pcs_w,
+                                            False)
+        for ix, iy, iz, w in gen:
+            ix = np.remainder(ix.astype('i8'), nmesh)
+            sel = (ix >= lo) & (ix < hi)      # own only this slab's rows
+            iy = np.remainder(iy.astype('i8')[sel], nmesh)
+            iz = np.remainder(iz.astype('i8')[sel], nmesh)
+            np.add.at(slab, (ix[sel] - lo, iy, iz), w[sel])
+    shm.close()
+    return r
+
+
+def _baseline_bin_worker(r):
+    import numpy as np
+    from oracle.fftpower import project_to_basis
+    lo, hi = _BASE['bin_ranges'][r]
+    coords = _BASE['coords']
+    sub = [coords[0][lo:hi], coords[1], coords[2]]
+    return project_to_basis(_BASE['p3d'][lo:hi], _BASE['geom'],
+                            _BASE['edges'], los=(0, 0, 1), coords=sub,
+                            _return_sums=True)
+
+
+def run_cpu_baseline(cfg, cores=8):
+    """Reference CPU path (restated oracle) at the benched mesh size on
+    `cores` host cores; returns particles/s for the full workload with
+    only the paint linearly rescaled from a bounded particle sample."""
+    import multiprocessing as mp
+    from multiprocessing import shared_memory
+    import scipy.fft
+    from oracle.mesh import MeshGeometry, complex_coords
+    from oracle.fftpower import (apply_compensation, compute_3d_power,
+                                 project_to_basis)
+
+    nmesh = cfg['nmesh']
+    box = float(cfg['box'])
+    resampler = cfg['resampler']
+    n_full = cfg['particles']
+    # bounded paint sample: ~nmesh^3/8 particles (>= 1e7), never more
+    # than the real count
+    n_s = int(min(n_full, max(nmesh ** 3 // 8, int(1e7))))
+    scale_n = n_full / n_s
+
     rng = numpy.random.RandomState(42)
-    pos = rng.uniform(0, box, size=(n, 3))
-    t0 = time.time()
-    fftpower_oracle(pos, Nmesh=nmesh, BoxSize=box, mode=cfg['mode'],
-                    resampler=cfg['resampler'],
-                    interlaced=cfg['interlaced'], compensated=True)
-    dt = time.time() - t0
+    ctx = mp.get_context('fork')
+
+    geom = MeshGeometry(Nmesh=nmesh, BoxSize=box)
+    shm = shared_memory.SharedMemory(create=True,
+                                     size=nmesh ** 3 * 8)
+    try:
+        mesh = numpy.ndarray((nmesh, nmesh, nmesh), dtype='f8',
+                             buffer=shm.buf)
+        mesh[:] = 0.0
+
+        pos = rng.uniform(0, box, size=(n_s, 3))
+
+        # partition by deposit x-range (ghost overlap from the window
+        # support, as the reference's decompose does)
+        nx_l = nmesh // cores
+        b0 = numpy.floor(pos[:, 0] / (box / nmesh)).astype('i8')
+        dmin, dmax = {'cic': (0, 1), 'tsc': (-1, 1),
+                      'pcs': (-1, 2)}[resampler]
+        parts = []
+        for r in range(cores):
+            lo, hi = r * nx_l, (r + 1) * nx_l
+            owners = numpy.zeros(n_s, dtype=bool)
+            for d in range(dmin, dmax + 1):
+                c = numpy.remainder(b0 + d, nmesh)
+                owners |= (c >= lo) & (c < hi)
+            parts.append(numpy.flatnonzero(owners))
+
+        _BASE.update(nmesh=nmesh, box=box, cores=cores, shm=shm.name,
+                     pos=pos, parts=parts, resampler=resampler)
+
+        t0 = time.time()
+        with ctx.Pool(cores) as pool:
+            pool.map(_baseline_paint_worker, range(cores))
+        t_paint = time.time() - t0
+
+        t0 = time.time()
+        nbar = n_s / float(nmesh) ** 3
+        mesh /= nbar                      # 1 + delta normalization
+        p3d = scipy.fft.rfftn(mesh, workers=cores) / float(nmesh) ** 3
+        t_fft = time.time() - t0
+
+        t0 = time.time()
+        apply_compensation(p3d, geom, resampler, cfg['interlaced'])
+        p3d = compute_3d_power(p3d, p3d, geom)
+        t_pow = time.time() - t0
+
+        dk = 2 * numpy.pi / box
+        kmax = numpy.pi * nmesh / box + dk / 2
+        kedges = numpy.arange(0., kmax, dk)
+        Nmu = 1 if cfg['mode'] == '1d' else cfg.get('Nmu', 5)
+        muedges = numpy.linspace(-1, 1, Nmu + 1, endpoint=True)
+        coords = complex_coords(geom)
+        step = (nmesh + cores - 1) // cores
+        _BASE.update(p3d=p3d, geom=geom, coords=coords,
+                     edges=[kedges, muedges],
+                     bin_ranges=[(r * step, min(nmesh, (r + 1) * step))
+                                 for r in range(cores)])
+        t0 = time.time()
+        with ctx.Pool(cores) as pool:
+            partials = pool.map(_baseline_bin_worker, range(cores))
+        sums = [sum(p[i] for p in partials) for i in range(4)]
+        for a in sums:                    # mu-fold (tiny)
+            a[..., -2] += a[..., -1]
+        t_bin = time.time() - t0
+    finally:
+        _BASE.clear()
+        shm.close()
+        shm.unlink()
+
+    t_total = scale_n * t_paint + t_fft + t_pow + t_bin
     return {
-        'value': n / dt,
+        'value': n_full / t_total,
         'unit': 'particles/s',
-        'cores': 1,
+        'cores': cores,
         'kind': 'port',
-        'sample': 'oracle FFTPower, %d uniform pts / %d^3 mesh '
-                  '(same ~1 pt/cell shape), %.1fs' % (n, nmesh, dt),
+        'sample': ('oracle FFTPower at the benched config: %d^3 mesh '
+                   '(full-size FFT %.1fs + compensate/power %.1fs + '
+                   'binning %.1fs measured as-is), paint timed on '
+                   '%.1e of %.1e uniform pts (%.1fs, slab-decomposed '
+                   'x%d workers) and scaled linearly in N'
+                   % (nmesh, t_fft, t_pow, t_bin, n_s, n_full,
+                      t_paint, cores)),
     }
 
 

@@ -106,7 +106,7 @@ def kedges_unique(geom, kmax=None):
 
 
 def project_to_basis(y3d, geom, edges, los=(0, 0, 1), poles=(),
-                     coords=None, hermitian=True):
+                     coords=None, hermitian=True, _return_sums=False):
     """
     Restates fftpower.py:507-701 for a single-process field: iterate
     y-z slabs along axis 0, digitize the coordinate norm^2 and mu, apply
@@ -213,6 +213,13 @@ def project_to_basis(y3d, geom, edges, los=(0, 0, 1), poles=(),
         musum.flat += numpy.bincount(multi_index,
                                      weights=(mu * hw).ravel(),
                                      minlength=musum.size)
+
+    if _return_sums:
+        # raw per-slab-range accumulators (pre-fold): lets callers split
+        # the field along axis 0 across workers and add the partials
+        # (bench.py's 8-core cpu_baseline harness); sums are linear so
+        # the composition is exact
+        return (xsum, musum, ysum, Nsum)
 
     # fold internal mu==1 bin into the last visible bin (:674-679)
     ysum[..., -2] += ysum[..., -1]
