@@ -78,6 +78,33 @@ double* nbk_internal_twiddles(int64_t N);
 // aliasing-corrected eq. 20 forms.  Shared by the standalone
 // nbk_compensate_f64 pass and the fused nbk_power_bin_f64 so both paths
 // are bit-identical.
+// single-axis factor of the separable compensation: nbk_comp_factor is
+// the product of three of these (as sequential divides; the fast bin
+// path composes them as multiplies — a last-ulp difference covered by
+// the 1e-10 fused-vs-unfused tolerance)
+__device__ __forceinline__ double nbk_comp_factor1(int window,
+                                                   int interlaced,
+                                                   double wi) {
+    if (interlaced) {
+        const double s = 0.5 * wi;
+        const double sc = (s == 0.0) ? 1.0 : sin(s) / s;
+        const double p = (window == NBK_WINDOW_CIC) ? sc * sc
+                       : (window == NBK_WINDOW_TSC) ? sc * sc * sc
+                                                    : sc * sc * sc * sc;
+        return 1.0 / p;
+    }
+    const double s2 = sin(0.5 * wi) * sin(0.5 * wi);
+    double d;
+    if (window == NBK_WINDOW_CIC)
+        d = 1.0 - 2.0 / 3.0 * s2;
+    else if (window == NBK_WINDOW_TSC)
+        d = 1.0 - s2 + 2.0 / 15.0 * s2 * s2;
+    else
+        d = 1.0 - 4.0 / 3.0 * s2 + 2.0 / 5.0 * s2 * s2
+            - 4.0 / 315.0 * s2 * s2 * s2;
+    return 1.0 / sqrt(d);
+}
+
 __device__ __forceinline__ double nbk_comp_factor(int window, int interlaced,
                                                   const double w[3]) {
     double corr = 1.0;
