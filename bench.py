@@ -39,19 +39,40 @@ HBM_PEAK = 8.0e12          # B/s, MI355X spec (MI355X_MICROARCH.md)
 BYTES_PER_PARTICLE = {'cic': 152.0, 'tsc': 456.0, 'pcs': 1048.0}
 
 
-def gather_bpp(resampler, nmesh, n_particles):
-    """Algorithmic bytes/particle of the ownership-gather paint
-    (nbk_paint_gather_f64, the big-mesh path): each particle's 24 B
-    position row is read once per source plane of its owner tiles (2
-    planes CIC, 4 TSC/PCS at shift 0) x a (RG+span)/RG row-halo factor,
-    plus the single plain mesh write amortized per particle."""
-    planes = {'cic': 2, 'tsc': 4, 'pcs': 4}[resampler]
-    xspan = {'cic': 1, 'tsc': 3, 'pcs': 3}[resampler]
-    n1 = n2 = nmesh
-    RG = 1
-    while RG * 2 * n2 <= 20480 and n1 % (RG * 2) == 0 and RG * 2 <= n1:
-        RG *= 2
-    reads = 24.0 * planes * (RG + xspan) / float(RG)
+def pick_tile(nx_local, n1, n2, pad, span):
+    """Mirror of the kernel's nbk_pick_tile (csrc/nbk_paint.hip):
+    the (P x-planes x RG y-rows) LDS tile minimizing the particle
+    re-read factor ((P+span)/P) * ((RG+span)/RG) within 160 KiB."""
+    budget = 20480 // (n2 + pad)
+    best = None
+    bP = bRG = 1
+    P = 1
+    while P <= nx_local and P <= budget and nx_local % P == 0:
+        RG = 1
+        while RG <= n1 and P * RG <= budget and n1 % RG == 0:
+            cost = (P + span) / P * (RG + span) / RG
+            if best is None or cost < best - 1e-12:
+                best = cost
+                bP, bRG = P, RG
+            RG *= 2
+        P *= 2
+    return bP, bRG
+
+
+def gather_bpp(resampler, nmesh, n_particles, interlaced=False,
+               fused=True):
+    """Algorithmic bytes/particle of the ownership-gather paint: each
+    particle's 24 B position row is read once per (P-plane x RG-row)
+    tile whose stencil reaches it — re-read factor
+    ((P+span)/P)*((RG+span)/RG) — plus the single plain mesh/spectrum
+    write amortized per particle."""
+    sh = bool(interlaced)
+    # xhi - xlo per window/shift (csrc/nbk_paint.hip source-span logic)
+    span = {('cic', False): 1, ('cic', True): 2,
+            ('tsc', False): 3, ('tsc', True): 2,
+            ('pcs', False): 3, ('pcs', True): 4}[(resampler, sh)]
+    P, RG = pick_tile(nmesh, nmesh, nmesh, 4 if fused else 0, span)
+    reads = 24.0 * (P + span) / P * (RG + span) / RG
     mesh_bytes = 8.0 * nmesh ** 3 / float(n_particles)
     return reads + mesh_bytes
 
@@ -365,7 +386,8 @@ def main():
     paint = prof.get('paint', {'ms': 0.0, 'calls': 0, 'units': 0})
     gather = paint_is_gather(nmesh, len(pos))
     if gather:
-        bpp = gather_bpp(cfg['resampler'], nmesh, max(1, len(pos)))
+        bpp = gather_bpp(cfg['resampler'], nmesh, max(1, len(pos)),
+                         interlaced=cfg['interlaced'])
     else:
         bpp = BYTES_PER_PARTICLE[cfg['resampler']]
     algo_bytes = paint['units'] * bpp

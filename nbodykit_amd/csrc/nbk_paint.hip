@@ -368,28 +368,34 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                                   (nx_local, n1, n2/2+1) cdouble when
                                   DOFFT */
                               int64_t x0, int64_t nx_local,
-                              int RG, int xlo, int xhi, int accumulate,
+                              int RG, int P, int xlo, int xhi,
+                              int accumulate,
                               const cdouble* __restrict__ table /* W_n2 */,
                               double scale)
 {
     constexpr int SUP = (WINDOW == NBK_WINDOW_CIC) ? 2
                       : (WINDOW == NBK_WINDOW_TSC) ? 3 : 4;
-    extern __shared__ double tile[];      // RG * (n2 [+4]) doubles
+    extern __shared__ double tile[];      // P * RG * (n2 [+4]) doubles
     const int64_t sp = DOFFT ? n2 + 4 : n2;   // padded row stride
     const int64_t tiles_per_plane = n1 / RG;
-    const int64_t ix = x0 + blockIdx.x / tiles_per_plane;
+    // block owns P consecutive x-planes x RG y-rows: a source plane
+    // feeding several owned planes is READ ONCE for all of them, so the
+    // per-particle read factor drops from (1 + span) planes to
+    // (P + span)/P — the host picks P ~ RG to balance the two halo
+    // factors within the 160 KiB LDS budget
+    const int64_t px0 = x0 + (blockIdx.x / tiles_per_plane) * P;
     const int64_t r0 = (blockIdx.x % tiles_per_plane) * RG;
     const int T = blockDim.x;
     const int t = threadIdx.x;
-    const int64_t win = (int64_t)RG * sp;
+    const int64_t win = (int64_t)P * RG * sp;
 
     for (int64_t w = t; w < win; w += T) tile[w] = 0.0;
     __syncthreads();
 
     // row intervals (wrapped) whose particles can deposit into the tile
     const int64_t rspan = (int64_t)RG + (xhi - xlo);
-    for (int dp = xlo; dp <= xhi; dp++) {
-        const int64_t p = wrap_idx(ix + dp, n0);
+    for (int dp = xlo; dp <= P - 1 + xhi; dp++) {
+        const int64_t p = wrap_idx(px0 + dp, n0);
         int64_t ivals[2][2];
         int niv;
         if (rspan >= n1) {
@@ -417,7 +423,9 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                                            b0, b1, b2);
                 #pragma unroll
                 for (int dx = 0; dx < SUP; dx++) {
-                    if (wrap_idx(b0 + dx, n0) != ix) continue;
+                    int64_t pl = wrap_idx(b0 + dx, n0) - px0;
+                    if (pl < 0) pl += n0;
+                    if (pl >= P) continue;
                     #pragma unroll
                     for (int dy = 0; dy < SUP; dy++) {
                         int64_t ly = wrap_idx(b1 + dy, n1) - r0;
@@ -427,8 +435,9 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                         #pragma unroll
                         for (int dz = 0; dz < SUP; dz++) {
                             const int64_t gz = wrap_idx(b2 + dz, n2);
-                            unsafeAtomicAdd(&tile[ly * sp + gz],
-                                            wxy * w2[dz]);
+                            unsafeAtomicAdd(
+                                &tile[(pl * RG + ly) * sp + gz],
+                                wxy * w2[dz]);
                         }
                     }
                 }
@@ -439,17 +448,23 @@ __global__ void kpaint_gather(const double* __restrict__ px,
 
     if (!DOFFT) {
         // flush the exclusively-owned tile with plain stores
-        double* dst = mesh + ((ix - x0) * n1 + r0) * n2;
         if (accumulate) {
             for (int64_t w = t; w < win; w += T) {
-                const int64_t r = w / sp, z = w - r * sp;
+                const int64_t pl = w / ((int64_t)RG * sp);
+                const int64_t rem = w - pl * RG * sp;
+                const int64_t r = rem / sp, z = rem - r * sp;
                 if (z < n2 && tile[w] != 0.0)
-                    dst[r * n2 + z] += tile[w];
+                    mesh[((px0 - x0 + pl) * n1 + r0 + r) * n2 + z]
+                        += tile[w];
             }
         } else {
             for (int64_t w = t; w < win; w += T) {
-                const int64_t r = w / sp, z = w - r * sp;
-                if (z < n2) dst[r * n2 + z] = tile[w];
+                const int64_t pl = w / ((int64_t)RG * sp);
+                const int64_t rem = w - pl * RG * sp;
+                const int64_t r = rem / sp, z = rem - r * sp;
+                if (z < n2)
+                    mesh[((px0 - x0 + pl) * n1 + r0 + r) * n2 + z]
+                        = tile[w];
             }
         }
         return;
@@ -466,9 +481,12 @@ __global__ void kpaint_gather(const double* __restrict__ px,
     const int lane = t & 63;
     const int wave = t >> 6;
     const int nw = T >> 6;
-    cdouble* out = (cdouble*)mesh + ((ix - x0) * n1 + r0) * (m + 1);
 
-    for (int r = wave; r < RG; r += nw) {
+    for (int r = wave; r < P * RG; r += nw) {
+        const int64_t pl = r / RG;
+        const int64_t rr = r - pl * RG;
+        cdouble* out = (cdouble*)mesh
+            + ((px0 - x0 + pl) * n1 + r0 + rr) * (m + 1);
         cdouble* z = (cdouble*)&tile[(int64_t)r * sp];
 
         for (int j = lane; j < m; j += 64) {
@@ -540,7 +558,7 @@ __global__ void kpaint_gather(const double* __restrict__ px,
             const cdouble D = csub(Zk, cconj(Zm));
             const cdouble O = {0.5 * D.im, -0.5 * D.re};  // D * (-i/2)
             const cdouble X = cadd(E, cmul(table[k], O));
-            out[(int64_t)r * (m + 1) + k] = cscale(X, scale);
+            out[k] = cscale(X, scale);
         }
         __builtin_amdgcn_wave_barrier();
     }
@@ -793,6 +811,31 @@ extern "C" int nbk_readout_f64(const double* pos, int64_t n,
     return NBK_OK;
 }
 
+
+// pick the gather tile shape (P x-planes x RG y-rows): minimize the
+// particle re-read factor (P+sx)/P * (RG+sy)/RG (sx/sy = source span
+// beyond the tile from the window support and interlace shift) within
+// the 160 KiB LDS budget; powers of two dividing the local dims
+static void nbk_pick_tile(int64_t nx_local, int64_t n1, int64_t n2,
+                          int64_t pad, int sx, int sy,
+                          int* P_out, int* RG_out)
+{
+    const int64_t budget = 20480 / (n2 + pad);
+    double best = 1e30;
+    int bP = 1, bRG = 1;
+    for (int64_t P = 1; P <= nx_local && P <= budget; P <<= 1) {
+        if (nx_local % P) break;
+        for (int64_t RG = 1; RG <= n1 && P * RG <= budget; RG <<= 1) {
+            if (n1 % RG) break;
+            const double cost = (double)(P + sx) / (double)P
+                              * (double)(RG + sy) / (double)RG;
+            if (cost < best - 1e-12) { best = cost; bP = (int)P; bRG = (int)RG; }
+        }
+    }
+    *P_out = bP;
+    *RG_out = bRG;
+}
+
 extern "C" int nbk_paint_gather_f64(const double* pos, const double* mass,
                                     int64_t n, const int64_t nmesh[3],
                                     const double box[3],
@@ -803,11 +846,7 @@ extern "C" int nbk_paint_gather_f64(const double* pos, const double* mass,
                                     void* stream)
 {
     const int64_t n0 = nmesh[0], n1 = nmesh[1], n2 = nmesh[2];
-    // largest power-of-two row group with RG * n2 <= 20480 f64 (160 KiB)
-    int64_t RG = 1;
-    while (RG * 2 * n2 <= 20480 && (n1 % (RG * 2)) == 0 && RG * 2 <= n1)
-        RG *= 2;
-    if (RG * n2 > 20480 || (n1 % RG)) {
+    if (n2 > 20480) {
         NBK_SET_ERR("nbk_paint_gather_f64: no LDS tile for n1=%lld "
                     "n2=%lld", (long long)n1, (long long)n2);
         return NBK_ERR_UNSUPPORTED;
@@ -828,9 +867,12 @@ extern "C" int nbk_paint_gather_f64(const double* pos, const double* mass,
     }
     const int xlo = -sup + 1 - dmax;
     const int xhi = -dmin;
+    const int span = xhi - xlo;
 
-    const int64_t grid = nx_local * (n1 / RG);
-    const size_t lds = (size_t)RG * n2 * sizeof(double);
+    int P, RG;
+    nbk_pick_tile(nx_local, n1, n2, 0, span, span, &P, &RG);
+    const int64_t grid = (nx_local / P) * (n1 / RG);
+    const size_t lds = (size_t)P * RG * n2 * sizeof(double);
     hipStream_t s = (hipStream_t)stream;
     if (lds > 64 * 1024) {
         const void* fns[3] = {
@@ -849,7 +891,7 @@ extern "C" int nbk_paint_gather_f64(const double* pos, const double* mass,
                            dim3(1024), lds, s, pos, pos + n, pos + 2 * n, \
                            mass, n, n0, n1, n2, \
                            n0 / box[0], n1 / box[1], n2 / box[2], shift, \
-                           rowtab, mesh, x0, nx_local, (int)RG, xlo, xhi, \
+                           rowtab, mesh, x0, nx_local, RG, P, xlo, xhi, \
                            accumulate, (const cdouble*)nullptr, 1.0)
     if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GATHER(NBK_WINDOW_CIC);
     else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GATHER(NBK_WINDOW_TSC);
@@ -875,12 +917,7 @@ extern "C" int nbk_paint_gather_fft_f64(const double* pos,
                     "FFT length", (long long)n2);
         return NBK_ERR_UNSUPPORTED;
     }
-    // padded tile: RG * (n2 + 4) f64 within 160 KiB
-    int64_t RG = 1;
-    while (RG * 2 * (n2 + 4) <= 20480 && (n1 % (RG * 2)) == 0
-           && RG * 2 <= n1)
-        RG *= 2;
-    if (RG * (n2 + 4) > 20480 || (n1 % RG)) {
+    if (n2 + 4 > 20480) {
         NBK_SET_ERR("nbk_paint_gather_fft_f64: no LDS tile for n1=%lld "
                     "n2=%lld", (long long)n1, (long long)n2);
         return NBK_ERR_UNSUPPORTED;
@@ -904,9 +941,12 @@ extern "C" int nbk_paint_gather_fft_f64(const double* pos,
     }
     const int xlo = -sup + 1 - dmax;
     const int xhi = -dmin;
+    const int span = xhi - xlo;
 
-    const int64_t grid = nx_local * (n1 / RG);
-    const size_t lds = (size_t)RG * (n2 + 4) * sizeof(double);
+    int P, RG;
+    nbk_pick_tile(nx_local, n1, n2, 4, span, span, &P, &RG);
+    const int64_t grid = (nx_local / P) * (n1 / RG);
+    const size_t lds = (size_t)P * RG * (n2 + 4) * sizeof(double);
     hipStream_t s = (hipStream_t)stream;
     if (lds > 64 * 1024) {
         const void* fns[3] = {
@@ -925,7 +965,7 @@ extern "C" int nbk_paint_gather_fft_f64(const double* pos,
                            dim3(1024), lds, s, pos, pos + n, pos + 2 * n, \
                            mass, n, n0, n1, n2, \
                            n0 / box[0], n1 / box[1], n2 / box[2], shift, \
-                           rowtab, zspec, x0, nx_local, (int)RG, xlo, \
+                           rowtab, zspec, x0, nx_local, RG, P, xlo, \
                            xhi, 0, (const cdouble*)table, scale)
     if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GFFT(NBK_WINDOW_CIC);
     else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GFFT(NBK_WINDOW_TSC);
