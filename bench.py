@@ -547,7 +547,9 @@ def run_cpu_baseline(cfg, cores=8):
         # support, as the reference's decompose does)
         nx_l = nmesh // cores
         b0 = numpy.floor(pos[:, 0] / (box / nmesh)).astype('i8')
-        dmin, dmax = {'cic': (0, 1), 'tsc': (-1, 1),
+        # deposit-cell range relative to floor(x/H): TSC's base is
+        # floor(u+0.5)-1, so its cells span [floor(u)-1, floor(u)+2]
+        dmin, dmax = {'cic': (0, 1), 'tsc': (-1, 2),
                       'pcs': (-1, 2)}[resampler]
         parts = []
         for r in range(cores):
