@@ -85,6 +85,11 @@ def paint_is_gather(nmesh, n_local):
             and n_local >= go['sort_two_level_min_n']
             and nmesh <= 20480)
 
+# (workload, gather?) -> measured HBM bytes per benched paint launch,
+# from rocprofv3 --pmc on THIS round's kernels (profiles/r02_pmc_*);
+# None/absent = unmeasured.  Refresh whenever the paint kernel changes.
+PMC_TRAFFIC_BYTES = {}
+
 WORKLOADS = {
     # BASELINE.json configs (C1 is the CPU-oracle plumbing config)
     'c2': dict(catalog='uniform', particles=int(1e7), nmesh=256,
@@ -273,8 +278,11 @@ def gen_lognormal(n_total, nmesh, box, rank, ws, seed, bias=2.0,
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument('--gpus', type=int, default=1)
-    ap.add_argument('--steps', type=int, default=3)
-    ap.add_argument('--warmup', type=int, default=1)
+    # default K large enough that the timed region dominates the GPU
+    # activity of the run (r01 verdict item: sparse SMI sampling can
+    # miss a too-short timed window)
+    ap.add_argument('--steps', type=int, default=25)
+    ap.add_argument('--warmup', type=int, default=2)
     ap.add_argument('--workload', default='c4',
                     choices=sorted(WORKLOADS))
     ap.add_argument('--particles', type=int, default=None)
@@ -375,6 +383,29 @@ def main():
     elapsed = comm.allreduce(elapsed, op='max')
 
     prof = profiling.summary()
+
+    # one untimed pure-paint launch (real-mesh gather, no fused z-FFT)
+    # so the bench line can quote BOTH fractions (r01 verdict item 6):
+    # the fused kernel contains FFT work its byte model doesn't count
+    pure_paint = None
+    gather0 = paint_is_gather(nmesh, len(pos))
+    if rank == 0 and ws == 1 and gather0:
+        profiling.reset()
+        with set_options(paint_chunk_size=1 << 30):
+            make_mesh(cat).compute(mode='real')
+        pp = profiling.summary().get('paint')
+        if pp and pp['ms'] > 0:
+            bpp0 = gather_bpp(cfg['resampler'], nmesh,
+                              max(1, len(pos)),
+                              interlaced=cfg['interlaced'], fused=False)
+            ach0 = pp['units'] * bpp0 / (pp['ms'] * 1e-3)
+            pure_paint = {
+                'kernel': 'nbk_paint_gather_f64[%s]' % cfg['resampler'],
+                'ms_per_launch': pp['ms'] / pp['calls'],
+                'algorithmic_B_per_particle': bpp0,
+                'achieved_GBps': ach0 / 1e9,
+                'frac': ach0 / HBM_PEAK,
+            }
     profiling.disable()
 
     n_global = comm.allreduce(len(pos)) + (comm.allreduce(len(pos2))
@@ -392,9 +423,14 @@ def main():
         bpp = BYTES_PER_PARTICLE[cfg['resampler']]
     algo_bytes = paint['units'] * bpp
     achieved = algo_bytes / (paint['ms'] * 1e-3) if paint['ms'] > 0 else 0.
-    traffic = None
-    if args.traffic_bytes is not None:
-        traffic = args.traffic_bytes
+    # measured HBM bytes per paint launch from rocprofv3 --pmc
+    # (FETCH_SIZE doubled per the gfx950 wide coalesced-read half-count,
+    # plus WRITE_SIZE, collected in separate passes — see
+    # profiles/r02_pmc_summary.txt for the runs these come from);
+    # --traffic-bytes overrides for fresh measurements
+    traffic = args.traffic_bytes
+    if traffic is None:
+        traffic = PMC_TRAFFIC_BYTES.get((args.workload, gather))
 
     cpu_baseline = None
     if rank == 0 and ws == 1 and not args.no_cpu_baseline:
@@ -440,6 +476,7 @@ def main():
                 'paint_ms_per_launch': (paint['ms'] / paint['calls']
                                         if paint['calls'] else None),
                 'algorithmic_B_per_particle': bpp,
+                'pure_paint': pure_paint,
             },
             'cpu_baseline': cpu_baseline,
         }
