@@ -31,6 +31,128 @@ def _int_freqs(N):
     return numpy.fft.fftfreq(N) * N
 
 
+# ---- FFT pass dispatch -------------------------------------------------
+# Power-of-two lengths run the radix-4 LDS kernels directly.  Other EVEN
+# lengths (N <= 2048) run Bluestein's algorithm composed from the SAME
+# kernels: the chirp multiplies are torch elementwise ops and the
+# convolution transforms are nbk_fft_c_strided at M = next power of two
+# >= 2N-1 — no rocFFT/hipFFT anywhere.  This is the capability fallback
+# matching FFTW-backed pmesh's arbitrary-Nmesh support
+# (nbodykit/base/mesh.py:50); every BASELINE config is a power of two,
+# so the fallback is never on the benchmark path.  Odd lengths are
+# rejected at ParticleMesh construction (the Nyquist-as-negative
+# coordinate and Hermitian-weight conventions assume even dims).
+
+def _is_pow2_len(n):
+    return 8 <= n <= 4096 and (n & (n - 1)) == 0
+
+
+_CHIRP_CACHE = {}
+
+
+def _chirp(N):
+    """w[n] = exp(-i pi n^2 / N), exponent reduced mod 2N in integer
+    arithmetic so the angle stays small (full f64 accuracy at any N)."""
+    import torch
+    t = _CHIRP_CACHE.get(N)
+    if t is None:
+        n = numpy.arange(N, dtype='i8')
+        ang = -numpy.pi * ((n * n) % (2 * N)) / float(N)
+        t = torch.as_tensor(numpy.exp(1j * ang)).to('cuda')
+        _CHIRP_CACHE[N] = t
+    return t
+
+
+def _fft_pow2_axis1(t, sign, s):
+    """in-place kernel pass along axis 1 of a contiguous (A, M, B)
+    complex128 tensor"""
+    lib = hiplib.require()
+    A, M, B = t.shape
+    hiplib.check(lib.nbk_fft_c_strided(
+        hiplib.dptr(t), M, B, A, M * B, B, sign, s),
+        'nbk_fft_c_strided')
+
+
+def _bluestein_axis1(t, sign, s):
+    """Unnormalized DFT (sign=-1) / inverse DFT (sign=+1) along axis 1
+    of a contiguous (A, N, B) complex128 tensor; returns a new tensor.
+    X_k = w_k sum_n (x_n w_n) conj(w)_{k-n}: a circular convolution of
+    length M >= 2N-1 done with the power-of-two kernel passes."""
+    import torch
+    A, N, B = t.shape
+    if sign > 0:
+        return torch.conj(_bluestein_axis1(torch.conj(t), -1, s))
+    M = 8
+    while M < 2 * N - 1:
+        M *= 2
+    if M > 4096:
+        raise ValueError(
+            "FFT length %d unsupported (Bluestein needs a convolution "
+            "length <= 4096; use a power of two or N <= 2048)" % N)
+    w = _chirp(N)
+    a = torch.zeros((A, M, B), dtype=torch.complex128, device='cuda')
+    a[:, :N, :] = t * w.view(1, N, 1)
+    wc = torch.conj(w)
+    b = torch.zeros((1, M, 1), dtype=torch.complex128, device='cuda')
+    b[0, :N, 0] = wc
+    b[0, M - N + 1:, 0] = torch.flip(wc[1:], (0,))
+    _fft_pow2_axis1(b, -1, s)
+    _fft_pow2_axis1(a, -1, s)
+    a *= b
+    _fft_pow2_axis1(a, +1, s)          # unnormalized inverse
+    out = a[:, :N, :].clone()
+    out *= w.view(1, N, 1) * (1.0 / M)
+    return out
+
+
+def fft_axis1(t, sign, s):
+    """forward (-1) / unnormalized-inverse (+1) complex pass along axis
+    1 of a contiguous (A, N, B) complex128 tensor, in place."""
+    N = int(t.shape[1])
+    if _is_pow2_len(N):
+        _fft_pow2_axis1(t, sign, s)
+    else:
+        t.copy_(_bluestein_axis1(t, sign, s))
+
+
+def fft_r2c_z(real2d, cplx2d, scale, s):
+    """(L, nz) f64 lines -> (L, nzh) normalized half-spectra."""
+    import torch
+    lib = hiplib.require()
+    L, nz = real2d.shape
+    if _is_pow2_len(nz):
+        hiplib.check(lib.nbk_fft_r2c_z(
+            hiplib.dptr(real2d), hiplib.dptr(cplx2d), L, nz, scale, s),
+            'nbk_fft_r2c_z')
+        return
+    full = real2d.to(torch.complex128).view(L, nz, 1)
+    X = _bluestein_axis1(full, -1, s)
+    cplx2d[...] = X[:, :nz // 2 + 1, 0] * scale
+
+
+def fft_c2r_z(cplx2d, real2d, nz, s):
+    """(L, nzh) half-spectra -> (L, nz) f64 lines, unnormalized inverse
+    (FFTW/numpy c2r: the self-conjugate DC/Nyquist imaginary parts are
+    dropped)."""
+    import torch
+    lib = hiplib.require()
+    L = cplx2d.shape[0]
+    if _is_pow2_len(nz):
+        hiplib.check(lib.nbk_fft_c2r_z(
+            hiplib.dptr(cplx2d), hiplib.dptr(real2d), L, nz, s),
+            'nbk_fft_c2r_z')
+        return
+    nzh = nz // 2 + 1
+    h = cplx2d.clone()
+    h.imag[:, 0] = 0.0
+    h.imag[:, nz // 2] = 0.0
+    full = torch.empty((L, nz), dtype=torch.complex128, device='cuda')
+    full[:, :nzh] = h
+    full[:, nzh:] = torch.conj(torch.flip(h[:, 1:nz - nzh + 1], (1,)))
+    y = _bluestein_axis1(full.view(L, nz, 1), +1, s)
+    real2d[...] = y[:, :, 0].real
+
+
 # ---- pencil-transpose helpers (pure tensor ops + one alltoall; factored
 # out so the reshape logic is CPU-testable under gloo) -------------------
 
@@ -129,6 +251,13 @@ class ParticleMesh(object):
         self.dtype = numpy.dtype(dtype)
 
         ws = self.comm.size
+        for ax in range(3):
+            if int(self.Nmesh[ax]) % 2:
+                raise ValueError(
+                    "Nmesh must be even on every axis (the compressed "
+                    "half-spectrum's Nyquist-as-negative and Hermitian "
+                    "conventions assume even dims); got Nmesh[%d]=%d"
+                    % (ax, self.Nmesh[ax]))
         for ax in (0, 1):
             if ws > 1 and self.Nmesh[ax] % ws != 0:
                 raise ValueError(
@@ -178,20 +307,16 @@ def _r2c_finish(cplx, pm, s):
     distributed) over an existing z half-spectrum — the tail of
     RealField.r2c, shared with the fused paint+z-FFT path
     (source/mesh/catalog.py to_complex_field)."""
-    lib = hiplib.require()
+    hiplib.require()
     nx_l, ny, nzh = cplx.shape
-    hiplib.check(lib.nbk_fft_c_strided(
-        hiplib.dptr(cplx), ny, nzh, nx_l, ny * nzh, nzh, -1, s),
-        'nbk_fft_c_strided(y)')
+    fft_axis1(cplx.view(nx_l, ny, nzh), -1, s)          # y pass
     ws = pm.comm.size
     if ws > 1:
         cplx = transpose_x_to_y(cplx, ws, nx_l, pm.ny_local, nzh)
         n_inner = pm.ny_local * nzh
     else:
         n_inner = ny * nzh
-    hiplib.check(lib.nbk_fft_c_strided(
-        hiplib.dptr(cplx), int(pm.Nmesh[0]), n_inner, 1, 0, n_inner,
-        -1, s), 'nbk_fft_c_strided(x)')
+    fft_axis1(cplx.view(1, int(pm.Nmesh[0]), n_inner), -1, s)  # x pass
     return cplx
 
 
@@ -290,9 +415,8 @@ class RealField(_FieldBase):
         cplx = torch.empty((nx_l, ny, nzh), dtype=torch.complex128,
                            device='cuda')
         s = self._stream()
-        hiplib.check(lib.nbk_fft_r2c_z(
-            hiplib.dptr(self.value), hiplib.dptr(cplx),
-            nx_l * ny, nz, scale, s), 'nbk_fft_r2c_z')
+        fft_r2c_z(self.value.view(nx_l * ny, nz),
+                  cplx.view(nx_l * ny, nzh), scale, s)
         cplx = _r2c_finish(cplx, pm, s)
         f = ComplexField(pm, tensor=cplx)
         f.attrs.update(self.attrs)
@@ -485,23 +609,18 @@ class ComplexField(_FieldBase):
 
         cplx = self.value.clone()    # passes are in-place; keep self intact
         n_inner = cplx.shape[1] * nzh
-        hiplib.check(lib.nbk_fft_c_strided(
-            hiplib.dptr(cplx), nx, n_inner, 1, 0, n_inner, +1, s),
-            'nbk_fft_c_strided(x,inv)')
+        fft_axis1(cplx.view(1, nx, n_inner), +1, s)      # x pass
 
         if ws > 1:
             cplx = transpose_y_to_x(cplx, ws, pm.nx_local, pm.ny_local, nzh)
         nx_l = cplx.shape[0]
 
-        hiplib.check(lib.nbk_fft_c_strided(
-            hiplib.dptr(cplx), ny, nzh, nx_l, ny * nzh, nzh, +1, s),
-            'nbk_fft_c_strided(y,inv)')
+        fft_axis1(cplx.view(nx_l, ny, nzh), +1, s)       # y pass
 
         real = torch.empty((nx_l, ny, nz), dtype=torch.float64,
                            device='cuda')
-        hiplib.check(lib.nbk_fft_c2r_z(
-            hiplib.dptr(cplx), hiplib.dptr(real), nx_l * ny, nz, s),
-            'nbk_fft_c2r_z')
+        fft_c2r_z(cplx.view(nx_l * ny, nzh),
+                  real.view(nx_l * ny, nz), nz, s)
 
         if isinstance(out, RealField):
             out.value.copy_(real)

@@ -926,3 +926,46 @@ def test_at_scale_512_tsc_interlaced_spot():
     ok = numpy.isfinite(ref) & (numpy.abs(ref) > 0)
     rel = numpy.abs(got[ok] - ref[ok]) / numpy.abs(ref[ok])
     assert rel.max() < 1e-10, '512^3 interlaced parity: %g' % rel.max()
+
+
+def test_nonpow2_mesh_parity():
+    """Nmesh=96 (2^5 x 3): the Bluestein fallback composed from the
+    power-of-two kernels (pm.fft_axis1 / fft_r2c_z) against the oracle.
+    FFTW-backed pmesh accepts arbitrary Nmesh (base/mesh.py:50); every
+    BASELINE config is a power of two, so this is capability coverage."""
+    n = 50000
+    pos = numpy.random.RandomState(3).uniform(0, 300., size=(n, 3))
+    cat = ArrayCatalog({'Position': pos})
+    mesh = cat.to_mesh(Nmesh=96, BoxSize=300., dtype='f8',
+                       compensated=True, resampler='tsc')
+    r = FFTPower(mesh, mode='1d')
+    want = fftpower_oracle(pos, Nmesh=96, BoxSize=300., mode='1d',
+                           resampler='tsc', compensated=True)
+    assert_array_equal(r.power['modes'], want['modes'])
+    got = r.power['power'].real
+    ref = want['power'].real
+    ok = numpy.isfinite(ref) & (numpy.abs(ref) > 0)
+    rel = numpy.abs(got[ok] - ref[ok]) / numpy.abs(ref[ok])
+    assert rel.max() < 1e-9, 'non-pow2 parity: %g' % rel.max()
+
+
+def test_nonpow2_roundtrip_and_rfftn():
+    """96^3 r2c against numpy rfftn and the c2r round trip."""
+    import torch
+    from nbodykit_amd.pm import ParticleMesh, RealField
+    pm = ParticleMesh(BoxSize=100., Nmesh=96)
+    rng = numpy.random.RandomState(11)
+    arr = rng.normal(size=(96, 96, 96))
+    f = RealField(pm, tensor=torch.as_tensor(arr).to('cuda'))
+    c = f.r2c()
+    want = numpy.fft.rfftn(arr) / 96.0 ** 3
+    assert_allclose(c.value.cpu().numpy(), want, rtol=0, atol=1e-13)
+    back = c.c2r()
+    assert_allclose(back.value.cpu().numpy(), arr, rtol=0, atol=1e-11)
+
+
+def test_odd_nmesh_rejected():
+    from nbodykit_amd.pm import ParticleMesh
+    import pytest as _pytest
+    with _pytest.raises(ValueError, match='even'):
+        ParticleMesh(BoxSize=100., Nmesh=97)
