@@ -622,8 +622,9 @@ def test_fused_power_matches_unfused(kwargs):
     r_fused = FFTPower(mesh, **kwargs)
     noop = mesh.apply(lambda x, v: v, kind='wavenumber', mode='complex')
     r_plain = FFTPower(noop, **kwargs)
-    # tolerances allow the atomic-accumulation ordering roundoff (the
-    # two paths are per-element bit-identical; bin sums are not ordered)
+    # tolerances allow the atomic/run-merged accumulation ordering
+    # roundoff and the fast kernel's per-axis-product compensation
+    # composition (vs chained divides) — last-ulp differences
     scale = numpy.nanmax(numpy.abs(r_plain.power['power']))
     assert_allclose(r_fused.power['power'], r_plain.power['power'],
                     rtol=1e-10, atol=1e-12 * scale, equal_nan=True)
