@@ -337,10 +337,21 @@ extern "C" int nbk_fft_c_strided(double* cplx, int64_t nfft, int64_t stride,
 
     // inner-tile width: 4 columns x 16 B = 64 B contiguous per element
     // row; LDS = nfft*TI*16 B <= 64 KiB keeps 2 blocks (8 waves) per CU.
-    // (TI=8 with 128 KiB LDS was tried: full 128 B lines but 1 block/CU
-    // — measured 16.2 vs 13.0 ms per 1024^3 pass; latency hiding wins.)
-    int TI = 4;
-    while ((int64_t)nfft * TI * (int64_t)sizeof(cdouble) > 65536 && TI > 1)
+    // (TI=8 with 128 KiB LDS was tried under radix-2: full 128 B lines
+    // but 1 block/CU — measured 16.2 vs 13.0 ms per 1024^3 pass;
+    // latency hiding won.  NBK_FFT_TI re-opens the experiment for the
+    // radix-4 network.)
+    static int TI0 = 0;
+    if (!TI0) {
+        const char* e = getenv("NBK_FFT_TI");
+        TI0 = e ? atoi(e) : 4;
+        if (TI0 < 1 || TI0 > 16) TI0 = 4;
+    }
+    int TI = TI0;
+    // clamp to the 64 KiB two-blocks-per-CU budget unless the env
+    // explicitly asked for a bigger tile (gfx950 LDS tops at 160 KiB)
+    const int64_t lds_cap = (TI0 != 4) ? 160 * 1024 : 65536;
+    while ((int64_t)nfft * TI * (int64_t)sizeof(cdouble) > lds_cap && TI > 1)
         TI >>= 1;
     if (TI > n_inner) TI = (int)n_inner;
     const int tiles = (int)((n_inner + TI - 1) / TI);
