@@ -52,12 +52,23 @@ __device__ __forceinline__ int64_t wrap_idx(int64_t i, int64_t n) {
     return i < 0 ? i + n : i;
 }
 
-// signed integer frequency of global index g on an axis of size n
-// (numpy fftfreq order: Nyquist negative; meshtools.py:150-153)
+// signed frequency of global index g on an axis of size n (numpy
+// fftfreq order: Nyquist negative; meshtools.py:150-153).  The value
+// reproduces numpy's ``fftfreq(n) * n`` BIT-EXACTLY, which computes
+// (m * fl(1/n)) * n — a reciprocal multiply, NOT a division.  At
+// non-power-of-two n this is NOT the exact integer (fftfreq(96)[7]*96
+// = 7 -+ 1 ulp); the Python/oracle coordinate arrays carry that
+// rounding, and modes sitting exactly on a k-bin edge digitize by it
+// (verified: the reciprocal form matches fftfreq*n for every m at
+// n = 24/48/96/100/640/1536/2048; plain division does not).  For
+// power-of-two n both forms are the exact integer.
 __device__ __forceinline__ double freq_full(int64_t g, int64_t n) {
-    return (double)(g < n / 2 ? g : g - n);
+    const int64_t m = (g < n / 2) ? g : g - n;
+    return ((double)m * (1.0 / (double)n)) * (double)n;
 }
-// compressed (last) axis: indices 0..n/2, Nyquist stored negative
+// compressed (last) axis: indices 0..n/2, Nyquist stored negative.
+// These come from ``arange`` (+ an integer Nyquist assignment) on the
+// Python side, so they ARE exact integers at any n.
 __device__ __forceinline__ double freq_half(int64_t g, int64_t n) {
     return (double)(g == n / 2 ? -(n / 2) : g);
 }
