@@ -821,14 +821,23 @@ static void nbk_pick_tile(int64_t nx_local, int64_t n1, int64_t n2,
                           int* P_out, int* RG_out)
 {
     const int64_t budget = 20480 / (n2 + pad);
+    // NBK_PAINT_P forces the plane count (tuning); RG fills the budget
+    static int forceP = -1;
+    if (forceP < 0) {
+        const char* e = getenv("NBK_PAINT_P");
+        forceP = e ? atoi(e) : 0;
+        if (forceP < 0 || forceP > 16) forceP = 0;
+    }
     double best = 1e30;
     int bP = 1, bRG = 1;
     for (int64_t P = 1; P <= nx_local && P <= budget; P <<= 1) {
         if (nx_local % P) break;
+        if (forceP && P != forceP) continue;
         for (int64_t RG = 1; RG <= n1 && P * RG <= budget; RG <<= 1) {
             if (n1 % RG) break;
-            const double cost = (double)(P + sx) / (double)P
-                              * (double)(RG + sy) / (double)RG;
+            double cost = (double)(P + sx) / (double)P
+                        * (double)(RG + sy) / (double)RG;
+            if (forceP) cost = 1.0 / (double)RG;   // forced: max RG
             if (cost < best - 1e-12) { best = cost; bP = (int)P; bRG = (int)RG; }
         }
     }
