@@ -392,7 +392,10 @@ def main():
     if rank == 0 and ws == 1 and gather0:
         profiling.reset()
         with set_options(paint_chunk_size=1 << 30):
-            make_mesh(cat).compute(mode='real')
+            # to_real_field directly: compute('real') on a compensated
+            # mesh applies the k-space action and would route through
+            # the FUSED complex path instead of the pure gather paint
+            make_mesh(cat).to_real_field()
         pp = profiling.summary().get('paint')
         if pp and pp['ms'] > 0:
             bpp0 = gather_bpp(cfg['resampler'], nmesh,

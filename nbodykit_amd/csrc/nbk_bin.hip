@@ -611,9 +611,12 @@ static int launch_bin(const double* d1, const double* d2, BinArgs& A,
     static int64_t gcap = 0;
     if (!gcap) {
         const char* e = getenv("NBK_BIN_GRID");
-        gcap = e ? atoll(e) : 8192;   // measured best (2048/1024 lose
-                                      // parallelism faster than they
-                                      // save flush atomics)
+        gcap = e ? atoll(e) : 4096;   // 8192 was best for the old
+                                      // per-element kernel; with
+                                      // kbin_run's register merging
+                                      // fewer blocks (= fewer LDS
+                                      // flushes) measured ~1 ms better
+                                      // at C4 (r02 tune sweep)
         if (gcap < 256 || gcap > 65536) gcap = 8192;
     }
     if (g > gcap) g = gcap;
