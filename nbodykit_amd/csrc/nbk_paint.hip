@@ -355,7 +355,11 @@ __global__ void kpaint_tiled(const double* __restrict__ px,
 // kfft_r2c_z) and the z half-spectrum is written directly — the real
 // mesh never touches HBM.  The row stride is padded (+4 doubles) so the
 // butterflies of different rows land in different LDS banks.
-template <int WINDOW, bool DOFFT>
+// PT: compile-time plane count when > 0 (PT=1 restores the
+// single-plane kernel's indexing with zero multi-plane overhead — the
+// CIC path is VALU+read balanced and pays for every extra op);
+// PT=0 = runtime P.
+template <int WINDOW, bool DOFFT, int PT>
 __global__ void kpaint_gather(const double* __restrict__ px,
                               const double* __restrict__ py,
                               const double* __restrict__ pz,
@@ -375,6 +379,7 @@ __global__ void kpaint_gather(const double* __restrict__ px,
 {
     constexpr int SUP = (WINDOW == NBK_WINDOW_CIC) ? 2
                       : (WINDOW == NBK_WINDOW_TSC) ? 3 : 4;
+    if (PT > 0) P = PT;                   // compile-time plane count
     extern __shared__ double tile[];      // P * RG * (n2 [+4]) doubles
     const int64_t sp = DOFFT ? n2 + 4 : n2;   // padded row stride
     const int64_t tiles_per_plane = n1 / RG;
@@ -423,9 +428,15 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                                            b0, b1, b2);
                 #pragma unroll
                 for (int dx = 0; dx < SUP; dx++) {
-                    int64_t pl = wrap_idx(b0 + dx, n0) - px0;
-                    if (pl < 0) pl += n0;
-                    if (pl >= P) continue;
+                    int64_t pl;
+                    if (PT == 1) {
+                        if (wrap_idx(b0 + dx, n0) != px0) continue;
+                        pl = 0;
+                    } else {
+                        pl = wrap_idx(b0 + dx, n0) - px0;
+                        if (pl < 0) pl += n0;
+                        if (pl >= P) continue;
+                    }
                     #pragma unroll
                     for (int dy = 0; dy < SUP; dy++) {
                         int64_t ly = wrap_idx(b1 + dy, n1) - r0;
@@ -450,8 +461,9 @@ __global__ void kpaint_gather(const double* __restrict__ px,
         // flush the exclusively-owned tile with plain stores
         if (accumulate) {
             for (int64_t w = t; w < win; w += T) {
-                const int64_t pl = w / ((int64_t)RG * sp);
-                const int64_t rem = w - pl * RG * sp;
+                const int64_t pl = (PT == 1) ? 0
+                    : w / ((int64_t)RG * sp);
+                const int64_t rem = (PT == 1) ? w : w - pl * RG * sp;
                 const int64_t r = rem / sp, z = rem - r * sp;
                 if (z < n2 && tile[w] != 0.0)
                     mesh[((px0 - x0 + pl) * n1 + r0 + r) * n2 + z]
@@ -459,8 +471,9 @@ __global__ void kpaint_gather(const double* __restrict__ px,
             }
         } else {
             for (int64_t w = t; w < win; w += T) {
-                const int64_t pl = w / ((int64_t)RG * sp);
-                const int64_t rem = w - pl * RG * sp;
+                const int64_t pl = (PT == 1) ? 0
+                    : w / ((int64_t)RG * sp);
+                const int64_t rem = (PT == 1) ? w : w - pl * RG * sp;
                 const int64_t r = rem / sp, z = rem - r * sp;
                 if (z < n2)
                     mesh[((px0 - x0 + pl) * n1 + r0 + r) * n2 + z]
@@ -483,8 +496,8 @@ __global__ void kpaint_gather(const double* __restrict__ px,
     const int nw = T >> 6;
 
     for (int r = wave; r < P * RG; r += nw) {
-        const int64_t pl = r / RG;
-        const int64_t rr = r - pl * RG;
+        const int64_t pl = (PT == 1) ? 0 : r / RG;
+        const int64_t rr = (PT == 1) ? r : r - pl * RG;
         cdouble* out = (cdouble*)mesh
             + ((px0 - x0 + pl) * n1 + r0 + rr) * (m + 1);
         cdouble* z = (cdouble*)&tile[(int64_t)r * sp];
@@ -884,27 +897,39 @@ extern "C" int nbk_paint_gather_f64(const double* pos, const double* mass,
     const size_t lds = (size_t)P * RG * n2 * sizeof(double);
     hipStream_t s = (hipStream_t)stream;
     if (lds > 64 * 1024) {
-        const void* fns[3] = {
+        const void* fns[6] = {
             reinterpret_cast<const void*>(
-                &kpaint_gather<NBK_WINDOW_CIC, false>),
+                &kpaint_gather<NBK_WINDOW_CIC, false, 0>),
             reinterpret_cast<const void*>(
-                &kpaint_gather<NBK_WINDOW_TSC, false>),
+                &kpaint_gather<NBK_WINDOW_TSC, false, 0>),
             reinterpret_cast<const void*>(
-                &kpaint_gather<NBK_WINDOW_PCS, false>)};
-        (void)hipFuncSetAttribute(fns[window],
+                &kpaint_gather<NBK_WINDOW_PCS, false, 0>),
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_CIC, false, 1>),
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_TSC, false, 1>),
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_PCS, false, 1>)};
+        (void)hipFuncSetAttribute(fns[window + (P == 1 ? 3 : 0)],
             hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     }
-    #define NBK_LAUNCH_GATHER(W) \
-        hipLaunchKernelGGL((kpaint_gather<W, false>), \
+    #define NBK_LAUNCH_GATHER(W, PT) \
+        hipLaunchKernelGGL((kpaint_gather<W, false, PT>), \
                            dim3((uint32_t)grid), \
                            dim3(1024), lds, s, pos, pos + n, pos + 2 * n, \
                            mass, n, n0, n1, n2, \
                            n0 / box[0], n1 / box[1], n2 / box[2], shift, \
                            rowtab, mesh, x0, nx_local, RG, P, xlo, xhi, \
                            accumulate, (const cdouble*)nullptr, 1.0)
-    if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GATHER(NBK_WINDOW_CIC);
-    else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GATHER(NBK_WINDOW_TSC);
-    else NBK_LAUNCH_GATHER(NBK_WINDOW_PCS);
+    if (P == 1) {
+        if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GATHER(NBK_WINDOW_CIC, 1);
+        else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GATHER(NBK_WINDOW_TSC, 1);
+        else NBK_LAUNCH_GATHER(NBK_WINDOW_PCS, 1);
+    } else {
+        if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GATHER(NBK_WINDOW_CIC, 0);
+        else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GATHER(NBK_WINDOW_TSC, 0);
+        else NBK_LAUNCH_GATHER(NBK_WINDOW_PCS, 0);
+    }
     #undef NBK_LAUNCH_GATHER
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
@@ -958,27 +983,39 @@ extern "C" int nbk_paint_gather_fft_f64(const double* pos,
     const size_t lds = (size_t)P * RG * (n2 + 4) * sizeof(double);
     hipStream_t s = (hipStream_t)stream;
     if (lds > 64 * 1024) {
-        const void* fns[3] = {
+        const void* fns[6] = {
             reinterpret_cast<const void*>(
-                &kpaint_gather<NBK_WINDOW_CIC, true>),
+                &kpaint_gather<NBK_WINDOW_CIC, true, 0>),
             reinterpret_cast<const void*>(
-                &kpaint_gather<NBK_WINDOW_TSC, true>),
+                &kpaint_gather<NBK_WINDOW_TSC, true, 0>),
             reinterpret_cast<const void*>(
-                &kpaint_gather<NBK_WINDOW_PCS, true>)};
-        (void)hipFuncSetAttribute(fns[window],
+                &kpaint_gather<NBK_WINDOW_PCS, true, 0>),
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_CIC, true, 1>),
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_TSC, true, 1>),
+            reinterpret_cast<const void*>(
+                &kpaint_gather<NBK_WINDOW_PCS, true, 1>)};
+        (void)hipFuncSetAttribute(fns[window + (P == 1 ? 3 : 0)],
             hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     }
-    #define NBK_LAUNCH_GFFT(W) \
-        hipLaunchKernelGGL((kpaint_gather<W, true>), \
+    #define NBK_LAUNCH_GFFT(W, PT) \
+        hipLaunchKernelGGL((kpaint_gather<W, true, PT>), \
                            dim3((uint32_t)grid), \
                            dim3(1024), lds, s, pos, pos + n, pos + 2 * n, \
                            mass, n, n0, n1, n2, \
                            n0 / box[0], n1 / box[1], n2 / box[2], shift, \
                            rowtab, zspec, x0, nx_local, RG, P, xlo, \
                            xhi, 0, (const cdouble*)table, scale)
-    if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GFFT(NBK_WINDOW_CIC);
-    else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GFFT(NBK_WINDOW_TSC);
-    else NBK_LAUNCH_GFFT(NBK_WINDOW_PCS);
+    if (P == 1) {
+        if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GFFT(NBK_WINDOW_CIC, 1);
+        else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GFFT(NBK_WINDOW_TSC, 1);
+        else NBK_LAUNCH_GFFT(NBK_WINDOW_PCS, 1);
+    } else {
+        if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GFFT(NBK_WINDOW_CIC, 0);
+        else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GFFT(NBK_WINDOW_TSC, 0);
+        else NBK_LAUNCH_GFFT(NBK_WINDOW_PCS, 0);
+    }
     #undef NBK_LAUNCH_GFFT
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;

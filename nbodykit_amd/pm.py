@@ -685,10 +685,13 @@ def spectral_resample(cfield, new_pm):
     overlapping Fourier modes (pmesh ``resample`` semantics, used by
     MeshSource.compute(Nmesh=...) at nbodykit/base/mesh.py:320-330).
     Modes are dimensionless (1/N^3-normalized r2c), so a straight copy
-    preserves large-scale amplitudes.  Single-rank only for now."""
+    preserves large-scale amplitudes.
+
+    Multi-rank: both fields are y-partitioned with x and z fully local,
+    so only the kept y-rows cross ranks — each rank extracts its local
+    source rows inside the kept band, the (small) row blocks are
+    allgathered, and each rank fills the dest rows it owns."""
     import torch
-    if cfield.pm.comm.size > 1:
-        raise NotImplementedError("resampling is single-rank for now")
     out = ComplexField(new_pm)
     src = cfield.value
     dst = out.value
@@ -700,13 +703,43 @@ def spectral_resample(cfield, new_pm):
     # compressed axis: 0..hz (Nyquist of the smaller mesh included)
     zsl_s = slice(0, hz + 1)
     zsl_d = slice(0, hz + 1)
-    for xs, xd in ((slice(0, hx), slice(0, hx)),
-                   (slice(n_src[0] - hx, n_src[0]),
-                    slice(n_dst[0] - hx, n_dst[0]))):
-        for ys, yd in ((slice(0, hy), slice(0, hy)),
-                       (slice(n_src[1] - hy, n_src[1]),
-                        slice(n_dst[1] - hy, n_dst[1]))):
-            dst[xd, yd, zsl_d] = src[xs, ys, zsl_s]
+    comm = cfield.pm.comm
+    if comm.size == 1:
+        for xs, xd in ((slice(0, hx), slice(0, hx)),
+                       (slice(n_src[0] - hx, n_src[0]),
+                        slice(n_dst[0] - hx, n_dst[0]))):
+            for ys, yd in ((slice(0, hy), slice(0, hy)),
+                           (slice(n_src[1] - hy, n_src[1]),
+                            slice(n_dst[1] - hy, n_dst[1]))):
+                dst[xd, yd, zsl_d] = src[xs, ys, zsl_s]
+        out.attrs.update(cfield.attrs)
+        return out
+
+    assert comm.size == new_pm.comm.size
+    # (global source y row -> global dest y row) for the kept band
+    ymap = {}
+    for j in range(hy):
+        ymap[j] = j
+        ymap[n_src[1] - hy + j] = n_dst[1] - hy + j
+    ys0 = cfield.pm.y_start
+    nyl_s = cfield.pm.ny_local
+    yd0 = new_pm.y_start
+    nyl_d = new_pm.ny_local
+    # kept x rows of the (fully local) first axis
+    xk_s = list(range(hx)) + list(range(n_src[0] - hx, n_src[0]))
+    xk_d = list(range(hx)) + list(range(n_dst[0] - hx, n_dst[0]))
+    local = []
+    for sy in range(ys0, ys0 + nyl_s):
+        dy = ymap.get(sy)
+        if dy is None:
+            continue
+        block = src[xk_s, sy - ys0, zsl_s].cpu().numpy()
+        local.append((dy, block))
+    for chunk in comm.allgather(local):
+        for dy, block in chunk:
+            if yd0 <= dy < yd0 + nyl_d:
+                dst[xk_d, dy - yd0, zsl_d] = \
+                    torch.as_tensor(block).to(dst.device)
     out.attrs.update(cfield.attrs)
     return out
 
