@@ -41,16 +41,21 @@ BYTES_PER_PARTICLE = {'cic': 152.0, 'tsc': 456.0, 'pcs': 1048.0}
 
 def pick_tile(nx_local, n1, n2, pad, span):
     """Mirror of the kernel's nbk_pick_tile (csrc/nbk_paint.hip):
-    the (P x-planes x RG y-rows) LDS tile minimizing the particle
-    re-read factor ((P+span)/P) * ((RG+span)/RG) within 160 KiB."""
+    the (P x-planes x RG y-rows) LDS tile.  Narrow-span windows (CIC)
+    run the single-plane specialized kernel with max RG (measured
+    faster); wide-span windows minimize the particle re-read factor
+    ((P+span)/P) * ((RG+span)/RG) within 160 KiB."""
     budget = 20480 // (n2 + pad)
     best = None
     bP = bRG = 1
     P = 1
     while P <= nx_local and P <= budget and nx_local % P == 0:
+        if span <= 1 and P > 1:
+            break
         RG = 1
         while RG <= n1 and P * RG <= budget and n1 % RG == 0:
-            cost = (P + span) / P * (RG + span) / RG
+            cost = (1.0 / RG) if span <= 1 \
+                else (P + span) / P * (RG + span) / RG
             if best is None or cost < best - 1e-12:
                 best = cost
                 bP, bRG = P, RG

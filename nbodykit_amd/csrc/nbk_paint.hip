@@ -843,14 +843,23 @@ static void nbk_pick_tile(int64_t nx_local, int64_t n1, int64_t n2,
     }
     double best = 1e30;
     int bP = 1, bRG = 1;
+    // Measured policy (r02 A/B at C4/C3): narrow-span windows (CIC)
+    // run fastest single-plane (the PT=1 specialized kernel: pure
+    // paint 13.8 vs 15.7 ms, fused 19.9 vs 20.4 at C4) — the extra
+    // source-plane read is cheaper than the multi-plane bookkeeping;
+    // wide-span windows (TSC/PCS, 4 source planes single-plane) win
+    // with the balanced multi-plane tile (C3 fused 4.8 vs 5.1 ms).
+    const int prefer_single = (sx <= 1) && !forceP;
     for (int64_t P = 1; P <= nx_local && P <= budget; P <<= 1) {
         if (nx_local % P) break;
         if (forceP && P != forceP) continue;
+        if (prefer_single && P > 1) break;
         for (int64_t RG = 1; RG <= n1 && P * RG <= budget; RG <<= 1) {
             if (n1 % RG) break;
             double cost = (double)(P + sx) / (double)P
                         * (double)(RG + sy) / (double)RG;
-            if (forceP) cost = 1.0 / (double)RG;   // forced: max RG
+            if (forceP || prefer_single)
+                cost = 1.0 / (double)RG;           // max RG at this P
             if (cost < best - 1e-12) { best = cost; bP = (int)P; bRG = (int)RG; }
         }
     }
