@@ -91,9 +91,15 @@ def paint_is_gather(nmesh, n_local):
             and nmesh <= 20480)
 
 # (workload, gather?) -> measured HBM bytes per benched paint launch,
-# from rocprofv3 --pmc on THIS round's kernels (profiles/r02_pmc_*);
-# None/absent = unmeasured.  Refresh whenever the paint kernel changes.
-PMC_TRAFFIC_BYTES = {}
+# from rocprofv3 --pmc on THIS round's kernels: FETCH_SIZE doubled per
+# the gfx950 wide-read half-count + WRITE_SIZE from its own pass
+# (profiles/r02_pmc_summary.txt documents the collection; raw CSVs in
+# gpurun_out/pmc_final_*).  C4 fused paint: 27.77x... fetch-corrected
+# 27.77 + 8.61 write = 36.4 GB DRAM-side per launch — BELOW the
+# 59.6 B/particle algorithmic model because the second source-plane
+# re-reads are served by L2, not HBM.  Refresh whenever the paint
+# kernel changes.
+PMC_TRAFFIC_BYTES = {('c4', True): 36.38e9}
 
 WORKLOADS = {
     # BASELINE.json configs (C1 is the CPU-oracle plumbing config)
@@ -477,10 +483,8 @@ def main():
                 'unit': 'GB/s',
                 'frac': achieved / HBM_PEAK,
                 'traffic': traffic,
-                'kernel': ('nbk_paint_gather_fft_f64[%s]'
-                           if gather and ws == 1 else
-                           'nbk_paint_gather_f64[%s]' if gather else
-                           'nbk_paint_f64[%s]') % cfg['resampler'],
+                'kernel': ('nbk_paint_gather_fft_f64[%s]' if gather
+                           else 'nbk_paint_f64[%s]') % cfg['resampler'],
                 'paint_ms_per_launch': (paint['ms'] / paint['calls']
                                         if paint['calls'] else None),
                 'algorithmic_B_per_particle': bpp,
