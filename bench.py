@@ -94,7 +94,7 @@ def paint_is_gather(nmesh, n_local):
 # from rocprofv3 --pmc on THIS round's kernels: FETCH_SIZE doubled per
 # the gfx950 wide-read half-count + WRITE_SIZE from its own pass
 # (profiles/r02_pmc_summary.txt documents the collection; raw CSVs in
-# gpurun_out/pmc_final_*).  C4 fused paint: 27.77x... fetch-corrected
+# gpurun_out/pmc_final_*).  C4 fused paint: x2-corrected fetch
 # 27.77 + 8.61 write = 36.4 GB DRAM-side per launch — BELOW the
 # 59.6 B/particle algorithmic model because the second source-plane
 # re-reads are served by L2, not HBM.  Refresh whenever the paint
