@@ -324,7 +324,12 @@ __global__ void kbin_run(const double* __restrict__ data,
 
     const int64_t n3[3] = {A.n0, A.n1, A.n2};
     const int64_t nlines = A.d0 * A.d1;
-    constexpr int NL = 8;
+    // NL=32 lines per group: at d2 ~= n2/2+1 (513 for 1024^3) a group
+    // holds NL*d2 elements, and with per-thread runs of R the active
+    // thread count is NL*d2/R — NL=8 left half a 1024-thread block
+    // IDLE at C4 (4104 elements / 8 = 513 workers).  32 lines = 16416
+    // elements = 2 full base iterations for every thread.
+    constexpr int NL = 32;
     constexpr int R = 8;
     __shared__ double lsxy[NL], ldot[NL], lc1[NL], lc2[NL];
     __shared__ unsigned char lz0[NL];
@@ -603,7 +608,10 @@ static int launch_bin(const double* d1, const double* d2, BinArgs& A,
     const size_t lds_bytes = ((size_t)NB * nfields + nx_edges + nmu_edges)
                              * sizeof(double);
     const int64_t nlines = A.d0 * A.d1;
-    const int64_t ngroups = (nlines + 7) / 8;     // NL = 8 in the kernel
+    const int64_t ngroups = (nlines + 31) / 32;   // NL in kbin_run (the
+                                                  // legacy kbin uses 8 —
+                                                  // grid is capped far
+                                                  // below either)
     int64_t g = ngroups;
     // bounded: each block's LDS histogram flush costs NB*nfields global
     // atomics, and the atomic pipe runs at ~25 G op/s (count_probe) —
