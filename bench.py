@@ -489,6 +489,13 @@ def main():
                                         if paint['calls'] else None),
                 'algorithmic_B_per_particle': bpp,
                 'pure_paint': pure_paint,
+                # mesh-cells/s FFT'd, per axis pass, through the y+x
+                # strided passes (HIP events; units = Nmesh^3 per pass;
+                # the z pass runs fused inside the paint kernel)
+                'fft_cells_per_s': (
+                    prof['fft_strided']['units']
+                    / (prof['fft_strided']['ms'] * 1e-3)
+                    if prof.get('fft_strided', {}).get('ms') else None),
             },
             'cpu_baseline': cpu_baseline,
         }

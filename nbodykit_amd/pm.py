@@ -307,16 +307,20 @@ def _r2c_finish(cplx, pm, s):
     distributed) over an existing z half-spectrum — the tail of
     RealField.r2c, shared with the fused paint+z-FFT path
     (source/mesh/catalog.py to_complex_field)."""
+    from nbodykit_amd import profiling
     hiplib.require()
     nx_l, ny, nzh = cplx.shape
-    fft_axis1(cplx.view(nx_l, ny, nzh), -1, s)          # y pass
+    ncells = int(pm.Nmesh[0]) * int(pm.Nmesh[1]) * int(pm.Nmesh[2])
+    with profiling.collect('fft_strided', ncells):
+        fft_axis1(cplx.view(nx_l, ny, nzh), -1, s)      # y pass
     ws = pm.comm.size
     if ws > 1:
         cplx = transpose_x_to_y(cplx, ws, nx_l, pm.ny_local, nzh)
         n_inner = pm.ny_local * nzh
     else:
         n_inner = ny * nzh
-    fft_axis1(cplx.view(1, int(pm.Nmesh[0]), n_inner), -1, s)  # x pass
+    with profiling.collect('fft_strided', ncells):
+        fft_axis1(cplx.view(1, int(pm.Nmesh[0]), n_inner), -1, s)  # x
     return cplx
 
 
