@@ -970,3 +970,34 @@ def test_odd_nmesh_rejected():
     import pytest as _pytest
     with _pytest.raises(ValueError, match='even'):
         ParticleMesh(BoxSize=100., Nmesh=97)
+
+
+@pytest.mark.timeout(600)
+def test_pcs_midscale_parity():
+    """PCS (support 4, 64 deposits/particle) at a 256^3 mesh with the
+    gather path engaged — the widest window was previously only
+    parity-tested on toy meshes."""
+    n = int(5e5)
+    pos = numpy.random.RandomState(13).uniform(0, 750., size=(n, 3))
+    cat = ArrayCatalog({'Position': pos})
+    with set_options(sort_min_n=100000, sort_two_level_min_n=100000,
+                     sort_two_level_min_cells=1 << 23):
+        mesh = cat.to_mesh(Nmesh=256, BoxSize=750., dtype='f8',
+                           compensated=True, resampler='pcs')
+        assert mesh.to_complex_field() is not NotImplemented
+        r = FFTPower(mesh, mode='2d', Nmu=4, poles=[0, 2])
+    want = fftpower_oracle(pos, Nmesh=256, BoxSize=750., mode='2d',
+                           Nmu=4, poles=[0, 2], resampler='pcs',
+                           compensated=True)
+    assert_array_equal(r.power['modes'], want['modes'])
+    got = numpy.nan_to_num(r.power['power'].real)
+    ref = numpy.nan_to_num(want['power'].real)
+    ok = numpy.isfinite(ref) & (numpy.abs(ref) > 0)
+    rel = numpy.abs(got[ok] - ref[ok]) / numpy.abs(ref[ok])
+    assert rel.max() < 1e-9, 'PCS midscale parity: %g' % rel.max()
+    for ell in (0, 2):
+        g = r.poles['power_%d' % ell].real
+        f = want['poles'][ell].real
+        ok = numpy.isfinite(f) & (numpy.abs(f) > 0)
+        rel = numpy.abs(g[ok] - f[ok]) / numpy.abs(f[ok])
+        assert rel.max() < 1e-9, 'PCS pole %d: %g' % (ell, rel.max())
