@@ -64,35 +64,6 @@ __device__ __forceinline__ int bitrev(int j, int bits) {
     return (int)(__brev((unsigned)j) >> (32 - bits));
 }
 
-// in-place radix-2 DIT stages over an LDS buffer holding TI interleaved
-// columns: element (j, c) at buf[j*TI + c].  table = W_{2*m} device table
-// (so stage twiddles are W_len^p = table[2*p*(m/len)]).  INV: conjugate.
-template <bool INV>
-__device__ void lds_fft(cdouble* buf, int m, int bits, int TI,
-                        const cdouble* __restrict__ table) {
-    const int T = blockDim.x;
-    const int tid = threadIdx.x;
-    for (int len = 2; len <= m; len <<= 1) {
-        const int half = len >> 1;
-        const int tw = m / len;              // W_m exponent step
-        for (int w = tid; w < (m >> 1) * TI; w += T) {
-            const int c = w % TI;
-            const int j = w / TI;
-            const int grp = j / half;
-            const int pos = j % half;
-            const int i0 = (grp * len + pos) * TI + c;
-            const int i1 = i0 + half * TI;
-            cdouble wv = table[2 * pos * tw];   // W_m^{pos*tw} from W_{2m}
-            if (INV) wv.im = -wv.im;
-            const cdouble u = buf[i0];
-            const cdouble v = cmul(buf[i1], wv);
-            buf[i0] = cadd(u, v);
-            buf[i1] = csub(u, v);
-        }
-        __syncthreads();
-    }
-}
-
 // Radix-4 variant: pairs of radix-2 DIT stages fused into one 4-point
 // butterfly over the SAME bit-reversed input order, halving the
 // __syncthreads count and LDS round trips (the strided passes are
