@@ -10,9 +10,9 @@
 //   nbk_fft_c_strided — in-place complex pass along a strided axis
 // with the RCCL alltoall pencil transpose between ranks (Python side).
 //
-// Each line is transformed in LDS by an in-place radix-2
-// decimation-in-time network: coalesced natural-order global loads are
-// scattered bit-reversed into LDS, log2(N) butterfly stages follow, and
+// Each line is transformed in LDS by an in-place radix-4 DIT network
+// (fused radix-2 pairs): coalesced natural-order global loads are
+// scattered bit-reversed into LDS, log4(N) butterfly stages follow, and
 // the natural-order result is stored back coalesced.  Twiddles come from
 // a cached per-length device table (W_N^j, j = 0..N/2) computed once on
 // host; stage twiddles index it as W_len^p = W_N^{p*N/len}.
