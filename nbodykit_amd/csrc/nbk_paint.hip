@@ -390,6 +390,9 @@ __global__ void kpaint_gather(const double* __restrict__ px,
     // factors within the 160 KiB LDS budget
     const int64_t px0 = x0 + (blockIdx.x / tiles_per_plane) * P;
     const int64_t r0 = (blockIdx.x % tiles_per_plane) * RG;
+    // log2 of the packed-complex z length (m = n2/2), for the DOFFT
+    // bit-reversed deposit slots
+    const int zbits = DOFFT ? (31 - __clz((unsigned)(n2 >> 1))) : 0;
     const int T = blockDim.x;
     const int t = threadIdx.x;
     const int64_t win = (int64_t)P * RG * sp;
@@ -446,8 +449,20 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                         #pragma unroll
                         for (int dz = 0; dz < SUP; dz++) {
                             const int64_t gz = wrap_idx(b2 + dz, n2);
+                            // DOFFT: deposit the packed-complex pair
+                            // (gz>>1) at its BIT-REVERSED slot — the
+                            // in-tile DIT network then needs no swap
+                            // pass and the untwiddle reads natural
+                            // order (bit-reversed LDS reads are
+                            // 32-bank conflicted); one __brev per
+                            // deposit, compile-time gated
+                            const int64_t zslot = DOFFT
+                                ? ((int64_t)nbk_bitrev(
+                                       (int)(gz >> 1), zbits) << 1)
+                                  | (gz & 1)
+                                : gz;
                             unsafeAtomicAdd(
-                                &tile[(pl * RG + ly) * sp + gz],
+                                &tile[(pl * RG + ly) * sp + zslot],
                                 wxy * w2[dz]);
                         }
                     }
@@ -502,44 +517,15 @@ __global__ void kpaint_gather(const double* __restrict__ px,
             + ((px0 - x0 + pl) * n1 + r0 + rr) * (m + 1);
         cdouble* z = (cdouble*)&tile[(int64_t)r * sp];
 
-        // fused radix-4 DIF stages: NATURAL-order input (as deposited),
-        // bit-reversed output — kills the whole bit-reversal swap pass
-        // the DIT network needed.  Two radix-2 DIF stages (len, len/2)
-        // fold into one 4-point butterfly; odd log2(m) ends on a
-        // multiply-free radix-2 stage.  Validated element-exact
-        // against numpy fft for m = 4..2048.
+        // fused radix-4 DIT stages over the ALREADY BIT-REVERSED tile
+        // (the deposits wrote packed-complex pairs at bit-reversed
+        // slots): no swap pass, natural-order output, conflict-free
+        // untwiddle reads.  Algebra as lds_fft4 in nbk_fft.hip,
+        // validated element-exact against numpy; odd log2(m) runs one
+        // multiply-free radix-2 stage first.
         {
-            int ln = m;
-            while (ln >= 4) {
-                const int L = ln >> 1;
-                const int h = L >> 1;
-                const int tw = m / L;              // W_ln^p = table[p*tw]
-                for (int q = lane; q < (m >> 2); q += 64) {
-                    const int blk = q / h;
-                    const int p = q - blk * h;
-                    const int i0 = blk * ln + p;
-                    const cdouble w1 = table[p * tw];
-                    const cdouble w2 = table[2 * p * tw];
-                    const cdouble x0 = z[i0];
-                    const cdouble x1 = z[i0 + h];
-                    const cdouble x2 = z[i0 + 2 * h];
-                    const cdouble x3 = z[i0 + 3 * h];
-                    const cdouble a0 = cadd(x0, x2);
-                    const cdouble a2 = cmul(csub(x0, x2), w1);
-                    const cdouble a1 = cadd(x1, x3);
-                    const cdouble d3 = csub(x1, x3);
-                    // (x1-x3) * (-i*w1)
-                    const cdouble mi = {d3.im, -d3.re};
-                    const cdouble a3 = cmul(mi, w1);
-                    z[i0] = cadd(a0, a1);
-                    z[i0 + h] = cmul(csub(a0, a1), w2);
-                    z[i0 + 2 * h] = cadd(a2, a3);
-                    z[i0 + 3 * h] = cmul(csub(a2, a3), w2);
-                }
-                __builtin_amdgcn_wave_barrier();
-                ln >>= 2;
-            }
-            if (ln == 2) {
+            int len = 2;
+            if (bits & 1) {
                 for (int q = lane; q < (m >> 1); q += 64) {
                     const int i0 = 2 * q;
                     const cdouble u = z[i0];
@@ -548,15 +534,45 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                     z[i0 + 1] = csub(u, v);
                 }
                 __builtin_amdgcn_wave_barrier();
+                len = 4;
+            }
+            for (; 2 * len <= m; len <<= 2) {
+                const int h = len >> 1;
+                const int tw = m / len;
+                for (int q = lane; q < (m >> 2); q += 64) {
+                    const int grp = q / h;
+                    const int pos = q - grp * h;
+                    const int base = grp * (len << 1) + pos;
+                    const cdouble w1 = table[pos * tw];
+                    const cdouble w2 = table[2 * pos * tw];
+                    const cdouble w3 = cmul(w1, w2);
+                    const cdouble x0 = z[base];
+                    const cdouble x1 = z[base + h];
+                    const cdouble x2 = z[base + 2 * h];
+                    const cdouble x3 = z[base + 3 * h];
+                    const cdouble b1 = cmul(x2, w1);
+                    const cdouble b2 = cmul(x1, w2);
+                    const cdouble b3 = cmul(x3, w3);
+                    const cdouble e0 = cadd(x0, b2);
+                    const cdouble e1 = csub(x0, b2);
+                    const cdouble o0 = cadd(b1, b3);
+                    const cdouble o1 = csub(b1, b3);
+                    z[base] = cadd(e0, o0);
+                    z[base + 2 * h] = csub(e0, o0);
+                    const cdouble io1 = {o1.im, -o1.re};   // -i*o1
+                    z[base + h] = cadd(e1, io1);
+                    z[base + 3 * h] = csub(e1, io1);
+                }
+                __builtin_amdgcn_wave_barrier();
             }
         }
 
         // untwiddle split: X[k] = E[k] + W_n2^k O[k], k = 0..m, written
-        // straight to the z half-spectrum; the packed spectrum sits in
-        // bit-reversed order after the DIF network
+        // straight to the z half-spectrum (natural order — no bank
+        // conflicts)
         for (int k = lane; k <= m; k += 64) {
-            const cdouble Zk = z[nbk_bitrev(k == m ? 0 : k, bits)];
-            const cdouble Zm = z[nbk_bitrev((m - k) % m, bits)];
+            const cdouble Zk = z[k == m ? 0 : k];
+            const cdouble Zm = z[(m - k) % m];
             const cdouble E = cscale(cadd(Zk, cconj(Zm)), 0.5);
             const cdouble D = csub(Zk, cconj(Zm));
             const cdouble O = {0.5 * D.im, -0.5 * D.re};  // D * (-i/2)
