@@ -390,9 +390,6 @@ __global__ void kpaint_gather(const double* __restrict__ px,
     // factors within the 160 KiB LDS budget
     const int64_t px0 = x0 + (blockIdx.x / tiles_per_plane) * P;
     const int64_t r0 = (blockIdx.x % tiles_per_plane) * RG;
-    // log2 of the packed-complex z length (m = n2/2), for the DOFFT
-    // bit-reversed deposit slots
-    const int zbits = DOFFT ? (31 - __clz((unsigned)(n2 >> 1))) : 0;
     const int T = blockDim.x;
     const int t = threadIdx.x;
     const int64_t win = (int64_t)P * RG * sp;
@@ -449,20 +446,8 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                         #pragma unroll
                         for (int dz = 0; dz < SUP; dz++) {
                             const int64_t gz = wrap_idx(b2 + dz, n2);
-                            // DOFFT: deposit the packed-complex pair
-                            // (gz>>1) at its BIT-REVERSED slot — the
-                            // in-tile DIT network then needs no swap
-                            // pass and the untwiddle reads natural
-                            // order (bit-reversed LDS reads are
-                            // 32-bank conflicted); one __brev per
-                            // deposit, compile-time gated
-                            const int64_t zslot = DOFFT
-                                ? ((int64_t)nbk_bitrev(
-                                       (int)(gz >> 1), zbits) << 1)
-                                  | (gz & 1)
-                                : gz;
                             unsafeAtomicAdd(
-                                &tile[(pl * RG + ly) * sp + zslot],
+                                &tile[(pl * RG + ly) * sp + gz],
                                 wxy * w2[dz]);
                         }
                     }
@@ -517,15 +502,25 @@ __global__ void kpaint_gather(const double* __restrict__ px,
             + ((px0 - x0 + pl) * n1 + r0 + rr) * (m + 1);
         cdouble* z = (cdouble*)&tile[(int64_t)r * sp];
 
-        // fused radix-4 DIT stages over the ALREADY BIT-REVERSED tile
-        // (the deposits wrote packed-complex pairs at bit-reversed
-        // slots): no swap pass, natural-order output, conflict-free
-        // untwiddle reads.  Algebra as lds_fft4 in nbk_fft.hip,
-        // validated element-exact against numpy; odd log2(m) runs one
-        // multiply-free radix-2 stage first.
+        for (int j = lane; j < m; j += 64) {
+            const int jr = nbk_bitrev(j, bits);
+            if (j < jr) {
+                const cdouble a = z[j];
+                z[j] = z[jr];
+                z[jr] = a;
+            }
+        }
+        __builtin_amdgcn_wave_barrier();
+
+        // fused radix-4 stages (same bit-reversed order; pairs of
+        // radix-2 stages become one 4-point butterfly — half the LDS
+        // round trips; algebra as lds_fft4 in nbk_fft.hip, validated
+        // element-exact against numpy).  Odd log2(m): one multiply-free
+        // radix-2 stage first.
         {
+            const int fbits = 31 - __clz((unsigned)m);
             int len = 2;
-            if (bits & 1) {
+            if (fbits & 1) {
                 for (int q = lane; q < (m >> 1); q += 64) {
                     const int i0 = 2 * q;
                     const cdouble u = z[i0];
@@ -568,8 +563,7 @@ __global__ void kpaint_gather(const double* __restrict__ px,
         }
 
         // untwiddle split: X[k] = E[k] + W_n2^k O[k], k = 0..m, written
-        // straight to the z half-spectrum (natural order — no bank
-        // conflicts)
+        // straight to the z half-spectrum
         for (int k = lane; k <= m; k += 64) {
             const cdouble Zk = z[k == m ? 0 : k];
             const cdouble Zm = z[(m - k) % m];
