@@ -326,11 +326,13 @@ __global__ void kbin_run(const double* __restrict__ data,
     const int64_t nlines = A.d0 * A.d1;
     // NL=32 lines per group: at d2 ~= n2/2+1 (513 for 1024^3) a group
     // holds NL*d2 elements, and with per-thread runs of R the active
-    // thread count is NL*d2/R — NL=8 left half a 1024-thread block
-    // IDLE at C4 (4104 elements / 8 = 513 workers).  32 lines = 16416
-    // elements = 2 full base iterations for every thread.
+    // thread count is NL*d2/R — NL=8/R=8 left half a 1024-thread block
+    // IDLE at C4 (4104/8 = 513 workers).  NL=32/R=16 gives 16416/16 =
+    // 1026 active threads AND halves the flush rate (measured: R=16
+    // -2 ms/step at C4 over R=8; R=32 would idle half the block
+    // again).
     constexpr int NL = 32;
-    constexpr int R = 8;
+    constexpr int R = 16;
     __shared__ double lsxy[NL], ldot[NL], lc1[NL], lc2[NL];
     __shared__ unsigned char lz0[NL];
     const int64_t ngroups = (nlines + NL - 1) / NL;
