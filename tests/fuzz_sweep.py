@@ -1,0 +1,78 @@
+"""One-off extended fuzz sweep (GPU): many random FFTPower + FFTCorr
+configs vs the oracle.  The committed test covers 8 seeds; this tool
+sweeps more for bug hunting.  Usage: python tests/fuzz_sweep.py [N]"""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy
+
+from nbodykit_amd.lab import ArrayCatalog, FFTPower, FFTCorr
+from oracle import fftpower_oracle
+from oracle.fftpower import fftcorr_oracle
+
+nseeds = int(sys.argv[1]) if len(sys.argv) > 1 else 64
+bad = 0
+for seed in range(nseeds):
+    rng = numpy.random.RandomState(5000 + seed)
+    nmesh = int(rng.choice([32, 48, 64, 96, 128, 160]))
+    box = float(rng.uniform(100., 1000.))
+    window = str(rng.choice(['cic', 'tsc', 'pcs']))
+    interlaced = bool(rng.randint(2))
+    compensated = bool(rng.randint(2))
+    corr = bool(seed % 4 == 3)
+    mode = str(rng.choice(['1d', '2d']))
+    Nmu = int(rng.choice([3, 5]))
+    poles = [0, 2] if rng.randint(2) else []
+    kmin = float(rng.choice([0.0, 0.02]))
+    n = int(rng.randint(20000, 120000))
+    pos = rng.uniform(0, box, size=(n, 3))
+    weight = rng.uniform(0.5, 2.0, size=n) if rng.randint(2) else None
+    cfg = dict(nmesh=nmesh, box=round(box, 1), window=window,
+               interlaced=interlaced, compensated=compensated,
+               mode=mode, poles=poles, corr=corr)
+    try:
+        cat = ArrayCatalog({'Position': pos} if weight is None
+                           else {'Position': pos, 'Weight': weight})
+        mesh = cat.to_mesh(Nmesh=nmesh, BoxSize=box, dtype='f8',
+                           compensated=compensated, resampler=window,
+                           interlaced=interlaced)
+        kw = dict(mode=mode, poles=poles)
+        if mode == '2d':
+            kw['Nmu'] = Nmu
+        if corr:
+            r = FFTCorr(mesh, **kw)
+            want = fftcorr_oracle(pos, weight=weight, Nmesh=nmesh,
+                                  BoxSize=box, resampler=window,
+                                  compensated=compensated,
+                                  interlaced=interlaced, Nmu=Nmu,
+                                  poles=poles, mode=mode)
+            got = numpy.nan_to_num(numpy.ravel(r.corr['corr'].real))
+            ref = numpy.nan_to_num(numpy.ravel(want['corr'].real))
+            modes_ok = numpy.array_equal(r.corr['modes'], want['modes'])
+        else:
+            r = FFTPower(mesh, kmin=kmin, **kw)
+            want = fftpower_oracle(pos, weight=weight, Nmesh=nmesh,
+                                   BoxSize=box, resampler=window,
+                                   compensated=compensated,
+                                   interlaced=interlaced, Nmu=Nmu,
+                                   poles=poles, kmin=kmin, mode=mode)
+            got = numpy.nan_to_num(numpy.ravel(r.power['power'].real))
+            ref = numpy.nan_to_num(numpy.ravel(want['power'].real))
+            modes_ok = numpy.array_equal(r.power['modes'],
+                                         want['modes'])
+        ok = numpy.isfinite(ref) & (numpy.abs(ref) > 0)
+        rel = (numpy.abs(got[ok] - ref[ok])
+               / numpy.abs(ref[ok])).max() if ok.any() else 0.0
+        status = 'OK' if (modes_ok and rel < 1e-9) else 'FAIL'
+        if status == 'FAIL':
+            bad += 1
+        print('seed %3d %-4s rel=%.2e modes=%s %s'
+              % (seed, status, rel, modes_ok, cfg), flush=True)
+    except Exception as e:
+        bad += 1
+        print('seed %3d EXC %r %s' % (seed, e, cfg), flush=True)
+print('RESULT:', 'FAIL %d/%d' % (bad, nseeds) if bad else
+      'ALL %d PASS' % nseeds, flush=True)
+sys.exit(1 if bad else 0)
