@@ -172,7 +172,13 @@ __global__ void kbin(const double* __restrict__ data,
         const double fy = f3[1];
         const double fz = f3[2];
 
-        const double kx = fx * A.k0x, ky = fy * A.k0y, kz = fz * A.k0z;
+        double kx = fx * A.k0x, ky = fy * A.k0y, kz = fz * A.k0z;
+        if (A.real_field) {
+            // fl(fl(f*L)/N): the Python/oracle real-coordinate rounding
+            kx /= (double)A.n0;
+            ky /= (double)A.n1;
+            kz /= (double)A.n2;
+        }
         const double k2 = kx * kx + ky * ky + kz * kz;
         // modes beyond the last edge land in the (Nx+1) overflow row,
         // which project_to_basis discards (fftpower.py:666-668 keeps
@@ -588,11 +594,17 @@ static int launch_bin(const double* d1, const double* d2, BinArgs& A,
     if (axis_map) { A.a0 = axis_map[0]; A.a1 = axis_map[1]; A.a2 = axis_map[2]; }
     else { A.a0 = 0; A.a1 = 1; A.a2 = 2; }
     // coordinate scale per axis: wavenumber 2 pi f / L for the complex
-    // field, relative position f * H for the real field
+    // field; for the REAL field the Python/oracle coordinate recipe is
+    // fl(fl(f * L) / N) — multiply by the box FIRST, divide by N in
+    // the kernel — NOT f * fl(L/N): the configuration lattice spacing
+    // equals the default r-bin width, so nearly every point sits
+    // exactly on a bin edge and the rounding chain must match
+    // bit-for-bit (fuzz seeds 51/55/63 caught the f*H form flipping
+    // r-bin membership across dozens of lattice shells)
     if (real_field) {
-        A.k0x = box[0] / nmesh[0];
-        A.k0y = box[1] / nmesh[1];
-        A.k0z = box[2] / nmesh[2];
+        A.k0x = box[0];
+        A.k0y = box[1];
+        A.k0z = box[2];
     } else {
         A.k0x = 2.0 * M_PI / box[0];
         A.k0y = 2.0 * M_PI / box[1];
