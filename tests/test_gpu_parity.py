@@ -1004,60 +1004,18 @@ def test_pcs_midscale_parity():
 
 
 @pytest.mark.timeout(900)
-@pytest.mark.parametrize('seed', range(8))
+@pytest.mark.parametrize('seed', list(range(8)) + [51, 55, 63])
 def test_random_config_fuzz(seed):
-    """Randomized config fuzz: random mesh size (incl. non-pow2), box,
-    window, interlacing, mode, poles, weights and k-edges — product vs
-    oracle on the same inputs.  Catches interaction bugs the curated
-    tests miss."""
-    rng = numpy.random.RandomState(1000 + seed)
-    nmesh = int(rng.choice([32, 48, 64, 96, 128]))
-    box = float(rng.uniform(100., 1000.))
-    window = str(rng.choice(['cic', 'tsc', 'pcs']))
-    interlaced = bool(rng.randint(2))
-    compensated = bool(rng.randint(2))
-    mode = str(rng.choice(['1d', '2d']))
-    Nmu = int(rng.choice([3, 5]))
-    poles = [0, 2] if rng.randint(2) else []
-    kmin = float(rng.choice([0.0, 0.02]))
-    dk = None if rng.randint(2) else 2.5 * 2 * numpy.pi / box
-    n = int(rng.randint(20000, 120000))
-    pos = rng.uniform(0, box, size=(n, 3))
-    weight = rng.uniform(0.5, 2.0, size=n) if rng.randint(2) else None
-
-    cat = ArrayCatalog({'Position': pos} if weight is None
-                       else {'Position': pos, 'Weight': weight})
-    mesh = cat.to_mesh(Nmesh=nmesh, BoxSize=box, dtype='f8',
-                       compensated=compensated, resampler=window,
-                       interlaced=interlaced)
-    kw = dict(mode=mode, poles=poles, kmin=kmin)
-    if dk is not None:
-        kw['dk'] = dk
-    if mode == '2d':
-        kw['Nmu'] = Nmu
-    r = FFTPower(mesh, **kw)
-    want = fftpower_oracle(pos, weight=weight, Nmesh=nmesh, BoxSize=box,
-                           resampler=window, compensated=compensated,
-                           interlaced=interlaced, Nmu=Nmu, poles=poles,
-                           kmin=kmin, mode=mode,
-                           **({'dk': dk} if dk is not None else {}))
-    assert_array_equal(r.power['modes'], want['modes']), \
-        ('config', nmesh, box, window, interlaced, compensated, mode)
-    got = numpy.nan_to_num(numpy.ravel(r.power['power'].real))
-    ref = numpy.nan_to_num(numpy.ravel(want['power'].real))
-    ok = numpy.isfinite(ref) & (numpy.abs(ref) > 0)
-    rel = numpy.abs(got[ok] - ref[ok]) / numpy.abs(ref[ok])
-    assert rel.max() < 1e-9, (
-        'fuzz parity %g at config nmesh=%d box=%.1f %s interlaced=%s '
-        'compensated=%s mode=%s poles=%s' % (
-            rel.max(), nmesh, box, window, interlaced, compensated,
-            mode, poles))
-    for ell in poles:
-        g = r.poles['power_%d' % ell].real
-        f = want['poles'][ell].real
-        okp = numpy.isfinite(f) & (numpy.abs(f) > 0)
-        relp = numpy.abs(g[okp] - f[okp]) / numpy.abs(f[okp])
-        assert relp.max() < 1e-9, ('pole', ell, relp.max())
+    """Randomized config cases vs the oracle (tests/fuzz_sweep.py):
+    mesh size incl. non-pow2, box, window, interlacing, mode, poles,
+    weights; every 4th seed is an FFTCorr.  Seeds 51/55/63 pin the
+    real-field coordinate-rounding bug the wider sweep caught (the
+    configuration lattice sits exactly on r-bin edges, so the kernel
+    must reproduce fl(fl(f*L)/N) bit-for-bit)."""
+    from tests.fuzz_sweep import run_case
+    rel, modes_ok, cfg = run_case(seed)
+    assert modes_ok, cfg
+    assert rel < 1e-9, (rel, cfg)
 
 
 @pytest.mark.timeout(900)
