@@ -78,10 +78,12 @@ def run_case(seed):
 
 def main():
     nseeds = int(sys.argv[1]) if len(sys.argv) > 1 else 64
+    gen = run_case2 if (len(sys.argv) > 2 and sys.argv[2] == 'v2') \
+        else run_case
     bad = 0
     for seed in range(nseeds):
         try:
-            rel, modes_ok, cfg = run_case(seed)
+            rel, modes_ok, cfg = gen(seed)
             status = 'OK' if (modes_ok and rel < 1e-9) else 'FAIL'
             if status == 'FAIL':
                 bad += 1
@@ -97,3 +99,74 @@ def main():
 
 if __name__ == '__main__':
     main()
+
+
+def run_case2(seed):
+    """Second-generation cases: cross power (second catalog), los along
+    any axis, Selection columns, kmax, and dk=0 unique edges join the
+    mix.  Kept separate from run_case so its pinned seeds stay stable."""
+    from nbodykit_amd.lab import ArrayCatalog, FFTPower
+    from oracle import fftpower_oracle
+
+    rng = numpy.random.RandomState(9000 + seed)
+    nmesh = int(rng.choice([32, 48, 64, 96, 128]))
+    box = float(rng.uniform(100., 1000.))
+    window = str(rng.choice(['cic', 'tsc', 'pcs']))
+    interlaced = bool(rng.randint(2))
+    compensated = bool(rng.randint(2))
+    mode = str(rng.choice(['1d', '2d']))
+    Nmu = int(rng.choice([3, 5]))
+    los = [[0, 0, 1], [0, 1, 0], [1, 0, 0]][rng.randint(3)]
+    poles = ([0, 2] if (los == [0, 0, 1] and rng.randint(2)) else [])
+    cross = bool(rng.randint(2))
+    use_sel = bool(rng.randint(2))
+    kmax = float(rng.uniform(0.3, 0.8)) if rng.randint(2) else None
+    dk0 = bool(rng.randint(4) == 0 and nmesh <= 64)
+    n = int(rng.randint(20000, 90000))
+    pos = rng.uniform(0, box, size=(n, 3))
+    sel = rng.rand(n) < 0.8 if use_sel else None
+    cfg = dict(nmesh=nmesh, box=round(box, 1), window=window,
+               interlaced=interlaced, compensated=compensated,
+               mode=mode, los=los, poles=poles, cross=cross,
+               sel=use_sel, kmax=kmax, dk0=dk0)
+
+    def make(p, s_):
+        d = {'Position': p}
+        if s_ is not None:
+            d['Selection'] = s_
+        c = ArrayCatalog(d)
+        return c.to_mesh(Nmesh=nmesh, BoxSize=box, dtype='f8',
+                         compensated=compensated, resampler=window,
+                         interlaced=interlaced)
+
+    mesh = make(pos, sel)
+    second = None
+    pos2 = None
+    if cross:
+        pos2 = numpy.random.RandomState(9500 + seed).uniform(
+            0, box, size=(n // 2, 3))
+        second = make(pos2, None)
+    kw = dict(mode=mode, poles=poles, los=los)
+    if mode == '2d':
+        kw['Nmu'] = Nmu
+    if kmax is not None:
+        kw['kmax'] = kmax
+    if dk0:
+        kw['dk'] = 0
+    r = FFTPower(mesh, second=second, **kw)
+    opos = pos[sel] if sel is not None else pos
+    want = fftpower_oracle(opos, second_position=pos2, Nmesh=nmesh,
+                           BoxSize=box, resampler=window,
+                           compensated=compensated,
+                           interlaced=interlaced, Nmu=Nmu, poles=poles,
+                           los=los, mode=mode,
+                           **(dict(kmax=kmax) if kmax is not None
+                              else {}),
+                           **(dict(dk=0) if dk0 else {}))
+    got = numpy.nan_to_num(numpy.ravel(r.power['power'].real))
+    ref = numpy.nan_to_num(numpy.ravel(want['power'].real))
+    modes_ok = numpy.array_equal(r.power['modes'], want['modes'])
+    ok = numpy.isfinite(ref) & (numpy.abs(ref) > 0)
+    rel = (numpy.abs(got[ok] - ref[ok])
+           / numpy.abs(ref[ok])).max() if ok.any() else 0.0
+    return float(rel), bool(modes_ok), cfg
