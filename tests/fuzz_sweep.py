@@ -97,8 +97,6 @@ def main():
     sys.exit(1 if bad else 0)
 
 
-if __name__ == '__main__':
-    main()
 
 
 def run_case2(seed):
@@ -170,3 +168,7 @@ def run_case2(seed):
     rel = (numpy.abs(got[ok] - ref[ok])
            / numpy.abs(ref[ok])).max() if ok.any() else 0.0
     return float(rel), bool(modes_ok), cfg
+
+
+if __name__ == '__main__':
+    main()
