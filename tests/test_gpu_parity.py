@@ -1042,3 +1042,15 @@ def test_2048_mesh_capability():
     assert chisq < 1.5, 'chi2/dof at 2048^3: %g' % chisq
     import torch
     torch.cuda.empty_cache()
+
+
+@pytest.mark.timeout(900)
+@pytest.mark.parametrize('seed', range(6))
+def test_random_config_fuzz_gen2(seed):
+    """Second-generation fuzz (tests/fuzz_sweep.run_case2): cross
+    power, los along any axis, Selection columns, kmax and dk=0 unique
+    edges.  64-seed sweep was clean on hardware (r02)."""
+    from tests.fuzz_sweep import run_case2
+    rel, modes_ok, cfg = run_case2(seed)
+    assert modes_ok, cfg
+    assert rel < 1e-9, (rel, cfg)
