@@ -78,8 +78,8 @@ def run_case(seed):
 
 def main():
     nseeds = int(sys.argv[1]) if len(sys.argv) > 1 else 64
-    gen = run_case2 if (len(sys.argv) > 2 and sys.argv[2] == 'v2') \
-        else run_case
+    variant = sys.argv[2] if len(sys.argv) > 2 else 'v1'
+    gen = {'v2': run_case2, 'v3': run_case3}.get(variant, run_case)
     bad = 0
     for seed in range(nseeds):
         try:
@@ -172,3 +172,36 @@ def run_case2(seed):
 
 if __name__ == '__main__':
     main()
+
+
+def run_case3(seed):
+    """Reconstruction fuzz: random mesh (incl. non-pow2), box, bias,
+    growth rate, smoothing and scheme — FFTRecon's displacement solve +
+    readout + shifted re-paint vs the oracle."""
+    from nbodykit_amd.lab import ArrayCatalog, FFTRecon
+    from oracle import fftrecon_oracle
+
+    rng = numpy.random.RandomState(12000 + seed)
+    nmesh = int(rng.choice([16, 24, 32, 48]))
+    box = float(rng.uniform(50., 300.))
+    bias = float(rng.uniform(1.0, 2.5))
+    f = float(rng.uniform(0.0, 0.9))
+    R = float(rng.uniform(box / 16, box / 4))
+    scheme = str(rng.choice(['LGS', 'LF2', 'LRR']))
+    nd = int(rng.randint(3000, 20000))
+    nr = int(rng.randint(6000, 40000))
+    dpos = rng.uniform(0, box, size=(nd, 3))
+    rpos = rng.uniform(0, box, size=(nr, 3))
+    cfg = dict(nmesh=nmesh, box=round(box, 1), bias=round(bias, 2),
+               f=round(f, 2), R=round(R, 1), scheme=scheme)
+
+    data = ArrayCatalog({'Position': dpos})
+    ran = ArrayCatalog({'Position': rpos})
+    recon = FFTRecon(data, ran, Nmesh=nmesh, BoxSize=box, bias=bias,
+                     f=f, R=R, scheme=scheme)
+    got = numpy.asarray(recon.compute(mode='real'))
+    want = fftrecon_oracle(dpos, rpos, Nmesh=nmesh, BoxSize=box,
+                           bias=bias, f=f, R=R, scheme=scheme)
+    scale = max(1e-30, numpy.abs(want).max())
+    rel = float(numpy.abs(got - want).max() / scale)
+    return rel, True, cfg
