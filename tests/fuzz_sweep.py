@@ -170,8 +170,6 @@ def run_case2(seed):
     return float(rel), bool(modes_ok), cfg
 
 
-if __name__ == '__main__':
-    main()
 
 
 def run_case3(seed):
@@ -205,3 +203,7 @@ def run_case3(seed):
     scale = max(1e-30, numpy.abs(want).max())
     rel = float(numpy.abs(got - want).max() / scale)
     return rel, True, cfg
+
+
+if __name__ == '__main__':
+    main()
