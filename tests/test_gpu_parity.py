@@ -1054,3 +1054,14 @@ def test_random_config_fuzz_gen2(seed):
     rel, modes_ok, cfg = run_case2(seed)
     assert modes_ok, cfg
     assert rel < 1e-9, (rel, cfg)
+
+
+@pytest.mark.timeout(900)
+@pytest.mark.parametrize('seed', range(4))
+def test_random_recon_fuzz(seed):
+    """Reconstruction fuzz (tests/fuzz_sweep.run_case3): random mesh,
+    bias, growth rate, smoothing, scheme — vs the oracle.  32-seed
+    sweep was clean on hardware (r02)."""
+    from tests.fuzz_sweep import run_case3
+    rel, _, cfg = run_case3(seed)
+    assert rel < 1e-9, (rel, cfg)
