@@ -79,7 +79,8 @@ def run_case(seed):
 def main():
     nseeds = int(sys.argv[1]) if len(sys.argv) > 1 else 64
     variant = sys.argv[2] if len(sys.argv) > 2 else 'v1'
-    gen = {'v2': run_case2, 'v3': run_case3}.get(variant, run_case)
+    gen = {'v2': run_case2, 'v3': run_case3,
+           'v4': run_case4}.get(variant, run_case)
     bad = 0
     for seed in range(nseeds):
         try:
@@ -203,6 +204,66 @@ def run_case3(seed):
     scale = max(1e-30, numpy.abs(want).max())
     rel = float(numpy.abs(got - want).max() / scale)
     return rel, True, cfg
+
+
+
+
+def run_case4(seed):
+    """Survey (FKP) fuzz: random survey geometry, mesh, windows, poles,
+    P0, dk — ConvolvedFFTPower vs the independent-Ylm oracle."""
+    import warnings
+    from nbodykit_amd.lab import (ArrayCatalog, FKPCatalog,
+                                  ConvolvedFFTPower)
+    from oracle.convpower import convpower_oracle
+
+    rng = numpy.random.RandomState(15000 + seed)
+    nmesh = int(rng.choice([32, 48, 64]))
+    lo = rng.uniform(800., 1500., size=3)
+    span = rng.uniform(150., 350., size=3)
+    ndata = int(rng.randint(1500, 6000))
+    nran = 10 * ndata
+    nbar = ndata / span.prod()
+    P0 = float(rng.choice([5e3, 1e4, 2e4]))
+    window = str(rng.choice(['cic', 'tsc']))
+    compensated = bool(rng.randint(2))
+    poles = [0, 2, 4] if rng.randint(2) else [0, 2]
+    dk = float(rng.choice([0.04, 0.05, 0.08]))
+    box = float(numpy.ceil(span.max() * 1.1 / 10) * 10)
+    center = lo + span / 2
+    cfg = dict(nmesh=nmesh, box=box, window=window,
+               compensated=compensated, poles=poles, P0=P0, dk=dk)
+
+    dpos = lo + rng.uniform(0., 1., size=(ndata, 3)) * span
+    rpos = lo + rng.uniform(0., 1., size=(nran, 3)) * span
+    data = ArrayCatalog({'Position': dpos,
+                         'NZ': numpy.full(ndata, nbar)})
+    ran = ArrayCatalog({'Position': rpos,
+                        'NZ': numpy.full(nran, nbar)})
+    cat = FKPCatalog(data, ran, P0=P0, BoxSize=box, BoxPad=0.02)
+    with warnings.catch_warnings():
+        warnings.simplefilter('ignore')
+        mesh = cat.to_mesh(Nmesh=nmesh, BoxCenter=list(center),
+                           dtype='f8', compensated=compensated,
+                           resampler=window)
+    r = ConvolvedFFTPower(mesh, poles=poles, dk=dk)
+
+    fkp_d = 1.0 / (1.0 + P0 * nbar)
+    o = convpower_oracle(dpos, rpos, poles, Nmesh=nmesh, BoxSize=box,
+                         BoxCenter=list(center),
+                         nbar_data=numpy.full(ndata, nbar),
+                         nbar_ran=numpy.full(nran, nbar),
+                         data_fkp=numpy.full(ndata, fkp_d),
+                         ran_fkp=numpy.full(nran, fkp_d),
+                         resampler=window, compensated=compensated,
+                         dk=dk)
+    modes_ok = numpy.array_equal(r.poles['modes'], o['modes'])
+    rel = 0.0
+    scale = numpy.nanmax(numpy.abs(o['power_0']))
+    for ell in poles:
+        g = numpy.nan_to_num(numpy.asarray(r.poles['power_%d' % ell]))
+        f = numpy.nan_to_num(numpy.asarray(o['power_%d' % ell]))
+        rel = max(rel, float(numpy.abs(g - f).max() / scale))
+    return rel, modes_ok, cfg
 
 
 if __name__ == '__main__':
