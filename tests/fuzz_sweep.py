@@ -81,11 +81,13 @@ def main():
     variant = sys.argv[2] if len(sys.argv) > 2 else 'v1'
     gen = {'v2': run_case2, 'v3': run_case3,
            'v4': run_case4}.get(variant, run_case)
+    # v4 poles are stored as c8 (the reference's dtype) => float32-level
+    thr = 5e-6 if variant == 'v4' else 1e-9
     bad = 0
     for seed in range(nseeds):
         try:
             rel, modes_ok, cfg = gen(seed)
-            status = 'OK' if (modes_ok and rel < 1e-9) else 'FAIL'
+            status = 'OK' if (modes_ok and rel < thr) else 'FAIL'
             if status == 'FAIL':
                 bad += 1
             print('seed %3d %-4s rel=%.2e modes=%s %s'
