@@ -250,13 +250,18 @@ def run_case4(seed):
     r = ConvolvedFFTPower(mesh, poles=poles, dk=dk)
 
     fkp_d = 1.0 / (1.0 + P0 * nbar)
+    # ConvolvedFFTPower ALWAYS applies the window compensation — it
+    # strips the mesh's actions and calls _get_compensation itself
+    # (reference convpower/fkp.py:428-434), so the mesh's own
+    # ``compensated`` flag (randomized above) must be IRRELEVANT; the
+    # oracle always compensates.
     o = convpower_oracle(dpos, rpos, poles, Nmesh=nmesh, BoxSize=box,
                          BoxCenter=list(center),
                          nbar_data=numpy.full(ndata, nbar),
                          nbar_ran=numpy.full(nran, nbar),
                          data_fkp=numpy.full(ndata, fkp_d),
                          ran_fkp=numpy.full(nran, fkp_d),
-                         resampler=window, compensated=compensated,
+                         resampler=window, compensated=True,
                          dk=dk)
     modes_ok = numpy.array_equal(r.poles['modes'], o['modes'])
     rel = 0.0
