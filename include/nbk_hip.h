@@ -101,8 +101,11 @@ int nbk_paint_sorted_f64(const double* pos, const double* mass, int64_t n,
 /* ownership-gather paint for cell-sorted input: requires the row table
  * produced by nbk_bucket_fine_f64 (rowtab[n0*n1+1]: start of each
  * (ix, iy) row in the sorted SoA arrays, sentinel n at the end).  Each
- * block owns an exclusive LDS mesh tile (RG y-rows x n2 of one x-plane)
- * and gathers from the source rows whose stencils reach it — deposits
+ * block owns an exclusive LDS mesh tile (P x-planes x RG y-rows x n2;
+ * the tile shape is picked per window span — single-plane max-RG for
+ * CIC, balanced multi-plane for TSC/PCS — within the 160 KiB LDS
+ * budget) and gathers from the source rows whose stencils reach it —
+ * deposits
  * are LDS f64 adds and the flush is plain stores, NO global atomics
  * (the global atomic pipe is the ~25 G op/s bound on the scatter
  * kernels above).  accumulate=0 overwrites the slab (fresh mesh, saves
