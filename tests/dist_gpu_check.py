@@ -181,6 +181,29 @@ def main():
     if max(e5, e5m) > 1e-10:
         failures.append('resample')
 
+    # ---- 6. skewed distribution: every particle in rank 0's slab ------
+    # post-routing the other ranks are (nearly) empty — the forced
+    # empty-rank row table of the fused path must hold up
+    n6 = 120000
+    full6 = numpy.random.RandomState(23).uniform(0, 512., size=(n6, 3))
+    full6[:, 0] *= (512. / ws) / 512.          # squeeze x into slab 0
+    lo = rank * n6 // ws
+    hi = (rank + 1) * n6 // ws
+    cat6 = ArrayCatalog({'Position': full6[lo:hi]}, comm=comm)
+    with set_options(**small_gates):
+        mesh6 = cat6.to_mesh(Nmesh=128, BoxSize=512., dtype='f8',
+                             compensated=True, resampler='cic')
+        assert mesh6.to_complex_field() is not NotImplemented
+        r6 = FFTPower(mesh6, mode='1d')
+    want6 = fftpower_oracle(full6, Nmesh=128, BoxSize=512., mode='1d',
+                            resampler='cic', compensated=True)
+    e6 = rel_err(r6.power['power'].real, want6['power'].real)
+    m6 = numpy.array_equal(r6.power['modes'], want6['modes'])
+    results['skewed'] = e6
+    log('6. skewed-slab ws=%d: rel err %.3g, modes %s' % (ws, e6, m6))
+    if e6 > RTOL or not m6:
+        failures.append('skewed')
+
     ok = not failures
     # every rank must agree
     all_ok = bool(min(comm.allgather(int(ok))))
