@@ -315,6 +315,34 @@ int nbk_power_bin_f64(const double* c1, const double* c2, double volume,
                       double* xsum, double* musum, double* Nsum,
                       double* ysum, void* stream);
 
+/* fused x-pass FFT + compensate + auto-power + binning: consumes the
+ * PRE-x-pass complex field (z and y axes already transformed, pencil
+ * transpose already applied when distributed: element (j, c) of the
+ * x-line through flattened local (y, zh) column c lives at
+ * data[j * n_inner + c]), runs the final strided x FFT per line in LDS
+ * and feeds the results straight into the project_to_basis sums — the
+ * finished complex field is never written to HBM and columns wholly
+ * beyond the last k-edge are skipped before their loads.  Equivalent to
+ * nbk_fft_c_strided(axis 0) + nbk_power_bin_f64(c1 == c2) with
+ * clear_zero semantics; bin assignment is bit-identical, the x-FFT
+ * element values are bit-identical to the unfused pass.  kedges are
+ * SQUARED k edges (as in nbk_bin_power_f64); out_sums is the contiguous
+ * [xsum|musum|Nsum|ysum] block of (nx_edges+1)*(nmu_edges+1) doubles
+ * each (ysum: nell planar re/im pairs).  nmesh[0] must be a power of
+ * two in [8, 4096]; fails with NBK_ERR_UNSUPPORTED when the histogram +
+ * FFT tile exceed the 160 KiB LDS (callers fall back to the unfused
+ * sequence). */
+int nbk_fft_x_bin_f64(const double* data, const int64_t nmesh[3],
+                      int64_t n_inner, int64_t y_off,
+                      const double box[3],
+                      int window1, int interlaced1,
+                      int clear_zero, double volume,
+                      const double* kedges, int64_t nx_edges,
+                      const double* muedges, int64_t nmu_edges,
+                      const double los[3],
+                      const int* ells, int nell,
+                      double* out_sums, void* stream);
+
 /* small helpers ------------------------------------------------------ */
 /* out[i] += a[i]  (f64, n elements) */
 int nbk_axpy_f64(double* out, const double* a, double alpha, int64_t n,

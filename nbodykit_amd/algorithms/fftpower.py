@@ -132,6 +132,30 @@ class FFTBase(object):
         attrs['shotnoise'] = Pshot
         return c1, c2, attrs
 
+    def _compute_deferred_x(self, mesh):
+        """Pre-x-pass spectrum ``(tensor, n_inner, mesh_attrs)`` for the
+        deferred-x fused FFT+binning path (nbk_fft_x_bin_f64): the z and
+        y FFT passes (+ pencil transpose when distributed) run here, the
+        final x pass runs inside the binning kernel.  Computed WITHOUT
+        the compensation action (the kernel applies it on the fly —
+        _fuse_info already certified compensation is the mesh's only
+        action).  Collective: the fused-paint gates inside
+        to_complex_field are rank-invariant and the fallback
+        (to_real_field + r2c_defer_x) is taken by all ranks together."""
+        from nbodykit_amd.pm import r2c_defer_x
+        res = mesh.to_complex_field(_defer_x=True)
+        if res is NotImplemented:
+            real = mesh.to_real_field(normalize=True)
+            tensor, n_inner = r2c_defer_x(real)
+            res = (tensor, n_inner, dict(real.attrs))
+        tensor, n_inner, mattrs = res
+        attrs = {}
+        attrs.update(self.attrs)
+        attrs.update({'N1': mattrs.get('N', 0), 'N2': mattrs.get('N', 0)})
+        attrs['shotnoise'] = mattrs.get('shotnoise', 0) \
+            if self.first is self.second else 0
+        return tensor, n_inner, attrs
+
 
 class FFTPower(FFTBase):
     """Periodic-box 1d/2d power spectrum and multipoles via FFT
@@ -180,17 +204,48 @@ class FFTPower(FFTBase):
         fuse2 = fuse1 if self.second is self.first \
             else _fuse_info(self.second)
         fused = fuse1 is not None and fuse2 is not None
-        if fused:
-            c1, c2, attrs = self._compute_complex_pair()
-            y3d = c1     # coordinate/metadata source only
-        else:
-            y3d, attrs = self._compute_3d_power(self.first, self.second)
 
         dk = self.attrs['dk']
         kmin = self.attrs['kmin']
         kmax = self.attrs['kmax']
+
+        # deferred-x path (auto power, dk>0): the final x FFT pass runs
+        # INSIDE the binning kernel (nbk_fft_x_bin_f64) — the finished
+        # complex field is never materialized.  Edge arrays must be
+        # known up front, so the dk == 0 unique-edges path stays on the
+        # standard flow.
+        defer = (fused and self.second is self.first and dk > 0
+                 and _xbin_ok(self.first, self.attrs['Nmesh']))
+        if defer:
+            kmax_eff = kmax
+            if kmax_eff is None:
+                kmax_eff = (numpy.pi * self.attrs['Nmesh'].min()
+                            / self.attrs['BoxSize'].max() + dk / 2)
+            kedges_d = numpy.arange(kmin, kmax_eff, dk)
+            n0 = int(self.attrs['Nmesh'][0])
+            poles_l = list(self.attrs['poles'])
+            nell = len(poles_l) + (0 not in poles_l)
+            defer = len(kedges_d) >= 1 and _xbin_lds_fits(
+                n0, len(kedges_d), self.attrs['Nmu'] + 1, nell)
+
+        if defer:
+            tensor, n_inner, attrs = self._compute_deferred_x(self.first)
+            pm = self.first.pm
+            Nmesh_arr = numpy.asarray(pm.Nmesh)
+            BoxSize_arr = numpy.asarray(pm.BoxSize)
+            y3d = None
+        else:
+            if fused:
+                c1, c2, attrs = self._compute_complex_pair()
+                y3d = c1     # coordinate/metadata source only
+            else:
+                y3d, attrs = self._compute_3d_power(self.first,
+                                                    self.second)
+            Nmesh_arr = y3d.Nmesh
+            BoxSize_arr = y3d.BoxSize
+
         if kmax is None:
-            kmax = numpy.pi * y3d.Nmesh.min() / y3d.BoxSize.max() + dk / 2
+            kmax = numpy.pi * Nmesh_arr.min() / BoxSize_arr.max() + dk / 2
 
         if dk > 0:
             kedges = numpy.arange(kmin, kmax, dk)
@@ -203,7 +258,12 @@ class FFTPower(FFTBase):
                                  endpoint=True)
         edges = [kedges, muedges]
         coords = [kcoords, None]
-        if fused:
+        if defer:
+            result, pole_result = _project_power_xbin(
+                tensor, pm, n_inner, fuse1,
+                volume=self.attrs['BoxSize'].prod(), edges=edges,
+                poles=self.attrs['poles'], los=self.attrs['los'])
+        elif fused:
             result, pole_result = _project_power_fused(
                 c1, c2, fuse1, fuse2,
                 volume=self.attrs['BoxSize'].prod(), edges=edges,
@@ -545,6 +605,91 @@ def _project_power_fused(c1, c2, comp1, comp2, volume, edges,
         hiplib.dptr(sums), hiplib.dptr(sums[NB:]),
         hiplib.dptr(sums[2 * NB:]), hiplib.dptr(sums[3 * NB:]),
         hiplib.cur_stream()), 'nbk_power_bin_f64')
+
+    torch.cuda.synchronize()
+    host = sums.cpu().numpy()
+    host = comm.allreduce(host)
+    return _fold_bin_sums(host, Nx, Nmu, Nell, ell_idx, do_poles,
+                          real_field=False)
+
+
+def _xbin_ok(mesh, Nmesh):
+    """True when the deferred-x fused FFT+binning kernel can serve this
+    mesh: non-interlaced, already at the target Nmesh (no spectral
+    resample), power-of-two axes in [8, 4096] (the in-LDS x-line FFT),
+    and not disabled via NBK_NO_XBIN=1.  Every term is rank-invariant,
+    so the decision is collective."""
+    import os
+    if os.environ.get('NBK_NO_XBIN', '0') == '1':
+        return False
+    if getattr(mesh, 'interlaced', False):
+        return False
+    pmN = numpy.asarray(mesh.pm.Nmesh)
+    if not numpy.array_equal(pmN, numpy.asarray(Nmesh)):
+        return False
+    for n in pmN:
+        n = int(n)
+        if n < 8 or n > 4096 or (n & (n - 1)):
+            return False
+    return True
+
+
+def _xbin_lds_fits(n0, nx_edges, nmu_edges, nell):
+    """Mirror of nbk_fft_x_bin_f64's LDS budget: histograms + edge
+    arrays + kx/compensation tables + the TI=1 FFT tile must fit the
+    gfx950 160 KiB LDS (the kernel itself would return
+    NBK_ERR_UNSUPPORTED; gate here so the standard path is taken
+    without a failed launch)."""
+    NB = (nx_edges + 1) * (nmu_edges + 1)
+    fixed = (NB * (3 + 2 * nell) + nx_edges + nmu_edges + 2 * n0) * 8
+    return fixed + n0 * 16 <= 160 * 1024
+
+
+def _project_power_xbin(tensor, pm, n_inner, comp1, volume, edges,
+                        los=[0, 0, 1], poles=[]):
+    """project_to_basis of |comp1(F_x(pre_x))|^2 V with the zero mode
+    cleared, fused into the final x-axis FFT pass (nbk_fft_x_bin_f64):
+    the finished complex field never exists in HBM and columns wholly
+    beyond the last k-edge are skipped before their loads.  Auto power
+    only.  Bin assignment is bit-identical to the unfused path (same
+    k2/mu groupings); the x-FFT element values are bit-identical to
+    nbk_fft_c_strided's."""
+    import torch
+    comm = pm.comm
+    lib = hiplib.require()
+
+    xedges, muedges = edges
+    Nx = len(xedges) - 1
+    Nmu = len(muedges) - 1
+
+    poles = list(poles)
+    do_poles = len(poles) > 0
+    _poles = [0] + sorted(poles) if 0 not in poles else sorted(poles)
+    ell_idx = [_poles.index(l) for l in poles]
+    Nell = len(_poles)
+    if any(ell < 0 for ell in _poles):
+        raise ValueError("in `project_to_basis`, multipole numbers must "
+                         "be non-negative integers")
+
+    NB = (Nx + 2) * (Nmu + 2)
+    nfields = 3 + 2 * Nell
+
+    dev = 'cuda'
+    k2edges_t = torch.as_tensor(numpy.asarray(xedges, dtype='f8') ** 2) \
+        .to(dev)
+    muedges_t = torch.as_tensor(numpy.asarray(muedges, dtype='f8')).to(dev)
+    sums = torch.zeros(nfields * NB, dtype=torch.float64, device=dev)
+
+    win1, interl1 = comp1
+    y_off = pm.y_start if comm.size > 1 else 0
+    hiplib.check(lib.nbk_fft_x_bin_f64(
+        hiplib.dptr(tensor), hiplib.i64_arr(pm.Nmesh),
+        int(n_inner), int(y_off), hiplib.f64_arr(pm.BoxSize),
+        int(win1), int(interl1), 1, float(volume),
+        hiplib.dptr(k2edges_t), len(xedges),
+        hiplib.dptr(muedges_t), len(muedges),
+        hiplib.f64_arr(los), hiplib.int_arr(_poles), Nell,
+        hiplib.dptr(sums), hiplib.cur_stream()), 'nbk_fft_x_bin_f64')
 
     torch.cuda.synchronize()
     host = sums.cpu().numpy()

@@ -258,6 +258,211 @@ int check_len(int64_t n, const char* who) {
     return NBK_OK;
 }
 
+// ---- fused x-pass FFT + compensate + power + project_to_basis --------
+// (auto power, non-interlaced — the FFTPower flagship path): the LAST
+// strided pass of the forward 3D FFT feeds each transformed x-line
+// straight into the project_to_basis sums, so the finished complex
+// field is never written to HBM and the separate binning pass never
+// reads it (~17 GB saved per 1024^3 FFTPower), and x-lines wholly
+// beyond the last k-edge (ky^2+kz^2 >= k2_last — ~21% of columns at
+// the default kmax, the part of the y-z plane outside the inscribed
+// circle) are skipped before their loads are even issued.
+//
+// Bin assignment is bit-identical to kbin_run/kbin (nbk_bin.hip):
+//   k2 = (kx*kx + ky*ky) + kz*kz      [fl(fl(kx2+ky2)+kz2)]
+//   mu = ((kx lx + ky ly) + kz lz)/|k|
+// with the x-dependent terms (kx = fl(fx*k0x), fl(kx*kx),
+// fl(kx*losx)) recomputed per element from an n0-entry LDS kx table,
+// and the compensation composed as (cx*cy)*cz per-axis products,
+// exactly the fast kbin_run composition.  The x-line FFT is the same
+// bit-reversed-load + lds_fft4 network as kfft_c_strided, so element
+// values match the unfused x-pass bit-for-bit.
+#ifndef NBK_MAX_ELL
+#define NBK_MAX_ELL 8
+#endif
+
+struct XBinArgs {
+    int64_t n0, n1, n2;        // global REAL mesh dims
+    int64_t nzh;               // n2/2 + 1
+    int64_t n_inner;           // local flattened (y, zh) column count
+    int64_t y_off;             // global y offset of the local block
+    double k0x, k0y, k0z;      // 2 pi / BoxSize
+    double losx, losy, losz;
+    int nx_edges, nmu_edges;   // edge COUNTS (bins + 1)
+    int nell;
+    int ells[NBK_MAX_ELL];
+    int win1, interl1;         // -1: no compensation
+    int clear_zero;
+    double volume;
+    int TI, tiles;
+};
+
+// numpy.digitize(x, edges): IDENTICAL to nbk_bin.hip `dig` — keep in
+// lockstep (bin-edge parity depends on it)
+__device__ __forceinline__ int digx(const double* __restrict__ edges,
+                                    int nedges, double x) {
+    int lo = 0, hi = nedges;
+    while (lo < hi) {
+        const int mid = (lo + hi) >> 1;
+        if (edges[mid] <= x) lo = mid + 1;
+        else hi = mid;
+    }
+    return lo;
+}
+
+__global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
+                          const double* __restrict__ k2edges_g,
+                          const double* __restrict__ muedges_g,
+                          const cdouble* __restrict__ table /* W_2n0 */,
+                          double* __restrict__ gout)
+{
+    const int NB = (A.nx_edges + 1) * (A.nmu_edges + 1);
+    const int nfields = 3 + 2 * A.nell;
+    const int n0 = (int)A.n0;
+    const int TI = A.TI;
+    const int T = blockDim.x;
+    const int t = threadIdx.x;
+    const int bits = 31 - __clz((unsigned)n0);
+
+    // LDS: [fft buf (16B-aligned, first) | hist | k2 edges | mu edges |
+    //        kx table | x-compensation table]
+    extern __shared__ cdouble smem[];                  // (shared decl
+                                                       // with kfft_c2r_z)
+    cdouble* buf = smem;                               // n0 * TI
+    double* h = (double*)smem + 2 * (size_t)n0 * TI;   // NB * nfields
+    double* ke = h + (size_t)NB * nfields;             // nx_edges
+    double* me = ke + A.nx_edges;                      // nmu_edges
+    double* kxv = me + A.nmu_edges;                    // n0
+    double* cxv = kxv + n0;                            // n0
+    __shared__ double cky2[16], ckz2[16], ckyl[16], ckzl[16],
+                      ccy[16], ccz[16];
+    __shared__ unsigned char cw2[16], czl[16], cskip[16];
+    __shared__ int s_nlive;
+
+    const bool comp_on = A.win1 >= 0;
+    for (int i = t; i < NB * nfields; i += T) h[i] = 0.0;
+    for (int i = t; i < A.nx_edges; i += T) ke[i] = k2edges_g[i];
+    for (int i = t; i < A.nmu_edges; i += T) me[i] = muedges_g[i];
+    for (int j = t; j < n0; j += T) {
+        const double fx = freq_full(j, A.n0);
+        kxv[j] = fx * A.k0x;
+        cxv[j] = comp_on
+            ? nbk_comp_factor1(A.win1, A.interl1,
+                               2.0 * M_PI * fx / (double)A.n0)
+            : 1.0;
+    }
+    __syncthreads();
+    const double k2last = ke[A.nx_edges - 1];
+
+    for (int64_t tile = blockIdx.x; tile < A.tiles; tile += gridDim.x) {
+        const int64_t c0 = tile * TI;
+        __syncthreads();                // prior tile's bin phase done
+        if (t < TI) {
+            const int64_t cg = c0 + t;
+            int live = 0;
+            if (cg < A.n_inner) {
+                const int64_t iyl = cg / A.nzh;
+                const int64_t iz = cg - iyl * A.nzh;
+                const double fy = freq_full(iyl + A.y_off, A.n1);
+                const double fz = freq_half(iz, A.n2);
+                const double ky = fy * A.k0y;
+                const double kz = fz * A.k0z;
+                cky2[t] = ky * ky;
+                ckz2[t] = kz * kz;
+                ckyl[t] = ky * A.losy;
+                ckzl[t] = kz * A.losz;
+                cw2[t] = (fz > 0.0);
+                czl[t] = (fy == 0.0 && fz == 0.0);
+                if (comp_on) {
+                    ccy[t] = nbk_comp_factor1(A.win1, A.interl1,
+                                 2.0 * M_PI * fy / (double)A.n1);
+                    ccz[t] = nbk_comp_factor1(A.win1, A.interl1,
+                                 2.0 * M_PI * fz / (double)A.n2);
+                } else {
+                    ccy[t] = 1.0;
+                    ccz[t] = 1.0;
+                }
+                // min over the line is at kx = 0 exactly, where
+                // fl(0 + ky2) = ky2; fl-addition is monotone, so
+                // ky2 + kz2 >= k2last skips every element of the column
+                live = (cky2[t] + ckz2[t] < k2last) ? 1 : 0;
+            }
+            cskip[t] = (unsigned char)(!live);
+        }
+        if (t == 0) s_nlive = 0;
+        __syncthreads();
+        if (t < TI && !cskip[t]) atomicAdd(&s_nlive, 1);
+        __syncthreads();
+        if (s_nlive == 0) continue;
+
+        const cdouble* g = (const cdouble*)data + c0;
+        for (int w = t; w < n0 * TI; w += T) {
+            const int c = w % TI;
+            const int j = w / TI;
+            cdouble v = {0.0, 0.0};
+            if (!cskip[c])
+                v = g[(int64_t)j * A.n_inner + c];
+            buf[bitrev(j, bits) * TI + c] = v;
+        }
+        __syncthreads();
+
+        lds_fft4<false>(buf, n0, TI, table);
+
+        for (int w = t; w < n0 * TI; w += T) {
+            const int c = w % TI;
+            if (cskip[c]) continue;
+            const int j = w / TI;
+            const double kx = kxv[j];
+            const double k2 = (kx * kx + cky2[c]) + ckz2[c];
+            if (k2 >= k2last) continue;
+            const double kmag = sqrt(k2);
+            double mu = (kx * A.losx + ckyl[c]) + ckzl[c];
+            mu = (kmag == 0.0) ? 0.0 : mu / kmag;
+            const bool nonsingular = cw2[c];
+            const double wgt = nonsingular ? 2.0 : 1.0;
+            const int bx = digx(ke, A.nx_edges, k2);
+            const int bmu = digx(me, A.nmu_edges, mu);
+            const int bin = bx * (A.nmu_edges + 1) + bmu;
+
+            cdouble a = buf[j * TI + c];
+            if (comp_on)
+                a = cscale(a, (cxv[j] * ccy[c]) * ccz[c]);
+            const cdouble p = cmul(a, cconj(a));
+            cdouble v = {p.re * A.volume, p.im * A.volume};
+            if (A.clear_zero && czl[c] && j == 0)
+                v = {0.0, 0.0};
+
+            atomicAdd(&h[bin], kmag * wgt);
+            atomicAdd(&h[NB + bin], mu * wgt);
+            atomicAdd(&h[2 * NB + bin], wgt);
+            double Pm1 = 0.0, P = 1.0;
+            int e = 0;
+            for (int l = 0; e < A.nell; l++) {
+                if (l > 0) {
+                    const double Pn = ((2 * l - 1) * mu * P
+                                       - (l - 1) * Pm1) / l;
+                    Pm1 = P;
+                    P = Pn;
+                }
+                if (l == A.ells[e]) {
+                    cdouble wy = cscale(v, P);
+                    if (nonsingular) {
+                        if (l % 2) wy = {0.0, 2.0 * wy.im};
+                        else wy = {2.0 * wy.re, 0.0};
+                    }
+                    wy = cscale(wy, 2.0 * l + 1.0);
+                    atomicAdd(&h[(3 + 2 * e) * NB + bin], wy.re);
+                    atomicAdd(&h[(3 + 2 * e + 1) * NB + bin], wy.im);
+                    e++;
+                }
+            }
+        }
+    }
+    __syncthreads();
+    for (int i = t; i < NB * nfields; i += T)
+        if (h[i] != 0.0) atomicAdd(&gout[i], h[i]);
+}
+
 }  // namespace
 
 extern "C" int nbk_fft_r2c_z(const double* real, double* cplx,
@@ -364,6 +569,101 @@ extern "C" int nbk_fft_c_strided(double* cplx, int64_t nfft, int64_t stride,
                            dim3(block), shmem, (hipStream_t)stream,
                            cplx, (int)nfft, stride, outer_stride, n_inner,
                            TI, tiles, (const cdouble*)table);
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}
+
+extern "C" int nbk_fft_x_bin_f64(const double* data,
+                                 const int64_t nmesh[3],
+                                 int64_t n_inner, int64_t y_off,
+                                 const double box[3],
+                                 int window1, int interlaced1,
+                                 int clear_zero, double volume,
+                                 const double* k2edges, int64_t nx_edges,
+                                 const double* muedges, int64_t nmu_edges,
+                                 const double los[3],
+                                 const int* ells, int nell,
+                                 double* out_sums, void* stream)
+{
+    int rc = check_len(nmesh[0], "nbk_fft_x_bin_f64");
+    if (rc) return rc;
+    if (n_inner == 0) return NBK_OK;
+    if (nell > NBK_MAX_ELL) {
+        NBK_SET_ERR("nbk_fft_x_bin_f64: at most %d multipoles",
+                    NBK_MAX_ELL);
+        return NBK_ERR_ARG;
+    }
+    if (window1 > 2) {
+        NBK_SET_ERR("nbk_fft_x_bin_f64: bad window %d", window1);
+        return NBK_ERR_ARG;
+    }
+    double* table = get_twiddles(2 * nmesh[0]);
+    if (!table) { NBK_SET_ERR("twiddle alloc failed"); return NBK_ERR_HIP; }
+
+    XBinArgs A;
+    A.n0 = nmesh[0]; A.n1 = nmesh[1]; A.n2 = nmesh[2];
+    A.nzh = nmesh[2] / 2 + 1;
+    A.n_inner = n_inner;
+    A.y_off = y_off;
+    A.k0x = 2.0 * M_PI / box[0];
+    A.k0y = 2.0 * M_PI / box[1];
+    A.k0z = 2.0 * M_PI / box[2];
+    A.losx = los[0]; A.losy = los[1]; A.losz = los[2];
+    A.nx_edges = (int)nx_edges;
+    A.nmu_edges = (int)nmu_edges;
+    A.nell = nell;
+    for (int e = 0; e < NBK_MAX_ELL; e++)
+        A.ells[e] = e < nell ? ells[e] : -1;
+    A.win1 = window1;
+    A.interl1 = interlaced1;
+    A.clear_zero = clear_zero;
+    A.volume = volume;
+
+    const int64_t NB = (nx_edges + 1) * (nmu_edges + 1);
+    const int nfields = 3 + 2 * nell;
+    const size_t fixed = ((size_t)NB * nfields + nx_edges + nmu_edges
+                          + 2 * nmesh[0]) * sizeof(double);
+    static int TI0 = 0;
+    if (!TI0) {
+        const char* e = getenv("NBK_XBIN_TI");
+        TI0 = e ? atoi(e) : 4;
+        if (TI0 < 1 || TI0 > 16) TI0 = 4;
+    }
+    int TI = TI0;
+    while (TI > 1 && fixed + (size_t)nmesh[0] * TI * sizeof(cdouble)
+                     > 160 * 1024)
+        TI >>= 1;
+    const size_t shmem = fixed + (size_t)nmesh[0] * TI * sizeof(cdouble);
+    if (shmem > 160 * 1024) {
+        NBK_SET_ERR("nbk_fft_x_bin_f64: LDS budget exceeded "
+                    "(%zu B) — use the unfused path", shmem);
+        return NBK_ERR_UNSUPPORTED;
+    }
+    if (TI > n_inner) TI = (int)n_inner;
+    A.TI = TI;
+    A.tiles = (int)((n_inner + TI - 1) / TI);
+
+    // persistent grid: each block flushes one LDS histogram at the end
+    // (the global atomic pipe runs ~25 G op/s — cap the flush count)
+    static int64_t gcap = 0;
+    if (!gcap) {
+        const char* e = getenv("NBK_XBIN_GRID");
+        gcap = e ? atoll(e) : 2048;
+        if (gcap < 256 || gcap > 65536) gcap = 2048;
+    }
+    int64_t g = A.tiles;
+    if (g > gcap) g = gcap;
+
+    static size_t raised = 0;
+    if (shmem > 64 * 1024 && shmem > raised) {
+        (void)hipFuncSetAttribute(
+            reinterpret_cast<const void*>(&kxfft_bin),
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);
+        raised = shmem;
+    }
+    hipLaunchKernelGGL(kxfft_bin, dim3((uint32_t)g), dim3(1024), shmem,
+                       (hipStream_t)stream, data, A, k2edges, muedges,
+                       (const cdouble*)table, out_sums);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

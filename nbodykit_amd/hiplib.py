@@ -24,7 +24,7 @@ EXPORTED_SYMBOLS = [
     'nbk_paint_gather_fft_f64',
     'nbk_fft_r2c_z', 'nbk_fft_c2r_z', 'nbk_fft_c_strided',
     'nbk_compensate_f64', 'nbk_interlace_combine_f64', 'nbk_power3d_f64',
-    'nbk_bin_power_f64', 'nbk_power_bin_f64',
+    'nbk_bin_power_f64', 'nbk_power_bin_f64', 'nbk_fft_x_bin_f64',
     'nbk_axpy_f64', 'nbk_scale_f64',
 ]
 
@@ -132,6 +132,14 @@ def _declare(lib):
                                       ctypes.c_int,
                                       c_void, c_void, c_void, c_void,
                                       c_void]
+    lib.nbk_fft_x_bin_f64.restype = ctypes.c_int
+    lib.nbk_fft_x_bin_f64.argtypes = [c_void, c_i64_p, c_i64, c_i64,
+                                      c_f64_p,
+                                      ctypes.c_int, ctypes.c_int,
+                                      ctypes.c_int, c_f64,
+                                      c_void, c_i64, c_void, c_i64,
+                                      c_f64_p, c_int_p, ctypes.c_int,
+                                      c_void, c_void]
     lib.nbk_axpy_f64.restype = ctypes.c_int
     lib.nbk_axpy_f64.argtypes = [c_void, c_void, c_f64, c_i64, c_void]
     lib.nbk_scale_f64.restype = ctypes.c_int

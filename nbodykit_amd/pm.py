@@ -302,11 +302,13 @@ class ParticleMesh(object):
                 and numpy.array_equal(self.BoxSize, other.BoxSize))
 
 
-def _r2c_finish(cplx, pm, s):
-    """y and x strided passes (+ the RCCL pencil transpose when
-    distributed) over an existing z half-spectrum — the tail of
-    RealField.r2c, shared with the fused paint+z-FFT path
-    (source/mesh/catalog.py to_complex_field)."""
+def _r2c_y_transpose(cplx, pm, s):
+    """y strided pass (+ the RCCL pencil transpose when distributed)
+    over an existing z half-spectrum — the head of _r2c_finish, shared
+    with the deferred-x binning path (FFTPower feeds the pre-x-pass
+    field straight into nbk_fft_x_bin_f64).  Returns (cplx, n_inner):
+    the x lines run through flattened local (y, zh) column c at
+    flat[j * n_inner + c]."""
     from nbodykit_amd import profiling
     hiplib.require()
     nx_l, ny, nzh = cplx.shape
@@ -319,9 +321,40 @@ def _r2c_finish(cplx, pm, s):
         n_inner = pm.ny_local * nzh
     else:
         n_inner = ny * nzh
+    return cplx, n_inner
+
+
+def _r2c_finish(cplx, pm, s):
+    """y and x strided passes (+ the RCCL pencil transpose when
+    distributed) over an existing z half-spectrum — the tail of
+    RealField.r2c, shared with the fused paint+z-FFT path
+    (source/mesh/catalog.py to_complex_field)."""
+    from nbodykit_amd import profiling
+    cplx, n_inner = _r2c_y_transpose(cplx, pm, s)
+    ncells = int(pm.Nmesh[0]) * int(pm.Nmesh[1]) * int(pm.Nmesh[2])
     with profiling.collect('fft_strided', ncells):
         fft_axis1(cplx.view(1, int(pm.Nmesh[0]), n_inner), -1, s)  # x
     return cplx
+
+
+def r2c_defer_x(real_field):
+    """z + y passes (+ transpose) of RealField.r2c WITHOUT the final x
+    pass: the deferred-x head for meshes that materialized a real field
+    (the chunked paint path).  Returns (tensor, n_inner); the 1/Ntotal
+    forward normalization is already applied (it folds into the z
+    pass), so the tensor feeds nbk_fft_x_bin_f64 directly."""
+    import torch
+    hiplib.require()
+    pm = real_field.pm
+    nx_l, ny, nz = real_field.value.shape
+    nzh = nz // 2 + 1
+    scale = 1.0 / float(numpy.prod(pm.Nmesh))
+    cplx = torch.empty((nx_l, ny, nzh), dtype=torch.complex128,
+                       device='cuda')
+    s = hiplib.cur_stream()
+    fft_r2c_z(real_field.value.view(nx_l * ny, nz),
+              cplx.view(nx_l * ny, nzh), scale, s)
+    return _r2c_y_transpose(cplx, pm, s)
 
 
 class _FieldBase(object):

@@ -484,13 +484,20 @@ class CatalogMesh(MeshSource):
 
         return toret
 
-    def to_complex_field(self, out=None):
+    def to_complex_field(self, out=None, _defer_x=False):
         """Fused paint -> forward-FFT fast path: the z pass runs inside
         the gather paint's tile flush, so the real mesh never exists in
         HBM; interlaced meshes additionally skip the c2r + re-r2c round
         trip of the real path (combine directly in k).  Numerically the
         FFT is the same radix-2 code as nbk_fft_r2c_z; the 1/N^3 and
         1/nbar (normalize) factors fold into the kernel's output scale.
+
+        ``_defer_x`` (internal, FFTPower's fused x-FFT+binning path;
+        only valid for non-interlaced meshes): stop BEFORE the final x
+        strided pass and return ``(tensor, n_inner, attrs)`` — the
+        pre-x-pass spectrum nbk_fft_x_bin_f64 consumes — instead of a
+        ComplexField.  Returns NotImplemented under exactly the same
+        (collective) gates as the normal path.
 
         Multi-rank: every gate below the collective allgather is
         rank-invariant (pm geometry or globally-reduced sizes), so all
@@ -607,9 +614,9 @@ class CatalogMesh(MeshSource):
         nzh = n2 // 2 + 1
         shape = (pm.nx_local, int(pm.Nmesh[1]), nzh)
 
-        from nbodykit_amd.pm import _r2c_finish
+        from nbodykit_amd.pm import _r2c_finish, _r2c_y_transpose
 
-        def one(shift):
+        def one(shift, finish=True):
             z = torch.empty(shape, dtype=torch.complex128, device='cuda')
             # time only the fused kernel as 'paint' (the y/x passes are
             # FFT work, not paint work — the bench roofline reads this)
@@ -620,7 +627,19 @@ class CatalogMesh(MeshSource):
                     hiplib.dptr(rowtab), hiplib.dptr(z), pm.x_start,
                     pm.nx_local, scale, stream),
                     'nbk_paint_gather_fft_f64')
-            return _r2c_finish(z, pm, stream)
+            if finish:
+                return _r2c_finish(z, pm, stream)
+            return _r2c_y_transpose(z, pm, stream)
+
+        if _defer_x:
+            assert not interlaced, \
+                "_defer_x is only valid for non-interlaced meshes"
+            tensor, n_inner = one(0.0, finish=False)
+            with numpy.errstate(divide='ignore', invalid='ignore'):
+                shotnoise = float(numpy.prod(pm.BoxSize)) * W2 / W ** 2
+            return tensor, n_inner, {'shotnoise': shotnoise, 'N': N,
+                                     'W': W, 'W2': W2,
+                                     'num_per_cell': nbar}
 
         cplx = one(0.0)
         if interlaced:

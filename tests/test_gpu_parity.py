@@ -1079,3 +1079,139 @@ def test_random_fkp_fuzz(seed):
     rel, modes_ok, cfg = run_case4(seed)
     assert modes_ok, cfg
     assert rel < 5e-6, (rel, cfg)       # poles stored as c8
+
+
+# ---- deferred-x fused FFT + binning path (nbk_fft_x_bin_f64) ----------
+# The final x-axis FFT pass runs inside the binning kernel for auto
+# power on non-interlaced power-of-two meshes: the finished complex
+# field is never materialized.  Parity bar: bit-identical bin counts,
+# power to the roundoff tolerances of the other fused-vs-unfused tests.
+
+def _trace_xbin(monkeypatch):
+    from nbodykit_amd.algorithms import fftpower as fmod
+    calls = []
+    orig = fmod._project_power_xbin
+
+    def wrapper(*a, **kw):
+        calls.append(1)
+        return orig(*a, **kw)
+
+    monkeypatch.setattr(fmod, '_project_power_xbin', wrapper)
+    return calls
+
+
+XBIN_CASES = [
+    dict(mode='1d'),
+    dict(mode='1d', poles=[0, 2, 4]),
+    dict(mode='2d', Nmu=5),
+    dict(mode='2d', Nmu=3, poles=[0, 2], los=[1, 0, 0]),
+    dict(mode='1d', kmin=0.05, kmax=1.0, dk=0.02),
+]
+
+
+@pytest.mark.parametrize('kwargs', XBIN_CASES,
+                         ids=['1d', 'poles', '2d', '2dlosx', 'kwin'])
+def test_xbin_matches_unfused(kwargs, monkeypatch):
+    """Deferred-x path vs the standard fused path (NBK_NO_XBIN=1) on
+    the same mesh — the x-FFT element values are bit-identical, so only
+    accumulation-order roundoff may differ."""
+    cat = UniformCatalog(nbar=3e-3, BoxSize=128., seed=11)
+    mesh = cat.to_mesh(Nmesh=128, dtype='f8', compensated=True)
+    calls = _trace_xbin(monkeypatch)
+    r_x = FFTPower(mesh, **kwargs)
+    assert calls, 'deferred-x path did not engage'
+    monkeypatch.setenv('NBK_NO_XBIN', '1')
+    r_ref = FFTPower(mesh, **kwargs)
+    scale = numpy.nanmax(numpy.abs(r_ref.power['power']))
+    assert_array_equal(r_x.power['modes'], r_ref.power['modes'])
+    assert_allclose(r_x.power['power'], r_ref.power['power'],
+                    rtol=1e-10, atol=1e-12 * scale, equal_nan=True)
+    assert_allclose(numpy.nan_to_num(r_x.power['k']),
+                    numpy.nan_to_num(r_ref.power['k']),
+                    rtol=1e-10, atol=1e-12)
+    if 'mu' in r_ref.power.variables:
+        assert_allclose(numpy.nan_to_num(r_x.power['mu']),
+                        numpy.nan_to_num(r_ref.power['mu']),
+                        rtol=1e-10, atol=1e-12)
+    for ell in kwargs.get('poles', []):
+        assert_allclose(r_x.poles['power_%d' % ell],
+                        r_ref.poles['power_%d' % ell],
+                        rtol=1e-10, atol=1e-12 * scale, equal_nan=True)
+
+
+@pytest.mark.parametrize('resampler,compensated',
+                         [('tsc', True), ('pcs', True), ('cic', False)])
+def test_xbin_windows_match_unfused(resampler, compensated, monkeypatch):
+    cat = UniformCatalog(nbar=1e-2, BoxSize=64., seed=12)
+    mesh = cat.to_mesh(Nmesh=64, dtype='f8', compensated=compensated,
+                       resampler=resampler)
+    calls = _trace_xbin(monkeypatch)
+    r_x = FFTPower(mesh, mode='1d')
+    assert calls
+    monkeypatch.setenv('NBK_NO_XBIN', '1')
+    r_ref = FFTPower(mesh, mode='1d')
+    scale = numpy.nanmax(numpy.abs(r_ref.power['power']))
+    assert_array_equal(r_x.power['modes'], r_ref.power['modes'])
+    assert_allclose(r_x.power['power'], r_ref.power['power'],
+                    rtol=1e-10, atol=1e-12 * scale, equal_nan=True)
+
+
+def test_xbin_through_gather_paint(monkeypatch):
+    """Deferred-x composed with the fused paint+z-FFT head: the
+    pre-x-pass tensor comes straight out of the gather paint (the C4
+    composition).  Compare against the fully unfused pipeline."""
+    from nbodykit_amd import set_options
+    cat = UniformCatalog(nbar=0.7, BoxSize=64., seed=13)   # ~1.8e5 pts
+    calls = _trace_xbin(monkeypatch)
+    with set_options(sort_min_n=1024, sort_two_level_min_n=1024,
+                     sort_two_level_min_cells=1):
+        r_x = FFTPower(cat, mode='1d', Nmesh=64)
+    assert calls
+    monkeypatch.setenv('NBK_NO_XBIN', '1')
+    r_ref = FFTPower(cat, mode='1d', Nmesh=64)
+    scale = numpy.nanmax(numpy.abs(r_ref.power['power']))
+    assert_array_equal(r_x.power['modes'], r_ref.power['modes'])
+    assert_allclose(r_x.power['power'], r_ref.power['power'],
+                    rtol=1e-10, atol=1e-12 * scale, equal_nan=True)
+    assert_allclose(r_x.attrs['shotnoise'], r_ref.attrs['shotnoise'],
+                    rtol=1e-12)
+
+
+def test_xbin_gates(monkeypatch):
+    """Configurations the deferred-x kernel must NOT serve fall back to
+    the standard path: interlaced meshes, cross power, dk=0 unique
+    edges, non-power-of-two meshes, and NBK_NO_XBIN=1."""
+    calls = _trace_xbin(monkeypatch)
+    cat = UniformCatalog(nbar=1e-2, BoxSize=64., seed=14)
+    mesh_i = cat.to_mesh(Nmesh=64, dtype='f8', compensated=True,
+                         interlaced=True)
+    FFTPower(mesh_i, mode='1d')
+    assert not calls, 'interlaced mesh must not defer'
+    cat2 = UniformCatalog(nbar=1e-2, BoxSize=64., seed=15)
+    m1 = cat.to_mesh(Nmesh=64, dtype='f8', compensated=True)
+    m2 = cat2.to_mesh(Nmesh=64, dtype='f8', compensated=True)
+    FFTPower(m1, mode='1d', second=m2)
+    assert not calls, 'cross power must not defer'
+    FFTPower(m1, mode='1d', dk=0)
+    assert not calls, 'dk=0 unique edges must not defer'
+    m96 = cat.to_mesh(Nmesh=96, dtype='f8', compensated=True)
+    FFTPower(m96, mode='1d')
+    assert not calls, 'non-power-of-two mesh must not defer'
+    monkeypatch.setenv('NBK_NO_XBIN', '1')
+    FFTPower(m1, mode='1d')
+    assert not calls, 'NBK_NO_XBIN=1 must not defer'
+    monkeypatch.delenv('NBK_NO_XBIN')
+    FFTPower(m1, mode='1d')
+    assert calls, 'plain auto 1d must defer'
+
+
+def test_xbin_matches_oracle():
+    """Deferred-x FFTPower vs the CPU oracle end-to-end (the engaged
+    default path for plain auto power)."""
+    nbar, box, nmesh, seed = 1e-2, 128., 128, 21
+    cat = UniformCatalog(nbar=nbar, BoxSize=box, seed=seed)
+    r = FFTPower(cat, mode='1d', Nmesh=nmesh)
+    pos = uniform_positions(nbar, box, seed)
+    want = fftpower_oracle(pos, Nmesh=nmesh, BoxSize=box, mode='1d',
+                           resampler='cic', compensated=True)
+    check_parity(r, want)
