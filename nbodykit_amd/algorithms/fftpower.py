@@ -641,8 +641,8 @@ def _xbin_lds_fits(n0, nx_edges, nmu_edges, nell):
     NBK_ERR_UNSUPPORTED; gate here so the standard path is taken
     without a failed launch)."""
     NB = (nx_edges + 1) * (nmu_edges + 1)
-    fixed = (NB * (3 + 2 * nell) + nx_edges + nmu_edges + 2 * n0) * 8
-    return fixed + n0 * 16 <= 160 * 1024
+    fixed = (NB * (3 + 2 * nell) + nx_edges + nmu_edges + n0) * 8
+    return fixed + n0 * 2 * 16 <= 160 * 1024   # TI >= 1, pitch TI+1
 
 
 def _project_power_xbin(tensor, pm, n_inner, comp1, volume, edges,

@@ -76,8 +76,11 @@ __device__ __forceinline__ int bitrev(int j, int bits) {
 //   e0 = x0 + b2, e1 = x0 - b2, o0 = b1 + b3, o1 = b1 - b3
 //   y = (e0+o0, e1 -/+ i*o1, e0-o0, e1 +/- i*o1)   (forward/inverse)
 // Validated element-exact against numpy fft for m = 4..2048.
+// PITCH is the LDS storage row stride in cdoubles (>= TI; the fused
+// x+bin kernel pads to TI+1 so its per-thread j-runs read across LDS
+// banks instead of landing on two)
 template <bool INV>
-__device__ void lds_fft4(cdouble* buf, int m, int TI,
+__device__ void lds_fft4(cdouble* buf, int m, int TI, int PITCH,
                          const cdouble* __restrict__ table) {
     const int T = blockDim.x;
     const int tid = threadIdx.x;
@@ -87,8 +90,8 @@ __device__ void lds_fft4(cdouble* buf, int m, int TI,
         for (int w = tid; w < (m >> 1) * TI; w += T) {
             const int c = w % TI;
             const int j = w / TI;
-            const int i0 = (2 * j) * TI + c;
-            const int i1 = i0 + TI;
+            const int i0 = (2 * j) * PITCH + c;
+            const int i1 = i0 + PITCH;
             const cdouble u = buf[i0];
             const cdouble v = buf[i1];
             buf[i0] = cadd(u, v);
@@ -105,8 +108,8 @@ __device__ void lds_fft4(cdouble* buf, int m, int TI,
             const int j = w / TI;
             const int grp = j / h;
             const int pos = j - grp * h;
-            const int base = (grp * (len << 1) + pos) * TI + c;
-            const int s = h * TI;
+            const int base = (grp * (len << 1) + pos) * PITCH + c;
+            const int s = h * PITCH;
             cdouble w1 = table[pos * tw];
             cdouble w2 = table[2 * pos * tw];
             if (INV) { w1.im = -w1.im; w2.im = -w2.im; }
@@ -151,7 +154,7 @@ __global__ void kfft_r2c_z(const double* __restrict__ real,
         buf[bitrev(q, bits)] = g[q];
     __syncthreads();
 
-    lds_fft4<false>(buf, m, 1, table);
+    lds_fft4<false>(buf, m, 1, 1, table);
 
     // untwiddle split: X[k] = E[k] + W_nz^k * O[k], k = 0..m
     cdouble* out = (cdouble*)cplx + line * (m + 1);
@@ -205,7 +208,7 @@ __global__ void kfft_c2r_z(const double* __restrict__ cplx,
     }
     __syncthreads();
 
-    lds_fft4<true>(buf, m, 1, table);
+    lds_fft4<true>(buf, m, 1, 1, table);
 
     // unpack: line[2t] = 2 Re(z[t]), line[2t+1] = 2 Im(z[t])
     cdouble* out = (cdouble*)(real + line * nz);
@@ -239,7 +242,7 @@ __global__ void kfft_c_strided(double* __restrict__ data,
     }
     __syncthreads();
 
-    lds_fft4<INV>(buf, nfft, TI, table);
+    lds_fft4<INV>(buf, nfft, TI, TI, table);
 
     for (int w = threadIdx.x; w < nfft * TI; w += blockDim.x) {
         const int c = w % TI;
@@ -320,20 +323,23 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
     const int nfields = 3 + 2 * A.nell;
     const int n0 = (int)A.n0;
     const int TI = A.TI;
+    const int W = TI + 1;          // padded LDS pitch: the bin phase's
+                                   // per-thread j-runs then stride an
+                                   // odd multiple of 16 B across banks
     const int T = blockDim.x;
     const int t = threadIdx.x;
     const int bits = 31 - __clz((unsigned)n0);
 
-    // LDS: [fft buf (16B-aligned, first) | hist | k2 edges | mu edges |
-    //        kx table | x-compensation table]
+    // LDS: [fft buf (16B-aligned, first, pitch W) | hist | k2 edges |
+    //        mu edges | x-compensation table]  (kx is recomputed per
+    //        element — same fl arithmetic as the coordinate tables)
     extern __shared__ cdouble smem[];                  // (shared decl
                                                        // with kfft_c2r_z)
-    cdouble* buf = smem;                               // n0 * TI
-    double* h = (double*)smem + 2 * (size_t)n0 * TI;   // NB * nfields
+    cdouble* buf = smem;                               // n0 * W
+    double* h = (double*)smem + 2 * (size_t)n0 * W;    // NB * nfields
     double* ke = h + (size_t)NB * nfields;             // nx_edges
     double* me = ke + A.nx_edges;                      // nmu_edges
-    double* kxv = me + A.nmu_edges;                    // n0
-    double* cxv = kxv + n0;                            // n0
+    double* cxv = me + A.nmu_edges;                    // n0
     __shared__ double cky2[16], ckz2[16], ckyl[16], ckzl[16],
                       ccy[16], ccz[16];
     __shared__ unsigned char cw2[16], czl[16], cskip[16];
@@ -345,7 +351,6 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
     for (int i = t; i < A.nmu_edges; i += T) me[i] = muedges_g[i];
     for (int j = t; j < n0; j += T) {
         const double fx = freq_full(j, A.n0);
-        kxv[j] = fx * A.k0x;
         cxv[j] = comp_on
             ? nbk_comp_factor1(A.win1, A.interl1,
                                2.0 * M_PI * fx / (double)A.n0)
@@ -353,6 +358,13 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
     }
     __syncthreads();
     const double k2last = ke[A.nx_edges - 1];
+
+    // bin-phase mapping: contiguous j-runs per thread at wave-uniform
+    // column c (kbin_run-style register merging + warm-start digitize)
+    const int CPT = T / TI;                 // threads per column
+    const int RB = (n0 + CPT - 1) / CPT;    // j-run length per thread
+    const int my_c = t / CPT;
+    const int my_j0 = (t - my_c * CPT) * RB;
 
     for (int64_t tile = blockIdx.x; tile < A.tiles; tile += gridDim.x) {
         const int64_t c0 = tile * TI;
@@ -402,58 +414,116 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
             cdouble v = {0.0, 0.0};
             if (!cskip[c])
                 v = g[(int64_t)j * A.n_inner + c];
-            buf[bitrev(j, bits) * TI + c] = v;
+            buf[bitrev(j, bits) * W + c] = v;
         }
         __syncthreads();
 
-        lds_fft4<false>(buf, n0, TI, table);
+        lds_fft4<false>(buf, n0, TI, W, table);
 
-        for (int w = t; w < n0 * TI; w += T) {
-            const int c = w % TI;
-            if (cskip[c]) continue;
-            const int j = w / TI;
-            const double kx = kxv[j];
-            const double k2 = (kx * kx + cky2[c]) + ckz2[c];
-            if (k2 >= k2last) continue;
-            const double kmag = sqrt(k2);
-            double mu = (kx * A.losx + ckyl[c]) + ckzl[c];
-            mu = (kmag == 0.0) ? 0.0 : mu / kmag;
+        // run-merged bin phase: this thread walks j = my_j0 .. +RB-1 of
+        // column my_c, accumulating same-bin neighbours in registers
+        // and warm-starting the digitize (|k| varies slowly along the
+        // line) — the kbin_run recipe, reading LDS instead of HBM
+        if (my_c < TI && !cskip[my_c] && my_j0 < n0) {
+            const int c = my_c;
+            const double ky2 = cky2[c];
+            const double kz2 = ckz2[c];
+            const double kyl = ckyl[c];
+            const double kzl = ckzl[c];
+            const double ccyv = ccy[c];
+            const double cczv = ccz[c];
             const bool nonsingular = cw2[c];
             const double wgt = nonsingular ? 2.0 : 1.0;
-            const int bx = digx(ke, A.nx_edges, k2);
-            const int bmu = digx(me, A.nmu_edges, mu);
-            const int bin = bx * (A.nmu_edges + 1) + bmu;
+            const bool col_zero = czl[c];
+            const int jend = (my_j0 + RB < n0) ? my_j0 + RB : n0;
 
-            cdouble a = buf[j * TI + c];
-            if (comp_on)
-                a = cscale(a, (cxv[j] * ccy[c]) * ccz[c]);
-            const cdouble p = cmul(a, cconj(a));
-            cdouble v = {p.re * A.volume, p.im * A.volume};
-            if (A.clear_zero && czl[c] && j == 0)
-                v = {0.0, 0.0};
+            int cbin = -1, cbx = -1, cbmu = -1;
+            double ak = 0.0, amu = 0.0, aw = 0.0;
+            cdouble ay[NBK_MAX_ELL];
+            for (int e = 0; e < A.nell; e++) ay[e] = {0.0, 0.0};
 
-            atomicAdd(&h[bin], kmag * wgt);
-            atomicAdd(&h[NB + bin], mu * wgt);
-            atomicAdd(&h[2 * NB + bin], wgt);
-            double Pm1 = 0.0, P = 1.0;
-            int e = 0;
-            for (int l = 0; e < A.nell; l++) {
-                if (l > 0) {
-                    const double Pn = ((2 * l - 1) * mu * P
-                                       - (l - 1) * Pm1) / l;
-                    Pm1 = P;
-                    P = Pn;
-                }
-                if (l == A.ells[e]) {
-                    cdouble wy = cscale(v, P);
-                    if (nonsingular) {
-                        if (l % 2) wy = {0.0, 2.0 * wy.im};
-                        else wy = {2.0 * wy.re, 0.0};
+            for (int j = my_j0; j < jend; j++) {
+                const double fx = freq_full(j, A.n0);
+                const double kx = fx * A.k0x;
+                const double k2 = (kx * kx + ky2) + kz2;
+                if (k2 >= k2last) continue;
+                const double kmag = sqrt(k2);
+                double mu = (kx * A.losx + kyl) + kzl;
+                mu = (kmag == 0.0) ? 0.0 : mu / kmag;
+
+                int bx;
+                if (cbx >= 0
+                    && (cbx == 0 || ke[cbx - 1] <= k2)
+                    && (cbx == A.nx_edges || k2 < ke[cbx]))
+                    bx = cbx;
+                else
+                    bx = digx(ke, A.nx_edges, k2);
+                int bmu;
+                if (cbmu >= 0
+                    && (cbmu == 0 || me[cbmu - 1] <= mu)
+                    && (cbmu == A.nmu_edges || mu < me[cbmu]))
+                    bmu = cbmu;
+                else
+                    bmu = digx(me, A.nmu_edges, mu);
+                const int bin = bx * (A.nmu_edges + 1) + bmu;
+
+                cdouble a = buf[j * W + c];
+                if (comp_on)
+                    a = cscale(a, (cxv[j] * ccyv) * cczv);
+                const cdouble p = cmul(a, cconj(a));
+                cdouble v = {p.re * A.volume, p.im * A.volume};
+                if (A.clear_zero && col_zero && j == 0)
+                    v = {0.0, 0.0};
+
+                if (bin != cbin) {
+                    if (cbin >= 0) {
+                        atomicAdd(&h[cbin], ak);
+                        atomicAdd(&h[NB + cbin], amu);
+                        atomicAdd(&h[2 * NB + cbin], aw);
+                        for (int e = 0; e < A.nell; e++) {
+                            atomicAdd(&h[(3 + 2 * e) * NB + cbin],
+                                      ay[e].re);
+                            atomicAdd(&h[(3 + 2 * e + 1) * NB + cbin],
+                                      ay[e].im);
+                        }
                     }
-                    wy = cscale(wy, 2.0 * l + 1.0);
-                    atomicAdd(&h[(3 + 2 * e) * NB + bin], wy.re);
-                    atomicAdd(&h[(3 + 2 * e + 1) * NB + bin], wy.im);
-                    e++;
+                    cbin = bin; cbx = bx; cbmu = bmu;
+                    ak = 0.0; amu = 0.0; aw = 0.0;
+                    for (int e = 0; e < A.nell; e++) ay[e] = {0.0, 0.0};
+                }
+                ak += kmag * wgt;
+                amu += mu * wgt;
+                aw += wgt;
+
+                double Pm1 = 0.0, P = 1.0;
+                int e = 0;
+                for (int l = 0; e < A.nell; l++) {
+                    if (l > 0) {
+                        const double Pn = ((2 * l - 1) * mu * P
+                                           - (l - 1) * Pm1) / l;
+                        Pm1 = P;
+                        P = Pn;
+                    }
+                    if (l == A.ells[e]) {
+                        cdouble wy = cscale(v, P);
+                        if (nonsingular) {
+                            if (l % 2) wy = {0.0, 2.0 * wy.im};
+                            else wy = {2.0 * wy.re, 0.0};
+                        }
+                        wy = cscale(wy, 2.0 * l + 1.0);
+                        ay[e].re += wy.re;
+                        ay[e].im += wy.im;
+                        e++;
+                    }
+                }
+            }
+            if (cbin >= 0) {
+                atomicAdd(&h[cbin], ak);
+                atomicAdd(&h[NB + cbin], amu);
+                atomicAdd(&h[2 * NB + cbin], aw);
+                for (int e = 0; e < A.nell; e++) {
+                    atomicAdd(&h[(3 + 2 * e) * NB + cbin], ay[e].re);
+                    atomicAdd(&h[(3 + 2 * e + 1) * NB + cbin], ay[e].im);
                 }
             }
         }
@@ -622,7 +692,7 @@ extern "C" int nbk_fft_x_bin_f64(const double* data,
     const int64_t NB = (nx_edges + 1) * (nmu_edges + 1);
     const int nfields = 3 + 2 * nell;
     const size_t fixed = ((size_t)NB * nfields + nx_edges + nmu_edges
-                          + 2 * nmesh[0]) * sizeof(double);
+                          + nmesh[0]) * sizeof(double);
     static int TI0 = 0;
     if (!TI0) {
         const char* e = getenv("NBK_XBIN_TI");
@@ -630,16 +700,17 @@ extern "C" int nbk_fft_x_bin_f64(const double* data,
         if (TI0 < 1 || TI0 > 16) TI0 = 4;
     }
     int TI = TI0;
-    while (TI > 1 && fixed + (size_t)nmesh[0] * TI * sizeof(cdouble)
+    // buf pitch is TI+1 (bank-spread padding for the bin phase's runs)
+    while (TI > 1 && fixed + (size_t)nmesh[0] * (TI + 1) * sizeof(cdouble)
                      > 160 * 1024)
         TI >>= 1;
-    const size_t shmem = fixed + (size_t)nmesh[0] * TI * sizeof(cdouble);
+    const size_t shmem = fixed
+        + (size_t)nmesh[0] * (TI + 1) * sizeof(cdouble);
     if (shmem > 160 * 1024) {
         NBK_SET_ERR("nbk_fft_x_bin_f64: LDS budget exceeded "
                     "(%zu B) — use the unfused path", shmem);
         return NBK_ERR_UNSUPPORTED;
     }
-    if (TI > n_inner) TI = (int)n_inner;
     A.TI = TI;
     A.tiles = (int)((n_inner + TI - 1) / TI);
 
