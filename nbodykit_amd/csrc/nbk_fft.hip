@@ -234,11 +234,31 @@ __global__ void kfft_c_strided(double* __restrict__ data,
     cdouble* g = (cdouble*)data + o * ostride + c0;
     const int ncol = (int)min((int64_t)TI, n_inner - c0);
 
-    for (int w = threadIdx.x; w < nfft * TI; w += blockDim.x) {
-        const int c = w % TI;
-        const int j = w / TI;
-        if (c < ncol)
-            buf[bitrev(j, bits) * TI + c] = g[(int64_t)j * stride + c];
+    // register-staged loads, 8 per thread per round: every global load
+    // of a round issues BEFORE the first LDS store, so the whole batch
+    // is in flight together.  The pass is latency-bound (each row is
+    // one 64 B line megabytes from the next); interleaving load + LDS
+    // store per element serializes on the waitcnt and was measured at
+    // <3 GB/s per block.
+    const int total = nfft * TI;
+    for (int base = 0; base < total; base += (int)blockDim.x * 8) {
+        cdouble r[8];
+        #pragma unroll
+        for (int q = 0; q < 8; q++) {
+            const int w = base + q * (int)blockDim.x + (int)threadIdx.x;
+            if (w < total) {
+                const int c = w % TI;
+                r[q] = (c < ncol)
+                    ? g[(int64_t)(w / TI) * stride + c]
+                    : cdouble{0.0, 0.0};
+            }
+        }
+        #pragma unroll
+        for (int q = 0; q < 8; q++) {
+            const int w = base + q * (int)blockDim.x + (int)threadIdx.x;
+            if (w < total)
+                buf[bitrev(w / TI, bits) * TI + (w % TI)] = r[q];
+        }
     }
     __syncthreads();
 
@@ -407,14 +427,29 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
         __syncthreads();
         if (s_nlive == 0) continue;
 
+        // register-staged loads (see kfft_c_strided: the pass is
+        // latency-bound, every load of a round must issue before the
+        // first LDS store)
         const cdouble* g = (const cdouble*)data + c0;
-        for (int w = t; w < n0 * TI; w += T) {
-            const int c = w % TI;
-            const int j = w / TI;
-            cdouble v = {0.0, 0.0};
-            if (!cskip[c])
-                v = g[(int64_t)j * A.n_inner + c];
-            buf[bitrev(j, bits) * W + c] = v;
+        const int total = n0 * TI;
+        for (int base = 0; base < total; base += T * 8) {
+            cdouble r[8];
+            #pragma unroll
+            for (int q = 0; q < 8; q++) {
+                const int w = base + q * T + t;
+                if (w < total) {
+                    const int c = w % TI;
+                    r[q] = cskip[c]
+                        ? cdouble{0.0, 0.0}
+                        : g[(int64_t)(w / TI) * A.n_inner + c];
+                }
+            }
+            #pragma unroll
+            for (int q = 0; q < 8; q++) {
+                const int w = base + q * T + t;
+                if (w < total)
+                    buf[bitrev(w / TI, bits) * W + (w % TI)] = r[q];
+            }
         }
         __syncthreads();
 
