@@ -333,6 +333,10 @@ __device__ __forceinline__ int digx(const double* __restrict__ edges,
     return lo;
 }
 
+// PHASES: bit 1 = run the FFT network, bit 2 = run the bin phase
+// (3 = the real kernel; other values exist only for the perf
+// decomposition tool, selected via NBK_XBIN_PHASES)
+template <int PHASES>
 __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
                           const double* __restrict__ k2edges_g,
                           const double* __restrict__ muedges_g,
@@ -453,13 +457,14 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
         }
         __syncthreads();
 
-        lds_fft4<false>(buf, n0, TI, W, table);
+        if (PHASES & 1)
+            lds_fft4<false>(buf, n0, TI, W, table);
 
         // run-merged bin phase: this thread walks j = my_j0 .. +RB-1 of
         // column my_c, accumulating same-bin neighbours in registers
         // and warm-starting the digitize (|k| varies slowly along the
         // line) — the kbin_run recipe, reading LDS instead of HBM
-        if (my_c < TI && !cskip[my_c] && my_j0 < n0) {
+        if ((PHASES & 2) && my_c < TI && !cskip[my_c] && my_j0 < n0) {
             const int c = my_c;
             const double ky2 = cky2[c];
             const double kz2 = ckz2[c];
@@ -760,16 +765,48 @@ extern "C" int nbk_fft_x_bin_f64(const double* data,
     int64_t g = A.tiles;
     if (g > gcap) g = gcap;
 
+    // NBK_XBIN_PHASES: perf-decomposition only (1 = loads+FFT,
+    // 2 = loads+bin, 0 = loads only); the product always runs 3
+    static int phases = 0;
+    if (!phases) {
+        const char* e = getenv("NBK_XBIN_PHASES");
+        phases = e ? atoi(e) : 3;
+        if (phases < 0 || phases > 3) phases = 3;
+        phases |= 4;            // mark initialized
+    }
+    const void* fn =
+        (phases & 3) == 3 ? reinterpret_cast<const void*>(&kxfft_bin<3>)
+        : (phases & 3) == 1 ? reinterpret_cast<const void*>(&kxfft_bin<1>)
+        : (phases & 3) == 2 ? reinterpret_cast<const void*>(&kxfft_bin<2>)
+        : reinterpret_cast<const void*>(&kxfft_bin<0>);
     static size_t raised = 0;
     if (shmem > 64 * 1024 && shmem > raised) {
         (void)hipFuncSetAttribute(
-            reinterpret_cast<const void*>(&kxfft_bin),
-            hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);
+            fn, hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);
         raised = shmem;
     }
-    hipLaunchKernelGGL(kxfft_bin, dim3((uint32_t)g), dim3(1024), shmem,
-                       (hipStream_t)stream, data, A, k2edges, muedges,
-                       (const cdouble*)table, out_sums);
+    switch (phases & 3) {
+    case 3:
+        hipLaunchKernelGGL(kxfft_bin<3>, dim3((uint32_t)g), dim3(1024),
+                           shmem, (hipStream_t)stream, data, A, k2edges,
+                           muedges, (const cdouble*)table, out_sums);
+        break;
+    case 1:
+        hipLaunchKernelGGL(kxfft_bin<1>, dim3((uint32_t)g), dim3(1024),
+                           shmem, (hipStream_t)stream, data, A, k2edges,
+                           muedges, (const cdouble*)table, out_sums);
+        break;
+    case 2:
+        hipLaunchKernelGGL(kxfft_bin<2>, dim3((uint32_t)g), dim3(1024),
+                           shmem, (hipStream_t)stream, data, A, k2edges,
+                           muedges, (const cdouble*)table, out_sums);
+        break;
+    default:
+        hipLaunchKernelGGL(kxfft_bin<0>, dim3((uint32_t)g), dim3(1024),
+                           shmem, (hipStream_t)stream, data, A, k2edges,
+                           muedges, (const cdouble*)table, out_sums);
+        break;
+    }
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

@@ -20,7 +20,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(
 def one_config(reps=3):
     from nbodykit_amd import hiplib
     lib = hiplib.require()
-    n0 = n1 = n2 = 1024
+    n0 = n1 = n2 = int(os.environ.get('NBK_XBIN_N', '1024'))
     nzh = n2 // 2 + 1
     box = 5000.0
     torch.manual_seed(7)
@@ -87,8 +87,10 @@ def one_config(reps=3):
     t_xbin = time_fn(run_xbin, 'xbin')
     t_xpass = time_fn(run_xpass, 'xpass')
     t_kbin = time_fn(run_kbin, 'kbin')
-    out = {'TI': os.environ.get('NBK_XBIN_TI', '4'),
+    out = {'N': n0,
+           'TI': os.environ.get('NBK_XBIN_TI', '4'),
            'GRID': os.environ.get('NBK_XBIN_GRID', '2048'),
+           'PH': os.environ.get('NBK_XBIN_PHASES', '3'),
            'xbin_ms': round(t_xbin, 3),
            'xpass_ms': round(t_xpass, 3),
            'kbin_ms': round(t_kbin, 3),
@@ -102,8 +104,16 @@ if __name__ == '__main__':
         one_config(int(sys.argv[1]) if len(sys.argv) > 1 else 3)
         sys.exit(0)
     env = dict(os.environ, NBK_XBIN_CHILD='1')
-    for ti in ('2', '4'):
-        for grid in ('1024', '2048', '4096', '8192'):
-            e = dict(env, NBK_XBIN_TI=ti, NBK_XBIN_GRID=grid)
-            subprocess.run([sys.executable, os.path.abspath(__file__),
-                            '3'], env=e, check=False)
+    # phase decomposition at the C4 geometry, then a 512^3 probe (the
+    # smaller field spans ~8x fewer DRAM pages — a large per-byte speed
+    # jump would implicate address-translation misses)
+    for ph in ('0', '1', '2', '3'):
+        e = dict(env, NBK_XBIN_TI='4', NBK_XBIN_GRID='4096',
+                 NBK_XBIN_PHASES=ph)
+        subprocess.run([sys.executable, os.path.abspath(__file__), '3'],
+                       env=e, check=False)
+    for n in ('512', '1024'):
+        e = dict(env, NBK_XBIN_TI='4', NBK_XBIN_GRID='4096',
+                 NBK_XBIN_N=n)
+        subprocess.run([sys.executable, os.path.abspath(__file__), '3'],
+                       env=e, check=False)
