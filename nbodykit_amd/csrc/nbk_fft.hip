@@ -335,8 +335,23 @@ __device__ __forceinline__ int digx(const double* __restrict__ edges,
 
 // PHASES: bit 1 = run the FFT network, bit 2 = run the bin phase
 // (3 = the real kernel; other values exist only for the perf
-// decomposition tool, selected via NBK_XBIN_PHASES)
-template <int PHASES>
+// decomposition tool, NBK_XBIN_PHASES).  MAXE bounds the per-thread
+// multipole accumulator (register pressure: nell == 1 — every plain
+// 1d/2d run — must not pay for 8).
+//
+// Software pipeline: tile t+1's strided global loads are issued right
+// after its column constants publish, BEFORE tile t's FFT + bin phases
+// run, so the HBM latency hides under compute (measured phases at C4:
+// loads 2.4 ms + FFT 2.9 ms + bin 4.7 ms, perfectly serial without
+// this).  Column constants and live-counts are double-buffered; the
+// FFT tile is carried across the compute phases in registers (4
+// cdoubles per thread — the launcher caps n0*TI at 4*blockDim).
+//
+// Auto power only, so the per-element value is real EXACTLY:
+// Im(a conj(a)) = fl(-re*im) + fl(im*re) = 0 under -ffp-contract=off.
+// The imaginary multipole sums are therefore never accumulated (their
+// output planes stay zero, matching the unfused kernel's zeros).
+template <int PHASES, int MAXE>
 __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
                           const double* __restrict__ k2edges_g,
                           const double* __restrict__ muedges_g,
@@ -348,8 +363,8 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
     const int n0 = (int)A.n0;
     const int TI = A.TI;
     const int W = TI + 1;          // padded LDS pitch: the bin phase's
-                                   // per-thread j-runs then stride an
-                                   // odd multiple of 16 B across banks
+                                   // per-thread j-runs stride across
+                                   // banks instead of landing on two
     const int T = blockDim.x;
     const int t = threadIdx.x;
     const int bits = 31 - __clz((unsigned)n0);
@@ -364,10 +379,10 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
     double* ke = h + (size_t)NB * nfields;             // nx_edges
     double* me = ke + A.nx_edges;                      // nmu_edges
     double* cxv = me + A.nmu_edges;                    // n0
-    __shared__ double cky2[16], ckz2[16], ckyl[16], ckzl[16],
-                      ccy[16], ccz[16];
-    __shared__ unsigned char cw2[16], czl[16], cskip[16];
-    __shared__ int s_nlive;
+    __shared__ double cky2[2][16], ckz2[2][16], ckyl[2][16],
+                      ckzl[2][16], ccy[2][16], ccz[2][16];
+    __shared__ unsigned char cw2[2][16], czl[2][16], cskip[2][16];
+    __shared__ int s_nlive[2];
 
     const bool comp_on = A.win1 >= 0;
     for (int i = t; i < NB * nfields; i += T) h[i] = 0.0;
@@ -380,8 +395,9 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
                                2.0 * M_PI * fx / (double)A.n0)
             : 1.0;
     }
-    __syncthreads();
-    const double k2last = ke[A.nx_edges - 1];
+    // the skip threshold, read straight from global (one scalar; the
+    // LDS copy is not published until the first sync)
+    const double k2last = k2edges_g[A.nx_edges - 1];
 
     // bin-phase mapping: contiguous j-runs per thread at wave-uniform
     // column c (kbin_run-style register merging + warm-start digitize)
@@ -390,184 +406,197 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
     const int my_c = t / CPT;
     const int my_j0 = (t - my_c * CPT) * RB;
 
-    for (int64_t tile = blockIdx.x; tile < A.tiles; tile += gridDim.x) {
-        const int64_t c0 = tile * TI;
-        __syncthreads();                // prior tile's bin phase done
-        if (t < TI) {
-            const int64_t cg = c0 + t;
-            int live = 0;
-            if (cg < A.n_inner) {
-                const int64_t iyl = cg / A.nzh;
-                const int64_t iz = cg - iyl * A.nzh;
-                const double fy = freq_full(iyl + A.y_off, A.n1);
-                const double fz = freq_half(iz, A.n2);
-                const double ky = fy * A.k0y;
-                const double kz = fz * A.k0z;
-                cky2[t] = ky * ky;
-                ckz2[t] = kz * kz;
-                ckyl[t] = ky * A.losy;
-                ckzl[t] = kz * A.losz;
-                cw2[t] = (fz > 0.0);
-                czl[t] = (fy == 0.0 && fz == 0.0);
-                if (comp_on) {
-                    ccy[t] = nbk_comp_factor1(A.win1, A.interl1,
-                                 2.0 * M_PI * fy / (double)A.n1);
-                    ccz[t] = nbk_comp_factor1(A.win1, A.interl1,
-                                 2.0 * M_PI * fz / (double)A.n2);
-                } else {
-                    ccy[t] = 1.0;
-                    ccz[t] = 1.0;
-                }
-                // min over the line is at kx = 0 exactly, where
-                // fl(0 + ky2) = ky2; fl-addition is monotone, so
-                // ky2 + kz2 >= k2last skips every element of the column
-                live = (cky2[t] + ckz2[t] < k2last) ? 1 : 0;
-            }
-            cskip[t] = (unsigned char)(!live);
-        }
-        if (t == 0) s_nlive = 0;
-        __syncthreads();
-        if (t < TI && !cskip[t]) atomicAdd(&s_nlive, 1);
-        __syncthreads();
-        if (s_nlive == 0) continue;
-
-        // register-staged loads (see kfft_c_strided: the pass is
-        // latency-bound, every load of a round must issue before the
-        // first LDS store)
-        const cdouble* g = (const cdouble*)data + c0;
-        const int total = n0 * TI;
-        for (int base = 0; base < total; base += T * 8) {
-            cdouble r[8];
-            #pragma unroll
-            for (int q = 0; q < 8; q++) {
-                const int w = base + q * T + t;
-                if (w < total) {
-                    const int c = w % TI;
-                    r[q] = cskip[c]
-                        ? cdouble{0.0, 0.0}
-                        : g[(int64_t)(w / TI) * A.n_inner + c];
-                }
-            }
-            #pragma unroll
-            for (int q = 0; q < 8; q++) {
-                const int w = base + q * T + t;
-                if (w < total)
-                    buf[bitrev(w / TI, bits) * W + (w % TI)] = r[q];
-            }
-        }
-        __syncthreads();
-
-        if (PHASES & 1)
-            lds_fft4<false>(buf, n0, TI, W, table);
-
-        // run-merged bin phase: this thread walks j = my_j0 .. +RB-1 of
-        // column my_c, accumulating same-bin neighbours in registers
-        // and warm-starting the digitize (|k| varies slowly along the
-        // line) — the kbin_run recipe, reading LDS instead of HBM
-        if ((PHASES & 2) && my_c < TI && !cskip[my_c] && my_j0 < n0) {
-            const int c = my_c;
-            const double ky2 = cky2[c];
-            const double kz2 = ckz2[c];
-            const double kyl = ckyl[c];
-            const double kzl = ckzl[c];
-            const double ccyv = ccy[c];
-            const double cczv = ccz[c];
-            const bool nonsingular = cw2[c];
-            const double wgt = nonsingular ? 2.0 : 1.0;
-            const bool col_zero = czl[c];
-            const int jend = (my_j0 + RB < n0) ? my_j0 + RB : n0;
-
-            int cbin = -1, cbx = -1, cbmu = -1;
-            double ak = 0.0, amu = 0.0, aw = 0.0;
-            cdouble ay[NBK_MAX_ELL];
-            for (int e = 0; e < A.nell; e++) ay[e] = {0.0, 0.0};
-
-            for (int j = my_j0; j < jend; j++) {
-                const double fx = freq_full(j, A.n0);
-                const double kx = fx * A.k0x;
-                const double k2 = (kx * kx + ky2) + kz2;
-                if (k2 >= k2last) continue;
-                const double kmag = sqrt(k2);
-                double mu = (kx * A.losx + kyl) + kzl;
-                mu = (kmag == 0.0) ? 0.0 : mu / kmag;
-
-                int bx;
-                if (cbx >= 0
-                    && (cbx == 0 || ke[cbx - 1] <= k2)
-                    && (cbx == A.nx_edges || k2 < ke[cbx]))
-                    bx = cbx;
-                else
-                    bx = digx(ke, A.nx_edges, k2);
-                int bmu;
-                if (cbmu >= 0
-                    && (cbmu == 0 || me[cbmu - 1] <= mu)
-                    && (cbmu == A.nmu_edges || mu < me[cbmu]))
-                    bmu = cbmu;
-                else
-                    bmu = digx(me, A.nmu_edges, mu);
-                const int bin = bx * (A.nmu_edges + 1) + bmu;
-
-                cdouble a = buf[j * W + c];
-                if (comp_on)
-                    a = cscale(a, (cxv[j] * ccyv) * cczv);
-                const cdouble p = cmul(a, cconj(a));
-                cdouble v = {p.re * A.volume, p.im * A.volume};
-                if (A.clear_zero && col_zero && j == 0)
-                    v = {0.0, 0.0};
-
-                if (bin != cbin) {
-                    if (cbin >= 0) {
-                        atomicAdd(&h[cbin], ak);
-                        atomicAdd(&h[NB + cbin], amu);
-                        atomicAdd(&h[2 * NB + cbin], aw);
-                        for (int e = 0; e < A.nell; e++) {
-                            atomicAdd(&h[(3 + 2 * e) * NB + cbin],
-                                      ay[e].re);
-                            atomicAdd(&h[(3 + 2 * e + 1) * NB + cbin],
-                                      ay[e].im);
-                        }
-                    }
-                    cbin = bin; cbx = bx; cbmu = bmu;
-                    ak = 0.0; amu = 0.0; aw = 0.0;
-                    for (int e = 0; e < A.nell; e++) ay[e] = {0.0, 0.0};
-                }
-                ak += kmag * wgt;
-                amu += mu * wgt;
-                aw += wgt;
-
-                double Pm1 = 0.0, P = 1.0;
-                int e = 0;
-                for (int l = 0; e < A.nell; l++) {
-                    if (l > 0) {
-                        const double Pn = ((2 * l - 1) * mu * P
-                                           - (l - 1) * Pm1) / l;
-                        Pm1 = P;
-                        P = Pn;
-                    }
-                    if (l == A.ells[e]) {
-                        cdouble wy = cscale(v, P);
-                        if (nonsingular) {
-                            if (l % 2) wy = {0.0, 2.0 * wy.im};
-                            else wy = {2.0 * wy.re, 0.0};
-                        }
-                        wy = cscale(wy, 2.0 * l + 1.0);
-                        ay[e].re += wy.re;
-                        ay[e].im += wy.im;
-                        e++;
-                    }
-                }
-            }
-            if (cbin >= 0) {
-                atomicAdd(&h[cbin], ak);
-                atomicAdd(&h[NB + cbin], amu);
-                atomicAdd(&h[2 * NB + cbin], aw);
-                for (int e = 0; e < A.nell; e++) {
-                    atomicAdd(&h[(3 + 2 * e) * NB + cbin], ay[e].re);
-                    atomicAdd(&h[(3 + 2 * e + 1) * NB + cbin], ay[e].im);
-                }
-            }
-        }
+    // ---- pipeline stages (macros so the prologue and loop share the
+    //      exact code) --------------------------------------------------
+#define XBIN_SETUP(tile_, q_)                                              \
+    if (t < 64) {                                                          \
+        int live_ = 0;                                                     \
+        if (t < TI) {                                                      \
+            const int64_t cg_ = (int64_t)(tile_) * TI + t;                 \
+            if ((tile_) < A.tiles && cg_ < A.n_inner) {                    \
+                const int64_t iyl_ = cg_ / A.nzh;                          \
+                const int64_t iz_ = cg_ - iyl_ * A.nzh;                    \
+                const double fy_ = freq_full(iyl_ + A.y_off, A.n1);        \
+                const double fz_ = freq_half(iz_, A.n2);                   \
+                const double ky_ = fy_ * A.k0y;                            \
+                const double kz_ = fz_ * A.k0z;                            \
+                cky2[q_][t] = ky_ * ky_;                                   \
+                ckz2[q_][t] = kz_ * kz_;                                   \
+                ckyl[q_][t] = ky_ * A.losy;                                \
+                ckzl[q_][t] = kz_ * A.losz;                                \
+                cw2[q_][t] = (fz_ > 0.0);                                  \
+                czl[q_][t] = (fy_ == 0.0 && fz_ == 0.0);                   \
+                if (comp_on) {                                             \
+                    ccy[q_][t] = nbk_comp_factor1(A.win1, A.interl1,       \
+                        2.0 * M_PI * fy_ / (double)A.n1);                  \
+                    ccz[q_][t] = nbk_comp_factor1(A.win1, A.interl1,       \
+                        2.0 * M_PI * fz_ / (double)A.n2);                  \
+                } else {                                                   \
+                    ccy[q_][t] = 1.0;                                      \
+                    ccz[q_][t] = 1.0;                                      \
+                }                                                          \
+                /* the line's k2 minimum is at kx = 0 exactly, where   */  \
+                /* fl(0 + ky2) = ky2; fl-addition is monotone           */  \
+                live_ = (cky2[q_][t] + ckz2[q_][t] < k2last) ? 1 : 0;      \
+            }                                                              \
+            cskip[q_][t] = (unsigned char)(!live_);                        \
+        }                                                                  \
+        const unsigned long long m_ = __ballot(live_);                     \
+        if (t == 0) s_nlive[q_] = __popcll(m_);                            \
     }
+
+#define XBIN_LOAD(tile_, q_)                                               \
+    {                                                                      \
+        const cdouble* g_ = (const cdouble*)data + (int64_t)(tile_) * TI;  \
+        _Pragma("unroll")                                                  \
+        for (int q = 0; q < 4; q++) {                                      \
+            const int w_ = q * T + t;                                      \
+            if (w_ < n0 * TI) {                                            \
+                const int c_ = w_ % TI;                                    \
+                r[q] = cskip[q_][c_]                                       \
+                    ? cdouble{0.0, 0.0}                                    \
+                    : g_[(int64_t)(w_ / TI) * A.n_inner + c_];             \
+            }                                                              \
+        }                                                                  \
+    }
+
+#define XBIN_STORE                                                         \
+    _Pragma("unroll")                                                      \
+    for (int q = 0; q < 4; q++) {                                          \
+        const int w_ = q * T + t;                                          \
+        if (w_ < n0 * TI)                                                  \
+            buf[bitrev(w_ / TI, bits) * W + (w_ % TI)] = r[q];             \
+    }
+
+    cdouble r[4] = {};
+    int64_t tile = blockIdx.x;
+    int p = 0;
+    XBIN_SETUP(tile, 0)
+    __syncthreads();            // publishes h/ke/me/cxv and consts[0]
+    if (tile < A.tiles)
+        XBIN_LOAD(tile, 0)
+
+    for (; tile < A.tiles; tile += gridDim.x, p ^= 1) {
+        XBIN_STORE              // regs -> buf (prev bin done: end sync)
+        const int64_t nxt = tile + gridDim.x;
+        XBIN_SETUP(nxt, p ^ 1)
+        __syncthreads();        // buf ready; consts[p^1] published
+        if (nxt < A.tiles)
+            XBIN_LOAD(nxt, p ^ 1)   // in flight under FFT + bin below
+
+        if (s_nlive[p] > 0) {
+            if (PHASES & 1)
+                lds_fft4<false>(buf, n0, TI, W, table);
+
+            // run-merged bin phase: walk j = my_j0 .. +RB-1 of column
+            // my_c, merging same-bin neighbours in registers and
+            // warm-starting the digitize (the kbin_run recipe, reading
+            // LDS instead of HBM)
+            if ((PHASES & 2) && my_c < TI && !cskip[p][my_c]
+                && my_j0 < n0) {
+                const int c = my_c;
+                const double ky2 = cky2[p][c];
+                const double kz2 = ckz2[p][c];
+                const double kyl = ckyl[p][c];
+                const double kzl = ckzl[p][c];
+                const double ccyv = ccy[p][c];
+                const double cczv = ccz[p][c];
+                const bool nonsingular = cw2[p][c];
+                const double wgt = nonsingular ? 2.0 : 1.0;
+                const bool col_zero = czl[p][c];
+                const int jend = (my_j0 + RB < n0) ? my_j0 + RB : n0;
+
+                int cbin = -1, cbx = -1, cbmu = -1;
+                double ak = 0.0, amu = 0.0, aw = 0.0;
+                double ay[MAXE];
+                for (int e = 0; e < MAXE; e++) ay[e] = 0.0;
+
+                for (int j = my_j0; j < jend; j++) {
+                    const double fx = freq_full(j, A.n0);
+                    const double kx = fx * A.k0x;
+                    const double k2 = (kx * kx + ky2) + kz2;
+                    if (k2 >= k2last) continue;
+                    const double kmag = sqrt(k2);
+                    double mu = (kx * A.losx + kyl) + kzl;
+                    mu = (kmag == 0.0) ? 0.0 : mu / kmag;
+
+                    int bx;
+                    if (cbx >= 0
+                        && (cbx == 0 || ke[cbx - 1] <= k2)
+                        && (cbx == A.nx_edges || k2 < ke[cbx]))
+                        bx = cbx;
+                    else
+                        bx = digx(ke, A.nx_edges, k2);
+                    int bmu;
+                    if (cbmu >= 0
+                        && (cbmu == 0 || me[cbmu - 1] <= mu)
+                        && (cbmu == A.nmu_edges || mu < me[cbmu]))
+                        bmu = cbmu;
+                    else
+                        bmu = digx(me, A.nmu_edges, mu);
+                    const int bin = bx * (A.nmu_edges + 1) + bmu;
+
+                    cdouble a = buf[j * W + c];
+                    if (comp_on)
+                        a = cscale(a, (cxv[j] * ccyv) * cczv);
+                    // auto power: Im(a conj(a)) == 0 exactly
+                    double vre = (a.re * a.re - a.im * (-a.im))
+                                 * A.volume;
+                    if (A.clear_zero && col_zero && j == 0)
+                        vre = 0.0;
+
+                    if (bin != cbin) {
+                        if (cbin >= 0) {
+                            atomicAdd(&h[cbin], ak);
+                            atomicAdd(&h[NB + cbin], amu);
+                            atomicAdd(&h[2 * NB + cbin], aw);
+                            for (int e = 0; e < A.nell && e < MAXE; e++)
+                                atomicAdd(&h[(3 + 2 * e) * NB + cbin],
+                                          ay[e]);
+                        }
+                        cbin = bin; cbx = bx; cbmu = bmu;
+                        ak = 0.0; amu = 0.0; aw = 0.0;
+                        for (int e = 0; e < MAXE; e++) ay[e] = 0.0;
+                    }
+                    ak += kmag * wgt;
+                    amu += mu * wgt;
+                    aw += wgt;
+
+                    double Pm1 = 0.0, P = 1.0;
+                    int e = 0;
+                    for (int l = 0; e < A.nell && e < MAXE; l++) {
+                        if (l > 0) {
+                            const double Pn = ((2 * l - 1) * mu * P
+                                               - (l - 1) * Pm1) / l;
+                            Pm1 = P;
+                            P = Pn;
+                        }
+                        if (l == A.ells[e]) {
+                            double wr = vre * P;
+                            if (nonsingular) {
+                                // conjugate-pair parity: odd ell keeps
+                                // only the (zero) imaginary part
+                                wr = (l % 2) ? 0.0 : 2.0 * wr;
+                            }
+                            ay[e] += wr * (2.0 * l + 1.0);
+                            e++;
+                        }
+                    }
+                }
+                if (cbin >= 0) {
+                    atomicAdd(&h[cbin], ak);
+                    atomicAdd(&h[NB + cbin], amu);
+                    atomicAdd(&h[2 * NB + cbin], aw);
+                    for (int e = 0; e < A.nell && e < MAXE; e++)
+                        atomicAdd(&h[(3 + 2 * e) * NB + cbin], ay[e]);
+                }
+            }
+        }
+        __syncthreads();        // bin done before the next tile's store
+    }
+#undef XBIN_SETUP
+#undef XBIN_LOAD
+#undef XBIN_STORE
     __syncthreads();
     for (int i = t; i < NB * nfields; i += T)
         if (h[i] != 0.0) atomicAdd(&gout[i], h[i]);
@@ -740,10 +769,18 @@ extern "C" int nbk_fft_x_bin_f64(const double* data,
         if (TI0 < 1 || TI0 > 16) TI0 = 4;
     }
     int TI = TI0;
+    // the pipeline carries a tile in 4 registers per thread
+    while (TI > 1 && (int64_t)nmesh[0] * TI > 4 * 1024)
+        TI >>= 1;
     // buf pitch is TI+1 (bank-spread padding for the bin phase's runs)
     while (TI > 1 && fixed + (size_t)nmesh[0] * (TI + 1) * sizeof(cdouble)
                      > 160 * 1024)
         TI >>= 1;
+    if ((int64_t)nmesh[0] * TI > 4 * 1024) {
+        NBK_SET_ERR("nbk_fft_x_bin_f64: n0 too large for the register "
+                    "pipeline — use the unfused path");
+        return NBK_ERR_UNSUPPORTED;
+    }
     const size_t shmem = fixed
         + (size_t)nmesh[0] * (TI + 1) * sizeof(cdouble);
     if (shmem > 160 * 1024) {
@@ -774,39 +811,51 @@ extern "C" int nbk_fft_x_bin_f64(const double* data,
         if (phases < 0 || phases > 3) phases = 3;
         phases |= 4;            // mark initialized
     }
-    const void* fn =
-        (phases & 3) == 3 ? reinterpret_cast<const void*>(&kxfft_bin<3>)
-        : (phases & 3) == 1 ? reinterpret_cast<const void*>(&kxfft_bin<1>)
-        : (phases & 3) == 2 ? reinterpret_cast<const void*>(&kxfft_bin<2>)
-        : reinterpret_cast<const void*>(&kxfft_bin<0>);
+    // MAXE sizes the per-thread multipole registers: <.,1> serves every
+    // plain 1d/2d run, <.,8> any poles request; the decomposition
+    // phases only exist at MAXE = 8
+    const void* fn;
+    if ((phases & 3) == 3)
+        fn = nell <= 1
+            ? reinterpret_cast<const void*>(&kxfft_bin<3, 1>)
+            : reinterpret_cast<const void*>(&kxfft_bin<3, 8>);
+    else if ((phases & 3) == 1)
+        fn = reinterpret_cast<const void*>(&kxfft_bin<1, 8>);
+    else if ((phases & 3) == 2)
+        fn = reinterpret_cast<const void*>(&kxfft_bin<2, 8>);
+    else
+        fn = reinterpret_cast<const void*>(&kxfft_bin<0, 8>);
     static size_t raised = 0;
     if (shmem > 64 * 1024 && shmem > raised) {
         (void)hipFuncSetAttribute(
             fn, hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);
         raised = shmem;
     }
-    switch (phases & 3) {
-    case 3:
-        hipLaunchKernelGGL(kxfft_bin<3>, dim3((uint32_t)g), dim3(1024),
-                           shmem, (hipStream_t)stream, data, A, k2edges,
-                           muedges, (const cdouble*)table, out_sums);
-        break;
-    case 1:
-        hipLaunchKernelGGL(kxfft_bin<1>, dim3((uint32_t)g), dim3(1024),
-                           shmem, (hipStream_t)stream, data, A, k2edges,
-                           muedges, (const cdouble*)table, out_sums);
-        break;
-    case 2:
-        hipLaunchKernelGGL(kxfft_bin<2>, dim3((uint32_t)g), dim3(1024),
-                           shmem, (hipStream_t)stream, data, A, k2edges,
-                           muedges, (const cdouble*)table, out_sums);
-        break;
-    default:
-        hipLaunchKernelGGL(kxfft_bin<0>, dim3((uint32_t)g), dim3(1024),
-                           shmem, (hipStream_t)stream, data, A, k2edges,
-                           muedges, (const cdouble*)table, out_sums);
-        break;
-    }
+    if ((phases & 3) == 3 && nell <= 1)
+        hipLaunchKernelGGL((kxfft_bin<3, 1>), dim3((uint32_t)g),
+                           dim3(1024), shmem, (hipStream_t)stream, data,
+                           A, k2edges, muedges, (const cdouble*)table,
+                           out_sums);
+    else if ((phases & 3) == 3)
+        hipLaunchKernelGGL((kxfft_bin<3, 8>), dim3((uint32_t)g),
+                           dim3(1024), shmem, (hipStream_t)stream, data,
+                           A, k2edges, muedges, (const cdouble*)table,
+                           out_sums);
+    else if ((phases & 3) == 1)
+        hipLaunchKernelGGL((kxfft_bin<1, 8>), dim3((uint32_t)g),
+                           dim3(1024), shmem, (hipStream_t)stream, data,
+                           A, k2edges, muedges, (const cdouble*)table,
+                           out_sums);
+    else if ((phases & 3) == 2)
+        hipLaunchKernelGGL((kxfft_bin<2, 8>), dim3((uint32_t)g),
+                           dim3(1024), shmem, (hipStream_t)stream, data,
+                           A, k2edges, muedges, (const cdouble*)table,
+                           out_sums);
+    else
+        hipLaunchKernelGGL((kxfft_bin<0, 8>), dim3((uint32_t)g),
+                           dim3(1024), shmem, (hipStream_t)stream, data,
+                           A, k2edges, muedges, (const cdouble*)table,
+                           out_sums);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }
