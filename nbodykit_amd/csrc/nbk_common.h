@@ -73,6 +73,17 @@ __device__ __forceinline__ double freq_half(int64_t g, int64_t n) {
     return (double)(g == n / 2 ? -(n / 2) : g);
 }
 
+// Block barrier that waits only on LDS traffic (lgkmcnt), NOT on
+// in-flight global loads (vmcnt) — plain __syncthreads emits
+// s_waitcnt vmcnt(0) before s_barrier, which would force any
+// software-pipelined global loads to complete at the first barrier
+// they cross.  Safe whenever the only cross-thread state the barrier
+// orders is LDS (all the FFT kernels: global reads are consumed
+// through registers, whose use carries its own vmcnt wait).
+__device__ __forceinline__ void nbk_sync_lds() {
+    asm volatile("s_waitcnt lgkmcnt(0)\n\ts_barrier" ::: "memory");
+}
+
 // bit-reversal for the in-tile FFT (shared with nbk_fft.hip's local copy)
 __device__ __forceinline__ int nbk_bitrev(int j, int bits) {
     return (int)(__brev((unsigned)j) >> (32 - bits));

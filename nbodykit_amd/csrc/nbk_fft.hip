@@ -97,7 +97,7 @@ __device__ void lds_fft4(cdouble* buf, int m, int TI, int PITCH,
             buf[i0] = cadd(u, v);
             buf[i1] = csub(u, v);
         }
-        __syncthreads();
+        nbk_sync_lds();
         len = 4;
     }
     for (; 2 * len <= m; len <<= 2) {
@@ -132,7 +132,7 @@ __device__ void lds_fft4(cdouble* buf, int m, int TI, int PITCH,
             buf[base + s] = cadd(e1, io1);
             buf[base + 3 * s] = csub(e1, io1);
         }
-        __syncthreads();
+        nbk_sync_lds();
     }
 }
 
@@ -472,7 +472,7 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
     int64_t tile = blockIdx.x;
     int p = 0;
     XBIN_SETUP(tile, 0)
-    __syncthreads();            // publishes h/ke/me/cxv and consts[0]
+    nbk_sync_lds();             // publishes h/ke/me/cxv and consts[0]
     if (tile < A.tiles)
         XBIN_LOAD(tile, 0)
 
@@ -480,7 +480,12 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
         XBIN_STORE              // regs -> buf (prev bin done: end sync)
         const int64_t nxt = tile + gridDim.x;
         XBIN_SETUP(nxt, p ^ 1)
-        __syncthreads();        // buf ready; consts[p^1] published
+        nbk_sync_lds();         // buf ready; consts[p^1] published
+                                // (LDS-only wait: the loads issued for
+                                // THIS tile were consumed by the store
+                                // above; the next tile's loads, issued
+                                // below, must survive every barrier
+                                // until the next store)
         if (nxt < A.tiles)
             XBIN_LOAD(nxt, p ^ 1)   // in flight under FFT + bin below
 
@@ -592,12 +597,12 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
                 }
             }
         }
-        __syncthreads();        // bin done before the next tile's store
+        nbk_sync_lds();         // bin done before the next tile's store
     }
 #undef XBIN_SETUP
 #undef XBIN_LOAD
 #undef XBIN_STORE
-    __syncthreads();
+    nbk_sync_lds();
     for (int i = t; i < NB * nfields; i += T)
         if (h[i] != 0.0) atomicAdd(&gout[i], h[i]);
 }
