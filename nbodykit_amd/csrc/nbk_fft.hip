@@ -468,17 +468,34 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
             buf[bitrev(w_ / TI, bits) * W + (w_ % TI)] = r[q];             \
     }
 
+    // CONTIGUOUS tile chunks per block (not grid-strided): adjacent
+    // tiles are adjacent (y, z) columns, whose elements land in the
+    // same k bins for many tiles in a row — the run state below
+    // (cbin + partial sums) carries ACROSS tiles, so the contended
+    // LDS histogram atomics fire on real bin transitions only
+    // (grid-strided mapping re-flushed every tile: ~8K atomics/tile
+    // dominated the bin phase)
+    const int64_t chunk = (A.tiles + gridDim.x - 1) / gridDim.x;
+    const int64_t tend = ((int64_t)(blockIdx.x + 1) * chunk < A.tiles)
+        ? (int64_t)(blockIdx.x + 1) * chunk : A.tiles;
     cdouble r[4] = {};
-    int64_t tile = blockIdx.x;
+    int64_t tile = (int64_t)blockIdx.x * chunk;
     int p = 0;
+
+    // cross-tile run state of this thread's (column-slot, j-window)
+    int cbin = -1, cbx = -1, cbmu = -1;
+    double ak = 0.0, amu = 0.0, aw = 0.0;
+    double ay[MAXE];
+    for (int e = 0; e < MAXE; e++) ay[e] = 0.0;
+
     XBIN_SETUP(tile, 0)
     nbk_sync_lds();             // publishes h/ke/me/cxv and consts[0]
-    if (tile < A.tiles)
+    if (tile < tend)
         XBIN_LOAD(tile, 0)
 
-    for (; tile < A.tiles; tile += gridDim.x, p ^= 1) {
+    for (; tile < tend; tile++, p ^= 1) {
         XBIN_STORE              // regs -> buf (prev bin done: end sync)
-        const int64_t nxt = tile + gridDim.x;
+        const int64_t nxt = tile + 1;
         XBIN_SETUP(nxt, p ^ 1)
         nbk_sync_lds();         // buf ready; consts[p^1] published
                                 // (LDS-only wait: the loads issued for
@@ -486,7 +503,7 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
                                 // above; the next tile's loads, issued
                                 // below, must survive every barrier
                                 // until the next store)
-        if (nxt < A.tiles)
+        if (nxt < tend)
             XBIN_LOAD(nxt, p ^ 1)   // in flight under FFT + bin below
 
         if (s_nlive[p] > 0) {
@@ -510,11 +527,6 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
                 const double wgt = nonsingular ? 2.0 : 1.0;
                 const bool col_zero = czl[p][c];
                 const int jend = (my_j0 + RB < n0) ? my_j0 + RB : n0;
-
-                int cbin = -1, cbx = -1, cbmu = -1;
-                double ak = 0.0, amu = 0.0, aw = 0.0;
-                double ay[MAXE];
-                for (int e = 0; e < MAXE; e++) ay[e] = 0.0;
 
                 for (int j = my_j0; j < jend; j++) {
                     const double fx = freq_full(j, A.n0);
@@ -588,16 +600,18 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
                         }
                     }
                 }
-                if (cbin >= 0) {
-                    atomicAdd(&h[cbin], ak);
-                    atomicAdd(&h[NB + cbin], amu);
-                    atomicAdd(&h[2 * NB + cbin], aw);
-                    for (int e = 0; e < A.nell && e < MAXE; e++)
-                        atomicAdd(&h[(3 + 2 * e) * NB + cbin], ay[e]);
-                }
+                // no per-tile flush: cbin and its partial sums carry
+                // into the next (adjacent) tile's columns
             }
         }
         nbk_sync_lds();         // bin done before the next tile's store
+    }
+    if (cbin >= 0) {            // final flush of the cross-tile run
+        atomicAdd(&h[cbin], ak);
+        atomicAdd(&h[NB + cbin], amu);
+        atomicAdd(&h[2 * NB + cbin], aw);
+        for (int e = 0; e < A.nell && e < MAXE; e++)
+            atomicAdd(&h[(3 + 2 * e) * NB + cbin], ay[e]);
     }
 #undef XBIN_SETUP
 #undef XBIN_LOAD
