@@ -399,14 +399,12 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
     // LDS copy is not published until the first sync)
     const double k2last = k2edges_g[A.nx_edges - 1];
 
-    // bin-phase mapping: column-major linear passes — a wave's 64 lanes
-    // cover 64 CONSECUTIVE j of (mostly) one column, so equal-bin lanes
-    // are contiguous and a head-flag segmented wave scan merges them
-    // before any histogram atomic; the LDS tile reads stride W*16 =
-    // 80 B between lanes (8-way banked, vs 32-way for the blocked
-    // per-thread-run mapping this replaces)
-    const int lane = t & 63;
-    const int npass = (n0 * TI + T - 1) / T;
+    // bin-phase mapping: contiguous j-runs per thread at wave-uniform
+    // column c (kbin_run-style register merging + warm-start digitize)
+    const int CPT = T / TI;                 // threads per column
+    const int RB = (n0 + CPT - 1) / CPT;    // j-run length per thread
+    const int my_c = t / CPT;
+    const int my_j0 = (t - my_c * CPT) * RB;
 
     // ---- pipeline stages (macros so the prologue and loop share the
     //      exact code) --------------------------------------------------
@@ -471,14 +469,24 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
     }
 
     // CONTIGUOUS tile chunks per block (not grid-strided): adjacent
-    // tiles are adjacent (y, z) columns — better L2/TLB locality for
-    // the pipelined strided loads
+    // tiles are adjacent (y, z) columns, whose elements land in the
+    // same k bins for many tiles in a row — the run state below
+    // (cbin + partial sums) carries ACROSS tiles, so the contended
+    // LDS histogram atomics fire on real bin transitions only
+    // (grid-strided mapping re-flushed every tile: ~8K atomics/tile
+    // dominated the bin phase)
     const int64_t chunk = (A.tiles + gridDim.x - 1) / gridDim.x;
     const int64_t tend = ((int64_t)(blockIdx.x + 1) * chunk < A.tiles)
         ? (int64_t)(blockIdx.x + 1) * chunk : A.tiles;
     cdouble r[4] = {};
     int64_t tile = (int64_t)blockIdx.x * chunk;
     int p = 0;
+
+    // cross-tile run state of this thread's (column-slot, j-window)
+    int cbin = -1, cbx = -1, cbmu = -1;
+    double ak = 0.0, amu = 0.0, aw = 0.0;
+    double ay[MAXE];
+    for (int e = 0; e < MAXE; e++) ay[e] = 0.0;
 
     XBIN_SETUP(tile, 0)
     nbk_sync_lds();             // publishes h/ke/me/cxv and consts[0]
@@ -502,114 +510,108 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
             if (PHASES & 1)
                 lds_fft4<false>(buf, n0, TI, W, table);
 
-            // wave-segmented bin phase: pass q covers T consecutive
-            // column-major elements e = q*T + t (c = e/n0, j = e mod
-            // n0), so a wave's 64 lanes hold 64 consecutive j of one
-            // column (straddling a column boundary is fine — merging
-            // is keyed on the bin index alone).  Equal-bin lanes are
-            // contiguous, and a head-flag segmented inclusive scan
-            // collapses them so only segment TAILS touch the LDS
-            // histograms (~10x fewer contended atomics than
-            // per-thread flushing).  Summation order differs from the
-            // unfused kernel at the last ulp only (covered by the
-            // 1e-10 fused-vs-unfused tolerance); Nsum stays exact
-            // (small integers in f64 add exactly in any order).
-            if (PHASES & 2) {
-                for (int q = 0; q < npass; q++) {
-                    const int e0 = q * T + t;
-                    int bin = -1;
-                    bool nonsingular = false;
-                    double kmag = 0.0, mu = 0.0, wgt = 0.0, vre = 0.0;
-                    const int c = e0 >> bits;        // n0 = 1 << bits
-                    const int j = e0 & (n0 - 1);
-                    if (c < TI && !cskip[p][c]) {
-                        const double fx = freq_full(j, A.n0);
-                        const double kx = fx * A.k0x;
-                        const double k2 = (kx * kx + cky2[p][c])
-                                          + ckz2[p][c];
-                        if (k2 < k2last) {
-                            kmag = sqrt(k2);
-                            mu = (kx * A.losx + ckyl[p][c]) + ckzl[p][c];
-                            mu = (kmag == 0.0) ? 0.0 : mu / kmag;
-                            nonsingular = cw2[p][c];
-                            wgt = nonsingular ? 2.0 : 1.0;
-                            const int bx = digx(ke, A.nx_edges, k2);
-                            const int bmu = digx(me, A.nmu_edges, mu);
-                            bin = bx * (A.nmu_edges + 1) + bmu;
-                            cdouble a = buf[j * W + c];
-                            if (comp_on)
-                                a = cscale(a, (cxv[j] * ccy[p][c])
-                                              * ccz[p][c]);
-                            // auto power: Im(a conj(a)) == 0 exactly
-                            vre = (a.re * a.re - a.im * (-a.im))
-                                  * A.volume;
-                            if (A.clear_zero && czl[p][c] && j == 0)
-                                vre = 0.0;
-                        }
-                    }
-                    double vkm = kmag * wgt;
-                    double vmu = mu * wgt;
-                    double vw = wgt;
-                    double vy[MAXE];
-                    {
-                        double Pm1 = 0.0, P = 1.0;
-                        int e = 0;
-                        for (int l = 0; e < A.nell && e < MAXE; l++) {
-                            if (l > 0) {
-                                const double Pn = ((2 * l - 1) * mu * P
-                                                   - (l - 1) * Pm1) / l;
-                                Pm1 = P;
-                                P = Pn;
-                            }
-                            if (l == A.ells[e]) {
-                                double wr = vre * P;
-                                if (nonsingular)
-                                    wr = (l % 2) ? 0.0 : 2.0 * wr;
-                                vy[e] = wr * (2.0 * l + 1.0);
-                                e++;
-                            }
-                        }
-                        for (; e < MAXE; e++) vy[e] = 0.0;
-                    }
+            // run-merged bin phase: walk j = my_j0 .. +RB-1 of column
+            // my_c, merging same-bin neighbours in registers and
+            // warm-starting the digitize (the kbin_run recipe, reading
+            // LDS instead of HBM)
+            if ((PHASES & 2) && my_c < TI && !cskip[p][my_c]
+                && my_j0 < n0) {
+                const int c = my_c;
+                const double ky2 = cky2[p][c];
+                const double kz2 = ckz2[p][c];
+                const double kyl = ckyl[p][c];
+                const double kzl = ckzl[p][c];
+                const double ccyv = ccy[p][c];
+                const double cczv = ccz[p][c];
+                const bool nonsingular = cw2[p][c];
+                const double wgt = nonsingular ? 2.0 : 1.0;
+                const bool col_zero = czl[p][c];
+                const int jend = (my_j0 + RB < n0) ? my_j0 + RB : n0;
 
-                    // head-flag segmented inclusive sum across the wave
-                    const int binup = __shfl_up(bin, 1, 64);
-                    int sid = (lane == 0 || binup != bin) ? lane : 0;
-                    #pragma unroll
-                    for (int d = 1; d < 64; d <<= 1) {
-                        const int u = __shfl_up(sid, d, 64);
-                        if (lane >= d && u > sid) sid = u;
-                    }
-                    #pragma unroll
-                    for (int d = 1; d < 64; d <<= 1) {
-                        const double ukm = __shfl_up(vkm, d, 64);
-                        const double umu = __shfl_up(vmu, d, 64);
-                        const double uw = __shfl_up(vw, d, 64);
-                        double uy[MAXE];
-                        for (int e = 0; e < MAXE; e++)
-                            uy[e] = __shfl_up(vy[e], d, 64);
-                        if (lane >= d && lane - d >= sid) {
-                            vkm += ukm;
-                            vmu += umu;
-                            vw += uw;
-                            for (int e = 0; e < MAXE; e++)
-                                vy[e] += uy[e];
+                for (int j = my_j0; j < jend; j++) {
+                    const double fx = freq_full(j, A.n0);
+                    const double kx = fx * A.k0x;
+                    const double k2 = (kx * kx + ky2) + kz2;
+                    if (k2 >= k2last) continue;
+                    const double kmag = sqrt(k2);
+                    double mu = (kx * A.losx + kyl) + kzl;
+                    mu = (kmag == 0.0) ? 0.0 : mu / kmag;
+
+                    int bx;
+                    if (cbx >= 0
+                        && (cbx == 0 || ke[cbx - 1] <= k2)
+                        && (cbx == A.nx_edges || k2 < ke[cbx]))
+                        bx = cbx;
+                    else
+                        bx = digx(ke, A.nx_edges, k2);
+                    int bmu;
+                    if (cbmu >= 0
+                        && (cbmu == 0 || me[cbmu - 1] <= mu)
+                        && (cbmu == A.nmu_edges || mu < me[cbmu]))
+                        bmu = cbmu;
+                    else
+                        bmu = digx(me, A.nmu_edges, mu);
+                    const int bin = bx * (A.nmu_edges + 1) + bmu;
+
+                    cdouble a = buf[j * W + c];
+                    if (comp_on)
+                        a = cscale(a, (cxv[j] * ccyv) * cczv);
+                    // auto power: Im(a conj(a)) == 0 exactly
+                    double vre = (a.re * a.re - a.im * (-a.im))
+                                 * A.volume;
+                    if (A.clear_zero && col_zero && j == 0)
+                        vre = 0.0;
+
+                    if (bin != cbin) {
+                        if (cbin >= 0) {
+                            atomicAdd(&h[cbin], ak);
+                            atomicAdd(&h[NB + cbin], amu);
+                            atomicAdd(&h[2 * NB + cbin], aw);
+                            for (int e = 0; e < A.nell && e < MAXE; e++)
+                                atomicAdd(&h[(3 + 2 * e) * NB + cbin],
+                                          ay[e]);
                         }
+                        cbin = bin; cbx = bx; cbmu = bmu;
+                        ak = 0.0; amu = 0.0; aw = 0.0;
+                        for (int e = 0; e < MAXE; e++) ay[e] = 0.0;
                     }
-                    const int bindn = __shfl_down(bin, 1, 64);
-                    const bool tail = (lane == 63) || (bindn != bin);
-                    if (tail && bin >= 0) {
-                        atomicAdd(&h[bin], vkm);
-                        atomicAdd(&h[NB + bin], vmu);
-                        atomicAdd(&h[2 * NB + bin], vw);
-                        for (int e = 0; e < A.nell && e < MAXE; e++)
-                            atomicAdd(&h[(3 + 2 * e) * NB + bin],
-                                      vy[e]);
+                    ak += kmag * wgt;
+                    amu += mu * wgt;
+                    aw += wgt;
+
+                    double Pm1 = 0.0, P = 1.0;
+                    int e = 0;
+                    for (int l = 0; e < A.nell && e < MAXE; l++) {
+                        if (l > 0) {
+                            const double Pn = ((2 * l - 1) * mu * P
+                                               - (l - 1) * Pm1) / l;
+                            Pm1 = P;
+                            P = Pn;
+                        }
+                        if (l == A.ells[e]) {
+                            double wr = vre * P;
+                            if (nonsingular) {
+                                // conjugate-pair parity: odd ell keeps
+                                // only the (zero) imaginary part
+                                wr = (l % 2) ? 0.0 : 2.0 * wr;
+                            }
+                            ay[e] += wr * (2.0 * l + 1.0);
+                            e++;
+                        }
                     }
                 }
+                // no per-tile flush: cbin and its partial sums carry
+                // into the next (adjacent) tile's columns
             }
         }
         nbk_sync_lds();         // bin done before the next tile's store
+    }
+    if (cbin >= 0) {            // final flush of the cross-tile run
+        atomicAdd(&h[cbin], ak);
+        atomicAdd(&h[NB + cbin], amu);
+        atomicAdd(&h[2 * NB + cbin], aw);
+        for (int e = 0; e < A.nell && e < MAXE; e++)
+            atomicAdd(&h[(3 + 2 * e) * NB + cbin], ay[e]);
     }
 #undef XBIN_SETUP
 #undef XBIN_LOAD
