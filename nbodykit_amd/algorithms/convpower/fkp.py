@@ -130,7 +130,8 @@ class ConvolvedFFTPower(object):
             N = [int(n) for n in pm.Nmesh]
             k0 = 2 * numpy.pi / pm.BoxSize
             fz = numpy.arange(N[2] // 2 + 1, dtype='f8')
-            fz[-1] = -(N[2] // 2)
+            if N[2] % 2 == 0:
+                fz[-1] = -(N[2] // 2)
             k = [(_int_freqs(N[0]) * k0[0]).reshape(-1, 1, 1),
                  (_int_freqs(N[1]) * k0[1]).reshape(1, -1, 1),
                  (fz * k0[2]).reshape(1, 1, -1)]

@@ -63,14 +63,19 @@ __device__ __forceinline__ int64_t wrap_idx(int64_t i, int64_t n) {
 // n = 24/48/96/100/640/1536/2048; plain division does not).  For
 // power-of-two n both forms are the exact integer.
 __device__ __forceinline__ double freq_full(int64_t g, int64_t n) {
-    const int64_t m = (g < n / 2) ? g : g - n;
+    // positive frequencies run to (n-1)/2 inclusive — for even n that
+    // is n/2 - 1 with the Nyquist stored negative, for odd n there is
+    // no Nyquist plane (numpy fftfreq at any parity)
+    const int64_t m = (g < n - n / 2) ? g : g - n;
     return ((double)m * (1.0 / (double)n)) * (double)n;
 }
-// compressed (last) axis: indices 0..n/2, Nyquist stored negative.
-// These come from ``arange`` (+ an integer Nyquist assignment) on the
-// Python side, so they ARE exact integers at any n.
+// compressed (last) axis: indices 0..n/2, Nyquist stored negative for
+// EVEN n (odd n has no self-conjugate Nyquist plane and every g > 0 is
+// a positive frequency).  These come from ``arange`` (+ an integer
+// Nyquist assignment) on the Python side, so they ARE exact integers
+// at any n.
 __device__ __forceinline__ double freq_half(int64_t g, int64_t n) {
-    return (double)(g == n / 2 ? -(n / 2) : g);
+    return (double)((n % 2 == 0 && g == n / 2) ? -(n / 2) : g);
 }
 
 // Block barrier that waits only on LDS traffic (lgkmcnt), NOT on

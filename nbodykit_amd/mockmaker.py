@@ -72,7 +72,8 @@ def gaussian_real_fields(nmesh, boxsize, linear_power, seed,
     fx = numpy.fft.fftfreq(N[0]) * N[0]
     fy = numpy.fft.fftfreq(N[1]) * N[1]
     fz = numpy.arange(N[2] // 2 + 1, dtype='f8')
-    fz[-1] = -(N[2] // 2)
+    if N[2] % 2 == 0:
+        fz[-1] = -(N[2] // 2)
     kx = (fx * k0[0]).reshape(-1, 1, 1)
     ky = (fy * k0[1]).reshape(1, -1, 1)
     kz = (fz * k0[2]).reshape(1, 1, -1)

@@ -145,7 +145,8 @@ def fft_c2r_z(cplx2d, real2d, nz, s):
     nzh = nz // 2 + 1
     h = cplx2d.clone()
     h.imag[:, 0] = 0.0
-    h.imag[:, nz // 2] = 0.0
+    if nz % 2 == 0:          # odd nz has no self-conjugate Nyquist bin
+        h.imag[:, nz // 2] = 0.0
     full = torch.empty((L, nz), dtype=torch.complex128, device='cuda')
     full[:, :nzh] = h
     full[:, nzh:] = torch.conj(torch.flip(h[:, 1:nz - nzh + 1], (1,)))
@@ -251,13 +252,6 @@ class ParticleMesh(object):
         self.dtype = numpy.dtype(dtype)
 
         ws = self.comm.size
-        for ax in range(3):
-            if int(self.Nmesh[ax]) % 2:
-                raise ValueError(
-                    "Nmesh must be even on every axis (the compressed "
-                    "half-spectrum's Nyquist-as-negative and Hermitian "
-                    "conventions assume even dims); got Nmesh[%d]=%d"
-                    % (ax, self.Nmesh[ax]))
         for ax in (0, 1):
             if ws > 1 and self.Nmesh[ax] % ws != 0:
                 raise ValueError(
@@ -617,7 +611,8 @@ class ComplexField(_FieldBase):
         fx = _int_freqs(N[0])
         fy = _int_freqs(N[1])
         fz = numpy.arange(N[2] // 2 + 1, dtype='f8')
-        fz[-1] = -(N[2] // 2)
+        if N[2] % 2 == 0:
+            fz[-1] = -(N[2] // 2)     # Nyquist negative (even only)
         f = [fx, fy, fz]
         out = []
         for i in range(3):

@@ -660,7 +660,8 @@ class CatalogMesh(MeshSource):
             # the field is y-partitioned, and c(-k) needs rows owned by
             # other ranks: allgather the two tiny planes (16 B * Nx * Ny
             # each), project, keep the local y-block.
-            for kz in (0, n2 // 2):
+            self_conj = (0, n2 // 2) if n2 % 2 == 0 else (0,)
+            for kz in self_conj:
                 if ws > 1:
                     from nbodykit_amd.pm import all_gather_tensor
                     A = cplx[:, :, kz].contiguous()

@@ -55,7 +55,8 @@ def complex_coords(geom):
     fx = _int_freqs(Nx)
     fy = _int_freqs(Ny)
     fz = numpy.arange(Nz // 2 + 1, dtype='f8')
-    fz[-1] = -(Nz // 2)                      # Nyquist as negative
+    if Nz % 2 == 0:
+        fz[-1] = -(Nz // 2)                  # Nyquist as negative (even)
     k0 = 2 * numpy.pi / geom.BoxSize
     return [
         (fx * k0[0]).reshape(Nx, 1, 1),

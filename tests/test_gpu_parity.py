@@ -965,11 +965,71 @@ def test_nonpow2_roundtrip_and_rfftn():
     assert_allclose(back.value.cpu().numpy(), arr, rtol=0, atol=1e-11)
 
 
-def test_odd_nmesh_rejected():
-    from nbodykit_amd.pm import ParticleMesh
-    import pytest as _pytest
-    with _pytest.raises(ValueError, match='even'):
-        ParticleMesh(BoxSize=100., Nmesh=97)
+@pytest.mark.parametrize('nmesh', [27, 45])
+def test_odd_mesh_parity(nmesh):
+    """ODD Nmesh (no Nyquist plane on any axis): full FFTPower vs the
+    oracle.  FFTW-backed pmesh accepts any Nmesh; odd lengths run the
+    Bluestein fallback with the parity-dependent conventions switched
+    (no Nyquist-as-negative, no self-conjugate z plane beyond DC)."""
+    cat = UniformCatalog(nbar=2e-3, BoxSize=64., seed=5)
+    r = FFTPower(cat, mode='1d', Nmesh=nmesh)
+    pos = uniform_positions(2e-3, 64., 5)
+    want = fftpower_oracle(pos, Nmesh=nmesh, BoxSize=64., mode='1d',
+                           resampler='cic', compensated=True)
+    check_parity(r, want)
+
+
+def test_odd_mesh_interlaced_poles():
+    """Odd Nmesh with TSC + interlacing (the k-space combine's
+    self-conjugate projection reduces to the DC plane only) and
+    multipoles, 2d."""
+    n = 30000
+    pos = numpy.random.RandomState(7).uniform(0, 120., size=(n, 3))
+    cat = ArrayCatalog({'Position': pos})
+    mesh = cat.to_mesh(Nmesh=45, BoxSize=120., dtype='f8',
+                       compensated=True, resampler='tsc',
+                       interlaced=True)
+    r = FFTPower(mesh, mode='2d', Nmu=4, poles=[0, 2])
+    want = fftpower_oracle(pos, Nmesh=45, BoxSize=120., mode='2d',
+                           Nmu=4, poles=[0, 2], resampler='tsc',
+                           compensated=True, interlaced=True)
+    check_parity(r, want, poles=[0, 2])
+
+
+def test_odd_roundtrip_and_rfftn():
+    """27^3 and 45^3 r2c against numpy rfftn and the c2r round trip."""
+    import torch
+    from nbodykit_amd.pm import ParticleMesh, RealField
+    for nm in (27, 45):
+        pm = ParticleMesh(BoxSize=100., Nmesh=nm)
+        rng = numpy.random.RandomState(nm)
+        arr = rng.normal(size=(nm, nm, nm))
+        f = RealField(pm, tensor=torch.as_tensor(arr).to('cuda'))
+        c = f.r2c()
+        want = numpy.fft.rfftn(arr) / float(nm) ** 3
+        assert_allclose(c.value.cpu().numpy(), want, rtol=0, atol=1e-13)
+        back = c.c2r()
+        assert_allclose(back.value.cpu().numpy(), arr, rtol=0,
+                        atol=1e-11)
+
+
+def test_odd_mesh_fftcorr():
+    """Odd-mesh FFTCorr (configuration-space binning + c2r on odd
+    lengths) vs the oracle."""
+    from nbodykit_amd.lab import FFTCorr
+    from oracle import fftcorr_oracle
+    n = 20000
+    pos = numpy.random.RandomState(9).uniform(0, 100., size=(n, 3))
+    cat = ArrayCatalog({'Position': pos})
+    r = FFTCorr(cat, mode='1d', Nmesh=33, BoxSize=100.)
+    want = fftcorr_oracle(pos, Nmesh=33, BoxSize=100., mode='1d',
+                          resampler='cic', compensated=True)
+    assert_array_equal(r.corr['modes'], want['modes'])
+    got = r.corr['corr']
+    ref = want['corr']
+    ok = numpy.isfinite(ref) & (numpy.abs(ref) > 1e-12)
+    rel = numpy.abs(got[ok] - ref[ok]) / numpy.abs(ref[ok])
+    assert rel.max() < 1e-9, 'odd FFTCorr parity: %g' % rel.max()
 
 
 @pytest.mark.timeout(600)
