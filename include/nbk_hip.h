@@ -332,7 +332,16 @@ int nbk_power_bin_f64(const double* c1, const double* c2, double volume,
  * two in [8, 4096]; fails with NBK_ERR_UNSUPPORTED when the histogram +
  * FFT tile exceed the 160 KiB LDS (callers fall back to the unfused
  * sequence). */
-int nbk_fft_x_bin_f64(const double* data, const int64_t nmesh[3],
+/* data2: NULL for a plain auto power; for an INTERLACED mesh the
+ * half-cell-shifted paint's pre-x-pass field — both tiles are FFT'd
+ * and combined as c = a/2 + b/2 exp(i k.H/2) before compensation and
+ * binning.  In that mode the self-conjugate z planes (iz = 0 and, for
+ * even n2, the Nyquist plane) are SKIPPED — their Hermitian projection
+ * couples columns, so the caller handles those two planes with the
+ * standalone kernels and the same out_sums buffer (all the binning
+ * entry points accumulate). */
+int nbk_fft_x_bin_f64(const double* data, const double* data2,
+                      const int64_t nmesh[3],
                       int64_t n_inner, int64_t y_off,
                       const double box[3],
                       int window1, int interlaced1,

@@ -145,16 +145,20 @@ class FFTBase(object):
         from nbodykit_amd.pm import r2c_defer_x
         res = mesh.to_complex_field(_defer_x=True)
         if res is NotImplemented:
+            # chunked-paint fallback: for an interlaced mesh
+            # to_real_field returns the COMBINED mesh (the reference's
+            # c2r round trip), so a single pre-x field with the
+            # interlaced compensation is exactly right
             real = mesh.to_real_field(normalize=True)
             tensor, n_inner = r2c_defer_x(real)
-            res = (tensor, n_inner, dict(real.attrs))
-        tensor, n_inner, mattrs = res
+            res = (tensor, None, n_inner, dict(real.attrs))
+        tensor, tensor2, n_inner, mattrs = res
         attrs = {}
         attrs.update(self.attrs)
         attrs.update({'N1': mattrs.get('N', 0), 'N2': mattrs.get('N', 0)})
         attrs['shotnoise'] = mattrs.get('shotnoise', 0) \
             if self.first is self.second else 0
-        return tensor, n_inner, attrs
+        return tensor, tensor2, n_inner, attrs
 
 
 class FFTPower(FFTBase):
@@ -226,10 +230,12 @@ class FFTPower(FFTBase):
             poles_l = list(self.attrs['poles'])
             nell = len(poles_l) + (0 not in poles_l)
             defer = len(kedges_d) >= 1 and _xbin_lds_fits(
-                n0, len(kedges_d), self.attrs['Nmu'] + 1, nell)
+                n0, len(kedges_d), self.attrs['Nmu'] + 1, nell,
+                il=bool(getattr(self.first, 'interlaced', False)))
 
         if defer:
-            tensor, n_inner, attrs = self._compute_deferred_x(self.first)
+            tensor, tensor2, n_inner, attrs = \
+                self._compute_deferred_x(self.first)
             pm = self.first.pm
             Nmesh_arr = numpy.asarray(pm.Nmesh)
             BoxSize_arr = numpy.asarray(pm.BoxSize)
@@ -260,7 +266,7 @@ class FFTPower(FFTBase):
         coords = [kcoords, None]
         if defer:
             result, pole_result = _project_power_xbin(
-                tensor, pm, n_inner, fuse1,
+                tensor, tensor2, pm, n_inner, fuse1,
                 volume=self.attrs['BoxSize'].prod(), edges=edges,
                 poles=self.attrs['poles'], los=self.attrs['los'])
         elif fused:
@@ -622,8 +628,6 @@ def _xbin_ok(mesh, Nmesh):
     import os
     if os.environ.get('NBK_NO_XBIN', '0') == '1':
         return False
-    if getattr(mesh, 'interlaced', False):
-        return False
     pmN = numpy.asarray(mesh.pm.Nmesh)
     if not numpy.array_equal(pmN, numpy.asarray(Nmesh)):
         return False
@@ -634,28 +638,41 @@ def _xbin_ok(mesh, Nmesh):
     return True
 
 
-def _xbin_lds_fits(n0, nx_edges, nmu_edges, nell):
+def _xbin_lds_fits(n0, nx_edges, nmu_edges, nell, il=False):
     """Mirror of nbk_fft_x_bin_f64's LDS budget: histograms + edge
-    arrays + kx/compensation tables + the TI=1 FFT tile must fit the
-    gfx950 160 KiB LDS (the kernel itself would return
-    NBK_ERR_UNSUPPORTED; gate here so the standard path is taken
-    without a failed launch)."""
+    arrays + compensation (+ interlace phase) tables + the TI=1 FFT
+    tile(s) must fit the gfx950 160 KiB LDS, and the register pipeline
+    caps n0*TI (the kernel itself would return NBK_ERR_UNSUPPORTED;
+    gate here so the standard path is taken without a failed
+    launch)."""
     NB = (nx_edges + 1) * (nmu_edges + 1)
     fixed = (NB * (3 + 2 * nell) + nx_edges + nmu_edges + n0) * 8
-    return fixed + n0 * 2 * 16 <= 160 * 1024   # TI >= 1, pitch TI+1
+    if il:
+        fixed += n0 * 16                       # x phase table
+        if n0 > 2048:
+            return False
+    nbufs = 2 if il else 1
+    return fixed + nbufs * n0 * 2 * 16 <= 160 * 1024  # TI=1, pitch 2
 
 
-def _project_power_xbin(tensor, pm, n_inner, comp1, volume, edges,
-                        los=[0, 0, 1], poles=[]):
+def _project_power_xbin(tensor, tensor2, pm, n_inner, comp1, volume,
+                        edges, los=[0, 0, 1], poles=[]):
     """project_to_basis of |comp1(F_x(pre_x))|^2 V with the zero mode
     cleared, fused into the final x-axis FFT pass (nbk_fft_x_bin_f64):
     the finished complex field never exists in HBM and columns wholly
     beyond the last k-edge are skipped before their loads.  Auto power
     only.  Bin assignment is bit-identical to the unfused path (same
     k2/mu groupings); the x-FFT element values are bit-identical to
-    nbk_fft_c_strided's."""
+    nbk_fft_c_strided's.
+
+    ``tensor2`` (interlaced meshes): the half-cell-shifted paint's
+    pre-x field — the kernel combines the pair per element; the two
+    self-conjugate z planes (whose Hermitian projection couples
+    columns) are skipped there and handled here with the standalone
+    kernels, accumulating into the same sums buffer."""
     import torch
     comm = pm.comm
+    ws = comm.size
     lib = hiplib.require()
 
     xedges, muedges = edges
@@ -681,9 +698,11 @@ def _project_power_xbin(tensor, pm, n_inner, comp1, volume, edges,
     sums = torch.zeros(nfields * NB, dtype=torch.float64, device=dev)
 
     win1, interl1 = comp1
-    y_off = pm.y_start if comm.size > 1 else 0
+    y_off = pm.y_start if ws > 1 else 0
     hiplib.check(lib.nbk_fft_x_bin_f64(
-        hiplib.dptr(tensor), hiplib.i64_arr(pm.Nmesh),
+        hiplib.dptr(tensor),
+        hiplib.dptr(tensor2) if tensor2 is not None else None,
+        hiplib.i64_arr(pm.Nmesh),
         int(n_inner), int(y_off), hiplib.f64_arr(pm.BoxSize),
         int(win1), int(interl1), 1, float(volume),
         hiplib.dptr(k2edges_t), len(xedges),
@@ -691,11 +710,77 @@ def _project_power_xbin(tensor, pm, n_inner, comp1, volume, edges,
         hiplib.f64_arr(los), hiplib.int_arr(_poles), Nell,
         hiplib.dptr(sums), hiplib.cur_stream()), 'nbk_fft_x_bin_f64')
 
+    if tensor2 is not None:
+        _xbin_conj_planes(tensor, tensor2, pm, n_inner, y_off, comp1,
+                          volume, k2edges_t, xedges, muedges_t, muedges,
+                          los, _poles, Nell, NB, sums)
+
     torch.cuda.synchronize()
     host = sums.cpu().numpy()
     host = comm.allreduce(host)
     return _fold_bin_sums(host, Nx, Nmu, Nell, ell_idx, do_poles,
                           real_field=False)
+
+
+def _xbin_conj_planes(tensor, tensor2, pm, n_inner, y_off, comp1,
+                      volume, k2edges_t, xedges, muedges_t, muedges,
+                      los, _poles, Nell, NB, sums):
+    """The interlaced pair's self-conjugate z planes (kz = 0 and, for
+    even n2, the Nyquist plane): x-FFT both paints' plane slices,
+    combine with exp(i k.H/2), apply the Hermitian projection
+    c <- (c + conj(c(-k)))/2 over the (kx, ky) mirror — allgathered
+    when y is partitioned — and bin with the standalone fused kernel
+    (source/mesh/catalog.py applies the same projection on the
+    materialized field; the reference gets it from the c2r + re-r2c
+    round trip)."""
+    import torch
+    from nbodykit_amd.pm import fft_axis1, all_gather_tensor, _int_freqs
+    lib = hiplib.require()
+    comm = pm.comm
+    ws = comm.size
+    s = hiplib.cur_stream()
+    n0, n1, n2 = (int(x) for x in pm.Nmesh)
+    nzh = n2 // 2 + 1
+    nyl = int(n_inner) // nzh
+    win1, interl1 = comp1
+
+    planes = [0] + ([n2 // 2] if n2 % 2 == 0 else [])
+    fx = _int_freqs(n0)
+    fy = _int_freqs(n1)[y_off:y_off + nyl]
+    for iz in planes:
+        fz = float(iz) if iz == 0 else -float(n2 // 2)
+        A = tensor.view(n0, -1)[:, iz::nzh].contiguous()
+        B = tensor2.view(n0, -1)[:, iz::nzh].contiguous()
+        fft_axis1(A.view(1, n0, nyl), -1, s)
+        fft_axis1(B.view(1, n0, nyl), -1, s)
+        kH = 2 * numpy.pi * (fx[:, None] / n0 + fy[None, :] / n1
+                             + fz / n2)
+        ph = torch.as_tensor(numpy.exp(0.5j * kH)).to('cuda')
+        C = 0.5 * A + 0.5 * B * ph
+        if ws > 1:
+            parts = [torch.empty_like(C) for _ in range(ws)]
+            all_gather_tensor(parts, C.contiguous())
+            Cf = torch.cat(parts, dim=1)
+            M = torch.conj(torch.roll(torch.flip(Cf, (0, 1)),
+                                      (1, 1), (0, 1)))
+            C = (0.5 * (Cf + M))[:, y_off:y_off + nyl]
+        else:
+            M = torch.conj(torch.roll(torch.flip(C, (0, 1)),
+                                      (1, 1), (0, 1)))
+            C = 0.5 * (C + M)
+        C = C.reshape(n0, nyl, 1).contiguous()
+        hiplib.check(lib.nbk_power_bin_f64(
+            hiplib.dptr(C), None, float(volume),
+            int(win1), int(interl1), int(win1), int(interl1), 1,
+            hiplib.i64_arr(pm.Nmesh), hiplib.f64_arr(pm.BoxSize),
+            hiplib.i64_arr([n0, nyl, 1]),
+            hiplib.i64_arr([0, y_off, iz]), None,
+            hiplib.dptr(k2edges_t), len(xedges),
+            hiplib.dptr(muedges_t), len(muedges),
+            hiplib.f64_arr(los), hiplib.int_arr(_poles), Nell,
+            hiplib.dptr(sums), hiplib.dptr(sums[NB:]),
+            hiplib.dptr(sums[2 * NB:]), hiplib.dptr(sums[3 * NB:]),
+            hiplib.cur_stream()), 'nbk_power_bin_f64')
 
 
 def _cast_source(source, BoxSize, Nmesh):

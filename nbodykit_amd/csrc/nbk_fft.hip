@@ -351,8 +351,17 @@ __device__ __forceinline__ int digx(const double* __restrict__ edges,
 // Im(a conj(a)) = fl(-re*im) + fl(im*re) = 0 under -ffp-contract=off.
 // The imaginary multipole sums are therefore never accumulated (their
 // output planes stay zero, matching the unfused kernel's zeros).
-template <int PHASES, int MAXE>
-__global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
+// IL: interlaced-pair mode — a SECOND pre-x field (the half-cell
+// shifted paint) rides along; both tiles are FFT'd and combined per
+// element as c = a/2 + b/2 exp(i k.H/2) (nbk_interlace_combine's
+// formula, phases composed from an LDS x-table times a per-column
+// factor) before compensation and binning.  The self-conjugate z
+// planes (iz = 0 and, for even n2, the Nyquist plane) are SKIPPED —
+// their Hermitian projection couples (-kx,-ky) across columns, so the
+// host handles those two planes with the standalone kernels.
+template <int PHASES, int MAXE, bool IL>
+__global__ void kxfft_bin(const double* __restrict__ data,
+                          const double* __restrict__ data2, XBinArgs A,
                           const double* __restrict__ k2edges_g,
                           const double* __restrict__ muedges_g,
                           const cdouble* __restrict__ table /* W_2n0 */,
@@ -375,12 +384,16 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
     extern __shared__ cdouble smem[];                  // (shared decl
                                                        // with kfft_c2r_z)
     cdouble* buf = smem;                               // n0 * W
-    double* h = (double*)smem + 2 * (size_t)n0 * W;    // NB * nfields
+    cdouble* buf2 = IL ? smem + (size_t)n0 * W : nullptr;
+    cdouble* pxv = IL ? smem + 2 * (size_t)n0 * W : nullptr;  // n0
+    double* h = (double*)smem
+        + 2 * (size_t)n0 * W * (IL ? 2 : 1)
+        + (IL ? 2 * (size_t)n0 : 0);                   // NB * nfields
     double* ke = h + (size_t)NB * nfields;             // nx_edges
     double* me = ke + A.nx_edges;                      // nmu_edges
     double* cxv = me + A.nmu_edges;                    // n0
     __shared__ double cky2[2][16], ckz2[2][16], ckyl[2][16],
-                      ckzl[2][16], ccy[2][16], ccz[2][16];
+                      ckzl[2][16], ccy[2][16], ccz[2][16], cph[2][32];
     __shared__ unsigned char cw2[2][16], czl[2][16], cskip[2][16];
     __shared__ int s_nlive[2];
 
@@ -394,6 +407,11 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
             ? nbk_comp_factor1(A.win1, A.interl1,
                                2.0 * M_PI * fx / (double)A.n0)
             : 1.0;
+        if (IL) {
+            double sp, cp;
+            sincos(M_PI * fx / (double)A.n0, &sp, &cp);
+            pxv[j] = {cp, sp};      // exp(i kx Hx / 2)
+        }
     }
     // the skip threshold, read straight from global (one scalar; the
     // LDS copy is not published until the first sync)
@@ -426,6 +444,13 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
                 ckzl[q_][t] = kz_ * A.losz;                                \
                 cw2[q_][t] = (fz_ > 0.0);                                  \
                 czl[q_][t] = (fy_ == 0.0 && fz_ == 0.0);                   \
+                if (IL) {                                                  \
+                    double sp_, cp_;                                       \
+                    sincos(M_PI * (fy_ / (double)A.n1                      \
+                                   + fz_ / (double)A.n2), &sp_, &cp_);     \
+                    cph[q_][2 * t] = cp_;                                  \
+                    cph[q_][2 * t + 1] = sp_;                              \
+                }                                                          \
                 if (comp_on) {                                             \
                     ccy[q_][t] = nbk_comp_factor1(A.win1, A.interl1,       \
                         2.0 * M_PI * fy_ / (double)A.n1);                  \
@@ -438,6 +463,10 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
                 /* the line's k2 minimum is at kx = 0 exactly, where   */  \
                 /* fl(0 + ky2) = ky2; fl-addition is monotone           */  \
                 live_ = (cky2[q_][t] + ckz2[q_][t] < k2last) ? 1 : 0;      \
+                /* IL: the self-conjugate planes go to the host path */    \
+                if (IL && (iz_ == 0                                        \
+                           || (A.n2 % 2 == 0 && iz_ == A.nzh - 1)))        \
+                    live_ = 0;                                             \
             }                                                              \
             cskip[q_][t] = (unsigned char)(!live_);                        \
         }                                                                  \
@@ -448,24 +477,32 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
 #define XBIN_LOAD(tile_, q_)                                               \
     {                                                                      \
         const cdouble* g_ = (const cdouble*)data + (int64_t)(tile_) * TI;  \
+        const cdouble* g2_ = IL                                            \
+            ? (const cdouble*)data2 + (int64_t)(tile_) * TI : nullptr;     \
         _Pragma("unroll")                                                  \
-        for (int q = 0; q < 4; q++) {                                      \
+        for (int q = 0; q < NLD; q++) {                                    \
             const int w_ = q * T + t;                                      \
             if (w_ < n0 * TI) {                                            \
                 const int c_ = w_ % TI;                                    \
-                r[q] = cskip[q_][c_]                                       \
-                    ? cdouble{0.0, 0.0}                                    \
-                    : g_[(int64_t)(w_ / TI) * A.n_inner + c_];             \
+                const bool dead_ = cskip[q_][c_];                          \
+                const int64_t at_ = (int64_t)(w_ / TI) * A.n_inner + c_;   \
+                r[q] = dead_ ? cdouble{0.0, 0.0} : g_[at_];                \
+                if (IL)                                                    \
+                    r2[q] = dead_ ? cdouble{0.0, 0.0} : g2_[at_];          \
             }                                                              \
         }                                                                  \
     }
 
 #define XBIN_STORE                                                         \
     _Pragma("unroll")                                                      \
-    for (int q = 0; q < 4; q++) {                                          \
+    for (int q = 0; q < NLD; q++) {                                        \
         const int w_ = q * T + t;                                          \
-        if (w_ < n0 * TI)                                                  \
-            buf[bitrev(w_ / TI, bits) * W + (w_ % TI)] = r[q];             \
+        if (w_ < n0 * TI) {                                                \
+            const int at_ = bitrev(w_ / TI, bits) * W + (w_ % TI);         \
+            buf[at_] = r[q];                                               \
+            if (IL)                                                        \
+                buf2[at_] = r2[q];                                         \
+        }                                                                  \
     }
 
     // CONTIGUOUS tile chunks per block (not grid-strided): adjacent
@@ -478,7 +515,11 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
     const int64_t chunk = (A.tiles + gridDim.x - 1) / gridDim.x;
     const int64_t tend = ((int64_t)(blockIdx.x + 1) * chunk < A.tiles)
         ? (int64_t)(blockIdx.x + 1) * chunk : A.tiles;
-    cdouble r[4] = {};
+    // IL carries two tiles in registers — the launcher halves the
+    // n0*TI cap so the held-register count stays constant
+    constexpr int NLD = IL ? 2 : 4;
+    cdouble r[NLD] = {};
+    cdouble r2[IL ? NLD : 1] = {};
     int64_t tile = (int64_t)blockIdx.x * chunk;
     int p = 0;
 
@@ -507,8 +548,11 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
             XBIN_LOAD(nxt, p ^ 1)   // in flight under FFT + bin below
 
         if (s_nlive[p] > 0) {
-            if (PHASES & 1)
+            if (PHASES & 1) {
                 lds_fft4<false>(buf, n0, TI, W, table);
+                if (IL)
+                    lds_fft4<false>(buf2, n0, TI, W, table);
+            }
 
             // run-merged bin phase: walk j = my_j0 .. +RB-1 of column
             // my_c, merging same-bin neighbours in registers and
@@ -523,6 +567,9 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
                 const double kzl = ckzl[p][c];
                 const double ccyv = ccy[p][c];
                 const double cczv = ccz[p][c];
+                const cdouble phc = IL
+                    ? cdouble{cph[p][2 * c], cph[p][2 * c + 1]}
+                    : cdouble{1.0, 0.0};
                 const bool nonsingular = cw2[p][c];
                 const double wgt = nonsingular ? 2.0 : 1.0;
                 const bool col_zero = czl[p][c];
@@ -554,6 +601,15 @@ __global__ void kxfft_bin(const double* __restrict__ data, XBinArgs A,
                     const int bin = bx * (A.nmu_edges + 1) + bmu;
 
                     cdouble a = buf[j * W + c];
+                    if (IL) {
+                        // c = a/2 + b/2 exp(i k.H/2) — the
+                        // nbk_interlace_combine formula with the phase
+                        // composed as px[j] * pyz[c]
+                        const cdouble b = buf2[j * W + c];
+                        const cdouble ph = cmul(pxv[j], phc);
+                        a = cadd(cscale(a, 0.5),
+                                 cscale(cmul(b, ph), 0.5));
+                    }
                     if (comp_on)
                         a = cscale(a, (cxv[j] * ccyv) * cczv);
                     // auto power: Im(a conj(a)) == 0 exactly
@@ -731,7 +787,7 @@ extern "C" int nbk_fft_c_strided(double* cplx, int64_t nfft, int64_t stride,
     return NBK_OK;
 }
 
-extern "C" int nbk_fft_x_bin_f64(const double* data,
+extern "C" int nbk_fft_x_bin_f64(const double* data, const double* data2,
                                  const int64_t nmesh[3],
                                  int64_t n_inner, int64_t y_off,
                                  const double box[3],
@@ -743,6 +799,7 @@ extern "C" int nbk_fft_x_bin_f64(const double* data,
                                  const int* ells, int nell,
                                  double* out_sums, void* stream)
 {
+    const bool il = data2 != nullptr;
     int rc = check_len(nmesh[0], "nbk_fft_x_bin_f64");
     if (rc) return rc;
     if (n_inner == 0) return NBK_OK;
@@ -788,20 +845,26 @@ extern "C" int nbk_fft_x_bin_f64(const double* data,
         if (TI0 < 1 || TI0 > 16) TI0 = 4;
     }
     int TI = TI0;
-    // the pipeline carries a tile in 4 registers per thread
-    while (TI > 1 && (int64_t)nmesh[0] * TI > 4 * 1024)
+    // the pipeline carries a tile in 4 registers per thread (2 + 2
+    // when an interlaced pair rides along)
+    const int64_t regcap = il ? 2 * 1024 : 4 * 1024;
+    while (TI > 1 && (int64_t)nmesh[0] * TI > regcap)
         TI >>= 1;
-    // buf pitch is TI+1 (bank-spread padding for the bin phase's runs)
-    while (TI > 1 && fixed + (size_t)nmesh[0] * (TI + 1) * sizeof(cdouble)
-                     > 160 * 1024)
+    // buf pitch is TI+1 (bank-spread padding for the bin phase's
+    // runs); interlaced mode holds two tiles + the x phase table
+    const size_t fixed_il = fixed
+        + (il ? (size_t)nmesh[0] * sizeof(cdouble) : 0);
+    const int nbufs = il ? 2 : 1;
+    while (TI > 1 && fixed_il + (size_t)nbufs * nmesh[0] * (TI + 1)
+                         * sizeof(cdouble) > 160 * 1024)
         TI >>= 1;
-    if ((int64_t)nmesh[0] * TI > 4 * 1024) {
+    if ((int64_t)nmesh[0] * TI > regcap) {
         NBK_SET_ERR("nbk_fft_x_bin_f64: n0 too large for the register "
                     "pipeline — use the unfused path");
         return NBK_ERR_UNSUPPORTED;
     }
-    const size_t shmem = fixed
-        + (size_t)nmesh[0] * (TI + 1) * sizeof(cdouble);
+    const size_t shmem = fixed_il
+        + (size_t)nbufs * nmesh[0] * (TI + 1) * sizeof(cdouble);
     if (shmem > 160 * 1024) {
         NBK_SET_ERR("nbk_fft_x_bin_f64: LDS budget exceeded "
                     "(%zu B) — use the unfused path", shmem);
@@ -830,51 +893,75 @@ extern "C" int nbk_fft_x_bin_f64(const double* data,
         if (phases < 0 || phases > 3) phases = 3;
         phases |= 4;            // mark initialized
     }
-    // MAXE sizes the per-thread multipole registers: <.,1> serves every
-    // plain 1d/2d run, <.,8> any poles request; the decomposition
-    // phases only exist at MAXE = 8
-    const void* fn;
-    if ((phases & 3) == 3)
-        fn = nell <= 1
-            ? reinterpret_cast<const void*>(&kxfft_bin<3, 1>)
-            : reinterpret_cast<const void*>(&kxfft_bin<3, 8>);
-    else if ((phases & 3) == 1)
-        fn = reinterpret_cast<const void*>(&kxfft_bin<1, 8>);
-    else if ((phases & 3) == 2)
-        fn = reinterpret_cast<const void*>(&kxfft_bin<2, 8>);
+    // MAXE sizes the per-thread multipole registers: <.,1,.> serves
+    // every plain 1d/2d run, <.,8,.> any poles request; the perf
+    // decomposition phases exist only at MAXE = 8, non-interlaced
+    const int ph = phases & 3;
+    int variant;               // 0..3: <3,1,F> <3,8,F> <3,1,T> <3,8,T>
+    if (ph != 3)
+        variant = 4 + ph;      // 4..6: phases 0..2 at <ph,8,F>
     else
-        fn = reinterpret_cast<const void*>(&kxfft_bin<0, 8>);
-    static size_t raised = 0;
-    if (shmem > 64 * 1024 && shmem > raised) {
+        variant = (il ? 2 : 0) + (nell <= 1 ? 0 : 1);
+    const void* fns[7] = {
+        reinterpret_cast<const void*>(&kxfft_bin<3, 1, false>),
+        reinterpret_cast<const void*>(&kxfft_bin<3, 8, false>),
+        reinterpret_cast<const void*>(&kxfft_bin<3, 1, true>),
+        reinterpret_cast<const void*>(&kxfft_bin<3, 8, true>),
+        reinterpret_cast<const void*>(&kxfft_bin<0, 8, false>),
+        reinterpret_cast<const void*>(&kxfft_bin<1, 8, false>),
+        reinterpret_cast<const void*>(&kxfft_bin<2, 8, false>),
+    };
+    static size_t raised[7] = {};
+    if (shmem > 64 * 1024 && shmem > raised[variant]) {
         (void)hipFuncSetAttribute(
-            fn, hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);
-        raised = shmem;
+            fns[variant], hipFuncAttributeMaxDynamicSharedMemorySize,
+            (int)shmem);
+        raised[variant] = shmem;
     }
-    if ((phases & 3) == 3 && nell <= 1)
-        hipLaunchKernelGGL((kxfft_bin<3, 1>), dim3((uint32_t)g),
+    switch (variant) {
+    case 0:
+        hipLaunchKernelGGL((kxfft_bin<3, 1, false>), dim3((uint32_t)g),
                            dim3(1024), shmem, (hipStream_t)stream, data,
-                           A, k2edges, muedges, (const cdouble*)table,
-                           out_sums);
-    else if ((phases & 3) == 3)
-        hipLaunchKernelGGL((kxfft_bin<3, 8>), dim3((uint32_t)g),
+                           data2, A, k2edges, muedges,
+                           (const cdouble*)table, out_sums);
+        break;
+    case 1:
+        hipLaunchKernelGGL((kxfft_bin<3, 8, false>), dim3((uint32_t)g),
                            dim3(1024), shmem, (hipStream_t)stream, data,
-                           A, k2edges, muedges, (const cdouble*)table,
-                           out_sums);
-    else if ((phases & 3) == 1)
-        hipLaunchKernelGGL((kxfft_bin<1, 8>), dim3((uint32_t)g),
+                           data2, A, k2edges, muedges,
+                           (const cdouble*)table, out_sums);
+        break;
+    case 2:
+        hipLaunchKernelGGL((kxfft_bin<3, 1, true>), dim3((uint32_t)g),
                            dim3(1024), shmem, (hipStream_t)stream, data,
-                           A, k2edges, muedges, (const cdouble*)table,
-                           out_sums);
-    else if ((phases & 3) == 2)
-        hipLaunchKernelGGL((kxfft_bin<2, 8>), dim3((uint32_t)g),
+                           data2, A, k2edges, muedges,
+                           (const cdouble*)table, out_sums);
+        break;
+    case 3:
+        hipLaunchKernelGGL((kxfft_bin<3, 8, true>), dim3((uint32_t)g),
                            dim3(1024), shmem, (hipStream_t)stream, data,
-                           A, k2edges, muedges, (const cdouble*)table,
-                           out_sums);
-    else
-        hipLaunchKernelGGL((kxfft_bin<0, 8>), dim3((uint32_t)g),
+                           data2, A, k2edges, muedges,
+                           (const cdouble*)table, out_sums);
+        break;
+    case 4:
+        hipLaunchKernelGGL((kxfft_bin<0, 8, false>), dim3((uint32_t)g),
                            dim3(1024), shmem, (hipStream_t)stream, data,
-                           A, k2edges, muedges, (const cdouble*)table,
-                           out_sums);
+                           data2, A, k2edges, muedges,
+                           (const cdouble*)table, out_sums);
+        break;
+    case 5:
+        hipLaunchKernelGGL((kxfft_bin<1, 8, false>), dim3((uint32_t)g),
+                           dim3(1024), shmem, (hipStream_t)stream, data,
+                           data2, A, k2edges, muedges,
+                           (const cdouble*)table, out_sums);
+        break;
+    default:
+        hipLaunchKernelGGL((kxfft_bin<2, 8, false>), dim3((uint32_t)g),
+                           dim3(1024), shmem, (hipStream_t)stream, data,
+                           data2, A, k2edges, muedges,
+                           (const cdouble*)table, out_sums);
+        break;
+    }
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

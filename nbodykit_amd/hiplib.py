@@ -133,8 +133,8 @@ def _declare(lib):
                                       c_void, c_void, c_void, c_void,
                                       c_void]
     lib.nbk_fft_x_bin_f64.restype = ctypes.c_int
-    lib.nbk_fft_x_bin_f64.argtypes = [c_void, c_i64_p, c_i64, c_i64,
-                                      c_f64_p,
+    lib.nbk_fft_x_bin_f64.argtypes = [c_void, c_void, c_i64_p, c_i64,
+                                      c_i64, c_f64_p,
                                       ctypes.c_int, ctypes.c_int,
                                       ctypes.c_int, c_f64,
                                       c_void, c_i64, c_void, c_i64,

@@ -632,14 +632,15 @@ class CatalogMesh(MeshSource):
             return _r2c_y_transpose(z, pm, stream)
 
         if _defer_x:
-            assert not interlaced, \
-                "_defer_x is only valid for non-interlaced meshes"
             tensor, n_inner = one(0.0, finish=False)
+            tensor2 = None
+            if interlaced:
+                tensor2, _ = one(0.5, finish=False)
             with numpy.errstate(divide='ignore', invalid='ignore'):
                 shotnoise = float(numpy.prod(pm.BoxSize)) * W2 / W ** 2
-            return tensor, n_inner, {'shotnoise': shotnoise, 'N': N,
-                                     'W': W, 'W2': W2,
-                                     'num_per_cell': nbar}
+            return tensor, tensor2, n_inner, {'shotnoise': shotnoise,
+                                              'N': N, 'W': W, 'W2': W2,
+                                              'num_per_cell': nbar}
 
         cplx = one(0.0)
         if interlaced:

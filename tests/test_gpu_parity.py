@@ -1243,10 +1243,6 @@ def test_xbin_gates(monkeypatch):
     edges, non-power-of-two meshes, and NBK_NO_XBIN=1."""
     calls = _trace_xbin(monkeypatch)
     cat = UniformCatalog(nbar=1e-2, BoxSize=64., seed=14)
-    mesh_i = cat.to_mesh(Nmesh=64, dtype='f8', compensated=True,
-                         interlaced=True)
-    FFTPower(mesh_i, mode='1d')
-    assert not calls, 'interlaced mesh must not defer'
     cat2 = UniformCatalog(nbar=1e-2, BoxSize=64., seed=15)
     m1 = cat.to_mesh(Nmesh=64, dtype='f8', compensated=True)
     m2 = cat2.to_mesh(Nmesh=64, dtype='f8', compensated=True)
@@ -1274,4 +1270,46 @@ def test_xbin_matches_oracle():
     pos = uniform_positions(nbar, box, seed)
     want = fftpower_oracle(pos, Nmesh=nmesh, BoxSize=box, mode='1d',
                            resampler='cic', compensated=True)
+    check_parity(r, want)
+
+
+@pytest.mark.parametrize('kwargs,mkw', [
+    (dict(mode='1d'), dict(resampler='cic')),
+    (dict(mode='2d', Nmu=4, poles=[0, 2]), dict(resampler='tsc')),
+    (dict(mode='1d'), dict(resampler='pcs', compensated=False)),
+], ids=['cic1d', 'tsc2dpoles', 'pcs-nocomp'])
+def test_xbin_interlaced_matches_unfused(kwargs, mkw, monkeypatch):
+    """Interlaced deferred-x: the kernel FFTs BOTH paints' tiles,
+    combines with exp(i k.H/2) and bins; the self-conjugate planes go
+    through the host path.  Against the standard interlaced pipeline
+    (combine kernel + projection + nbk_power_bin_f64)."""
+    cat = UniformCatalog(nbar=1e-2, BoxSize=128., seed=16)
+    m = dict(Nmesh=128, dtype='f8', compensated=True, interlaced=True)
+    m.update(mkw)
+    mesh = cat.to_mesh(**m)
+    calls = _trace_xbin(monkeypatch)
+    r_x = FFTPower(mesh, **kwargs)
+    assert calls, 'interlaced deferred-x did not engage'
+    monkeypatch.setenv('NBK_NO_XBIN', '1')
+    r_ref = FFTPower(mesh, **kwargs)
+    scale = numpy.nanmax(numpy.abs(r_ref.power['power']))
+    assert_array_equal(r_x.power['modes'], r_ref.power['modes'])
+    assert_allclose(r_x.power['power'], r_ref.power['power'],
+                    rtol=1e-10, atol=1e-11 * scale, equal_nan=True)
+    for ell in kwargs.get('poles', []):
+        assert_allclose(r_x.poles['power_%d' % ell],
+                        r_ref.poles['power_%d' % ell],
+                        rtol=1e-10, atol=1e-11 * scale, equal_nan=True)
+
+
+def test_xbin_interlaced_matches_oracle():
+    nbar, box, nmesh, seed = 1e-2, 96., 64, 17
+    cat = UniformCatalog(nbar=nbar, BoxSize=box, seed=seed)
+    mesh = cat.to_mesh(Nmesh=nmesh, dtype='f8', compensated=True,
+                       interlaced=True, resampler='tsc')
+    r = FFTPower(mesh, mode='1d')
+    pos = uniform_positions(nbar, box, seed)
+    want = fftpower_oracle(pos, Nmesh=nmesh, BoxSize=box, mode='1d',
+                           resampler='tsc', compensated=True,
+                           interlaced=True)
     check_parity(r, want)
