@@ -79,9 +79,13 @@ __device__ __forceinline__ int bitrev(int j, int bits) {
 // PITCH is the LDS storage row stride in cdoubles (>= TI; the fused
 // x+bin kernel pads to TI+1 so its per-thread j-runs read across LDS
 // banks instead of landing on two)
+// mstop: run stages while 2*len <= mstop (pass m for the full
+// transform; m/2 leaves the FINAL radix-4 stage to the caller — the
+// fused x+bin kernel runs it in registers and never writes it back)
 template <bool INV>
 __device__ void lds_fft4(cdouble* buf, int m, int TI, int PITCH,
-                         const cdouble* __restrict__ table) {
+                         const cdouble* __restrict__ table,
+                         int mstop) {
     const int T = blockDim.x;
     const int tid = threadIdx.x;
     const int bits = 31 - __clz((unsigned)m);
@@ -100,7 +104,7 @@ __device__ void lds_fft4(cdouble* buf, int m, int TI, int PITCH,
         nbk_sync_lds();
         len = 4;
     }
-    for (; 2 * len <= m; len <<= 2) {
+    for (; 2 * len <= mstop; len <<= 2) {
         const int h = len >> 1;
         const int tw = m / len;              // W_{2L}^p = table[p*tw]
         for (int w = tid; w < (m >> 2) * TI; w += T) {
@@ -154,7 +158,7 @@ __global__ void kfft_r2c_z(const double* __restrict__ real,
         buf[bitrev(q, bits)] = g[q];
     __syncthreads();
 
-    lds_fft4<false>(buf, m, 1, 1, table);
+    lds_fft4<false>(buf, m, 1, 1, table, m);
 
     // untwiddle split: X[k] = E[k] + W_nz^k * O[k], k = 0..m
     cdouble* out = (cdouble*)cplx + line * (m + 1);
@@ -208,7 +212,7 @@ __global__ void kfft_c2r_z(const double* __restrict__ cplx,
     }
     __syncthreads();
 
-    lds_fft4<true>(buf, m, 1, 1, table);
+    lds_fft4<true>(buf, m, 1, 1, table, m);
 
     // unpack: line[2t] = 2 Re(z[t]), line[2t+1] = 2 Im(z[t])
     cdouble* out = (cdouble*)(real + line * nz);
@@ -262,7 +266,7 @@ __global__ void kfft_c_strided(double* __restrict__ data,
     }
     __syncthreads();
 
-    lds_fft4<INV>(buf, nfft, TI, TI, table);
+    lds_fft4<INV>(buf, nfft, TI, TI, table, nfft);
 
     for (int w = threadIdx.x; w < nfft * TI; w += blockDim.x) {
         const int c = w % TI;
@@ -417,12 +421,6 @@ __global__ void kxfft_bin(const double* __restrict__ data,
     // LDS copy is not published until the first sync)
     const double k2last = k2edges_g[A.nx_edges - 1];
 
-    // bin-phase mapping: contiguous j-runs per thread at wave-uniform
-    // column c (kbin_run-style register merging + warm-start digitize)
-    const int CPT = T / TI;                 // threads per column
-    const int RB = (n0 + CPT - 1) / CPT;    // j-run length per thread
-    const int my_c = t / CPT;
-    const int my_j0 = (t - my_c * CPT) * RB;
 
     // ---- pipeline stages (macros so the prologue and loop share the
     //      exact code) --------------------------------------------------
@@ -523,12 +521,6 @@ __global__ void kxfft_bin(const double* __restrict__ data,
     int64_t tile = (int64_t)blockIdx.x * chunk;
     int p = 0;
 
-    // cross-tile run state of this thread's (column-slot, j-window)
-    int cbin = -1, cbx = -1, cbmu = -1;
-    double ak = 0.0, amu = 0.0, aw = 0.0;
-    double ay[MAXE];
-    for (int e = 0; e < MAXE; e++) ay[e] = 0.0;
-
     XBIN_SETUP(tile, 0)
     nbk_sync_lds();             // publishes h/ke/me/cxv and consts[0]
     if (tile < tend)
@@ -548,19 +540,51 @@ __global__ void kxfft_bin(const double* __restrict__ data,
             XBIN_LOAD(nxt, p ^ 1)   // in flight under FFT + bin below
 
         if (s_nlive[p] > 0) {
+            // all stages but the last (for IL, the shifted field runs
+            // its FULL transform — its elements are read back per
+            // element in the fused last stage below)
             if (PHASES & 1) {
-                lds_fft4<false>(buf, n0, TI, W, table);
+                lds_fft4<false>(buf, n0, TI, W, table, n0 / 2);
                 if (IL)
-                    lds_fft4<false>(buf2, n0, TI, W, table);
+                    lds_fft4<false>(buf2, n0, TI, W, table, n0);
             }
 
-            // run-merged bin phase: walk j = my_j0 .. +RB-1 of column
-            // my_c, merging same-bin neighbours in registers and
-            // warm-starting the digitize (the kbin_run recipe, reading
-            // LDS instead of HBM)
-            if ((PHASES & 2) && my_c < TI && !cskip[p][my_c]
-                && my_j0 < n0) {
-                const int c = my_c;
+            // FUSED FINAL STAGE + binning: the last radix-4 butterfly's
+            // four outputs (j = pos, pos+n0/4, pos+n0/2, pos+3n0/4 of
+            // column c = w%TI) are binned straight from REGISTERS —
+            // the finished line is never written back to LDS and the
+            // bin phase re-reads nothing (the v4 blocked-run readback
+            // was 16-way bank-conflicted and dominated the kernel).
+            if (PHASES & 2)
+            for (int w = t; w < (n0 >> 2) * TI; w += T) {
+                const int c = w % TI;
+                if (cskip[p][c]) continue;
+                const int pos = w / TI;
+                const int len = n0 >> 1;     // final stage geometry
+                const int s_ = (len >> 1) * W;
+                const int base = pos * W + c;
+                const int tw = 2;            // m / len
+                const cdouble w1 = table[pos * tw];
+                const cdouble w2 = table[2 * pos * tw];
+                const cdouble w3 = cmul(w1, w2);
+                const cdouble x0 = buf[base];
+                const cdouble x1 = buf[base + s_];
+                const cdouble x2 = buf[base + 2 * s_];
+                const cdouble x3 = buf[base + 3 * s_];
+                const cdouble b1 = cmul(x2, w1);
+                const cdouble b2 = cmul(x1, w2);
+                const cdouble b3 = cmul(x3, w3);
+                const cdouble e0 = cadd(x0, b2);
+                const cdouble e1 = csub(x0, b2);
+                const cdouble o0 = cadd(b1, b3);
+                const cdouble o1 = csub(b1, b3);
+                const cdouble io1 = {o1.im, -o1.re};
+                cdouble y4[4];
+                y4[0] = cadd(e0, o0);
+                y4[1] = cadd(e1, io1);
+                y4[2] = csub(e0, o0);
+                y4[3] = csub(e1, io1);
+
                 const double ky2 = cky2[p][c];
                 const double kz2 = ckz2[p][c];
                 const double kyl = ckyl[p][c];
@@ -573,9 +597,10 @@ __global__ void kxfft_bin(const double* __restrict__ data,
                 const bool nonsingular = cw2[p][c];
                 const double wgt = nonsingular ? 2.0 : 1.0;
                 const bool col_zero = czl[p][c];
-                const int jend = (my_j0 + RB < n0) ? my_j0 + RB : n0;
 
-                for (int j = my_j0; j < jend; j++) {
+                #pragma unroll
+                for (int q4 = 0; q4 < 4; q4++) {
+                    const int j = pos + q4 * (n0 >> 2);
                     const double fx = freq_full(j, A.n0);
                     const double kx = fx * A.k0x;
                     const double k2 = (kx * kx + ky2) + kz2;
@@ -583,28 +608,12 @@ __global__ void kxfft_bin(const double* __restrict__ data,
                     const double kmag = sqrt(k2);
                     double mu = (kx * A.losx + kyl) + kzl;
                     mu = (kmag == 0.0) ? 0.0 : mu / kmag;
-
-                    int bx;
-                    if (cbx >= 0
-                        && (cbx == 0 || ke[cbx - 1] <= k2)
-                        && (cbx == A.nx_edges || k2 < ke[cbx]))
-                        bx = cbx;
-                    else
-                        bx = digx(ke, A.nx_edges, k2);
-                    int bmu;
-                    if (cbmu >= 0
-                        && (cbmu == 0 || me[cbmu - 1] <= mu)
-                        && (cbmu == A.nmu_edges || mu < me[cbmu]))
-                        bmu = cbmu;
-                    else
-                        bmu = digx(me, A.nmu_edges, mu);
+                    const int bx = digx(ke, A.nx_edges, k2);
+                    const int bmu = digx(me, A.nmu_edges, mu);
                     const int bin = bx * (A.nmu_edges + 1) + bmu;
 
-                    cdouble a = buf[j * W + c];
+                    cdouble a = y4[q4];
                     if (IL) {
-                        // c = a/2 + b/2 exp(i k.H/2) — the
-                        // nbk_interlace_combine formula with the phase
-                        // composed as px[j] * pyz[c]
                         const cdouble b = buf2[j * W + c];
                         const cdouble ph = cmul(pxv[j], phc);
                         a = cadd(cscale(a, 0.5),
@@ -618,23 +627,9 @@ __global__ void kxfft_bin(const double* __restrict__ data,
                     if (A.clear_zero && col_zero && j == 0)
                         vre = 0.0;
 
-                    if (bin != cbin) {
-                        if (cbin >= 0) {
-                            atomicAdd(&h[cbin], ak);
-                            atomicAdd(&h[NB + cbin], amu);
-                            atomicAdd(&h[2 * NB + cbin], aw);
-                            for (int e = 0; e < A.nell && e < MAXE; e++)
-                                atomicAdd(&h[(3 + 2 * e) * NB + cbin],
-                                          ay[e]);
-                        }
-                        cbin = bin; cbx = bx; cbmu = bmu;
-                        ak = 0.0; amu = 0.0; aw = 0.0;
-                        for (int e = 0; e < MAXE; e++) ay[e] = 0.0;
-                    }
-                    ak += kmag * wgt;
-                    amu += mu * wgt;
-                    aw += wgt;
-
+                    atomicAdd(&h[bin], kmag * wgt);
+                    atomicAdd(&h[NB + bin], mu * wgt);
+                    atomicAdd(&h[2 * NB + bin], wgt);
                     double Pm1 = 0.0, P = 1.0;
                     int e = 0;
                     for (int l = 0; e < A.nell && e < MAXE; l++) {
@@ -646,28 +641,17 @@ __global__ void kxfft_bin(const double* __restrict__ data,
                         }
                         if (l == A.ells[e]) {
                             double wr = vre * P;
-                            if (nonsingular) {
-                                // conjugate-pair parity: odd ell keeps
-                                // only the (zero) imaginary part
+                            if (nonsingular)
                                 wr = (l % 2) ? 0.0 : 2.0 * wr;
-                            }
-                            ay[e] += wr * (2.0 * l + 1.0);
+                            atomicAdd(&h[(3 + 2 * e) * NB + bin],
+                                      wr * (2.0 * l + 1.0));
                             e++;
                         }
                     }
                 }
-                // no per-tile flush: cbin and its partial sums carry
-                // into the next (adjacent) tile's columns
             }
         }
         nbk_sync_lds();         // bin done before the next tile's store
-    }
-    if (cbin >= 0) {            // final flush of the cross-tile run
-        atomicAdd(&h[cbin], ak);
-        atomicAdd(&h[NB + cbin], amu);
-        atomicAdd(&h[2 * NB + cbin], aw);
-        for (int e = 0; e < A.nell && e < MAXE; e++)
-            atomicAdd(&h[(3 + 2 * e) * NB + cbin], ay[e]);
     }
 #undef XBIN_SETUP
 #undef XBIN_LOAD
