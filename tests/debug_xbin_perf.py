@@ -47,7 +47,7 @@ def one_config(reps=3):
 
     def run_xbin():
         hiplib.check(lib.nbk_fft_x_bin_f64(
-            hiplib.dptr(data), nmesh, n_inner, 0, boxa,
+            hiplib.dptr(data), None, nmesh, n_inner, 0, boxa,
             0, 0, 1, vol,
             hiplib.dptr(k2e), len(kedges), hiplib.dptr(mue), len(muedges),
             los, ells, 1, hiplib.dptr(sums), hiplib.cur_stream()),
