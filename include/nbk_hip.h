@@ -340,6 +340,11 @@ int nbk_power_bin_f64(const double* c1, const double* c2, double volume,
  * couples columns, so the caller handles those two planes with the
  * standalone kernels and the same out_sums buffer (all the binning
  * entry points accumulate). */
+/* dig_hints = {k_lo, k_invd, mu_lo, mu_invd}: the uniform edge-grid
+ * parameters (kedges = arange(k_lo, ., 1/k_invd), muedges =
+ * linspace(mu_lo, ., 1/mu_invd)) used only as digitize STARTING
+ * GUESSES — assignment is corrected against the exact edge arrays, so
+ * it stays bit-identical to numpy.digitize. */
 int nbk_fft_x_bin_f64(const double* data, const double* data2,
                       const int64_t nmesh[3],
                       int64_t n_inner, int64_t y_off,
@@ -348,6 +353,7 @@ int nbk_fft_x_bin_f64(const double* data, const double* data2,
                       int clear_zero, double volume,
                       const double* kedges, int64_t nx_edges,
                       const double* muedges, int64_t nmu_edges,
+                      const double dig_hints[4],
                       const double los[3],
                       const int* ells, int nell,
                       double* out_sums, void* stream);

@@ -219,6 +219,7 @@ class FFTPower(FFTBase):
         # known up front, so the dk == 0 unique-edges path stays on the
         # standard flow.
         defer = (fused and self.second is self.first and dk > 0
+                 and kmin >= 0
                  and _xbin_ok(self.first, self.attrs['Nmesh']))
         if defer:
             kmax_eff = kmax
@@ -699,6 +700,13 @@ def _project_power_xbin(tensor, tensor2, pm, n_inner, comp1, volume,
 
     win1, interl1 = comp1
     y_off = pm.y_start if ws > 1 else 0
+    # uniform-grid digitize guesses (run() gated on dk > 0, kmin >= 0;
+    # the kernel corrects the guess against the exact edges, so these
+    # never change the assignment)
+    dk_g = float(xedges[1] - xedges[0]) if Nx >= 1 else 1.0
+    dmu_g = float(muedges[1] - muedges[0]) if Nmu >= 1 else 1.0
+    hints = hiplib.f64_arr([float(xedges[0]), 1.0 / dk_g,
+                            float(muedges[0]), 1.0 / dmu_g])
     hiplib.check(lib.nbk_fft_x_bin_f64(
         hiplib.dptr(tensor),
         hiplib.dptr(tensor2) if tensor2 is not None else None,
@@ -706,7 +714,7 @@ def _project_power_xbin(tensor, tensor2, pm, n_inner, comp1, volume,
         int(n_inner), int(y_off), hiplib.f64_arr(pm.BoxSize),
         int(win1), int(interl1), 1, float(volume),
         hiplib.dptr(k2edges_t), len(xedges),
-        hiplib.dptr(muedges_t), len(muedges),
+        hiplib.dptr(muedges_t), len(muedges), hints,
         hiplib.f64_arr(los), hiplib.int_arr(_poles), Nell,
         hiplib.dptr(sums), hiplib.cur_stream()), 'nbk_fft_x_bin_f64')
 

@@ -138,6 +138,7 @@ def _declare(lib):
                                       ctypes.c_int, ctypes.c_int,
                                       ctypes.c_int, c_f64,
                                       c_void, c_i64, c_void, c_i64,
+                                      c_f64_p,
                                       c_f64_p, c_int_p, ctypes.c_int,
                                       c_void, c_void]
     lib.nbk_axpy_f64.restype = ctypes.c_int

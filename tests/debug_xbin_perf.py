@@ -50,6 +50,7 @@ def one_config(reps=3):
             hiplib.dptr(data), None, nmesh, n_inner, 0, boxa,
             0, 0, 1, vol,
             hiplib.dptr(k2e), len(kedges), hiplib.dptr(mue), len(muedges),
+            hiplib.f64_arr([0.0, 1.0 / dk, -1.0, 0.5]),
             los, ells, 1, hiplib.dptr(sums), hiplib.cur_stream()),
             'nbk_fft_x_bin_f64')
 
