@@ -112,11 +112,17 @@ int nbk_paint_sorted_f64(const double* pos, const double* mass, int64_t n,
  * the read), accumulate=1 adds (hold semantics for chunked paints).
  * Returns NBK_ERR_UNSUPPORTED when no LDS tile fits (fall back to
  * nbk_paint_f64). */
+/* pair_gs >= 0: `rowtab` is the PAIR-BUCKET table of the duplicating
+ * sort (nbk_psort_*): [(n0/2)*(n1>>pair_gs)+1] exclusive bases, one
+ * contiguous range per (x-plane pair, y-group); the tile geometry is
+ * pinned to (1 plane x 1<<pair_gs rows) and the deposit masks drop the
+ * duplicated out-of-tile copies.  pair_gs = -1: per-row table from
+ * nbk_bucket_fine_f64. */
 int nbk_paint_gather_f64(const double* pos, const double* mass, int64_t n,
                          const int64_t nmesh[3], const double box[3],
                          int window, double shift, const int* rowtab,
                          double* mesh, int64_t x0, int64_t nx_local,
-                         int accumulate, void* stream);
+                         int accumulate, int pair_gs, void* stream);
 
 /* gather paint with the forward z-axis FFT fused into the tile flush:
  * writes the z half-spectrum ((nx_local, n1, n2/2+1) interleaved c128,
@@ -129,7 +135,7 @@ int nbk_paint_gather_fft_f64(const double* pos, const double* mass,
                              const double box[3],
                              int window, double shift, const int* rowtab,
                              double* zspec, int64_t x0, int64_t nx_local,
-                             double scale, void* stream);
+                             double scale, int pair_gs, void* stream);
 
 /* paint locality sort --------------------------------------------------
  * Two-pass counting sort of particles by coarse mesh cell
@@ -184,6 +190,22 @@ int nbk_xsort_count_f64(const double* pos_aos, int64_t n, int chunk,
                         const int64_t nmesh[3], const double box[3],
                         int ys, int* mat, int* scrambled_flag,
                         void* stream);
+/* PAIR-BUCKET duplicating sort (see nbk_paint_gather_f64 pair_gs):
+ * counting sort by (x-plane pair, y row-group) with each particle
+ * duplicated into every y-group rows [iy+dlo, iy+dhi] of its deposit
+ * stencil touch (<= 2 copies; dhi-dlo must be < 1<<ys).  Same
+ * count-matrix/scan/scatter structure as nbk_xsort_*; n_out =
+ * bucket_bases[nbuck] after nbk_scan_matrix_i32. */
+int nbk_psort_count_f64(const double* pos_aos, int64_t n, int chunk,
+                        const int64_t nmesh[3], const double box[3],
+                        int ys, int dlo, int dhi, int* mat, void* stream);
+int nbk_psort_scatter_f64(const double* pos_aos, const double* mass,
+                          int64_t n, int chunk, const int64_t nmesh[3],
+                          const double box[3], int ys, int dlo, int dhi,
+                          const int* bases, int64_t n_out,
+                          double* pos_out, double* mass_out,
+                          void* stream);
+
 int nbk_scan_matrix_i32(const int* mat, int64_t nblocks, int64_t nbuck,
                         int* colsum_tmp, int* bases, int* bucket_bases,
                         void* stream);

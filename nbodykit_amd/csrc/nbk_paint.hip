@@ -375,7 +375,14 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                               int RG, int P, int xlo, int xhi,
                               int accumulate,
                               const cdouble* __restrict__ table /* W_n2 */,
-                              double scale)
+                              double scale,
+                              int gs /* >= 0: rowtab is the PAIR-BUCKET
+                                  table [(n0/2)*(n1>>gs)+1] of the
+                                  duplicating sort (1<<gs == RG); the
+                                  tile reads whole (pair, group) ranges
+                                  and the deposit masks drop the
+                                  out-of-tile copies.  -1: per-row
+                                  table (nbk_bucket_fine_f64) */)
 {
     constexpr int SUP = (WINDOW == NBK_WINDOW_CIC) ? 2
                       : (WINDOW == NBK_WINDOW_TSC) ? 3 : 4;
@@ -399,11 +406,26 @@ __global__ void kpaint_gather(const double* __restrict__ px,
 
     // row intervals (wrapped) whose particles can deposit into the tile
     const int64_t rspan = (int64_t)RG + (xhi - xlo);
+    int64_t prev_pair = -1;
     for (int dp = xlo; dp <= P - 1 + xhi; dp++) {
         const int64_t p = wrap_idx(px0 + dp, n0);
         int64_t ivals[2][2];
         int niv;
-        if (rspan >= n1) {
+        if (gs >= 0) {
+            // PAIR-BUCKET mode: the duplicating sort already placed a
+            // copy of every stencil-relevant particle in this tile's
+            // own (pair, group) bucket — one contiguous range, no
+            // halo-row lookups.  Consecutive source planes share a
+            // pair: read each pair once.
+            const int64_t q = p >> 1;
+            if (q == prev_pair) continue;
+            prev_pair = q;
+            const int64_t ng = n1 >> gs;
+            const int64_t bidx = q * ng + (r0 >> gs);
+            ivals[0][0] = rowtab[bidx];
+            ivals[0][1] = rowtab[bidx + 1];
+            niv = 1;
+        } else if (rspan >= n1) {
             ivals[0][0] = 0; ivals[0][1] = n1 - 1; niv = 1;
         } else {
             const int64_t a = wrap_idx(r0 + xlo, n1);
@@ -415,8 +437,10 @@ __global__ void kpaint_gather(const double* __restrict__ px,
             }
         }
         for (int v = 0; v < niv; v++) {
-            const int64_t i0 = rowtab[p * n1 + ivals[v][0]];
-            const int64_t i1 = rowtab[p * n1 + ivals[v][1] + 1];
+            const int64_t i0 = (gs >= 0) ? ivals[v][0]
+                : rowtab[p * n1 + ivals[v][0]];
+            const int64_t i1 = (gs >= 0) ? ivals[v][1]
+                : rowtab[p * n1 + ivals[v][1] + 1];
             for (int64_t i = i0 + t; i < i1; i += T) {
                 const double u0 = px[i] * invH0 + shift;
                 const double u1 = py[i] * invH1 + shift;
@@ -874,7 +898,7 @@ extern "C" int nbk_paint_gather_f64(const double* pos, const double* mass,
                                     const int* rowtab,
                                     double* mesh, int64_t x0,
                                     int64_t nx_local, int accumulate,
-                                    void* stream)
+                                    int pair_gs, void* stream)
 {
     const int64_t n0 = nmesh[0], n1 = nmesh[1], n2 = nmesh[2];
     if (n2 > 20480) {
@@ -901,7 +925,18 @@ extern "C" int nbk_paint_gather_f64(const double* pos, const double* mass,
     const int span = xhi - xlo;
 
     int P, RG;
-    nbk_pick_tile(nx_local, n1, n2, 0, span, span, &P, &RG);
+    if (pair_gs >= 0) {
+        // pair-bucket mode: tile geometry is pinned by the sort's
+        // group size (one plane x one y-group)
+        P = 1;
+        RG = 1 << pair_gs;
+        if ((int64_t)RG * n2 * 8 > 160 * 1024 || n1 % RG) {
+            NBK_SET_ERR("nbk_paint_gather_f64: bad pair_gs=%d", pair_gs);
+            return NBK_ERR_ARG;
+        }
+    } else {
+        nbk_pick_tile(nx_local, n1, n2, 0, span, span, &P, &RG);
+    }
     const int64_t grid = (nx_local / P) * (n1 / RG);
     const size_t lds = (size_t)P * RG * n2 * sizeof(double);
     hipStream_t s = (hipStream_t)stream;
@@ -929,7 +964,8 @@ extern "C" int nbk_paint_gather_f64(const double* pos, const double* mass,
                            mass, n, n0, n1, n2, \
                            n0 / box[0], n1 / box[1], n2 / box[2], shift, \
                            rowtab, mesh, x0, nx_local, RG, P, xlo, xhi, \
-                           accumulate, (const cdouble*)nullptr, 1.0)
+                           accumulate, (const cdouble*)nullptr, 1.0, \
+                           pair_gs)
     if (P == 1) {
         if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GATHER(NBK_WINDOW_CIC, 1);
         else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GATHER(NBK_WINDOW_TSC, 1);
@@ -952,7 +988,7 @@ extern "C" int nbk_paint_gather_fft_f64(const double* pos,
                                         const int* rowtab,
                                         double* zspec, int64_t x0,
                                         int64_t nx_local, double scale,
-                                        void* stream)
+                                        int pair_gs, void* stream)
 {
     const int64_t n0 = nmesh[0], n1 = nmesh[1], n2 = nmesh[2];
     if (n2 < 8 || n2 > 4096 || (n2 & (n2 - 1))) {
@@ -987,7 +1023,17 @@ extern "C" int nbk_paint_gather_fft_f64(const double* pos,
     const int span = xhi - xlo;
 
     int P, RG;
-    nbk_pick_tile(nx_local, n1, n2, 4, span, span, &P, &RG);
+    if (pair_gs >= 0) {
+        P = 1;
+        RG = 1 << pair_gs;
+        if ((int64_t)RG * (n2 + 4) * 8 > 160 * 1024 || n1 % RG) {
+            NBK_SET_ERR("nbk_paint_gather_fft_f64: bad pair_gs=%d",
+                        pair_gs);
+            return NBK_ERR_ARG;
+        }
+    } else {
+        nbk_pick_tile(nx_local, n1, n2, 4, span, span, &P, &RG);
+    }
     const int64_t grid = (nx_local / P) * (n1 / RG);
     const size_t lds = (size_t)P * RG * (n2 + 4) * sizeof(double);
     hipStream_t s = (hipStream_t)stream;
@@ -1015,7 +1061,7 @@ extern "C" int nbk_paint_gather_fft_f64(const double* pos,
                            mass, n, n0, n1, n2, \
                            n0 / box[0], n1 / box[1], n2 / box[2], shift, \
                            rowtab, zspec, x0, nx_local, RG, P, xlo, \
-                           xhi, 0, (const cdouble*)table, scale)
+                           xhi, 0, (const cdouble*)table, scale, pair_gs)
     if (P == 1) {
         if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GFFT(NBK_WINDOW_CIC, 1);
         else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GFFT(NBK_WINDOW_TSC, 1);

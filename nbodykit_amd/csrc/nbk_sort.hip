@@ -428,6 +428,91 @@ __global__ void kbucket_fine_rows(const double* __restrict__ px,
     }
 }
 
+
+// ---- PAIR-BUCKET sort (the fine-pass eliminator) ---------------------
+// Key = (x-plane PAIR, y row-GROUP) with nbuck = (n0/2) * (n1 >> ys)
+// <= 40960 (i32 LDS histograms), and particles DUPLICATED into every
+// y-group their deposit stencil touches (rows [iy+dlo, iy+dhi]; the
+// stencil spans <= 5 rows << the group size, so <= 2 copies).  The
+// ownership-gather paint then reads each (pair, group) bucket range
+// directly — its deposit masks drop the out-of-tile copies — and the
+// whole per-row fine pass (56 GB of moves at C4) disappears.  The cost
+// moves into the paint's source reads: a pair bucket holds ~3 planes
+// of particles, read by each of the pair's 2 tiles.
+__device__ __forceinline__ void pair_keys(double x, double y,
+                                          double invH0, double invH1,
+                                          int64_t n0, int64_t n1, int ys,
+                                          int dlo, int dhi,
+                                          int64_t* k1, int64_t* k2) {
+    const int64_t ix = wrap_idx((int64_t)floor(x * invH0), n0);
+    const int64_t iy = (int64_t)floor(y * invH1);
+    const int64_t ng = n1 >> ys;
+    const int64_t g1 = wrap_idx(iy + dlo, n1) >> ys;
+    const int64_t g2 = wrap_idx(iy + dhi, n1) >> ys;
+    const int64_t base = (ix >> 1) * ng;
+    *k1 = base + g1;
+    *k2 = (g2 == g1) ? -1 : base + g2;
+}
+
+__global__ void kpsort_count(const double* __restrict__ pos, int64_t n,
+                             int chunk, int64_t n0, int64_t n1,
+                             double invH0, double invH1,
+                             int ys, int dlo, int dhi, int64_t nbuck,
+                             int* __restrict__ mat)
+{
+    extern __shared__ int hist[];   // nbuck ints
+    for (int64_t b = threadIdx.x; b < nbuck; b += blockDim.x) hist[b] = 0;
+    __syncthreads();
+    const int64_t beg = (int64_t)blockIdx.x * chunk;
+    const int64_t end = (beg + chunk < n) ? beg + chunk : n;
+    for (int64_t i = beg + threadIdx.x; i < end; i += blockDim.x) {
+        int64_t k1, k2;
+        pair_keys(pos[3 * i], pos[3 * i + 1], invH0, invH1, n0, n1,
+                  ys, dlo, dhi, &k1, &k2);
+        atomicAdd(&hist[k1], 1);
+        if (k2 >= 0) atomicAdd(&hist[k2], 1);
+    }
+    __syncthreads();
+    for (int64_t b = threadIdx.x; b < nbuck; b += blockDim.x)
+        mat[(int64_t)blockIdx.x * nbuck + b] = hist[b];
+}
+
+__global__ void kpsort_scatter(const double* __restrict__ pos,
+                               const double* __restrict__ mass, int64_t n,
+                               int chunk, int64_t n0, int64_t n1,
+                               double invH0, double invH1,
+                               int ys, int dlo, int dhi, int64_t nbuck,
+                               const int* __restrict__ bases,
+                               int64_t n_out,
+                               double* __restrict__ ox,
+                               double* __restrict__ oy,
+                               double* __restrict__ oz,
+                               double* __restrict__ om)
+{
+    extern __shared__ int cur[];    // nbuck running cursors
+    for (int64_t b = threadIdx.x; b < nbuck; b += blockDim.x)
+        cur[b] = bases[(int64_t)blockIdx.x * nbuck + b];
+    __syncthreads();
+    const int64_t beg = (int64_t)blockIdx.x * chunk;
+    const int64_t end = (beg + chunk < n) ? beg + chunk : n;
+    for (int64_t i = beg + threadIdx.x; i < end; i += blockDim.x) {
+        const double x = pos[3 * i], y = pos[3 * i + 1],
+                     z = pos[3 * i + 2];
+        int64_t k1, k2;
+        pair_keys(x, y, invH0, invH1, n0, n1, ys, dlo, dhi, &k1, &k2);
+        const double m = mass ? mass[i] : 0.0;
+        int64_t t = (int64_t)atomicAdd(&cur[k1], 1);
+        ox[t] = x; oy[t] = y; oz[t] = z;
+        if (mass) om[t] = m;
+        if (k2 >= 0) {
+            t = (int64_t)atomicAdd(&cur[k2], 1);
+            ox[t] = x; oy[t] = y; oz[t] = z;
+            if (mass) om[t] = m;
+        }
+    }
+    (void)n_out;
+}
+
 int sgrid(int64_t n) {
     int64_t g = (n + 255) / 256;
     if (g > 1048576) g = 1048576;
@@ -604,6 +689,61 @@ extern "C" int nbk_bucket_fine_f64(const double* pos_soa,
                        nmesh[1] / box[1], nmesh[2] / box[2], ys,
                        bucket_bases, soa_out, soa_out + n,
                        soa_out + 2 * n, mass_out, rowtab);
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}
+
+extern "C" int nbk_psort_count_f64(const double* pos_aos, int64_t n,
+                                   int chunk, const int64_t nmesh[3],
+                                   const double box[3], int ys,
+                                   int dlo, int dhi, int* mat,
+                                   void* stream)
+{
+    if (n == 0) return NBK_OK;
+    const int64_t nbuck = (nmesh[0] >> 1) * (nmesh[1] >> ys);
+    if (nbuck > NBK_SORT_LDS_INTS || (nmesh[0] & 1)
+        || (nmesh[1] % ((int64_t)1 << ys))
+        || (dhi - dlo) >= ((int64_t)1 << ys)) {
+        NBK_SET_ERR("nbk_psort_count_f64: bad geometry ys=%d", ys);
+        return NBK_ERR_ARG;
+    }
+    const int64_t nblocks = (n + chunk - 1) / chunk;
+    const size_t lds = (size_t)nbuck * sizeof(int);
+    raise_lds(reinterpret_cast<const void*>(&kpsort_count), lds);
+    hipLaunchKernelGGL(kpsort_count, dim3((uint32_t)nblocks), dim3(1024),
+                       lds, (hipStream_t)stream, pos_aos, n, chunk,
+                       nmesh[0], nmesh[1],
+                       nmesh[0] / box[0], nmesh[1] / box[1],
+                       ys, dlo, dhi, nbuck, mat);
+    NBK_CHECK_HIP(hipGetLastError());
+    return NBK_OK;
+}
+
+extern "C" int nbk_psort_scatter_f64(const double* pos_aos,
+                                     const double* mass, int64_t n,
+                                     int chunk, const int64_t nmesh[3],
+                                     const double box[3], int ys,
+                                     int dlo, int dhi,
+                                     const int* bases, int64_t n_out,
+                                     double* pos_out, double* mass_out,
+                                     void* stream)
+{
+    if (n == 0) return NBK_OK;
+    const int64_t nbuck = (nmesh[0] >> 1) * (nmesh[1] >> ys);
+    if (nbuck > NBK_SORT_LDS_INTS) {
+        NBK_SET_ERR("nbk_psort_scatter_f64: bad ys=%d", ys);
+        return NBK_ERR_ARG;
+    }
+    const int64_t nblocks = (n + chunk - 1) / chunk;
+    const size_t lds = (size_t)nbuck * sizeof(int);
+    raise_lds(reinterpret_cast<const void*>(&kpsort_scatter), lds);
+    hipLaunchKernelGGL(kpsort_scatter, dim3((uint32_t)nblocks),
+                       dim3(1024), lds, (hipStream_t)stream, pos_aos,
+                       mass, n, chunk, nmesh[0], nmesh[1],
+                       nmesh[0] / box[0], nmesh[1] / box[1],
+                       ys, dlo, dhi, nbuck, bases, n_out,
+                       pos_out, pos_out + n_out, pos_out + 2 * n_out,
+                       mass_out);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

@@ -21,6 +21,7 @@ EXPORTED_SYMBOLS = [
     'nbk_bucket_count_f64', 'nbk_bucket_scatter_f64',
     'nbk_xsort_count_f64', 'nbk_xsort_scatter_f64',
     'nbk_bucket_fine_f64', 'nbk_scan_matrix_i32', 'nbk_paint_gather_f64',
+    'nbk_psort_count_f64', 'nbk_psort_scatter_f64',
     'nbk_paint_gather_fft_f64',
     'nbk_fft_r2c_z', 'nbk_fft_c2r_z', 'nbk_fft_c_strided',
     'nbk_compensate_f64', 'nbk_interlace_combine_f64', 'nbk_power3d_f64',
@@ -79,13 +80,25 @@ def _declare(lib):
     lib.nbk_paint_gather_f64.argtypes = [c_void, c_void, c_i64, c_i64_p,
                                          c_f64_p, ctypes.c_int, c_f64,
                                          c_void, c_void, c_i64, c_i64,
-                                         ctypes.c_int, c_void]
+                                         ctypes.c_int, ctypes.c_int,
+                                         c_void]
     lib.nbk_paint_gather_fft_f64.restype = ctypes.c_int
     lib.nbk_paint_gather_fft_f64.argtypes = [c_void, c_void, c_i64,
                                              c_i64_p, c_f64_p,
                                              ctypes.c_int, c_f64, c_void,
                                              c_void, c_i64, c_i64, c_f64,
-                                             c_void]
+                                             ctypes.c_int, c_void]
+    lib.nbk_psort_count_f64.restype = ctypes.c_int
+    lib.nbk_psort_count_f64.argtypes = [c_void, c_i64, ctypes.c_int,
+                                        c_i64_p, c_f64_p, ctypes.c_int,
+                                        ctypes.c_int, ctypes.c_int,
+                                        c_void, c_void]
+    lib.nbk_psort_scatter_f64.restype = ctypes.c_int
+    lib.nbk_psort_scatter_f64.argtypes = [c_void, c_void, c_i64,
+                                          ctypes.c_int, c_i64_p, c_f64_p,
+                                          ctypes.c_int, ctypes.c_int,
+                                          ctypes.c_int, c_void, c_i64,
+                                          c_void, c_void, c_void]
     lib.nbk_paint_sorted_f64.restype = ctypes.c_int
     lib.nbk_paint_sorted_f64.argtypes = lib.nbk_paint_f64.argtypes
     lib.nbk_readout_f64.restype = ctypes.c_int

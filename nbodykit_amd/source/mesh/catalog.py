@@ -52,6 +52,11 @@ def _is_trivial_true(sel):
 
 
 
+def _pair_enabled():
+    import os
+    return os.environ.get('NBK_SORT_PAIR', '1') != '0'
+
+
 def _two_level_ys(pm):
     """The y-group shift of the two-level locality sort, or None when no
     LDS budget fits this mesh (coarse histogram n0*(n1>>ys) ints and fine
@@ -73,10 +78,44 @@ def _two_level_ys(pm):
     return None
 
 
-def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False):
-    """Return (pos_soa, mass, cell_sorted, rowtab) for the deposit
-    kernel,
-    bucket-sorting the chunk by mesh cell when it arrives scrambled.
+_PAIR_GHOST = {
+    # deposit rows relative to iy = floor(y/H): [Dlo, Dhi] covering the
+    # stencil at BOTH interlacing shifts when interlaced (the one sorted
+    # array feeds both paints); derived from the launchers' (dmin, dmax)
+    # with Dhi = dmax + support - 1
+    ('cic', False): (0, 1), ('cic', True): (0, 2),
+    ('tsc', False): (-1, 1), ('tsc', True): (-1, 2),
+    ('pcs', False): (-1, 2), ('pcs', True): (-1, 3),
+}
+
+
+def _pair_gs(pm):
+    """The y-group shift of the PAIR-BUCKET duplicating sort, or None
+    when the geometry does not admit it (tile = 1 plane x 1<<gs rows x
+    (n2+4) doubles in LDS; (n0/2)*(n1>>gs) buckets <= 40960)."""
+    n0, n1, n2 = (int(x) for x in pm.Nmesh)
+    if n0 % 2:
+        return None
+    best = None
+    for gs in range(1, max(1, n1.bit_length())):
+        rg = 1 << gs
+        if rg > n1 or n1 % rg:
+            break
+        if rg * (n2 + 4) * 8 > 160 * 1024:
+            break
+        if (n0 >> 1) * (n1 >> gs) <= 40960:
+            best = gs
+    return best
+
+
+def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
+                       window='cic', interlaced=False):
+    """Return (pos_soa, mass, cell_sorted, table) for the deposit
+    kernel, bucket-sorting the chunk by mesh cell when it arrives
+    scrambled.  ``table`` is a per-row rowtab tensor (legacy two-level
+    sort), a ``(bucket_bases, gs, n_out)`` tuple (pair-bucket
+    duplicating sort — the default for big meshes; NBK_SORT_PAIR=0
+    reverts), or None (scatter-paint fallback).
 
     ``force_rowtab`` makes the two-level path run regardless of the size
     thresholds and always emit the row table (the fused paint+z-FFT path
@@ -101,8 +140,13 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False):
     n = len(pos_t)
     n0, n1, n2 = (int(x) for x in pm.Nmesh)
     if force_rowtab and n == 0:
-        # empty rank (post-routing): an all-zero row table means every
-        # row range is empty and the gather kernel paints zeros
+        # empty rank (post-routing): an all-zero table means every
+        # range is empty and the gather kernel paints zeros
+        gs0 = _pair_gs(pm) if _pair_enabled() else None
+        if gs0 is not None:
+            btab = torch.zeros((n0 >> 1) * (n1 >> gs0) + 1,
+                               dtype=torch.int32, device='cuda')
+            return pos_t.t().contiguous(), mass_t, True, (btab, gs0, 0)
         rowtab = torch.zeros(n0 * n1 + 1, dtype=torch.int32,
                              device='cuda')
         return pos_t.t().contiguous(), mass_t, True, rowtab
@@ -153,11 +197,53 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False):
     # kernel (LDS count + block scan + placement) emitting the exact
     # cell order.  ys balances the two LDS budgets: coarse histogram
     # n0*(n1>>ys) ints vs fine window (1<<ys)*n2 ints, both <= 40960.
+    big = (force_rowtab
+           or (n >= _global_options['sort_two_level_min_n']
+               and ncells > _global_options['sort_two_level_min_cells']))
+
+    # PAIR-BUCKET duplicating sort: one counting sort by (x-plane pair,
+    # y row-group) with stencil-boundary particles duplicated into both
+    # touched groups — the per-row fine pass (the single biggest sort
+    # cost) disappears, and the gather paint reads whole bucket ranges
+    # with its deposit masks dropping the out-of-tile copies.
+    gs = _pair_gs(pm) if (big and _pair_enabled() and n2 <= 20480) \
+        else None
+    if gs is not None:
+        dlo, dhi = _PAIR_GHOST[(window, bool(interlaced))]
+        if dhi - dlo >= (1 << gs):
+            gs = None
+    if gs is not None:
+        nbuck = (n0 >> 1) * (n1 >> gs)
+        CH = 262144
+        nblocks = (n + CH - 1) // CH
+        mat = torch.empty(nblocks * nbuck, dtype=torch.int32,
+                          device='cuda')
+        hiplib.check(lib.nbk_psort_count_f64(
+            hiplib.dptr(pos_in), n, CH, nmesh, box, gs, dlo, dhi,
+            hiplib.dptr(mat), stream), 'nbk_psort_count_f64')
+        colsum = torch.empty(nbuck, dtype=torch.int32, device='cuda')
+        bases = torch.empty(nblocks * nbuck, dtype=torch.int32,
+                            device='cuda')
+        bucket_bases = torch.empty(nbuck + 1, dtype=torch.int32,
+                                   device='cuda')
+        hiplib.check(lib.nbk_scan_matrix_i32(
+            hiplib.dptr(mat), nblocks, nbuck, hiplib.dptr(colsum),
+            hiplib.dptr(bases), hiplib.dptr(bucket_bases), stream),
+            'nbk_scan_matrix_i32')
+        n_out = int(bucket_bases[-1].item())
+        out = torch.empty(3 * n_out, dtype=torch.float64, device='cuda')
+        out_m = None
+        if mass_t is not None:
+            out_m = torch.empty(n_out, dtype=torch.float64,
+                                device='cuda')
+        hiplib.check(lib.nbk_psort_scatter_f64(
+            hiplib.dptr(pos_in), hiplib.dptr(mass_t), n, CH, nmesh, box,
+            gs, dlo, dhi, hiplib.dptr(bases), n_out, hiplib.dptr(out),
+            hiplib.dptr(out_m), stream), 'nbk_psort_scatter_f64')
+        return out, out_m, True, (bucket_bases, gs, n_out)
+
     ys_fine = _two_level_ys(pm)
-    use_two = ys_fine is not None and (
-        force_rowtab
-        or (n >= _global_options['sort_two_level_min_n']
-            and ncells > _global_options['sort_two_level_min_cells']))
+    use_two = ys_fine is not None and big
     if use_two:
         ys = ys_fine
         nbuck = n0 * (n1 >> ys)
@@ -408,26 +494,31 @@ class CatalogMesh(MeshSource):
             n = len(pos_t)
             if n > 0:
                 pos_soa, mass_t, sorted_, rowtab = _prepare_particles(
-                    pos_t, mass_t, pm)
+                    pos_t, mass_t, pm, window=self.resampler,
+                    interlaced=interlaced)
+                if isinstance(rowtab, tuple):
+                    table, pair_gs, n_eff = rowtab
+                else:
+                    table, pair_gs, n_eff = rowtab, -1, n
                 # fresh fields take plain stores on the first chunk
                 acc = 0 if (i == 0 and out is None) else 1
                 tgt = (real1 if interlaced else toret).value
                 with profiling.collect('paint', n * (1 + interlaced)):
-                    if rowtab is not None:
+                    if table is not None:
                         hiplib.check(lib.nbk_paint_gather_f64(
-                            hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n,
-                            nmesh, box, window_id, 0.0,
-                            hiplib.dptr(rowtab), hiplib.dptr(tgt),
-                            pm.x_start, pm.nx_local, acc, stream),
-                            'nbk_paint_gather_f64')
+                            hiplib.dptr(pos_soa), hiplib.dptr(mass_t),
+                            n_eff, nmesh, box, window_id, 0.0,
+                            hiplib.dptr(table), hiplib.dptr(tgt),
+                            pm.x_start, pm.nx_local, acc, pair_gs,
+                            stream), 'nbk_paint_gather_f64')
                         if interlaced:
                             hiplib.check(lib.nbk_paint_gather_f64(
                                 hiplib.dptr(pos_soa), hiplib.dptr(mass_t),
-                                n, nmesh, box, window_id, 0.5,
-                                hiplib.dptr(rowtab),
+                                n_eff, nmesh, box, window_id, 0.5,
+                                hiplib.dptr(table),
                                 hiplib.dptr(real2.value),
-                                pm.x_start, pm.nx_local, acc, stream),
-                                'nbk_paint_gather_f64')
+                                pm.x_start, pm.nx_local, acc, pair_gs,
+                                stream), 'nbk_paint_gather_f64')
                     else:
                         paint_fn = (lib.nbk_paint_sorted_f64 if sorted_
                                     else lib.nbk_paint_f64)
@@ -606,8 +697,13 @@ class CatalogMesh(MeshSource):
             pos_t, mass_t = self._route(pos_t, mass_t)
 
         pos_soa, mass_t, sorted_, rowtab = _prepare_particles(
-            pos_t, mass_t, pm, force_rowtab=True)
+            pos_t, mass_t, pm, force_rowtab=True,
+            window=self.resampler, interlaced=interlaced)
         assert rowtab is not None
+        if isinstance(rowtab, tuple):
+            table, pair_gs, n_eff = rowtab
+        else:
+            table, pair_gs, n_eff = rowtab, -1, len(pos_t)
         n_routed = len(pos_t)
 
         scale = 1.0 / float(numpy.prod(pm.Nmesh)) / nbar
@@ -622,10 +718,10 @@ class CatalogMesh(MeshSource):
             # FFT work, not paint work — the bench roofline reads this)
             with profiling.collect('paint', n_routed):
                 hiplib.check(lib.nbk_paint_gather_fft_f64(
-                    hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n_routed,
+                    hiplib.dptr(pos_soa), hiplib.dptr(mass_t), n_eff,
                     nmesh, box, window_id, float(shift),
-                    hiplib.dptr(rowtab), hiplib.dptr(z), pm.x_start,
-                    pm.nx_local, scale, stream),
+                    hiplib.dptr(table), hiplib.dptr(z), pm.x_start,
+                    pm.nx_local, scale, pair_gs, stream),
                     'nbk_paint_gather_fft_f64')
             if finish:
                 return _r2c_finish(z, pm, stream)

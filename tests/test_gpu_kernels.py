@@ -483,6 +483,10 @@ def test_gather_paint_slab_partition():
                      sort_two_level_min_cells=1):
         soa, mass, sorted_, rowtab = _prepare_particles(pos_t, None, pm)
     assert rowtab is not None
+    if isinstance(rowtab, tuple):
+        table, pair_gs, n_eff = rowtab
+    else:
+        table, pair_gs, n_eff = rowtab, -1, n
 
     lib = hiplib.require()
     nmesh = hiplib.i64_arr(pm.Nmesh)
@@ -490,8 +494,8 @@ def test_gather_paint_slab_partition():
 
     full = torch.zeros((N, N, N), dtype=torch.float64, device='cuda')
     hiplib.check(lib.nbk_paint_gather_f64(
-        hiplib.dptr(soa), None, n, nmesh, box, 0, 0.0,
-        hiplib.dptr(rowtab), hiplib.dptr(full), 0, N, 0,
+        hiplib.dptr(soa), None, n_eff, nmesh, box, 0, 0.0,
+        hiplib.dptr(table), hiplib.dptr(full), 0, N, 0, pair_gs,
         hiplib.cur_stream()), 'gather full')
 
     parts = []
@@ -499,9 +503,9 @@ def test_gather_paint_slab_partition():
         slab = torch.zeros((N // 2, N, N), dtype=torch.float64,
                            device='cuda')
         hiplib.check(lib.nbk_paint_gather_f64(
-            hiplib.dptr(soa), None, n, nmesh, box, 0, 0.0,
-            hiplib.dptr(rowtab), hiplib.dptr(slab), x0, N // 2, 0,
-            hiplib.cur_stream()), 'gather slab')
+            hiplib.dptr(soa), None, n_eff, nmesh, box, 0, 0.0,
+            hiplib.dptr(table), hiplib.dptr(slab), x0, N // 2, 0,
+            pair_gs, hiplib.cur_stream()), 'gather slab')
         parts.append(slab)
     stacked = torch.cat(parts, dim=0)
     torch.cuda.synchronize()
