@@ -142,7 +142,8 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
     if force_rowtab and n == 0:
         # empty rank (post-routing): an all-zero table means every
         # range is empty and the gather kernel paints zeros
-        gs0 = _pair_gs(pm) if _pair_enabled() else None
+        gs0 = _pair_gs(pm) if (_pair_enabled() and window == 'cic'
+                               and not interlaced) else None
         if gs0 is not None:
             btab = torch.zeros((n0 >> 1) * (n1 >> gs0) + 1,
                                dtype=torch.int32, device='cuda')
@@ -206,8 +207,15 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
     # touched groups — the per-row fine pass (the single biggest sort
     # cost) disappears, and the gather paint reads whole bucket ranges
     # with its deposit masks dropping the out-of-tile copies.
-    gs = _pair_gs(pm) if (big and _pair_enabled() and n2 <= 20480) \
-        else None
+    # Pair mode pays off when a tile's source planes span a single
+    # plane pair boundary (CIC, shift 0: planes [p-1, p] — bucket read
+    # by 3 tiles, L2-served).  Wider stencils (TSC/PCS, or CIC's
+    # interlaced half-cell shift) span 3+ planes: each bucket gets read
+    # by 5-6 tiles and the measured C3 step LOST 2.6 ms — those
+    # windows keep the two-level row-table sort.
+    pair_ok = (window == 'cic' and not interlaced)
+    gs = _pair_gs(pm) if (big and pair_ok and _pair_enabled()
+                          and n2 <= 20480) else None
     if gs is not None:
         dlo, dhi = _PAIR_GHOST[(window, bool(interlaced))]
         if dhi - dlo >= (1 << gs):
