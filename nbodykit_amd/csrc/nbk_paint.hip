@@ -395,18 +395,8 @@ __global__ void kpaint_gather(const double* __restrict__ px,
     // per-particle read factor drops from (1 + span) planes to
     // (P + span)/P — the host picks P ~ RG to balance the two halo
     // factors within the 160 KiB LDS budget
-    // pair mode (gs >= 0) dispatches GROUP-major: the 2-3 tiles that
-    // scan the same (pair, group) bucket become adjacent in launch
-    // order, so the bucket's particle reads hit L2 instead of going to
-    // DRAM three times.  Row-table mode keeps plane-major order (its
-    // tiles share source ROWS with their row-group neighbours).
-    const int64_t nplanes = (int64_t)gridDim.x / tiles_per_plane;
-    const int64_t px0 = gs >= 0
-        ? x0 + (blockIdx.x % nplanes) * P
-        : x0 + (blockIdx.x / tiles_per_plane) * P;
-    const int64_t r0 = gs >= 0
-        ? (blockIdx.x / nplanes) * RG
-        : (blockIdx.x % tiles_per_plane) * RG;
+    const int64_t px0 = x0 + (blockIdx.x / tiles_per_plane) * P;
+    const int64_t r0 = (blockIdx.x % tiles_per_plane) * RG;
     const int T = blockDim.x;
     const int t = threadIdx.x;
     const int64_t win = (int64_t)P * RG * sp;
