@@ -66,18 +66,41 @@ def pick_tile(nx_local, n1, n2, pad, span):
 
 def gather_bpp(resampler, nmesh, n_particles, interlaced=False,
                fused=True):
-    """Algorithmic bytes/particle of the ownership-gather paint: each
-    particle's 24 B position row is read once per (P-plane x RG-row)
-    tile whose stencil reaches it — re-read factor
-    ((P+span)/P)*((RG+span)/RG) — plus the single plain mesh/spectrum
-    write amortized per particle."""
+    """Algorithmic bytes/particle of the ownership-gather paint, under
+    the byte model of the sort mode that actually feeds it.
+
+    PAIR-BUCKET mode (CIC non-interlaced, the default): each (pair,
+    group) bucket — holding the pair's two planes of particles plus the
+    ~1/RG y-stencil duplicates — is scanned by the 3 tiles whose
+    stencils reach the pair, so every 24 B position row is read 3 x
+    (1 + span/RG) times, plus the single mesh/spectrum write.  (The
+    scheme trades these L2-friendly re-reads for the eliminated 56 GB
+    per-row fine sort; DESIGN.md has the minimum-traffic model
+    alongside.)
+
+    Row-table mode (TSC/PCS/interlaced): read once per (P-plane x
+    RG-row) tile whose stencil reaches it — re-read factor
+    ((P+span)/P)*((RG+span)/RG)."""
     sh = bool(interlaced)
     # xhi - xlo per window/shift (csrc/nbk_paint.hip source-span logic)
     span = {('cic', False): 1, ('cic', True): 2,
             ('tsc', False): 3, ('tsc', True): 2,
             ('pcs', False): 3, ('pcs', True): 4}[(resampler, sh)]
-    P, RG = pick_tile(nmesh, nmesh, nmesh, 4 if fused else 0, span)
-    reads = 24.0 * (P + span) / P * (RG + span) / RG
+    import os
+    pair = (resampler == 'cic' and not sh
+            and os.environ.get('NBK_SORT_PAIR', '1') != '0')
+    if pair:
+        from nbodykit_amd.source.mesh.catalog import _pair_gs
+        class _G:  # pm stand-in for the gs picker
+            Nmesh = [nmesh, nmesh, nmesh]
+        gs = _pair_gs(_G)
+        pair = gs is not None
+        if pair:
+            RG = 1 << gs
+            reads = 24.0 * 3.0 * (1.0 + span / float(RG))
+    if not pair:
+        P, RG = pick_tile(nmesh, nmesh, nmesh, 4 if fused else 0, span)
+        reads = 24.0 * (P + span) / P * (RG + span) / RG
     mesh_bytes = 8.0 * nmesh ** 3 / float(n_particles)
     return reads + mesh_bytes
 
@@ -99,7 +122,9 @@ def paint_is_gather(nmesh, n_local):
 # 59.6 B/particle algorithmic model because the second source-plane
 # re-reads are served by L2, not HBM.  Refresh whenever the paint
 # kernel changes.
-PMC_TRAFFIC_BYTES = {('c4', True): 36.38e9}
+PMC_TRAFFIC_BYTES = {('c4', True): 48.5e9}  # r02 pair-sort build:
+# x2-corrected fetch 39.9 + 8.6 write per C4 fused-paint launch
+# (profiles/r02_pair_pmc.txt)
 
 WORKLOADS = {
     # BASELINE.json configs (C1 is the CPU-oracle plumbing config)
