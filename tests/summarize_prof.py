@@ -42,6 +42,8 @@ def summarize_pmc(files):
                 c = row.get('Counter_Name') or row.get('Counter-Name', '')
                 v = float(row.get('Counter_Value')
                           or row.get('Counter-Value') or 0)
+                k = k.replace('void ', '').replace(
+                    '(anonymous namespace)::', '')
                 acc[k.split('(')[0][:70]][c] += v
                 calls[k.split('(')[0][:70]][c] += 1
     for k in sorted(acc, key=lambda k: -max(acc[k].values())):
