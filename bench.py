@@ -122,9 +122,11 @@ def paint_is_gather(nmesh, n_local):
 # 59.6 B/particle algorithmic model because the second source-plane
 # re-reads are served by L2, not HBM.  Refresh whenever the paint
 # kernel changes.
-PMC_TRAFFIC_BYTES = {('c4', True): 48.5e9}  # r02 pair-sort build:
-# x2-corrected fetch 39.9 + 8.6 write per C4 fused-paint launch
-# (profiles/r02_pair_pmc.txt)
+PMC_TRAFFIC_BYTES = {('c4', True): 69.4e9}  # r02 pair-sort build:
+# x2-corrected FETCH 30.4*2 = 60.8 + WRITE 8.6 per C4 fused-paint
+# launch (profiles/r02_pair_pmc.txt; the corrected fetch sits BELOW the
+# 76.5 GB algorithmic 3-tile read model because L2 serves part of the
+# shared-bucket re-reads)
 
 WORKLOADS = {
     # BASELINE.json configs (C1 is the CPU-oracle plumbing config)
