@@ -319,7 +319,7 @@ def main():
     # default K large enough that the timed region dominates the GPU
     # activity of the run (r01 verdict item: sparse SMI sampling can
     # miss a too-short timed window)
-    ap.add_argument('--steps', type=int, default=25)
+    ap.add_argument("--steps", type=int, default=50)
     ap.add_argument('--warmup', type=int, default=2)
     ap.add_argument('--workload', default='c4',
                     choices=sorted(WORKLOADS))
