@@ -406,3 +406,38 @@ def test_catalog_view():
     import numpy
     numpy.testing.assert_array_equal(numpy.asarray(view['Position']),
                                      numpy.asarray(source['Position']))
+
+
+def test_xbin_lds_gate_budget():
+    """The Python LDS-budget gate for the deferred-x kernel must mirror
+    the kernel's own check (a mismatch means a failed launch instead of
+    a clean fallback)."""
+    from nbodykit_amd.algorithms.fftpower import _xbin_lds_fits
+    # C4 1d: 512 k-edges, Nmu=1 (2 mu edges) at n0=1024 fits
+    assert _xbin_lds_fits(1024, 512, 2, 1)
+    # 2d Nmu=5 at 1024^3 with ~512 edges does NOT (falls back)
+    assert not _xbin_lds_fits(1024, 512, 6, 1)
+    # interlaced pair needs two tiles + the phase table
+    assert _xbin_lds_fits(512, 256, 2, 1, il=True)
+    assert not _xbin_lds_fits(4096, 2048, 2, 1, il=True)
+
+
+def test_pair_sort_gate_geometry():
+    """Pair-bucket sort geometry picker: bucket count within the LDS
+    ceiling, group == tile rows, graceful None for unsupported
+    meshes."""
+    from nbodykit_amd.source.mesh.catalog import _pair_gs
+
+    class _PM:
+        def __init__(self, n):
+            self.Nmesh = [n, n, n]
+
+    assert _pair_gs(_PM(1024)) == 4          # 16-row groups, 32768 buckets
+    assert _pair_gs(_PM(512)) == 5           # 32-row groups
+    gs256 = _pair_gs(_PM(256))
+    assert gs256 is not None and (256 >> 1) * (256 >> gs256) <= 40960
+    assert _pair_gs(_PM(2048)) is None       # beyond the LDS ceiling
+
+    class _Odd:
+        Nmesh = [27, 27, 27]
+    assert _pair_gs(_Odd) is None            # odd n0 cannot pair
