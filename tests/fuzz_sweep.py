@@ -22,7 +22,7 @@ def run_case(seed):
     from oracle.fftpower import fftcorr_oracle
 
     rng = numpy.random.RandomState(5000 + seed)
-    nmesh = int(rng.choice([32, 48, 64, 96, 128, 160]))
+    nmesh = int(rng.choice([27, 32, 45, 48, 64, 96, 128, 160]))
     box = float(rng.uniform(100., 1000.))
     window = str(rng.choice(['cic', 'tsc', 'pcs']))
     interlaced = bool(rng.randint(2))
@@ -110,7 +110,7 @@ def run_case2(seed):
     from oracle import fftpower_oracle
 
     rng = numpy.random.RandomState(9000 + seed)
-    nmesh = int(rng.choice([32, 48, 64, 96, 128]))
+    nmesh = int(rng.choice([27, 32, 45, 48, 64, 96, 128]))
     box = float(rng.uniform(100., 1000.))
     window = str(rng.choice(['cic', 'tsc', 'pcs']))
     interlaced = bool(rng.randint(2))
@@ -183,7 +183,7 @@ def run_case3(seed):
     from oracle import fftrecon_oracle
 
     rng = numpy.random.RandomState(12000 + seed)
-    nmesh = int(rng.choice([16, 24, 32, 48]))
+    nmesh = int(rng.choice([16, 24, 27, 32, 45, 48]))
     box = float(rng.uniform(50., 300.))
     bias = float(rng.uniform(1.0, 2.5))
     f = float(rng.uniform(0.0, 0.9))
@@ -219,7 +219,7 @@ def run_case4(seed):
     from oracle.convpower import convpower_oracle
 
     rng = numpy.random.RandomState(15000 + seed)
-    nmesh = int(rng.choice([32, 48, 64]))
+    nmesh = int(rng.choice([27, 32, 48, 64]))
     lo = rng.uniform(800., 1500., size=3)
     span = rng.uniform(150., 350., size=3)
     ndata = int(rng.randint(1500, 6000))
