@@ -525,12 +525,8 @@ __global__ void kxfft_bin(const double* __restrict__ data,
     }
 
     // CONTIGUOUS tile chunks per block (not grid-strided): adjacent
-    // tiles are adjacent (y, z) columns, whose elements land in the
-    // same k bins for many tiles in a row — the run state below
-    // (cbin + partial sums) carries ACROSS tiles, so the contended
-    // LDS histogram atomics fire on real bin transitions only
-    // (grid-strided mapping re-flushed every tile: ~8K atomics/tile
-    // dominated the bin phase)
+    // tiles are adjacent (y, z) columns — L2/TLB locality for the
+    // pipelined strided loads
     const int64_t chunk = (A.tiles + gridDim.x - 1) / gridDim.x;
     const int64_t tend = ((int64_t)(blockIdx.x + 1) * chunk < A.tiles)
         ? (int64_t)(blockIdx.x + 1) * chunk : A.tiles;
