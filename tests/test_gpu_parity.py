@@ -1313,3 +1313,36 @@ def test_xbin_interlaced_matches_oracle():
                            resampler='tsc', compensated=True,
                            interlaced=True)
     check_parity(r, want)
+
+
+def test_diagonal_los_parity():
+    """Arbitrary (non-axis) unit line-of-sight: mu = k.los/|k| with a
+    diagonal los — covers the general dot-product path in the fused
+    binning kernels vs the oracle (the reference accepts any unit
+    vector, fftpower.py:177-182)."""
+    s3 = 1.0 / numpy.sqrt(3.0)
+    los = [s3, s3, s3]
+    cat = UniformCatalog(nbar=3e-3, BoxSize=100., seed=23)
+    r = FFTPower(cat, mode='2d', Nmu=4, Nmesh=64, los=los,
+                 poles=[0, 2])
+    pos = uniform_positions(3e-3, 100., 23)
+    want = fftpower_oracle(pos, Nmesh=64, BoxSize=100., mode='2d',
+                           Nmu=4, poles=[0, 2], los=los,
+                           resampler='cic', compensated=True)
+    check_parity(r, want, poles=[0, 2])
+
+
+def test_poles_without_monopole():
+    """poles=[2, 4] (no explicit 0): the internal ell list prepends 0
+    for the normalization but only the requested poles are reported
+    (fftpower.py:616-620 semantics)."""
+    cat = UniformCatalog(nbar=3e-3, BoxSize=100., seed=24)
+    r = FFTPower(cat, mode='1d', Nmesh=64, poles=[2, 4])
+    assert 'power_2' in r.poles.variables
+    assert 'power_4' in r.poles.variables
+    assert 'power_0' not in r.poles.variables
+    pos = uniform_positions(3e-3, 100., 24)
+    want = fftpower_oracle(pos, Nmesh=64, BoxSize=100., mode='1d',
+                           poles=[2, 4], resampler='cic',
+                           compensated=True)
+    check_parity(r, want, poles=[2, 4])
