@@ -91,9 +91,7 @@ def gather_bpp(resampler, nmesh, n_particles, interlaced=False,
             and os.environ.get('NBK_SORT_PAIR', '1') != '0')
     if pair:
         from nbodykit_amd.source.mesh.catalog import _pair_gs
-        class _G:  # pm stand-in for the gs picker
-            Nmesh = [nmesh, nmesh, nmesh]
-        gs = _pair_gs(_G)
+        gs = _pair_gs([nmesh, nmesh, nmesh])
         pair = gs is not None
         if pair:
             RG = 1 << gs

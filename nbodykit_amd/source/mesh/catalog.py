@@ -89,11 +89,12 @@ _PAIR_GHOST = {
 }
 
 
-def _pair_gs(pm):
+def _pair_gs(nmesh):
     """The y-group shift of the PAIR-BUCKET duplicating sort, or None
     when the geometry does not admit it (tile = 1 plane x 1<<gs rows x
-    (n2+4) doubles in LDS; (n0/2)*(n1>>gs) buckets <= 40960)."""
-    n0, n1, n2 = (int(x) for x in pm.Nmesh)
+    (n2+4) doubles in LDS; (n0/2)*(n1>>gs) buckets <= 40960).
+    ``nmesh`` is any 3-sequence (pm.Nmesh or a plain triple)."""
+    n0, n1, n2 = (int(x) for x in nmesh)
     if n0 % 2:
         return None
     best = None
@@ -121,8 +122,10 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
     thresholds and always emit the row table (the fused paint+z-FFT path
     has committed collectively and must not fall back per-rank: ws>1
     ranks can end up with arbitrarily few local particles after
-    routing).  The caller must have checked ``_two_level_ys(pm)`` is not
-    None and n2 <= 20480.
+    routing).  The caller must have checked ``_two_level_ys(pm)`` is
+    not None and n2 <= 20480; whether the pair-bucket or the two-level
+    row-table pipeline serves the request is decided here from
+    (pm, window, interlaced) alone, so the choice is rank-invariant.
 
     The deposit kernel's wave-merge and its L2 locality both depend on
     nearby-in-space particles being nearby-in-memory; a cell-ordered
@@ -142,7 +145,7 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
     if force_rowtab and n == 0:
         # empty rank (post-routing): an all-zero table means every
         # range is empty and the gather kernel paints zeros
-        gs0 = _pair_gs(pm) if (_pair_enabled() and window == 'cic'
+        gs0 = _pair_gs(pm.Nmesh) if (_pair_enabled() and window == 'cic'
                                and not interlaced) else None
         if gs0 is not None:
             btab = torch.zeros((n0 >> 1) * (n1 >> gs0) + 1,
@@ -214,7 +217,7 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
     # by 5-6 tiles and the measured C3 step LOST 2.6 ms — those
     # windows keep the two-level row-table sort.
     pair_ok = (window == 'cic' and not interlaced)
-    gs = _pair_gs(pm) if (big and pair_ok and _pair_enabled()
+    gs = _pair_gs(pm.Nmesh) if (big and pair_ok and _pair_enabled()
                           and n2 <= 20480) else None
     if gs is not None:
         dlo, dhi = _PAIR_GHOST[(window, bool(interlaced))]

@@ -428,16 +428,9 @@ def test_pair_sort_gate_geometry():
     meshes."""
     from nbodykit_amd.source.mesh.catalog import _pair_gs
 
-    class _PM:
-        def __init__(self, n):
-            self.Nmesh = [n, n, n]
-
-    assert _pair_gs(_PM(1024)) == 4          # 16-row groups, 32768 buckets
-    assert _pair_gs(_PM(512)) == 5           # 32-row groups
-    gs256 = _pair_gs(_PM(256))
+    assert _pair_gs([1024] * 3) == 4         # 16-row groups, 32768 buckets
+    assert _pair_gs([512] * 3) == 5          # 32-row groups
+    gs256 = _pair_gs([256] * 3)
     assert gs256 is not None and (256 >> 1) * (256 >> gs256) <= 40960
-    assert _pair_gs(_PM(2048)) is None       # beyond the LDS ceiling
-
-    class _Odd:
-        Nmesh = [27, 27, 27]
-    assert _pair_gs(_Odd) is None            # odd n0 cannot pair
+    assert _pair_gs([2048] * 3) is None      # beyond the LDS ceiling
+    assert _pair_gs([27] * 3) is None        # odd n0 cannot pair
