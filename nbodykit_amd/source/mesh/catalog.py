@@ -57,6 +57,18 @@ def _pair_enabled():
     return os.environ.get('NBK_SORT_PAIR', '1') != '0'
 
 
+def _sort_chunk():
+    """Per-block chunk of the counting sorts (NBK_SORT_CHUNK overrides;
+    exists so tests can prove the sorted output's consumers are
+    invariant to the chunking)."""
+    import os
+    try:
+        v = int(os.environ.get('NBK_SORT_CHUNK', '262144'))
+    except ValueError:
+        v = 262144
+    return max(4096, min(v, 1 << 22))
+
+
 def _two_level_ys(pm):
     """The y-group shift of the two-level locality sort, or None when no
     LDS budget fits this mesh (coarse histogram n0*(n1>>ys) ints and fine
@@ -225,7 +237,7 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
             gs = None
     if gs is not None:
         nbuck = (n0 >> 1) * (n1 >> gs)
-        CH = 262144
+        CH = _sort_chunk()
         nblocks = (n + CH - 1) // CH
         mat = torch.empty(nblocks * nbuck, dtype=torch.int32,
                           device='cuda')
@@ -258,7 +270,7 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
     if use_two:
         ys = ys_fine
         nbuck = n0 * (n1 >> ys)
-        CH = 262144
+        CH = _sort_chunk()
         nblocks = (n + CH - 1) // CH
         mat = torch.empty(nblocks * nbuck, dtype=torch.int32,
                           device='cuda')

@@ -1346,3 +1346,26 @@ def test_poles_without_monopole():
                            poles=[2, 4], resampler='cic',
                            compensated=True)
     check_parity(r, want, poles=[2, 4])
+
+
+def test_sort_chunk_invariance(monkeypatch):
+    """The counting sorts process particles in fixed-size chunks
+    (count-matrix rows); P(k) must be invariant to the chunk size —
+    pins the atomic-free placement across chunk boundaries for BOTH
+    the pair-bucket and the two-level pipelines."""
+    from nbodykit_amd import set_options
+    cat = UniformCatalog(nbar=0.7, BoxSize=64., seed=31)   # ~1.8e5 pts
+    with set_options(sort_min_n=1024, sort_two_level_min_n=1024,
+                     sort_two_level_min_cells=1):
+        r_def = FFTPower(cat, mode='1d', Nmesh=64)
+        monkeypatch.setenv('NBK_SORT_CHUNK', '10000')
+        r_small = FFTPower(cat, mode='1d', Nmesh=64)
+        monkeypatch.setenv('NBK_SORT_PAIR', '0')
+        r_small_rows = FFTPower(cat, mode='1d', Nmesh=64)
+        monkeypatch.setenv('NBK_SORT_CHUNK', '262144')
+        r_def_rows = FFTPower(cat, mode='1d', Nmesh=64)
+    scale = numpy.nanmax(numpy.abs(r_def.power['power']))
+    for r in (r_small, r_small_rows, r_def_rows):
+        assert_array_equal(r.power['modes'], r_def.power['modes'])
+        assert_allclose(r.power['power'], r_def.power['power'],
+                        rtol=1e-11, atol=1e-12 * scale, equal_nan=True)
