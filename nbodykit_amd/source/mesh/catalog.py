@@ -57,16 +57,22 @@ def _pair_enabled():
     return os.environ.get('NBK_SORT_PAIR', '1') != '0'
 
 
-def _sort_chunk():
-    """Per-block chunk of the counting sorts (NBK_SORT_CHUNK overrides;
-    exists so tests can prove the sorted output's consumers are
-    invariant to the chunking)."""
+def _sort_chunk(n):
+    """Per-block chunk of the counting sorts: sized for ~512 blocks
+    (the count matrix shrinks and per-block bandwidth improves with
+    bigger chunks — measured 55.8 -> 53.5 ms/step at C4 going
+    256K -> 2M — while the grid must still fill 256 CUs).
+    NBK_SORT_CHUNK overrides; also lets tests prove the consumers are
+    invariant to the chunking."""
     import os
-    try:
-        v = int(os.environ.get('NBK_SORT_CHUNK', '262144'))
-    except ValueError:
-        v = 262144
-    return max(4096, min(v, 1 << 22))
+    e = os.environ.get('NBK_SORT_CHUNK')
+    if e:
+        try:
+            return max(4096, min(int(e), 1 << 22))
+        except ValueError:
+            pass
+    target = -(-int(n) // 512)          # ceil(n / 512 blocks)
+    return max(262144, min(target, 1 << 21))
 
 
 def _two_level_ys(pm):
@@ -237,7 +243,7 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
             gs = None
     if gs is not None:
         nbuck = (n0 >> 1) * (n1 >> gs)
-        CH = _sort_chunk()
+        CH = _sort_chunk(n)
         nblocks = (n + CH - 1) // CH
         mat = torch.empty(nblocks * nbuck, dtype=torch.int32,
                           device='cuda')
@@ -270,7 +276,7 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
     if use_two:
         ys = ys_fine
         nbuck = n0 * (n1 >> ys)
-        CH = _sort_chunk()
+        CH = _sort_chunk(n)
         nblocks = (n + CH - 1) // CH
         mat = torch.empty(nblocks * nbuck, dtype=torch.int32,
                           device='cuda')
