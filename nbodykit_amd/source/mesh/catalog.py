@@ -72,7 +72,7 @@ def _sort_chunk(n):
         except ValueError:
             pass
     target = -(-int(n) // 512)          # ceil(n / 512 blocks)
-    return max(262144, min(target, 1 << 21))
+    return max(32768, min(target, 1 << 21))
 
 
 def _two_level_ys(pm):
