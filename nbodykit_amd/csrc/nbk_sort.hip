@@ -187,14 +187,38 @@ __global__ void kxsort_scatter(const double* __restrict__ pos,
 // read+write for the bases: ~1.5 passes over mat vs the ~8 of the
 // transpose+cumsum+sub+transpose torch chain it replaces.
 
-__global__ void kscan_colsum(const int* __restrict__ mat, int64_t nblocks,
-                             int64_t nbuck, int* __restrict__ colsum)
+// segmented column sums: with only nbuck (~32K) threads the serial
+// per-bucket chunk walks ran at 2 waves/CU and cost ~3 ms/step at C4;
+// splitting each column into NSEG segments gives NSEG x the threads
+// (full occupancy) at ~1.5 extra matrix passes of traffic.
+#define NBK_SCAN_SEG 8
+
+__global__ void kscan_partial(const int* __restrict__ mat,
+                              int64_t nblocks, int64_t nbuck,
+                              int64_t cs /* chunks per segment */,
+                              int* __restrict__ partial /* NSEG*nbuck */)
+{
+    const int64_t flat = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+    if (flat >= nbuck * NBK_SCAN_SEG) return;
+    const int64_t s = flat / nbuck;
+    const int64_t b = flat - s * nbuck;
+    const int64_t c0 = s * cs;
+    const int64_t c1 = (c0 + cs < nblocks) ? c0 + cs : nblocks;
+    int acc = 0;
+    for (int64_t c = c0; c < c1; c++)
+        acc += mat[c * nbuck + b];
+    partial[s * nbuck + b] = acc;
+}
+
+__global__ void kscan_fold(const int* __restrict__ partial,
+                           int64_t nbuck, int* __restrict__ colsum)
 {
     const int64_t b = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
     if (b >= nbuck) return;
     int s = 0;
-    for (int64_t c = 0; c < nblocks; c++)
-        s += mat[c * nbuck + b];
+    #pragma unroll
+    for (int j = 0; j < NBK_SCAN_SEG; j++)
+        s += partial[(int64_t)j * nbuck + b];
     colsum[b] = s;
 }
 
@@ -246,14 +270,21 @@ __global__ void kscan_exclusive(int* __restrict__ colsum, int64_t nbuck,
 }
 
 __global__ void kscan_bases(const int* __restrict__ mat, int64_t nblocks,
-                            int64_t nbuck,
+                            int64_t nbuck, int64_t cs,
+                            const int* __restrict__ partial,
                             const int* __restrict__ excl,
                             int* __restrict__ bases)
 {
-    const int64_t b = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-    if (b >= nbuck) return;
+    const int64_t flat = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+    if (flat >= nbuck * NBK_SCAN_SEG) return;
+    const int64_t s = flat / nbuck;
+    const int64_t b = flat - s * nbuck;
     int run = excl[b];
-    for (int64_t c = 0; c < nblocks; c++) {
+    for (int64_t j = 0; j < s; j++)
+        run += partial[j * nbuck + b];
+    const int64_t c0 = s * cs;
+    const int64_t c1 = (c0 + cs < nblocks) ? c0 + cs : nblocks;
+    for (int64_t c = c0; c < c1; c++) {
         bases[c * nbuck + b] = run;
         run += mat[c * nbuck + b];
     }
@@ -629,6 +660,8 @@ extern "C" int nbk_xsort_scatter_f64(const double* pos_aos,
     return NBK_OK;
 }
 
+/* colsum_tmp must hold (NBK_SCAN_SEG + 1) * nbuck ints: segment
+ * partials first, the column-sum/exclusive-base vector last. */
 extern "C" int nbk_scan_matrix_i32(const int* mat, int64_t nblocks,
                                    int64_t nbuck, int* colsum_tmp,
                                    int* bases, int* bucket_bases,
@@ -637,13 +670,19 @@ extern "C" int nbk_scan_matrix_i32(const int* mat, int64_t nblocks,
     if (nblocks == 0 || nbuck == 0) return NBK_OK;
     hipStream_t s = (hipStream_t)stream;
     const int T = 256;
+    int* partial = colsum_tmp;
+    int* colsum = colsum_tmp + NBK_SCAN_SEG * nbuck;
+    const int64_t cs = (nblocks + NBK_SCAN_SEG - 1) / NBK_SCAN_SEG;
+    const uint32_t gs_ = (uint32_t)((nbuck * NBK_SCAN_SEG + T - 1) / T);
     const uint32_t g = (uint32_t)((nbuck + T - 1) / T);
-    hipLaunchKernelGGL(kscan_colsum, dim3(g), dim3(T), 0, s,
-                       mat, nblocks, nbuck, colsum_tmp);
+    hipLaunchKernelGGL(kscan_partial, dim3(gs_), dim3(T), 0, s,
+                       mat, nblocks, nbuck, cs, partial);
+    hipLaunchKernelGGL(kscan_fold, dim3(g), dim3(T), 0, s,
+                       partial, nbuck, colsum);
     hipLaunchKernelGGL(kscan_exclusive, dim3(1), dim3(1024), 0, s,
-                       colsum_tmp, nbuck, bucket_bases);
-    hipLaunchKernelGGL(kscan_bases, dim3(g), dim3(T), 0, s,
-                       mat, nblocks, nbuck, colsum_tmp, bases);
+                       colsum, nbuck, bucket_bases);
+    hipLaunchKernelGGL(kscan_bases, dim3(gs_), dim3(T), 0, s,
+                       mat, nblocks, nbuck, cs, partial, colsum, bases);
     NBK_CHECK_HIP(hipGetLastError());
     return NBK_OK;
 }

@@ -250,7 +250,8 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
         hiplib.check(lib.nbk_psort_count_f64(
             hiplib.dptr(pos_in), n, CH, nmesh, box, gs, dlo, dhi,
             hiplib.dptr(mat), stream), 'nbk_psort_count_f64')
-        colsum = torch.empty(nbuck, dtype=torch.int32, device='cuda')
+        colsum = torch.empty(9 * nbuck, dtype=torch.int32,
+                             device='cuda')   # 8 segment partials + sums
         bases = torch.empty(nblocks * nbuck, dtype=torch.int32,
                             device='cuda')
         bucket_bases = torch.empty(nbuck + 1, dtype=torch.int32,
@@ -291,7 +292,8 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
             return pos_t.t().contiguous(), mass_t, True, None
         # one fused device scan of the count matrix (replaces a ~4 GB
         # torch transpose/cumsum/sub chain with ~1.5 coalesced passes)
-        colsum = torch.empty(nbuck, dtype=torch.int32, device='cuda')
+        colsum = torch.empty(9 * nbuck, dtype=torch.int32,
+                             device='cuda')   # 8 segment partials + sums
         bases = torch.empty(nblocks * nbuck, dtype=torch.int32,
                             device='cuda')
         bucket_bases = torch.empty(nbuck + 1, dtype=torch.int32,
