@@ -57,13 +57,14 @@ def _pair_enabled():
     return os.environ.get('NBK_SORT_PAIR', '1') != '0'
 
 
-def _sort_chunk(n):
+def _sort_chunk(n, nbuck):
     """Per-block chunk of the counting sorts: sized for ~512 blocks
     (the count matrix shrinks and per-block bandwidth improves with
     bigger chunks — measured 55.8 -> 53.5 ms/step at C4 going
-    256K -> 2M — while the grid must still fill 256 CUs).
-    NBK_SORT_CHUNK overrides; also lets tests prove the consumers are
-    invariant to the chunking."""
+    256K -> 2M — while the grid must still fill 256 CUs), floored so
+    the per-block LDS histogram zero/flush (nbuck ints) stays a small
+    fraction of the counting work.  NBK_SORT_CHUNK overrides; also
+    lets tests prove the consumers are invariant to the chunking."""
     import os
     e = os.environ.get('NBK_SORT_CHUNK')
     if e:
@@ -72,7 +73,7 @@ def _sort_chunk(n):
         except ValueError:
             pass
     target = -(-int(n) // 512)          # ceil(n / 512 blocks)
-    return max(32768, min(target, 1 << 21))
+    return max(16384, 4 * int(nbuck), min(target, 1 << 21))
 
 
 def _two_level_ys(pm):
@@ -243,7 +244,7 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
             gs = None
     if gs is not None:
         nbuck = (n0 >> 1) * (n1 >> gs)
-        CH = _sort_chunk(n)
+        CH = _sort_chunk(n, nbuck)
         nblocks = (n + CH - 1) // CH
         mat = torch.empty(nblocks * nbuck, dtype=torch.int32,
                           device='cuda')
@@ -277,7 +278,7 @@ def _prepare_particles(pos_t, mass_t, pm, force_rowtab=False,
     if use_two:
         ys = ys_fine
         nbuck = n0 * (n1 >> ys)
-        CH = _sort_chunk(n)
+        CH = _sort_chunk(n, nbuck)
         nblocks = (n + CH - 1) // CH
         mat = torch.empty(nblocks * nbuck, dtype=torch.int32,
                           device='cuda')
