@@ -376,13 +376,16 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                               int accumulate,
                               const cdouble* __restrict__ table /* W_n2 */,
                               double scale,
-                              int gs /* >= 0: rowtab is the PAIR-BUCKET
+                              int gs, /* >= 0: rowtab is the PAIR-BUCKET
                                   table [(n0/2)*(n1>>gs)+1] of the
                                   duplicating sort (1<<gs == RG); the
                                   tile reads whole (pair, group) ranges
                                   and the deposit masks drop the
                                   out-of-tile copies.  -1: per-row
-                                  table (nbk_bucket_fine_f64) */)
+                                  table (nbk_bucket_fine_f64) */
+                              int phases /* perf decomposition only
+                                  (NBK_PAINT_PHASES): bit1 deposits,
+                                  bit2 FFT+flush; 3 = real kernel */)
 {
     constexpr int SUP = (WINDOW == NBK_WINDOW_CIC) ? 2
                       : (WINDOW == NBK_WINDOW_TSC) ? 3 : 4;
@@ -446,6 +449,12 @@ __global__ void kpaint_gather(const double* __restrict__ px,
                 const double u1 = py[i] * invH1 + shift;
                 const double u2 = pz[i] * invH2 + shift;
                 const double m = mass ? mass[i] : 1.0;
+                if (!(phases & 1)) {
+                    // reads-only decomposition mode: keep the loads
+                    // observable, skip the deposit work
+                    if (u0 + u1 + u2 + m == 1e308) tile[0] += 1.0;
+                    continue;
+                }
                 double w0[SUP], w1[SUP], w2[SUP];
                 int64_t b0, b1, b2;
                 paint_weights<WINDOW, SUP>(u0, u1, u2, w0, w1, w2,
@@ -480,6 +489,8 @@ __global__ void kpaint_gather(const double* __restrict__ px,
         }
     }
     __syncthreads();
+
+    if (!(phases & 2)) return;       // decomposition mode: no flush/FFT
 
     if (!DOFFT) {
         // flush the exclusively-owned tile with plain stores
@@ -853,6 +864,18 @@ extern "C" int nbk_readout_f64(const double* pos, int64_t n,
 // particle re-read factor (P+sx)/P * (RG+sy)/RG (sx/sy = source span
 // beyond the tile from the window support and interlace shift) within
 // the 160 KiB LDS budget; powers of two dividing the local dims
+// NBK_PAINT_PHASES: perf decomposition only (1 = reads+deposits,
+// 0 = reads only, 2 = flush/FFT on an empty tile); default 3
+static int nbk_paint_phases(void) {
+    static int v = -1;
+    if (v < 0) {
+        const char* e = getenv("NBK_PAINT_PHASES");
+        v = e ? atoi(e) : 3;
+        if (v < 0 || v > 3) v = 3;
+    }
+    return v;
+}
+
 static void nbk_pick_tile(int64_t nx_local, int64_t n1, int64_t n2,
                           int64_t pad, int sx, int sy,
                           int* P_out, int* RG_out)
@@ -965,7 +988,7 @@ extern "C" int nbk_paint_gather_f64(const double* pos, const double* mass,
                            n0 / box[0], n1 / box[1], n2 / box[2], shift, \
                            rowtab, mesh, x0, nx_local, RG, P, xlo, xhi, \
                            accumulate, (const cdouble*)nullptr, 1.0, \
-                           pair_gs)
+                           pair_gs, nbk_paint_phases())
     if (P == 1) {
         if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GATHER(NBK_WINDOW_CIC, 1);
         else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GATHER(NBK_WINDOW_TSC, 1);
@@ -1061,7 +1084,8 @@ extern "C" int nbk_paint_gather_fft_f64(const double* pos,
                            mass, n, n0, n1, n2, \
                            n0 / box[0], n1 / box[1], n2 / box[2], shift, \
                            rowtab, zspec, x0, nx_local, RG, P, xlo, \
-                           xhi, 0, (const cdouble*)table, scale, pair_gs)
+                           xhi, 0, (const cdouble*)table, scale, \
+                           pair_gs, nbk_paint_phases())
     if (P == 1) {
         if (window == NBK_WINDOW_CIC) NBK_LAUNCH_GFFT(NBK_WINDOW_CIC, 1);
         else if (window == NBK_WINDOW_TSC) NBK_LAUNCH_GFFT(NBK_WINDOW_TSC, 1);
