@@ -1149,6 +1149,9 @@ def test_random_fkp_fuzz(seed):
 
 def _trace_xbin(monkeypatch):
     from nbodykit_amd.algorithms import fftpower as fmod
+    # these tests A/B the deferred path against NBK_NO_XBIN themselves —
+    # an externally preset value must not leak in
+    monkeypatch.delenv('NBK_NO_XBIN', raising=False)
     calls = []
     orig = fmod._project_power_xbin
 
