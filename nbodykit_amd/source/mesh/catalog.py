@@ -847,18 +847,36 @@ def route_particles(pos_t, mass_t, pm, dmin, dmax):
     n0 = int(pm.Nmesh[0])
     nx_l = pm.nx_local
 
-    idx_list = []
-    rank_list = []
-    for d in range(dmin, dmax + 1):
-        cell = torch.remainder(fu + d, n0)
-        rank_list.append(torch.div(cell, nx_l, rounding_mode='floor'))
-        idx_list.append(torch.arange(len(pos_t), device=pos_t.device))
-    ranks = torch.cat(rank_list)
-    idxs = torch.cat(idx_list)
-    # dedup (particle, rank) pairs
-    keys = torch.unique(idxs * ws + ranks)
-    idxs = torch.div(keys, ws, rounding_mode='floor')
-    ranks = keys - idxs * ws
+    if dmax - dmin < nx_l:
+        # The ghost window spans (dmax - dmin) cells < one slab, so a
+        # particle's destination set is ONE or TWO (wrap-adjacent)
+        # ranks: dedup analytically instead of torch.unique over all
+        # (particle, rank) pairs — the sort-based unique plus a second
+        # full argsort cost ~2 large device sorts per chunk per step,
+        # which would dominate the per-rank step at 8 GPUs.
+        r_lo = torch.div(torch.remainder(fu + dmin, n0), nx_l,
+                         rounding_mode='floor')
+        r_hi = torch.div(torch.remainder(fu + dmax, n0), nx_l,
+                         rounding_mode='floor')
+        two = r_hi != r_lo
+        ar = torch.arange(len(pos_t), device=pos_t.device)
+        idxs = torch.cat([ar, ar[two]])
+        ranks = torch.cat([r_lo, r_hi[two]])
+    else:
+        # degenerate slabs (nx_local smaller than the stencil): the
+        # generic per-offset route with pair dedup
+        idx_list = []
+        rank_list = []
+        for d in range(dmin, dmax + 1):
+            cell = torch.remainder(fu + d, n0)
+            rank_list.append(torch.div(cell, nx_l,
+                                       rounding_mode='floor'))
+            idx_list.append(torch.arange(len(pos_t),
+                                         device=pos_t.device))
+        keys = torch.unique(torch.cat(idx_list) * ws
+                            + torch.cat(rank_list))
+        idxs = torch.div(keys, ws, rounding_mode='floor')
+        ranks = keys - idxs * ws
 
     order = torch.argsort(ranks, stable=True)
     idxs = idxs[order]
