@@ -32,16 +32,18 @@ def _int_freqs(N):
 
 
 # ---- FFT pass dispatch -------------------------------------------------
-# Power-of-two lengths run the radix-4 LDS kernels directly.  Other EVEN
-# lengths (N <= 2048) run Bluestein's algorithm composed from the SAME
-# kernels: the chirp multiplies are torch elementwise ops and the
-# convolution transforms are nbk_fft_c_strided at M = next power of two
-# >= 2N-1 — no rocFFT/hipFFT anywhere.  This is the capability fallback
-# matching FFTW-backed pmesh's arbitrary-Nmesh support
-# (nbodykit/base/mesh.py:50); every BASELINE config is a power of two,
-# so the fallback is never on the benchmark path.  Odd lengths are
-# rejected at ParticleMesh construction (the Nyquist-as-negative
-# coordinate and Hermitian-weight conventions assume even dims).
+# Power-of-two lengths run the radix-4 LDS kernels directly.  Every
+# other length (N <= 2048, even OR odd) runs Bluestein's algorithm
+# composed from the SAME kernels: the chirp multiplies are torch
+# elementwise ops and the convolution transforms are nbk_fft_c_strided
+# at M = next power of two >= 2N-1 — no rocFFT/hipFFT anywhere.  This
+# is the capability fallback matching FFTW-backed pmesh's
+# arbitrary-Nmesh support (nbodykit/base/mesh.py:50); every BASELINE
+# config is a power of two, so the fallback is never on the benchmark
+# path.  Odd lengths carry parity-correct conventions throughout: no
+# Nyquist-as-negative coordinate, no self-conjugate z plane beyond DC
+# (freq helpers in csrc/nbk_common.h, coordinate builders here and in
+# the oracle).
 
 def _is_pow2_len(n):
     return 8 <= n <= 4096 and (n & (n - 1)) == 0
