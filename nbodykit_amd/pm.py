@@ -14,8 +14,10 @@ Layouts (DESIGN.md "Data layout in HBM"):
   (Nx, Ny_local, Nz/2+1) — full x, y-partitioned (pfft's transposed
   convention; the coordinates travel with the block so downstream
   consumers are layout-agnostic, like fftpower.py:570-605).
-- Coordinates follow the Nyquist-as-negative convention
-  (nbodykit/meshtools.py:150-153).
+- Coordinates follow the Nyquist-as-negative convention for even axis
+  lengths (nbodykit/meshtools.py:150-153); odd lengths have no Nyquist
+  plane and keep all stored frequencies positive (numpy fftfreq at any
+  parity).
 
 Compute requires the HIP extension and a GPU (hiplib.require()); field
 construction and metadata work anywhere so the API can be exercised in
