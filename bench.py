@@ -523,6 +523,9 @@ def main():
                     if prof.get('fft_strided', {}).get('ms') else None),
             },
             'cpu_baseline': cpu_baseline,
+            # diagnostic: HBM high-water mark on rank 0 (288 GB/GPU)
+            'peak_hbm_gb': round(
+                torch.cuda.max_memory_allocated() / 1e9, 2),
         }
         print(json.dumps(out), flush=True)
 
