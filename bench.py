@@ -514,9 +514,12 @@ def main():
                                         if paint['calls'] else None),
                 'algorithmic_B_per_particle': bpp,
                 'pure_paint': pure_paint,
-                # mesh-cells/s FFT'd, per axis pass, through the y+x
-                # strided passes (HIP events; units = Nmesh^3 per pass;
-                # the z pass runs fused inside the paint kernel)
+                # mesh-cells/s FFT'd through the STANDALONE strided
+                # pass (HIP events; units = Nmesh^3 per pass).  In the
+                # deferred path that is the y pass only: the z pass
+                # runs fused inside the paint kernel and the x pass
+                # inside the binning kernel (nbk_fft_x_bin_f64), so
+                # neither appears here.
                 'fft_cells_per_s': (
                     prof['fft_strided']['units']
                     / (prof['fft_strided']['ms'] * 1e-3)
