@@ -114,17 +114,12 @@ def paint_is_gather(nmesh, n_local):
 # (workload, gather?) -> measured HBM bytes per benched paint launch,
 # from rocprofv3 --pmc on THIS round's kernels: FETCH_SIZE doubled per
 # the gfx950 wide-read half-count + WRITE_SIZE from its own pass
-# (profiles/r02_pmc_summary.txt documents the collection; raw CSVs in
-# gpurun_out/pmc_final_*).  C4 fused paint: x2-corrected fetch
-# 27.77 + 8.61 write = 36.4 GB DRAM-side per launch — BELOW the
-# 59.6 B/particle algorithmic model because the second source-plane
-# re-reads are served by L2, not HBM.  Refresh whenever the paint
-# kernel changes.
-PMC_TRAFFIC_BYTES = {('c4', True): 69.4e9}  # r02 pair-sort build:
-# x2-corrected FETCH 30.4*2 = 60.8 + WRITE 8.6 per C4 fused-paint
-# launch (profiles/r02_pair_pmc.txt; the corrected fetch sits BELOW the
-# 76.5 GB algorithmic 3-tile read model because L2 serves part of the
-# shared-bucket re-reads)
+# (profiles/r02_pair_pmc.txt documents the collection).  C4 fused
+# paint, pair-sort build: x2-corrected FETCH 30.4*2 = 60.8 + WRITE 8.6
+# per launch.  The corrected fetch sits BELOW the 76.5 GB algorithmic
+# 3-tile read model because L2 serves part of the shared-bucket
+# re-reads.  Refresh whenever the paint kernel changes.
+PMC_TRAFFIC_BYTES = {('c4', True): 69.4e9}
 
 WORKLOADS = {
     # BASELINE.json configs (C1 is the CPU-oracle plumbing config)
