@@ -902,14 +902,22 @@ def paint_raw(pos_t, pm, resampler='cic'):
     if pm.comm.size > 1:
         dmin, dmax = _GHOST_RANGE[(resampler, False)]
         pos_t, _ = route_particles(pos_t, None, pm, dmin, dmax)
-    pos_soa, _, sorted_, rowtab = _prepare_particles(pos_t, None, pm)
+    # the window MUST reach the sort: the pair-bucket pipeline's
+    # duplication range is stencil-dependent (a CIC-range sort under a
+    # TSC paint would drop group-boundary deposits)
+    pos_soa, _, sorted_, rowtab = _prepare_particles(
+        pos_t, None, pm, window=resampler, interlaced=False)
     if rowtab is not None:
+        if isinstance(rowtab, tuple):
+            table, pair_gs, n_eff = rowtab
+        else:
+            table, pair_gs, n_eff = rowtab, -1, len(pos_t)
         hiplib.check(lib.nbk_paint_gather_f64(
-            hiplib.dptr(pos_soa), None, len(pos_t),
+            hiplib.dptr(pos_soa), None, n_eff,
             hiplib.i64_arr(pm.Nmesh), hiplib.f64_arr(pm.BoxSize),
-            hiplib.WINDOW_IDS[resampler], 0.0, hiplib.dptr(rowtab),
+            hiplib.WINDOW_IDS[resampler], 0.0, hiplib.dptr(table),
             hiplib.dptr(field.value), pm.x_start, pm.nx_local, 0,
-            hiplib.cur_stream()), 'nbk_paint_gather_f64')
+            pair_gs, hiplib.cur_stream()), 'nbk_paint_gather_f64')
         return field
     paint_fn = lib.nbk_paint_sorted_f64 if sorted_ else lib.nbk_paint_f64
     hiplib.check(paint_fn(

@@ -1372,3 +1372,31 @@ def test_sort_chunk_invariance(monkeypatch):
         assert_array_equal(r.power['modes'], r_def.power['modes'])
         assert_allclose(r.power['power'], r_def.power['power'],
                         rtol=1e-11, atol=1e-12 * scale, equal_nan=True)
+
+
+@pytest.mark.parametrize('resampler', ['cic', 'tsc', 'pcs'])
+def test_paint_raw_window_reaches_sort(resampler):
+    """paint_raw (the reconstruction driver's bare paint) must pass its
+    ACTUAL window to the particle sort: the pair-bucket pipeline's
+    duplication range is stencil-dependent, and a CIC-range sort under
+    a TSC/PCS paint would silently drop group-boundary deposits
+    (regression: the gather branch also carried a stale kernel call
+    signature only reachable above the sort thresholds)."""
+    import torch
+    from nbodykit_amd import set_options
+    from nbodykit_amd.source.mesh.catalog import paint_raw
+    from nbodykit_amd.pm import ParticleMesh
+    pm = ParticleMesh(BoxSize=64., Nmesh=64)
+    rng = numpy.random.RandomState(41)
+    n = 150000
+    pos_t = torch.as_tensor(rng.uniform(0, 64., size=(n, 3))).to('cuda')
+    # scatter-path reference (thresholds high: no sort path engages)
+    ref = paint_raw(pos_t, pm, resampler=resampler)
+    with set_options(sort_min_n=1024, sort_two_level_min_n=1024,
+                     sort_two_level_min_cells=1):
+        got = paint_raw(pos_t, pm, resampler=resampler)
+    a = got.value.cpu().numpy()
+    b = ref.value.cpu().numpy()
+    assert abs(a.sum() - n) < 1e-5 * n, 'mass not conserved'
+    assert_allclose(a, b, rtol=1e-11, atol=1e-11,
+                    err_msg='gather/pair paint_raw differs from scatter')
