@@ -23,10 +23,12 @@ def tcf(self, out=None):
 cmod.CatalogMesh.to_complex_field = tcf
 
 orig_prep = cmod._prepare_particles
-def prep(pos_t, mass_t, pm, force_rowtab=False):
-    r = orig_prep(pos_t, mass_t, pm, force_rowtab)
-    print('[trace] prepare n=%d force=%s rowtab=%s sorted=%s'
-          % (len(pos_t), force_rowtab, r[3] is not None, r[2]), flush=True)
+def prep(pos_t, mass_t, pm, force_rowtab=False, **kw):
+    r = orig_prep(pos_t, mass_t, pm, force_rowtab, **kw)
+    print('[trace] prepare n=%d force=%s table=%s sorted=%s'
+          % (len(pos_t), force_rowtab,
+             'pair' if isinstance(r[3], tuple) else r[3] is not None,
+             r[2]), flush=True)
     return r
 cmod._prepare_particles = prep
 
