@@ -1400,3 +1400,25 @@ def test_paint_raw_window_reaches_sort(resampler):
     assert abs(a.sum() - n) < 1e-5 * n, 'mass not conserved'
     assert_allclose(a, b, rtol=1e-11, atol=1e-11,
                     err_msg='gather/pair paint_raw differs from scatter')
+
+
+def test_weighted_pair_sort_paint():
+    """Weighted catalog through the pair-bucket sort (the mass array is
+    duplicated alongside the positions — otherwise untested above the
+    thresholds): P(k) must match the scatter-path reference."""
+    from nbodykit_amd import set_options
+    rng = numpy.random.RandomState(44)
+    n = 160000
+    cat = ArrayCatalog({'Position': rng.uniform(0, 64., size=(n, 3)),
+                        'Weight': rng.exponential(size=n) + 0.1})
+    kw = dict(mode='1d', Nmesh=64)
+    r_ref = FFTPower(cat, **kw)               # scatter path (small n)
+    with set_options(sort_min_n=1024, sort_two_level_min_n=1024,
+                     sort_two_level_min_cells=1):
+        r_pair = FFTPower(cat, **kw)          # pair sort + gather
+    scale = numpy.nanmax(numpy.abs(r_ref.power['power']))
+    assert_array_equal(r_pair.power['modes'], r_ref.power['modes'])
+    assert_allclose(r_pair.power['power'], r_ref.power['power'],
+                    rtol=1e-10, atol=1e-11 * scale, equal_nan=True)
+    assert_allclose(r_pair.attrs['shotnoise'], r_ref.attrs['shotnoise'],
+                    rtol=1e-12)
