@@ -1411,7 +1411,7 @@ def test_weighted_pair_sort_paint():
     n = 160000
     cat = ArrayCatalog({'Position': rng.uniform(0, 64., size=(n, 3)),
                         'Weight': rng.exponential(size=n) + 0.1})
-    kw = dict(mode='1d', Nmesh=64)
+    kw = dict(mode='1d', Nmesh=64, BoxSize=64.)
     r_ref = FFTPower(cat, **kw)               # scatter path (small n)
     with set_options(sort_min_n=1024, sort_two_level_min_n=1024,
                      sort_two_level_min_cells=1):
