@@ -100,10 +100,13 @@ def _two_level_ys(pm):
 _PAIR_GHOST = {
     # deposit rows relative to iy = floor(y/H): [Dlo, Dhi] covering the
     # stencil at BOTH interlacing shifts when interlaced (the one sorted
-    # array feeds both paints); derived from the launchers' (dmin, dmax)
-    # with Dhi = dmax + support - 1
+    # array feeds both paints); Dlo = dmin, Dhi = dmax + support - 1
+    # from the paint launchers' (dmin, dmax) source-span logic
+    # (csrc/nbk_paint.hip).  Only the CIC rows are exercised today (the
+    # pair sort is gated to CIC non-interlaced); the others are kept
+    # correct for any future widening of that gate.
     ('cic', False): (0, 1), ('cic', True): (0, 2),
-    ('tsc', False): (-1, 1), ('tsc', True): (-1, 2),
+    ('tsc', False): (-1, 2), ('tsc', True): (-1, 2),
     ('pcs', False): (-1, 2), ('pcs', True): (-1, 3),
 }
 
