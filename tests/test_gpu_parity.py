@@ -1424,15 +1424,19 @@ def test_weighted_pair_sort_paint():
                     rtol=1e-12)
 
 
-def test_noncubic_mesh_parity():
-    """Non-cubic Nmesh (per-axis dims differ): exercises every
-    per-axis code path — freq conventions, the sort geometry pickers,
-    tile divisions, per-axis compensation — vs the oracle."""
+@pytest.mark.parametrize('nmesh', [[64, 32, 128], [27, 32, 45],
+                                   [32, 45, 64]],
+                         ids=['pow2', 'mixed-odd', 'odd-mid'])
+def test_noncubic_mesh_parity(nmesh):
+    """Non-cubic Nmesh (per-axis dims differ, including MIXED odd/even
+    axes — each axis carries its own parity conventions): exercises
+    every per-axis code path — freq conventions, the sort geometry
+    pickers, tile divisions, per-axis compensation — vs the oracle."""
     nbar, box, seed = 2e-3, 64., 5
     cat = UniformCatalog(nbar=nbar, BoxSize=box, seed=seed)
-    r = FFTPower(cat, mode='1d', Nmesh=[64, 32, 128])
+    r = FFTPower(cat, mode='1d', Nmesh=nmesh)
     pos = uniform_positions(nbar, box, seed)
-    want = fftpower_oracle(pos, Nmesh=[64, 32, 128], BoxSize=box,
+    want = fftpower_oracle(pos, Nmesh=nmesh, BoxSize=box,
                            mode='1d', resampler='cic', compensated=True)
     check_parity(r, want)
 
