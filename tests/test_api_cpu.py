@@ -434,3 +434,15 @@ def test_pair_sort_gate_geometry():
     assert gs256 is not None and (256 >> 1) * (256 >> gs256) <= 40960
     assert _pair_gs([2048] * 3) is None      # beyond the LDS ceiling
     assert _pair_gs([27] * 3) is None        # odd n0 cannot pair
+
+
+def test_mismatched_boxsize_raises():
+    """Meshes with different BoxSize raise ValueError at FFTPower
+    construction (reference test_fftpower.py:123-135) — checked before
+    any compute, so it works without a GPU."""
+    import pytest as _pytest
+    from nbodykit_amd.lab import UniformCatalog, FFTPower
+    m1 = UniformCatalog(nbar=1e-4, BoxSize=128., seed=1).to_mesh(Nmesh=16)
+    m2 = UniformCatalog(nbar=1e-4, BoxSize=256., seed=1).to_mesh(Nmesh=16)
+    with _pytest.raises(ValueError):
+        FFTPower(m1, mode='1d', second=m2)

@@ -1454,3 +1454,22 @@ def test_anisotropic_box_parity():
                            mode='2d', Nmu=4, resampler='cic',
                            compensated=True)
     check_parity(r, want)
+
+
+def test_fftcorr_unique_edges():
+    """FFTCorr with dr=0: unique configuration-space separations as bin
+    edges (reference test_fftcorr.py:28-36)."""
+    from nbodykit_amd.lab import FFTCorr
+    from oracle import fftcorr_oracle
+    n = 20000
+    pos = numpy.random.RandomState(10).uniform(0, 100., size=(n, 3))
+    cat = ArrayCatalog({'Position': pos})
+    r = FFTCorr(cat, mode='1d', Nmesh=16, BoxSize=100., dr=0)
+    want = fftcorr_oracle(pos, Nmesh=16, BoxSize=100., mode='1d',
+                          resampler='cic', compensated=True, dr=0)
+    assert_array_equal(r.corr['modes'], want['modes'])
+    got = r.corr['corr']
+    ref = want['corr']
+    ok = numpy.isfinite(ref) & (numpy.abs(ref) > 1e-12)
+    rel = numpy.abs(got[ok] - ref[ok]) / numpy.abs(ref[ok])
+    assert rel.max() < 1e-9, 'dr=0 FFTCorr parity: %g' % rel.max()
