@@ -105,6 +105,32 @@ def kedges_unique(geom, kmax=None):
     return edges, fx
 
 
+def redges_unique(geom, rmax):
+    """dr=0 unique-separation edges for FFTCorr: the configuration-space
+    analogue of ``kedges_unique`` (reference fftcorr.py:96-99 calling
+    fftpower.py:732-769 with x = RealField coords and x0 = H)."""
+    x = real_coords(geom)
+    x0 = geom.BoxSize / geom.Nmesh
+
+    fx2 = sum(xi ** 2 for xi in x).ravel()
+
+    def unique_binned(values, binning):
+        ints = numpy.int64(values / binning + 0.5)
+        _, ind = numpy.unique(ints, return_index=True)
+        return values[ind]
+
+    fx = unique_binned(fx2, (x0.min() * 0.05) ** 2) ** 0.5
+    fx = fx[fx < rmax]
+    fx = unique_binned(fx, x0.min() * 1e-5)
+
+    width = numpy.diff(fx)
+    edges = fx.copy()
+    edges[1:] -= width * 0.5
+    edges = numpy.append(edges, [fx[-1] + width[-1] * 0.5])
+    edges[0] = 0
+    return edges, fx
+
+
 def project_to_basis(y3d, geom, edges, los=(0, 0, 1), poles=(),
                      coords=None, hermitian=True, _return_sums=False):
     """
@@ -375,7 +401,11 @@ def fftcorr_oracle(position, Nmesh, BoxSize, mode='1d',
         dr = float(geom.BoxSize.min() / geom.Nmesh.max())
     if rmax is None:
         rmax = 0.5 * float(geom.BoxSize.min()) + dr / 2
-    redges = numpy.arange(rmin, rmax, dr)
+    rcoords = None
+    if dr > 0:
+        redges = numpy.arange(rmin, rmax, dr)
+    else:
+        redges, rcoords = redges_unique(geom, rmax)
     muedges = numpy.linspace(0, 1, Nmu + 1, endpoint=True)
 
     result, pole_result = project_to_basis(
@@ -391,7 +421,7 @@ def fftcorr_oracle(position, Nmesh, BoxSize, mode='1d',
         'Nmesh': geom.Nmesh.copy(), 'BoxSize': geom.BoxSize.copy(),
     }
     out = {
-        'redges': redges, 'muedges': muedges,
+        'redges': redges, 'muedges': muedges, 'rcoords': rcoords,
         'r': numpy.squeeze(xmean_2d) if mode == '1d' else xmean_2d,
         'mu': None if mode == '1d' else mumean_2d,
         'corr': numpy.squeeze(y2d) if mode == '1d' else y2d,
