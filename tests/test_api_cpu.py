@@ -446,3 +446,33 @@ def test_mismatched_boxsize_raises():
     m2 = UniformCatalog(nbar=1e-4, BoxSize=256., seed=1).to_mesh(Nmesh=16)
     with _pytest.raises(ValueError):
         FFTPower(m1, mode='1d', second=m2)
+
+
+def test_redges_unique_matches_product():
+    """Oracle's dr=0 unique-separation edges agree bit-for-bit with the
+    product's host-side _find_unique_edges on the same coordinate grid
+    (reference fftcorr.py:96-99 -> fftpower.py:732-769)."""
+    import numpy
+    from oracle.fftpower import redges_unique
+    from oracle.mesh import MeshGeometry, real_coords
+    from nbodykit_amd.algorithms.fftpower import _find_unique_edges
+
+    class SerialComm:
+        size = 1
+        rank = 0
+
+        def allgather(self, x):
+            return [x]
+
+        def allreduce(self, x, op=None):
+            return x
+
+    for nmesh, box in [(16, 100.), (12, 64.), (17, 80.)]:
+        geom = MeshGeometry(nmesh, box, dtype='f8')
+        rmax = 0.5 * box
+        e1, c1 = redges_unique(geom, rmax)
+        e2, c2 = _find_unique_edges(real_coords(geom),
+                                    geom.BoxSize / geom.Nmesh,
+                                    rmax, SerialComm())
+        assert numpy.array_equal(e1, e2)
+        assert numpy.array_equal(c1, c2)
