@@ -346,3 +346,31 @@ def test_gloo_fft_scheme_world4():
 def test_gloo_particle_exchange_world4():
     res = _run_world('_body_exchange', world=4)
     assert all(res)
+
+
+@pytest.mark.timeout(300)
+def test_gloo_world3_transpose_and_exchange():
+    """Odd (non-power-of-two) world size: the pencil transpose and the
+    particle exchange are ws-generic — cover ws=3 explicitly."""
+    res = _run_world('_body_transpose', world=3)
+    assert all(fwd and bwd for fwd, bwd in res)
+    res = _run_world('_body_exchange', world=3)
+    assert all(res)
+
+
+def _body_divisibility_gate(comm):
+    from nbodykit_amd.pm import ParticleMesh
+    try:
+        ParticleMesh(BoxSize=100., Nmesh=16, comm=comm)
+    except ValueError as e:
+        return 'divisible' in str(e)
+    return False
+
+
+@pytest.mark.timeout(300)
+def test_gloo_world3_nmesh_gate():
+    """Nmesh not divisible by world size raises the documented
+    ValueError on every rank (pm.py slab-partition contract) instead of
+    silently mis-partitioning."""
+    res = _run_world('_body_divisibility_gate', world=3)
+    assert all(res)
