@@ -80,7 +80,7 @@ def main():
     nseeds = int(sys.argv[1]) if len(sys.argv) > 1 else 64
     variant = sys.argv[2] if len(sys.argv) > 2 else 'v1'
     seed0 = int(sys.argv[3]) if len(sys.argv) > 3 else 0
-    gen = {'v2': run_case2, 'v3': run_case3,
+    gen = {'v2': run_case2, 'v3': run_case3, 'v5': run_case5,
            'v4': run_case4}.get(variant, run_case)
     # v4 poles are stored as c8 (the reference's dtype) => float32-level
     thr = 5e-6 if variant == 'v4' else 1e-9
@@ -276,3 +276,76 @@ def run_case4(seed):
 
 if __name__ == '__main__':
     main()
+
+
+def run_case5(seed):
+    """FFTCorr fuzz: xi(r)/xi(r,mu)/xi_ell over random meshes, windows,
+    interlacing, los, dr (incl. dr=0 unique separations), rmax, cross —
+    vs fftcorr_oracle (reference algorithms/fftcorr.py:15-235)."""
+    from nbodykit_amd.lab import ArrayCatalog, FFTCorr
+    from oracle import fftcorr_oracle
+
+    rng = numpy.random.RandomState(21000 + seed)
+    nmesh = int(rng.choice([16, 24, 27, 32, 45, 48, 64]))
+    box = float(rng.uniform(80., 600.))
+    window = str(rng.choice(['cic', 'tsc', 'pcs']))
+    interlaced = bool(rng.randint(2))
+    compensated = bool(rng.randint(2))
+    mode = str(rng.choice(['1d', '2d']))
+    Nmu = int(rng.choice([3, 5]))
+    los = [[0, 0, 1], [0, 1, 0], [1, 0, 0]][rng.randint(3)]
+    poles = ([0, 2] if (los == [0, 0, 1] and rng.randint(2)) else [])
+    cross = bool(rng.randint(2))
+    dr0 = bool(rng.randint(4) == 0 and nmesh <= 48)
+    rmax = float(rng.uniform(0.25, 0.45)) * box if rng.randint(2) \
+        else None
+    n = int(rng.randint(15000, 60000))
+    pos = rng.uniform(0, box, size=(n, 3))
+    cfg = dict(nmesh=nmesh, box=round(box, 1), window=window,
+               interlaced=interlaced, compensated=compensated,
+               mode=mode, los=los, poles=poles, cross=cross, dr0=dr0,
+               rmax=None if rmax is None else round(rmax, 1))
+
+    def make(p):
+        c = ArrayCatalog({'Position': p})
+        return c.to_mesh(Nmesh=nmesh, BoxSize=box, dtype='f8',
+                         compensated=compensated, resampler=window,
+                         interlaced=interlaced)
+
+    second = None
+    pos2 = None
+    if cross:
+        pos2 = numpy.random.RandomState(21500 + seed).uniform(
+            0, box, size=(n // 2, 3))
+        second = make(pos2)
+    kw = dict(mode=mode, los=los, poles=poles)
+    if mode == '2d':
+        kw['Nmu'] = Nmu
+    if dr0:
+        kw['dr'] = 0
+    if rmax is not None:
+        kw['rmax'] = rmax
+    r = FFTCorr(make(pos), second=second, **kw)
+    want = fftcorr_oracle(pos, second_position=pos2, Nmesh=nmesh,
+                          BoxSize=box, resampler=window,
+                          compensated=compensated,
+                          interlaced=interlaced, Nmu=Nmu, poles=poles,
+                          los=los, mode=mode,
+                          **(dict(dr=0) if dr0 else {}),
+                          **(dict(rmax=rmax) if rmax is not None
+                             else {}))
+    got = numpy.nan_to_num(numpy.ravel(r.corr['corr']))
+    ref = numpy.nan_to_num(numpy.ravel(want['corr']))
+    modes_ok = numpy.array_equal(r.corr['modes'], want['modes'])
+    ok = numpy.isfinite(ref) & (numpy.abs(ref) > 1e-12)
+    rel = (numpy.abs(got[ok] - ref[ok])
+           / numpy.abs(ref[ok])).max() if ok.any() else 0.0
+    if poles and want['poles'] is not None:
+        for ell in poles:
+            g = numpy.nan_to_num(r.poles['corr_%d' % ell])
+            f = numpy.nan_to_num(want['poles'][ell])
+            ok = numpy.isfinite(f) & (numpy.abs(f) > 1e-12)
+            if ok.any():
+                rel = max(rel, float((numpy.abs(g[ok] - f[ok])
+                                      / numpy.abs(f[ok])).max()))
+    return float(rel), bool(modes_ok), cfg
