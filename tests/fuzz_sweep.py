@@ -274,8 +274,6 @@ def run_case4(seed):
     return rel, modes_ok, cfg
 
 
-if __name__ == '__main__':
-    main()
 
 
 def run_case5(seed):
@@ -349,3 +347,7 @@ def run_case5(seed):
                 rel = max(rel, float((numpy.abs(g[ok] - f[ok])
                                       / numpy.abs(f[ok])).max()))
     return float(rel), bool(modes_ok), cfg
+
+
+if __name__ == '__main__':
+    main()
