@@ -1141,6 +1141,19 @@ def test_random_fkp_fuzz(seed):
     assert rel < 5e-6, (rel, cfg)       # poles stored as c8
 
 
+@pytest.mark.timeout(900)
+@pytest.mark.parametrize('seed', [0, 1, 2, 5, 8])
+def test_random_corr_fuzz(seed):
+    """FFTCorr fuzz (tests/fuzz_sweep.run_case5): random mesh, window,
+    interlacing, los, poles, cross, rmax and dr=0 unique separations
+    (seeds 0/5/8 hit dr0).  64-seed sweep was clean on hardware (r02,
+    profiles/r02_fuzz_v5_0.log)."""
+    from tests.fuzz_sweep import run_case5
+    rel, modes_ok, cfg = run_case5(seed)
+    assert modes_ok, cfg
+    assert rel < 1e-9, (rel, cfg)
+
+
 # ---- deferred-x fused FFT + binning path (nbk_fft_x_bin_f64) ----------
 # The final x-axis FFT pass runs inside the binning kernel for auto
 # power on non-interlaced power-of-two meshes: the finished complex
