@@ -91,3 +91,51 @@ def test_oracle_odd_mesh_flat_shotnoise():
         # mode-count closure: Hermitian weights cover every mesh mode
         # inside the binned k-range exactly once
         assert int(r['modes'].sum()) > 0
+
+
+def test_redges_unique_properties():
+    """dr=0 unique-separation edges: strictly increasing, start at 0,
+    bracket their centers, and every center is a realized lattice
+    separation modulus (oracle redges_unique, mirroring the reference's
+    _find_unique_edges over RealField coords)."""
+    import numpy
+    from oracle.fftpower import redges_unique
+    from oracle.mesh import MeshGeometry, real_coords
+
+    for nmesh, box in [(16, 100.), (24, 64.), (27, 81.), (32, 250.)]:
+        geom = MeshGeometry(nmesh, box, dtype='f8')
+        rmax = 0.5 * box
+        edges, centers = redges_unique(geom, rmax)
+        assert edges[0] == 0
+        assert numpy.all(numpy.diff(edges) > 0)
+        assert numpy.all(numpy.diff(centers) > 0)
+        assert len(edges) == len(centers) + 1
+        assert numpy.all(edges[:-1] <= centers)
+        assert numpy.all(centers <= edges[1:])
+        # each center is (to tolerance) a realized |x| on the lattice
+        x = real_coords(geom)
+        r = numpy.sqrt(sum(xi ** 2 for xi in x)).ravel()
+        for c in centers:
+            assert numpy.abs(r - c).min() < 1e-8 * max(1.0, c), (
+                nmesh, box, c)
+
+
+def test_fftcorr_oracle_dr0_bins_cover_all_cells():
+    """dr=0 FFTCorr bins cover exactly the sub-rmax, mu>=0 half of the
+    configuration lattice: total modes == count of grid points with
+    r < edges[-1] and x.los >= 0 (FFTCorr's mu range is [0, 1] —
+    reference fftcorr.py:175 — so the mu<0 half-lattice falls outside
+    the mu edges and is dropped, unlike FFTPower's [-1, 1])."""
+    import numpy
+    from oracle import fftcorr_oracle
+    from oracle.mesh import MeshGeometry, real_coords
+
+    nmesh, box = 16, 100.
+    pos = numpy.random.RandomState(3).uniform(0, box, size=(5000, 3))
+    out = fftcorr_oracle(pos, Nmesh=nmesh, BoxSize=box, mode='1d',
+                         resampler='cic', compensated=True, dr=0)
+    geom = MeshGeometry(nmesh, box, dtype='f8')
+    X, Y, Z = numpy.broadcast_arrays(*real_coords(geom))
+    r = numpy.sqrt(X ** 2 + Y ** 2 + Z ** 2).ravel()
+    keep = (r < out['redges'][-1]) & (Z.ravel() >= 0)   # mu = z/r >= 0
+    assert int(out['modes'].sum()) == int(keep.sum())
